@@ -1,0 +1,38 @@
+// Framebuffer sources: synthetic pattern generator + X11/XShm capture.
+#pragma once
+
+#include <memory>
+#include <string>
+
+#include "include/hipflux/common.h"
+
+namespace hipflux {
+
+class FrameSource {
+ public:
+  virtual ~FrameSource() = default;
+  // Acquire the next framebuffer. The returned pointers stay valid until the
+  // next acquire() or destruction. Returns false on unrecoverable error.
+  virtual bool acquire(RawFrame& out) = 0;
+  virtual int width() const = 0;
+  virtual int height() const = 0;
+};
+
+// Synthetic BGRX generator for benches and GPU boxes without an X server
+// (the BASELINE measurement surface: "synthetic framebuffer with random
+// pixel content"). Patterns:
+//   "noise"   — every pixel re-randomized every frame (worst case: 100%
+//               damage, defeats any gating — used by bench.py)
+//   "desktop" — static random background + moving window + blinking cursor
+//               (exercises damage gating / paint-over)
+//   "static"  — one random frame, never changes (exercises paint-over)
+std::unique_ptr<FrameSource> make_synthetic_source(int width, int height,
+                                                   const std::string& pattern,
+                                                   uint64_t seed = 0x5eed);
+
+// X11 XShm capture of (x, y, w, h) from `display`; nullptr if the display
+// cannot be opened. w/h of 0 = full root window.
+std::unique_ptr<FrameSource> make_x11_source(const std::string& display,
+                                             int x, int y, int w, int h);
+
+}  // namespace hipflux

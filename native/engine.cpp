@@ -1,0 +1,216 @@
+#include "engine.h"
+
+#include <algorithm>
+#include <chrono>
+#include <cstdio>
+
+namespace hipflux {
+
+ScreenCapture::~ScreenCapture() { stop_capture(); }
+
+void ScreenCapture::start_capture(StripeCallback cb,
+                                  const CaptureSettings& settings) {
+  stop_capture();
+  settings_ = settings;
+  cb_ = std::move(cb);
+  fps_.store(settings.target_fps);
+  bitrate_kbps_.store(settings.video_bitrate_kbps);
+  crf_.store(settings.video_crf);
+  jpeg_quality_.store(settings.jpeg_quality);
+  vbv_mult_.store(settings.vbv_multiplier);
+  stop_.store(false);
+  running_.store(true);
+  thread_ = std::thread([this] { run(); });
+}
+
+void ScreenCapture::stop_capture() {
+  stop_.store(true);
+  if (thread_.joinable()) thread_.join();
+  running_.store(false);
+}
+
+void ScreenCapture::update_capture_region(int x, int y, int w, int h) {
+  std::lock_guard<std::mutex> lk(region_mutex_);
+  region_[0] = x; region_[1] = y; region_[2] = w; region_[3] = h;
+  region_changed_.store(true);
+}
+
+void pack_wire_stripe(const EncodedStripe& s, std::vector<uint8_t>& out) {
+  if (s.type == StripeType::kH264) {
+    // [0x04, keyflag, frame_id:u16be, y:u16be, w:u16be, h:u16be] (10 bytes)
+    out.push_back(0x04);
+    out.push_back(s.is_keyframe ? 1 : 0);
+    out.push_back(static_cast<uint8_t>((s.frame_id >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.frame_id & 0xFF));
+    out.push_back(static_cast<uint8_t>((s.y >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.y & 0xFF));
+    out.push_back(static_cast<uint8_t>((s.width >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.width & 0xFF));
+    out.push_back(static_cast<uint8_t>((s.height >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.height & 0xFF));
+  } else {
+    // [0x03, flags, frame_id:u16be, y:u16be] (6 bytes)
+    out.push_back(0x03);
+    out.push_back(s.is_keyframe ? 1 : 0);
+    out.push_back(static_cast<uint8_t>((s.frame_id >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.frame_id & 0xFF));
+    out.push_back(static_cast<uint8_t>((s.y >> 8) & 0xFF));
+    out.push_back(static_cast<uint8_t>(s.y & 0xFF));
+  }
+  out.insert(out.end(), s.data, s.data + s.size);
+}
+
+void ScreenCapture::run() {
+  using clock = std::chrono::steady_clock;
+
+  // --- frame source -------------------------------------------------------
+  std::unique_ptr<FrameSource> src;
+  std::string backend = settings_.capture_backend;
+  if (backend == "auto")
+    backend = settings_.display.empty() ? "synthetic" : "x11";
+  if (backend == "x11") {
+    src = make_x11_source(settings_.display, settings_.capture_x,
+                          settings_.capture_y, settings_.capture_width,
+                          settings_.capture_height);
+    if (!src) {
+      std::fprintf(stderr,
+                   "hipflux: cannot open X display '%s', falling back to "
+                   "synthetic source\n",
+                   settings_.display.c_str());
+    }
+  }
+  if (!src) {
+    // synthetic pattern override: "synthetic:<pattern>" in capture_backend
+    std::string pattern = "desktop";
+    auto pos = settings_.capture_backend.find(':');
+    if (pos != std::string::npos)
+      pattern = settings_.capture_backend.substr(pos + 1);
+    src = make_synthetic_source(settings_.capture_width,
+                                settings_.capture_height, pattern);
+  }
+
+  // --- pipeline ------------------------------------------------------------
+  std::unique_ptr<EncodePipeline> pipeline;
+  if (!settings_.use_cpu && settings_.gpu_id >= 0)
+    pipeline = make_hip_pipeline(settings_);
+  if (!pipeline) {
+    pipeline = settings_.output_mode == 0 ? make_cpu_jpeg_pipeline(settings_)
+                                          : make_cpu_h264_pipeline(settings_);
+  }
+  if (!pipeline) {
+    std::fprintf(stderr,
+                 "hipflux: requested pipeline unavailable, using cpu-jpeg\n");
+    pipeline = make_cpu_jpeg_pipeline(settings_);
+  }
+  pipeline_name_ = pipeline->name();
+
+  DamageTracker damage;
+  damage.reset(src->width(), src->height());
+
+  const int stripe_h = std::max(16, settings_.stripe_height & ~15);
+  uint32_t frame_id = 0;
+  int paintover_pending = 0;   // remaining paint-over burst frames
+  bool paintover_done = false; // paint-over completed for current still period
+  double keyframe_due_ms = 0;  // next forced keyframe time (0 = disabled)
+  if (settings_.keyframe_interval_s > 0)
+    keyframe_due_ms = now_ms() + settings_.keyframe_interval_s * 1000.0;
+
+  std::vector<uint8_t> wire;
+  auto next_tick = clock::now();
+
+  while (!stop_.load()) {
+    double fps = std::max(1.0, fps_.load());
+    next_tick += std::chrono::microseconds(static_cast<int64_t>(1e6 / fps));
+
+    RawFrame frame;
+    if (!src->acquire(frame)) break;
+    frames_captured_.fetch_add(1);
+
+    // damage update (skipped in fullframe mode to save the diff cost)
+    bool force_all = settings_.video_fullframe;
+    if (!force_all)
+      damage.update(frame.data, frame.stride, settings_.damage_block_threshold,
+                    settings_.damage_block_duration);
+
+    bool idr = idr_requested_.exchange(false);
+    double tnow = now_ms();
+    if (keyframe_due_ms > 0 && tnow >= keyframe_due_ms) {
+      idr = true;
+      keyframe_due_ms = tnow + settings_.keyframe_interval_s * 1000.0;
+    }
+
+    // paint-over state machine
+    bool paintover_frame = false;
+    if (!force_all && settings_.use_paint_over_quality) {
+      if (damage.consecutive_still_frames() == 0) {
+        paintover_done = false;
+        paintover_pending = 0;
+      } else if (!paintover_done &&
+                 damage.consecutive_still_frames() >=
+                     settings_.paint_over_trigger_frames) {
+        paintover_pending = settings_.output_mode == 0
+                                ? 1
+                                : settings_.video_paintover_burst_frames;
+        paintover_done = true;
+      }
+      if (paintover_pending > 0) {
+        paintover_frame = true;
+        --paintover_pending;
+      }
+    }
+
+    // build stripe jobs
+    FrameContext ctx;
+    ctx.frame_id = frame_id;
+    ctx.idr = idr;
+    ctx.paintover = paintover_frame;
+    ctx.jpeg_quality = paintover_frame ? settings_.jpeg_paintover_quality
+                                       : jpeg_quality_.load();
+    ctx.crf = paintover_frame ? settings_.video_paintover_crf : crf_.load();
+    ctx.bitrate_kbps = bitrate_kbps_.load();
+    bool any = false;
+    for (int y = 0; y < frame.height; y += stripe_h) {
+      StripeJob job;
+      job.y0 = y;
+      job.y1 = std::min(y + stripe_h, frame.height);
+      job.encode = force_all || idr || paintover_frame ||
+                   damage.stripe_damaged(job.y0, job.y1);
+      job.paintover = paintover_frame;
+      any |= job.encode;
+      ctx.stripes.push_back(job);
+    }
+
+    if (any) {
+      double t0 = now_ms();
+      pipeline->encode_frame(frame, ctx, [&](EncodedStripe& s) {
+        s.capture_ts_ms = frame.ts_ms;
+        s.encode_done_ms = now_ms();
+        if (settings_.omit_stripe_headers) {
+          if (cb_) cb_(s);
+        } else {
+          wire.clear();
+          pack_wire_stripe(s, wire);
+          EncodedStripe ws = s;
+          ws.data = wire.data();
+          ws.size = wire.size();
+          if (cb_) cb_(ws);
+        }
+        stripes_emitted_.fetch_add(1);
+      });
+      last_encode_ms_.store(now_ms() - t0);
+      frames_encoded_.fetch_add(1);
+      ++frame_id;
+    }
+
+    // pacing (skip sleeping if we're behind)
+    auto now = clock::now();
+    if (next_tick > now) {
+      std::this_thread::sleep_until(next_tick);
+    } else {
+      next_tick = now;
+    }
+  }
+  running_.store(false);
+}
+
+}  // namespace hipflux
