@@ -255,6 +255,13 @@ inline void rgb_to_ycbcr(float r, float g, float b, float& y, float& cb, float& 
   cr = 0.5f * r - 0.418688f * g - 0.081312f * b + 128.f;
 }
 
+// Round to the 8-bit sample grid — matches the HIP path, which stores
+// Y/Cb/Cr planes as uint8 before the DCT (so CPU and GPU bitstreams agree).
+inline float to_u8_sample(float v) {
+  v = std::lrintf(std::min(std::max(v, 0.f), 255.f));
+  return v;
+}
+
 }  // namespace
 
 void jpeg_quality_tables(int quality, uint8_t qy[64], uint8_t qc[64]) {
@@ -297,6 +304,7 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
             sample(mx * 16 + xx, my * 16 + yy, r, g, b);
             rgb_to_ycbcr(r, g, b, Y[yy * 16 + xx], Cb[yy * 16 + xx],
                          Cr[yy * 16 + xx]);
+            Y[yy * 16 + xx] = to_u8_sample(Y[yy * 16 + xx]);
           }
         // 4 luma blocks
         for (int by = 0; by < 2; ++by)
@@ -317,7 +325,7 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
                         src[(2 * yy) * 16 + 2 * xx + 1] +
                         src[(2 * yy + 1) * 16 + 2 * xx] +
                         src[(2 * yy + 1) * 16 + 2 * xx + 1];
-              plane[yy * 8 + xx] = s * 0.25f - 128.f;
+              plane[yy * 8 + xx] = to_u8_sample(s * 0.25f) - 128.f;
             }
           fdct8x8_quant(plane, qc, blk);
           if (c == 0)
@@ -337,9 +345,9 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
             float r, g, b, y_, cb, cr;
             sample(mx * 8 + xx, my * 8 + yy, r, g, b);
             rgb_to_ycbcr(r, g, b, y_, cb, cr);
-            plane[0][yy * 8 + xx] = y_ - 128.f;
-            plane[1][yy * 8 + xx] = cb - 128.f;
-            plane[2][yy * 8 + xx] = cr - 128.f;
+            plane[0][yy * 8 + xx] = to_u8_sample(y_) - 128.f;
+            plane[1][yy * 8 + xx] = to_u8_sample(cb) - 128.f;
+            plane[2][yy * 8 + xx] = to_u8_sample(cr) - 128.f;
           }
         fdct8x8_quant(plane[0], qy, blk);
         dcY = encode_block(bw, blk, dcY, dc_luma(), ac_luma());
