@@ -19,7 +19,9 @@ from dataclasses import dataclass
 try:
     from hipflux._native import (  # noqa: F401
         CaptureSettings,
+        H264Encoder,
         ScreenCapture,
+        bgrx_to_yuv420,
         hip_device_count,
         jpeg_encode,
     )

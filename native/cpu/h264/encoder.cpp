@@ -1,0 +1,684 @@
+#include "encoder.h"
+
+#include <algorithm>
+#include <cassert>
+#include <cmath>
+#include <cstring>
+
+#include "bitwriter.h"
+#include "cavlc.h"
+#include "transform.h"
+
+namespace hipflux {
+namespace h264 {
+namespace {
+
+inline uint8_t clip8(int v) {
+  return static_cast<uint8_t>(v < 0 ? 0 : (v > 255 ? 255 : v));
+}
+
+// blkIdx (Z-order) -> 4x4 block coords within MB
+inline void blk_xy(int blk, int& bx, int& by) {
+  bx = 2 * ((blk >> 2) & 1) + (blk & 1);
+  by = 2 * (blk >> 3) + ((blk >> 1) & 1);
+}
+
+struct Plane {
+  std::vector<uint8_t> data;
+  int pitch = 0;
+  void alloc(int w, int h) {
+    pitch = w;
+    data.assign(static_cast<size_t>(w) * h, 0);
+  }
+  uint8_t* row(int y) { return data.data() + static_cast<size_t>(y) * pitch; }
+  const uint8_t* row(int y) const {
+    return data.data() + static_cast<size_t>(y) * pitch;
+  }
+};
+
+struct Frame {
+  Plane y, cb, cr;
+  void alloc(int yw, int yh) {
+    y.alloc(yw, yh);
+    cb.alloc(yw / 2, yh / 2);
+    cr.alloc(yw / 2, yh / 2);
+  }
+};
+
+// ---- per-row entropy/pred context ----------------------------------------
+struct RowCtx {
+  bool have_left = false;
+  bool left_is_inter = false;
+  int left_mvx = 0, left_mvy = 0;  // quarter-pel
+  uint8_t left_luma_nc[4] = {0, 0, 0, 0};   // totals of left MB col-3 blocks
+  uint8_t left_cb_nc[2] = {0, 0}, left_cr_nc[2] = {0, 0};
+  int skip_run = 0;
+};
+
+struct MbTotals {
+  uint8_t luma[16] = {};
+  uint8_t cb[4] = {}, cr[4] = {};
+};
+
+}  // namespace
+
+struct StripeEncoder::Impl {
+  int w, h, mbw, mbh, yw, yh;
+  Frame ref, cur;        // reference (prev recon) and current recon
+  Frame src;             // padded source planes
+  uint32_t frame_num = 0;
+  uint32_t idr_pic_id = 0;
+  bool need_idr = true;
+  int level_idc = 0;
+
+  Impl(int width, int height) : w(width), h(height) {
+    mbw = (w + 15) / 16;
+    mbh = (h + 15) / 16;
+    yw = mbw * 16;
+    yh = mbh * 16;
+    ref.alloc(yw, yh);
+    cur.alloc(yw, yh);
+    src.alloc(yw, yh);
+    // level from MB count (MaxFS) and 60 fps MaxMBPS, generous
+    int fs = mbw * mbh;
+    if (fs <= 1620) level_idc = 31;        // <= 480p
+    else if (fs <= 3600) level_idc = 32;
+    else if (fs <= 8192) level_idc = 42;   // 1080p60
+    else if (fs <= 22080) level_idc = 51;  // 4K
+    else level_idc = 52;
+  }
+
+  // ---- headers ------------------------------------------------------------
+  void write_sps(std::vector<uint8_t>& out) {
+    BitWriter b;
+    b.u(66, 8);   // profile_idc: Baseline
+    b.u(1, 1);    // constraint_set0
+    b.u(1, 1);    // constraint_set1 (constrained baseline)
+    b.u(0, 6);    // constraint_set2..5 + reserved
+    b.u(level_idc, 8);
+    b.ue(0);      // sps id
+    b.ue(12);     // log2_max_frame_num_minus4 -> 16-bit frame_num
+    b.ue(2);      // pic_order_cnt_type = 2
+    b.ue(1);      // max_num_ref_frames
+    b.u(0, 1);    // gaps_in_frame_num_value_allowed
+    b.ue(mbw - 1);
+    b.ue(mbh - 1);
+    b.u(1, 1);    // frame_mbs_only
+    b.u(1, 1);    // direct_8x8_inference
+    int crop_r = (yw - w) / 2, crop_b = (yh - h) / 2;
+    if (crop_r || crop_b) {
+      b.u(1, 1);
+      b.ue(0);
+      b.ue(crop_r);
+      b.ue(0);
+      b.ue(crop_b);
+    } else {
+      b.u(0, 1);
+    }
+    // VUI: declare BT.601 full-range to match our CSC
+    b.u(1, 1);    // vui_parameters_present
+    b.u(0, 1);    // aspect_ratio_info_present
+    b.u(0, 1);    // overscan_info_present
+    b.u(1, 1);    // video_signal_type_present
+    b.u(5, 3);    // video_format: unspecified
+    b.u(1, 1);    // video_full_range_flag
+    b.u(1, 1);    // colour_description_present
+    b.u(6, 8);    // colour_primaries: SMPTE 170M
+    b.u(6, 8);    // transfer_characteristics
+    b.u(6, 8);    // matrix_coefficients: BT.601
+    b.u(0, 1);    // chroma_loc_info_present
+    b.u(0, 1);    // timing_info_present
+    b.u(0, 1);    // nal_hrd_parameters_present
+    b.u(0, 1);    // vcl_hrd_parameters_present
+    b.u(0, 1);    // pic_struct_present
+    b.u(0, 1);    // bitstream_restriction
+    b.rbsp_trailing();
+    b.emit_nal(out, 3, 7);
+  }
+
+  void write_pps(std::vector<uint8_t>& out) {
+    BitWriter b;
+    b.ue(0);      // pps id
+    b.ue(0);      // sps id
+    b.u(0, 1);    // entropy_coding_mode: CAVLC
+    b.u(0, 1);    // bottom_field_pic_order_in_frame_present
+    b.ue(0);      // num_slice_groups_minus1
+    b.ue(0);      // num_ref_idx_l0_default_active_minus1
+    b.ue(0);      // num_ref_idx_l1_default_active_minus1
+    b.u(0, 1);    // weighted_pred
+    b.u(0, 2);    // weighted_bipred_idc
+    b.se(0);      // pic_init_qp_minus26
+    b.se(0);      // pic_init_qs_minus26
+    b.se(0);      // chroma_qp_index_offset
+    b.u(1, 1);    // deblocking_filter_control_present
+    b.u(0, 1);    // constrained_intra_pred
+    b.u(0, 1);    // redundant_pic_cnt_present
+    b.rbsp_trailing();
+    b.emit_nal(out, 3, 8);
+  }
+
+  void write_slice_header(BitWriter& b, bool idr, int mb_row, int qp) {
+    b.ue(mb_row * mbw);          // first_mb_in_slice
+    b.ue(idr ? 7 : 5);           // slice_type: I / P (all-slices-same form)
+    b.ue(0);                     // pps id
+    b.u(frame_num & 0xFFFF, 16); // frame_num
+    if (idr) b.ue(idr_pic_id);
+    if (!idr) {
+      b.u(0, 1);                 // num_ref_idx_active_override
+      b.u(0, 1);                 // ref_pic_list_modification_flag_l0
+    }
+    // dec_ref_pic_marking (nal_ref_idc != 0)
+    if (idr) {
+      b.u(0, 1);                 // no_output_of_prior_pics
+      b.u(0, 1);                 // long_term_reference
+    } else {
+      b.u(0, 1);                 // adaptive_ref_pic_marking_mode
+    }
+    b.se(qp - 26);               // slice_qp_delta
+    b.ue(1);                     // disable_deblocking_filter_idc = 1
+  }
+
+  // ---- intra 16x16 macroblock ---------------------------------------------
+  // Encodes src MB at (mbx, mby) as I16x16 into bw; reconstructs into cur.
+  // Returns totals for nC bookkeeping. p_slice: emit P-slice mb_type offset.
+  void encode_i16(BitWriter& bw, int mbx, int mby, int qp, bool p_slice,
+                  RowCtx& ctx, MbTotals& tot) {
+    const int x0 = mbx * 16, y0 = mby * 16;
+    const int cx0 = mbx * 8, cy0 = mby * 8;
+    const bool left = ctx.have_left;
+
+    // ----- luma prediction: H (mode 1) vs DC (mode 2)
+    uint8_t predH[256], predDC[256];
+    int dc = 128;
+    if (left) {
+      int s = 0;
+      for (int r = 0; r < 16; ++r) s += cur.y.row(y0 + r)[x0 - 1];
+      dc = (s + 8) >> 4;
+    }
+    std::memset(predDC, dc, 256);
+    long costH = 1 << 30, costDC = 0;
+    if (left) {
+      costH = 0;
+      for (int r = 0; r < 16; ++r) {
+        uint8_t v = cur.y.row(y0 + r)[x0 - 1];
+        std::memset(predH + 16 * r, v, 16);
+      }
+      for (int r = 0; r < 16; ++r) {
+        const uint8_t* s = src.y.row(y0 + r) + x0;
+        for (int c = 0; c < 16; ++c) {
+          costH += std::abs(int(s[c]) - int(predH[16 * r + c]));
+        }
+      }
+    }
+    for (int r = 0; r < 16; ++r) {
+      const uint8_t* s = src.y.row(y0 + r) + x0;
+      for (int c = 0; c < 16; ++c) costDC += std::abs(int(s[c]) - dc);
+    }
+    const int luma_mode = (left && costH < costDC) ? 1 : 2;
+    const uint8_t* pred = luma_mode == 1 ? predH : predDC;
+
+    // ----- luma transform: 16 blocks, DC Hadamard path
+    int coef[16][16];  // raster coeffs per block
+    int dcs[16];       // dc per block, raster (by*4+bx)
+    for (int by = 0; by < 4; ++by)
+      for (int bx = 0; bx < 4; ++bx) {
+        int resid[16];
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = by * 4 + r, xx = bx * 4 + c;
+            resid[4 * r + c] =
+                int(src.y.row(y0 + yy)[x0 + xx]) - int(pred[16 * yy + xx]);
+          }
+        fdct4x4(resid, coef[by * 4 + bx]);
+        dcs[by * 4 + bx] = coef[by * 4 + bx][0];
+      }
+    int had[16];
+    hadamard4x4_fwd(dcs, had);
+    int qdc_r[16];  // quantized DC, raster
+    for (int i = 0; i < 16; ++i) qdc_r[i] = quant_dc(had[i], qp, true);
+    int zz_dc[16];
+    for (int i = 0; i < 16; ++i) zz_dc[i] = qdc_r[kZigzag4[i]];
+    cap_coeffs(zz_dc, 16);
+    for (int i = 0; i < 16; ++i) qdc_r[kZigzag4[i]] = zz_dc[i];
+
+    int zz_ac[16][15];
+    bool any_ac = false;
+    for (int b = 0; b < 16; ++b) {
+      for (int i = 1; i < 16; ++i) {
+        int pos = kZigzag4[i];
+        zz_ac[b][i - 1] = quant_coeff(coef[b][pos], qp,
+                                      coeff_class(pos >> 2, pos & 3), true);
+      }
+      cap_coeffs(zz_ac[b], 15);
+      for (int i = 0; i < 15; ++i) any_ac |= zz_ac[b][i] != 0;
+    }
+    const int cbp_luma = any_ac ? 15 : 0;
+
+    // ----- luma reconstruction
+    int ihad[16], dcrec[16];
+    hadamard4x4_inv(qdc_r, ihad);
+    for (int i = 0; i < 16; ++i) dcrec[i] = dequant_luma_dc(ihad[i], qp);
+    for (int by = 0; by < 4; ++by)
+      for (int bx = 0; bx < 4; ++bx) {
+        int b = by * 4 + bx;
+        int dq[16] = {};
+        dq[0] = dcrec[b];
+        if (cbp_luma) {
+          for (int i = 1; i < 16; ++i) {
+            int pos = kZigzag4[i];
+            dq[pos] = dequant_coeff(zz_ac[b][i - 1], qp,
+                                    coeff_class(pos >> 2, pos & 3));
+          }
+        }
+        int rec[16];
+        idct4x4(dq, rec);
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = by * 4 + r, xx = bx * 4 + c;
+            cur.y.row(y0 + yy)[x0 + xx] =
+                clip8(rec[4 * r + c] + pred[16 * yy + xx]);
+          }
+      }
+
+    // ----- chroma: mode DC (0) vs H (1); shared by Cb/Cr
+    const int qpc = chroma_qp(qp);
+    uint8_t cpred[2][64];
+    long ccostH = 1 << 30, ccostDC = 0;
+    uint8_t cpredH[2][64], cpredDC[2][64];
+    for (int comp = 0; comp < 2; ++comp) {
+      Plane& rc = comp ? cur.cr : cur.cb;
+      Plane& sp = comp ? src.cr : src.cb;
+      for (int sub = 0; sub < 4; ++sub) {
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int d = 128;
+        if (left) {
+          int s = 0;
+          for (int r = 0; r < 4; ++r) s += rc.row(cy0 + scy + r)[cx0 - 1];
+          d = (s + 2) >> 2;
+        }
+        for (int r = 0; r < 4; ++r)
+          std::memset(cpredDC[comp] + 8 * (scy + r) + scx, d, 4);
+      }
+      if (left)
+        for (int r = 0; r < 8; ++r)
+          std::memset(cpredH[comp] + 8 * r, rc.row(cy0 + r)[cx0 - 1], 8);
+      for (int r = 0; r < 8; ++r)
+        for (int c = 0; c < 8; ++c) {
+          int s = sp.row(cy0 + r)[cx0 + c];
+          ccostDC += std::abs(s - int(cpredDC[comp][8 * r + c]));
+          if (left) {
+            if (comp == 0 && r == 0 && c == 0) ccostH = 0;
+            ccostH += std::abs(s - int(cpredH[comp][8 * r + c]));
+          }
+        }
+    }
+    const int chroma_mode = (left && ccostH < ccostDC) ? 1 : 0;
+    std::memcpy(cpred[0], chroma_mode ? cpredH[0] : cpredDC[0], 64);
+    std::memcpy(cpred[1], chroma_mode ? cpredH[1] : cpredDC[1], 64);
+
+    int ccoef[2][4][16], cdc[2][4];
+    int czz_ac[2][4][15];
+    bool c_any_ac = false;
+    for (int comp = 0; comp < 2; ++comp) {
+      Plane& sp = comp ? src.cr : src.cb;
+      for (int sub = 0; sub < 4; ++sub) {
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int resid[16];
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c)
+            resid[4 * r + c] = int(sp.row(cy0 + scy + r)[cx0 + scx + c]) -
+                               int(cpred[comp][8 * (scy + r) + scx + c]);
+        fdct4x4(resid, ccoef[comp][sub]);
+        cdc[comp][sub] = ccoef[comp][sub][0];
+        for (int i = 1; i < 16; ++i) {
+          int pos = kZigzag4[i];
+          czz_ac[comp][sub][i - 1] = quant_coeff(
+              ccoef[comp][sub][pos], qpc, coeff_class(pos >> 2, pos & 3), true);
+        }
+        cap_coeffs(czz_ac[comp][sub], 15);
+        for (int i = 0; i < 15; ++i) c_any_ac |= czz_ac[comp][sub][i] != 0;
+      }
+    }
+    // 2x2 Hadamard + quant of chroma DC
+    int cqdc[2][4];
+    bool c_any_dc = false;
+    for (int comp = 0; comp < 2; ++comp) {
+      int* d = cdc[comp];
+      int w0 = d[0] + d[1] + d[2] + d[3];
+      int w1 = d[0] - d[1] + d[2] - d[3];
+      int w2 = d[0] + d[1] - d[2] - d[3];
+      int w3 = d[0] - d[1] - d[2] + d[3];
+      cqdc[comp][0] = quant_dc(w0, qpc, true);
+      cqdc[comp][1] = quant_dc(w1, qpc, true);
+      cqdc[comp][2] = quant_dc(w2, qpc, true);
+      cqdc[comp][3] = quant_dc(w3, qpc, true);
+      for (int i = 0; i < 4; ++i) c_any_dc |= cqdc[comp][i] != 0;
+    }
+    const int cbp_chroma = c_any_ac ? 2 : (c_any_dc ? 1 : 0);
+
+    // ----- chroma recon
+    for (int comp = 0; comp < 2; ++comp) {
+      Plane& rc = comp ? cur.cr : cur.cb;
+      int dcq[4] = {0, 0, 0, 0};
+      if (cbp_chroma >= 1) {
+        int* q = cqdc[comp];
+        int w0 = q[0] + q[1] + q[2] + q[3];
+        int w1 = q[0] - q[1] + q[2] - q[3];
+        int w2 = q[0] + q[1] - q[2] - q[3];
+        int w3 = q[0] - q[1] - q[2] + q[3];
+        dcq[0] = dequant_chroma_dc(w0, qpc);
+        dcq[1] = dequant_chroma_dc(w1, qpc);
+        dcq[2] = dequant_chroma_dc(w2, qpc);
+        dcq[3] = dequant_chroma_dc(w3, qpc);
+      }
+      for (int sub = 0; sub < 4; ++sub) {
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int dq[16] = {};
+        dq[0] = dcq[sub];
+        if (cbp_chroma == 2)
+          for (int i = 1; i < 16; ++i) {
+            int pos = kZigzag4[i];
+            dq[pos] = dequant_coeff(czz_ac[comp][sub][i - 1], qpc,
+                                    coeff_class(pos >> 2, pos & 3));
+          }
+        int rec[16];
+        idct4x4(dq, rec);
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c)
+            rc.row(cy0 + scy + r)[cx0 + scx + c] = clip8(
+                rec[4 * r + c] + cpred[comp][8 * (scy + r) + scx + c]);
+      }
+    }
+
+    // ----- entropy
+    // I-slice mb_type for I16x16: 1 + predMode + 4*cbp_chroma + 12*cbp_luma01
+    // (predMode = Intra16x16PredMode: ours is 1=H or 2=DC).
+    // In P slices intra mb_types are offset by 5.
+    const int i16_type =
+        1 + luma_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0);
+    bw.ue(p_slice ? 5 + i16_type : i16_type);
+    bw.ue(chroma_mode);   // intra_chroma_pred_mode
+    bw.se(0);             // mb_qp_delta
+    // DC block: nC from neighboring 4x4 blk(0,0)
+    int nC_dc = blk_nc(ctx, tot, 0, true);
+    int tc_dc = cavlc_residual(bw, zz_dc, 16, nC_dc);
+    (void)tc_dc;
+    MbTotals newtot;
+    if (cbp_luma) {
+      // bitstream block order is Z-order; zz_ac is stored raster (by*4+bx)
+      for (int blk = 0; blk < 16; ++blk) {
+        int bx, by;
+        blk_xy(blk, bx, by);
+        int r = by * 4 + bx;
+        int nC = luma_nc_for(ctx, newtot, bx, by);
+        int tc = cavlc_residual(bw, zz_ac[r], 15, nC);
+        newtot.luma[r] = static_cast<uint8_t>(tc);
+      }
+    }
+    if (cbp_chroma > 0) {
+      int zz[4];
+      for (int comp = 0; comp < 2; ++comp) {
+        for (int i = 0; i < 4; ++i) zz[i] = cqdc[comp][i];
+        cavlc_residual(bw, zz, 4, -1);
+      }
+    }
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp) {
+        uint8_t* t = comp ? newtot.cr : newtot.cb;
+        for (int sub = 0; sub < 4; ++sub) {
+          int cx = sub & 1, cy = sub >> 1;
+          int nC = chroma_nc_for(ctx, t, comp, cx, cy);
+          int tc = cavlc_residual(bw, czz_ac[comp][sub], 15, nC);
+          t[cy * 2 + cx] = static_cast<uint8_t>(tc);
+        }
+      }
+    }
+    tot = newtot;
+  }
+
+  // nC for luma blk (bx,by): top row always unavailable (slice-per-row);
+  // left within MB, or col-3 of left MB.
+  int luma_nc_for(const RowCtx& ctx, const MbTotals& cur_tot, int bx,
+                  int by) const {
+    if (bx > 0) return cur_tot.luma[by * 4 + (bx - 1)];
+    if (ctx.have_left) return ctx.left_luma_nc[by];
+    return 0;
+  }
+  int blk_nc(const RowCtx& ctx, const MbTotals& cur_tot, int blk,
+             bool /*dc*/) const {
+    int bx, by;
+    blk_xy(blk, bx, by);
+    return luma_nc_for(ctx, cur_tot, bx, by);
+  }
+  int chroma_nc_for(const RowCtx& ctx, const uint8_t* t, int comp, int cx,
+                    int cy) const {
+    if (cx > 0) return t[cy * 2];
+    if (ctx.have_left)
+      return comp ? ctx.left_cr_nc[cy] : ctx.left_cb_nc[cy];
+    return 0;
+  }
+
+  // ---- P macroblock helpers -----------------------------------------------
+  long sad16(const uint8_t* a, int ap, const uint8_t* b, int bp) const {
+    long s = 0;
+    for (int r = 0; r < 16; ++r) {
+      const uint8_t* pa = a + static_cast<size_t>(r) * ap;
+      const uint8_t* pb = b + static_cast<size_t>(r) * bp;
+      for (int c = 0; c < 16; ++c) s += std::abs(int(pa[c]) - int(pb[c]));
+    }
+    return s;
+  }
+
+  void copy_mb_from_ref(int mbx, int mby, int mvx, int mvy) {
+    int x0 = mbx * 16, y0 = mby * 16;
+    for (int r = 0; r < 16; ++r)
+      std::memcpy(cur.y.row(y0 + r) + x0,
+                  ref.y.row(y0 + r + mvy) + x0 + mvx, 16);
+    int cx0 = mbx * 8, cy0 = mby * 8, cmx = mvx / 2, cmy = mvy / 2;
+    for (int r = 0; r < 8; ++r) {
+      std::memcpy(cur.cb.row(cy0 + r) + cx0,
+                  ref.cb.row(cy0 + r + cmy) + cx0 + cmx, 8);
+      std::memcpy(cur.cr.row(cy0 + r) + cx0,
+                  ref.cr.row(cy0 + r + cmy) + cx0 + cmx, 8);
+    }
+  }
+
+  // ---- frame encode --------------------------------------------------------
+  void encode(const uint8_t* sy, int syp, const uint8_t* scb,
+              const uint8_t* scr, int scp, int qp, bool force_idr,
+              std::vector<uint8_t>& out, EncodeStats* stats) {
+    qp = std::clamp(qp, 0, 51);
+    // pad source into aligned planes
+    for (int r = 0; r < yh; ++r) {
+      int sr = std::min(r, h - 1);
+      std::memcpy(src.y.row(r), sy + static_cast<size_t>(sr) * syp, w);
+      uint8_t e = src.y.row(r)[w - 1];
+      std::memset(src.y.row(r) + w, e, yw - w);
+    }
+    int cw = (w + 1) / 2, ch = (h + 1) / 2, cwp = yw / 2, chp = yh / 2;
+    for (int r = 0; r < chp; ++r) {
+      int sr = std::min(r, ch - 1);
+      std::memcpy(src.cb.row(r), scb + static_cast<size_t>(sr) * scp, cw);
+      std::memcpy(src.cr.row(r), scr + static_cast<size_t>(sr) * scp, cw);
+      std::memset(src.cb.row(r) + cw, src.cb.row(r)[cw - 1], cwp - cw);
+      std::memset(src.cr.row(r) + cw, src.cr.row(r)[cw - 1], cwp - cw);
+    }
+
+    bool idr = force_idr || need_idr;
+    if (idr) {
+      frame_num = 0;
+      write_sps(out);
+      write_pps(out);
+      ++idr_pic_id;
+      need_idr = false;
+    }
+    if (stats) {
+      *stats = EncodeStats{};
+      stats->frame_qp = qp;
+      stats->is_idr = idr;
+    }
+
+    const long skip_thresh = 48L << (qp / 6);       // tuned-for-screen default
+    const long inter_thresh = 2 * skip_thresh;
+
+    for (int mb_row = 0; mb_row < mbh; ++mb_row) {
+      BitWriter b;
+      write_slice_header(b, idr, mb_row, qp);
+      RowCtx ctx;
+      MbTotals tot;
+      for (int mbx = 0; mbx < mbw; ++mbx) {
+        if (idr) {
+          encode_i16(b, mbx, mb_row, qp, false, ctx, tot);
+          ctx.have_left = true;
+          ctx.left_is_inter = false;
+          for (int by = 0; by < 4; ++by)
+            ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
+          for (int cy = 0; cy < 2; ++cy) {
+            ctx.left_cb_nc[cy] = tot.cb[cy * 2 + 1];
+            ctx.left_cr_nc[cy] = tot.cr[cy * 2 + 1];
+          }
+          if (stats) ++stats->mb_intra;
+          continue;
+        }
+        // ---- P slice decisions
+        int x0 = mbx * 16, y0 = mb_row * 16;
+        long sad0 = sad16(src.y.row(y0) + x0, src.y.pitch,
+                          ref.y.row(y0) + x0, ref.y.pitch);
+        if (sad0 <= skip_thresh) {
+          // P_Skip (skip MV is always (0,0) here: top neighbor is in
+          // another slice, which forces the zero-MV rule)
+          copy_mb_from_ref(mbx, mb_row, 0, 0);
+          ++ctx.skip_run;
+          ctx.have_left = true;
+          ctx.left_is_inter = true;
+          ctx.left_mvx = 0;
+          ctx.left_mvy = 0;
+          std::memset(ctx.left_luma_nc, 0, 4);
+          std::memset(ctx.left_cb_nc, 0, 2);
+          std::memset(ctx.left_cr_nc, 0, 2);
+          if (stats) ++stats->mb_skip;
+          continue;
+        }
+        // small even-step diamond search around (0,0), clamped to frame
+        int best_mvx = 0, best_mvy = 0;
+        long best = sad0;
+        static const int pat[8][2] = {{-2, 0}, {2, 0},  {0, -2}, {0, 2},
+                                      {-2, -2}, {2, 2}, {-2, 2}, {2, -2}};
+        for (int iter = 0; iter < 8; ++iter) {
+          int bmx = best_mvx, bmy = best_mvy;
+          bool improved = false;
+          for (auto& p : pat) {
+            int mx = bmx + p[0], my = bmy + p[1];
+            if (x0 + mx < 0 || y0 + my < 0 || x0 + mx + 16 > yw ||
+                y0 + my + 16 > yh || std::abs(mx) > 16 || std::abs(my) > 16)
+              continue;
+            long s = sad16(src.y.row(y0) + x0, src.y.pitch,
+                           ref.y.row(y0 + my) + x0 + mx, ref.y.pitch);
+            if (s < best) {
+              best = s;
+              best_mvx = mx;
+              best_mvy = my;
+              improved = true;
+            }
+          }
+          if (!improved) break;
+        }
+        if (best <= inter_thresh && !(best_mvx == 0 && best_mvy == 0)) {
+          // P_L0_16x16, zero residual
+          flush_skip_run(b, ctx);
+          b.ue(0);  // mb_type P_L0_16x16
+          int mvpx = ctx.have_left && ctx.left_is_inter ? ctx.left_mvx : 0;
+          int mvpy = ctx.have_left && ctx.left_is_inter ? ctx.left_mvy : 0;
+          b.se(best_mvx * 4 - mvpx);
+          b.se(best_mvy * 4 - mvpy);
+          b.ue(0);  // coded_block_pattern = 0 (inter me(v): codeNum 0)
+          copy_mb_from_ref(mbx, mb_row, best_mvx, best_mvy);
+          ctx.have_left = true;
+          ctx.left_is_inter = true;
+          ctx.left_mvx = best_mvx * 4;
+          ctx.left_mvy = best_mvy * 4;
+          std::memset(ctx.left_luma_nc, 0, 4);
+          std::memset(ctx.left_cb_nc, 0, 2);
+          std::memset(ctx.left_cr_nc, 0, 2);
+          if (stats) ++stats->mb_inter;
+        } else {
+          flush_skip_run(b, ctx);
+          encode_i16(b, mbx, mb_row, qp, true, ctx, tot);
+          ctx.have_left = true;
+          ctx.left_is_inter = false;
+          for (int by = 0; by < 4; ++by)
+            ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
+          for (int cy = 0; cy < 2; ++cy) {
+            ctx.left_cb_nc[cy] = tot.cb[cy * 2 + 1];
+            ctx.left_cr_nc[cy] = tot.cr[cy * 2 + 1];
+          }
+          if (stats) ++stats->mb_intra;
+        }
+      }
+      if (!idr && ctx.skip_run > 0) {
+        b.ue(ctx.skip_run);  // trailing skip run
+        ctx.skip_run = 0;
+      }
+      b.rbsp_trailing();
+      b.emit_nal(out, idr ? 3 : 2, idr ? 5 : 1, mb_row == 0);
+    }
+
+    std::swap(ref, cur);
+    ++frame_num;
+    if (stats) stats->bytes = out.size();
+  }
+
+  void flush_skip_run(BitWriter& b, RowCtx& ctx) {
+    b.ue(ctx.skip_run);
+    ctx.skip_run = 0;
+  }
+};
+
+StripeEncoder::StripeEncoder(int width, int height)
+    : impl_(new Impl(width, height)), width_(width), height_(height) {}
+StripeEncoder::~StripeEncoder() = default;
+
+void StripeEncoder::encode_frame(const uint8_t* y, int ypitch,
+                                 const uint8_t* cb, const uint8_t* cr,
+                                 int cpitch, int qp, bool force_idr,
+                                 std::vector<uint8_t>& out,
+                                 EncodeStats* stats) {
+  impl_->encode(y, ypitch, cb, cr, cpitch, qp, force_idr, out, stats);
+}
+
+const uint8_t* StripeEncoder::recon_y() const { return impl_->ref.y.data.data(); }
+const uint8_t* StripeEncoder::recon_cb() const { return impl_->ref.cb.data.data(); }
+const uint8_t* StripeEncoder::recon_cr() const { return impl_->ref.cr.data.data(); }
+int StripeEncoder::recon_ypitch() const { return impl_->ref.y.pitch; }
+int StripeEncoder::recon_cpitch() const { return impl_->ref.cb.pitch; }
+
+void bgrx_to_yuv420(const uint8_t* bgrx, int stride, int width, int height,
+                    uint8_t* y, int ypitch, uint8_t* cb, uint8_t* cr,
+                    int cpitch) {
+  auto clampf = [](float v) {
+    return static_cast<uint8_t>(std::lrintf(std::min(std::max(v, 0.f), 255.f)));
+  };
+  int cw = (width + 1) / 2, chh = (height + 1) / 2;
+  for (int cy = 0; cy < chh; ++cy)
+    for (int cx = 0; cx < cw; ++cx) {
+      float cbs = 0, crs = 0;
+      for (int dy = 0; dy < 2; ++dy)
+        for (int dx = 0; dx < 2; ++dx) {
+          int xx = std::min(cx * 2 + dx, width - 1);
+          int yy = std::min(cy * 2 + dy, height - 1);
+          const uint8_t* p = bgrx + static_cast<size_t>(yy) * stride + xx * 4;
+          float r = p[2], g = p[1], b = p[0];
+          float yv = 0.299f * r + 0.587f * g + 0.114f * b;
+          cbs += -0.168736f * r - 0.331264f * g + 0.5f * b + 128.f;
+          crs += 0.5f * r - 0.418688f * g - 0.081312f * b + 128.f;
+          if (cy * 2 + dy < height && cx * 2 + dx < width)
+            y[static_cast<size_t>(cy * 2 + dy) * ypitch + cx * 2 + dx] =
+                clampf(yv);
+        }
+      cb[static_cast<size_t>(cy) * cpitch + cx] = clampf(cbs * 0.25f);
+      cr[static_cast<size_t>(cy) * cpitch + cx] = clampf(crs * 0.25f);
+    }
+}
+
+}  // namespace h264
+}  // namespace hipflux

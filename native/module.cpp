@@ -6,6 +6,8 @@
 #include <pybind11/stl.h>
 
 #include "cpu/damage.h"
+#include "cpu/h264/cavlc.h"
+#include "cpu/h264/encoder.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
 
@@ -124,6 +126,121 @@ PYBIND11_MODULE(_native, m) {
       py::arg("bgrx"), py::arg("width"), py::arg("height"),
       py::arg("quality") = 80, py::arg("fullcolor") = false,
       "Encode a BGRX buffer as baseline JPEG (CPU reference path).");
+
+  // ---- H.264 stripe encoder (direct access for tests + conformance) ------
+  struct PyH264 {
+    h264::StripeEncoder enc;
+    int w, h, ypitch, cpitch;
+    std::vector<uint8_t> yuv;
+    PyH264(int width, int height)
+        : enc(width, height), w(width), h(height) {
+      ypitch = (w + 15) & ~15;
+      cpitch = ypitch / 2;
+      int yh = (h + 15) & ~15;
+      yuv.resize(static_cast<size_t>(ypitch) * yh * 3 / 2);
+    }
+  };
+  py::class_<PyH264>(m, "H264Encoder")
+      .def(py::init<int, int>(), py::arg("width"), py::arg("height"))
+      .def(
+          "encode",
+          [](PyH264& self, py::buffer bgrx, int qp, bool idr) {
+            py::buffer_info info = bgrx.request();
+            if (info.size < static_cast<ssize_t>(self.w) * self.h * 4)
+              throw std::runtime_error("buffer too small");
+            std::vector<uint8_t> out;
+            h264::EncodeStats st;
+            {
+              py::gil_scoped_release rel;
+              uint8_t* y = self.yuv.data();
+              int yh = (self.h + 15) & ~15;
+              uint8_t* cb = y + static_cast<size_t>(self.ypitch) * yh;
+              uint8_t* cr = cb + static_cast<size_t>(self.cpitch) * (yh / 2);
+              h264::bgrx_to_yuv420(static_cast<const uint8_t*>(info.ptr),
+                                   self.w * 4, self.w, self.h, y, self.ypitch,
+                                   cb, cr, self.cpitch);
+              self.enc.encode_frame(y, self.ypitch, cb, cr, self.cpitch, qp,
+                                    idr, out, &st);
+            }
+            py::dict d;
+            d["data"] = py::bytes(reinterpret_cast<const char*>(out.data()),
+                                  out.size());
+            d["is_idr"] = st.is_idr;
+            d["mb_intra"] = st.mb_intra;
+            d["mb_inter"] = st.mb_inter;
+            d["mb_skip"] = st.mb_skip;
+            return d;
+          },
+          py::arg("bgrx"), py::arg("qp") = 26, py::arg("idr") = false)
+      .def(
+          "encode_yuv",
+          [](PyH264& self, py::buffer ybuf, py::buffer cbbuf, py::buffer crbuf,
+             int qp, bool idr) {
+            auto yi = ybuf.request(), cbi = cbbuf.request(),
+                 cri = crbuf.request();
+            std::vector<uint8_t> out;
+            h264::EncodeStats st;
+            {
+              py::gil_scoped_release rel;
+              self.enc.encode_frame(static_cast<const uint8_t*>(yi.ptr),
+                                    self.w, static_cast<const uint8_t*>(cbi.ptr),
+                                    static_cast<const uint8_t*>(cri.ptr),
+                                    (self.w + 1) / 2, qp, idr, out, &st);
+            }
+            return py::bytes(reinterpret_cast<const char*>(out.data()),
+                             out.size());
+          },
+          py::arg("y"), py::arg("cb"), py::arg("cr"), py::arg("qp") = 26,
+          py::arg("idr") = false,
+          "Encode from tightly-packed YUV420 planes (pitch = width).")
+      .def("recon",
+           [](PyH264& self) {
+             // returns (y, cb, cr, ypitch, cpitch) of the reference frame
+             int yh = (self.h + 15) & ~15;
+             py::bytes y(reinterpret_cast<const char*>(self.enc.recon_y()),
+                         static_cast<size_t>(self.enc.recon_ypitch()) * yh);
+             py::bytes cb(reinterpret_cast<const char*>(self.enc.recon_cb()),
+                          static_cast<size_t>(self.enc.recon_cpitch()) * yh / 2);
+             py::bytes cr(reinterpret_cast<const char*>(self.enc.recon_cr()),
+                          static_cast<size_t>(self.enc.recon_cpitch()) * yh / 2);
+             return py::make_tuple(y, cb, cr, self.enc.recon_ypitch(),
+                                   self.enc.recon_cpitch());
+           });
+
+  m.def(
+      "bgrx_to_yuv420",
+      [](py::buffer bgrx, int width, int height) {
+        py::buffer_info info = bgrx.request();
+        int cw = (width + 1) / 2, ch = (height + 1) / 2;
+        std::vector<uint8_t> y(static_cast<size_t>(width) * height),
+            cb(static_cast<size_t>(cw) * ch), cr(static_cast<size_t>(cw) * ch);
+        h264::bgrx_to_yuv420(static_cast<const uint8_t*>(info.ptr), width * 4,
+                             width, height, y.data(), width, cb.data(),
+                             cr.data(), cw);
+        return py::make_tuple(
+            py::bytes(reinterpret_cast<char*>(y.data()), y.size()),
+            py::bytes(reinterpret_cast<char*>(cb.data()), cb.size()),
+            py::bytes(reinterpret_cast<char*>(cr.data()), cr.size()));
+      },
+      py::arg("bgrx"), py::arg("width"), py::arg("height"));
+
+  m.def(
+      "_cavlc_bits",
+      [](std::vector<int> zz, int nC) {
+        h264::BitWriter bw;
+        int tmp[16] = {};
+        for (size_t i = 0; i < zz.size() && i < 16; ++i) tmp[i] = zz[i];
+        h264::cavlc_residual(bw, tmp, static_cast<int>(zz.size()), nC);
+        size_t nbits = bw.bit_count();
+        bw.put_bit(1);
+        while (bw.bit_count() % 8) bw.put_bit(0);
+        std::string s;
+        const auto& bytes = bw.bytes();
+        for (size_t i = 0; i < nbits; ++i)
+          s += ((bytes[i / 8] >> (7 - i % 8)) & 1) ? '1' : '0';
+        return s;
+      },
+      "debug: CAVLC-encode one zigzag block, return bit string");
 
   m.def("hip_device_count", &hip_device_count,
         "Number of usable HIP devices (0 on CPU-only hosts).");

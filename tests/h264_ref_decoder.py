@@ -1,0 +1,615 @@
+"""Reference H.264 decoder for the subset emitted by hipflux (pure Python).
+
+Implements Annex-B parsing, SPS/PPS/slice headers, CAVLC residual decoding
+(tables loaded from tests/data/cavlc_tables.json — same verified source as
+the C++ header), I16x16 (H/DC) + chroma intra, P_Skip / P_L0_16x16, inverse
+transforms and reconstruction, for streams with:
+  Constrained Baseline, CAVLC, frame_mbs_only, one slice per MB row,
+  pic_order_cnt_type=2, 1 reference frame, deblocking disabled.
+
+Used by tests to verify that decoding an encoder-produced stream yields a
+frame BIT-EXACTLY equal to the encoder's own reconstruction — which
+exercises every entropy/ transform/prediction path end to end.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+
+import numpy as np
+
+_TAB = json.load(open(os.path.join(os.path.dirname(__file__), "data",
+                                   "cavlc_tables.json")))
+
+ZZ4 = [0, 1, 4, 8, 5, 2, 3, 6, 9, 12, 13, 10, 7, 11, 14, 15]
+
+QUANT_V = [[10, 16, 13], [11, 18, 14], [13, 20, 16],
+           [14, 23, 18], [16, 25, 20], [18, 29, 23]]
+
+CHROMA_QP = {30: 29, 31: 30, 32: 31, 33: 32, 34: 32, 35: 33, 36: 34, 37: 34,
+             38: 35, 39: 35, 40: 36, 41: 36, 42: 37, 43: 37, 44: 37, 45: 38,
+             46: 38, 47: 38, 48: 39, 49: 39, 50: 39, 51: 39}
+
+
+def chroma_qp(qp):
+    return qp if qp < 30 else CHROMA_QP[qp]
+
+
+def coeff_class(i, j):
+    ei, ej = (i & 1) == 0, (j & 1) == 0
+    return 0 if (ei and ej) else (1 if (not ei and not ej) else 2)
+
+
+def build_decode_tree(entries):
+    """entries: list of (symbol, len, bits) -> dict[(len,bits)] = symbol"""
+    return {(ln, bits): sym for sym, ln, bits in entries}
+
+
+def _ct_entries(arr):
+    out = []
+    for tc in range(len(arr)):
+        for t1 in range(4):
+            ln, bits = arr[tc][t1]
+            if ln:
+                out.append(((tc, t1), ln, bits))
+    return out
+
+
+CT_TABLES = [build_decode_tree(_ct_entries(a)) for a in _TAB["coeff_token"]]
+CT_CDC = build_decode_tree(_ct_entries(_TAB["coeff_token_cdc"]))
+TZ_TABLES = {int(k): build_decode_tree([(i, ln, bits)
+                                        for i, (ln, bits) in enumerate(v)])
+             for k, v in _TAB["total_zeros"].items()}
+TZ_CDC = {int(k): build_decode_tree([(i, ln, bits)
+                                     for i, (ln, bits) in enumerate(v)])
+          for k, v in _TAB["total_zeros_cdc"].items()}
+RB_TABLES = {int(k): build_decode_tree([(i, ln, bits)
+                                        for i, (ln, bits) in enumerate(v)
+                                        if ln])
+             for k, v in _TAB["run_before"].items()}
+
+
+class BitReader:
+    def __init__(self, data: bytes):
+        self.data = data
+        self.pos = 0  # bit position
+
+    def bit(self) -> int:
+        b = (self.data[self.pos >> 3] >> (7 - (self.pos & 7))) & 1
+        self.pos += 1
+        return b
+
+    def u(self, n: int) -> int:
+        v = 0
+        for _ in range(n):
+            v = (v << 1) | self.bit()
+        return v
+
+    def ue(self) -> int:
+        zeros = 0
+        while self.bit() == 0:
+            zeros += 1
+            if zeros > 31:
+                raise ValueError("bad exp-golomb")
+        return (1 << zeros) - 1 + (self.u(zeros) if zeros else 0)
+
+    def se(self) -> int:
+        cn = self.ue()
+        return (cn + 1) // 2 if cn & 1 else -(cn // 2)
+
+    def vlc(self, tree: dict):
+        ln, bits = 0, 0
+        while ln < 20:
+            bits = (bits << 1) | self.bit()
+            ln += 1
+            if (ln, bits) in tree:
+                return tree[(ln, bits)]
+        raise ValueError(f"VLC decode failed at bit {self.pos}")
+
+    def more_rbsp_data(self) -> bool:
+        # true if there are bits beyond the rbsp_stop_bit
+        total = len(self.data) * 8
+        if self.pos >= total:
+            return False
+        # find last 1-bit in the stream (the stop bit)
+        last = total - 1
+        while last >= 0:
+            if (self.data[last >> 3] >> (7 - (last & 7))) & 1:
+                break
+            last -= 1
+        return self.pos < last
+
+
+def split_nals(data: bytes):
+    """Annex-B -> list of NAL payloads (emulation prevention removed)."""
+    nals = []
+    i = 0
+    starts = []
+    while True:
+        j = data.find(b"\x00\x00\x01", i)
+        if j < 0:
+            break
+        starts.append(j + 3)
+        i = j + 3
+    for k, s in enumerate(starts):
+        e = len(data)
+        if k + 1 < len(starts):
+            e = starts[k + 1] - 3
+            while e > s and data[e - 1] == 0:
+                e -= 1
+        raw = data[s:e]
+        # remove emulation prevention
+        out = bytearray()
+        zeros = 0
+        for b in raw:
+            if zeros >= 2 and b == 3:
+                zeros = 0
+                continue
+            out.append(b)
+            zeros = zeros + 1 if b == 0 else 0
+        nals.append(bytes(out))
+    return nals
+
+
+def residual_cavlc(br: BitReader, n: int, nC: int):
+    """Decode one CAVLC residual block -> (zigzag list of n coeffs, tc)."""
+    if nC == -1:
+        tc, t1 = br.vlc(CT_CDC)
+    elif nC < 2:
+        tc, t1 = br.vlc(CT_TABLES[0])
+    elif nC < 4:
+        tc, t1 = br.vlc(CT_TABLES[1])
+    elif nC < 8:
+        tc, t1 = br.vlc(CT_TABLES[2])
+    else:
+        v = br.u(6)
+        if v == 3:
+            tc, t1 = 0, 0
+        else:
+            tc, t1 = (v >> 2) + 1, v & 3
+    zz = [0] * n
+    if tc == 0:
+        return zz, 0
+    levels = []
+    for _ in range(t1):
+        levels.append(1 if br.bit() == 0 else -1)
+    suffix_len = 1 if (tc > 10 and t1 < 3) else 0
+    for idx in range(tc - t1):
+        # level_prefix
+        prefix = 0
+        while br.bit() == 0:
+            prefix += 1
+            if prefix > 18:
+                raise ValueError("bad level prefix")
+        if suffix_len == 0:
+            if prefix < 14:
+                code = prefix
+            elif prefix == 14:
+                code = 14 + br.u(4)
+            else:
+                code = 30 + br.u(12)
+        else:
+            if prefix < 15:
+                code = (prefix << suffix_len) | br.u(suffix_len)
+            else:
+                code = (15 << suffix_len) + br.u(12)
+        level = (code >> 1) + 1 if (code & 1) == 0 else -((code + 1) >> 1)
+        if idx == 0 and t1 < 3:
+            level = level + 1 if level > 0 else level - 1
+        levels.append(level)
+        if suffix_len == 0:
+            suffix_len = 1
+        if abs(level) > (3 << (suffix_len - 1)) and suffix_len < 6:
+            suffix_len += 1
+    # levels[] is high-frequency-first
+    if tc < n:
+        if nC == -1:
+            total_zeros = br.vlc(TZ_CDC[tc])
+        else:
+            total_zeros = br.vlc(TZ_TABLES[tc])
+    else:
+        total_zeros = 0
+    runs = []
+    zeros_left = total_zeros
+    for k in range(tc - 1):
+        if zeros_left > 0:
+            run = br.vlc(RB_TABLES[min(zeros_left, 7)])
+            zeros_left -= run
+        else:
+            run = 0
+        runs.append(run)
+    runs.append(zeros_left)  # run before the lowest-frequency coefficient
+    pos = -1
+    # place from lowest frequency: reconstruct positions
+    # levels are hf-first; runs are hf-first (gap below each coeff)
+    coeffs_lf = list(reversed(levels))       # low-frequency first
+    runs_lf = list(reversed(runs))
+    for c, r in zip(coeffs_lf, runs_lf):
+        pos += r + 1
+        zz[pos] = c
+    return zz, tc
+
+
+def idct4(blk):
+    out = np.zeros(16, dtype=np.int64)
+    tmp = np.zeros(16, dtype=np.int64)
+    for i in range(4):
+        a, b, c, d = blk[4 * i:4 * i + 4]
+        e0, e1 = a + c, a - c
+        e2, e3 = (b >> 1) - d, b + (d >> 1)
+        tmp[4 * i:4 * i + 4] = [e0 + e3, e1 + e2, e1 - e2, e0 - e3]
+    for j in range(4):
+        a, b, c, d = tmp[j], tmp[4 + j], tmp[8 + j], tmp[12 + j]
+        e0, e1 = a + c, a - c
+        e2, e3 = (b >> 1) - d, b + (d >> 1)
+        out[j] = (e0 + e3 + 32) >> 6
+        out[4 + j] = (e1 + e2 + 32) >> 6
+        out[8 + j] = (e1 - e2 + 32) >> 6
+        out[12 + j] = (e0 - e3 + 32) >> 6
+    return out
+
+
+def hadamard4_inv(blk):
+    tmp = [0] * 16
+    out = [0] * 16
+    for i in range(4):
+        r = blk[4 * i:4 * i + 4]
+        s03, d03 = r[0] + r[3], r[0] - r[3]
+        s12, d12 = r[1] + r[2], r[1] - r[2]
+        tmp[4 * i:4 * i + 4] = [s03 + s12, d03 + d12, s03 - s12, d03 - d12]
+    for j in range(4):
+        a, b, c, d = tmp[j], tmp[4 + j], tmp[8 + j], tmp[12 + j]
+        s03, d03 = a + d, a - d
+        s12, d12 = b + c, b - c
+        out[j] = s03 + s12
+        out[4 + j] = d03 + d12
+        out[8 + j] = s03 - s12
+        out[12 + j] = d03 - d12
+    return out
+
+
+def dequant_ac(level, qp, cls):
+    return (level * QUANT_V[qp % 6][cls]) << (qp // 6)
+
+
+def dequant_luma_dc(c, qp):
+    v = QUANT_V[qp % 6][0]
+    if qp >= 12:
+        return (c * v) << (qp // 6 - 2)
+    return (c * v + (1 << (1 - qp // 6))) >> (2 - qp // 6)
+
+
+def dequant_chroma_dc(c, qp):
+    v = QUANT_V[qp % 6][0]
+    if qp >= 6:
+        return (c * v) << (qp // 6 - 1)
+    return (c * v) >> 1
+
+
+def blk_xy(blk):
+    bx = 2 * ((blk >> 2) & 1) + (blk & 1)
+    by = 2 * (blk >> 3) + ((blk >> 1) & 1)
+    return bx, by
+
+
+class Decoder:
+    def __init__(self):
+        self.sps = None
+        self.pps = None
+        self.y = self.cb = self.cr = None
+        self.ref_y = self.ref_cb = self.ref_cr = None
+        self.frames = []  # decoded (cropped) frames as (y, cb, cr)
+
+    # ---- headers -----------------------------------------------------------
+    def parse_sps(self, br: BitReader):
+        s = {}
+        s["profile_idc"] = br.u(8)
+        br.u(8)  # constraint flags + reserved
+        s["level_idc"] = br.u(8)
+        assert br.ue() == 0
+        s["log2_max_frame_num"] = br.ue() + 4
+        s["poc_type"] = br.ue()
+        assert s["poc_type"] == 2
+        s["max_num_ref_frames"] = br.ue()
+        br.u(1)
+        s["mbw"] = br.ue() + 1
+        s["mbh"] = br.ue() + 1
+        assert br.u(1) == 1  # frame_mbs_only
+        br.u(1)  # direct_8x8
+        if br.u(1):  # cropping
+            cl, cr_, ct, cb_ = br.ue(), br.ue(), br.ue(), br.ue()
+            s["crop"] = (cl * 2, cr_ * 2, ct * 2, cb_ * 2)
+        else:
+            s["crop"] = (0, 0, 0, 0)
+        self.sps = s
+
+    def parse_pps(self, br: BitReader):
+        p = {}
+        assert br.ue() == 0 and br.ue() == 0
+        assert br.u(1) == 0  # CAVLC
+        br.u(1)
+        assert br.ue() == 0  # one slice group
+        p["num_ref_idx_l0"] = br.ue() + 1
+        br.ue()
+        br.u(1)
+        br.u(2)
+        p["pic_init_qp"] = br.se() + 26
+        br.se()
+        p["chroma_qp_offset"] = br.se()
+        p["deblocking_control"] = br.u(1)
+        assert br.u(1) == 0  # constrained_intra_pred
+        br.u(1)
+        self.pps = p
+
+    # ---- slice -------------------------------------------------------------
+    def decode_slice(self, br: BitReader, nal_type: int):
+        s = self.sps
+        mbw, mbh = s["mbw"], s["mbh"]
+        W, H = mbw * 16, mbh * 16
+        idr = nal_type == 5
+        first_mb = br.ue()
+        slice_type = br.ue()
+        is_i = slice_type in (2, 7)
+        is_p = slice_type in (0, 5)
+        assert is_i or is_p
+        assert br.ue() == 0  # pps id
+        br.u(s["log2_max_frame_num"])  # frame_num
+        if idr:
+            br.ue()  # idr_pic_id
+        if is_p:
+            if br.u(1):  # num_ref_idx_override
+                br.ue()
+            assert br.u(1) == 0  # no ref list modification
+        # dec_ref_pic_marking
+        if idr:
+            br.u(1)
+            br.u(1)
+        else:
+            assert br.u(1) == 0
+        qp = self.pps["pic_init_qp"] + br.se()
+        if self.pps["deblocking_control"]:
+            dbf = br.ue()
+            assert dbf == 1, "decoder subset requires deblocking disabled"
+
+        if idr and first_mb == 0:
+            self.y = np.zeros((H, W), np.int32)
+            self.cb = np.zeros((H // 2, W // 2), np.int32)
+            self.cr = np.zeros((H // 2, W // 2), np.int32)
+        if self.y is None:
+            raise ValueError("P slice before any IDR")
+
+        mb_row = first_mb // mbw
+        assert first_mb % mbw == 0, "subset: slice starts at row boundary"
+
+        ctx = {
+            "left_avail": False,
+            "left_inter": False,
+            "left_mv": (0, 0),
+            "left_luma_nc": [0] * 4,
+            "left_cb_nc": [0] * 2,
+            "left_cr_nc": [0] * 2,
+        }
+        mbx = 0
+        skip_left = 0
+        if is_p:
+            skip_left = br.ue()
+        while mbx < mbw:
+            if is_p and skip_left > 0:
+                self.decode_skip(mbx, mb_row)
+                ctx.update(left_avail=True, left_inter=True, left_mv=(0, 0),
+                           left_luma_nc=[0] * 4, left_cb_nc=[0] * 2,
+                           left_cr_nc=[0] * 2)
+                skip_left -= 1
+                mbx += 1
+                continue
+            if is_p and not br.more_rbsp_data():
+                break  # trailing skips consumed the rest
+            mb_type = br.ue()
+            if is_p and mb_type < 5:
+                assert mb_type == 0, f"subset: P mb_type {mb_type}"
+                self.decode_p16(br, mbx, mb_row, ctx)
+            else:
+                it = mb_type - 5 if is_p else mb_type
+                assert 1 <= it <= 24, f"subset: mb_type {mb_type}"
+                self.decode_i16(br, mbx, mb_row, it, qp, ctx)
+            mbx += 1
+            if is_p and mbx < mbw and br.more_rbsp_data():
+                skip_left = br.ue()
+        return mb_row
+
+    def decode_skip(self, mbx, mby):
+        x0, y0 = mbx * 16, mby * 16
+        cx0, cy0 = mbx * 8, mby * 8
+        self.y[y0:y0 + 16, x0:x0 + 16] = self.ref_y[y0:y0 + 16, x0:x0 + 16]
+        self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = self.ref_cb[cy0:cy0 + 8,
+                                                        cx0:cx0 + 8]
+        self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = self.ref_cr[cy0:cy0 + 8,
+                                                        cx0:cx0 + 8]
+
+    def decode_p16(self, br, mbx, mby, ctx):
+        mvdx, mvdy = br.se(), br.se()
+        cbp_cn = br.ue()
+        assert cbp_cn == 0, "subset: inter residual not supported"
+        mvpx, mvpy = ctx["left_mv"] if (ctx["left_avail"] and
+                                        ctx["left_inter"]) else (0, 0)
+        mvx, mvy = mvdx + mvpx, mvdy + mvpy
+        assert mvx % 8 == 0 and mvy % 8 == 0, "subset: even integer MVs"
+        ix, iy = mvx // 4, mvy // 4
+        x0, y0 = mbx * 16, mby * 16
+        cx0, cy0 = mbx * 8, mby * 8
+        self.y[y0:y0 + 16, x0:x0 + 16] = \
+            self.ref_y[y0 + iy:y0 + iy + 16, x0 + ix:x0 + ix + 16]
+        self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = \
+            self.ref_cb[cy0 + iy // 2:cy0 + iy // 2 + 8,
+                        cx0 + ix // 2:cx0 + ix // 2 + 8]
+        self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = \
+            self.ref_cr[cy0 + iy // 2:cy0 + iy // 2 + 8,
+                        cx0 + ix // 2:cx0 + ix // 2 + 8]
+        ctx.update(left_avail=True, left_inter=True, left_mv=(mvx, mvy),
+                   left_luma_nc=[0] * 4, left_cb_nc=[0] * 2,
+                   left_cr_nc=[0] * 2)
+
+    def decode_i16(self, br, mbx, mby, i16_type, qp, ctx):
+        t = i16_type - 1
+        pred_mode = t % 4
+        cbp_chroma = (t // 4) % 3
+        cbp_luma = 15 if t >= 12 else 0
+        chroma_mode = br.ue()
+        qp = qp + br.se()  # mb_qp_delta
+        x0, y0 = mbx * 16, mby * 16
+        cx0, cy0 = mbx * 8, mby * 8
+
+        # luma prediction
+        assert pred_mode in (1, 2), "subset: V/Plane luma pred not emitted"
+        pred = np.zeros((16, 16), np.int32)
+        if pred_mode == 1:  # H
+            assert ctx["left_avail"]
+            col = self.y[y0:y0 + 16, x0 - 1]
+            pred[:] = col[:, None]
+        else:  # DC
+            if ctx["left_avail"]:
+                dc = (int(self.y[y0:y0 + 16, x0 - 1].sum()) + 8) >> 4
+            else:
+                dc = 128
+            pred[:] = dc
+
+        # DC block
+        nC = ctx["left_luma_nc"][0] if ctx["left_avail"] else 0
+        new_luma_nc = np.zeros((4, 4), np.int32)  # per (by,bx) totals
+        zz_dc, _tc = residual_cavlc(br, 16, nC)
+        raster_dc = [0] * 16
+        for i in range(16):
+            raster_dc[ZZ4[i]] = zz_dc[i]
+        ih = hadamard4_inv(raster_dc)
+        dcrec = [dequant_luma_dc(c, qp) for c in ih]
+
+        ac = np.zeros((16, 16), np.int64)  # [blk][raster pos]
+        if cbp_luma:
+            for blk in range(16):
+                bx, by = blk_xy(blk)
+                if bx > 0:
+                    nC = int(new_luma_nc[by, bx - 1])
+                elif ctx["left_avail"]:
+                    nC = ctx["left_luma_nc"][by]
+                else:
+                    nC = 0
+                zz, tc = residual_cavlc(br, 15, nC)
+                new_luma_nc[by, bx] = tc
+                for i in range(1, 16):
+                    ac[blk][ZZ4[i]] = dequant_ac(zz[i - 1], qp,
+                                                 coeff_class(ZZ4[i] >> 2,
+                                                             ZZ4[i] & 3))
+        for blk in range(16):
+            bx, by = blk_xy(blk)
+            coeffs = ac[blk].copy()
+            coeffs[0] = dcrec[by * 4 + bx]
+            rec = idct4(coeffs).reshape(4, 4)
+            self.y[y0 + by * 4:y0 + by * 4 + 4, x0 + bx * 4:x0 + bx * 4 + 4] = \
+                np.clip(rec + pred[by * 4:by * 4 + 4, bx * 4:bx * 4 + 4],
+                        0, 255)
+
+        # chroma
+        qpc = chroma_qp(qp + self.pps["chroma_qp_offset"])
+        assert chroma_mode in (0, 1), "subset: chroma V/Plane not emitted"
+        new_cb_nc = np.zeros((2, 2), np.int32)
+        new_cr_nc = np.zeros((2, 2), np.int32)
+        cpred = {}
+        for comp, plane in (("cb", self.cb), ("cr", self.cr)):
+            p = np.zeros((8, 8), np.int32)
+            if chroma_mode == 1:  # H
+                assert ctx["left_avail"]
+                p[:] = plane[cy0:cy0 + 8, cx0 - 1][:, None]
+            else:
+                for sub in range(4):
+                    scx, scy = (sub & 1) * 4, (sub >> 1) * 4
+                    if ctx["left_avail"]:
+                        d = (int(plane[cy0 + scy:cy0 + scy + 4,
+                                       cx0 - 1].sum()) + 2) >> 2
+                    else:
+                        d = 128
+                    p[scy:scy + 4, scx:scx + 4] = d
+            cpred[comp] = p
+
+        cdc = {"cb": [0] * 4, "cr": [0] * 4}
+        if cbp_chroma >= 1:
+            for comp in ("cb", "cr"):
+                zz, _ = residual_cavlc(br, 4, -1)
+                q = zz  # raster 2x2 (c00,c01,c10,c11)
+                w0 = q[0] + q[1] + q[2] + q[3]
+                w1 = q[0] - q[1] + q[2] - q[3]
+                w2 = q[0] + q[1] - q[2] - q[3]
+                w3 = q[0] - q[1] - q[2] + q[3]
+                cdc[comp] = [dequant_chroma_dc(w, qpc)
+                             for w in (w0, w1, w2, w3)]
+        cac = {"cb": np.zeros((4, 16), np.int64),
+               "cr": np.zeros((4, 16), np.int64)}
+        if cbp_chroma == 2:
+            for comp, newnc, leftnc in (("cb", new_cb_nc, ctx["left_cb_nc"]),
+                                        ("cr", new_cr_nc, ctx["left_cr_nc"])):
+                for sub in range(4):
+                    cx, cy = sub & 1, sub >> 1
+                    if cx > 0:
+                        nC = int(newnc[cy, 0])
+                    elif ctx["left_avail"]:
+                        nC = leftnc[cy]
+                    else:
+                        nC = 0
+                    zz, tc = residual_cavlc(br, 15, nC)
+                    newnc[cy, cx] = tc
+                    for i in range(1, 16):
+                        cac[comp][sub][ZZ4[i]] = dequant_ac(
+                            zz[i - 1], qpc,
+                            coeff_class(ZZ4[i] >> 2, ZZ4[i] & 3))
+        for comp, plane in (("cb", self.cb), ("cr", self.cr)):
+            for sub in range(4):
+                scx, scy = (sub & 1) * 4, (sub >> 1) * 4
+                coeffs = cac[comp][sub].copy()
+                coeffs[0] = cdc[comp][sub]
+                rec = idct4(coeffs).reshape(4, 4)
+                plane[cy0 + scy:cy0 + scy + 4, cx0 + scx:cx0 + scx + 4] = \
+                    np.clip(rec + cpred[comp][scy:scy + 4, scx:scx + 4],
+                            0, 255)
+
+        ctx.update(left_avail=True, left_inter=False,
+                   left_luma_nc=[int(new_luma_nc[by, 3]) for by in range(4)],
+                   left_cb_nc=[int(new_cb_nc[cy, 1]) for cy in range(2)],
+                   left_cr_nc=[int(new_cr_nc[cy, 1]) for cy in range(2)])
+
+    # ---- top level ---------------------------------------------------------
+    def decode(self, data: bytes):
+        """Decode an Annex-B stream; returns list of (y, cb, cr) uint8 frames
+        (cropped to the SPS-declared size)."""
+        rows_done = set()
+        mbh = None
+        for nal in split_nals(data):
+            nal_type = nal[0] & 0x1F
+            br = BitReader(nal[1:])
+            if nal_type == 7:
+                self.parse_sps(br)
+                mbh = self.sps["mbh"]
+            elif nal_type == 8:
+                self.parse_pps(br)
+            elif nal_type in (1, 5):
+                row = self.decode_slice(br, nal_type)
+                rows_done.add(row)
+                if len(rows_done) == mbh:
+                    self.finish_frame()
+                    rows_done = set()
+        return self.frames
+
+    def finish_frame(self):
+        s = self.sps
+        cl, cr_, ct, cb_ = s["crop"]
+        W, H = s["mbw"] * 16, s["mbh"] * 16
+        w, h = W - cl - cr_, H - ct - cb_
+        self.frames.append((
+            self.y[ct:ct + h, cl:cl + w].astype(np.uint8),
+            self.cb[ct // 2:(ct + h) // 2, cl // 2:(cl + w) // 2]
+                .astype(np.uint8),
+            self.cr[ct // 2:(ct + h) // 2, cl // 2:(cl + w) // 2]
+                .astype(np.uint8),
+        ))
+        self.ref_y = self.y.copy()
+        self.ref_cb = self.cb.copy()
+        self.ref_cr = self.cr.copy()

@@ -273,6 +273,7 @@ def encode_residual(zigzag16, nC, bs):
     suffix_len = 1 if (tc > 10 and t1 < 3) else 0
     first = True
     for idx in range(tc - t1 - 1, -1, -1):
+        true_level = coeffs[idx]   # suffix growth uses the TRUE value (9.2.2.1)
         level = coeffs[idx]
         if first and t1 < 3:
             level = level - 1 if level > 0 else level + 1
@@ -297,7 +298,7 @@ def encode_residual(zigzag16, nC, bs):
                 bs.put(code - (15 << suffix_len), 12)
         if suffix_len == 0:
             suffix_len = 1
-        if abs(level) > (3 << (suffix_len - 1)) and suffix_len < 6:
+        if abs(true_level) > (3 << (suffix_len - 1)) and suffix_len < 6:
             suffix_len += 1
     # total zeros
     maxcoeff = 4 if nC == -1 else 16
