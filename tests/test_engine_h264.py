@@ -1,0 +1,94 @@
+"""Engine-level striped H.264: wire framing, per-stripe independent streams,
+damage-gated stripe emission, decodability of every stripe stream."""
+
+import threading
+import time
+
+import numpy as np
+import pytest
+
+hipflux = pytest.importorskip("hipflux")
+if not hipflux.native_available():
+    pytest.skip("hipflux native module not built", allow_module_level=True)
+
+from h264_ref_decoder import Decoder
+
+
+class Collector:
+    def __init__(self):
+        self.stripes = []
+        self.lock = threading.Lock()
+
+    def __call__(self, data, frame_id, y, width, height, is_keyframe,
+                 capture_ts_ms, encode_done_ms, stripe_type):
+        with self.lock:
+            self.stripes.append(dict(data=bytes(data), frame_id=frame_id,
+                                     y=y, width=width, height=height,
+                                     key=is_keyframe, type=stripe_type))
+
+
+def settings(**kw):
+    s = hipflux.CaptureSettings()
+    s.capture_width = 320
+    s.capture_height = 192
+    s.target_fps = 30
+    s.output_mode = 1          # H.264
+    s.use_cpu = True
+    s.gpu_id = -1
+    s.capture_backend = "synthetic:desktop"
+    s.stripe_height = 64
+    s.damage_block_duration = 1
+    s.video_crf = 26
+    for k, v in kw.items():
+        setattr(s, k, v)
+    return s
+
+
+def capture(s, seconds):
+    col = Collector()
+    cap = hipflux.ScreenCapture()
+    cap.start_capture(col, s)
+    time.sleep(seconds)
+    name = cap.pipeline
+    cap.stop_capture()
+    return col, name
+
+
+def test_striped_h264_stream_decodes():
+    col, name = capture(settings(), 0.7)
+    assert name == "cpu-h264"
+    assert col.stripes
+    st = col.stripes[0]
+    data = st["data"]
+    assert data[0] == 0x04 and len(data) > 10
+    # reassemble each stripe row's stream and decode it
+    rows = {}
+    for s in col.stripes:
+        rows.setdefault(s["y"], b"")
+        rows[s["y"]] += s["data"][10:]
+    assert set(rows) == {0, 64, 128}
+    for y, stream in rows.items():
+        frames = Decoder().decode(stream)
+        assert frames, f"stripe y={y} produced no decodable frames"
+        dy, _, _ = frames[0]
+        assert dy.shape == (64, 320)
+
+
+def test_first_frame_stripes_are_keyframes():
+    col, _ = capture(settings(), 0.4)
+    f0 = [s for s in col.stripes if s["frame_id"] == 0]
+    assert f0 and all(s["key"] for s in f0)
+    # header keyflag matches
+    for s in f0:
+        assert s["data"][1] == 1
+
+
+def test_damage_gated_stripes():
+    col, _ = capture(settings(capture_backend="synthetic:desktop",
+                              capture_height=448,
+                              use_paint_over_quality=False), 0.6)
+    per_frame = {}
+    for s in col.stripes:
+        per_frame.setdefault(s["frame_id"], []).append(s["y"])
+    later = [len(v) for f, v in per_frame.items() if f > 0]
+    assert later and min(later) < 3, "damage gating never skipped a stripe"
