@@ -1,0 +1,184 @@
+#include "gpu_entropy.h"
+
+#include "../../hip/h264_gpu_layout.h"
+#include "bitwriter.h"
+#include "cavlc.h"
+#include "headers.h"
+#include "transform.h"
+
+namespace hipflux {
+namespace h264 {
+namespace {
+
+using h264gpu::kChromaAcOff;
+using h264gpu::kChromaDcOff;
+using h264gpu::kLevelsPerMb;
+using h264gpu::kLumaAcOff;
+using h264gpu::kLumaDcOff;
+using h264gpu::kMetaPerMb;
+
+inline void blk_xy(int blk, int& bx, int& by) {
+  bx = 2 * ((blk >> 2) & 1) + (blk & 1);
+  by = 2 * (blk >> 3) + ((blk >> 1) & 1);
+}
+
+struct RowCtx {
+  bool have_left = false;
+  bool left_is_inter = false;
+  int left_mvx = 0, left_mvy = 0;
+  uint8_t left_luma_nc[4] = {};
+  uint8_t left_cb_nc[2] = {};
+  uint8_t left_cr_nc[2] = {};
+  int skip_run = 0;
+};
+
+// Encode one intra MB from the GPU level buffer. Mirrors
+// encoder.cpp::encode_i16's entropy section exactly.
+void entropy_i16(BitWriter& bw, const int16_t* L, int m0, int qp,
+                 bool p_slice, RowCtx& ctx) {
+  const int luma_mode = (m0 >> 2) & 7;
+  const int chroma_mode = (m0 >> 5) & 7;
+
+  // gather zigzag blocks + cbp from levels
+  int zz_dc[16];
+  for (int i = 0; i < 16; ++i) zz_dc[i] = L[kLumaDcOff + kZigzag4[i]];
+  int zz_ac[16][15];
+  bool any_ac = false;
+  for (int r = 0; r < 16; ++r) {
+    for (int i = 1; i < 16; ++i) {
+      zz_ac[r][i - 1] = L[kLumaAcOff + r * 16 + kZigzag4[i]];
+      any_ac |= zz_ac[r][i - 1] != 0;
+    }
+  }
+  const int cbp_luma = any_ac ? 15 : 0;
+  int cdc[2][4];
+  bool any_cdc = false;
+  for (int comp = 0; comp < 2; ++comp)
+    for (int i = 0; i < 4; ++i) {
+      cdc[comp][i] = L[kChromaDcOff + comp * 4 + i];
+      any_cdc |= cdc[comp][i] != 0;
+    }
+  int czz[8][15];
+  bool any_cac = false;
+  for (int b = 0; b < 8; ++b)
+    for (int i = 1; i < 16; ++i) {
+      czz[b][i - 1] = L[kChromaAcOff + b * 16 + kZigzag4[i]];
+      any_cac |= czz[b][i - 1] != 0;
+    }
+  const int cbp_chroma = any_cac ? 2 : (any_cdc ? 1 : 0);
+
+  const int i16_type =
+      1 + luma_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0);
+  bw.ue(p_slice ? 5 + i16_type : i16_type);
+  bw.ue(chroma_mode);
+  bw.se(0);  // mb_qp_delta
+
+  uint8_t new_luma[16] = {};
+  uint8_t new_cb[4] = {}, new_cr[4] = {};
+  // DC block nC = blk(0,0) context
+  int nC = ctx.have_left ? ctx.left_luma_nc[0] : 0;
+  cavlc_residual(bw, zz_dc, 16, nC);
+  if (cbp_luma) {
+    for (int blk = 0; blk < 16; ++blk) {
+      int bx, by;
+      blk_xy(blk, bx, by);
+      int r = by * 4 + bx;
+      int n;
+      if (bx > 0)
+        n = new_luma[by * 4 + bx - 1];
+      else
+        n = ctx.have_left ? ctx.left_luma_nc[by] : 0;
+      int tc = cavlc_residual(bw, zz_ac[r], 15, n);
+      new_luma[r] = static_cast<uint8_t>(tc);
+    }
+  }
+  if (cbp_chroma > 0) {
+    cavlc_residual(bw, cdc[0], 4, -1);
+    cavlc_residual(bw, cdc[1], 4, -1);
+  }
+  if (cbp_chroma == 2) {
+    for (int comp = 0; comp < 2; ++comp) {
+      uint8_t* t = comp ? new_cr : new_cb;
+      const uint8_t* lt = comp ? ctx.left_cr_nc : ctx.left_cb_nc;
+      for (int sub = 0; sub < 4; ++sub) {
+        int cx = sub & 1, cy = sub >> 1;
+        int n = cx > 0 ? t[cy * 2] : (ctx.have_left ? lt[cy] : 0);
+        int tc = cavlc_residual(bw, czz[comp * 4 + sub], 15, n);
+        t[cy * 2 + cx] = static_cast<uint8_t>(tc);
+      }
+    }
+  }
+  ctx.have_left = true;
+  ctx.left_is_inter = false;
+  for (int by = 0; by < 4; ++by) ctx.left_luma_nc[by] = new_luma[by * 4 + 3];
+  for (int cy = 0; cy < 2; ++cy) {
+    ctx.left_cb_nc[cy] = new_cb[cy * 2 + 1];
+    ctx.left_cr_nc[cy] = new_cr[cy * 2 + 1];
+  }
+}
+
+}  // namespace
+
+void encode_stripe_from_gpu(const GpuStripeParams& p,
+                            std::vector<uint8_t>& out) {
+  const int mbh = p.n_mb_rows;
+  const int mbw_stripe = (p.width + 15) / 16;
+  if (p.idr) {
+    write_sps_nal(out, mbw_stripe, mbh, p.width, p.height);
+    write_pps_nal(out);
+  }
+  for (int row = 0; row < mbh; ++row) {
+    BitWriter b;
+    write_slice_header_bits(b, p.idr, row * mbw_stripe, p.frame_num,
+                            p.idr_pic_id, p.qp);
+    RowCtx ctx;
+    const size_t mb_base = (size_t)(p.mb_row0 + row) * p.mbw;
+    for (int mbx = 0; mbx < mbw_stripe; ++mbx) {
+      const int16_t* L = p.levels + (mb_base + mbx) * kLevelsPerMb;
+      const int* M = p.meta + (mb_base + mbx) * kMetaPerMb;
+      int m0 = M[0];
+      int mode = p.idr ? h264gpu::kIntra : (m0 & 3);
+      if (!p.idr && mode == h264gpu::kSkip) {
+        ++ctx.skip_run;
+        ctx.have_left = true;
+        ctx.left_is_inter = true;
+        ctx.left_mvx = 0;
+        ctx.left_mvy = 0;
+        std::fill(ctx.left_luma_nc, ctx.left_luma_nc + 4, 0);
+        ctx.left_cb_nc[0] = ctx.left_cb_nc[1] = 0;
+        ctx.left_cr_nc[0] = ctx.left_cr_nc[1] = 0;
+        continue;
+      }
+      if (!p.idr) {
+        b.ue(ctx.skip_run);
+        ctx.skip_run = 0;
+      }
+      if (!p.idr && mode == h264gpu::kInter) {
+        int m1 = M[1];
+        int mvx = (int16_t)(m1 & 0xFFFF);
+        int mvy = m1 >> 16;
+        b.ue(0);  // P_L0_16x16
+        int mvpx = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvx : 0;
+        int mvpy = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvy : 0;
+        b.se(mvx - mvpx);
+        b.se(mvy - mvpy);
+        b.ue(0);  // cbp = 0
+        ctx.have_left = true;
+        ctx.left_is_inter = true;
+        ctx.left_mvx = mvx;
+        ctx.left_mvy = mvy;
+        std::fill(ctx.left_luma_nc, ctx.left_luma_nc + 4, 0);
+        ctx.left_cb_nc[0] = ctx.left_cb_nc[1] = 0;
+        ctx.left_cr_nc[0] = ctx.left_cr_nc[1] = 0;
+        continue;
+      }
+      entropy_i16(b, L, m0, p.qp, !p.idr, ctx);
+    }
+    if (!p.idr && ctx.skip_run > 0) b.ue(ctx.skip_run);
+    b.rbsp_trailing();
+    b.emit_nal(out, p.idr ? 3 : 2, p.idr ? 5 : 1, row == 0);
+  }
+}
+
+}  // namespace h264
+}  // namespace hipflux

@@ -1,0 +1,30 @@
+// CAVLC entropy packing from GPU-produced level/meta buffers
+// (layout: native/hip/h264_gpu_layout.h). One call packs one stripe's
+// frame: SPS/PPS (on IDR) + one slice NAL per MB row.
+#pragma once
+
+#include <cstdint>
+#include <vector>
+
+namespace hipflux {
+namespace h264 {
+
+struct GpuStripeParams {
+  const int16_t* levels;   // frame-wide levels buffer (host copy)
+  const int* meta;         // frame-wide meta buffer (host copy)
+  int mbw;                 // frame MBs per row
+  int mb_row0;             // first absolute MB row of this stripe
+  int n_mb_rows;           // MB rows in this stripe
+  int width;               // stripe logical width (pixels)
+  int height;              // stripe logical height (pixels)
+  int qp;
+  bool idr;
+  uint32_t frame_num;
+  uint32_t idr_pic_id;
+};
+
+void encode_stripe_from_gpu(const GpuStripeParams& p,
+                            std::vector<uint8_t>& out);
+
+}  // namespace h264
+}  // namespace hipflux

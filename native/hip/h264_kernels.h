@@ -1,0 +1,24 @@
+#pragma once
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#include "h264_gpu_layout.h"
+
+namespace hipflux {
+namespace h264gpu {
+
+void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
+                    const uint8_t* refY, int mbw, int n_jobs,
+                    const RowJob* d_jobs, int* d_meta, hipStream_t stream);
+
+void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
+                      const uint8_t* srcCr, int ypitch, int cpitch, int w,
+                      int h, const uint8_t* refY, const uint8_t* refCb,
+                      const uint8_t* refCr, uint8_t* curY, uint8_t* curCb,
+                      uint8_t* curCr, int mbw, int n_jobs,
+                      const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
+                      hipStream_t stream);
+
+}  // namespace h264gpu
+}  // namespace hipflux

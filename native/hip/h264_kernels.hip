@@ -1,0 +1,641 @@
+// CDNA4 (gfx950) H.264 encode kernels.
+//
+// Parallel structure (see cpu/h264/encoder.h for the slice design):
+//  * one slice per MB row -> every MB row is independent; k_h264_rows runs
+//    ONE WAVE (64 lanes) per scheduled row, walking its MBs left-to-right
+//    (the only true dependency: H/DC intra prediction + recon feedback).
+//  * within an MB, 64 lanes process 4 4x4 blocks at a time, one lane per
+//    coefficient; the 4x4 forward/inverse transforms are done with
+//    cross-lane __shfl butterflies inside each 16-lane group (no LDS
+//    round-trips, wave-synchronous by construction).
+//  * k_me does P-frame mode decision (skip / inter MV / intra) for all MBs
+//    of scheduled P rows in parallel, one workgroup per MB, before the row
+//    kernel runs.
+//
+// Integer semantics match cpu/h264/transform.h exactly (same quant/dequant
+// formulas, same floor shifts) so GPU streams decode bit-identically under
+// tests/h264_ref_decoder.py.
+#include <hip/hip_runtime.h>
+
+#include "h264_gpu_layout.h"
+#include "h264_kernels.h"
+
+namespace hipflux {
+namespace h264gpu {
+
+__constant__ int c_quant_mf[6][3] = {
+    {13107, 5243, 8066}, {11916, 4660, 7490}, {10082, 4194, 6554},
+    {9362, 3647, 5825},  {8192, 3355, 5243},  {7282, 2893, 4559}};
+__constant__ int c_dequant_v[6][3] = {{10, 16, 13}, {11, 18, 14},
+                                      {13, 20, 16}, {14, 23, 18},
+                                      {16, 25, 20}, {18, 29, 23}};
+// zigzag index of each raster position (inverse of the scan)
+__constant__ int c_zz_of_pos[16] = {0, 1, 5, 6, 2, 4, 7, 12,
+                                    3, 8, 11, 13, 9, 10, 14, 15};
+__constant__ int c_chroma_qp[22] = {29, 30, 31, 32, 32, 33, 34, 34,
+                                    35, 35, 36, 36, 37, 37, 37, 38,
+                                    38, 38, 39, 39, 39, 39};
+
+__device__ inline int dev_chroma_qp(int qp) {
+  return qp < 30 ? qp : c_chroma_qp[qp - 30];
+}
+
+__device__ inline int coeff_cls(int pos) {
+  int i = pos >> 2, j = pos & 3;
+  bool ei = (i & 1) == 0, ej = (j & 1) == 0;
+  return (ei && ej) ? 0 : ((!ei && !ej) ? 1 : 2);
+}
+
+__device__ inline int quant_coeff(int w, int qp, int cls) {
+  int qbits = 15 + qp / 6;
+  int f = (1 << qbits) / 3;  // intra
+  int az = abs(w);
+  int level = (az * c_quant_mf[qp % 6][cls] + f) >> qbits;
+  level = min(level, 2063);
+  return w < 0 ? -level : level;
+}
+
+__device__ inline int quant_dc_v(int w, int qp) {
+  int qbits = 15 + qp / 6;
+  int f = (1 << qbits) / 3;
+  int az = abs(w);
+  int level = (az * c_quant_mf[qp % 6][0] + 2 * f) >> (qbits + 1);
+  level = min(level, 2063);
+  return w < 0 ? -level : level;
+}
+
+__device__ inline int dequant_c(int level, int qp, int cls) {
+  return (level * c_dequant_v[qp % 6][cls]) << (qp / 6);
+}
+
+__device__ inline int dequant_luma_dc_v(int c, int qp) {
+  int v = c_dequant_v[qp % 6][0];
+  if (qp >= 12) return (c * v) << (qp / 6 - 2);
+  return (c * v + (1 << (1 - qp / 6))) >> (2 - qp / 6);
+}
+
+__device__ inline int dequant_chroma_dc_v(int c, int qp) {
+  int v = c_dequant_v[qp % 6][0];
+  if (qp >= 6) return (c * v) << (qp / 6 - 1);
+  return (c * v) >> 1;
+}
+
+__device__ inline uint8_t clip8(int v) { return (uint8_t)max(0, min(255, v)); }
+
+// ---------------------------------------------------------------------------
+// 16-lane-group 4x4 transforms via shuffles. lane c in [0,16): r=c>>2, x=c&3.
+// base = (lane & ~15) is the group's first lane in the wave.
+
+__device__ inline int fdct4_wave(int v, int lane) {
+  int base = lane & ~15;
+  int r = (lane >> 2) & 3, u = lane & 3;
+  int s0 = __shfl(v, base + r * 4 + 0);
+  int s1 = __shfl(v, base + r * 4 + 1);
+  int s2 = __shfl(v, base + r * 4 + 2);
+  int s3 = __shfl(v, base + r * 4 + 3);
+  int t;
+  switch (u) {
+    case 0: t = s0 + s1 + s2 + s3; break;
+    case 1: t = 2 * s0 + s1 - s2 - 2 * s3; break;
+    case 2: t = s0 - s1 - s2 + s3; break;
+    default: t = s0 - 2 * s1 + 2 * s2 - s3; break;
+  }
+  int vv = r;  // now do columns: lane (vv,u) reads t[y][u] from lanes y*4+u
+  int c0 = __shfl(t, base + 0 * 4 + u);
+  int c1 = __shfl(t, base + 1 * 4 + u);
+  int c2 = __shfl(t, base + 2 * 4 + u);
+  int c3 = __shfl(t, base + 3 * 4 + u);
+  switch (vv) {
+    case 0: return c0 + c1 + c2 + c3;
+    case 1: return 2 * c0 + c1 - c2 - 2 * c3;
+    case 2: return c0 - c1 - c2 + c3;
+    default: return c0 - 2 * c1 + 2 * c2 - c3;
+  }
+}
+
+// inverse core transform; input dequantized coeff per lane; result includes
+// (x+32)>>6
+__device__ inline int idct4_wave(int d, int lane) {
+  int base = lane & ~15;
+  int r = (lane >> 2) & 3, x = lane & 3;
+  int d0 = __shfl(d, base + r * 4 + 0);
+  int d1 = __shfl(d, base + r * 4 + 1);
+  int d2 = __shfl(d, base + r * 4 + 2);
+  int d3 = __shfl(d, base + r * 4 + 3);
+  int e0 = d0 + d2, e1 = d0 - d2;
+  int e2 = (d1 >> 1) - d3, e3 = d1 + (d3 >> 1);
+  int t;
+  switch (x) {
+    case 0: t = e0 + e3; break;
+    case 1: t = e1 + e2; break;
+    case 2: t = e1 - e2; break;
+    default: t = e0 - e3; break;
+  }
+  int c0 = __shfl(t, base + 0 * 4 + x);
+  int c1 = __shfl(t, base + 1 * 4 + x);
+  int c2 = __shfl(t, base + 2 * 4 + x);
+  int c3 = __shfl(t, base + 3 * 4 + x);
+  int f0 = c0 + c2, f1 = c0 - c2;
+  int f2 = (c1 >> 1) - c3, f3 = c1 + (c3 >> 1);
+  int o;
+  switch (r) {
+    case 0: o = f0 + f3; break;
+    case 1: o = f1 + f2; break;
+    case 2: o = f1 - f2; break;
+    default: o = f0 - f3; break;
+  }
+  return (o + 32) >> 6;
+}
+
+// 4x4 Hadamard (fwd includes >>1) on lanes 0..15 of the wave
+__device__ inline int hadamard4_wave(int v, int lane, bool fwd) {
+  int r = (lane >> 2) & 3, u = lane & 3;
+  int base = lane & ~15;
+  int s0 = __shfl(v, base + r * 4 + 0);
+  int s1 = __shfl(v, base + r * 4 + 1);
+  int s2 = __shfl(v, base + r * 4 + 2);
+  int s3 = __shfl(v, base + r * 4 + 3);
+  int a03 = s0 + s3, d03 = s0 - s3, a12 = s1 + s2, d12 = s1 - s2;
+  int t;
+  switch (u) {
+    case 0: t = a03 + a12; break;
+    case 1: t = d03 + d12; break;
+    case 2: t = a03 - a12; break;
+    default: t = d03 - d12; break;
+  }
+  int c0 = __shfl(t, base + 0 * 4 + u);
+  int c1 = __shfl(t, base + 1 * 4 + u);
+  int c2 = __shfl(t, base + 2 * 4 + u);
+  int c3 = __shfl(t, base + 3 * 4 + u);
+  int b03 = c0 + c3, e03 = c0 - c3, b12 = c1 + c2, e12 = c1 - c2;
+  int o;
+  switch (r) {
+    case 0: o = b03 + b12; break;
+    case 1: o = e03 + e12; break;
+    case 2: o = b03 - b12; break;
+    default: o = e03 - e12; break;
+  }
+  return fwd ? (o >> 1) : o;
+}
+
+// cap nonzero count at 12 within a 16-lane group (zero highest zigzag)
+// active: whether this lane's coefficient participates (e.g. AC excludes 0)
+__device__ inline int cap12_group(int level, int zz, bool active, int lane) {
+  for (;;) {
+    unsigned long long m = __ballot(active && level != 0);
+    int cnt = __popcll((m >> (lane & ~15)) & 0xFFFFULL);
+    if (cnt <= 12) break;
+    int key = (active && level != 0) ? zz : -1;
+    int mx = key;
+    for (int d = 1; d < 16; d <<= 1) mx = max(mx, __shfl_xor(mx, d));
+    if (key == mx) level = 0;
+  }
+  return level;
+}
+
+__device__ inline int wave_sum_i(int v) {
+  for (int d = 1; d < 64; d <<= 1) v += __shfl_xor(v, d);
+  return v;
+}
+
+__device__ inline int wave_min_i(int v) {
+  for (int d = 1; d < 64; d <<= 1) v = min(v, __shfl_xor(v, d));
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// The row kernel. One wave per RowJob; walks MBs of the row.
+__global__ void __launch_bounds__(64) k_h264_rows(
+    const uint8_t* __restrict__ srcY, const uint8_t* __restrict__ srcCb,
+    const uint8_t* __restrict__ srcCr, int ypitch, int cpitch, int w, int h,
+    const uint8_t* __restrict__ refY, const uint8_t* __restrict__ refCb,
+    const uint8_t* __restrict__ refCr, uint8_t* __restrict__ curY,
+    uint8_t* __restrict__ curCb, uint8_t* __restrict__ curCr, int mbw,
+    const RowJob* __restrict__ jobs, int16_t* __restrict__ levels,
+    int* __restrict__ meta) {
+  const RowJob job = jobs[blockIdx.x];
+  const int lane = threadIdx.x;
+  const int qp = job.qp;
+  const int qpc = dev_chroma_qp(qp);
+  const bool i_slice = (job.flags & 1) != 0;
+  const int mby = job.mb_row;
+
+  __shared__ uint8_t s_left_y[16];       // left MB's rightmost recon column
+  __shared__ uint8_t s_left_cb[8];
+  __shared__ uint8_t s_left_cr[8];
+  __shared__ uint8_t s_left_y_new[16];   // staging (committed after recon)
+  __shared__ uint8_t s_left_cb_new[8];
+  __shared__ uint8_t s_left_cr_new[8];
+  __shared__ int s_dc[16];
+  __shared__ int16_t s_lvl[16 * 16];     // luma AC levels of current MB
+  __shared__ int s_dcrec[16];
+  __shared__ int s_cdc[8];               // chroma DCs (cb0..3, cr0..3)
+  __shared__ int s_cdcq[8];              // quantized (hadamard domain)
+  __shared__ int s_cdcrec[8];            // dequantized recon DC
+  __shared__ int16_t s_clvl[8 * 16];     // chroma AC levels
+  __shared__ uint8_t s_src[256];         // current MB source luma
+  __shared__ uint8_t s_csrc[128];        // source chroma (cb 8x8, cr 8x8)
+
+  bool have_left = false;
+
+  for (int mbx = 0; mbx < mbw; ++mbx) {
+    const int x0 = mbx * 16, y0 = mby * 16;
+    const int cx0 = mbx * 8, cy0 = mby * 8;
+    const size_t mb_index = (size_t)mby * mbw + mbx;
+    int16_t* L = levels + mb_index * kLevelsPerMb;
+    int* M = meta + mb_index * kMetaPerMb;
+
+    int mode = kIntra;
+    int mvx = 0, mvy = 0;
+    if (!i_slice) {
+      int m0 = M[0];
+      mode = m0 & 3;
+      int m1 = M[1];
+      mvx = (short)(m1 & 0xFFFF);
+      mvy = m1 >> 16;
+    }
+
+    if (mode != kIntra) {
+      // skip / inter: motion copy from ref (even integer MVs)
+      int ix = mvx >> 2, iy = mvy >> 2;
+      // luma: lane = r*4 + cq; each lane copies 4 bytes
+      {
+        int r = lane >> 2, cq = (lane & 3) * 4;
+        const uint8_t* s =
+            refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
+        uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
+        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
+        if (cq == 12) s_left_y[r] = s[3];  // no pred reads in copy MBs
+      }
+      // chroma: lanes 0..15 copy cb rows, 16..31 cr
+      if (lane < 32) {
+        int comp = lane >> 4;
+        int r = (lane & 15) >> 1, cq = (lane & 1) * 4;
+        const uint8_t* sp = comp ? refCr : refCb;
+        uint8_t* dp = comp ? curCr : curCb;
+        const uint8_t* s =
+            sp + (size_t)(cy0 + (iy >> 1) + r) * cpitch + cx0 + (ix >> 1) + cq;
+        uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
+        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
+        if (cq == 4) {
+          if (comp) s_left_cr[r] = s[3]; else s_left_cb[r] = s[3];
+        }
+      }
+      __syncthreads();
+      have_left = true;
+      continue;
+    }
+
+    // ---------------- intra I16x16 ----------------
+    // load src MB (clamped edge replication)
+    int r = lane >> 2, cq = (lane & 3) * 4;
+    int srow = min(y0 + r, h - 1);
+    int p0, p1, p2, p3;
+    {
+      const uint8_t* s = srcY + (size_t)srow * ypitch;
+      p0 = s[min(x0 + cq + 0, w - 1)];
+      p1 = s[min(x0 + cq + 1, w - 1)];
+      p2 = s[min(x0 + cq + 2, w - 1)];
+      p3 = s[min(x0 + cq + 3, w - 1)];
+    }
+    // DC value from left recon
+    int dcval = 128;
+    if (have_left) {
+      int part = (lane < 16) ? s_left_y[lane] : 0;
+      int tot = wave_sum_i(part);
+      dcval = (tot + 8) >> 4;
+    }
+    // costs
+    int costH = 0, costDC = 0;
+    {
+      int lv = have_left ? s_left_y[r] : 0;
+      costDC = abs(p0 - dcval) + abs(p1 - dcval) + abs(p2 - dcval) +
+               abs(p3 - dcval);
+      costH = abs(p0 - lv) + abs(p1 - lv) + abs(p2 - lv) + abs(p3 - lv);
+      costDC = wave_sum_i(costDC);
+      costH = wave_sum_i(costH);
+    }
+    const int luma_mode = (have_left && costH < costDC) ? 1 : 2;
+
+    // stash src into LDS for per-block passes
+    s_src[r * 16 + cq + 0] = (uint8_t)p0;
+    s_src[r * 16 + cq + 1] = (uint8_t)p1;
+    s_src[r * 16 + cq + 2] = (uint8_t)p2;
+    s_src[r * 16 + cq + 3] = (uint8_t)p3;
+    __syncthreads();
+
+    // 4 passes x 4 blocks: forward DCT + quant AC, collect DC
+    const int g = lane >> 4;        // group = block within pass
+    const int c = lane & 15;        // coeff position (raster)
+    const int zz = c_zz_of_pos[c];
+    for (int pass = 0; pass < 4; ++pass) {
+      int blk = pass * 4 + g;       // raster block index
+      int bx = blk & 3, by = blk >> 2;
+      int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
+      int pred = luma_mode == 1 ? s_left_y[py] : dcval;
+      int resid = (int)s_src[py * 16 + px] - pred;
+      int coefv = fdct4_wave(resid, lane);
+      if (c == 0) s_dc[blk] = coefv;
+      int lvl = (c == 0) ? 0 : quant_coeff(coefv, qp, coeff_cls(c));
+      lvl = cap12_group(lvl, zz, c != 0, lane);
+      s_lvl[blk * 16 + c] = (int16_t)lvl;
+      L[kLumaAcOff + blk * 16 + c] = (int16_t)lvl;
+    }
+    __syncthreads();
+
+    // luma DC: hadamard + quant + inverse + dequant (lanes 0..15)
+    {
+      int v = (lane < 16) ? s_dc[lane] : 0;
+      int had = hadamard4_wave(v, lane, true);
+      int qdc = quant_dc_v(had, qp);
+      qdc = cap12_group(qdc, zz, lane < 16, lane);
+      // NOTE: DC block max coeffs 16, cap at 12 shared helper (active only
+      // for lanes 0..15)
+      if (lane < 16) L[kLumaDcOff + lane] = (int16_t)qdc;
+      int ih = hadamard4_wave(qdc, lane, false);
+      if (lane < 16) s_dcrec[lane] = dequant_luma_dc_v(ih, qp);
+    }
+    __syncthreads();
+
+    // cbp_luma: any AC nonzero?
+    int anyl = 0;
+    for (int i = lane; i < 256; i += 64) anyl |= s_lvl[i] != 0;
+    anyl = wave_sum_i(anyl);
+    const int cbp_luma = anyl ? 15 : 0;
+
+    // recon passes
+    for (int pass = 0; pass < 4; ++pass) {
+      int blk = pass * 4 + g;
+      int bx = blk & 3, by = blk >> 2;
+      int d = 0;
+      if (c == 0) {
+        d = s_dcrec[blk];
+      } else if (cbp_luma) {
+        d = dequant_c(s_lvl[blk * 16 + c], qp, coeff_cls(c));
+      }
+      int rec = idct4_wave(d, lane);
+      int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
+      int pred = luma_mode == 1 ? s_left_y[py] : dcval;
+      uint8_t pix = clip8(rec + pred);
+      curY[(size_t)(y0 + py) * ypitch + x0 + px] = pix;
+      if (px == 15) s_left_y_new[py] = pix;  // stage next MB's left column
+    }
+    __syncthreads();
+    if (lane < 16) s_left_y[lane] = s_left_y_new[lane];
+    __syncthreads();
+
+    // ---------------- chroma ----------------
+    if (lane < 32) {
+      int comp = lane >> 4;
+      int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
+      const uint8_t* sp = comp ? srcCr : srcCb;
+      int cw = (w + 1) / 2, chh = (h + 1) / 2;
+      int sr = min(cy0 + rr, chh - 1);
+      for (int k = 0; k < 4; ++k)
+        s_csrc[comp * 64 + rr * 8 + ccq + k] =
+            sp[(size_t)sr * cpitch + min(cx0 + ccq + k, cw - 1)];
+    }
+    __syncthreads();
+
+    // chroma mode decision: DC(0) vs H(1), shared across components.
+    // lane -> (comp, row, 2 columns); DC pred of a pixel = avg of the left
+    // 4 recon pixels of its 4x4 sub-block row range (top unavailable).
+    int ccH = 0, ccDC = 0;
+    {
+      int comp = lane >> 5;           // 0 cb, 1 cr
+      int idx = lane & 31;            // 8 rows x 4 col-pairs
+      int rr = idx >> 2, ccq = (idx & 3) * 2;
+      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
+      int scy4 = (rr >> 2) * 4;
+      int dd = 128;
+      if (have_left)
+        dd = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
+      int lv = have_left ? lc[rr] : 0;
+      for (int k = 0; k < 2; ++k) {
+        int sv = s_csrc[comp * 64 + rr * 8 + ccq + k];
+        ccDC += abs(sv - dd);
+        ccH += abs(sv - lv);
+      }
+      ccDC = wave_sum_i(ccDC);
+      ccH = wave_sum_i(ccH);
+    }
+    const int chroma_mode = (have_left && ccH < ccDC) ? 1 : 0;
+
+    // chroma transform: 2 passes x 4 groups (pass0: cb0..3, pass1: cr0..3)
+    for (int pass = 0; pass < 2; ++pass) {
+      int comp = pass;
+      int sub = g;                     // sub-block raster in 8x8
+      int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+      int rr = scy + (c >> 2), cc2 = scx + (c & 3);
+      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
+      int pred;
+      if (chroma_mode == 1) {
+        pred = lc[rr];
+      } else {
+        pred = 128;
+        if (have_left) {
+          int scy4 = (rr >> 2) * 4;
+          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
+        }
+      }
+      int resid = (int)s_csrc[comp * 64 + rr * 8 + cc2] - pred;
+      int coefv = fdct4_wave(resid, lane);
+      if (c == 0) s_cdc[comp * 4 + sub] = coefv;
+      int lvl = (c == 0) ? 0 : quant_coeff(coefv, qpc, coeff_cls(c));
+      lvl = cap12_group(lvl, zz, c != 0, lane);
+      s_clvl[(comp * 4 + sub) * 16 + c] = (int16_t)lvl;
+      L[kChromaAcOff + (comp * 4 + sub) * 16 + c] = (int16_t)lvl;
+    }
+    __syncthreads();
+
+    // chroma DC 2x2 hadamard + quant + recon scaling (lane 0: cb, lane 1: cr)
+    if (lane < 2) {
+      int* d = &s_cdc[lane * 4];
+      int w0 = d[0] + d[1] + d[2] + d[3];
+      int w1 = d[0] - d[1] + d[2] - d[3];
+      int w2 = d[0] + d[1] - d[2] - d[3];
+      int w3 = d[0] - d[1] - d[2] + d[3];
+      int q0 = quant_dc_v(w0, qpc), q1 = quant_dc_v(w1, qpc);
+      int q2 = quant_dc_v(w2, qpc), q3 = quant_dc_v(w3, qpc);
+      s_cdcq[lane * 4 + 0] = q0;
+      s_cdcq[lane * 4 + 1] = q1;
+      s_cdcq[lane * 4 + 2] = q2;
+      s_cdcq[lane * 4 + 3] = q3;
+      L[kChromaDcOff + lane * 4 + 0] = (int16_t)q0;
+      L[kChromaDcOff + lane * 4 + 1] = (int16_t)q1;
+      L[kChromaDcOff + lane * 4 + 2] = (int16_t)q2;
+      L[kChromaDcOff + lane * 4 + 3] = (int16_t)q3;
+    }
+    __syncthreads();
+    // cbp_chroma
+    int anyc_ac = 0, anyc_dc = 0;
+    for (int i = lane; i < 128; i += 64) anyc_ac |= s_clvl[i] != 0;
+    if (lane < 8) anyc_dc = s_cdcq[lane] != 0;
+    anyc_ac = wave_sum_i(anyc_ac);
+    anyc_dc = wave_sum_i(anyc_dc);
+    const int cbp_chroma = anyc_ac ? 2 : (anyc_dc ? 1 : 0);
+    if (lane < 2) {
+      int* q = &s_cdcq[lane * 4];
+      int dq[4] = {0, 0, 0, 0};
+      if (cbp_chroma >= 1) {
+        int w0 = q[0] + q[1] + q[2] + q[3];
+        int w1 = q[0] - q[1] + q[2] - q[3];
+        int w2 = q[0] + q[1] - q[2] - q[3];
+        int w3 = q[0] - q[1] - q[2] + q[3];
+        dq[0] = dequant_chroma_dc_v(w0, qpc);
+        dq[1] = dequant_chroma_dc_v(w1, qpc);
+        dq[2] = dequant_chroma_dc_v(w2, qpc);
+        dq[3] = dequant_chroma_dc_v(w3, qpc);
+      }
+      for (int k = 0; k < 4; ++k) s_cdcrec[lane * 4 + k] = dq[k];
+    }
+    __syncthreads();
+
+    // chroma recon
+    for (int pass = 0; pass < 2; ++pass) {
+      int comp = pass;
+      int sub = g;
+      int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+      int rr = scy + (c >> 2), cc2 = scx + (c & 3);
+      int d = (c == 0) ? s_cdcrec[comp * 4 + sub]
+                       : (cbp_chroma == 2
+                              ? dequant_c(s_clvl[(comp * 4 + sub) * 16 + c],
+                                          qpc, coeff_cls(c))
+                              : 0);
+      int rec = idct4_wave(d, lane);
+      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
+      int pred;
+      if (chroma_mode == 1) {
+        pred = lc[rr];
+      } else {
+        pred = 128;
+        if (have_left) {
+          int scy4 = (rr >> 2) * 4;
+          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
+        }
+      }
+      uint8_t pix = clip8(rec + pred);
+      uint8_t* dp = comp ? curCr : curCb;
+      dp[(size_t)(cy0 + rr) * cpitch + cx0 + cc2] = pix;
+      if (cc2 == 7) {
+        if (comp) s_left_cr_new[rr] = pix; else s_left_cb_new[rr] = pix;
+      }
+    }
+    __syncthreads();
+    if (lane < 8) {
+      s_left_cb[lane] = s_left_cb_new[lane];
+      s_left_cr[lane] = s_left_cr_new[lane];
+    }
+
+    // meta out
+    if (lane == 0) {
+      M[0] = kIntra | (luma_mode << 2) | (chroma_mode << 5);
+      M[1] = 0;
+    }
+    __syncthreads();
+    have_left = true;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// P-frame mode decision: one workgroup (64 lanes) per MB of each P row.
+// Produces meta: skip / inter(mv) / intra. Thresholds match the CPU
+// encoder (cpu/h264/encoder.cpp).
+__global__ void __launch_bounds__(64) k_h264_me(
+    const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
+    const uint8_t* __restrict__ refY, int mbw, int frame_w_mb16,
+    const RowJob* __restrict__ jobs, int* __restrict__ meta) {
+  const int job_idx = blockIdx.x / mbw;
+  const int mbx = blockIdx.x % mbw;
+  const RowJob job = jobs[job_idx];
+  if (job.flags & 1) return;  // I rows have no ME
+  const int lane = threadIdx.x;
+  const int mby = job.mb_row;
+  const int x0 = mbx * 16, y0 = mby * 16;
+  const size_t mb_index = (size_t)mby * mbw + mbx;
+
+  const int r = lane >> 2, cq = (lane & 3) * 4;
+  // source pixels (clamped)
+  int sp[4];
+  {
+    const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+    sp[0] = s[min(x0 + cq + 0, w - 1)];
+    sp[1] = s[min(x0 + cq + 1, w - 1)];
+    sp[2] = s[min(x0 + cq + 2, w - 1)];
+    sp[3] = s[min(x0 + cq + 3, w - 1)];
+  }
+  auto sad_at = [&](int mx, int my) -> int {
+    const uint8_t* rp = refY + (size_t)(y0 + my + r) * ypitch + x0 + mx + cq;
+    int s = abs(sp[0] - rp[0]) + abs(sp[1] - rp[1]) + abs(sp[2] - rp[2]) +
+            abs(sp[3] - rp[3]);
+    return wave_sum_i(s);
+  };
+
+  const int qp = job.qp;
+  const int skip_thresh = 48 << (qp / 6);
+  const int inter_thresh = 2 * skip_thresh;
+
+  int sad0 = sad_at(0, 0);
+  int mode, bmx = 0, bmy = 0;
+  if (sad0 <= skip_thresh) {
+    mode = kSkip;
+  } else {
+    int best = sad0;
+    static const int pat[8][2] = {{-2, 0}, {2, 0},  {0, -2}, {0, 2},
+                                  {-2, -2}, {2, 2}, {-2, 2}, {2, -2}};
+    for (int iter = 0; iter < 8; ++iter) {
+      int cx = bmx, cy = bmy;
+      bool improved = false;
+      for (int p = 0; p < 8; ++p) {
+        int mx = cx + pat[p][0], my = cy + pat[p][1];
+        if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
+            y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1 ||
+            abs(mx) > 16 || abs(my) > 16)
+          continue;
+        int s = sad_at(mx, my);
+        if (s < best) {
+          best = s;
+          bmx = mx;
+          bmy = my;
+          improved = true;
+        }
+      }
+      if (!improved) break;
+    }
+    if (best <= inter_thresh && !(bmx == 0 && bmy == 0))
+      mode = kInter;
+    else
+      mode = kIntra;
+  }
+  if (lane == 0) {
+    meta[mb_index * kMetaPerMb + 0] = mode;
+    meta[mb_index * kMetaPerMb + 1] =
+        ((bmx * 4) & 0xFFFF) | ((bmy * 4) << 16);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers
+void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
+                    const uint8_t* refY, int mbw, int n_jobs,
+                    const RowJob* d_jobs, int* d_meta, hipStream_t stream) {
+  if (n_jobs == 0) return;
+  hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * mbw), dim3(64), 0, stream,
+                     srcY, ypitch, w, h, refY, mbw, mbw * 16, d_jobs, d_meta);
+}
+
+void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
+                      const uint8_t* srcCr, int ypitch, int cpitch, int w,
+                      int h, const uint8_t* refY, const uint8_t* refCb,
+                      const uint8_t* refCr, uint8_t* curY, uint8_t* curCb,
+                      uint8_t* curCr, int mbw, int n_jobs,
+                      const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
+                      hipStream_t stream) {
+  if (n_jobs == 0) return;
+  hipLaunchKernelGGL(k_h264_rows, dim3(n_jobs), dim3(64), 0, stream, srcY,
+                     srcCb, srcCr, ypitch, cpitch, w, h, refY, refCb, refCr,
+                     curY, curCb, curCr, mbw, d_jobs, d_levels, d_meta);
+}
+
+}  // namespace h264gpu
+}  // namespace hipflux

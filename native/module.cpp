@@ -225,6 +225,69 @@ PYBIND11_MODULE(_native, m) {
       py::arg("bgrx"), py::arg("width"), py::arg("height"));
 
   m.def(
+      "_pipeline_encode",
+      [](const std::string& kind, py::list frames, int w, int h, int qp,
+         int stripe_h, int output_mode) {
+        CaptureSettings s;
+        s.capture_width = w;
+        s.capture_height = h;
+        s.output_mode = output_mode;
+        s.stripe_height = stripe_h;
+        s.video_crf = qp;
+        s.jpeg_quality = qp;
+        s.use_cpu = kind == "cpu";
+        s.gpu_id = kind == "cpu" ? -1 : 0;
+        std::unique_ptr<EncodePipeline> p;
+        if (kind == "gpu") {
+          p = make_hip_pipeline(s);
+          if (!p) throw std::runtime_error("no HIP pipeline available");
+        } else {
+          p = output_mode == 1 ? make_cpu_h264_pipeline(s)
+                               : make_cpu_jpeg_pipeline(s);
+        }
+        py::list result;
+        std::vector<std::vector<uint8_t>> copies;
+        for (size_t i = 0; i < frames.size(); ++i) {
+          py::buffer buf = frames[i].cast<py::buffer>();
+          py::buffer_info info = buf.request();
+          if (info.size < static_cast<ssize_t>(w) * h * 4)
+            throw std::runtime_error("frame buffer too small");
+          RawFrame f;
+          f.data = static_cast<const uint8_t*>(info.ptr);
+          f.width = w;
+          f.height = h;
+          f.stride = w * 4;
+          f.ts_ms = now_ms();
+          FrameContext ctx;
+          ctx.frame_id = static_cast<uint32_t>(i);
+          ctx.idr = (i == 0);
+          ctx.crf = qp;
+          ctx.jpeg_quality = qp;
+          for (int y = 0; y < h; y += stripe_h) {
+            StripeJob j;
+            j.y0 = y;
+            j.y1 = std::min(y + stripe_h, h);
+            j.encode = true;
+            ctx.stripes.push_back(j);
+          }
+          py::list frame_out;
+          {
+            p->encode_frame(f, ctx, [&](EncodedStripe& st) {
+              frame_out.append(py::make_tuple(
+                  py::bytes(reinterpret_cast<const char*>(st.data), st.size),
+                  st.y, st.height, st.is_keyframe));
+            });
+          }
+          result.append(frame_out);
+        }
+        return result;
+      },
+      py::arg("kind"), py::arg("frames"), py::arg("w"), py::arg("h"),
+      py::arg("qp") = 26, py::arg("stripe_h") = 64,
+      py::arg("output_mode") = 1,
+      "Test hook: run frames through a named encode pipeline.");
+
+  m.def(
       "_cavlc_bits",
       [](std::vector<int> zz, int nC) {
         h264::BitWriter bw;

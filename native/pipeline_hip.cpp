@@ -5,11 +5,14 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstring>
 #include <map>
 #include <stdexcept>
 
+#include "cpu/h264/gpu_entropy.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
+#include "hip/h264_kernels.h"
 #include "hip/jpeg_kernels.h"
 #include "thread_pool.h"
 
@@ -206,6 +209,248 @@ class HipJpegPipeline : public EncodePipeline {
   std::map<void*, bool> registered_;
 };
 
+// ---------------------------------------------------------------------------
+// HIP H.264 pipeline: CSC + per-MB-row intra/inter kernels on the GPU,
+// stripe-parallel CAVLC packing on the CPU. Each stripe is an independent
+// bitstream with its own frame_num / recon region (SURVEY.md §5.7 seam).
+class HipH264Pipeline : public EncodePipeline {
+ public:
+  explicit HipH264Pipeline(const CaptureSettings& s)
+      : settings_(s),
+        pool_(std::max(2u, std::thread::hardware_concurrency() / 2)) {
+    HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    stripe_h_ = std::max(16, s.stripe_height & ~15);
+    alloc_for(s.capture_width, s.capture_height);
+  }
+
+  ~HipH264Pipeline() override {
+    (void)hipStreamSynchronize(stream_);
+    for (auto& kv : registered_) (void)hipHostUnregister(kv.first);
+    for (void* p : device_ptrs_)
+      if (p) (void)hipFree(p);
+    if (h_levels_) (void)hipHostFree(h_levels_);
+    if (h_meta_) (void)hipHostFree(h_meta_);
+    if (h_jobs_) (void)hipHostFree(h_jobs_);
+  }
+
+  void encode_frame(const RawFrame& frame, const FrameContext& ctx,
+                    const Emit& emit) override {
+    if (frame.width != w_ || frame.height != h_)
+      alloc_for(frame.width, frame.height);
+    const int qp = std::min(51, std::max(0, ctx.crf));
+
+    // upload + CSC
+    const uint8_t* src = frame.data;
+    size_t frame_bytes = static_cast<size_t>(frame.stride) * frame.height;
+    if (!registered_.count(const_cast<uint8_t*>(src))) {
+      hipError_t e = hipHostRegister(const_cast<uint8_t*>(src), frame_bytes,
+                                     hipHostRegisterDefault);
+      registered_[const_cast<uint8_t*>(src)] = (e == hipSuccess);
+      if (e != hipSuccess) (void)hipGetLastError();
+    }
+    HIP_CHECK(hipMemcpyAsync(d_frame_, src, frame_bytes,
+                             hipMemcpyHostToDevice, stream_));
+    launch_bgrx_to_planes(d_frame_, w_, h_, frame.stride / 4, d_srcY_,
+                          d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
+                          stream_);
+
+    // build row jobs for scheduled stripes
+    struct SJob {
+      int idx;       // stripe index
+      int y0, y1;    // pixel bounds (logical)
+      bool idr;
+    };
+    std::vector<SJob> sjobs;
+    int n_jobs = 0;
+    for (size_t i = 0; i < ctx.stripes.size(); ++i) {
+      const auto& st = ctx.stripes[i];
+      if (!st.encode) continue;
+      auto& state = stripes_[st.y0 / stripe_h_];
+      bool idr = ctx.idr || state.need_idr;
+      sjobs.push_back({static_cast<int>(st.y0 / stripe_h_), st.y0, st.y1,
+                       idr});
+      int row0 = st.y0 / 16;
+      int rows = (std::min(st.y1, mbh_ * 16) - st.y0 + 15) / 16;
+      // stripe's padded pixel bounds for ME clamping
+      int sy0 = st.y0, sy1 = (row0 + rows) * 16;
+      for (int r = 0; r < rows; ++r) {
+        h_jobs_[n_jobs].mb_row = row0 + r;
+        h_jobs_[n_jobs].qp = qp;
+        h_jobs_[n_jobs].flags = idr ? 1 : 0;
+        h_jobs_[n_jobs].stripe_y0 = sy0;
+        h_jobs_[n_jobs].stripe_y1 = sy1;
+        ++n_jobs;
+      }
+    }
+    if (n_jobs == 0) return;
+
+    HIP_CHECK(hipMemcpyAsync(d_jobs_, h_jobs_,
+                             sizeof(h264gpu::RowJob) * n_jobs,
+                             hipMemcpyHostToDevice, stream_));
+    h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, mbw_, n_jobs,
+                            d_jobs_, d_meta_, stream_);
+    h264gpu::launch_h264_rows(d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_,
+                              w_, h_, d_refY_, d_refCb_, d_refCr_, d_curY_,
+                              d_curCb_, d_curCr_, mbw_, n_jobs, d_jobs_,
+                              d_levels_, d_meta_, stream_);
+
+    // readback levels + meta, then refresh ref regions for encoded stripes
+    size_t level_bytes = static_cast<size_t>(mbw_) * mbh_ *
+                         h264gpu::kLevelsPerMb * sizeof(int16_t);
+    size_t meta_bytes = static_cast<size_t>(mbw_) * mbh_ *
+                        h264gpu::kMetaPerMb * sizeof(int);
+    HIP_CHECK(hipMemcpyAsync(h_levels_, d_levels_, level_bytes,
+                             hipMemcpyDeviceToHost, stream_));
+    HIP_CHECK(hipMemcpyAsync(h_meta_, d_meta_, meta_bytes,
+                             hipMemcpyDeviceToHost, stream_));
+    for (const auto& sj : sjobs) {
+      int row0 = sj.y0 / 16;
+      int rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
+      copy_region(d_refY_, d_curY_, ypitch_, row0 * 16, rows * 16);
+      copy_region(d_refCb_, d_curCb_, cpitch_, row0 * 8, rows * 8);
+      copy_region(d_refCr_, d_curCr_, cpitch_, row0 * 8, rows * 8);
+    }
+    HIP_CHECK(hipStreamSynchronize(stream_));
+
+    // stripe-parallel CPU entropy
+    struct Out {
+      std::vector<uint8_t> bytes;
+      int y0 = 0, h = 0;
+      bool idr = false;
+    };
+    std::vector<Out> outs(sjobs.size());
+    for (size_t i = 0; i < sjobs.size(); ++i) {
+      const auto& sj = sjobs[i];
+      auto& state = stripes_[sj.idx];
+      outs[i].y0 = sj.y0;
+      outs[i].h = std::min(sj.y1, h_) - sj.y0;
+      outs[i].idr = sj.idr;
+      if (sj.idr) {
+        state.frame_num = 0;
+        ++state.idr_pic_id;
+        state.need_idr = false;
+      }
+      h264::GpuStripeParams p;
+      p.levels = h_levels_;
+      p.meta = h_meta_;
+      p.mbw = mbw_;
+      p.mb_row0 = sj.y0 / 16;
+      p.n_mb_rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
+      p.width = w_;
+      p.height = outs[i].h;
+      p.qp = qp;
+      p.idr = sj.idr;
+      p.frame_num = state.frame_num;
+      p.idr_pic_id = state.idr_pic_id;
+      pool_.submit([&outs, i, p] {
+        h264::encode_stripe_from_gpu(p, outs[i].bytes);
+      });
+      ++state.frame_num;
+    }
+    pool_.wait_all();
+    for (auto& o : outs) {
+      if (o.bytes.empty()) continue;
+      EncodedStripe s;
+      s.type = StripeType::kH264;
+      s.data = o.bytes.data();
+      s.size = o.bytes.size();
+      s.frame_id = ctx.frame_id;
+      s.y = o.y0;
+      s.width = w_;
+      s.height = o.h;
+      s.is_keyframe = o.idr;
+      emit(s);
+    }
+  }
+
+  const char* name() const override { return "hip-h264"; }
+
+ private:
+  struct StripeState {
+    uint32_t frame_num = 0;
+    uint32_t idr_pic_id = 0;
+    bool need_idr = true;
+  };
+
+  void copy_region(uint8_t* dst, const uint8_t* src, int pitch, int y0,
+                   int rows) {
+    HIP_CHECK(hipMemcpyAsync(dst + static_cast<size_t>(y0) * pitch,
+                             src + static_cast<size_t>(y0) * pitch,
+                             static_cast<size_t>(rows) * pitch,
+                             hipMemcpyDeviceToDevice, stream_));
+  }
+
+  void alloc_for(int w, int h) {
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    for (void* p : device_ptrs_)
+      if (p) (void)hipFree(p);
+    device_ptrs_.clear();
+    if (h_levels_) (void)hipHostFree(h_levels_);
+    if (h_meta_) (void)hipHostFree(h_meta_);
+    if (h_jobs_) (void)hipHostFree(h_jobs_);
+    w_ = w;
+    h_ = h;
+    mbw_ = (w + 15) / 16;
+    mbh_ = (h + 15) / 16;
+    ypitch_ = mbw_ * 16;
+    cpitch_ = mbw_ * 8;
+    size_t ysz = static_cast<size_t>(ypitch_) * mbh_ * 16;
+    size_t csz = static_cast<size_t>(cpitch_) * mbh_ * 8;
+    auto dalloc = [&](size_t bytes) {
+      void* p = nullptr;
+      HIP_CHECK(hipMalloc(&p, bytes));
+      device_ptrs_.push_back(p);
+      return static_cast<uint8_t*>(p);
+    };
+    d_frame_ = dalloc(static_cast<size_t>(w) * h * 4);
+    d_srcY_ = dalloc(ysz);
+    d_srcCb_ = dalloc(csz);
+    d_srcCr_ = dalloc(csz);
+    d_refY_ = dalloc(ysz);
+    d_refCb_ = dalloc(csz);
+    d_refCr_ = dalloc(csz);
+    d_curY_ = dalloc(ysz);
+    d_curCb_ = dalloc(csz);
+    d_curCr_ = dalloc(csz);
+    size_t level_bytes = static_cast<size_t>(mbw_) * mbh_ *
+                         h264gpu::kLevelsPerMb * sizeof(int16_t);
+    size_t meta_bytes =
+        static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb * sizeof(int);
+    d_levels_ = reinterpret_cast<int16_t*>(dalloc(level_bytes));
+    d_meta_ = reinterpret_cast<int*>(dalloc(meta_bytes));
+    d_jobs_ = reinterpret_cast<h264gpu::RowJob*>(
+        dalloc(sizeof(h264gpu::RowJob) * mbh_));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_levels_), level_bytes,
+                            hipHostMallocDefault));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_meta_), meta_bytes,
+                            hipHostMallocDefault));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_jobs_),
+                            sizeof(h264gpu::RowJob) * mbh_,
+                            hipHostMallocDefault));
+    stripes_.assign((h + stripe_h_ - 1) / stripe_h_, StripeState{});
+  }
+
+  CaptureSettings settings_;
+  ThreadPool pool_;
+  hipStream_t stream_{};
+  int stripe_h_ = 64;
+  int w_ = 0, h_ = 0, mbw_ = 0, mbh_ = 0, ypitch_ = 0, cpitch_ = 0;
+  uint8_t *d_frame_ = nullptr, *d_srcY_ = nullptr, *d_srcCb_ = nullptr,
+          *d_srcCr_ = nullptr, *d_refY_ = nullptr, *d_refCb_ = nullptr,
+          *d_refCr_ = nullptr, *d_curY_ = nullptr, *d_curCb_ = nullptr,
+          *d_curCr_ = nullptr;
+  int16_t* d_levels_ = nullptr;
+  int* d_meta_ = nullptr;
+  h264gpu::RowJob* d_jobs_ = nullptr;
+  int16_t* h_levels_ = nullptr;
+  int* h_meta_ = nullptr;
+  h264gpu::RowJob* h_jobs_ = nullptr;
+  std::vector<void*> device_ptrs_;
+  std::vector<StripeState> stripes_;
+  std::map<void*, bool> registered_;
+};
+
 }  // namespace
 
 std::unique_ptr<EncodePipeline> make_hip_pipeline(const CaptureSettings& s) {
@@ -214,8 +459,7 @@ std::unique_ptr<EncodePipeline> make_hip_pipeline(const CaptureSettings& s) {
   if (s.gpu_id >= n) return nullptr;
   try {
     if (s.output_mode == 0) return std::make_unique<HipJpegPipeline>(s);
-    // H.264 HIP pipeline lands next; JPEG covers output_mode 0 only.
-    return nullptr;
+    return std::make_unique<HipH264Pipeline>(s);
   } catch (const std::exception& e) {
     std::fprintf(stderr, "hipflux: HIP pipeline init failed: %s\n", e.what());
     return nullptr;
