@@ -287,6 +287,77 @@ PYBIND11_MODULE(_native, m) {
       py::arg("output_mode") = 1,
       "Test hook: run frames through a named encode pipeline.");
 
+  // ---- persistent pipeline handle for benchmarking ------------------------
+  struct BenchPipeline {
+    std::unique_ptr<EncodePipeline> p;
+    int w, h, qp, stripe_h;
+    uint32_t frame_id = 0;
+    BenchPipeline(const std::string& kind, int width, int height, int qp_,
+                  int stripe, int output_mode, int gpu_id)
+        : w(width), h(height), qp(qp_), stripe_h(stripe) {
+      CaptureSettings s;
+      s.capture_width = w;
+      s.capture_height = h;
+      s.output_mode = output_mode;
+      s.stripe_height = stripe_h;
+      s.video_crf = qp;
+      s.jpeg_quality = qp;
+      s.gpu_id = gpu_id;
+      s.use_cpu = kind == "cpu";
+      if (kind == "gpu") {
+        p = make_hip_pipeline(s);
+        if (!p) throw std::runtime_error("no HIP pipeline available");
+      } else {
+        p = output_mode == 1 ? make_cpu_h264_pipeline(s)
+                             : make_cpu_jpeg_pipeline(s);
+      }
+    }
+  };
+  py::class_<BenchPipeline>(m, "BenchPipeline")
+      .def(py::init<const std::string&, int, int, int, int, int, int>(),
+           py::arg("kind"), py::arg("width"), py::arg("height"),
+           py::arg("qp") = 28, py::arg("stripe_height") = 64,
+           py::arg("output_mode") = 1, py::arg("gpu_id") = 0)
+      .def_property_readonly("pipeline",
+                             [](BenchPipeline& b) { return b.p->name(); })
+      .def(
+          "encode",
+          [](BenchPipeline& b, py::buffer bgrx, bool idr) {
+            py::buffer_info info = bgrx.request();
+            if (info.size < static_cast<ssize_t>(b.w) * b.h * 4)
+              throw std::runtime_error("frame buffer too small");
+            size_t total = 0;
+            int stripes = 0;
+            {
+              py::gil_scoped_release rel;
+              RawFrame f;
+              f.data = static_cast<const uint8_t*>(info.ptr);
+              f.width = b.w;
+              f.height = b.h;
+              f.stride = b.w * 4;
+              f.ts_ms = now_ms();
+              FrameContext ctx;
+              ctx.frame_id = b.frame_id++;
+              ctx.idr = idr;
+              ctx.crf = b.qp;
+              ctx.jpeg_quality = b.qp;
+              for (int y = 0; y < b.h; y += b.stripe_h) {
+                StripeJob j;
+                j.y0 = y;
+                j.y1 = std::min(y + b.stripe_h, b.h);
+                j.encode = true;
+                ctx.stripes.push_back(j);
+              }
+              b.p->encode_frame(f, ctx, [&](EncodedStripe& st) {
+                total += st.size;
+                ++stripes;
+              });
+            }
+            return py::make_tuple(total, stripes);
+          },
+          py::arg("bgrx"), py::arg("idr") = false,
+          "Encode one frame; returns (bitstream_bytes, stripes).");
+
   m.def(
       "_cavlc_bits",
       [](std::vector<int> zz, int nC) {

