@@ -97,8 +97,11 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("display", str, ":0", "X DISPLAY to capture/inject into.",
                fallback_env=("DISPLAY",)),
     SettingDef("capture_backend", str, "auto",
-               "Framebuffer source: auto (x11 if DISPLAY reachable else synthetic).",
-               allowed=("auto", "x11", "synthetic")),
+               "Framebuffer source: auto (x11 if DISPLAY reachable else "
+               "synthetic), x11, synthetic, or synthetic:<pattern>.",
+               normalize=lambda v: v if (
+                   v in ("auto", "x11", "synthetic")
+                   or str(v).startswith("synthetic:")) else "auto"),
     SettingDef("resolution", str, "1920x1080", "Initial capture WxH.", client=True),
     SettingDef("framerate", int, 60, "Target capture/encode fps.",
                value_range=(1, 240), client=True),

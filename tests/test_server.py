@@ -1,0 +1,163 @@
+"""End-to-end WS server test: live aiohttp server + websocket client
+receiving real encoded stripes, control verbs, input round-trip, auth,
+API surface. (Behavioral equivalent of the reference's integration
+test_protocol.py tier — SURVEY.md §4.)"""
+
+import asyncio
+import json
+
+import pytest
+
+hipflux = pytest.importorskip("hipflux")
+if not hipflux.native_available():
+    pytest.skip("hipflux native module not built", allow_module_level=True)
+
+import aiohttp
+from aiohttp import WSMsgType
+
+from selkies_amd.settings import load_settings
+from selkies_amd.stream_server import CentralizedStreamServer
+from selkies_amd.input_handler import RecordingBackend
+
+
+def make_server(**env):
+    base_env = {
+        "SELKIES_PORT": "0",
+        "SELKIES_CAPTURE_BACKEND": "synthetic:desktop",
+        "SELKIES_RESOLUTION": "320x192",
+        "SELKIES_FRAMERATE": "30",
+        "SELKIES_USE_CPU": "true",
+        "SELKIES_ENCODER": "h264enc-striped",
+    }
+    base_env.update(env)
+    settings = load_settings(argv=[], env=base_env)
+    server = CentralizedStreamServer(settings)
+    # force deterministic input backend
+    server.streaming.input.backend = RecordingBackend()
+    return server
+
+
+async def start_on_free_port(server):
+    from aiohttp import web
+    runner = web.AppRunner(server.app)
+    await runner.setup()
+    site = web.TCPSite(runner, "127.0.0.1", 0)
+    await site.start()
+    port = site._server.sockets[0].getsockname()[1]
+    return runner, port
+
+
+@pytest.fixture()
+def loop():
+    loop = asyncio.new_event_loop()
+    yield loop
+    loop.close()
+
+
+def test_video_flows_and_control(loop):
+    async def main():
+        server = make_server()
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                # API surface
+                async with sess.get(f"http://127.0.0.1:{port}/api/health") as r:
+                    assert (await r.json())["ok"]
+                async with sess.get(f"http://127.0.0.1:{port}/api/status") as r:
+                    st = await r.json()
+                    assert st["mode"] == "websockets"
+
+                async with sess.ws_connect(
+                        f"http://127.0.0.1:{port}/websockets") as ws:
+                    texts, frames = [], []
+                    deadline = asyncio.get_event_loop().time() + 8
+                    while asyncio.get_event_loop().time() < deadline:
+                        msg = await ws.receive(timeout=8)
+                        if msg.type == WSMsgType.TEXT:
+                            texts.append(msg.data)
+                        elif msg.type == WSMsgType.BINARY:
+                            frames.append(msg.data)
+                            if len(frames) >= 24:
+                                break
+                        else:
+                            break
+                    # handshake pushes
+                    assert any(t.startswith("MODE,") for t in texts)
+                    assert any(t.startswith("SETTINGS_PAYLOAD,") for t in texts)
+                    # the liveness floor the reference e2e asserts (>=24
+                    # binary frames)
+                    assert len(frames) >= 24
+                    assert frames[0][0] == 0x04  # H.264 stripe tag
+                    # ACK a frame id, send input, change a setting
+                    await ws.send_str("CLIENT_FRAME_ACK,0")
+                    await ws.send_str("kd,65")
+                    await ws.send_str("ku,65")
+                    await ws.send_str("m,10,20,0")
+                    await ws.send_str('SETTINGS,{"framerate": 15}')
+                    # wait for settings echo
+                    deadline = asyncio.get_event_loop().time() + 5
+                    got_echo = False
+                    while asyncio.get_event_loop().time() < deadline:
+                        msg = await ws.receive(timeout=5)
+                        if msg.type == WSMsgType.TEXT and \
+                                msg.data.startswith("SETTINGS_PAYLOAD,"):
+                            payload = json.loads(
+                                msg.data.split(",", 1)[1])
+                            assert payload["framerate"]["value"] == 15
+                            got_echo = True
+                            break
+                    assert got_echo
+                ev = server.streaming.input.backend.events
+                assert ("key", 65, True) in ev and ("key", 65, False) in ev
+                assert ("move", 10, 20) in ev
+        finally:
+            server.streaming.stop_capture()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_auth_token(loop):
+    async def main():
+        server = make_server(SELKIES_AUTH_TOKEN="sekrit")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.get(f"http://127.0.0.1:{port}/api/status") as r:
+                    assert r.status == 401
+                async with sess.get(
+                        f"http://127.0.0.1:{port}/api/status",
+                        headers={"Authorization": "Bearer sekrit"}) as r:
+                    assert r.status == 200
+                # health is always open
+                async with sess.get(f"http://127.0.0.1:{port}/api/health") as r:
+                    assert r.status == 200
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_jpeg_mode_stripes(loop):
+    async def main():
+        server = make_server(SELKIES_ENCODER="jpeg")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"http://127.0.0.1:{port}/ws") as ws:
+                    deadline = asyncio.get_event_loop().time() + 8
+                    got = None
+                    while asyncio.get_event_loop().time() < deadline:
+                        msg = await ws.receive(timeout=8)
+                        if msg.type == WSMsgType.BINARY:
+                            got = msg.data
+                            break
+                    assert got is not None and got[0] == 0x03
+                    # payload after the 6-byte header is a JFIF image
+                    assert got[6:8] == b"\xff\xd8"
+        finally:
+            server.streaming.stop_capture()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
