@@ -5,6 +5,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "audio.h"
 #include "cpu/damage.h"
 #include "cpu/h264/cavlc.h"
 #include "cpu/h264/encoder.h"
@@ -375,6 +376,72 @@ PYBIND11_MODULE(_native, m) {
         return s;
       },
       "debug: CAVLC-encode one zigzag block, return bit string");
+
+  // ---- audio (pcmflux contract) -------------------------------------------
+  py::class_<AudioCaptureSettings>(m, "AudioCaptureSettings")
+      .def(py::init<>())
+      .def_readwrite("device_name", &AudioCaptureSettings::device_name)
+      .def_readwrite("sample_rate", &AudioCaptureSettings::sample_rate)
+      .def_readwrite("channels", &AudioCaptureSettings::channels)
+      .def_readwrite("opus_bitrate", &AudioCaptureSettings::opus_bitrate)
+      .def_readwrite("frame_duration_ms",
+                     &AudioCaptureSettings::frame_duration_ms)
+      .def_readwrite("red_distance", &AudioCaptureSettings::red_distance)
+      .def_readwrite("omit_audio_header",
+                     &AudioCaptureSettings::omit_audio_header)
+      .def_readwrite("debug_logging", &AudioCaptureSettings::debug_logging);
+
+  py::class_<AudioCapture>(m, "AudioCapture")
+      .def(py::init<>())
+      .def(
+          "start_capture",
+          [](AudioCapture& self, const AudioCaptureSettings& s,
+             py::function cb) {
+            {
+              py::gil_scoped_release rel;
+              self.stop_capture();
+            }
+            self.clear_callback();
+            AudioCapture::Callback native = [cb](const AudioFrame& f) {
+              py::gil_scoped_acquire gil;
+              try {
+                cb(py::bytes(reinterpret_cast<const char*>(f.data), f.size),
+                   f.pts_ms);
+              } catch (py::error_already_set& e) {
+                e.discard_as_unraisable("hipflux audio callback");
+              }
+            };
+            py::gil_scoped_release rel;
+            self.start_capture(s, std::move(native));
+          },
+          py::arg("settings"), py::arg("callback"))
+      .def("stop_capture", &AudioCapture::stop_capture,
+           py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("is_capturing", &AudioCapture::is_capturing)
+      .def("update_audio_bitrate", &AudioCapture::update_audio_bitrate);
+
+  py::class_<AudioPlaybackSettings>(m, "AudioPlaybackSettings")
+      .def(py::init<>())
+      .def_readwrite("sample_rate", &AudioPlaybackSettings::sample_rate)
+      .def_readwrite("channels", &AudioPlaybackSettings::channels)
+      .def_readwrite("max_buffer_bytes",
+                     &AudioPlaybackSettings::max_buffer_bytes);
+
+  py::class_<AudioPlayback>(m, "AudioPlayback")
+      .def(py::init<const AudioPlaybackSettings&>())
+      .def("write",
+           [](AudioPlayback& self, py::bytes data) {
+             std::string s = data;
+             return self.write(
+                 reinterpret_cast<const uint8_t*>(s.data()), s.size());
+           })
+      .def("read",
+           [](AudioPlayback& self, size_t n) {
+             std::vector<uint8_t> out(n);
+             size_t got = self.read(out.data(), n);
+             return py::bytes(reinterpret_cast<char*>(out.data()), got);
+           })
+      .def_property_readonly("buffered", &AudioPlayback::buffered);
 
   m.def("hip_device_count", &hip_device_count,
         "Number of usable HIP devices (0 on CPU-only hosts).");

@@ -1,0 +1,279 @@
+/* selkies_amd HTML5 client.
+ *
+ * Speaks the binary stripe wire protocol (SURVEY.md §3.2):
+ *   0x03 JPEG stripe  [tag, flags, frame_id u16, y u16] + JFIF
+ *   0x04 H.264 stripe [tag, key, frame_id u16, y u16, w u16, h u16] + AnnexB
+ * H.264 stripes feed one WebCodecs VideoDecoder PER STRIPE ROW (each row is
+ * an independent bitstream); decoded rows composite onto the canvas.
+ * JPEG stripes decode via createImageBitmap. Input is captured and sent as
+ * the text verbs the server's InputDispatcher understands.
+ */
+"use strict";
+
+const canvas = document.getElementById("screen");
+const ctx2d = canvas.getContext("2d");
+const stateEl = document.getElementById("state");
+const statsEl = document.getElementById("stats");
+
+let ws = null;
+let serverSettings = {};
+let frameCount = 0, byteCount = 0, lastStats = performance.now();
+let lastAckedFrame = -1;
+
+/* ---------------- decode sinks ---------------- */
+
+class H264Row {
+  constructor(y) {
+    this.y = y;
+    this.decoder = null;
+    this.width = 0;
+    this.height = 0;
+  }
+  configure(w, h) {
+    if (this.decoder && this.width === w && this.height === h) return;
+    if (this.decoder) try { this.decoder.close(); } catch (e) {}
+    this.width = w; this.height = h;
+    this.decoder = new VideoDecoder({
+      output: (frame) => {
+        ctx2d.drawImage(frame, 0, this.y);
+        frame.close();
+      },
+      error: (e) => { console.warn("decoder error row", this.y, e);
+                      this.decoder = null; requestIdr(); },
+    });
+    this.decoder.configure({
+      codec: "avc1.42e028",
+      optimizeForLatency: true,
+    });
+  }
+  push(payload, key, frameId) {
+    if (!this.decoder) return;
+    try {
+      this.decoder.decode(new EncodedVideoChunk({
+        type: key ? "key" : "delta",
+        timestamp: frameId * 16667,
+        data: payload,
+      }));
+    } catch (e) { console.warn(e); requestIdr(); }
+  }
+}
+
+const h264Rows = new Map();   // y -> H264Row
+let jpegDrawQueue = Promise.resolve();
+
+function ensureCanvas(w, hTotal) {
+  if (canvas.width !== w) canvas.width = w;
+  if (hTotal > canvas.height) canvas.height = hTotal;
+}
+
+function onBinary(buf) {
+  const d = new Uint8Array(buf);
+  byteCount += d.length;
+  const tag = d[0];
+  if (tag === 0x04) {
+    const key = d[1] === 1;
+    const frameId = (d[2] << 8) | d[3];
+    const y = (d[4] << 8) | d[5];
+    const w = (d[6] << 8) | d[7];
+    const h = (d[8] << 8) | d[9];
+    ensureCanvas(w, y + h);
+    let row = h264Rows.get(y);
+    if (!row) { row = new H264Row(y); h264Rows.set(y, row); }
+    if (key) row.configure(w, h);
+    row.push(d.subarray(10), key, frameId);
+    noteFrame(frameId);
+  } else if (tag === 0x03) {
+    const frameId = (d[2] << 8) | d[3];
+    const y = (d[4] << 8) | d[5];
+    const blob = new Blob([d.subarray(6)], { type: "image/jpeg" });
+    jpegDrawQueue = jpegDrawQueue.then(async () => {
+      try {
+        const img = await createImageBitmap(blob);
+        ensureCanvas(img.width, y + img.height);
+        ctx2d.drawImage(img, 0, y);
+        img.close();
+      } catch (e) { console.warn("jpeg decode", e); }
+    });
+    noteFrame(frameId);
+  } else if (tag === 0x01) {
+    /* audio lands with the audio engine */
+  } else if (tag === 0x05) {
+    /* gzip'd control text: rare server->client path; ignore for now */
+  }
+}
+
+function noteFrame(frameId) {
+  frameCount++;
+  if (frameId !== lastAckedFrame && ws && ws.readyState === 1) {
+    lastAckedFrame = frameId;
+    ws.send("CLIENT_FRAME_ACK," + frameId);
+  }
+}
+
+function requestIdr() {
+  if (ws && ws.readyState === 1) ws.send("REQUEST_IDR,");
+}
+
+/* ---------------- input capture ---------------- */
+
+/* Browser key -> X11 keysym (reference keysym wire contract). Printable
+ * characters map via unicode (latin1 direct, others 0x01000000+cp);
+ * specials via this table (X11/keysymdef.h values). */
+const KEYSYMS = {
+  Backspace: 0xff08, Tab: 0xff09, Enter: 0xff0d, Escape: 0xff1b,
+  Delete: 0xffff, Home: 0xff50, End: 0xff57, PageUp: 0xff55,
+  PageDown: 0xff56, ArrowLeft: 0xff51, ArrowUp: 0xff52,
+  ArrowRight: 0xff53, ArrowDown: 0xff54, Insert: 0xff63,
+  ShiftLeft: 0xffe1, ShiftRight: 0xffe2, ControlLeft: 0xffe3,
+  ControlRight: 0xffe4, AltLeft: 0xffe9, AltRight: 0xffea,
+  MetaLeft: 0xffeb, MetaRight: 0xffec, CapsLock: 0xffe5,
+  F1: 0xffbe, F2: 0xffbf, F3: 0xffc0, F4: 0xffc1, F5: 0xffc2,
+  F6: 0xffc3, F7: 0xffc4, F8: 0xffc5, F9: 0xffc6, F10: 0xffc7,
+  F11: 0xffc8, F12: 0xffc9, Space: 0x20,
+};
+
+function keysymOf(ev) {
+  if (KEYSYMS[ev.code] !== undefined && ev.key.length !== 1)
+    return KEYSYMS[ev.code];
+  if (ev.key.length === 1) {
+    const cp = ev.key.codePointAt(0);
+    return cp < 0x100 ? cp : 0x01000000 + cp;
+  }
+  return KEYSYMS[ev.code];
+}
+
+let buttonMask = 0;
+
+function canvasPos(ev) {
+  const r = canvas.getBoundingClientRect();
+  const x = Math.round((ev.clientX - r.left) * canvas.width / r.width);
+  const y = Math.round((ev.clientY - r.top) * canvas.height / r.height);
+  return [Math.max(0, Math.min(canvas.width - 1, x)),
+          Math.max(0, Math.min(canvas.height - 1, y))];
+}
+
+function send(msg) { if (ws && ws.readyState === 1) ws.send(msg); }
+
+function hookInput() {
+  canvas.addEventListener("keydown", (ev) => {
+    const ks = keysymOf(ev);
+    if (ks !== undefined) { send("kd," + ks); ev.preventDefault(); }
+  });
+  canvas.addEventListener("keyup", (ev) => {
+    const ks = keysymOf(ev);
+    if (ks !== undefined) { send("ku," + ks); ev.preventDefault(); }
+  });
+  canvas.addEventListener("mousemove", (ev) => {
+    if (document.pointerLockElement === canvas) {
+      send(`m2,${ev.movementX},${ev.movementY},${buttonMask}`);
+    } else {
+      const [x, y] = canvasPos(ev);
+      send(`m,${x},${y},${buttonMask}`);
+    }
+  });
+  const maskBit = (b) => b === 0 ? 1 : b === 1 ? 2 : b === 2 ? 4 : 0;
+  canvas.addEventListener("mousedown", (ev) => {
+    canvas.focus();
+    buttonMask |= maskBit(ev.button);
+    const [x, y] = canvasPos(ev);
+    send(`m,${x},${y},${buttonMask}`);
+    ev.preventDefault();
+  });
+  canvas.addEventListener("mouseup", (ev) => {
+    buttonMask &= ~maskBit(ev.button);
+    const [x, y] = canvasPos(ev);
+    send(`m,${x},${y},${buttonMask}`);
+  });
+  canvas.addEventListener("wheel", (ev) => {
+    send("sw," + (ev.deltaY < 0 ? "u" : "d") + ",1");
+    ev.preventDefault();
+  }, { passive: false });
+  canvas.addEventListener("contextmenu", (ev) => ev.preventDefault());
+  window.addEventListener("blur", () => send("kr,"));
+}
+
+/* ---------------- settings UI ---------------- */
+
+function hookHud() {
+  const enc = document.getElementById("encoder");
+  const fps = document.getElementById("fps");
+  const crf = document.getElementById("crf");
+  enc.onchange = () => send('SETTINGS,' +
+      JSON.stringify({ encoder: enc.value }));
+  fps.onchange = () => send('SETTINGS,' +
+      JSON.stringify({ framerate: +fps.value }));
+  crf.onchange = () => send('SETTINGS,' +
+      JSON.stringify({ video_crf: +crf.value }));
+  document.getElementById("idr").onclick = requestIdr;
+}
+
+function applyServerSettings(payload) {
+  serverSettings = payload;
+  const set = (id, name) => {
+    if (payload[name]) document.getElementById(id).value =
+        payload[name].value;
+  };
+  set("encoder", "encoder");
+  set("fps", "framerate");
+  set("crf", "video_crf");
+  const res = payload.resolution && payload.resolution.value;
+  if (res) {
+    const [w, h] = res.split("x").map(Number);
+    canvas.width = w; canvas.height = h;
+  }
+}
+
+/* ---------------- transport ---------------- */
+
+function connect() {
+  const proto = location.protocol === "https:" ? "wss" : "ws";
+  ws = new WebSocket(`${proto}://${location.host}/websockets` +
+                     location.search);
+  ws.binaryType = "arraybuffer";
+  ws.onopen = () => { stateEl.textContent = "connected"; };
+  ws.onclose = () => {
+    stateEl.textContent = "reconnecting…";
+    for (const row of h264Rows.values())
+      if (row.decoder) try { row.decoder.close(); } catch (e) {}
+    h264Rows.clear();
+    setTimeout(connect, 1000);
+  };
+  ws.onmessage = (ev) => {
+    if (typeof ev.data === "string") {
+      const i = ev.data.indexOf(",");
+      const verb = i < 0 ? ev.data : ev.data.slice(0, i);
+      const rest = i < 0 ? "" : ev.data.slice(i + 1);
+      if (verb === "SETTINGS_PAYLOAD") {
+        try { applyServerSettings(JSON.parse(rest)); } catch (e) {}
+      } else if (verb === "clipboard") {
+        try { navigator.clipboard.writeText(atob(rest)); } catch (e) {}
+      }
+    } else {
+      onBinary(ev.data);
+    }
+  };
+}
+
+setInterval(() => {
+  const now = performance.now();
+  const dt = (now - lastStats) / 1000;
+  statsEl.textContent =
+      `stripes/s ${(frameCount / dt).toFixed(0)}  ` +
+      `mbps ${(byteCount * 8 / dt / 1e6).toFixed(2)}`;
+  frameCount = 0; byteCount = 0; lastStats = now;
+}, 2000);
+
+/* resize -> ask the server to match our window */
+let resizeTimer = null;
+window.addEventListener("resize", () => {
+  clearTimeout(resizeTimer);
+  resizeTimer = setTimeout(() => {
+    const el = document.getElementById("stage");
+    send(`r,${el.clientWidth}x${el.clientHeight}`);
+  }, 400);
+});
+
+hookInput();
+hookHud();
+connect();
+canvas.focus();
