@@ -79,6 +79,10 @@ class StreamingService:
         self._frame_clock = 0
         self._stats_task: Optional[asyncio.Task] = None
         self.frames_relayed = 0
+        self.audio: Optional[object] = None
+        self._audio_queue: Optional[asyncio.Queue] = None
+        self._audio_task: Optional[asyncio.Task] = None
+        self.mic_sink: Optional[object] = None
 
     # ---- capture lifecycle -------------------------------------------------
     def build_capture_settings(self) -> "hipflux.CaptureSettings":
@@ -134,6 +138,59 @@ class StreamingService:
             self.capture.stop_capture()
             self.capture = None
 
+    # ---- audio (shared encode, per-client delivery; reference
+    # _pcmflux_audio_callback drop-oldest contract, selkies.py:1613) --------
+    def start_audio(self):
+        if self.audio is not None or not self.settings.enable_audio:
+            return
+        from hipflux import _native
+        s = _native.AudioCaptureSettings()
+        s.device_name = self.settings.audio_device if \
+            self.settings.audio_device != "auto" else "synthetic"
+        s.channels = self.settings.audio_channels
+        s.opus_bitrate = self.settings.audio_bitrate
+        s.frame_duration_ms = self.settings.audio_frame_duration_ms
+        s.red_distance = self.settings.audio_red_distance
+        loop = asyncio.get_running_loop()
+        self._audio_queue = asyncio.Queue(maxsize=16)
+
+        def on_frame(data, pts_ms):
+            def put():
+                q = self._audio_queue
+                if q is None:
+                    return
+                if q.full():
+                    try:
+                        q.get_nowait()      # drop oldest
+                    except asyncio.QueueEmpty:
+                        pass
+                q.put_nowait(data)
+            loop.call_soon_threadsafe(put)
+
+        self.audio = _native.AudioCapture()
+        self.audio.start_capture(s, on_frame)
+        self._audio_task = loop.create_task(self._audio_sender())
+
+    def stop_audio(self):
+        if self.audio is not None:
+            self.audio.stop_capture()
+            self.audio = None
+        if self._audio_task is not None:
+            self._audio_task.cancel()
+            self._audio_task = None
+        self._audio_queue = None
+
+    async def _audio_sender(self):
+        while True:
+            data = await self._audio_queue.get()
+            for cs in list(self.clients.values()):
+                if cs.relay.dead:
+                    continue
+                try:
+                    await asyncio.wait_for(cs.ws.send_bytes(data), 1.0)
+                except Exception:
+                    pass
+
     def request_idr(self):
         if self.capture is not None:
             self.capture.request_idr_frame()
@@ -182,6 +239,7 @@ class StreamingService:
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload()))
             self.start_capture()
+            self.start_audio()
             self.request_idr()
 
             async for msg in ws:
@@ -198,6 +256,7 @@ class StreamingService:
             await relay.stop()
             if not self.clients:
                 self.stop_capture()
+                self.stop_audio()
         return ws
 
     async def _on_text(self, state: ClientState, text: str) -> Optional[str]:
@@ -231,7 +290,13 @@ class StreamingService:
             asyncio.get_running_loop().create_task(
                 self._handle_inflated(state, text))
         elif data[0] == P.TAG_MIC_PCM:
-            pass  # mic uplink lands with the audio engine
+            # client mic PCM -> virtual microphone sink
+            if self.settings.enable_microphone:
+                if self.mic_sink is None:
+                    from hipflux import _native
+                    ps = _native.AudioPlaybackSettings()
+                    self.mic_sink = _native.AudioPlayback(ps)
+                self.mic_sink.write(data[1:])
 
     async def _handle_inflated(self, state: ClientState, text: str):
         reply = await self._on_text(state, text)

@@ -96,10 +96,44 @@ function onBinary(buf) {
     });
     noteFrame(frameId);
   } else if (tag === 0x01) {
-    /* audio lands with the audio engine */
+    playAudioFrame(d);
   } else if (tag === 0x05) {
     /* gzip'd control text: rare server->client path; ignore for now */
   }
+}
+
+/* ---------------- audio playback ----------------
+ * wire: [0x01, n_red] + n_red x (u16 len + redundant payload) + primary.
+ * Payload is s16le PCM (WS is reliable, so RED payloads are skipped). */
+let audioCtx = null, audioTime = 0;
+const AUDIO_RATE = 48000, AUDIO_CH = 2;
+
+function playAudioFrame(d) {
+  let off = 2;
+  const nRed = d[1];
+  for (let i = 0; i < nRed; i++) {
+    const len = (d[off] << 8) | d[off + 1];
+    off += 2 + len;
+  }
+  const pcm = new Int16Array(d.buffer, d.byteOffset + off,
+                             (d.length - off) >> 1);
+  if (!audioCtx) {
+    try { audioCtx = new AudioContext({ sampleRate: AUDIO_RATE }); }
+    catch (e) { return; }
+  }
+  const framesN = pcm.length / AUDIO_CH;
+  const buf = audioCtx.createBuffer(AUDIO_CH, framesN, AUDIO_RATE);
+  for (let c = 0; c < AUDIO_CH; c++) {
+    const chan = buf.getChannelData(c);
+    for (let i = 0; i < framesN; i++) chan[i] = pcm[i * AUDIO_CH + c] / 32768;
+  }
+  const src = audioCtx.createBufferSource();
+  src.buffer = buf;
+  src.connect(audioCtx.destination);
+  const now = audioCtx.currentTime;
+  if (audioTime < now + 0.02) audioTime = now + 0.04;  // jitter buffer
+  src.start(audioTime);
+  audioTime += framesN / AUDIO_RATE;
 }
 
 function noteFrame(frameId) {

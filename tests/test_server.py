@@ -28,6 +28,7 @@ def make_server(**env):
         "SELKIES_FRAMERATE": "30",
         "SELKIES_USE_CPU": "true",
         "SELKIES_ENCODER": "h264enc-striped",
+        "SELKIES_ENABLE_AUDIO": "false",
     }
     base_env.update(env)
     settings = load_settings(argv=[], env=base_env)
@@ -150,14 +151,44 @@ def test_jpeg_mode_stripes(loop):
                     got = None
                     while asyncio.get_event_loop().time() < deadline:
                         msg = await ws.receive(timeout=8)
-                        if msg.type == WSMsgType.BINARY:
+                        if msg.type == WSMsgType.BINARY and \
+                                msg.data[0] == 0x03:
                             got = msg.data
                             break
-                    assert got is not None and got[0] == 0x03
+                    assert got is not None
                     # payload after the 6-byte header is a JFIF image
                     assert got[6:8] == b"\xff\xd8"
         finally:
             server.streaming.stop_capture()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_audio_broadcast(loop):
+    async def main():
+        server = make_server(SELKIES_ENABLE_AUDIO="true",
+                             SELKIES_AUDIO_RED_DISTANCE="1")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"http://127.0.0.1:{port}/ws") as ws:
+                    audio = []
+                    deadline = asyncio.get_event_loop().time() + 8
+                    while asyncio.get_event_loop().time() < deadline:
+                        msg = await ws.receive(timeout=8)
+                        if msg.type == WSMsgType.BINARY and \
+                                msg.data[0] == 0x01:
+                            audio.append(msg.data)
+                            if len(audio) >= 5:
+                                break
+                    assert len(audio) >= 5
+                    # RED depth reaches the configured distance
+                    assert audio[-1][1] == 1
+        finally:
+            server.streaming.stop_capture()
+            server.streaming.stop_audio()
             await runner.cleanup()
 
     loop.run_until_complete(main())
