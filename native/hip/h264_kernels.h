@@ -10,7 +10,8 @@ namespace h264gpu {
 
 void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
                     const uint8_t* refY, int mbw, int n_jobs,
-                    const RowJob* d_jobs, int* d_meta, hipStream_t stream);
+                    const RowJob* d_jobs, int* d_meta, hipStream_t stream,
+                    bool use_mfma = true);
 
 void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
                       const uint8_t* srcCr, int ypitch, int cpitch, int w,

@@ -615,13 +615,205 @@ __global__ void __launch_bounds__(64) k_h264_me(
 }
 
 // ---------------------------------------------------------------------------
+// MFMA-int8 full-search motion estimation.
+//
+// Scores ALL candidates of a ±8 even-step grid (9x9 = 81) by SSD using the
+// matrix cores:  SSD_n = S2 - 2*X_n + R2_n  with pixels centered to int8
+// (p-128; SSD is shift-invariant). Per 16-candidate group:
+//   X_n  : 4x mfma_i32_16x16x64_i8, A rows = broadcast src chunk (64 px),
+//          B cols = candidate chunks       -> D[0][n] = sum(a*b_n)
+//   R2_n : 4x mfma with A = B              -> diag D[n][n] = sum(b_n^2)
+// One wave per MB. The final skip/inter decision uses the winner's SAD so
+// thresholds stay identical to the CPU reference encoder.
+using i32x4 = __attribute__((__vector_size__(16))) int;
+
+__global__ void __launch_bounds__(64) k_h264_me_mfma(
+    const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
+    const uint8_t* __restrict__ refY, int mbw, int frame_w_mb16,
+    const RowJob* __restrict__ jobs, int* __restrict__ meta) {
+  const int job_idx = blockIdx.x / mbw;
+  const int mbx = blockIdx.x % mbw;
+  const RowJob job = jobs[job_idx];
+  if (job.flags & 1) return;
+  const int lane = threadIdx.x;
+  const int mby = job.mb_row;
+  const int x0 = mbx * 16, y0 = mby * 16;
+  const size_t mb_index = (size_t)mby * mbw + mbx;
+
+  // --- src MB into LDS (+ centered copy for fragments)
+  __shared__ int8_t s_a[256];
+  {
+    int r = lane >> 2, cq = (lane & 3) * 4;
+    const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+    for (int j = 0; j < 4; ++j)
+      s_a[r * 16 + cq + j] =
+          (int8_t)((int)s[min(x0 + cq + j, w - 1)] - 128);
+  }
+  __syncthreads();
+
+  // S2 = sum(a^2)
+  int s2 = 0;
+  {
+    int r = lane >> 2, cq = (lane & 3) * 4;
+    for (int j = 0; j < 4; ++j) {
+      int v = s_a[r * 16 + cq + j];
+      s2 += v * v;
+    }
+    s2 = wave_sum_i(s2);
+  }
+
+  // A fragment per k-chunk-group kg = lane>>4: bytes a_q[kg*16 .. +15]
+  // (identical for the 16 lanes of a group -> broadcast rows). Chunk q is
+  // selected per MFMA call; precompute all 4 chunks' fragments.
+  i32x4 afrag[4];
+  {
+    int kg = lane >> 4;
+    for (int q = 0; q < 4; ++q) {
+      // chunk q = MB rows 4q..4q+3; k = r*16+c (r<4). bytes kg*16..+15 of
+      // the chunk = row kg of the chunk.
+      const int8_t* p = &s_a[(4 * q + kg) * 16];
+      int words[4];
+      for (int t = 0; t < 4; ++t)
+        words[t] = (uint8_t)p[4 * t] | ((uint8_t)p[4 * t + 1] << 8) |
+                   ((uint8_t)p[4 * t + 2] << 16) |
+                   ((uint8_t)p[4 * t + 3] << 24);
+      afrag[q] = i32x4{words[0], words[1], words[2], words[3]};
+    }
+  }
+
+  // candidate grid: 81 candidates in 6 groups of up to 16
+  const int NC = 81;
+  int best_score = INT_MAX, best_mvx = 0, best_mvy = 0;
+  const int col = lane & 15, kg = lane >> 4;
+
+  for (int g0 = 0; g0 < NC; g0 += 16) {
+    // this lane's candidate (column) in the group
+    int cand = g0 + col;
+    int mx = 0, my = 0;
+    bool valid = cand < NC;
+    if (valid) {
+      mx = (cand % 9) * 2 - 8;
+      my = (cand / 9) * 2 - 8;
+      if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
+          y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1)
+        valid = false;
+    }
+    i32x4 cross = {0, 0, 0, 0};
+    i32x4 norm = {0, 0, 0, 0};
+    for (int q = 0; q < 4; ++q) {
+      // B fragment: lane holds B[k = kg*16 + j][col] = candidate col's
+      // chunk-q byte (row kg of chunk, cols 0..15 -> j)
+      int8_t bb[16];
+      if (valid) {
+        const uint8_t* rp =
+            refY + (size_t)(y0 + my + 4 * q + kg) * ypitch + x0 + mx;
+        for (int j = 0; j < 16; ++j) bb[j] = (int8_t)((int)rp[j] - 128);
+      } else {
+        for (int j = 0; j < 16; ++j) bb[j] = 0;
+      }
+      int words[4];
+      for (int t = 0; t < 4; ++t)
+        words[t] = (uint8_t)bb[4 * t] | ((uint8_t)bb[4 * t + 1] << 8) |
+                   ((uint8_t)bb[4 * t + 2] << 16) |
+                   ((uint8_t)bb[4 * t + 3] << 24);
+      i32x4 bfrag = i32x4{words[0], words[1], words[2], words[3]};
+      cross = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[q], bfrag, cross,
+                                                    0, 0, 0);
+      norm = __builtin_amdgcn_mfma_i32_16x16x64_i8(bfrag, bfrag, norm,
+                                                   0, 0, 0);
+    }
+    // D mapping (16x16): col = lane&15, row = (lane>>4)*4 + i.
+    // X_n lives in row 0 -> lanes kg==0, acc idx 0, col n.
+    // R2_n is the diagonal (n,n) -> lane ((n>>2)<<4)|n, acc idx n&3.
+    // Gather per-candidate scores into the lane owning candidate n = col.
+    for (int n_base = 0; n_base < 16; ++n_base) {
+      // broadcast X_n and R2_n from their owner lanes to everyone
+      int xl = __shfl(cross[0], n_base);                  // lane n_base
+      int diag_lane = ((n_base >> 2) << 4) | n_base;
+      int rl;
+      switch (n_base & 3) {
+        case 0: rl = __shfl(norm[0], diag_lane); break;
+        case 1: rl = __shfl(norm[1], diag_lane); break;
+        case 2: rl = __shfl(norm[2], diag_lane); break;
+        default: rl = __shfl(norm[3], diag_lane); break;
+      }
+      int cand2 = g0 + n_base;
+      if (cand2 < NC && lane == 0) {
+        int mx2 = (cand2 % 9) * 2 - 8, my2 = (cand2 / 9) * 2 - 8;
+        bool v2 = !(x0 + mx2 < 0 || x0 + mx2 + 16 > frame_w_mb16 ||
+                    y0 + my2 < job.stripe_y0 ||
+                    y0 + my2 + 16 > job.stripe_y1);
+        if (v2) {
+          int ssd = s2 - 2 * xl + rl;
+          // small center bias keeps MVs compact on flat content
+          int score = ssd + (abs(mx2) + abs(my2)) * 4;
+          if (score < best_score) {
+            best_score = score;
+            best_mvx = mx2;
+            best_mvy = my2;
+          }
+        }
+      }
+    }
+  }
+  best_score = __shfl(best_score, 0);
+  best_mvx = __shfl(best_mvx, 0);
+  best_mvy = __shfl(best_mvy, 0);
+
+  // SAD of (0,0) and of the SSD winner -> same thresholds as CPU encoder
+  const int r = lane >> 2, cq = (lane & 3) * 4;
+  int sp[4];
+  {
+    const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+    for (int j = 0; j < 4; ++j) sp[j] = s[min(x0 + cq + j, w - 1)];
+  }
+  auto sad_at = [&](int mx, int my) -> int {
+    const uint8_t* rp = refY + (size_t)(y0 + my + r) * ypitch + x0 + mx + cq;
+    int s = abs(sp[0] - rp[0]) + abs(sp[1] - rp[1]) + abs(sp[2] - rp[2]) +
+            abs(sp[3] - rp[3]);
+    return wave_sum_i(s);
+  };
+  const int qp = job.qp;
+  const int skip_thresh = 48 << (qp / 6);
+  const int inter_thresh = 2 * skip_thresh;
+  int sad0 = sad_at(0, 0);
+  int mode, omvx = 0, omvy = 0;
+  if (sad0 <= skip_thresh) {
+    mode = kSkip;
+  } else {
+    int best_sad = (best_mvx || best_mvy) ? sad_at(best_mvx, best_mvy)
+                                          : sad0;
+    if (best_sad <= inter_thresh && !(best_mvx == 0 && best_mvy == 0)) {
+      mode = kInter;
+      omvx = best_mvx;
+      omvy = best_mvy;
+    } else {
+      mode = kIntra;
+    }
+  }
+  if (lane == 0) {
+    meta[mb_index * kMetaPerMb + 0] = mode;
+    meta[mb_index * kMetaPerMb + 1] =
+        ((omvx * 4) & 0xFFFF) | ((omvy * 4) << 16);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host launchers
 void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
                     const uint8_t* refY, int mbw, int n_jobs,
-                    const RowJob* d_jobs, int* d_meta, hipStream_t stream) {
+                    const RowJob* d_jobs, int* d_meta, hipStream_t stream,
+                    bool use_mfma) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * mbw), dim3(64), 0, stream,
-                     srcY, ypitch, w, h, refY, mbw, mbw * 16, d_jobs, d_meta);
+  if (use_mfma) {
+    hipLaunchKernelGGL(k_h264_me_mfma, dim3(n_jobs * mbw), dim3(64), 0,
+                       stream, srcY, ypitch, w, h, refY, mbw, mbw * 16,
+                       d_jobs, d_meta);
+  } else {
+    hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * mbw), dim3(64), 0, stream,
+                       srcY, ypitch, w, h, refY, mbw, mbw * 16, d_jobs,
+                       d_meta);
+  }
 }
 
 void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,

@@ -138,6 +138,52 @@ def test_gpu_p_chain_no_drift():
         assert p > 30, f"frame {fi}: PSNR {p:.1f} too low (drift?)"
 
 
+def test_gpu_mfma_motion_search_exact():
+    """Frame 1 = frame 0 rolled by (dx=4, dy=-2). The MFMA full-search ME
+    must find mv=(-4,+2) for interior MBs; with cbp=0 inter coding the
+    decoded P frame then equals the rolled decoded IDR EXACTLY wherever
+    inter/skip applies. Any MFMA layout/scoring bug breaks this."""
+    require_gpu()
+    from scipy.ndimage import uniform_filter
+    w, h = 256, 128
+    rng = np.random.default_rng(9)
+    base = rng.integers(0, 256, (h, w), dtype=np.uint8).astype(np.float32)
+    smooth = uniform_filter(base, 3).astype(np.uint8)
+    f0 = np.zeros((h, w, 4), np.uint8)
+    for c in range(3):
+        f0[:, :, c] = smooth
+    f0[:, :, 3] = 255
+    dx, dy = 4, -2
+    f1 = np.roll(np.roll(f0, dy, axis=0), dx, axis=1)
+    out = _native._pipeline_encode(
+        "gpu", [np.ascontiguousarray(f0), np.ascontiguousarray(f1)],
+        w, h, 30, 64, 1)
+    rows = reassemble(out)
+    dec = decode_rows(rows, w, 64)
+    ys = sorted(dec)
+    d0 = np.concatenate([dec[y][0][0] for y in ys], 0)
+    d1 = np.concatenate([dec[y][1][0] for y in ys], 0)
+    expected = np.roll(np.roll(d0, dy, axis=0), dx, axis=1)
+    # valid MBs: not the first MB column (mv x out of frame) and not the
+    # last MB row of each stripe (mv y out of the stripe)
+    match = 0
+    total = 0
+    for mby in range(h // 16):
+        if (mby * 16) % 64 == 48:
+            continue
+        for mbx in range(1, w // 16):
+            a = d1[mby * 16:mby * 16 + 16, mbx * 16:mbx * 16 + 16]
+            b = expected[mby * 16:mby * 16 + 16, mbx * 16:mbx * 16 + 16]
+            total += 1
+            match += int(np.array_equal(a, b))
+    assert total > 0
+    frac = match / total
+    assert frac > 0.9, f"only {match}/{total} MBs motion-matched exactly"
+    # and the P frame must be far smaller than the IDR (pure MC copies)
+    sizes = [sum(len(t[0]) for t in fr) for fr in out]
+    assert sizes[1] < sizes[0] * 0.2, f"P frame too large: {sizes}"
+
+
 def test_gpu_h264_1080p_throughput():
     require_gpu()
     import time
