@@ -83,6 +83,10 @@ class StreamingService:
         self._audio_queue: Optional[asyncio.Queue] = None
         self._audio_task: Optional[asyncio.Task] = None
         self.mic_sink: Optional[object] = None
+        self.gamepads = None
+        if settings.enable_gamepad:
+            from .gamepad import GamepadHub
+            self.gamepads = GamepadHub()
 
     # ---- capture lifecycle -------------------------------------------------
     def build_capture_settings(self) -> "hipflux.CaptureSettings":
@@ -276,6 +280,9 @@ class StreamingService:
                 self.settings.build_client_settings_payload())
         if verb == "REQUEST_IDR":
             self.request_idr()
+            return None
+        if verb == "js" and self.gamepads is not None:
+            await self.gamepads.handle(text)
             return None
         return self.input.on_message(text)
 
