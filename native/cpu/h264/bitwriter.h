@@ -17,9 +17,18 @@ class BitWriter {
     nacc_ = 0;
   }
 
+  // 64-bit accumulator: append up to 32 bits per call, flush whole bytes.
   void u(uint32_t value, int nbits) {
     assert(nbits >= 0 && nbits <= 32);
-    for (int i = nbits - 1; i >= 0; --i) put_bit((value >> i) & 1);
+    if (nbits == 0) return;
+    acc_ = (acc_ << nbits) |
+           (static_cast<uint64_t>(value) &
+            ((nbits == 32 ? 0ull : (1ull << nbits)) - 1ull));
+    nacc_ += nbits;
+    while (nacc_ >= 8) {
+      buf_.push_back(static_cast<uint8_t>(acc_ >> (nacc_ - 8)));
+      nacc_ -= 8;
+    }
   }
 
   void ue(uint32_t v) {
@@ -37,18 +46,11 @@ class BitWriter {
     ue(cn);
   }
 
-  void put_bit(int b) {
-    acc_ = (acc_ << 1) | (b & 1);
-    if (++nacc_ == 8) {
-      buf_.push_back(static_cast<uint8_t>(acc_));
-      acc_ = 0;
-      nacc_ = 0;
-    }
-  }
+  void put_bit(int b) { u(b & 1, 1); }
 
   void rbsp_trailing() {
     put_bit(1);
-    while (nacc_ != 0) put_bit(0);
+    if (nacc_ != 0) u(0, 8 - nacc_);
   }
 
   size_t bit_count() const { return buf_.size() * 8 + nacc_; }

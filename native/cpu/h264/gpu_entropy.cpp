@@ -127,7 +127,13 @@ void encode_stripe_from_gpu(const GpuStripeParams& p,
     write_sps_nal(out, mbw_stripe, mbh, p.width, p.height);
     write_pps_nal(out);
   }
-  for (int row = 0; row < mbh; ++row) {
+  for (int row = 0; row < mbh; ++row) encode_row_nal_from_gpu(p, row, out);
+}
+
+void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
+                             std::vector<uint8_t>& out) {
+  const int mbw_stripe = (p.width + 15) / 16;
+  {
     BitWriter b;
     write_slice_header_bits(b, p.idr, row * mbw_stripe, p.frame_num,
                             p.idr_pic_id, p.qp);

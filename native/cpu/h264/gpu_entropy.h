@@ -26,5 +26,11 @@ struct GpuStripeParams {
 void encode_stripe_from_gpu(const GpuStripeParams& p,
                             std::vector<uint8_t>& out);
 
+// One MB-row slice NAL (row is stripe-relative). Used for row-parallel
+// entropy; concatenating SPS/PPS (idr) + all rows reproduces
+// encode_stripe_from_gpu's output exactly.
+void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
+                             std::vector<uint8_t>& out);
+
 }  // namespace h264
 }  // namespace hipflux

@@ -10,6 +10,7 @@
 #include <stdexcept>
 
 #include "cpu/h264/gpu_entropy.h"
+#include "cpu/h264/headers.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
 #include "hip/h264_kernels.h"
@@ -304,17 +305,36 @@ class HipH264Pipeline : public EncodePipeline {
                              hipMemcpyDeviceToHost, stream_));
     HIP_CHECK(hipMemcpyAsync(h_meta_, d_meta_, meta_bytes,
                              hipMemcpyDeviceToHost, stream_));
-    for (const auto& sj : sjobs) {
-      int row0 = sj.y0 / 16;
-      int rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
-      copy_region(d_refY_, d_curY_, ypitch_, row0 * 16, rows * 16);
-      copy_region(d_refCb_, d_curCb_, cpitch_, row0 * 8, rows * 8);
-      copy_region(d_refCr_, d_curCr_, cpitch_, row0 * 8, rows * 8);
+    // refresh ref from cur for encoded stripes (merge contiguous spans so
+    // the all-stripes case is 3 copies, not 3 x n_stripes)
+    {
+      int span_y0 = -1, span_y1 = -1;
+      auto flush_span = [&] {
+        if (span_y0 < 0) return;
+        int rows16 = (std::min(span_y1, mbh_ * 16) - span_y0 + 15) / 16;
+        copy_region(d_refY_, d_curY_, ypitch_, span_y0, rows16 * 16);
+        copy_region(d_refCb_, d_curCb_, cpitch_, span_y0 / 2, rows16 * 8);
+        copy_region(d_refCr_, d_curCr_, cpitch_, span_y0 / 2, rows16 * 8);
+        span_y0 = span_y1 = -1;
+      };
+      for (const auto& sj : sjobs) {
+        if (span_y1 == sj.y0) {
+          span_y1 = sj.y1;
+        } else {
+          flush_span();
+          span_y0 = sj.y0;
+          span_y1 = sj.y1;
+        }
+      }
+      flush_span();
     }
     HIP_CHECK(hipStreamSynchronize(stream_));
 
-    // stripe-parallel CPU entropy
+    // row-parallel CPU entropy (one task per MB row for load balance),
+    // concatenated per stripe afterwards
     struct Out {
+      std::vector<uint8_t> header;            // SPS/PPS on IDR
+      std::vector<std::vector<uint8_t>> rows;
       std::vector<uint8_t> bytes;
       int y0 = 0, h = 0;
       bool idr = false;
@@ -343,13 +363,25 @@ class HipH264Pipeline : public EncodePipeline {
       p.idr = sj.idr;
       p.frame_num = state.frame_num;
       p.idr_pic_id = state.idr_pic_id;
-      pool_.submit([&outs, i, p] {
-        h264::encode_stripe_from_gpu(p, outs[i].bytes);
-      });
       ++state.frame_num;
+      outs[i].rows.resize(p.n_mb_rows);
+      if (p.idr) {
+        h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
+                            p.n_mb_rows, p.width, p.height);
+        h264::write_pps_nal(outs[i].header);
+      }
+      for (int r = 0; r < p.n_mb_rows; ++r) {
+        auto* dst = &outs[i].rows[r];
+        pool_.submit([p, r, dst] {
+          h264::encode_row_nal_from_gpu(p, r, *dst);
+        });
+      }
     }
     pool_.wait_all();
     for (auto& o : outs) {
+      o.bytes = std::move(o.header);
+      for (auto& r : o.rows)
+        o.bytes.insert(o.bytes.end(), r.begin(), r.end());
       if (o.bytes.empty()) continue;
       EncodedStripe s;
       s.type = StripeType::kH264;
