@@ -72,7 +72,9 @@ class StreamingService:
         self.input = InputDispatcher(
             make_backend(settings.display if settings.enable_input else None),
             on_resize=self._on_resize,
+            on_dpi=self._on_dpi,
             on_bitrate=lambda kbps: self._tune("video_bitrate_kbps", kbps),
+            on_audio_bitrate=lambda bps: self._tune("audio_bitrate", bps),
             enable_input=settings.enable_input,
             enable_clipboard=settings.enable_clipboard,
         )
@@ -350,12 +352,25 @@ class StreamingService:
     def _on_resize(self, w: int, h: int):
         if not self.settings.enable_resize:
             return
-        w, h = max(16, w & ~1), max(16, h & ~1)
+        from . import display_utils
+        w, h = display_utils.align_dims_16(w, h)
         self.settings.set("resolution", f"{w}x{h}")
+        # resize the real display when capturing X11
+        backend = self.settings.capture_backend
+        if backend == "x11" or (backend == "auto" and self.settings.display):
+            display_utils.resize_display(w, h, self.settings.display)
         if self.capture is not None:
             self.stop_capture()
             self.start_capture()
             self.request_idr()
+
+    def _on_dpi(self, dpi: int):
+        try:
+            self.settings.set("dpi", dpi)
+        except Exception:
+            return
+        from . import display_utils
+        display_utils.set_dpi(dpi, self.settings.display)
 
     # ---- stats --------------------------------------------------------------
     def stats(self) -> dict:
