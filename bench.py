@@ -66,7 +66,9 @@ def main():
     # synthetic capture source: pre-generated random BGRX frames, cycled.
     # Every frame differs everywhere (worst case for a screen encoder).
     rng = np.random.default_rng(1234 + rank)
-    n_src = 24
+    # every source frame cycles through warmup at least once so one-time
+    # costs (pinned-memory registration) never land in the timed region
+    n_src = min(24, max(4, args.warmup))
     frames = [np.ascontiguousarray(
         rng.integers(0, 256, (args.height, args.width, 4), dtype=np.uint8))
         for _ in range(n_src)]

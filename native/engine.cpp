@@ -107,6 +107,13 @@ void ScreenCapture::run() {
   DamageTracker damage;
   damage.reset(src->width(), src->height());
 
+  // CBR rate control (video_cbr_mode): VBV-fullness PI controller mapping
+  // produced bytes to a per-frame QP within [min_qp, max_qp]. CRF mode
+  // uses the fixed crf value. (Reference knob semantics: SURVEY.md §2.3
+  // video_cbr_mode / vbv fields / min-max QP clamp.)
+  double rc_qp = static_cast<double>(crf_.load());
+  double vbv_fullness = 0.0;
+
   const int stripe_h = std::max(16, settings_.stripe_height & ~15);
   uint32_t frame_id = 0;
   int paintover_pending = 0;   // remaining paint-over burst frames
@@ -166,7 +173,15 @@ void ScreenCapture::run() {
     ctx.paintover = paintover_frame;
     ctx.jpeg_quality = paintover_frame ? settings_.jpeg_paintover_quality
                                        : jpeg_quality_.load();
-    ctx.crf = paintover_frame ? settings_.video_paintover_crf : crf_.load();
+    if (settings_.video_cbr_mode && settings_.output_mode == 1) {
+      ctx.crf = static_cast<int>(rc_qp + 0.5);
+      if (paintover_frame)
+        ctx.crf = std::min(ctx.crf, settings_.video_paintover_crf);
+    } else {
+      ctx.crf = paintover_frame ? settings_.video_paintover_crf : crf_.load();
+    }
+    ctx.crf = std::clamp(ctx.crf, settings_.video_min_qp,
+                         settings_.video_max_qp);
     ctx.bitrate_kbps = bitrate_kbps_.load();
     bool any = false;
     for (int y = 0; y < frame.height; y += stripe_h) {
@@ -182,7 +197,9 @@ void ScreenCapture::run() {
 
     if (any) {
       double t0 = now_ms();
+      size_t frame_bytes = 0;
       pipeline->encode_frame(frame, ctx, [&](EncodedStripe& s) {
+        frame_bytes += s.size;
         s.capture_ts_ms = frame.ts_ms;
         s.encode_done_ms = now_ms();
         if (settings_.omit_stripe_headers) {
@@ -200,6 +217,23 @@ void ScreenCapture::run() {
       last_encode_ms_.store(now_ms() - t0);
       frames_encoded_.fetch_add(1);
       ++frame_id;
+
+      if (settings_.video_cbr_mode && settings_.output_mode == 1) {
+        double target_bpf =
+            bitrate_kbps_.load() * 1000.0 / 8.0 / std::max(1.0, fps);
+        // VBV buffer: vbv_multiplier x ~quarter-second of stream
+        double vbv_size = target_bpf * std::max(0.2, vbv_mult_.load()) *
+                          std::max(2.0, fps / 4.0);
+        vbv_fullness += static_cast<double>(frame_bytes) - target_bpf;
+        vbv_fullness = std::clamp(vbv_fullness, 0.0, vbv_size);
+        double err = frame_bytes / std::max(1.0, target_bpf);
+        // proportional on instantaneous overshoot + integral via fullness
+        rc_qp += 0.6 * std::log2(std::max(0.05, err)) +
+                 2.0 * (vbv_fullness / vbv_size - 0.4);
+        rc_qp = std::clamp(rc_qp,
+                           static_cast<double>(settings_.video_min_qp),
+                           static_cast<double>(settings_.video_max_qp));
+      }
     }
 
     // pacing (skip sleeping if we're behind)

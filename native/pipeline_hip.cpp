@@ -222,11 +222,15 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     stripe_h_ = std::max(16, s.stripe_height & ~15);
+    batch_events_.resize(4);
+    for (auto& e : batch_events_)
+      HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
     alloc_for(s.capture_width, s.capture_height);
   }
 
   ~HipH264Pipeline() override {
     (void)hipStreamSynchronize(stream_);
+    for (auto& e : batch_events_) (void)hipEventDestroy(e);
     for (auto& kv : registered_) (void)hipHostUnregister(kv.first);
     for (void* p : device_ptrs_)
       if (p) (void)hipFree(p);
@@ -291,20 +295,42 @@ class HipH264Pipeline : public EncodePipeline {
                              hipMemcpyHostToDevice, stream_));
     h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, mbw_, n_jobs,
                             d_jobs_, d_meta_, stream_);
-    h264gpu::launch_h264_rows(d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_,
-                              w_, h_, d_refY_, d_refCb_, d_refCr_, d_curY_,
-                              d_curCb_, d_curCr_, mbw_, n_jobs, d_jobs_,
-                              d_levels_, d_meta_, stream_);
 
-    // readback levels + meta, then refresh ref regions for encoded stripes
-    size_t level_bytes = static_cast<size_t>(mbw_) * mbh_ *
-                         h264gpu::kLevelsPerMb * sizeof(int16_t);
-    size_t meta_bytes = static_cast<size_t>(mbw_) * mbh_ *
-                        h264gpu::kMetaPerMb * sizeof(int);
-    HIP_CHECK(hipMemcpyAsync(h_levels_, d_levels_, level_bytes,
-                             hipMemcpyDeviceToHost, stream_));
-    HIP_CHECK(hipMemcpyAsync(h_meta_, d_meta_, meta_bytes,
-                             hipMemcpyDeviceToHost, stream_));
+    // Row kernel in batches: as each batch's levels land on the host, its
+    // rows' CAVLC starts on the pool while later batches still compute —
+    // entropy overlaps GPU work inside the SAME frame (no added latency).
+    const int kBatches = std::min(n_jobs, 4);
+    struct Batch {
+      int job0, jobn;        // job index range
+      int row0, rown;        // absolute MB row range [row0, rown)
+    };
+    std::vector<Batch> batches;
+    for (int b = 0; b < kBatches; ++b) {
+      int j0 = n_jobs * b / kBatches, j1 = n_jobs * (b + 1) / kBatches;
+      if (j0 == j1) continue;
+      batches.push_back({j0, j1, h_jobs_[j0].mb_row,
+                         h_jobs_[j1 - 1].mb_row + 1});
+    }
+    const size_t lvl_row =
+        static_cast<size_t>(mbw_) * h264gpu::kLevelsPerMb * sizeof(int16_t);
+    const size_t meta_row =
+        static_cast<size_t>(mbw_) * h264gpu::kMetaPerMb * sizeof(int);
+    for (size_t b = 0; b < batches.size(); ++b) {
+      const Batch& bt = batches[b];
+      h264gpu::launch_h264_rows(
+          d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
+          d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
+          bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_, stream_);
+      HIP_CHECK(hipMemcpyAsync(
+          reinterpret_cast<uint8_t*>(h_levels_) + bt.row0 * lvl_row,
+          reinterpret_cast<uint8_t*>(d_levels_) + bt.row0 * lvl_row,
+          (bt.rown - bt.row0) * lvl_row, hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipMemcpyAsync(
+          reinterpret_cast<uint8_t*>(h_meta_) + bt.row0 * meta_row,
+          reinterpret_cast<uint8_t*>(d_meta_) + bt.row0 * meta_row,
+          (bt.rown - bt.row0) * meta_row, hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipEventRecord(batch_events_[b], stream_));
+    }
     // refresh ref from cur for encoded stripes (merge contiguous spans so
     // the all-stripes case is 3 copies, not 3 x n_stripes)
     {
@@ -328,56 +354,72 @@ class HipH264Pipeline : public EncodePipeline {
       }
       flush_span();
     }
-    HIP_CHECK(hipStreamSynchronize(stream_));
+    // NOTE: no stream sync here — entropy overlaps the remaining batches;
+    // the final sync below covers the ref-refresh copies.
 
-    // row-parallel CPU entropy (one task per MB row for load balance),
-    // concatenated per stripe afterwards
+    // row-parallel CPU entropy: per-stripe output shells + params first,
+    // then submit each batch's rows as its event completes.
     struct Out {
       std::vector<uint8_t> header;            // SPS/PPS on IDR
       std::vector<std::vector<uint8_t>> rows;
       std::vector<uint8_t> bytes;
+      h264::GpuStripeParams p{};
       int y0 = 0, h = 0;
       bool idr = false;
     };
     std::vector<Out> outs(sjobs.size());
-    for (size_t i = 0; i < sjobs.size(); ++i) {
-      const auto& sj = sjobs[i];
-      auto& state = stripes_[sj.idx];
-      outs[i].y0 = sj.y0;
-      outs[i].h = std::min(sj.y1, h_) - sj.y0;
-      outs[i].idr = sj.idr;
-      if (sj.idr) {
-        state.frame_num = 0;
-        ++state.idr_pic_id;
-        state.need_idr = false;
+    // job index -> (stripe index in outs, row within stripe)
+    std::vector<std::pair<int, int>> job_map(n_jobs);
+    {
+      int j = 0;
+      for (size_t i = 0; i < sjobs.size(); ++i) {
+        const auto& sj = sjobs[i];
+        auto& state = stripes_[sj.idx];
+        outs[i].y0 = sj.y0;
+        outs[i].h = std::min(sj.y1, h_) - sj.y0;
+        outs[i].idr = sj.idr;
+        if (sj.idr) {
+          state.frame_num = 0;
+          ++state.idr_pic_id;
+          state.need_idr = false;
+        }
+        h264::GpuStripeParams& p = outs[i].p;
+        p.levels = h_levels_;
+        p.meta = h_meta_;
+        p.mbw = mbw_;
+        p.mb_row0 = sj.y0 / 16;
+        p.n_mb_rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
+        p.width = w_;
+        p.height = outs[i].h;
+        p.qp = qp;
+        p.idr = sj.idr;
+        p.frame_num = state.frame_num;
+        p.idr_pic_id = state.idr_pic_id;
+        ++state.frame_num;
+        outs[i].rows.resize(p.n_mb_rows);
+        if (p.idr) {
+          h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
+                              p.n_mb_rows, p.width, p.height);
+          h264::write_pps_nal(outs[i].header);
+        }
+        for (int r = 0; r < p.n_mb_rows; ++r) job_map[j++] = {int(i), r};
       }
-      h264::GpuStripeParams p;
-      p.levels = h_levels_;
-      p.meta = h_meta_;
-      p.mbw = mbw_;
-      p.mb_row0 = sj.y0 / 16;
-      p.n_mb_rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
-      p.width = w_;
-      p.height = outs[i].h;
-      p.qp = qp;
-      p.idr = sj.idr;
-      p.frame_num = state.frame_num;
-      p.idr_pic_id = state.idr_pic_id;
-      ++state.frame_num;
-      outs[i].rows.resize(p.n_mb_rows);
-      if (p.idr) {
-        h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
-                            p.n_mb_rows, p.width, p.height);
-        h264::write_pps_nal(outs[i].header);
-      }
-      for (int r = 0; r < p.n_mb_rows; ++r) {
-        auto* dst = &outs[i].rows[r];
-        pool_.submit([p, r, dst] {
-          h264::encode_row_nal_from_gpu(p, r, *dst);
+    }
+    for (size_t b = 0; b < batches.size(); ++b) {
+      HIP_CHECK(hipEventSynchronize(batch_events_[b]));
+      for (int j = batches[b].job0; j < batches[b].jobn; ++j) {
+        auto [si, r] = job_map[j];
+        auto* dst = &outs[si].rows[r];
+        const h264::GpuStripeParams* pp = &outs[si].p;
+        pool_.submit([pp, r, dst] {
+          h264::encode_row_nal_from_gpu(*pp, r, *dst);
         });
       }
     }
     pool_.wait_all();
+    // ref-refresh copies (and everything else) must land before the next
+    // frame reuses the pinned staging buffers
+    HIP_CHECK(hipStreamSynchronize(stream_));
     for (auto& o : outs) {
       o.bytes = std::move(o.header);
       for (auto& r : o.rows)
@@ -478,6 +520,7 @@ class HipH264Pipeline : public EncodePipeline {
   int16_t* h_levels_ = nullptr;
   int* h_meta_ = nullptr;
   h264gpu::RowJob* h_jobs_ = nullptr;
+  std::vector<hipEvent_t> batch_events_;
   std::vector<void*> device_ptrs_;
   std::vector<StripeState> stripes_;
   std::map<void*, bool> registered_;
