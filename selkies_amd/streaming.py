@@ -69,12 +69,16 @@ class StreamingService:
         self.clients: dict[object, ClientState] = {}
         self.capture: Optional[hipflux.ScreenCapture] = None
         self.loop: Optional[asyncio.AbstractEventLoop] = None
+        from .clipboard import make_clipboard
+        self.clipboard = make_clipboard(settings.display)
         self.input = InputDispatcher(
             make_backend(settings.display if settings.enable_input else None),
             on_resize=self._on_resize,
             on_dpi=self._on_dpi,
             on_bitrate=lambda kbps: self._tune("video_bitrate_kbps", kbps),
             on_audio_bitrate=lambda bps: self._tune("audio_bitrate", bps),
+            on_clipboard=self.clipboard.write,
+            clipboard_read=self.clipboard.read,
             enable_input=settings.enable_input,
             enable_clipboard=settings.enable_clipboard,
         )

@@ -1,0 +1,71 @@
+"""Clipboard wire verbs + input dispatcher coverage."""
+
+import base64
+
+from selkies_amd.clipboard import MemoryClipboard
+from selkies_amd.input_handler import InputDispatcher, RecordingBackend
+
+
+def make_dispatcher():
+    cb = MemoryClipboard()
+    backend = RecordingBackend()
+    d = InputDispatcher(backend, on_clipboard=cb.write,
+                        clipboard_read=cb.read)
+    return d, backend, cb
+
+
+def test_clipboard_write_and_read():
+    d, _, cb = make_dispatcher()
+    payload = base64.b64encode("héllo 世界".encode()).decode()
+    d.on_message(f"cw,{payload}")
+    assert cb.read() == "héllo 世界"
+    reply = d.on_message("cr,")
+    verb, rest = reply.split(",", 1)
+    assert verb == "clipboard"
+    assert base64.b64decode(rest).decode() == "héllo 世界"
+
+
+def test_clipboard_disabled():
+    cb = MemoryClipboard()
+    d = InputDispatcher(RecordingBackend(), on_clipboard=cb.write,
+                        clipboard_read=cb.read, enable_clipboard=False)
+    payload = base64.b64encode(b"secret").decode()
+    d.on_message(f"cw,{payload}")
+    assert cb.read() == ""
+    assert d.on_message("cr,") is None
+
+
+def test_button_mask_diffing():
+    d, backend, _ = make_dispatcher()
+    d.on_message("m,5,5,1")       # left down
+    d.on_message("m,6,6,3")       # +middle
+    d.on_message("m,7,7,0")       # all up
+    btns = [e for e in backend.events if e[0] == "btn"]
+    assert ("btn", 1, True) in btns and ("btn", 2, True) in btns
+    assert ("btn", 1, False) in btns and ("btn", 2, False) in btns
+
+
+def test_scroll_and_relative():
+    d, backend, _ = make_dispatcher()
+    d.on_message("sw,u,2")
+    d.on_message("m2,-3,4,0")
+    assert backend.events.count(("btn", 4, True)) == 2
+    assert ("rel", -3, 4) in backend.events
+
+
+def test_atomic_char_typing():
+    d, backend, _ = make_dispatcher()
+    text = base64.b64encode("aÉ".encode()).decode()
+    d.on_message(f"co,{text}")
+    keys = [e for e in backend.events if e[0] == "key"]
+    assert ("key", ord("a"), True) in keys
+    assert ("key", 0xC9, True) in keys          # É latin-1 keysym
+
+
+def test_key_release_all_on_reset():
+    d, backend, _ = make_dispatcher()
+    d.on_message("kd,65")
+    d.on_message("kd,0xff08" if False else "kd,66")
+    d.on_message("kr,")
+    ups = [e for e in backend.events if e[0] == "key" and not e[2]]
+    assert ("key", 65, False) in ups and ("key", 66, False) in ups
