@@ -86,6 +86,21 @@ __device__ inline uint8_t clip8(int v) { return (uint8_t)max(0, min(255, v)); }
 // 16-lane-group 4x4 transforms via shuffles. lane c in [0,16): r=c>>2, x=c&3.
 // base = (lane & ~15) is the group's first lane in the wave.
 
+// branchless forward butterfly: out(u) for inputs s0..s3
+//   u0: (s0+s3)+(s1+s2)   u1: 2(s0-s3)+(s1-s2)
+//   u2: (s0+s3)-(s1+s2)   u3: (s0-s3)-2(s1-s2)
+// selected via cndmask-style arithmetic (no lane divergence).
+__device__ inline int fdct_bfly(int u, int s0, int s1, int s2, int s3) {
+  bool odd = u & 1;            // u1/u3 use differences, u0/u2 sums
+  bool hi = u & 2;             // u2/u3 subtract the second term
+  int a = odd ? (s0 - s3) : (s0 + s3);
+  int b = odd ? (s1 - s2) : (s1 + s2);
+  // weights: u0: a+b; u1: 2a+b; u2: a-b; u3: a-2b
+  int wa = (u == 1) ? 2 : 1;
+  int wb = (u == 3) ? 2 : 1;
+  return hi ? (wa * a - wb * b) : (wa * a + wb * b);
+}
+
 __device__ inline int fdct4_wave(int v, int lane) {
   int base = lane & ~15;
   int r = (lane >> 2) & 3, u = lane & 3;
@@ -93,28 +108,26 @@ __device__ inline int fdct4_wave(int v, int lane) {
   int s1 = __shfl(v, base + r * 4 + 1);
   int s2 = __shfl(v, base + r * 4 + 2);
   int s3 = __shfl(v, base + r * 4 + 3);
-  int t;
-  switch (u) {
-    case 0: t = s0 + s1 + s2 + s3; break;
-    case 1: t = 2 * s0 + s1 - s2 - 2 * s3; break;
-    case 2: t = s0 - s1 - s2 + s3; break;
-    default: t = s0 - 2 * s1 + 2 * s2 - s3; break;
-  }
-  int vv = r;  // now do columns: lane (vv,u) reads t[y][u] from lanes y*4+u
+  int t = fdct_bfly(u, s0, s1, s2, s3);
   int c0 = __shfl(t, base + 0 * 4 + u);
   int c1 = __shfl(t, base + 1 * 4 + u);
   int c2 = __shfl(t, base + 2 * 4 + u);
   int c3 = __shfl(t, base + 3 * 4 + u);
-  switch (vv) {
-    case 0: return c0 + c1 + c2 + c3;
-    case 1: return 2 * c0 + c1 - c2 - 2 * c3;
-    case 2: return c0 - c1 - c2 + c3;
-    default: return c0 - 2 * c1 + 2 * c2 - c3;
-  }
+  return fdct_bfly(r, c0, c1, c2, c3);
 }
 
 // inverse core transform; input dequantized coeff per lane; result includes
 // (x+32)>>6
+// branchless inverse butterfly: x0:e0+e3 x1:e1+e2 x2:e1-e2 x3:e0-e3
+__device__ inline int idct_bfly(int x, int d0, int d1, int d2, int d3) {
+  int e0 = d0 + d2, e1 = d0 - d2;
+  int e2 = (d1 >> 1) - d3, e3 = d1 + (d3 >> 1);
+  bool mid = (x == 1) || (x == 2);   // e1/e2 pair
+  int a = mid ? e1 : e0;
+  int b = mid ? e2 : e3;
+  return (x & 2) ? (a - b) : (a + b);
+}
+
 __device__ inline int idct4_wave(int d, int lane) {
   int base = lane & ~15;
   int r = (lane >> 2) & 3, x = lane & 3;
@@ -122,29 +135,20 @@ __device__ inline int idct4_wave(int d, int lane) {
   int d1 = __shfl(d, base + r * 4 + 1);
   int d2 = __shfl(d, base + r * 4 + 2);
   int d3 = __shfl(d, base + r * 4 + 3);
-  int e0 = d0 + d2, e1 = d0 - d2;
-  int e2 = (d1 >> 1) - d3, e3 = d1 + (d3 >> 1);
-  int t;
-  switch (x) {
-    case 0: t = e0 + e3; break;
-    case 1: t = e1 + e2; break;
-    case 2: t = e1 - e2; break;
-    default: t = e0 - e3; break;
-  }
+  int t = idct_bfly(x, d0, d1, d2, d3);
   int c0 = __shfl(t, base + 0 * 4 + x);
   int c1 = __shfl(t, base + 1 * 4 + x);
   int c2 = __shfl(t, base + 2 * 4 + x);
   int c3 = __shfl(t, base + 3 * 4 + x);
-  int f0 = c0 + c2, f1 = c0 - c2;
-  int f2 = (c1 >> 1) - c3, f3 = c1 + (c3 >> 1);
-  int o;
-  switch (r) {
-    case 0: o = f0 + f3; break;
-    case 1: o = f1 + f2; break;
-    case 2: o = f1 - f2; break;
-    default: o = f0 - f3; break;
-  }
-  return (o + 32) >> 6;
+  return (idct_bfly(r, c0, c1, c2, c3) + 32) >> 6;
+}
+
+// branchless Hadamard butterfly: u0:a+b u1:d03+d12 u2:a-b u3:d03-d12
+__device__ inline int had_bfly(int u, int s0, int s1, int s2, int s3) {
+  bool odd = u & 1;
+  int a = odd ? (s0 - s3) : (s0 + s3);
+  int b = odd ? (s1 - s2) : (s1 + s2);
+  return (u & 2) ? (a - b) : (a + b);
 }
 
 // 4x4 Hadamard (fwd includes >>1) on lanes 0..15 of the wave
@@ -155,27 +159,22 @@ __device__ inline int hadamard4_wave(int v, int lane, bool fwd) {
   int s1 = __shfl(v, base + r * 4 + 1);
   int s2 = __shfl(v, base + r * 4 + 2);
   int s3 = __shfl(v, base + r * 4 + 3);
-  int a03 = s0 + s3, d03 = s0 - s3, a12 = s1 + s2, d12 = s1 - s2;
-  int t;
-  switch (u) {
-    case 0: t = a03 + a12; break;
-    case 1: t = d03 + d12; break;
-    case 2: t = a03 - a12; break;
-    default: t = d03 - d12; break;
-  }
+  int t = had_bfly(u, s0, s1, s2, s3);
   int c0 = __shfl(t, base + 0 * 4 + u);
   int c1 = __shfl(t, base + 1 * 4 + u);
   int c2 = __shfl(t, base + 2 * 4 + u);
   int c3 = __shfl(t, base + 3 * 4 + u);
-  int b03 = c0 + c3, e03 = c0 - c3, b12 = c1 + c2, e12 = c1 - c2;
-  int o;
-  switch (r) {
-    case 0: o = b03 + b12; break;
-    case 1: o = e03 + e12; break;
-    case 2: o = b03 - b12; break;
-    default: o = e03 - e12; break;
-  }
+  int o = had_bfly(r, c0, c1, c2, c3);
   return fwd ? (o >> 1) : o;
+}
+
+// store a 16-lane group's int16 levels as paired u32 writes (even lanes)
+__device__ inline void store_lvl_pair(int16_t* base, int c, int lvl,
+                                      int lane) {
+  int nxt = __shfl(lvl, lane + 1);
+  if ((c & 1) == 0)
+    *reinterpret_cast<uint32_t*>(base + c) =
+        (uint16_t)lvl | ((uint32_t)(uint16_t)nxt << 16);
 }
 
 // cap nonzero count at 12 within a 16-lane group (zero highest zigzag)
@@ -339,7 +338,7 @@ __global__ void __launch_bounds__(64) k_h264_rows(
       int lvl = (c == 0) ? 0 : quant_coeff(coefv, qp, coeff_cls(c));
       lvl = cap12_group(lvl, zz, c != 0, lane);
       s_lvl[blk * 16 + c] = (int16_t)lvl;
-      L[kLumaAcOff + blk * 16 + c] = (int16_t)lvl;
+      store_lvl_pair(L + kLumaAcOff + blk * 16, c, lvl, lane);
     }
     __syncthreads();
 
@@ -349,9 +348,7 @@ __global__ void __launch_bounds__(64) k_h264_rows(
       int had = hadamard4_wave(v, lane, true);
       int qdc = quant_dc_v(had, qp);
       qdc = cap12_group(qdc, zz, lane < 16, lane);
-      // NOTE: DC block max coeffs 16, cap at 12 shared helper (active only
-      // for lanes 0..15)
-      if (lane < 16) L[kLumaDcOff + lane] = (int16_t)qdc;
+      if (lane < 16) store_lvl_pair(L + kLumaDcOff, lane, qdc, lane);
       int ih = hadamard4_wave(qdc, lane, false);
       if (lane < 16) s_dcrec[lane] = dequant_luma_dc_v(ih, qp);
     }
@@ -376,9 +373,17 @@ __global__ void __launch_bounds__(64) k_h264_rows(
       int rec = idct4_wave(d, lane);
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
       int pred = luma_mode == 1 ? s_left_y[py] : dcval;
-      uint8_t pix = clip8(rec + pred);
-      curY[(size_t)(y0 + py) * ypitch + x0 + px] = pix;
-      if (px == 15) s_left_y_new[py] = pix;  // stage next MB's left column
+      int pix = clip8(rec + pred);
+      // pack 4 recon bytes into one aligned u32 store (lane c%4==0)
+      int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
+          p3 = __shfl(pix, lane + 3);
+      if ((c & 3) == 0) {
+        *reinterpret_cast<uint32_t*>(
+            curY + (size_t)(y0 + py) * ypitch + x0 + px) =
+            (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
+            ((uint32_t)p3 << 24);
+      }
+      if (px == 15) s_left_y_new[py] = (uint8_t)pix;
     }
     __syncthreads();
     if (lane < 16) s_left_y[lane] = s_left_y_new[lane];
@@ -444,7 +449,7 @@ __global__ void __launch_bounds__(64) k_h264_rows(
       int lvl = (c == 0) ? 0 : quant_coeff(coefv, qpc, coeff_cls(c));
       lvl = cap12_group(lvl, zz, c != 0, lane);
       s_clvl[(comp * 4 + sub) * 16 + c] = (int16_t)lvl;
-      L[kChromaAcOff + (comp * 4 + sub) * 16 + c] = (int16_t)lvl;
+      store_lvl_pair(L + kChromaAcOff + (comp * 4 + sub) * 16, c, lvl, lane);
     }
     __syncthreads();
 
@@ -514,11 +519,19 @@ __global__ void __launch_bounds__(64) k_h264_rows(
           pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
         }
       }
-      uint8_t pix = clip8(rec + pred);
+      int pix = clip8(rec + pred);
       uint8_t* dp = comp ? curCr : curCb;
-      dp[(size_t)(cy0 + rr) * cpitch + cx0 + cc2] = pix;
+      int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
+          p3 = __shfl(pix, lane + 3);
+      if ((c & 3) == 0) {
+        *reinterpret_cast<uint32_t*>(
+            dp + (size_t)(cy0 + rr) * cpitch + cx0 + cc2) =
+            (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
+            ((uint32_t)p3 << 24);
+      }
       if (cc2 == 7) {
-        if (comp) s_left_cr_new[rr] = pix; else s_left_cb_new[rr] = pix;
+        if (comp) s_left_cr_new[rr] = (uint8_t)pix;
+        else s_left_cb_new[rr] = (uint8_t)pix;
       }
     }
     __syncthreads();

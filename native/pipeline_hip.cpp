@@ -296,10 +296,11 @@ class HipH264Pipeline : public EncodePipeline {
     h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, mbw_, n_jobs,
                             d_jobs_, d_meta_, stream_);
 
-    // Row kernel in batches: as each batch's levels land on the host, its
-    // rows' CAVLC starts on the pool while later batches still compute —
-    // entropy overlaps GPU work inside the SAME frame (no added latency).
-    const int kBatches = std::min(n_jobs, 4);
+    // Single batch: the row kernel's cost is per-row LATENCY (all rows run
+    // concurrently), so splitting into sequential batches multiplies GPU
+    // time (measured: 4 batches regressed 278->155 fps). Entropy overlap
+    // would need per-row completion signaling; revisit with stream-per-batch.
+    const int kBatches = 1;
     struct Batch {
       int job0, jobn;        // job index range
       int row0, rown;        // absolute MB row range [row0, rown)
