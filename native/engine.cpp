@@ -3,6 +3,8 @@
 #include <algorithm>
 #include <chrono>
 #include <cstdio>
+#include <map>
+#include <string>
 
 namespace hipflux {
 
@@ -123,6 +125,9 @@ void ScreenCapture::run() {
     keyframe_due_ms = now_ms() + settings_.keyframe_interval_s * 1000.0;
 
   std::vector<uint8_t> wire;
+  // raw-ES recording tap (reference recording_socket semantics,
+  // SURVEY.md §2.3): one elementary-stream file per stripe row.
+  std::map<int, FILE*> rec_files;  // y0 -> file
   auto next_tick = clock::now();
 
   while (!stop_.load()) {
@@ -195,11 +200,24 @@ void ScreenCapture::run() {
       ctx.stripes.push_back(job);
     }
 
+    auto record_stripe = [&](const EncodedStripe& s) {
+      if (settings_.recording_path.empty()) return;
+      FILE*& f = rec_files[s.y];
+      if (!f) {
+        std::string path = settings_.recording_path + ".s" +
+                           std::to_string(s.y) +
+                           (s.type == StripeType::kH264 ? ".h264" : ".mjpeg");
+        f = std::fopen(path.c_str(), "wb");
+      }
+      if (f) std::fwrite(s.data, 1, s.size, f);
+    };
+
     if (any) {
       double t0 = now_ms();
       size_t frame_bytes = 0;
       pipeline->encode_frame(frame, ctx, [&](EncodedStripe& s) {
         frame_bytes += s.size;
+        record_stripe(s);
         s.capture_ts_ms = frame.ts_ms;
         s.encode_done_ms = now_ms();
         if (settings_.omit_stripe_headers) {
@@ -244,6 +262,8 @@ void ScreenCapture::run() {
       next_tick = now;
     }
   }
+  for (auto& kv : rec_files)
+    if (kv.second) std::fclose(kv.second);
   running_.store(false);
 }
 
