@@ -3,6 +3,7 @@
 #include <algorithm>
 #include <chrono>
 #include <cstdio>
+#include <cstring>
 #include <map>
 #include <string>
 
@@ -36,6 +37,75 @@ void ScreenCapture::update_capture_region(int x, int y, int w, int h) {
   region_[0] = x; region_[1] = y; region_[2] = w; region_[3] = h;
   region_changed_.store(true);
 }
+
+namespace {
+
+// Watermark: raw ".bgra" file = u32 width, u32 height, then BGRA pixels.
+// (The Python layer converts PNG -> .bgra via PIL; keeps libpng out of the
+// native build.) Locations: 1=TL 2=TR 3=BL 4=BR 5=center 6=animated.
+struct Watermark {
+  int w = 0, h = 0;
+  std::vector<uint8_t> bgra;
+  bool load(const std::string& path) {
+    FILE* f = std::fopen(path.c_str(), "rb");
+    if (!f) return false;
+    uint32_t wh[2];
+    if (std::fread(wh, 4, 2, f) != 2) {
+      std::fclose(f);
+      return false;
+    }
+    w = static_cast<int>(wh[0]);
+    h = static_cast<int>(wh[1]);
+    if (w <= 0 || h <= 0 || w > 8192 || h > 8192) {
+      std::fclose(f);
+      return false;
+    }
+    bgra.resize(static_cast<size_t>(w) * h * 4);
+    size_t got = std::fread(bgra.data(), 1, bgra.size(), f);
+    std::fclose(f);
+    return got == bgra.size();
+  }
+
+  void composite(uint8_t* frame, int fw, int fh, int stride, int location,
+                 uint64_t frame_idx) const {
+    if (w == 0) return;
+    int x0 = 0, y0 = 0, margin = 16;
+    switch (location) {
+      case 1: x0 = margin; y0 = margin; break;
+      case 2: x0 = fw - w - margin; y0 = margin; break;
+      case 3: x0 = margin; y0 = fh - h - margin; break;
+      case 4: x0 = fw - w - margin; y0 = fh - h - margin; break;
+      case 5: x0 = (fw - w) / 2; y0 = (fh - h) / 2; break;
+      case 6: {  // bounce animation
+        int sx = std::max(1, fw - w), sy = std::max(1, fh - h);
+        x0 = static_cast<int>((frame_idx * 3) % (2 * sx));
+        y0 = static_cast<int>((frame_idx * 2) % (2 * sy));
+        if (x0 >= sx) x0 = 2 * sx - x0 - 1;
+        if (y0 >= sy) y0 = 2 * sy - y0 - 1;
+        break;
+      }
+      default: return;
+    }
+    for (int y = 0; y < h; ++y) {
+      int fy = y0 + y;
+      if (fy < 0 || fy >= fh) continue;
+      uint8_t* dst = frame + static_cast<size_t>(fy) * stride;
+      const uint8_t* src = bgra.data() + static_cast<size_t>(y) * w * 4;
+      for (int x = 0; x < w; ++x) {
+        int fx = x0 + x;
+        if (fx < 0 || fx >= fw) continue;
+        int a = src[x * 4 + 3];
+        if (a == 0) continue;
+        uint8_t* d = dst + fx * 4;
+        for (int c = 0; c < 3; ++c)
+          d[c] = static_cast<uint8_t>(
+              (src[x * 4 + c] * a + d[c] * (255 - a) + 127) / 255);
+      }
+    }
+  }
+};
+
+}  // namespace
 
 void pack_wire_stripe(const EncodedStripe& s, std::vector<uint8_t>& out) {
   if (s.type == StripeType::kH264) {
@@ -109,6 +179,15 @@ void ScreenCapture::run() {
   DamageTracker damage;
   damage.reset(src->width(), src->height());
 
+  Watermark watermark;
+  std::vector<uint8_t> wm_scratch;
+  if (!settings_.watermark_path.empty() && settings_.watermark_location > 0) {
+    if (!watermark.load(settings_.watermark_path))
+      std::fprintf(stderr, "hipflux: cannot load watermark %s\n",
+                   settings_.watermark_path.c_str());
+  }
+  uint64_t wm_frame = 0;
+
   // CBR rate control (video_cbr_mode): VBV-fullness PI controller mapping
   // produced bytes to a per-frame QP within [min_qp, max_qp]. CRF mode
   // uses the fixed crf value. (Reference knob semantics: SURVEY.md §2.3
@@ -137,6 +216,17 @@ void ScreenCapture::run() {
     RawFrame frame;
     if (!src->acquire(frame)) break;
     frames_captured_.fetch_add(1);
+
+    if (watermark.w > 0) {
+      // composite on a scratch copy (the source buffer may persist)
+      size_t bytes = static_cast<size_t>(frame.stride) * frame.height;
+      wm_scratch.resize(bytes);
+      std::memcpy(wm_scratch.data(), frame.data, bytes);
+      watermark.composite(wm_scratch.data(), frame.width, frame.height,
+                          frame.stride, settings_.watermark_location,
+                          wm_frame++);
+      frame.data = wm_scratch.data();
+    }
 
     // damage update (skipped in fullframe mode to save the diff cost)
     bool force_all = settings_.video_fullframe;

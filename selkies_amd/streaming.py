@@ -125,7 +125,34 @@ class StreamingService:
         cs.jpeg_quality = s.jpeg_quality
         cs.stripe_height = s.stripe_height
         cs.capture_cursor = s.capture_cursor
+        if s.watermark_path and s.watermark_location > 0:
+            raw = self._watermark_to_bgra(s.watermark_path)
+            if raw:
+                cs.watermark_path = raw
+                cs.watermark_location = s.watermark_location
         return cs
+
+    @staticmethod
+    def _watermark_to_bgra(path: str) -> Optional[str]:
+        """Convert a PNG/image watermark to the engine's raw .bgra format
+        (u32 w, u32 h, BGRA pixels) via PIL."""
+        if path.endswith(".bgra"):
+            return path
+        try:
+            import struct
+            from PIL import Image
+            img = Image.open(path).convert("RGBA")
+            out = path + ".bgra"
+            import numpy as np
+            arr = np.asarray(img)
+            bgra = arr[:, :, [2, 1, 0, 3]].tobytes()
+            with open(out, "wb") as f:
+                f.write(struct.pack("<II", img.width, img.height))
+                f.write(bgra)
+            return out
+        except Exception as exc:
+            logger.warning("watermark conversion failed: %r", exc)
+            return None
 
     def start_capture(self):
         if self.capture is not None and self.capture.is_capturing:
