@@ -187,6 +187,21 @@ def set_dpi(dpi: int, display: Optional[str] = None) -> bool:
         return False
 
 
+def resize_entrypoint(argv=None) -> int:
+    """`selkies-resize WxH` CLI (reference entrypoint,
+    display_utils.py:2200)."""
+    import sys
+    args = argv if argv is not None else sys.argv[1:]
+    if not args:
+        print("usage: selkies-resize WxH [display]")
+        return 2
+    w, _, h = args[0].partition("x")
+    display = args[1] if len(args) > 1 else None
+    ok = resize_display(int(w), int(h), display)
+    print("ok" if ok else "failed")
+    return 0 if ok else 1
+
+
 def parse_dri_node_to_index(path: str) -> int:
     """'/dev/dri/renderD129' -> 1 (reference parse_dri_node_to_index,
     display_utils.py:2223)."""

@@ -33,6 +33,12 @@ class CentralizedStreamServer:
     def __init__(self, settings: AppSettings):
         self.settings = settings
         self.streaming = StreamingService(settings)
+        from .transfers import TransferManager
+        dirs = [d.strip() for d in settings.file_transfers.split(",")]
+        self.transfers = TransferManager(
+            settings.upload_dir,
+            allow_upload="upload" in dirs,
+            allow_download="download" in dirs)
         self.app = web.Application(middlewares=[self._auth_middleware])
         self.started_at = time.time()
         self._runner: Optional[web.AppRunner] = None
@@ -50,6 +56,9 @@ class CentralizedStreamServer:
         app.router.add_get("/api/stats", self.handle_stats)
         app.router.add_get("/api/settings", self.handle_settings)
         app.router.add_get("/metrics", self.handle_metrics)
+        app.router.add_post("/api/upload", self.handle_upload)
+        app.router.add_get("/api/download", self.handle_download)
+        app.router.add_get("/api/files", self.handle_files)
         web_dir = os.path.join(os.path.dirname(__file__), "web")
         if os.path.isdir(web_dir):
             app.router.add_get("/", self._index)
@@ -125,6 +134,8 @@ class CentralizedStreamServer:
             self.settings.build_client_settings_payload())
 
     async def handle_metrics(self, request):
+        """Prometheus exposition (reference: /api/metrics gated by
+        --enable-metrics-http + bearer token)."""
         s = self.settings
         if not s.enable_metrics_http:
             raise web.HTTPNotFound()
@@ -132,19 +143,66 @@ class CentralizedStreamServer:
             if request.headers.get("Authorization") != \
                     f"Bearer {s.metrics_http_token}":
                 raise web.HTTPUnauthorized()
+        from prometheus_client import (CollectorRegistry, Counter, Gauge,
+                                       generate_latest)
         st = self.streaming.stats()
-        lines = [
-            "# TYPE selkies_clients gauge",
-            f"selkies_clients {st['clients']}",
-            "# TYPE selkies_frames_encoded counter",
-            f"selkies_frames_encoded {st['frames_encoded']}",
-            "# TYPE selkies_stripes_emitted counter",
-            f"selkies_stripes_emitted {st['stripes_emitted']}",
-            "# TYPE selkies_last_encode_ms gauge",
-            f"selkies_last_encode_ms {st['last_encode_ms']}",
-        ]
-        return web.Response(text="\n".join(lines) + "\n",
-                            content_type="text/plain")
+        reg = CollectorRegistry()
+        g = Gauge("selkies_clients", "connected clients", registry=reg)
+        g.set(st["clients"])
+        c = Counter("selkies_frames_encoded", "encoded frames", registry=reg)
+        c.inc(st["frames_encoded"])
+        c2 = Counter("selkies_stripes_emitted", "emitted stripes",
+                     registry=reg)
+        c2.inc(st["stripes_emitted"])
+        g2 = Gauge("selkies_last_encode_ms", "last frame encode ms",
+                   registry=reg)
+        g2.set(st["last_encode_ms"])
+        for gpu in gpu_stats_snapshot():
+            gg = Gauge(f"selkies_gpu_busy_percent_{gpu['card']}",
+                       "amdgpu busy", registry=reg)
+            gg.set(gpu.get("busy_percent", 0))
+        return web.Response(body=generate_latest(reg),
+                            content_type="text/plain; version=0.0.4")
+
+    # ---- file transfers -----------------------------------------------------
+    async def handle_upload(self, request):
+        name = request.query.get("name", "")
+        if not name:
+            raise web.HTTPBadRequest(reason="name required")
+
+        async def chunks():
+            async for chunk in request.content.iter_chunked(256 * 1024):
+                yield chunk
+
+        try:
+            result = await self.transfers.upload(name, chunks())
+        except PermissionError as exc:
+            raise web.HTTPForbidden(reason=str(exc))
+        return web.json_response(result)
+
+    async def handle_download(self, request):
+        name = request.query.get("name", "")
+        try:
+            path = self.transfers.resolve(name)
+            if not os.path.isfile(path):
+                raise web.HTTPNotFound()
+            resp = web.StreamResponse(headers={
+                "Content-Disposition":
+                    f'attachment; filename="{os.path.basename(path)}"'})
+            await resp.prepare(request)
+            async for chunk in self.transfers.stream_file(name):
+                await resp.write(chunk)
+            await resp.write_eof()
+            return resp
+        except PermissionError as exc:
+            raise web.HTTPForbidden(reason=str(exc))
+
+    async def handle_files(self, request):
+        try:
+            return web.json_response(
+                self.transfers.listdir(request.query.get("path", "")))
+        except PermissionError as exc:
+            raise web.HTTPForbidden(reason=str(exc))
 
     # ---- TLS hot reload -----------------------------------------------------
     def _build_ssl(self) -> Optional[ssl.SSLContext]:

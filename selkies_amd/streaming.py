@@ -278,6 +278,9 @@ class StreamingService:
             self.start_capture()
             self.start_audio()
             self.request_idr()
+            if self._stats_task is None or self._stats_task.done():
+                self._stats_task = asyncio.get_running_loop().create_task(
+                    self._stats_pusher())
 
             async for msg in ws:
                 if msg.type == WSMsgType.TEXT:
@@ -402,6 +405,31 @@ class StreamingService:
             return
         from . import display_utils
         display_utils.set_dpi(dpi, self.settings.display)
+
+    async def _stats_pusher(self):
+        """Periodic system/GPU stats pushed to clients (reference
+        _collect_system_stats_ws / _collect_gpu_stats_ws, selkies.py:5676)."""
+        import psutil
+        from .stream_server import gpu_stats_snapshot
+        while self.clients:
+            payload = {
+                "cpu_percent": psutil.cpu_percent(interval=None),
+                "mem_percent": psutil.virtual_memory().percent,
+                "gpus": gpu_stats_snapshot(),
+                "stream": {
+                    "fps_target": self.settings.framerate,
+                    "encoder": self.settings.encoder,
+                    "last_encode_ms":
+                        self.capture.last_encode_ms if self.capture else 0,
+                },
+            }
+            msg = P.encode_control("SYSTEM_STATS", payload)
+            for cs in list(self.clients.values()):
+                try:
+                    await cs.ws.send_str(msg)
+                except Exception:
+                    pass
+            await asyncio.sleep(2.0)
 
     # ---- stats --------------------------------------------------------------
     def stats(self) -> dict:
