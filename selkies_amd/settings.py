@@ -103,6 +103,8 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                    v in ("auto", "x11", "synthetic")
                    or str(v).startswith("synthetic:")) else "auto"),
     SettingDef("resolution", str, "1920x1080", "Initial capture WxH.", client=True),
+    SettingDef("resolution2", str, "1280x720",
+               "Second (extended) display WxH.", client=True),
     SettingDef("framerate", int, 60, "Target capture/encode fps.",
                value_range=(1, 240), client=True),
     SettingDef("capture_cursor", bool, True, "Composite the cursor into frames.", client=True),

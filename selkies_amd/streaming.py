@@ -36,9 +36,10 @@ logger = logging.getLogger("selkies.streaming")
 
 
 class ClientState:
-    def __init__(self, ws, relay: VideoRelay):
+    def __init__(self, ws, relay: VideoRelay, display: str = "primary"):
         self.ws = ws
         self.relay = relay
+        self.display = display
         self.last_acked_frame = -1
         self.last_sent_frame = -1
         self.ack_rtt_ms = 50.0
@@ -67,7 +68,7 @@ class StreamingService:
     def __init__(self, settings: AppSettings):
         self.settings = settings
         self.clients: dict[object, ClientState] = {}
-        self.capture: Optional[hipflux.ScreenCapture] = None
+        self.captures: dict[str, hipflux.ScreenCapture] = {}
         self.loop: Optional[asyncio.AbstractEventLoop] = None
         from .clipboard import make_clipboard
         self.clipboard = make_clipboard(settings.display)
@@ -94,11 +95,23 @@ class StreamingService:
             from .gamepad import GamepadHub
             self.gamepads = GamepadHub()
 
+    @property
+    def capture(self):
+        """Primary display capture (back-compat accessor)."""
+        return self.captures.get("primary")
+
     # ---- capture lifecycle -------------------------------------------------
-    def build_capture_settings(self) -> "hipflux.CaptureSettings":
+    def build_capture_settings(self, display: str = "primary"
+                               ) -> "hipflux.CaptureSettings":
         s = self.settings
         cs = hipflux.CaptureSettings()
-        w, h = s.resolution_wh
+        if display == "primary":
+            w, h = s.resolution_wh
+        else:
+            w, _, h = s.resolution2.lower().partition("x")
+            w, h = int(w), int(h)
+            # extended desktop: second display sits right of the primary
+            cs.capture_x = s.resolution_wh[0]
         cs.capture_width = w
         cs.capture_height = h
         cs.target_fps = float(s.framerate)
@@ -154,26 +167,32 @@ class StreamingService:
             logger.warning("watermark conversion failed: %r", exc)
             return None
 
-    def start_capture(self):
-        if self.capture is not None and self.capture.is_capturing:
+    def start_capture(self, display: str = "primary"):
+        cap = self.captures.get(display)
+        if cap is not None and cap.is_capturing:
             return
         self.loop = asyncio.get_running_loop()
-        self.capture = hipflux.ScreenCapture()
+        cap = hipflux.ScreenCapture()
+        self.captures[display] = cap
         loop = self.loop
 
         def on_stripe(data, frame_id, y, width, height, is_keyframe,
-                      capture_ts_ms, encode_done_ms, stripe_type):
+                      capture_ts_ms, encode_done_ms, stripe_type,
+                      _display=display):
             # native thread -> loop (the only allowed crossing)
-            loop.call_soon_threadsafe(self._fanout, data, frame_id, y,
-                                      is_keyframe)
+            loop.call_soon_threadsafe(self._fanout, _display, data,
+                                      frame_id, y, is_keyframe)
 
-        self.capture.start_capture(on_stripe, self.build_capture_settings())
-        logger.info("capture started (pipeline=%s)", self.capture.pipeline)
+        cap.start_capture(on_stripe, self.build_capture_settings(display))
+        logger.info("capture started for %s (pipeline=%s)", display,
+                    cap.pipeline)
 
-    def stop_capture(self):
-        if self.capture is not None:
-            self.capture.stop_capture()
-            self.capture = None
+    def stop_capture(self, display: Optional[str] = None):
+        names = [display] if display else list(self.captures)
+        for name in names:
+            cap = self.captures.pop(name, None)
+            if cap is not None:
+                cap.stop_capture()
 
     # ---- audio (shared encode, per-client delivery; reference
     # _pcmflux_audio_callback drop-oldest contract, selkies.py:1613) --------
@@ -228,15 +247,17 @@ class StreamingService:
                 except Exception:
                     pass
 
-    def request_idr(self):
-        if self.capture is not None:
-            self.capture.request_idr_frame()
+    def request_idr(self, display: Optional[str] = None):
+        for name, cap in self.captures.items():
+            if display is None or name == display:
+                cap.request_idr_frame()
 
     # ---- fan-out -----------------------------------------------------------
-    def _fanout(self, data: bytes, frame_id: int, y: int, is_keyframe: bool):
+    def _fanout(self, display: str, data: bytes, frame_id: int, y: int,
+                is_keyframe: bool):
         self.frames_relayed += 1
         for cs in list(self.clients.values()):
-            if cs.relay.dead:
+            if cs.relay.dead or cs.display != display:
                 continue
             if cs.paused and not is_keyframe:
                 continue
@@ -262,12 +283,17 @@ class StreamingService:
             heartbeat=30)
         await ws.prepare(request)
 
+        display = request.query.get("display", "primary")
+        if display not in ("primary", "display2"):
+            display = "primary"
+        if display == "display2" and not self.settings.second_display:
+            self.settings.set("second_display", True)
         relay = VideoRelay(
             send=ws.send_bytes,
-            request_idr=self.request_idr,
+            request_idr=lambda d=display: self.request_idr(d),
             bitrate_bps=self.settings.video_bitrate_kbps * 1000.0)
         relay.start()
-        state = ClientState(ws, relay)
+        state = ClientState(ws, relay, display)
         self.clients[ws] = state
 
         try:
@@ -275,9 +301,9 @@ class StreamingService:
             await ws.send_str(P.encode_control(
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload()))
-            self.start_capture()
+            self.start_capture(display)
             self.start_audio()
-            self.request_idr()
+            self.request_idr(display)
             if self._stats_task is None or self._stats_task.done():
                 self._stats_task = asyncio.get_running_loop().create_task(
                     self._stats_pusher())
@@ -294,6 +320,8 @@ class StreamingService:
         finally:
             self.clients.pop(ws, None)
             await relay.stop()
+            if not any(c.display == display for c in self.clients.values()):
+                self.stop_capture(display)
             if not self.clients:
                 self.stop_capture()
                 self.stop_audio()
@@ -360,19 +388,22 @@ class StreamingService:
                 continue
             self.settings.set(name, value)
             structural |= name in self.STRUCTURAL
-            if self.capture is not None and not structural:
-                if name == "framerate":
-                    self.capture.update_framerate(float(value))
-                elif name == "video_bitrate_kbps":
-                    self.capture.update_video_bitrate(int(value))
-                elif name == "video_crf":
-                    self.capture.update_crf(int(value))
-                elif name == "jpeg_quality":
-                    self.capture.update_jpeg_quality(int(value))
-        if structural and self.capture is not None:
-            logger.info("structural setting changed; restarting capture")
+            if not structural:
+                for cap in self.captures.values():
+                    if name == "framerate":
+                        cap.update_framerate(float(value))
+                    elif name == "video_bitrate_kbps":
+                        cap.update_video_bitrate(int(value))
+                    elif name == "video_crf":
+                        cap.update_crf(int(value))
+                    elif name == "jpeg_quality":
+                        cap.update_jpeg_quality(int(value))
+        if structural and self.captures:
+            logger.info("structural setting changed; restarting captures")
+            displays = list(self.captures)
             self.stop_capture()
-            self.start_capture()
+            for d in displays:
+                self.start_capture(d)
             self.request_idr()
 
     def _tune(self, name: str, value):
@@ -380,8 +411,9 @@ class StreamingService:
             self.settings.set(name, value)
         except Exception:
             return
-        if self.capture is not None and name == "video_bitrate_kbps":
-            self.capture.update_video_bitrate(int(value))
+        if name == "video_bitrate_kbps":
+            for cap in self.captures.values():
+                cap.update_video_bitrate(int(value))
 
     def _on_resize(self, w: int, h: int):
         if not self.settings.enable_resize:
@@ -393,9 +425,11 @@ class StreamingService:
         backend = self.settings.capture_backend
         if backend == "x11" or (backend == "auto" and self.settings.display):
             display_utils.resize_display(w, h, self.settings.display)
-        if self.capture is not None:
+        if self.captures:
+            displays = list(self.captures)
             self.stop_capture()
-            self.start_capture()
+            for d in displays:
+                self.start_capture(d)
             self.request_idr()
 
     def _on_dpi(self, dpi: int):
