@@ -8,6 +8,15 @@
 
 namespace hipflux {
 
+// Cursor snapshot (ARGB premultiplied-ish, as X11 XFixes delivers).
+struct CursorImage {
+  int width = 0, height = 0;
+  int hot_x = 0, hot_y = 0;
+  int x = 0, y = 0;              // current position
+  uint64_t serial = 0;           // changes when the shape changes
+  std::vector<uint32_t> argb;
+};
+
 class FrameSource {
  public:
   virtual ~FrameSource() = default;
@@ -16,6 +25,8 @@ class FrameSource {
   virtual bool acquire(RawFrame& out) = 0;
   virtual int width() const = 0;
   virtual int height() const = 0;
+  // Fill the current cursor state; false if the source has no cursor.
+  virtual bool cursor(CursorImage& out) { (void)out; return false; }
 };
 
 // Synthetic BGRX generator for benches and GPU boxes without an X server

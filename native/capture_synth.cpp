@@ -1,3 +1,4 @@
+#include <algorithm>
 #include <chrono>
 #include <cstring>
 #include <random>
@@ -65,6 +66,32 @@ class SyntheticSource : public FrameSource {
 
   int width() const override { return w_; }
   int height() const override { return h_; }
+
+  bool cursor(CursorImage& out) override {
+    // synthetic cursor: 12x16 arrow bouncing over the frame
+    out.width = 12;
+    out.height = 16;
+    out.hot_x = 0;
+    out.hot_y = 0;
+    out.serial = 1;
+    int sx = std::max(1, w_ - 12), sy = std::max(1, h_ - 16);
+    int px = static_cast<int>((frame_ * 5) % (2 * sx));
+    int py = static_cast<int>((frame_ * 4) % (2 * sy));
+    if (px >= sx) px = 2 * sx - px - 1;
+    if (py >= sy) py = 2 * sy - py - 1;
+    out.x = px;
+    out.y = py;
+    if (out.argb.empty()) {
+      out.argb.assign(12 * 16, 0);
+      for (int y = 0; y < 16; ++y)
+        for (int x = 0; x < 12; ++x)
+          if (x <= y && y < 14 - x / 3)
+            out.argb[y * 12 + x] =
+                (x == 0 || x == y || y == 13 - x / 3) ? 0xFF000000
+                                                      : 0xFFFFFFFF;
+    }
+    return true;
+  }
 
  private:
   void mask_x() {

@@ -92,6 +92,28 @@ class X11Source : public FrameSource {
   int width() const override { return w_; }
   int height() const override { return h_; }
 
+  bool cursor(CursorImage& out) override {
+    if (!have_xfixes_) {
+      int eb, er;
+      have_xfixes_ = XFixesQueryExtension(dpy_, &eb, &er) ? 1 : -1;
+    }
+    if (have_xfixes_ < 0) return false;
+    XFixesCursorImage* ci = XFixesGetCursorImage(dpy_);
+    if (!ci) return false;
+    out.width = ci->width;
+    out.height = ci->height;
+    out.hot_x = ci->xhot;
+    out.hot_y = ci->yhot;
+    out.x = ci->x - ci->xhot - x_;
+    out.y = ci->y - ci->yhot - y_;
+    out.serial = ci->cursor_serial;
+    out.argb.resize(static_cast<size_t>(ci->width) * ci->height);
+    for (size_t i = 0; i < out.argb.size(); ++i)
+      out.argb[i] = static_cast<uint32_t>(ci->pixels[i]);
+    XFree(ci);
+    return true;
+  }
+
  private:
   X11Source() = default;
   Display* dpy_ = nullptr;
@@ -99,6 +121,7 @@ class X11Source : public FrameSource {
   XImage* img_ = nullptr;
   XShmSegmentInfo shm_{};
   bool use_shm_ = false;
+  int have_xfixes_ = 0;
   int x_ = 0, y_ = 0, w_ = 0, h_ = 0;
 };
 

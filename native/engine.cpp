@@ -40,6 +40,10 @@ void ScreenCapture::update_capture_region(int x, int y, int w, int h) {
 
 namespace {
 
+inline uint8_t clip8u(long v) {
+  return static_cast<uint8_t>(v < 0 ? 0 : (v > 255 ? 255 : v));
+}
+
 // Watermark: raw ".bgra" file = u32 width, u32 height, then BGRA pixels.
 // (The Python layer converts PNG -> .bgra via PIL; keeps libpng out of the
 // native build.) Locations: 1=TL 2=TR 3=BL 4=BR 5=center 6=animated.
@@ -187,6 +191,7 @@ void ScreenCapture::run() {
                    settings_.watermark_path.c_str());
   }
   uint64_t wm_frame = 0;
+  uint64_t last_cursor_serial = ~0ULL;
 
   // CBR rate control (video_cbr_mode): VBV-fullness PI controller mapping
   // produced bytes to a per-frame QP within [min_qp, max_qp]. CRF mode
@@ -217,15 +222,47 @@ void ScreenCapture::run() {
     if (!src->acquire(frame)) break;
     frames_captured_.fetch_add(1);
 
-    if (watermark.w > 0) {
+    CursorImage cursor;
+    bool have_cursor = false;
+    if (settings_.capture_cursor || cursor_cb_)
+      have_cursor = src->cursor(cursor);
+
+    if (watermark.w > 0 || (settings_.capture_cursor && have_cursor)) {
       // composite on a scratch copy (the source buffer may persist)
       size_t bytes = static_cast<size_t>(frame.stride) * frame.height;
       wm_scratch.resize(bytes);
       std::memcpy(wm_scratch.data(), frame.data, bytes);
-      watermark.composite(wm_scratch.data(), frame.width, frame.height,
-                          frame.stride, settings_.watermark_location,
-                          wm_frame++);
+      if (watermark.w > 0)
+        watermark.composite(wm_scratch.data(), frame.width, frame.height,
+                            frame.stride, settings_.watermark_location,
+                            wm_frame++);
+      if (settings_.capture_cursor && have_cursor) {
+        for (int cy = 0; cy < cursor.height; ++cy) {
+          int fy = cursor.y + cy;
+          if (fy < 0 || fy >= frame.height) continue;
+          uint8_t* dst = wm_scratch.data() +
+                         static_cast<size_t>(fy) * frame.stride;
+          for (int cx = 0; cx < cursor.width; ++cx) {
+            int fx = cursor.x + cx;
+            if (fx < 0 || fx >= frame.width) continue;
+            uint32_t p = cursor.argb[static_cast<size_t>(cy) * cursor.width +
+                                     cx];
+            int a = p >> 24;
+            if (a == 0) continue;
+            uint8_t* d = dst + fx * 4;
+            // XFixes pixels are premultiplied ARGB
+            d[0] = clip8u(((p & 0xFF) * 255 + d[0] * (255 - a)) / 255);
+            d[1] = clip8u((((p >> 8) & 0xFF) * 255 + d[1] * (255 - a)) / 255);
+            d[2] = clip8u((((p >> 16) & 0xFF) * 255 + d[2] * (255 - a)) / 255);
+          }
+        }
+      }
       frame.data = wm_scratch.data();
+    }
+    if (cursor_cb_ && have_cursor && cursor.serial != last_cursor_serial) {
+      last_cursor_serial = cursor.serial;
+      cursor_cb_(cursor.width, cursor.height, cursor.hot_x, cursor.hot_y,
+                 cursor.argb.data(), cursor.argb.size());
     }
 
     // damage update (skipped in fullframe mode to save the diff cost)

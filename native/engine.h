@@ -70,6 +70,14 @@ class ScreenCapture {
   // Collapses concurrent requests into one per-frame flag.
   void request_idr_frame() { idr_requested_.store(true); }
 
+  // Cursor shape/position callback (native thread): fired when the cursor
+  // shape changes (reference set_cursor_callback contract).
+  using CursorCallback =
+      std::function<void(int w, int h, int hot_x, int hot_y,
+                         const uint32_t* argb, size_t count)>;
+  void set_cursor_callback(CursorCallback cb) { cursor_cb_ = std::move(cb); }
+  void clear_cursor_callback() { cursor_cb_ = nullptr; }
+
   // Live tunables (no restart).
   void update_framerate(double fps) { fps_.store(fps); }
   void update_video_bitrate(int kbps) { bitrate_kbps_.store(kbps); }
@@ -90,6 +98,7 @@ class ScreenCapture {
 
   CaptureSettings settings_;
   StripeCallback cb_;
+  CursorCallback cursor_cb_;
   std::thread thread_;
   std::atomic<bool> running_{false};
   std::atomic<bool> stop_{false};

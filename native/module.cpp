@@ -93,6 +93,25 @@ PYBIND11_MODULE(_native, m) {
           "stripe_type) runs on the native thread.")
       .def("stop_capture", &ScreenCapture::stop_capture,
            py::call_guard<py::gil_scoped_release>())
+      .def(
+          "set_cursor_callback",
+          [](ScreenCapture& self, py::function cb) {
+            // set BEFORE start_capture (callback read on capture thread)
+            self.clear_cursor_callback();
+            self.set_cursor_callback(
+                [cb](int w, int h, int hx, int hy, const uint32_t* argb,
+                     size_t n) {
+                  py::gil_scoped_acquire gil;
+                  try {
+                    cb(w, h, hx, hy,
+                       py::bytes(reinterpret_cast<const char*>(argb),
+                                 n * 4));
+                  } catch (py::error_already_set& e) {
+                    e.discard_as_unraisable("hipflux cursor callback");
+                  }
+                });
+          },
+          "callback(width, height, hot_x, hot_y, argb_bytes) on shape change")
       .def_property_readonly("is_capturing", &ScreenCapture::is_capturing)
       .def("request_idr_frame", &ScreenCapture::request_idr_frame)
       .def("update_framerate", &ScreenCapture::update_framerate)

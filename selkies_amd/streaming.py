@@ -183,6 +183,12 @@ class StreamingService:
             loop.call_soon_threadsafe(self._fanout, _display, data,
                                       frame_id, y, is_keyframe)
 
+        if display == "primary" and not self.settings.capture_cursor:
+            # client-side cursor rendering: push shape updates over control
+            def on_cursor(w, h, hx, hy, argb):
+                loop.call_soon_threadsafe(self._broadcast_cursor, w, h, hx,
+                                          hy, argb)
+            cap.set_cursor_callback(on_cursor)
         cap.start_capture(on_stripe, self.build_capture_settings(display))
         logger.info("capture started for %s (pipeline=%s)", display,
                     cap.pipeline)
@@ -263,6 +269,17 @@ class StreamingService:
                 continue
             if cs.relay.offer(data, y, is_keyframe):
                 cs.note_sent(frame_id)
+
+    def _broadcast_cursor(self, w, h, hx, hy, argb):
+        import base64
+        msg = P.encode_control("CURSOR", {
+            "width": w, "height": h, "hot_x": hx, "hot_y": hy,
+            "argb_b64": base64.b64encode(argb).decode()})
+        for cs in list(self.clients.values()):
+            try:
+                asyncio.get_running_loop().create_task(cs.ws.send_str(msg))
+            except Exception:
+                pass
 
     # ---- backpressure (reference selkies.py:2261 behavior) -----------------
     def _backpressure_tick(self):

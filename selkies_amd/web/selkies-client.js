@@ -257,6 +257,26 @@ function applyServerSettings(payload) {
   }
 }
 
+/* server-pushed cursor shape -> CSS cursor on the canvas */
+function applyCursor(c) {
+  const bin = atob(c.argb_b64);
+  const n = c.width * c.height;
+  const oc = document.createElement("canvas");
+  oc.width = c.width; oc.height = c.height;
+  const octx = oc.getContext("2d");
+  const img = octx.createImageData(c.width, c.height);
+  for (let i = 0; i < n; i++) {
+    // little-endian u32 ARGB -> bytes B,G,R,A
+    const b = bin.charCodeAt(i * 4), g = bin.charCodeAt(i * 4 + 1);
+    const r = bin.charCodeAt(i * 4 + 2), a = bin.charCodeAt(i * 4 + 3);
+    img.data[i * 4] = r; img.data[i * 4 + 1] = g;
+    img.data[i * 4 + 2] = b; img.data[i * 4 + 3] = a;
+  }
+  octx.putImageData(img, 0, 0);
+  canvas.style.cursor =
+      `url(${oc.toDataURL()}) ${c.hot_x} ${c.hot_y}, auto`;
+}
+
 /* ---------------- transport ---------------- */
 
 function connect() {
@@ -279,6 +299,8 @@ function connect() {
       const rest = i < 0 ? "" : ev.data.slice(i + 1);
       if (verb === "SETTINGS_PAYLOAD") {
         try { applyServerSettings(JSON.parse(rest)); } catch (e) {}
+      } else if (verb === "CURSOR") {
+        try { applyCursor(JSON.parse(rest)); } catch (e) {}
       } else if (verb === "clipboard") {
         try { navigator.clipboard.writeText(atob(rest)); } catch (e) {}
       }
