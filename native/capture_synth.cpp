@@ -36,13 +36,13 @@ class SyntheticSource : public FrameSource {
   }
 
   bool acquire(RawFrame& out) override {
+    ++frame_;  // also advances the synthetic cursor on static patterns
     if (pattern_ == "noise") {
-      uint64_t s = seed_ + ++frame_;
+      uint64_t s = seed_ + frame_;
       auto* p64 = reinterpret_cast<uint64_t*>(buf_.data());
       for (size_t i = 0; i < buf_.size() / 8; ++i) p64[i] = xs64(s);
       mask_x();
     } else if (pattern_ == "desktop") {
-      ++frame_;
       // moving 256x192 "window" bouncing over the static background
       int box_w = std::min(256, w_), box_h = std::min(192, h_);
       int span_x = std::max(1, w_ - box_w), span_y = std::max(1, h_ - box_h);
