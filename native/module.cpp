@@ -462,6 +462,39 @@ PYBIND11_MODULE(_native, m) {
            })
       .def_property_readonly("buffered", &AudioPlayback::buffered);
 
+  m.def(
+      "screenshot",
+      [](const std::string& backend, const std::string& display, int w,
+         int h) {
+        std::unique_ptr<FrameSource> src;
+        {
+          py::gil_scoped_release rel;
+          if (backend == "x11" || (backend == "auto" && !display.empty()))
+            src = make_x11_source(display, 0, 0, w, h);
+          if (!src) {
+            std::string pattern = "desktop";
+            auto pos = backend.find(':');
+            if (pos != std::string::npos) pattern = backend.substr(pos + 1);
+            src = make_synthetic_source(w > 0 ? w : 1920, h > 0 ? h : 1080,
+                                        pattern);
+          }
+        }
+        RawFrame f;
+        if (!src->acquire(f)) throw std::runtime_error("capture failed");
+        // tightly pack
+        std::vector<uint8_t> out(static_cast<size_t>(f.width) * f.height * 4);
+        for (int y = 0; y < f.height; ++y)
+          std::memcpy(out.data() + static_cast<size_t>(y) * f.width * 4,
+                      f.data + static_cast<size_t>(y) * f.stride,
+                      static_cast<size_t>(f.width) * 4);
+        return py::make_tuple(
+            py::bytes(reinterpret_cast<char*>(out.data()), out.size()),
+            f.width, f.height);
+      },
+      py::arg("backend") = "auto", py::arg("display") = "",
+      py::arg("w") = 0, py::arg("h") = 0,
+      "One-shot framebuffer grab -> (bgrx, w, h).");
+
   m.def("hip_device_count", &hip_device_count,
         "Number of usable HIP devices (0 on CPU-only hosts).");
   m.def("__version__", [] { return "0.1.0"; });

@@ -56,6 +56,8 @@ class CentralizedStreamServer:
         app.router.add_get("/api/stats", self.handle_stats)
         app.router.add_get("/api/settings", self.handle_settings)
         app.router.add_get("/metrics", self.handle_metrics)
+        from .computer_use import ComputerUseAPI
+        ComputerUseAPI(self.settings, self.streaming.input).register(app)
         app.router.add_post("/api/upload", self.handle_upload)
         app.router.add_get("/api/download", self.handle_download)
         app.router.add_get("/api/files", self.handle_files)
