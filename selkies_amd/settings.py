@@ -178,6 +178,8 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("enable_clipboard", bool, True, "Bidirectional clipboard sync.",
                client=True),
     SettingDef("enable_gamepad", bool, True, "Gamepad passthrough (interposer/uinput)."),
+    SettingDef("enable_shared", bool, False,
+               "Allow multiple controlling clients (shared input)."),
     SettingDef("enable_command_input", bool, False,
                "Allow the 'cmd' wire verb to run shell commands."),
 

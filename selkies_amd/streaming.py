@@ -36,10 +36,12 @@ logger = logging.getLogger("selkies.streaming")
 
 
 class ClientState:
-    def __init__(self, ws, relay: VideoRelay, display: str = "primary"):
+    def __init__(self, ws, relay: VideoRelay, display: str = "primary",
+                 role: str = "viewer"):
         self.ws = ws
         self.relay = relay
         self.display = display
+        self.role = role
         self.last_acked_frame = -1
         self.last_sent_frame = -1
         self.ack_rtt_ms = 50.0
@@ -303,6 +305,19 @@ class StreamingService:
         display = request.query.get("display", "primary")
         if display not in ("primary", "display2"):
             display = "primary"
+        # role/slot policy (reference roles: one controller, shared viewers;
+        # signaling_server.py allowed_client_slots behavior)
+        want_role = request.query.get("role", "")
+        has_controller = any(c.role == "controller"
+                             for c in self.clients.values())
+        if want_role == "viewer":
+            role = "viewer"
+        elif not has_controller:
+            role = "controller"
+        elif self.settings.enable_shared and want_role == "controller":
+            role = "controller"
+        else:
+            role = "viewer"
         if display == "display2" and not self.settings.second_display:
             self.settings.set("second_display", True)
         relay = VideoRelay(
@@ -310,11 +325,12 @@ class StreamingService:
             request_idr=lambda d=display: self.request_idr(d),
             bitrate_bps=self.settings.video_bitrate_kbps * 1000.0)
         relay.start()
-        state = ClientState(ws, relay, display)
+        state = ClientState(ws, relay, display, role)
         self.clients[ws] = state
 
         try:
             await ws.send_str(P.encode_control("MODE", "websockets"))
+            await ws.send_str(P.encode_control("ROLE", role))
             await ws.send_str(P.encode_control(
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload()))
@@ -337,6 +353,15 @@ class StreamingService:
         finally:
             self.clients.pop(ws, None)
             await relay.stop()
+            # promote the oldest remaining client when the controller leaves
+            if state.role == "controller" and self.clients:
+                nxt = next(iter(self.clients.values()))
+                nxt.role = "controller"
+                try:
+                    await nxt.ws.send_str(P.encode_control("ROLE",
+                                                           "controller"))
+                except Exception:
+                    pass
             if not any(c.display == display for c in self.clients.values()):
                 self.stop_capture(display)
             if not self.clients:
@@ -363,8 +388,11 @@ class StreamingService:
             self.request_idr()
             return None
         if verb == "js" and self.gamepads is not None:
-            await self.gamepads.handle(text)
+            if state.role == "controller" or self.settings.enable_shared:
+                await self.gamepads.handle(text)
             return None
+        if state.role != "controller" and not self.settings.enable_shared:
+            return None          # viewers cannot inject input
         return self.input.on_message(text)
 
     def _on_binary(self, state: ClientState, data: bytes):

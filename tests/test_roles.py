@@ -1,0 +1,69 @@
+"""Role/slot policy: one controller, viewers can't inject, promotion."""
+
+import asyncio
+
+import pytest
+
+hipflux = pytest.importorskip("hipflux")
+if not hipflux.native_available():
+    pytest.skip("hipflux native module not built", allow_module_level=True)
+
+import aiohttp
+from aiohttp import WSMsgType
+
+from selkies_amd.input_handler import RecordingBackend
+from test_server import make_server, start_on_free_port
+
+
+@pytest.fixture()
+def loop():
+    loop = asyncio.new_event_loop()
+    yield loop
+    loop.close()
+
+
+async def read_role(ws, timeout=5):
+    deadline = asyncio.get_event_loop().time() + timeout
+    while asyncio.get_event_loop().time() < deadline:
+        msg = await ws.receive(timeout=timeout)
+        if msg.type == WSMsgType.TEXT and msg.data.startswith("ROLE,"):
+            return msg.data.split(",", 1)[1]
+    return None
+
+
+def test_controller_assignment_and_promotion(loop):
+    async def main():
+        server = make_server()
+        backend = RecordingBackend()
+        server.streaming.input.backend = backend
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                url = f"http://127.0.0.1:{port}/websockets"
+                ws1 = await sess.ws_connect(url)
+                assert await read_role(ws1) == "controller"
+                ws2 = await sess.ws_connect(url)
+                assert await read_role(ws2) == "viewer"
+
+                # viewer input is ignored
+                await ws2.send_str("kd,120")
+                await asyncio.sleep(0.2)
+                assert ("key", 120, True) not in backend.events
+                # controller input lands
+                await ws1.send_str("kd,121")
+                await asyncio.sleep(0.2)
+                assert ("key", 121, True) in backend.events
+
+                # controller leaves -> viewer promoted
+                await ws1.close()
+                assert await read_role(ws2) == "controller"
+                await ws2.send_str("kd,122")
+                await asyncio.sleep(0.2)
+                assert ("key", 122, True) in backend.events
+                await ws2.close()
+        finally:
+            server.streaming.stop_capture()
+            server.streaming.stop_audio()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
