@@ -237,6 +237,7 @@ class HipH264Pipeline : public EncodePipeline {
     if (h_levels_) (void)hipHostFree(h_levels_);
     if (h_meta_) (void)hipHostFree(h_meta_);
     if (h_jobs_) (void)hipHostFree(h_jobs_);
+    if (h_stage_) (void)hipHostFree(h_stage_);
   }
 
   void encode_frame(const RawFrame& frame, const FrameContext& ctx,
@@ -245,14 +246,28 @@ class HipH264Pipeline : public EncodePipeline {
       alloc_for(frame.width, frame.height);
     const int qp = std::min(51, std::max(0, ctx.crf));
 
-    // upload + CSC
+    // upload + CSC. Prefer zero-copy DMA from the (registered) capture
+    // buffer; if registration failed, stage through a pinned buffer — an
+    // unpinned async copy silently degrades to a slow sync path.
     const uint8_t* src = frame.data;
     size_t frame_bytes = static_cast<size_t>(frame.stride) * frame.height;
-    if (!registered_.count(const_cast<uint8_t*>(src))) {
+    auto reg_it = registered_.find(const_cast<uint8_t*>(src));
+    if (reg_it == registered_.end()) {
       hipError_t e = hipHostRegister(const_cast<uint8_t*>(src), frame_bytes,
                                      hipHostRegisterDefault);
-      registered_[const_cast<uint8_t*>(src)] = (e == hipSuccess);
       if (e != hipSuccess) (void)hipGetLastError();
+      reg_it = registered_.emplace(const_cast<uint8_t*>(src),
+                                   e == hipSuccess).first;
+    }
+    if (!reg_it->second) {
+      if (h_stage_bytes_ < frame_bytes) {
+        if (h_stage_) (void)hipHostFree(h_stage_);
+        HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_stage_),
+                                frame_bytes, hipHostMallocDefault));
+        h_stage_bytes_ = frame_bytes;
+      }
+      std::memcpy(h_stage_, src, frame_bytes);
+      src = h_stage_;
     }
     HIP_CHECK(hipMemcpyAsync(d_frame_, src, frame_bytes,
                              hipMemcpyHostToDevice, stream_));
@@ -521,6 +536,8 @@ class HipH264Pipeline : public EncodePipeline {
   int16_t* h_levels_ = nullptr;
   int* h_meta_ = nullptr;
   h264gpu::RowJob* h_jobs_ = nullptr;
+  uint8_t* h_stage_ = nullptr;
+  size_t h_stage_bytes_ = 0;
   std::vector<hipEvent_t> batch_events_;
   std::vector<void*> device_ptrs_;
   std::vector<StripeState> stripes_;
