@@ -119,7 +119,14 @@ class StreamingService:
         cs.target_fps = float(s.framerate)
         cs.output_mode = 0 if s.encoder == "jpeg" else 1
         cs.use_cpu = bool(s.use_cpu)
-        cs.gpu_id = s.gpu_id if s.gpu_id >= 0 else 0
+        # per-session GPU placement: explicit session_gpus list round-robins
+        # displays/sessions over devices (SURVEY.md §5.8 multi-GPU facility)
+        gpus = [int(g) for g in s.session_gpus.split(",") if g.strip()]
+        if gpus:
+            idx = 0 if display == "primary" else 1
+            cs.gpu_id = gpus[idx % len(gpus)]
+        else:
+            cs.gpu_id = s.gpu_id if s.gpu_id >= 0 else 0
         cs.capture_backend = s.capture_backend
         cs.display = s.display
         cs.video_bitrate_kbps = s.video_bitrate_kbps
