@@ -91,6 +91,9 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("https_key", str, "", "TLS key path (hot-reloaded)."),
     SettingDef("allowed_ws_origins", str, "", "Comma list of allowed WS origins ('' = any)."),
     SettingDef("enable_metrics_http", bool, False, "Expose Prometheus /metrics."),
+    SettingDef("webrtc_udp_port", int, 0,
+               "UDP port for the WebRTC ICE-lite endpoint (0 = ephemeral).",
+               value_range=(0, 65535)),
     SettingDef("metrics_http_token", str, "", "Bearer token guarding /metrics."),
 
     # ---- display / capture ----

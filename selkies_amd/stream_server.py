@@ -40,6 +40,7 @@ class CentralizedStreamServer:
             allow_upload="upload" in dirs,
             allow_download="download" in dirs)
         self.app = web.Application(middlewares=[self._auth_middleware])
+        self.webrtc = None
         self.started_at = time.time()
         self._runner: Optional[web.AppRunner] = None
         self._ssl_ctx: Optional[ssl.SSLContext] = None
@@ -58,6 +59,7 @@ class CentralizedStreamServer:
         app.router.add_get("/metrics", self.handle_metrics)
         from .computer_use import ComputerUseAPI
         ComputerUseAPI(self.settings, self.streaming.input).register(app)
+        app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
         app.router.add_post("/api/upload", self.handle_upload)
         app.router.add_get("/api/download", self.handle_download)
         app.router.add_get("/api/files", self.handle_files)
@@ -166,6 +168,26 @@ class CentralizedStreamServer:
         return web.Response(body=generate_latest(reg),
                             content_type="text/plain; version=0.0.4")
 
+    # ---- WebRTC signaling ---------------------------------------------------
+    async def handle_webrtc_offer(self, request):
+        """HTTP signaling: browser POSTs its SDP offer, gets the ice-lite
+        answer. (Both transports are always live — WS is the default,
+        WebRTC the alternative; reference keeps them at feature parity.)"""
+        if self.webrtc is None:
+            from .webrtc_service import WebRTCService
+            self.webrtc = WebRTCService(self.settings, self.streaming)
+            await self.webrtc.start(self.settings.webrtc_udp_port)
+        try:
+            body = await request.json()
+            offer_sdp = body["sdp"]
+        except Exception:
+            raise web.HTTPBadRequest(reason="JSON {sdp} required")
+        try:
+            answer = self.webrtc.handle_offer(offer_sdp)
+        except Exception as exc:
+            raise web.HTTPBadRequest(reason=f"offer rejected: {exc}")
+        return web.json_response({"type": "answer", "sdp": answer})
+
     # ---- file transfers -----------------------------------------------------
     async def handle_upload(self, request):
         name = request.query.get("name", "")
@@ -258,6 +280,9 @@ class CentralizedStreamServer:
 
     async def stop(self):
         self.streaming.stop_capture()
+        self.streaming.stop_audio()
+        if self.webrtc is not None:
+            await self.webrtc.stop()
         if self._runner is not None:
             await self._runner.cleanup()
 

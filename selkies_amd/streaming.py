@@ -310,7 +310,7 @@ class StreamingService:
         await ws.prepare(request)
 
         display = request.query.get("display", "primary")
-        if display not in ("primary", "display2"):
+        if display not in ("primary", "display2", "none"):
             display = "primary"
         # role/slot policy (reference roles: one controller, shared viewers;
         # signaling_server.py allowed_client_slots behavior)
@@ -341,9 +341,10 @@ class StreamingService:
             await ws.send_str(P.encode_control(
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload()))
-            self.start_capture(display)
+            if display != "none":   # "none" = control-only (WebRTC video)
+                self.start_capture(display)
+                self.request_idr(display)
             self.start_audio()
-            self.request_idr(display)
             if self._stats_task is None or self._stats_task.done():
                 self._stats_task = asyncio.get_running_loop().create_task(
                     self._stats_pusher())

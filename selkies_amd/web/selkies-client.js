@@ -329,7 +329,54 @@ window.addEventListener("resize", () => {
   }, 400);
 });
 
+/* ---------------- WebRTC transport (?transport=webrtc) ----------------
+ * Video arrives over DTLS-SRTP from the ice-lite endpoint; input/control
+ * still ride the WebSocket (connected with display=none). */
+async function connectWebRTC() {
+  const pc = new RTCPeerConnection();
+  pc.addTransceiver("video", { direction: "recvonly" });
+  pc.ontrack = (ev) => {
+    const vid = document.createElement("video");
+    vid.autoplay = true; vid.playsInline = true; vid.muted = true;
+    vid.srcObject = ev.streams[0] || new MediaStream([ev.track]);
+    vid.play().catch(() => {});
+    const draw = () => {
+      if (vid.videoWidth) {
+        ensureCanvas(vid.videoWidth, vid.videoHeight);
+        ctx2d.drawImage(vid, 0, 0);
+      }
+      requestAnimationFrame(draw);
+    };
+    draw();
+    stateEl.textContent = "connected (webrtc)";
+  };
+  const offer = await pc.createOffer();
+  await pc.setLocalDescription(offer);
+  await new Promise((res) => {
+    if (pc.iceGatheringState === "complete") return res();
+    pc.onicegatheringstatechange = () =>
+        pc.iceGatheringState === "complete" && res();
+    setTimeout(res, 1000);
+  });
+  const r = await fetch("/api/webrtc/offer" + location.search, {
+    method: "POST",
+    headers: { "Content-Type": "application/json" },
+    body: JSON.stringify({ sdp: pc.localDescription.sdp }),
+  });
+  const ans = await r.json();
+  await pc.setRemoteDescription({ type: "answer", sdp: ans.sdp });
+}
+
 hookInput();
 hookHud();
+const params = new URLSearchParams(location.search);
+if (params.get("transport") === "webrtc") {
+  connectWebRTC().catch((e) => {
+    stateEl.textContent = "webrtc failed: " + e;
+  });
+  /* control-only WS (no video fan-out) */
+  params.set("display", "none");
+  history.replaceState(null, "", "?" + params.toString());
+}
 connect();
 canvas.focus();

@@ -1,0 +1,195 @@
+"""WebRTC streaming service: ICE-lite + DTLS-SRTP video transport.
+
+The reference's WebRTC mode (SURVEY.md §2.1 webrtc_mode.py + vendored
+stacks) rebuilt on this environment's primitives (ctypes-OpenSSL DTLS, own
+SRTP/RTP/ICE — selkies_amd/webrtc/). Round-1 scope: H.264 video sendonly
+over DTLS-SRTP with PLI/FIR-driven IDR and RTCP sender reports; signaling
+is plain HTTP (POST /api/webrtc/offer); input/control ride the WebSocket
+(connect with ?display=none for control-only).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import secrets
+import time
+from typing import Optional
+
+import hipflux
+
+from .webrtc import dtls, ice, rtp, sdp
+from .webrtc.srtp import SrtpSession, is_rtcp
+
+logger = logging.getLogger("selkies.webrtc")
+
+
+class PeerState:
+    def __init__(self, addr):
+        self.addr = addr
+        self.dtls: Optional[dtls.DtlsEndpoint] = None
+        self.srtp_out: Optional[SrtpSession] = None
+        self.srtp_in: Optional[SrtpSession] = None
+        self.connected = False
+        self.last_sr = 0.0
+
+
+class WebRTCService:
+    """One UDP endpoint serving all WebRTC peers (ice-lite)."""
+
+    def __init__(self, settings, streaming):
+        self.settings = settings
+        self.streaming = streaming       # reuse capture orchestration
+        self.cert = dtls.Certificate()
+        self.ufrag, self.pwd = ice.make_ice_credentials()
+        self.ssrc = secrets.randbits(31) | 1
+        self.packetizer = rtp.H264Packetizer(self.ssrc)
+        self.peers: dict[tuple, PeerState] = {}
+        self.transport = None
+        self.port = 0
+        self.host_ip = ice.default_host_ip()
+        self.capture: Optional[hipflux.ScreenCapture] = None
+        self._ts_base = time.monotonic()
+        self.frames_sent = 0
+
+    # ---- lifecycle ---------------------------------------------------------
+    async def start(self, port: int = 0):
+        loop = asyncio.get_running_loop()
+        self.transport, _ = await loop.create_datagram_endpoint(
+            lambda: _Proto(self), local_addr=("0.0.0.0", port))
+        self.port = self.transport.get_extra_info("sockname")[1]
+        logger.info("webrtc ice-lite endpoint on %s:%d", self.host_ip,
+                    self.port)
+
+    async def stop(self):
+        self.stop_video()
+        if self.transport:
+            self.transport.close()
+            self.transport = None
+        for p in self.peers.values():
+            if p.dtls:
+                p.dtls.close()
+        self.peers.clear()
+
+    # ---- signaling -----------------------------------------------------------
+    def handle_offer(self, offer_sdp: str) -> str:
+        offer = sdp.parse_offer(offer_sdp)
+        video = next((m for m in offer.media if m.kind == "video"), None)
+        if video is not None and video.h264_pts:
+            self.packetizer.pt = video.h264_pts[0][0]
+        answer = sdp.build_answer(offer, self.ufrag, self.pwd,
+                                  self.cert.fingerprint, self.host_ip,
+                                  self.port, self.ssrc)
+        return answer
+
+    # ---- media ---------------------------------------------------------------
+    def start_video(self):
+        if self.capture is not None and self.capture.is_capturing:
+            return
+        loop = asyncio.get_running_loop()
+        cs = self.streaming.build_capture_settings("primary")
+        cs.stripe_height = ((cs.capture_height + 15) & ~15)  # full frame
+        cs.output_mode = 1
+        self.capture = hipflux.ScreenCapture()
+
+        def on_stripe(data, frame_id, y, width, height, is_keyframe,
+                      capture_ts_ms, encode_done_ms, stripe_type):
+            payload = bytes(data[10:])  # drop the WS wire header
+            loop.call_soon_threadsafe(self._send_frame, payload)
+
+        self.capture.start_capture(on_stripe, cs)
+        logger.info("webrtc video capture started (pipeline=%s)",
+                    self.capture.pipeline)
+
+    def stop_video(self):
+        if self.capture is not None:
+            self.capture.stop_capture()
+            self.capture = None
+
+    def _send_frame(self, annexb: bytes):
+        if not any(p.connected for p in self.peers.values()):
+            return
+        ts90k = int((time.monotonic() - self._ts_base) * 90000)
+        packets = self.packetizer.packetize(annexb, ts90k)
+        self.frames_sent += 1
+        now = time.monotonic()
+        for peer in list(self.peers.values()):
+            if not peer.connected:
+                continue
+            for pkt in packets:
+                try:
+                    self.transport.sendto(peer.srtp_out.protect_rtp(pkt),
+                                          peer.addr)
+                except Exception as exc:
+                    logger.debug("srtp send failed: %r", exc)
+            if now - peer.last_sr > 1.0:
+                peer.last_sr = now
+                sr = rtp.build_sender_report(self.ssrc, ts90k,
+                                             self.packetizer.packets_sent,
+                                             self.packetizer.bytes_sent)
+                try:
+                    self.transport.sendto(peer.srtp_out.protect_rtcp(sr),
+                                          peer.addr)
+                except Exception:
+                    pass
+
+    # ---- datagram demux -------------------------------------------------------
+    def on_datagram(self, data: bytes, addr):
+        if ice.is_stun(data):
+            resp = ice.binding_response(data, addr, self.pwd)
+            self.transport.sendto(resp, addr)
+            if addr not in self.peers:
+                self.peers[addr] = PeerState(addr)
+                logger.info("webrtc peer candidate %s", addr)
+            return
+        peer = self.peers.get(addr)
+        if peer is None:
+            return
+        first = data[0] if data else 0
+        if 20 <= first <= 63:        # DTLS record
+            if peer.dtls is None:
+                peer.dtls = dtls.DtlsEndpoint(self.cert, server=True)
+            peer.dtls.put_datagram(data)
+            for out in peer.dtls.take_datagrams():
+                self.transport.sendto(out, addr)
+            if peer.dtls.handshake_done and not peer.connected:
+                (ck, cs), (sk, ss) = peer.dtls.export_srtp_keys()
+                # we are the DTLS server: send with server keys,
+                # receive with client keys
+                peer.srtp_out = SrtpSession(sk, ss)
+                peer.srtp_in = SrtpSession(ck, cs)
+                peer.connected = True
+                logger.info("webrtc peer %s connected (DTLS-SRTP up)", addr)
+                self.start_video()
+                if self.capture:
+                    self.capture.request_idr_frame()
+        elif first >= 128:           # SRTP/SRTCP
+            if peer.connected and is_rtcp(data):
+                try:
+                    plain = peer.srtp_in.unprotect_rtcp(data)
+                except ValueError:
+                    return
+                for report in rtp.parse_rtcp(plain):
+                    if report["type"] in ("PLI", "FIR"):
+                        if self.capture:
+                            self.capture.request_idr_frame()
+
+    def stats(self) -> dict:
+        return {
+            "port": self.port,
+            "peers": len(self.peers),
+            "connected": sum(p.connected for p in self.peers.values()),
+            "frames_sent": self.frames_sent,
+            "packets_sent": self.packetizer.packets_sent,
+        }
+
+
+class _Proto(asyncio.DatagramProtocol):
+    def __init__(self, svc: WebRTCService):
+        self.svc = svc
+
+    def datagram_received(self, data, addr):
+        try:
+            self.svc.on_datagram(data, addr)
+        except Exception as exc:
+            logger.warning("webrtc datagram error: %r", exc)
