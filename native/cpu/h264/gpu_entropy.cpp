@@ -186,5 +186,34 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
   }
 }
 
+void assemble_gpu_row_nal(const uint32_t* words, int bits, bool idr,
+                          bool long_startcode, std::vector<uint8_t>& out) {
+  int nbytes = (bits + 7) / 8;
+  std::vector<uint8_t> rbsp(nbytes + 1, 0);
+  for (int i = 0; i < nbytes; ++i)
+    rbsp[i] = static_cast<uint8_t>(words[i / 4] >> (24 - 8 * (i % 4)));
+  // rbsp stop bit (trailing bits beyond `bits` are zero by construction)
+  int stop_byte = bits / 8;
+  rbsp[stop_byte] |= 0x80 >> (bits % 8);
+  int total = stop_byte + 1;
+
+  if (long_startcode) out.push_back(0);
+  out.push_back(0);
+  out.push_back(0);
+  out.push_back(1);
+  out.push_back(static_cast<uint8_t>(
+      idr ? ((3 << 5) | 5) : ((2 << 5) | 1)));
+  int zeros = 0;
+  for (int i = 0; i < total; ++i) {
+    uint8_t b = rbsp[i];
+    if (zeros >= 2 && b <= 3) {
+      out.push_back(3);
+      zeros = 0;
+    }
+    out.push_back(b);
+    zeros = (b == 0) ? zeros + 1 : 0;
+  }
+}
+
 }  // namespace h264
 }  // namespace hipflux

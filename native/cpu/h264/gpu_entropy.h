@@ -32,5 +32,10 @@ void encode_stripe_from_gpu(const GpuStripeParams& p,
 void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
                              std::vector<uint8_t>& out);
 
+// Wrap a GPU-packed slice RBSP (MSB-first u32 words, `bits` long) into an
+// Annex-B NAL: stop bit, emulation prevention, start code + header.
+void assemble_gpu_row_nal(const uint32_t* words, int bits, bool idr,
+                          bool long_startcode, std::vector<uint8_t>& out);
+
 }  // namespace h264
 }  // namespace hipflux

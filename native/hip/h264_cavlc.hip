@@ -1,0 +1,429 @@
+// GPU CAVLC entropy stage (gfx950).
+//
+// One wave per MB-row slice. Three phases inside the kernel:
+//  A. each lane encodes independent bitstring ITEMS (slice header, MB
+//     headers, residual blocks) into 64-byte staging slots — legal because
+//     every CAVLC context (nC neighbor totals, skip runs, MV predictors,
+//     CBP) is derivable in parallel from the levels/meta buffers without
+//     decoding bits;
+//  B. a wave prefix-sum over item bit-lengths yields every item's global
+//     bit offset;
+//  C. lanes scatter their items into the row's output bitstream with
+//     atomicOr word writes (bit order: MSB-first within u32 words).
+// The CPU keeps only RBSP stop-bit, emulation prevention and NAL wrap.
+//
+// Bit-exactness contract: identical output to the CPU packer
+// (cpu/h264/gpu_entropy.cpp) for the same levels/meta — enforced by
+// tests/test_gpu_h264.py::test_gpu_cavlc_matches_cpu.
+#include <hip/hip_runtime.h>
+
+#include "cavlc_tables_gpu.h"
+#include "h264_gpu_layout.h"
+#include "h264_kernels.h"
+
+namespace hipflux {
+namespace h264gpu {
+
+__constant__ int c_zig4[16] = {0, 1, 4, 8, 5, 2, 3, 6,
+                               9, 12, 13, 10, 7, 11, 14, 15};
+
+// Z-order block -> raster index within the MB
+__constant__ int c_zorder_raster[16] = {0, 1, 4, 5, 2, 3, 6, 7,
+                                        8, 9, 12, 13, 10, 11, 14, 15};
+
+struct DevBW {
+  uint64_t acc = 0;
+  int na = 0;
+  uint32_t* out;
+  int word = 0;
+  __device__ void u(uint32_t v, int n) {
+    if (n == 0) return;
+    acc = (acc << n) |
+          (uint64_t)(v & (uint32_t)((n == 32 ? 0ull : (1ull << n)) - 1ull));
+    na += n;
+    while (na >= 32) {
+      out[word++] = (uint32_t)(acc >> (na - 32));
+      na -= 32;
+    }
+  }
+  __device__ void ue(uint32_t v) {
+    uint32_t cw = v + 1;
+    int len = 31 - __clz(cw);
+    u(0, len);
+    u(cw, len + 1);
+  }
+  __device__ void se(int v) {
+    ue(v > 0 ? 2 * (uint32_t)v - 1 : 2 * (uint32_t)(-v));
+  }
+  __device__ int flush() {
+    int bits = word * 32 + na;
+    if (na) out[word++] = (uint32_t)(acc << (32 - na));
+    return bits;
+  }
+};
+
+// ---- residual block encode (mirrors cpu/h264/cavlc.h exactly) -------------
+__device__ void dev_cavlc_residual(DevBW& bw, const int* zz, int n, int nC) {
+  int coeffs[16], pos[16], tc = 0;
+  for (int i = 0; i < n; ++i)
+    if (zz[i]) {
+      coeffs[tc] = zz[i];
+      pos[tc] = i;
+      ++tc;
+    }
+  int t1 = 0;
+  for (int k = tc - 1; k >= 0 && t1 < 3; --k) {
+    if (coeffs[k] == 1 || coeffs[k] == -1)
+      ++t1;
+    else
+      break;
+  }
+  unsigned packed;
+  if (nC == -1)
+    packed = c_ctcdc[tc * 4 + t1];
+  else if (nC < 2)
+    packed = c_ct0[tc * 4 + t1];
+  else if (nC < 4)
+    packed = c_ct1[tc * 4 + t1];
+  else if (nC < 8)
+    packed = c_ct2[tc * 4 + t1];
+  else
+    packed = VLCPACK(6, tc == 0 ? 3 : (((tc - 1) << 2) | t1));
+  bw.u(packed & 0xFFFF, packed >> 16);
+  if (tc == 0) return;
+
+  for (int k = tc - 1; k >= tc - t1; --k) bw.u(coeffs[k] > 0 ? 0 : 1, 1);
+
+  int suffix_len = (tc > 10 && t1 < 3) ? 1 : 0;
+  bool first = true;
+  for (int k = tc - t1 - 1; k >= 0; --k) {
+    const int true_level = coeffs[k];
+    int level = coeffs[k];
+    if (first && t1 < 3) level += level > 0 ? -1 : 1;
+    first = false;
+    int code = level > 0 ? 2 * level - 2 : -2 * level - 1;
+    if (suffix_len == 0) {
+      if (code < 14) {
+        bw.u(1, code + 1);
+      } else if (code < 30) {
+        bw.u(1, 15);
+        bw.u(code - 14, 4);
+      } else {
+        bw.u(1, 16);
+        bw.u(code - 30, 12);
+      }
+    } else {
+      int prefix = code >> suffix_len;
+      if (prefix < 15) {
+        bw.u(1, prefix + 1);
+        bw.u(code & ((1 << suffix_len) - 1), suffix_len);
+      } else {
+        bw.u(1, 16);
+        bw.u(code - (15 << suffix_len), 12);
+      }
+    }
+    if (suffix_len == 0) suffix_len = 1;
+    if (abs(true_level) > (3 << (suffix_len - 1)) && suffix_len < 6)
+      ++suffix_len;
+  }
+
+  int total_zeros = pos[tc - 1] + 1 - tc;
+  if (tc < n) {
+    unsigned p = (nC == -1) ? c_tzcdc[(tc - 1) * 4 + total_zeros]
+                            : c_tz[(tc - 1) * 16 + total_zeros];
+    bw.u(p & 0xFFFF, p >> 16);
+  }
+  int zeros_left = total_zeros;
+  for (int k = tc - 1; k > 0 && zeros_left > 0; --k) {
+    int run = pos[k] - pos[k - 1] - 1;
+    int zl = zeros_left < 7 ? zeros_left : 7;
+    unsigned p = c_rb[(zl - 1) * 15 + run];
+    bw.u(p & 0xFFFF, p >> 16);
+    zeros_left -= run;
+  }
+}
+
+// ---- per-MB derived state ---------------------------------------------------
+// totals of a block (0 if the MB is not intra)
+__device__ inline int blk_total(const int16_t* L, int off, int first,
+                                int count) {
+  int t = 0;
+  for (int i = first; i < first + count; ++i) t += L[off + i] != 0;
+  return t;
+}
+
+struct MbInfo {
+  int mode;          // kSkip/kInter/kIntra (I rows: kIntra)
+  int luma_mode, chroma_mode;
+  int mvx, mvy;
+  int cbp_luma, cbp_chroma;
+};
+
+__device__ MbInfo load_mb(const int16_t* levels, const int* meta,
+                          size_t mb_index, bool i_slice) {
+  MbInfo m{};
+  int m0 = meta[mb_index * kMetaPerMb + 0];
+  m.mode = i_slice ? kIntra : (m0 & 3);
+  m.luma_mode = (m0 >> 2) & 7;
+  m.chroma_mode = (m0 >> 5) & 7;
+  int m1 = meta[mb_index * kMetaPerMb + 1];
+  m.mvx = (short)(m1 & 0xFFFF);
+  m.mvy = m1 >> 16;
+  if (m.mode == kIntra) {
+    const int16_t* L = levels + mb_index * kLevelsPerMb;
+    int any_ac = 0;
+    for (int b = 0; b < 16 && !any_ac; ++b)
+      any_ac |= blk_total(L, kLumaAcOff + b * 16, 1, 15) != 0;
+    m.cbp_luma = any_ac ? 15 : 0;
+    int any_cac = 0, any_cdc = 0;
+    for (int b = 0; b < 8 && !any_cac; ++b)
+      any_cac |= blk_total(L, kChromaAcOff + b * 16, 1, 15) != 0;
+    for (int i = 0; i < 8 && !any_cdc; ++i)
+      any_cdc |= L[kChromaDcOff + i] != 0;
+    m.cbp_chroma = any_cac ? 2 : (any_cdc ? 1 : 0);
+  }
+  return m;
+}
+
+// nC context for luma block (bx,by) of MB mbx (top row never available)
+__device__ int luma_nc(const int16_t* levels, const int* meta,
+                       size_t row_base, int mbx, int bx, int by,
+                       bool i_slice) {
+  if (bx > 0) {
+    const int16_t* L = levels + (row_base + mbx) * kLevelsPerMb;
+    MbInfo self = load_mb(levels, meta, row_base + mbx, i_slice);
+    if (self.cbp_luma == 0) return 0;
+    int r = by * 4 + (bx - 1);
+    return blk_total(L, kLumaAcOff + r * 16, 1, 15);
+  }
+  if (mbx == 0) return 0;
+  MbInfo left = load_mb(levels, meta, row_base + mbx - 1, i_slice);
+  if (left.mode != kIntra || left.cbp_luma == 0) return 0;
+  const int16_t* L = levels + (row_base + mbx - 1) * kLevelsPerMb;
+  return blk_total(L, kLumaAcOff + (by * 4 + 3) * 16, 1, 15);
+}
+
+__device__ int chroma_nc(const int16_t* levels, const int* meta,
+                         size_t row_base, int mbx, int comp, int cx, int cy,
+                         bool i_slice) {
+  if (cx > 0) {
+    MbInfo self = load_mb(levels, meta, row_base + mbx, i_slice);
+    if (self.cbp_chroma != 2) return 0;
+    const int16_t* L = levels + (row_base + mbx) * kLevelsPerMb;
+    return blk_total(L, kChromaAcOff + (comp * 4 + cy * 2) * 16, 1, 15);
+  }
+  if (mbx == 0) return 0;
+  MbInfo left = load_mb(levels, meta, row_base + mbx - 1, i_slice);
+  if (left.mode != kIntra || left.cbp_chroma != 2) return 0;
+  const int16_t* L = levels + (row_base + mbx - 1) * kLevelsPerMb;
+  return blk_total(L, kChromaAcOff + (comp * 4 + cy * 2 + 1) * 16, 1, 15);
+}
+
+// ---- the kernel -------------------------------------------------------------
+__global__ void __launch_bounds__(64) k_h264_cavlc_rows(
+    const int16_t* __restrict__ levels, const int* __restrict__ meta,
+    int mbw, const RowJob* __restrict__ jobs,
+    uint32_t* __restrict__ stage,       // [row][item][kStageWordsPerItem]
+    int* __restrict__ nbits,            // [row][item]
+    uint32_t* __restrict__ out,         // [row][out_stride_words]
+    int out_stride_words,
+    int* __restrict__ out_bits) {       // [row]
+  const RowJob job = jobs[blockIdx.x];
+  const int lane = threadIdx.x;
+  const bool i_slice = (job.flags & 1) != 0;
+  const size_t row_base = (size_t)job.mb_row * mbw;
+  const int nitems = items_per_row(mbw);
+  uint32_t* row_stage =
+      stage + (size_t)blockIdx.x * nitems * kStageWordsPerItem;
+  int* row_nbits = nbits + (size_t)blockIdx.x * nitems;
+  uint32_t* row_out = out + (size_t)blockIdx.x * out_stride_words;
+
+  // ---- skip runs (P rows): s_skiprun[mb] = run before this MB; trailing
+  __shared__ short s_skiprun[1024];
+  __shared__ int s_trailing;
+  if (lane == 0) {
+    int run = 0;
+    for (int mb = 0; mb < mbw; ++mb) {
+      MbInfo m = load_mb(levels, meta, row_base + mb, i_slice);
+      if (!i_slice && m.mode == kSkip) {
+        s_skiprun[mb] = -1;  // skipped MB: no items
+        ++run;
+      } else {
+        s_skiprun[mb] = (short)run;
+        run = 0;
+      }
+    }
+    s_trailing = run;
+  }
+  __syncthreads();
+
+  // ---- phase A: encode items
+  for (int item = lane; item < nitems; item += 64) {
+    DevBW bw;
+    bw.out = row_stage + (size_t)item * kStageWordsPerItem;
+    int bits = 0;
+    if (item == 0) {
+      // slice header
+      bw.ue(job.first_mb);
+      bw.ue(i_slice ? 7 : 5);
+      bw.ue(0);
+      bw.u(job.frame_num & 0xFFFF, 16);
+      if (i_slice) bw.ue(job.idr_pic_id);
+      if (!i_slice) {
+        bw.u(0, 1);
+        bw.u(0, 1);
+      }
+      if (i_slice) {
+        bw.u(0, 1);
+        bw.u(0, 1);
+      } else {
+        bw.u(0, 1);
+      }
+      bw.se(job.qp - 26);
+      bw.ue(1);  // disable_deblocking_filter_idc
+      bits = bw.flush();
+    } else if (item == nitems - 1) {
+      if (!i_slice && s_trailing > 0) bw.ue(s_trailing);
+      bits = bw.flush();
+    } else {
+      int mb = (item - 1) / kSlotsPerMb;
+      int slot = (item - 1) % kSlotsPerMb;
+      MbInfo m = load_mb(levels, meta, row_base + mb, i_slice);
+      const int16_t* L = levels + (row_base + mb) * kLevelsPerMb;
+      if (s_skiprun[mb] < 0) {
+        bits = 0;  // skipped MB
+      } else if (slot == 0) {
+        if (!i_slice) bw.ue(s_skiprun[mb]);
+        if (m.mode == kInter) {
+          bw.ue(0);  // P_L0_16x16
+          int mvpx = 0, mvpy = 0;
+          if (mb > 0) {
+            MbInfo left = load_mb(levels, meta, row_base + mb - 1, i_slice);
+            if (left.mode != kIntra) {  // skip or inter both carry MVs
+              mvpx = left.mvx;
+              mvpy = left.mvy;
+            }
+          }
+          bw.se(m.mvx - mvpx);
+          bw.se(m.mvy - mvpy);
+          bw.ue(0);  // cbp = 0
+        } else {
+          int i16 = 1 + m.luma_mode + 4 * m.cbp_chroma +
+                    12 * (m.cbp_luma ? 1 : 0);
+          bw.ue(i_slice ? i16 : 5 + i16);
+          bw.ue(m.chroma_mode);
+          bw.se(0);  // mb_qp_delta
+        }
+        bits = bw.flush();
+      } else if (m.mode != kIntra) {
+        bits = 0;
+      } else if (slot == 1) {
+        // luma DC
+        int zz[16];
+        for (int i = 0; i < 16; ++i)
+          zz[i] = L[kLumaDcOff + c_zig4[i]];
+        int nC = luma_nc(levels, meta, row_base, mb, 0, 0, i_slice);
+        dev_cavlc_residual(bw, zz, 16, nC);
+        bits = bw.flush();
+      } else if (slot < 18) {
+        if (m.cbp_luma) {
+          int blk = slot - 2;                      // Z-order index
+          int r = c_zorder_raster[blk];
+          int bx = r & 3, by = r >> 2;
+          int zz[15];
+          for (int i = 1; i < 16; ++i)
+            zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
+          int nC = luma_nc(levels, meta, row_base, mb, bx, by, i_slice);
+          dev_cavlc_residual(bw, zz, 15, nC);
+          bits = bw.flush();
+        }
+      } else if (slot < 20) {
+        if (m.cbp_chroma > 0) {
+          int comp = slot - 18;
+          int zz[4];
+          for (int i = 0; i < 4; ++i) zz[i] = L[kChromaDcOff + comp * 4 + i];
+          dev_cavlc_residual(bw, zz, 4, -1);
+          bits = bw.flush();
+        }
+      } else {
+        if (m.cbp_chroma == 2) {
+          int b = slot - 20;                       // cb0..3 then cr0..3
+          int comp = b >> 2, sub = b & 3;
+          int zz[15];
+          for (int i = 1; i < 16; ++i)
+            zz[i - 1] = L[kChromaAcOff + (comp * 4 + sub) * 16 + c_zig4[i]];
+          int nC = chroma_nc(levels, meta, row_base, mb, comp, sub & 1,
+                             sub >> 1, i_slice);
+          dev_cavlc_residual(bw, zz, 15, nC);
+          bits = bw.flush();
+        }
+      }
+    }
+    row_nbits[item] = bits;
+  }
+  __syncthreads();
+
+  // ---- phase B: exclusive prefix sum of nbits (wave-strided)
+  __shared__ int s_carry;
+  if (lane == 0) s_carry = 0;
+  __syncthreads();
+  for (int base = 0; base < nitems; base += 64) {
+    int idx = base + lane;
+    int v = idx < nitems ? row_nbits[idx] : 0;
+    // inclusive scan within the wave
+    int inc = v;
+    for (int d = 1; d < 64; d <<= 1) {
+      int other = __shfl_up(inc, d);
+      if (lane >= d) inc += other;
+    }
+    int excl = inc - v + s_carry;
+    if (idx < nitems) row_nbits[idx] = excl;   // becomes the offset
+    __syncthreads();
+    if (lane == 63) s_carry += inc;
+    __syncthreads();
+  }
+  int total_bits = s_carry;
+  if (lane == 0) out_bits[blockIdx.x] = total_bits;
+
+  // ---- zero the needed output words, then scatter
+  int total_words = (total_bits + 31) / 32 + 1;
+  for (int wdx = lane; wdx < total_words; wdx += 64) row_out[wdx] = 0;
+  __syncthreads();
+
+  for (int item = lane; item < nitems; item += 64) {
+    int off = row_nbits[item];
+    int next_off = item + 1 < nitems
+                       ? row_nbits[item + 1]
+                       : total_bits;
+    int bits = next_off - off;
+    if (bits <= 0) continue;
+    const uint32_t* src = row_stage + (size_t)item * kStageWordsPerItem;
+    int nwords = (bits + 31) / 32;
+    int shift = off & 31;
+    int w0 = off >> 5;
+    for (int k = 0; k < nwords; ++k) {
+      uint32_t w = src[k];
+      if (k == nwords - 1 && (bits & 31))
+        w &= ~((1u << (32 - (bits & 31))) - 1u);  // mask tail garbage
+      if (shift == 0) {
+        atomicOr(&row_out[w0 + k], w);
+      } else {
+        atomicOr(&row_out[w0 + k], w >> shift);
+        atomicOr(&row_out[w0 + k + 1], w << (32 - shift));
+      }
+    }
+  }
+}
+
+void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
+                       int n_jobs, const RowJob* d_jobs, uint32_t* d_stage,
+                       int* d_nbits, uint32_t* d_out, int out_stride_words,
+                       int* d_out_bits, hipStream_t stream) {
+  if (n_jobs == 0) return;
+  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(64), 0, stream,
+                     d_levels, d_meta, mbw, d_jobs, d_stage, d_nbits, d_out,
+                     out_stride_words, d_out_bits);
+}
+
+}  // namespace h264gpu
+}  // namespace hipflux

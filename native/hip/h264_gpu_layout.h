@@ -33,7 +33,23 @@ struct RowJob {
   int flags;       // bit0: I slice (IDR)
   int stripe_y0;   // stripe pixel bounds (for ME clamping)
   int stripe_y1;
+  // slice-header fields for the GPU entropy kernel
+  int first_mb;    // first_mb_in_slice (stripe-relative)
+  int frame_num;
+  int idr_pic_id;
 };
+
+// GPU entropy staging layout: per row, items =
+//   [0] slice header, [1 + mb*28 + slot] per-MB items, [last] trailing
+//   skip run.  Per-MB slots: 0 header, 1 lumaDC, 2..17 lumaAC (Z-order),
+//   18..19 chromaDC, 20..27 chromaAC.
+constexpr int kSlotsPerMb = 28;
+constexpr int kStageWordsPerItem = 16;   // 64 B staging per item
+
+#if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
+__host__ __device__
+#endif
+inline int items_per_row(int mbw) { return 2 + mbw * kSlotsPerMb; }
 
 }  // namespace h264gpu
 }  // namespace hipflux

@@ -21,5 +21,10 @@ void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
                       const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
                       hipStream_t stream);
 
+void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
+                       int n_jobs, const RowJob* d_jobs, uint32_t* d_stage,
+                       int* d_nbits, uint32_t* d_out, int out_stride_words,
+                       int* d_out_bits, hipStream_t stream);
+
 }  // namespace h264gpu
 }  // namespace hipflux

@@ -5,6 +5,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <stdexcept>
@@ -238,6 +239,8 @@ class HipH264Pipeline : public EncodePipeline {
     if (h_meta_) (void)hipHostFree(h_meta_);
     if (h_jobs_) (void)hipHostFree(h_jobs_);
     if (h_stage_) (void)hipHostFree(h_stage_);
+    if (h_entout_) (void)hipHostFree(h_entout_);
+    if (h_outbits_) (void)hipHostFree(h_outbits_);
   }
 
   void encode_frame(const RawFrame& frame, const FrameContext& ctx,
@@ -275,11 +278,14 @@ class HipH264Pipeline : public EncodePipeline {
                           d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
                           stream_);
 
-    // build row jobs for scheduled stripes
+    // build row jobs for scheduled stripes (stripe stream state — IDR,
+    // frame_num, idr_pic_id — resolves here so the GPU entropy kernel gets
+    // final slice-header fields)
     struct SJob {
       int idx;       // stripe index
       int y0, y1;    // pixel bounds (logical)
       bool idr;
+      uint32_t frame_num, idr_pic_id;
     };
     std::vector<SJob> sjobs;
     int n_jobs = 0;
@@ -288,8 +294,15 @@ class HipH264Pipeline : public EncodePipeline {
       if (!st.encode) continue;
       auto& state = stripes_[st.y0 / stripe_h_];
       bool idr = ctx.idr || state.need_idr;
-      sjobs.push_back({static_cast<int>(st.y0 / stripe_h_), st.y0, st.y1,
-                       idr});
+      if (idr) {
+        state.frame_num = 0;
+        ++state.idr_pic_id;
+        state.need_idr = false;
+      }
+      SJob sj{static_cast<int>(st.y0 / stripe_h_), st.y0, st.y1, idr,
+              state.frame_num, state.idr_pic_id};
+      ++state.frame_num;
+      sjobs.push_back(sj);
       int row0 = st.y0 / 16;
       int rows = (std::min(st.y1, mbh_ * 16) - st.y0 + 15) / 16;
       // stripe's padded pixel bounds for ME clamping
@@ -297,9 +310,12 @@ class HipH264Pipeline : public EncodePipeline {
       for (int r = 0; r < rows; ++r) {
         h_jobs_[n_jobs].mb_row = row0 + r;
         h_jobs_[n_jobs].qp = qp;
-        h_jobs_[n_jobs].flags = idr ? 1 : 0;
+        h_jobs_[n_jobs].flags = sj.idr ? 1 : 0;
         h_jobs_[n_jobs].stripe_y0 = sy0;
         h_jobs_[n_jobs].stripe_y1 = sy1;
+        h_jobs_[n_jobs].first_mb = r * mbw_;
+        h_jobs_[n_jobs].frame_num = static_cast<int>(sj.frame_num);
+        h_jobs_[n_jobs].idr_pic_id = static_cast<int>(sj.idr_pic_id);
         ++n_jobs;
       }
     }
@@ -337,15 +353,28 @@ class HipH264Pipeline : public EncodePipeline {
           d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
           d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
           bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_, stream_);
-      HIP_CHECK(hipMemcpyAsync(
-          reinterpret_cast<uint8_t*>(h_levels_) + bt.row0 * lvl_row,
-          reinterpret_cast<uint8_t*>(d_levels_) + bt.row0 * lvl_row,
-          (bt.rown - bt.row0) * lvl_row, hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipMemcpyAsync(
-          reinterpret_cast<uint8_t*>(h_meta_) + bt.row0 * meta_row,
-          reinterpret_cast<uint8_t*>(d_meta_) + bt.row0 * meta_row,
-          (bt.rown - bt.row0) * meta_row, hipMemcpyDeviceToHost, stream_));
+      if (cpu_entropy_) {
+        HIP_CHECK(hipMemcpyAsync(
+            reinterpret_cast<uint8_t*>(h_levels_) + bt.row0 * lvl_row,
+            reinterpret_cast<uint8_t*>(d_levels_) + bt.row0 * lvl_row,
+            (bt.rown - bt.row0) * lvl_row, hipMemcpyDeviceToHost, stream_));
+        HIP_CHECK(hipMemcpyAsync(
+            reinterpret_cast<uint8_t*>(h_meta_) + bt.row0 * meta_row,
+            reinterpret_cast<uint8_t*>(d_meta_) + bt.row0 * meta_row,
+            (bt.rown - bt.row0) * meta_row, hipMemcpyDeviceToHost, stream_));
+      }
       HIP_CHECK(hipEventRecord(batch_events_[b], stream_));
+    }
+    if (!cpu_entropy_) {
+      h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs, d_jobs_,
+                                 d_stage_, d_nbits_, d_entout_,
+                                 ent_stride_words_, d_outbits_, stream_);
+      HIP_CHECK(hipMemcpyAsync(h_entout_, d_entout_,
+                               (size_t)n_jobs * ent_stride_words_ * 4,
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipMemcpyAsync(h_outbits_, d_outbits_,
+                               sizeof(int) * n_jobs, hipMemcpyDeviceToHost,
+                               stream_));
     }
     // refresh ref from cur for encoded stripes (merge contiguous spans so
     // the all-stripes case is 3 copies, not 3 x n_stripes)
@@ -373,8 +402,10 @@ class HipH264Pipeline : public EncodePipeline {
     // NOTE: no stream sync here — entropy overlaps the remaining batches;
     // the final sync below covers the ref-refresh copies.
 
-    // row-parallel CPU entropy: per-stripe output shells + params first,
-    // then submit each batch's rows as its event completes.
+    // entropy: GPU CAVLC by default (k_h264_cavlc_rows packs every row
+    // slice on-device; CPU only adds stop bit + emulation prevention +
+    // NAL headers). HIPFLUX_CPU_ENTROPY=1 selects the CPU packer (kept
+    // for tests and as reference).
     struct Out {
       std::vector<uint8_t> header;            // SPS/PPS on IDR
       std::vector<std::vector<uint8_t>> rows;
@@ -390,15 +421,9 @@ class HipH264Pipeline : public EncodePipeline {
       int j = 0;
       for (size_t i = 0; i < sjobs.size(); ++i) {
         const auto& sj = sjobs[i];
-        auto& state = stripes_[sj.idx];
         outs[i].y0 = sj.y0;
         outs[i].h = std::min(sj.y1, h_) - sj.y0;
         outs[i].idr = sj.idr;
-        if (sj.idr) {
-          state.frame_num = 0;
-          ++state.idr_pic_id;
-          state.need_idr = false;
-        }
         h264::GpuStripeParams& p = outs[i].p;
         p.levels = h_levels_;
         p.meta = h_meta_;
@@ -409,9 +434,8 @@ class HipH264Pipeline : public EncodePipeline {
         p.height = outs[i].h;
         p.qp = qp;
         p.idr = sj.idr;
-        p.frame_num = state.frame_num;
-        p.idr_pic_id = state.idr_pic_id;
-        ++state.frame_num;
+        p.frame_num = sj.frame_num;
+        p.idr_pic_id = sj.idr_pic_id;
         outs[i].rows.resize(p.n_mb_rows);
         if (p.idr) {
           h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
@@ -421,21 +445,29 @@ class HipH264Pipeline : public EncodePipeline {
         for (int r = 0; r < p.n_mb_rows; ++r) job_map[j++] = {int(i), r};
       }
     }
-    for (size_t b = 0; b < batches.size(); ++b) {
-      HIP_CHECK(hipEventSynchronize(batch_events_[b]));
-      for (int j = batches[b].job0; j < batches[b].jobn; ++j) {
+    if (cpu_entropy_) {
+      for (size_t b = 0; b < batches.size(); ++b) {
+        HIP_CHECK(hipEventSynchronize(batch_events_[b]));
+        for (int j = batches[b].job0; j < batches[b].jobn; ++j) {
+          auto [si, r] = job_map[j];
+          auto* dst = &outs[si].rows[r];
+          const h264::GpuStripeParams* pp = &outs[si].p;
+          pool_.submit([pp, r, dst] {
+            h264::encode_row_nal_from_gpu(*pp, r, *dst);
+          });
+        }
+      }
+      pool_.wait_all();
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    } else {
+      HIP_CHECK(hipStreamSynchronize(stream_));
+      for (int j = 0; j < n_jobs; ++j) {
         auto [si, r] = job_map[j];
-        auto* dst = &outs[si].rows[r];
-        const h264::GpuStripeParams* pp = &outs[si].p;
-        pool_.submit([pp, r, dst] {
-          h264::encode_row_nal_from_gpu(*pp, r, *dst);
-        });
+        h264::assemble_gpu_row_nal(
+            h_entout_ + (size_t)j * ent_stride_words_, h_outbits_[j],
+            outs[si].idr, r == 0, outs[si].rows[r]);
       }
     }
-    pool_.wait_all();
-    // ref-refresh copies (and everything else) must land before the next
-    // frame reuses the pinned staging buffers
-    HIP_CHECK(hipStreamSynchronize(stream_));
     for (auto& o : outs) {
       o.bytes = std::move(o.header);
       for (auto& r : o.rows)
@@ -518,6 +550,24 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_jobs_),
                             sizeof(h264gpu::RowJob) * mbh_,
                             hipHostMallocDefault));
+    // GPU entropy buffers
+    cpu_entropy_ = std::getenv("HIPFLUX_CPU_ENTROPY") != nullptr;
+    const int nitems = h264gpu::items_per_row(mbw_);
+    ent_stride_words_ = nitems * h264gpu::kStageWordsPerItem;
+    size_t stage_bytes =
+        (size_t)mbh_ * nitems * h264gpu::kStageWordsPerItem * 4;
+    d_stage_ = reinterpret_cast<uint32_t*>(dalloc(stage_bytes));
+    d_nbits_ = reinterpret_cast<int*>(dalloc((size_t)mbh_ * nitems * 4));
+    d_entout_ = reinterpret_cast<uint32_t*>(
+        dalloc((size_t)mbh_ * ent_stride_words_ * 4));
+    d_outbits_ = reinterpret_cast<int*>(dalloc(sizeof(int) * mbh_));
+    if (h_entout_) (void)hipHostFree(h_entout_);
+    if (h_outbits_) (void)hipHostFree(h_outbits_);
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_entout_),
+                            (size_t)mbh_ * ent_stride_words_ * 4,
+                            hipHostMallocDefault));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_outbits_),
+                            sizeof(int) * mbh_, hipHostMallocDefault));
     stripes_.assign((h + stripe_h_ - 1) / stripe_h_, StripeState{});
   }
 
@@ -538,6 +588,15 @@ class HipH264Pipeline : public EncodePipeline {
   h264gpu::RowJob* h_jobs_ = nullptr;
   uint8_t* h_stage_ = nullptr;
   size_t h_stage_bytes_ = 0;
+  // GPU entropy buffers
+  bool cpu_entropy_ = false;
+  uint32_t* d_stage_ = nullptr;
+  int* d_nbits_ = nullptr;
+  uint32_t* d_entout_ = nullptr;
+  int* d_outbits_ = nullptr;
+  uint32_t* h_entout_ = nullptr;
+  int* h_outbits_ = nullptr;
+  int ent_stride_words_ = 0;
   std::vector<hipEvent_t> batch_events_;
   std::vector<void*> device_ptrs_;
   std::vector<StripeState> stripes_;

@@ -185,6 +185,27 @@ def test_gpu_mfma_motion_search_exact():
     assert sizes[1] < sizes[0] * 0.5, f"P frame too large: {sizes}"
 
 
+def test_gpu_cavlc_matches_cpu_entropy():
+    """The GPU CAVLC kernel must emit byte-identical streams to the CPU
+    packer for the same levels/meta (IDR + P frames, mixed modes)."""
+    require_gpu()
+    import os
+    w, h, n = 320, 192, 5
+    frames = make_frames(w, h, n)
+    os.environ["HIPFLUX_CPU_ENTROPY"] = "1"
+    cpu_out = _native._pipeline_encode("gpu", frames, w, h, 26, 64, 1)
+    del os.environ["HIPFLUX_CPU_ENTROPY"]
+    gpu_out = _native._pipeline_encode("gpu", frames, w, h, 26, 64, 1)
+    for fi, (fa, fb) in enumerate(zip(cpu_out, gpu_out)):
+        sa = sorted(fa, key=lambda t: t[1])
+        sb = sorted(fb, key=lambda t: t[1])
+        assert len(sa) == len(sb)
+        for (da, ya, _, _), (db, yb, _, _) in zip(sa, sb):
+            assert ya == yb
+            assert bytes(da) == bytes(db), \
+                f"frame {fi} stripe {ya}: GPU CAVLC differs from CPU"
+
+
 def test_gpu_h264_1080p_throughput():
     require_gpu()
     import time
