@@ -152,71 +152,79 @@ __device__ inline int blk_total(const int16_t* L, int off, int first,
   return t;
 }
 
+// Per-MB info, precomputed once per row into LDS:
+//   flags: mode(2) | luma_mode<<2 | chroma_mode<<5 | cbp_luma1<<8 |
+//          cbp_chroma<<9
+//   totals: 16 luma-AC + 8 chroma-AC per-block nonzero counts
 struct MbInfo {
-  int mode;          // kSkip/kInter/kIntra (I rows: kIntra)
-  int luma_mode, chroma_mode;
-  int mvx, mvy;
-  int cbp_luma, cbp_chroma;
+  short flags;
+  short mvx, mvy;
+  uint8_t ltot[16];
+  uint8_t ctot[8];
 };
 
-__device__ MbInfo load_mb(const int16_t* levels, const int* meta,
-                          size_t mb_index, bool i_slice) {
-  MbInfo m{};
+#define MB_MODE(f) ((f) & 3)
+#define MB_LMODE(f) (((f) >> 2) & 7)
+#define MB_CMODE(f) (((f) >> 5) & 7)
+#define MB_CBPL(f) (((f) >> 8) & 1)
+#define MB_CBPC(f) (((f) >> 9) & 3)
+
+__device__ void precompute_mb(const int16_t* levels, const int* meta,
+                              size_t mb_index, bool i_slice, MbInfo* out) {
   int m0 = meta[mb_index * kMetaPerMb + 0];
-  m.mode = i_slice ? kIntra : (m0 & 3);
-  m.luma_mode = (m0 >> 2) & 7;
-  m.chroma_mode = (m0 >> 5) & 7;
+  int mode = i_slice ? kIntra : (m0 & 3);
   int m1 = meta[mb_index * kMetaPerMb + 1];
-  m.mvx = (short)(m1 & 0xFFFF);
-  m.mvy = m1 >> 16;
-  if (m.mode == kIntra) {
+  out->mvx = (short)(m1 & 0xFFFF);
+  out->mvy = (short)(m1 >> 16);
+  int cbp_luma1 = 0, cbp_chroma = 0;
+  if (mode == kIntra) {
     const int16_t* L = levels + mb_index * kLevelsPerMb;
     int any_ac = 0;
-    for (int b = 0; b < 16 && !any_ac; ++b)
-      any_ac |= blk_total(L, kLumaAcOff + b * 16, 1, 15) != 0;
-    m.cbp_luma = any_ac ? 15 : 0;
+    for (int b = 0; b < 16; ++b) {
+      int t = blk_total(L, kLumaAcOff + b * 16, 1, 15);
+      out->ltot[b] = (uint8_t)t;
+      any_ac |= t;
+    }
     int any_cac = 0, any_cdc = 0;
-    for (int b = 0; b < 8 && !any_cac; ++b)
-      any_cac |= blk_total(L, kChromaAcOff + b * 16, 1, 15) != 0;
-    for (int i = 0; i < 8 && !any_cdc; ++i)
-      any_cdc |= L[kChromaDcOff + i] != 0;
-    m.cbp_chroma = any_cac ? 2 : (any_cdc ? 1 : 0);
+    for (int b = 0; b < 8; ++b) {
+      int t = blk_total(L, kChromaAcOff + b * 16, 1, 15);
+      out->ctot[b] = (uint8_t)t;
+      any_cac |= t;
+    }
+    for (int i = 0; i < 8; ++i) any_cdc |= L[kChromaDcOff + i] != 0;
+    cbp_luma1 = any_ac ? 1 : 0;
+    cbp_chroma = any_cac ? 2 : (any_cdc ? 1 : 0);
+  } else {
+    for (int b = 0; b < 16; ++b) out->ltot[b] = 0;
+    for (int b = 0; b < 8; ++b) out->ctot[b] = 0;
   }
-  return m;
+  out->flags = (short)(mode | (((m0 >> 2) & 7) << 2) |
+                       (((m0 >> 5) & 7) << 5) | (cbp_luma1 << 8) |
+                       (cbp_chroma << 9));
 }
 
-// nC context for luma block (bx,by) of MB mbx (top row never available)
-__device__ int luma_nc(const int16_t* levels, const int* meta,
-                       size_t row_base, int mbx, int bx, int by,
-                       bool i_slice) {
+__device__ inline int lds_luma_nc(const MbInfo* info, int mbx, int bx,
+                                  int by) {
   if (bx > 0) {
-    const int16_t* L = levels + (row_base + mbx) * kLevelsPerMb;
-    MbInfo self = load_mb(levels, meta, row_base + mbx, i_slice);
-    if (self.cbp_luma == 0) return 0;
-    int r = by * 4 + (bx - 1);
-    return blk_total(L, kLumaAcOff + r * 16, 1, 15);
+    if (!MB_CBPL(info[mbx].flags)) return 0;
+    return info[mbx].ltot[by * 4 + (bx - 1)];
   }
   if (mbx == 0) return 0;
-  MbInfo left = load_mb(levels, meta, row_base + mbx - 1, i_slice);
-  if (left.mode != kIntra || left.cbp_luma == 0) return 0;
-  const int16_t* L = levels + (row_base + mbx - 1) * kLevelsPerMb;
-  return blk_total(L, kLumaAcOff + (by * 4 + 3) * 16, 1, 15);
+  const MbInfo& l = info[mbx - 1];
+  if (MB_MODE(l.flags) != kIntra || !MB_CBPL(l.flags)) return 0;
+  return l.ltot[by * 4 + 3];
 }
 
-__device__ int chroma_nc(const int16_t* levels, const int* meta,
-                         size_t row_base, int mbx, int comp, int cx, int cy,
-                         bool i_slice) {
+__device__ inline int lds_chroma_nc(const MbInfo* info, int mbx, int comp,
+                                    int cx, int cy) {
   if (cx > 0) {
-    MbInfo self = load_mb(levels, meta, row_base + mbx, i_slice);
-    if (self.cbp_chroma != 2) return 0;
-    const int16_t* L = levels + (row_base + mbx) * kLevelsPerMb;
-    return blk_total(L, kChromaAcOff + (comp * 4 + cy * 2) * 16, 1, 15);
+    if (MB_CBPC(info[mbx].flags) != 2) return 0;
+    return info[mbx].ctot[comp * 4 + cy * 2];
   }
   if (mbx == 0) return 0;
-  MbInfo left = load_mb(levels, meta, row_base + mbx - 1, i_slice);
-  if (left.mode != kIntra || left.cbp_chroma != 2) return 0;
-  const int16_t* L = levels + (row_base + mbx - 1) * kLevelsPerMb;
-  return blk_total(L, kChromaAcOff + (comp * 4 + cy * 2 + 1) * 16, 1, 15);
+  const MbInfo& l = info[mbx - 1];
+  if (MB_MODE(l.flags) != kIntra || MB_CBPC(l.flags) != 2) return 0;
+  return l.ctot[comp * 4 + cy * 2 + 1];
 }
 
 // ---- the kernel -------------------------------------------------------------
@@ -238,14 +246,17 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
   int* row_nbits = nbits + (size_t)blockIdx.x * nitems;
   uint32_t* row_out = out + (size_t)blockIdx.x * out_stride_words;
 
-  // ---- skip runs (P rows): s_skiprun[mb] = run before this MB; trailing
-  __shared__ short s_skiprun[1024];
+  // ---- precompute per-MB info into LDS (totals, cbp, modes, mvs)
+  __shared__ MbInfo s_mb[512];
+  __shared__ short s_skiprun[512];
   __shared__ int s_trailing;
+  for (int mb = lane; mb < mbw; mb += 64)
+    precompute_mb(levels, meta, row_base + mb, i_slice, &s_mb[mb]);
+  __syncthreads();
   if (lane == 0) {
     int run = 0;
     for (int mb = 0; mb < mbw; ++mb) {
-      MbInfo m = load_mb(levels, meta, row_base + mb, i_slice);
-      if (!i_slice && m.mode == kSkip) {
+      if (!i_slice && MB_MODE(s_mb[mb].flags) == kSkip) {
         s_skiprun[mb] = -1;  // skipped MB: no items
         ++run;
       } else {
@@ -288,57 +299,56 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
     } else {
       int mb = (item - 1) / kSlotsPerMb;
       int slot = (item - 1) % kSlotsPerMb;
-      MbInfo m = load_mb(levels, meta, row_base + mb, i_slice);
+      const MbInfo& m = s_mb[mb];
+      const int mode = MB_MODE(m.flags);
+      const int cbp_chroma = MB_CBPC(m.flags);
       const int16_t* L = levels + (row_base + mb) * kLevelsPerMb;
       if (s_skiprun[mb] < 0) {
         bits = 0;  // skipped MB
       } else if (slot == 0) {
         if (!i_slice) bw.ue(s_skiprun[mb]);
-        if (m.mode == kInter) {
+        if (mode == kInter) {
           bw.ue(0);  // P_L0_16x16
           int mvpx = 0, mvpy = 0;
-          if (mb > 0) {
-            MbInfo left = load_mb(levels, meta, row_base + mb - 1, i_slice);
-            if (left.mode != kIntra) {  // skip or inter both carry MVs
-              mvpx = left.mvx;
-              mvpy = left.mvy;
-            }
+          if (mb > 0 && MB_MODE(s_mb[mb - 1].flags) != kIntra) {
+            mvpx = s_mb[mb - 1].mvx;   // skip or inter both carry MVs
+            mvpy = s_mb[mb - 1].mvy;
           }
           bw.se(m.mvx - mvpx);
           bw.se(m.mvy - mvpy);
           bw.ue(0);  // cbp = 0
         } else {
-          int i16 = 1 + m.luma_mode + 4 * m.cbp_chroma +
-                    12 * (m.cbp_luma ? 1 : 0);
+          int i16 = 1 + MB_LMODE(m.flags) + 4 * cbp_chroma +
+                    12 * MB_CBPL(m.flags);
           bw.ue(i_slice ? i16 : 5 + i16);
-          bw.ue(m.chroma_mode);
+          bw.ue(MB_CMODE(m.flags));
           bw.se(0);  // mb_qp_delta
         }
         bits = bw.flush();
-      } else if (m.mode != kIntra) {
+      } else if (mode != kIntra) {
         bits = 0;
       } else if (slot == 1) {
         // luma DC
         int zz[16];
         for (int i = 0; i < 16; ++i)
           zz[i] = L[kLumaDcOff + c_zig4[i]];
-        int nC = luma_nc(levels, meta, row_base, mb, 0, 0, i_slice);
+        int nC = lds_luma_nc(s_mb, mb, 0, 0);
         dev_cavlc_residual(bw, zz, 16, nC);
         bits = bw.flush();
       } else if (slot < 18) {
-        if (m.cbp_luma) {
+        if (MB_CBPL(m.flags)) {
           int blk = slot - 2;                      // Z-order index
           int r = c_zorder_raster[blk];
           int bx = r & 3, by = r >> 2;
           int zz[15];
           for (int i = 1; i < 16; ++i)
             zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
-          int nC = luma_nc(levels, meta, row_base, mb, bx, by, i_slice);
+          int nC = lds_luma_nc(s_mb, mb, bx, by);
           dev_cavlc_residual(bw, zz, 15, nC);
           bits = bw.flush();
         }
       } else if (slot < 20) {
-        if (m.cbp_chroma > 0) {
+        if (cbp_chroma > 0) {
           int comp = slot - 18;
           int zz[4];
           for (int i = 0; i < 4; ++i) zz[i] = L[kChromaDcOff + comp * 4 + i];
@@ -346,14 +356,13 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
           bits = bw.flush();
         }
       } else {
-        if (m.cbp_chroma == 2) {
+        if (cbp_chroma == 2) {
           int b = slot - 20;                       // cb0..3 then cr0..3
           int comp = b >> 2, sub = b & 3;
           int zz[15];
           for (int i = 1; i < 16; ++i)
             zz[i - 1] = L[kChromaAcOff + (comp * 4 + sub) * 16 + c_zig4[i]];
-          int nC = chroma_nc(levels, meta, row_base, mb, comp, sub & 1,
-                             sub >> 1, i_slice);
+          int nC = lds_chroma_nc(s_mb, mb, comp, sub & 1, sub >> 1);
           dev_cavlc_residual(bw, zz, 15, nC);
           bits = bw.flush();
         }

@@ -66,7 +66,6 @@ def _parse_bool(raw: str) -> bool:
     return str(raw).strip().lower() in ("1", "true", "yes", "on")
 
 
-ENCODERS = ("x264enc", "x264enc-striped", "jpeg")
 # Wire names follow the reference encoder menu (reference selkies.py:144
 # PIXELFLUX_VIDEO_ENCODERS: h264enc, h264enc-striped, jpeg, openh264enc).
 # Ours: h264enc (full-frame H.264), h264enc-striped (stripe-parallel
