@@ -187,6 +187,7 @@ function canvasPos(ev) {
 }
 
 function send(msg) { if (ws && ws.readyState === 1) ws.send(msg); }
+window.skSend = send;   /* dashboard hook */
 
 function hookInput() {
   canvas.addEventListener("keydown", (ev) => {
@@ -298,11 +299,23 @@ function connect() {
       const verb = i < 0 ? ev.data : ev.data.slice(0, i);
       const rest = i < 0 ? "" : ev.data.slice(i + 1);
       if (verb === "SETTINGS_PAYLOAD") {
-        try { applyServerSettings(JSON.parse(rest)); } catch (e) {}
+        try {
+          const payload = JSON.parse(rest);
+          applyServerSettings(payload);
+          if (window.skOnSettings) window.skOnSettings(payload);
+        } catch (e) {}
+      } else if (verb === "SYSTEM_STATS") {
+        try {
+          if (window.skOnStats) window.skOnStats(JSON.parse(rest));
+        } catch (e) {}
       } else if (verb === "CURSOR") {
         try { applyCursor(JSON.parse(rest)); } catch (e) {}
       } else if (verb === "clipboard") {
-        try { navigator.clipboard.writeText(atob(rest)); } catch (e) {}
+        try {
+          const text = decodeURIComponent(escape(atob(rest)));
+          navigator.clipboard.writeText(text).catch(() => {});
+          if (window.skOnClipboard) window.skOnClipboard(text);
+        } catch (e) {}
       }
     } else {
       onBinary(ev.data);
