@@ -62,21 +62,29 @@ struct DevBW {
   }
 };
 
-// ---- residual block encode (mirrors cpu/h264/cavlc.h exactly) -------------
-__device__ void dev_cavlc_residual(DevBW& bw, const int* zz, int n, int nC) {
-  int coeffs[16], pos[16], tc = 0;
-  for (int i = 0; i < n; ++i)
-    if (zz[i]) {
-      coeffs[tc] = zz[i];
-      pos[tc] = i;
+// ---- residual block encode (mirrors cpu/h264/cavlc.h bit-exactly) ---------
+// Streaming over the zigzag values with STATIC trip counts (template N):
+// no runtime-indexed arrays -> everything stays in registers (a
+// coeffs[]/pos[] formulation spills to scratch and was 2.5x slower).
+template <int N>
+__device__ __forceinline__ void dev_cavlc_residual(DevBW& bw,
+                                                   const int* zz, int nC) {
+  // pass 1: tc, trailing ones, last nonzero position
+  int tc = 0, t1 = 0, last_pos = -1;
+  bool blocked = false;
+#pragma unroll
+  for (int i = N - 1; i >= 0; --i) {
+    int v = zz[i];
+    if (v) {
       ++tc;
+      if (last_pos < 0) last_pos = i;
+      if (!blocked) {
+        if ((v == 1 || v == -1) && t1 < 3)
+          ++t1;
+        else
+          blocked = true;
+      }
     }
-  int t1 = 0;
-  for (int k = tc - 1; k >= 0 && t1 < 3; --k) {
-    if (coeffs[k] == 1 || coeffs[k] == -1)
-      ++t1;
-    else
-      break;
   }
   unsigned packed;
   if (nC == -1)
@@ -92,54 +100,84 @@ __device__ void dev_cavlc_residual(DevBW& bw, const int* zz, int n, int nC) {
   bw.u(packed & 0xFFFF, packed >> 16);
   if (tc == 0) return;
 
-  for (int k = tc - 1; k >= tc - t1; --k) bw.u(coeffs[k] > 0 ? 0 : 1, 1);
-
-  int suffix_len = (tc > 10 && t1 < 3) ? 1 : 0;
-  bool first = true;
-  for (int k = tc - t1 - 1; k >= 0; --k) {
-    const int true_level = coeffs[k];
-    int level = coeffs[k];
-    if (first && t1 < 3) level += level > 0 ? -1 : 1;
-    first = false;
-    int code = level > 0 ? 2 * level - 2 : -2 * level - 1;
-    if (suffix_len == 0) {
-      if (code < 14) {
-        bw.u(1, code + 1);
-      } else if (code < 30) {
-        bw.u(1, 15);
-        bw.u(code - 14, 4);
-      } else {
-        bw.u(1, 16);
-        bw.u(code - 30, 12);
-      }
-    } else {
-      int prefix = code >> suffix_len;
-      if (prefix < 15) {
-        bw.u(1, prefix + 1);
-        bw.u(code & ((1 << suffix_len) - 1), suffix_len);
-      } else {
-        bw.u(1, 16);
-        bw.u(code - (15 << suffix_len), 12);
+  // pass 2: trailing-one signs (highest frequency first)
+  {
+    int seen = 0;
+#pragma unroll
+    for (int i = N - 1; i >= 0; --i) {
+      int v = zz[i];
+      if (v) {
+        if (seen < t1) bw.u(v > 0 ? 0 : 1, 1);
+        ++seen;
       }
     }
-    if (suffix_len == 0) suffix_len = 1;
-    if (abs(true_level) > (3 << (suffix_len - 1)) && suffix_len < 6)
-      ++suffix_len;
   }
 
-  int total_zeros = pos[tc - 1] + 1 - tc;
-  if (tc < n) {
+  // pass 3: remaining levels (highest frequency first)
+  {
+    int suffix_len = (tc > 10 && t1 < 3) ? 1 : 0;
+    int seen = 0;
+#pragma unroll
+    for (int i = N - 1; i >= 0; --i) {
+      int v = zz[i];
+      if (!v) continue;
+      ++seen;
+      if (seen <= t1) continue;
+      const int true_level = v;
+      int level = v;
+      if (seen == t1 + 1 && t1 < 3) level += level > 0 ? -1 : 1;
+      int code = level > 0 ? 2 * level - 2 : -2 * level - 1;
+      if (suffix_len == 0) {
+        if (code < 14) {
+          bw.u(1, code + 1);
+        } else if (code < 30) {
+          bw.u(1, 15);
+          bw.u(code - 14, 4);
+        } else {
+          bw.u(1, 16);
+          bw.u(code - 30, 12);
+        }
+      } else {
+        int prefix = code >> suffix_len;
+        if (prefix < 15) {
+          bw.u(1, prefix + 1);
+          bw.u(code & ((1 << suffix_len) - 1), suffix_len);
+        } else {
+          bw.u(1, 16);
+          bw.u(code - (15 << suffix_len), 12);
+        }
+      }
+      if (suffix_len == 0) suffix_len = 1;
+      if (abs(true_level) > (3 << (suffix_len - 1)) && suffix_len < 6)
+        ++suffix_len;
+    }
+  }
+
+  // pass 4: total_zeros
+  int total_zeros = last_pos + 1 - tc;
+  if (tc < N) {
     unsigned p = (nC == -1) ? c_tzcdc[(tc - 1) * 4 + total_zeros]
                             : c_tz[(tc - 1) * 16 + total_zeros];
     bw.u(p & 0xFFFF, p >> 16);
   }
-  int zeros_left = total_zeros;
-  for (int k = tc - 1; k > 0 && zeros_left > 0; --k) {
-    int run = pos[k] - pos[k - 1] - 1;
-    int zl = zeros_left < 7 ? zeros_left : 7;
-    unsigned p = c_rb[(zl - 1) * 15 + run];
-    bw.u(p & 0xFFFF, p >> 16);
-    zeros_left -= run;
+
+  // pass 5: run_before (gap below each nonzero, highest first; the lowest
+  // nonzero carries no run; stop once all zeros are accounted for)
+  {
+    int zeros_left = total_zeros;
+    int prev_pos = -1;
+#pragma unroll
+    for (int i = N - 1; i >= 0; --i) {
+      if (!zz[i]) continue;
+      if (prev_pos >= 0 && zeros_left > 0) {
+        int run = prev_pos - i - 1;
+        int zl = zeros_left < 7 ? zeros_left : 7;
+        unsigned p = c_rb[(zl - 1) * 15 + run];
+        bw.u(p & 0xFFFF, p >> 16);
+        zeros_left -= run;
+      }
+      prev_pos = i;
+    }
   }
 }
 
@@ -228,7 +266,7 @@ __device__ inline int lds_chroma_nc(const MbInfo* info, int mbx, int comp,
 }
 
 // ---- the kernel -------------------------------------------------------------
-__global__ void __launch_bounds__(64) k_h264_cavlc_rows(
+__global__ void __launch_bounds__(256) k_h264_cavlc_rows(
     const int16_t* __restrict__ levels, const int* __restrict__ meta,
     int mbw, const RowJob* __restrict__ jobs,
     uint32_t* __restrict__ stage,       // [row][item][kStageWordsPerItem]
@@ -237,7 +275,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
     int out_stride_words,
     int* __restrict__ out_bits) {       // [row]
   const RowJob job = jobs[blockIdx.x];
-  const int lane = threadIdx.x;
+  const int tid = threadIdx.x;            // 256 threads (4 waves)
   const bool i_slice = (job.flags & 1) != 0;
   const size_t row_base = (size_t)job.mb_row * mbw;
   const int nitems = items_per_row(mbw);
@@ -250,10 +288,10 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
   __shared__ MbInfo s_mb[512];
   __shared__ short s_skiprun[512];
   __shared__ int s_trailing;
-  for (int mb = lane; mb < mbw; mb += 64)
+  for (int mb = tid; mb < mbw; mb += 256)
     precompute_mb(levels, meta, row_base + mb, i_slice, &s_mb[mb]);
   __syncthreads();
-  if (lane == 0) {
+  if (tid == 0) {
     int run = 0;
     for (int mb = 0; mb < mbw; ++mb) {
       if (!i_slice && MB_MODE(s_mb[mb].flags) == kSkip) {
@@ -269,7 +307,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
   __syncthreads();
 
   // ---- phase A: encode items
-  for (int item = lane; item < nitems; item += 64) {
+  for (int item = tid; item < nitems; item += 256) {
     DevBW bw;
     bw.out = row_stage + (size_t)item * kStageWordsPerItem;
     int bits = 0;
@@ -333,7 +371,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
         for (int i = 0; i < 16; ++i)
           zz[i] = L[kLumaDcOff + c_zig4[i]];
         int nC = lds_luma_nc(s_mb, mb, 0, 0);
-        dev_cavlc_residual(bw, zz, 16, nC);
+        dev_cavlc_residual<16>(bw, zz, nC);
         bits = bw.flush();
       } else if (slot < 18) {
         if (MB_CBPL(m.flags)) {
@@ -344,7 +382,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
           for (int i = 1; i < 16; ++i)
             zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
           int nC = lds_luma_nc(s_mb, mb, bx, by);
-          dev_cavlc_residual(bw, zz, 15, nC);
+          dev_cavlc_residual<15>(bw, zz, nC);
           bits = bw.flush();
         }
       } else if (slot < 20) {
@@ -352,7 +390,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
           int comp = slot - 18;
           int zz[4];
           for (int i = 0; i < 4; ++i) zz[i] = L[kChromaDcOff + comp * 4 + i];
-          dev_cavlc_residual(bw, zz, 4, -1);
+          dev_cavlc_residual<4>(bw, zz, -1);
           bits = bw.flush();
         }
       } else {
@@ -363,7 +401,7 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
           for (int i = 1; i < 16; ++i)
             zz[i - 1] = L[kChromaAcOff + (comp * 4 + sub) * 16 + c_zig4[i]];
           int nC = lds_chroma_nc(s_mb, mb, comp, sub & 1, sub >> 1);
-          dev_cavlc_residual(bw, zz, 15, nC);
+          dev_cavlc_residual<15>(bw, zz, nC);
           bits = bw.flush();
         }
       }
@@ -372,34 +410,44 @@ __global__ void __launch_bounds__(64) k_h264_cavlc_rows(
   }
   __syncthreads();
 
-  // ---- phase B: exclusive prefix sum of nbits (wave-strided)
-  __shared__ int s_carry;
-  if (lane == 0) s_carry = 0;
-  __syncthreads();
-  for (int base = 0; base < nitems; base += 64) {
-    int idx = base + lane;
-    int v = idx < nitems ? row_nbits[idx] : 0;
-    // inclusive scan within the wave
-    int inc = v;
+  // ---- phase B: exclusive prefix sum via per-thread segments
+  const int seg_lo = (int)((long)nitems * tid / 256);
+  const int seg_hi = (int)((long)nitems * (tid + 1) / 256);
+  int seg_sum = 0;
+  for (int i = seg_lo; i < seg_hi; ++i) seg_sum += row_nbits[i];
+  __shared__ int s_total;
+  {
+    const int lane = tid & 63, wid = tid >> 6;
+    int inc = seg_sum;
     for (int d = 1; d < 64; d <<= 1) {
       int other = __shfl_up(inc, d);
       if (lane >= d) inc += other;
     }
-    int excl = inc - v + s_carry;
-    if (idx < nitems) row_nbits[idx] = excl;   // becomes the offset
+    __shared__ int s_wsum[4];
+    if (lane == 63) s_wsum[wid] = inc;
     __syncthreads();
-    if (lane == 63) s_carry += inc;
+    int wbase = 0;
+    for (int w = 0; w < wid; ++w) wbase += s_wsum[w];
+    int seg_off = wbase + inc - seg_sum;
+    if (tid == 255) s_total = wbase + inc;
+    // rewrite nbits -> offsets for this thread's segment
+    int off = seg_off;
+    for (int i = seg_lo; i < seg_hi; ++i) {
+      int t = row_nbits[i];
+      row_nbits[i] = off;
+      off += t;
+    }
     __syncthreads();
   }
-  int total_bits = s_carry;
-  if (lane == 0) out_bits[blockIdx.x] = total_bits;
+  int total_bits = s_total;
+  if (tid == 0) out_bits[blockIdx.x] = total_bits;
 
   // ---- zero the needed output words, then scatter
   int total_words = (total_bits + 31) / 32 + 1;
-  for (int wdx = lane; wdx < total_words; wdx += 64) row_out[wdx] = 0;
+  for (int wdx = tid; wdx < total_words; wdx += 256) row_out[wdx] = 0;
   __syncthreads();
 
-  for (int item = lane; item < nitems; item += 64) {
+  for (int item = tid; item < nitems; item += 256) {
     int off = row_nbits[item];
     int next_off = item + 1 < nitems
                        ? row_nbits[item + 1]
@@ -429,7 +477,7 @@ void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
                        int* d_nbits, uint32_t* d_out, int out_stride_words,
                        int* d_out_bits, hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(64), 0, stream,
+  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(256), 0, stream,
                      d_levels, d_meta, mbw, d_jobs, d_stage, d_nbits, d_out,
                      out_stride_words, d_out_bits);
 }
