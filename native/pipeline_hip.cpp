@@ -463,10 +463,16 @@ class HipH264Pipeline : public EncodePipeline {
       HIP_CHECK(hipStreamSynchronize(stream_));
       for (int j = 0; j < n_jobs; ++j) {
         auto [si, r] = job_map[j];
-        h264::assemble_gpu_row_nal(
-            h_entout_ + (size_t)j * ent_stride_words_, h_outbits_[j],
-            outs[si].idr, r == 0, outs[si].rows[r]);
+        const uint32_t* words = h_entout_ + (size_t)j * ent_stride_words_;
+        int bits = h_outbits_[j];
+        bool idr = outs[si].idr;
+        auto* dst = &outs[si].rows[r];
+        bool long_sc = r == 0;
+        pool_.submit([words, bits, idr, long_sc, dst] {
+          h264::assemble_gpu_row_nal(words, bits, idr, long_sc, *dst);
+        });
       }
+      pool_.wait_all();
     }
     for (auto& o : outs) {
       o.bytes = std::move(o.header);
