@@ -34,10 +34,11 @@ struct RowCtx {
 
 // Encode one intra MB from the GPU level buffer. Mirrors
 // encoder.cpp::encode_i16's entropy section exactly.
-void entropy_i16(BitWriter& bw, const int16_t* L, int m0, int qp,
+void entropy_i16(BitWriter& bw, const int16_t* L, int m0, int m1, int qp,
                  bool p_slice, RowCtx& ctx) {
   const int luma_mode = (m0 >> 2) & 7;
-  const int chroma_mode = (m0 >> 5) & 7;
+  // chroma pipeline reports its mode via meta word 1 (unused for intra)
+  const int chroma_mode = m1 & 7;
 
   // gather zigzag blocks + cbp from levels
   int zz_dc[16];
@@ -143,6 +144,7 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
       const int16_t* L = p.levels + (mb_base + mbx) * kLevelsPerMb;
       const int* M = p.meta + (mb_base + mbx) * kMetaPerMb;
       int m0 = M[0];
+      int m1 = M[1];
       int mode = p.idr ? h264gpu::kIntra : (m0 & 3);
       if (!p.idr && mode == h264gpu::kSkip) {
         ++ctx.skip_run;
@@ -160,7 +162,6 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
         ctx.skip_run = 0;
       }
       if (!p.idr && mode == h264gpu::kInter) {
-        int m1 = M[1];
         int mvx = (int16_t)(m1 & 0xFFFF);
         int mvy = m1 >> 16;
         b.ue(0);  // P_L0_16x16
@@ -178,7 +179,7 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
         ctx.left_cr_nc[0] = ctx.left_cr_nc[1] = 0;
         continue;
       }
-      entropy_i16(b, L, m0, p.qp, !p.idr, ctx);
+      entropy_i16(b, L, m0, m1, p.qp, !p.idr, ctx);
     }
     if (!p.idr && ctx.skip_run > 0) b.ue(ctx.skip_run);
     b.rbsp_trailing();

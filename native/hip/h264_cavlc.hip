@@ -236,9 +236,10 @@ __device__ void precompute_mb(const int16_t* levels, const int* meta,
     for (int b = 0; b < 16; ++b) out->ltot[b] = 0;
     for (int b = 0; b < 8; ++b) out->ctot[b] = 0;
   }
-  out->flags = (short)(mode | (((m0 >> 2) & 7) << 2) |
-                       (((m0 >> 5) & 7) << 5) | (cbp_luma1 << 8) |
-                       (cbp_chroma << 9));
+  // chroma mode rides in meta word 1 for intra MBs (see k_h264_rows)
+  int cmode = (mode == kIntra) ? (m1 & 7) : 0;
+  out->flags = (short)(mode | (((m0 >> 2) & 7) << 2) | (cmode << 5) |
+                       (cbp_luma1 << 8) | (cbp_chroma << 9));
 }
 
 __device__ inline int lds_luma_nc(const MbInfo* info, int mbx, int bx,
@@ -454,21 +455,29 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
                        : total_bits;
     int bits = next_off - off;
     if (bits <= 0) continue;
+    // Build each destination word fully (carry accumulator across source
+    // words), then: plain store for interior words, atomicOr only at the
+    // first/last destination word (the only ones a neighbor item shares).
     const uint32_t* src = row_stage + (size_t)item * kStageWordsPerItem;
     int nwords = (bits + 31) / 32;
     int shift = off & 31;
     int w0 = off >> 5;
+    int last_dst = (off + bits - 1) >> 5;
+    uint32_t carry = 0;
     for (int k = 0; k < nwords; ++k) {
       uint32_t w = src[k];
       if (k == nwords - 1 && (bits & 31))
         w &= ~((1u << (32 - (bits & 31))) - 1u);  // mask tail garbage
-      if (shift == 0) {
-        atomicOr(&row_out[w0 + k], w);
-      } else {
-        atomicOr(&row_out[w0 + k], w >> shift);
-        atomicOr(&row_out[w0 + k + 1], w << (32 - shift));
-      }
+      uint32_t val = shift ? (carry | (w >> shift)) : w;
+      int d = w0 + k;
+      if (d == w0 || d == last_dst)
+        atomicOr(&row_out[d], val);
+      else
+        row_out[d] = val;
+      carry = shift ? (w << (32 - shift)) : 0;
     }
+    if (shift && (w0 + nwords) <= last_dst)
+      atomicOr(&row_out[w0 + nwords], carry);
   }
 }
 

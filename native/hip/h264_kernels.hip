@@ -203,49 +203,45 @@ __device__ inline int wave_min_i(int v) {
 }
 
 // ---------------------------------------------------------------------------
-// The row kernel. One wave per RowJob; walks MBs of the row.
-__global__ void __launch_bounds__(64) k_h264_rows(
-    const uint8_t* __restrict__ srcY, const uint8_t* __restrict__ srcCb,
-    const uint8_t* __restrict__ srcCr, int ypitch, int cpitch, int w, int h,
-    const uint8_t* __restrict__ refY, const uint8_t* __restrict__ refCb,
-    const uint8_t* __restrict__ refCr, uint8_t* __restrict__ curY,
-    uint8_t* __restrict__ curCb, uint8_t* __restrict__ curCr, int mbw,
-    const RowJob* __restrict__ jobs, int16_t* __restrict__ levels,
-    int* __restrict__ meta) {
-  const RowJob job = jobs[blockIdx.x];
-  const int lane = threadIdx.x;
-  const int qp = job.qp;
-  const int qpc = dev_chroma_qp(qp);
-  const bool i_slice = (job.flags & 1) != 0;
-  const int mby = job.mb_row;
+// The row kernel: one workgroup of 128 threads per MB-row slice.
+// Wave 0 encodes the row's LUMA, wave 1 its CHROMA — the two pipelines
+// share no state (chroma_mode is reported through meta word 1, which is
+// unused for intra MBs), so the waves run fully decoupled and the per-MB
+// serial chain is the LUMA path only.
+//
+// Cross-lane data movement uses shuffles (source pixels, DC terms) instead
+// of LDS staging; the only LDS state is the left-neighbor recon columns,
+// updated once per MB behind a wave-level fence.
 
-  __shared__ uint8_t s_left_y[16];       // left MB's rightmost recon column
-  __shared__ uint8_t s_left_cb[8];
-  __shared__ uint8_t s_left_cr[8];
-  __shared__ uint8_t s_left_y_new[16];   // staging (committed after recon)
-  __shared__ uint8_t s_left_cb_new[8];
-  __shared__ uint8_t s_left_cr_new[8];
-  __shared__ int s_dc[16];
-  __shared__ int16_t s_lvl[16 * 16];     // luma AC levels of current MB
-  __shared__ int s_dcrec[16];
-  __shared__ int s_cdc[8];               // chroma DCs (cb0..3, cr0..3)
-  __shared__ int s_cdcq[8];              // quantized (hadamard domain)
-  __shared__ int s_cdcrec[8];            // dequantized recon DC
-  __shared__ int16_t s_clvl[8 * 16];     // chroma AC levels
-  __shared__ uint8_t s_src[256];         // current MB source luma
-  __shared__ uint8_t s_csrc[128];        // source chroma (cb 8x8, cr 8x8)
+__device__ inline void wave_lds_fence() {
+  // order our own LDS writes before subsequent cross-lane LDS reads
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
+}
 
+// ---- luma row pipeline (one wave) ----------------------------------------
+__device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
+                         int h, const uint8_t* __restrict__ refY,
+                         uint8_t* __restrict__ curY, int mbw, int mby,
+                         int qp, bool i_slice,
+                         const int16_t* __restrict__ levels_base,
+                         int* __restrict__ meta, int lane,
+                         uint8_t* s_left_y) {
+  const int y0 = mby * 16;
+  const int r = lane >> 2, cq = (lane & 3) * 4;
+  const int g = lane >> 4, c = lane & 15;
+  const int zz = c_zz_of_pos[c];
   bool have_left = false;
 
   for (int mbx = 0; mbx < mbw; ++mbx) {
-    const int x0 = mbx * 16, y0 = mby * 16;
-    const int cx0 = mbx * 8, cy0 = mby * 8;
+    const int x0 = mbx * 16;
     const size_t mb_index = (size_t)mby * mbw + mbx;
-    int16_t* L = levels + mb_index * kLevelsPerMb;
+    int16_t* L = const_cast<int16_t*>(levels_base) +
+                 mb_index * kLevelsPerMb;
     int* M = meta + mb_index * kMetaPerMb;
 
-    int mode = kIntra;
-    int mvx = 0, mvy = 0;
+    int mode = kIntra, mvx = 0, mvy = 0;
     if (!i_slice) {
       int m0 = M[0];
       mode = m0 & 3;
@@ -255,126 +251,95 @@ __global__ void __launch_bounds__(64) k_h264_rows(
     }
 
     if (mode != kIntra) {
-      // skip / inter: motion copy from ref (even integer MVs)
       int ix = mvx >> 2, iy = mvy >> 2;
-      // luma: lane = r*4 + cq; each lane copies 4 bytes
-      {
-        int r = lane >> 2, cq = (lane & 3) * 4;
-        const uint8_t* s =
-            refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
-        uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
-        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
-        if (cq == 12) s_left_y[r] = s[3];  // no pred reads in copy MBs
-      }
-      // chroma: lanes 0..15 copy cb rows, 16..31 cr
-      if (lane < 32) {
-        int comp = lane >> 4;
-        int r = (lane & 15) >> 1, cq = (lane & 1) * 4;
-        const uint8_t* sp = comp ? refCr : refCb;
-        uint8_t* dp = comp ? curCr : curCb;
-        const uint8_t* s =
-            sp + (size_t)(cy0 + (iy >> 1) + r) * cpitch + cx0 + (ix >> 1) + cq;
-        uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
-        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
-        if (cq == 4) {
-          if (comp) s_left_cr[r] = s[3]; else s_left_cb[r] = s[3];
-        }
-      }
-      __syncthreads();
+      const uint8_t* s = refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
+      uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
+      d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
+      if (cq == 12) s_left_y[r] = s[3];
+      wave_lds_fence();
       have_left = true;
       continue;
     }
 
-    // ---------------- intra I16x16 ----------------
-    // load src MB (clamped edge replication)
-    int r = lane >> 2, cq = (lane & 3) * 4;
-    int srow = min(y0 + r, h - 1);
-    int p0, p1, p2, p3;
+    // ---- source pixels: lane (r, cq..cq+3) as packed u32 (for shuffles)
+    uint32_t psrc;
     {
-      const uint8_t* s = srcY + (size_t)srow * ypitch;
-      p0 = s[min(x0 + cq + 0, w - 1)];
-      p1 = s[min(x0 + cq + 1, w - 1)];
-      p2 = s[min(x0 + cq + 2, w - 1)];
-      p3 = s[min(x0 + cq + 3, w - 1)];
+      const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+      psrc = (uint32_t)s[min(x0 + cq + 0, w - 1)] |
+             ((uint32_t)s[min(x0 + cq + 1, w - 1)] << 8) |
+             ((uint32_t)s[min(x0 + cq + 2, w - 1)] << 16) |
+             ((uint32_t)s[min(x0 + cq + 3, w - 1)] << 24);
     }
-    // DC value from left recon
+    // DC value + mode costs
     int dcval = 128;
     if (have_left) {
       int part = (lane < 16) ? s_left_y[lane] : 0;
-      int tot = wave_sum_i(part);
-      dcval = (tot + 8) >> 4;
+      dcval = (wave_sum_i(part) + 8) >> 4;
     }
-    // costs
+    int lv = have_left ? s_left_y[r] : 0;
     int costH = 0, costDC = 0;
-    {
-      int lv = have_left ? s_left_y[r] : 0;
-      costDC = abs(p0 - dcval) + abs(p1 - dcval) + abs(p2 - dcval) +
-               abs(p3 - dcval);
-      costH = abs(p0 - lv) + abs(p1 - lv) + abs(p2 - lv) + abs(p3 - lv);
-      costDC = wave_sum_i(costDC);
-      costH = wave_sum_i(costH);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      int pv = (psrc >> (8 * k)) & 0xFF;
+      costDC += abs(pv - dcval);
+      costH += abs(pv - lv);
     }
+    costDC = wave_sum_i(costDC);
+    costH = wave_sum_i(costH);
     const int luma_mode = (have_left && costH < costDC) ? 1 : 2;
 
-    // stash src into LDS for per-block passes
-    s_src[r * 16 + cq + 0] = (uint8_t)p0;
-    s_src[r * 16 + cq + 1] = (uint8_t)p1;
-    s_src[r * 16 + cq + 2] = (uint8_t)p2;
-    s_src[r * 16 + cq + 3] = (uint8_t)p3;
-    __syncthreads();
+    // fetch pixel (py,px) of the MB from the wave's packed registers
+    auto pix_at = [&](int py, int px) -> int {
+      uint32_t v = __shfl(psrc, py * 4 + (px >> 2));
+      return (v >> (8 * (px & 3))) & 0xFF;
+    };
 
-    // 4 passes x 4 blocks: forward DCT + quant AC, collect DC
-    const int g = lane >> 4;        // group = block within pass
-    const int c = lane & 15;        // coeff position (raster)
-    const int zz = c_zz_of_pos[c];
+    // ---- 4 passes x 4 blocks: fdct + quant (+ DC collection in regs)
+    int lvl_p[4];       // this lane's level per pass
+    int dcreg = 0;      // lanes 0..15: DC coefficient of block `lane`
+#pragma unroll
     for (int pass = 0; pass < 4; ++pass) {
-      int blk = pass * 4 + g;       // raster block index
+      int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
       int pred = luma_mode == 1 ? s_left_y[py] : dcval;
-      int resid = (int)s_src[py * 16 + px] - pred;
+      int resid = pix_at(py, px) - pred;
       int coefv = fdct4_wave(resid, lane);
-      if (c == 0) s_dc[blk] = coefv;
+      // collect DC of block (pass*4+gg) into lane (pass*4+gg)
+      int dcv = __shfl(coefv, (lane & 3) * 16);
+      if (lane < 16 && (lane >> 2) == pass) dcreg = dcv;
       int lvl = (c == 0) ? 0 : quant_coeff(coefv, qp, coeff_cls(c));
       lvl = cap12_group(lvl, zz, c != 0, lane);
-      s_lvl[blk * 16 + c] = (int16_t)lvl;
+      lvl_p[pass] = lvl;
       store_lvl_pair(L + kLumaAcOff + blk * 16, c, lvl, lane);
     }
-    __syncthreads();
 
-    // luma DC: hadamard + quant + inverse + dequant (lanes 0..15)
-    {
-      int v = (lane < 16) ? s_dc[lane] : 0;
-      int had = hadamard4_wave(v, lane, true);
-      int qdc = quant_dc_v(had, qp);
-      qdc = cap12_group(qdc, zz, lane < 16, lane);
-      if (lane < 16) store_lvl_pair(L + kLumaDcOff, lane, qdc, lane);
-      int ih = hadamard4_wave(qdc, lane, false);
-      if (lane < 16) s_dcrec[lane] = dequant_luma_dc_v(ih, qp);
-    }
-    __syncthreads();
+    // ---- luma DC: hadamard + quant + inverse + dequant, all in-wave
+    int had = hadamard4_wave(dcreg, lane, true);
+    int qdc = quant_dc_v(had, qp);
+    qdc = cap12_group(qdc, zz, lane < 16, lane);
+    if (lane < 16) store_lvl_pair(L + kLumaDcOff, lane, qdc, lane);
+    int ih = hadamard4_wave(qdc, lane, false);
+    int dcrec = dequant_luma_dc_v(ih, qp);  // valid on lanes 0..15
 
-    // cbp_luma: any AC nonzero?
-    int anyl = 0;
-    for (int i = lane; i < 256; i += 64) anyl |= s_lvl[i] != 0;
-    anyl = wave_sum_i(anyl);
-    const int cbp_luma = anyl ? 15 : 0;
+    // cbp_luma: any AC nonzero
+    int anyl = (lvl_p[0] | lvl_p[1] | lvl_p[2] | lvl_p[3]) != 0 && c != 0;
+    const int cbp_luma = __ballot(anyl) ? 15 : 0;
 
-    // recon passes
+    // ---- recon passes
+#pragma unroll
     for (int pass = 0; pass < 4; ++pass) {
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
       int d = 0;
-      if (c == 0) {
-        d = s_dcrec[blk];
-      } else if (cbp_luma) {
-        d = dequant_c(s_lvl[blk * 16 + c], qp, coeff_cls(c));
-      }
+      if (c == 0)
+        d = __shfl(dcrec, blk);
+      else if (cbp_luma)
+        d = dequant_c(lvl_p[pass], qp, coeff_cls(c));
       int rec = idct4_wave(d, lane);
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
       int pred = luma_mode == 1 ? s_left_y[py] : dcval;
       int pix = clip8(rec + pred);
-      // pack 4 recon bytes into one aligned u32 store (lane c%4==0)
       int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
           p3 = __shfl(pix, lane + 3);
       if ((c & 3) == 0) {
@@ -383,53 +348,120 @@ __global__ void __launch_bounds__(64) k_h264_rows(
             (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
             ((uint32_t)p3 << 24);
       }
-      if (px == 15) s_left_y_new[py] = (uint8_t)pix;
+      if (px == 15) s_left_y[py + 16] = (uint8_t)pix;  // staging half
     }
-    __syncthreads();
-    if (lane < 16) s_left_y[lane] = s_left_y_new[lane];
-    __syncthreads();
+    wave_lds_fence();
+    if (lane < 16) s_left_y[lane] = s_left_y[lane + 16];
+    wave_lds_fence();
 
-    // ---------------- chroma ----------------
+    if (lane == 0) M[0] = kIntra | (luma_mode << 2);
+    have_left = true;
+  }
+}
+
+// ---- chroma row pipeline (one wave) ---------------------------------------
+__device__ void chroma_row(const uint8_t* __restrict__ srcCb,
+                           const uint8_t* __restrict__ srcCr, int cpitch,
+                           int w, int h, const uint8_t* __restrict__ refCb,
+                           const uint8_t* __restrict__ refCr,
+                           uint8_t* __restrict__ curCb,
+                           uint8_t* __restrict__ curCr, int mbw, int mby,
+                           int qpc, bool i_slice,
+                           const int16_t* __restrict__ levels_base,
+                           int* __restrict__ meta, int lane,
+                           uint8_t* s_left_cb, uint8_t* s_left_cr) {
+  const int cy0 = mby * 8;
+  const int g = lane >> 4, c = lane & 15;
+  const int zz = c_zz_of_pos[c];
+  const int cw = (w + 1) / 2, chh = (h + 1) / 2;
+  bool have_left = false;
+
+  for (int mbx = 0; mbx < mbw; ++mbx) {
+    const int cx0 = mbx * 8;
+    const size_t mb_index = (size_t)mby * mbw + mbx;
+    int16_t* L = const_cast<int16_t*>(levels_base) +
+                 mb_index * kLevelsPerMb;
+    int* M = meta + mb_index * kMetaPerMb;
+
+    int mode = kIntra, mvx = 0, mvy = 0;
+    if (!i_slice) {
+      int m0 = M[0];
+      mode = m0 & 3;
+      int m1 = M[1];
+      mvx = (short)(m1 & 0xFFFF);
+      mvy = m1 >> 16;
+    }
+
+    if (mode != kIntra) {
+      int ix = (mvx >> 2) >> 1, iy = (mvy >> 2) >> 1;
+      if (lane < 32) {
+        int comp = lane >> 4;
+        int r = (lane & 15) >> 1, cq = (lane & 1) * 4;
+        const uint8_t* sp = comp ? refCr : refCb;
+        uint8_t* dp = comp ? curCr : curCb;
+        const uint8_t* s =
+            sp + (size_t)(cy0 + iy + r) * cpitch + cx0 + ix + cq;
+        uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
+        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
+        if (cq == 4) {
+          if (comp) s_left_cr[r] = s[3]; else s_left_cb[r] = s[3];
+        }
+      }
+      wave_lds_fence();
+      have_left = true;
+      continue;
+    }
+
+    // ---- source: loader lane (comp = lane>>4, row = (lane&15)>>1,
+    // 4 cols at (lane&1)*4) as packed u32; shuffles serve the passes.
+    uint32_t csrc = 0;
     if (lane < 32) {
       int comp = lane >> 4;
       int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
       const uint8_t* sp = comp ? srcCr : srcCb;
-      int cw = (w + 1) / 2, chh = (h + 1) / 2;
       int sr = min(cy0 + rr, chh - 1);
-      for (int k = 0; k < 4; ++k)
-        s_csrc[comp * 64 + rr * 8 + ccq + k] =
-            sp[(size_t)sr * cpitch + min(cx0 + ccq + k, cw - 1)];
+      csrc = (uint32_t)sp[(size_t)sr * cpitch + min(cx0 + ccq + 0, cw - 1)] |
+             ((uint32_t)sp[(size_t)sr * cpitch +
+                           min(cx0 + ccq + 1, cw - 1)] << 8) |
+             ((uint32_t)sp[(size_t)sr * cpitch +
+                           min(cx0 + ccq + 2, cw - 1)] << 16) |
+             ((uint32_t)sp[(size_t)sr * cpitch +
+                           min(cx0 + ccq + 3, cw - 1)] << 24);
     }
-    __syncthreads();
+    auto cpix_at = [&](int comp, int rr, int cc) -> int {
+      uint32_t v = __shfl(csrc, (comp << 4) | (rr << 1) | (cc >> 2));
+      return (v >> (8 * (cc & 3))) & 0xFF;
+    };
 
-    // chroma mode decision: DC(0) vs H(1), shared across components.
-    // lane -> (comp, row, 2 columns); DC pred of a pixel = avg of the left
-    // 4 recon pixels of its 4x4 sub-block row range (top unavailable).
+    // ---- mode decision: DC(0) vs H(1), summed across both components
     int ccH = 0, ccDC = 0;
     {
-      int comp = lane >> 5;           // 0 cb, 1 cr
-      int idx = lane & 31;            // 8 rows x 4 col-pairs
+      int comp = lane >> 5;
+      int idx = lane & 31;
       int rr = idx >> 2, ccq = (idx & 3) * 2;
       const uint8_t* lc = comp ? s_left_cr : s_left_cb;
       int scy4 = (rr >> 2) * 4;
       int dd = 128;
       if (have_left)
         dd = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
-      int lv = have_left ? lc[rr] : 0;
+      int lvv = have_left ? lc[rr] : 0;
       for (int k = 0; k < 2; ++k) {
-        int sv = s_csrc[comp * 64 + rr * 8 + ccq + k];
+        int sv = cpix_at(comp, rr, ccq + k);
         ccDC += abs(sv - dd);
-        ccH += abs(sv - lv);
+        ccH += abs(sv - lvv);
       }
       ccDC = wave_sum_i(ccDC);
       ccH = wave_sum_i(ccH);
     }
     const int chroma_mode = (have_left && ccH < ccDC) ? 1 : 0;
 
-    // chroma transform: 2 passes x 4 groups (pass0: cb0..3, pass1: cr0..3)
+    // ---- 2 passes x 4 sub-blocks: fdct + quant; DC terms kept in regs
+    int lvl_p[2];
+    int dcpass[2];          // lane holds shfl-collected DC (see below)
+#pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
       int comp = pass;
-      int sub = g;                     // sub-block raster in 8x8
+      int sub = g;
       int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
       int rr = scy + (c >> 2), cc2 = scx + (c & 3);
       const uint8_t* lc = comp ? s_left_cr : s_left_cb;
@@ -440,72 +472,74 @@ __global__ void __launch_bounds__(64) k_h264_rows(
         pred = 128;
         if (have_left) {
           int scy4 = (rr >> 2) * 4;
-          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
+          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2)
+                 >> 2;
         }
       }
-      int resid = (int)s_csrc[comp * 64 + rr * 8 + cc2] - pred;
+      int resid = cpix_at(comp, rr, cc2) - pred;
       int coefv = fdct4_wave(resid, lane);
-      if (c == 0) s_cdc[comp * 4 + sub] = coefv;
+      // every lane grabs all 4 sub-block DCs of this pass (sub k from
+      // lane k*16): lane's slot (lane&3) -> dcpass[pass]
+      dcpass[pass] = __shfl(coefv, (lane & 3) * 16);
       int lvl = (c == 0) ? 0 : quant_coeff(coefv, qpc, coeff_cls(c));
       lvl = cap12_group(lvl, zz, c != 0, lane);
-      s_clvl[(comp * 4 + sub) * 16 + c] = (int16_t)lvl;
+      lvl_p[pass] = lvl;
       store_lvl_pair(L + kChromaAcOff + (comp * 4 + sub) * 16, c, lvl, lane);
     }
-    __syncthreads();
 
-    // chroma DC 2x2 hadamard + quant + recon scaling (lane 0: cb, lane 1: cr)
-    if (lane < 2) {
-      int* d = &s_cdc[lane * 4];
-      int w0 = d[0] + d[1] + d[2] + d[3];
-      int w1 = d[0] - d[1] + d[2] - d[3];
-      int w2 = d[0] + d[1] - d[2] - d[3];
-      int w3 = d[0] - d[1] - d[2] + d[3];
+    // ---- chroma DC 2x2 hadamard/quant/recon, redundantly on all lanes
+    // lane's dcpass[p] holds dc of sub-block (lane&3) of comp p; gather the
+    // full set via 4 shuffles per comp.
+    int qdc_all[8], dcr_all[8];
+    bool any_cdc = false;
+#pragma unroll
+    for (int comp = 0; comp < 2; ++comp) {
+      int d0 = __shfl(dcpass[comp], 0);   // lane with slot 0
+      int d1 = __shfl(dcpass[comp], 1);
+      int d2 = __shfl(dcpass[comp], 2);
+      int d3 = __shfl(dcpass[comp], 3);
+      int w0 = d0 + d1 + d2 + d3, w1 = d0 - d1 + d2 - d3;
+      int w2 = d0 + d1 - d2 - d3, w3 = d0 - d1 - d2 + d3;
       int q0 = quant_dc_v(w0, qpc), q1 = quant_dc_v(w1, qpc);
       int q2 = quant_dc_v(w2, qpc), q3 = quant_dc_v(w3, qpc);
-      s_cdcq[lane * 4 + 0] = q0;
-      s_cdcq[lane * 4 + 1] = q1;
-      s_cdcq[lane * 4 + 2] = q2;
-      s_cdcq[lane * 4 + 3] = q3;
-      L[kChromaDcOff + lane * 4 + 0] = (int16_t)q0;
-      L[kChromaDcOff + lane * 4 + 1] = (int16_t)q1;
-      L[kChromaDcOff + lane * 4 + 2] = (int16_t)q2;
-      L[kChromaDcOff + lane * 4 + 3] = (int16_t)q3;
+      qdc_all[comp * 4 + 0] = q0;
+      qdc_all[comp * 4 + 1] = q1;
+      qdc_all[comp * 4 + 2] = q2;
+      qdc_all[comp * 4 + 3] = q3;
+      any_cdc |= (q0 | q1 | q2 | q3) != 0;
     }
-    __syncthreads();
-    // cbp_chroma
-    int anyc_ac = 0, anyc_dc = 0;
-    for (int i = lane; i < 128; i += 64) anyc_ac |= s_clvl[i] != 0;
-    if (lane < 8) anyc_dc = s_cdcq[lane] != 0;
-    anyc_ac = wave_sum_i(anyc_ac);
-    anyc_dc = wave_sum_i(anyc_dc);
-    const int cbp_chroma = anyc_ac ? 2 : (anyc_dc ? 1 : 0);
-    if (lane < 2) {
-      int* q = &s_cdcq[lane * 4];
-      int dq[4] = {0, 0, 0, 0};
+    if (lane < 8)
+      L[kChromaDcOff + lane] = (int16_t)qdc_all[lane];
+    int anyc = (lvl_p[0] | lvl_p[1]) != 0 && c != 0;
+    const int cbp_chroma = __ballot(anyc) ? 2 : (any_cdc ? 1 : 0);
+#pragma unroll
+    for (int comp = 0; comp < 2; ++comp) {
+      int dq0 = 0, dq1 = 0, dq2 = 0, dq3 = 0;
       if (cbp_chroma >= 1) {
-        int w0 = q[0] + q[1] + q[2] + q[3];
-        int w1 = q[0] - q[1] + q[2] - q[3];
-        int w2 = q[0] + q[1] - q[2] - q[3];
-        int w3 = q[0] - q[1] - q[2] + q[3];
-        dq[0] = dequant_chroma_dc_v(w0, qpc);
-        dq[1] = dequant_chroma_dc_v(w1, qpc);
-        dq[2] = dequant_chroma_dc_v(w2, qpc);
-        dq[3] = dequant_chroma_dc_v(w3, qpc);
+        int* q = &qdc_all[comp * 4];
+        int w0 = q[0] + q[1] + q[2] + q[3], w1 = q[0] - q[1] + q[2] - q[3];
+        int w2 = q[0] + q[1] - q[2] - q[3], w3 = q[0] - q[1] - q[2] + q[3];
+        dq0 = dequant_chroma_dc_v(w0, qpc);
+        dq1 = dequant_chroma_dc_v(w1, qpc);
+        dq2 = dequant_chroma_dc_v(w2, qpc);
+        dq3 = dequant_chroma_dc_v(w3, qpc);
       }
-      for (int k = 0; k < 4; ++k) s_cdcrec[lane * 4 + k] = dq[k];
+      dcr_all[comp * 4 + 0] = dq0;
+      dcr_all[comp * 4 + 1] = dq1;
+      dcr_all[comp * 4 + 2] = dq2;
+      dcr_all[comp * 4 + 3] = dq3;
     }
-    __syncthreads();
 
-    // chroma recon
+    // ---- recon
+#pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
       int comp = pass;
       int sub = g;
       int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
       int rr = scy + (c >> 2), cc2 = scx + (c & 3);
-      int d = (c == 0) ? s_cdcrec[comp * 4 + sub]
+      int d = (c == 0) ? dcr_all[comp * 4 + sub]
                        : (cbp_chroma == 2
-                              ? dequant_c(s_clvl[(comp * 4 + sub) * 16 + c],
-                                          qpc, coeff_cls(c))
+                              ? dequant_c(lvl_p[pass], qpc, coeff_cls(c))
                               : 0);
       int rec = idct4_wave(d, lane);
       const uint8_t* lc = comp ? s_left_cr : s_left_cb;
@@ -516,7 +550,8 @@ __global__ void __launch_bounds__(64) k_h264_rows(
         pred = 128;
         if (have_left) {
           int scy4 = (rr >> 2) * 4;
-          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
+          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2)
+                 >> 2;
         }
       }
       int pix = clip8(rec + pred);
@@ -530,30 +565,53 @@ __global__ void __launch_bounds__(64) k_h264_rows(
             ((uint32_t)p3 << 24);
       }
       if (cc2 == 7) {
-        if (comp) s_left_cr_new[rr] = (uint8_t)pix;
-        else s_left_cb_new[rr] = (uint8_t)pix;
+        if (comp) s_left_cr[rr + 8] = (uint8_t)pix;
+        else s_left_cb[rr + 8] = (uint8_t)pix;
       }
     }
-    __syncthreads();
+    wave_lds_fence();
     if (lane < 8) {
-      s_left_cb[lane] = s_left_cb_new[lane];
-      s_left_cr[lane] = s_left_cr_new[lane];
+      s_left_cb[lane] = s_left_cb[lane + 8];
+      s_left_cr[lane] = s_left_cr[lane + 8];
     }
+    wave_lds_fence();
 
-    // meta out
-    if (lane == 0) {
-      M[0] = kIntra | (luma_mode << 2) | (chroma_mode << 5);
-      M[1] = 0;
-    }
-    __syncthreads();
+    if (lane == 0) M[1] = chroma_mode;   // m1 is unused for intra MBs
     have_left = true;
   }
 }
 
+__global__ void __launch_bounds__(128) k_h264_rows(
+    const uint8_t* __restrict__ srcY, const uint8_t* __restrict__ srcCb,
+    const uint8_t* __restrict__ srcCr, int ypitch, int cpitch, int w, int h,
+    const uint8_t* __restrict__ refY, const uint8_t* __restrict__ refCb,
+    const uint8_t* __restrict__ refCr, uint8_t* __restrict__ curY,
+    uint8_t* __restrict__ curCb, uint8_t* __restrict__ curCr, int mbw,
+    const RowJob* __restrict__ jobs, int16_t* __restrict__ levels,
+    int* __restrict__ meta) {
+  const RowJob job = jobs[blockIdx.x];
+  const int qp = job.qp;
+  const bool i_slice = (job.flags & 1) != 0;
+  const int mby = job.mb_row;
+  const int lane = threadIdx.x & 63;
+
+  __shared__ uint8_t s_left_y[32];       // [0..15] current, [16..31] staging
+  __shared__ uint8_t s_left_cb[16];      // [0..7] current, [8..15] staging
+  __shared__ uint8_t s_left_cr[16];
+
+  if (threadIdx.x < 64) {
+    luma_row(srcY, ypitch, w, h, refY, curY, mbw, mby, qp, i_slice, levels,
+             meta, lane, s_left_y);
+  } else {
+    chroma_row(srcCb, srcCr, cpitch, w, h, refCb, refCr, curCb, curCr, mbw,
+               mby, dev_chroma_qp(qp), i_slice, levels, meta, lane,
+               s_left_cb, s_left_cr);
+  }
+}
+
 // ---------------------------------------------------------------------------
-// P-frame mode decision: one workgroup (64 lanes) per MB of each P row.
-// Produces meta: skip / inter(mv) / intra. Thresholds match the CPU
-// encoder (cpu/h264/encoder.cpp).
+// P-frame mode decision (diamond search reference path): one workgroup
+// (64 lanes) per MB of each P row. Thresholds match the CPU encoder.
 __global__ void __launch_bounds__(64) k_h264_me(
     const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
     const uint8_t* __restrict__ refY, int mbw, int frame_w_mb16,
@@ -568,7 +626,6 @@ __global__ void __launch_bounds__(64) k_h264_me(
   const size_t mb_index = (size_t)mby * mbw + mbx;
 
   const int r = lane >> 2, cq = (lane & 3) * 4;
-  // source pixels (clamped)
   int sp[4];
   {
     const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
@@ -599,8 +656,8 @@ __global__ void __launch_bounds__(64) k_h264_me(
     for (int iter = 0; iter < 8; ++iter) {
       int cx = bmx, cy = bmy;
       bool improved = false;
-      for (int p = 0; p < 8; ++p) {
-        int mx = cx + pat[p][0], my = cy + pat[p][1];
+      for (int pi = 0; pi < 8; ++pi) {
+        int mx = cx + pat[pi][0], my = cy + pat[pi][1];
         if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
             y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1 ||
             abs(mx) > 16 || abs(my) > 16)
