@@ -894,7 +894,7 @@ void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
                       const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
                       hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_rows, dim3(n_jobs), dim3(64), 0, stream, srcY,
+  hipLaunchKernelGGL(k_h264_rows, dim3(n_jobs), dim3(128), 0, stream, srcY,
                      srcCb, srcCr, ypitch, cpitch, w, h, refY, refCb, refCr,
                      curY, curCb, curCr, mbw, d_jobs, d_levels, d_meta);
 }
