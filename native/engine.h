@@ -45,6 +45,16 @@ class EncodePipeline {
   virtual void encode_frame(const RawFrame& frame, const FrameContext& ctx,
                             const Emit& emit) = 0;
   virtual const char* name() const = 0;
+
+  // Test-only introspection: reconstruction planes + (GPU) level/meta
+  // buffers of the last encoded frame. Returns false when unsupported.
+  struct DebugDump {
+    int w = 0, h = 0, ypitch = 0, cpitch = 0;
+    std::vector<uint8_t> y, cb, cr;
+    std::vector<int16_t> levels;
+    std::vector<int> meta;
+  };
+  virtual bool debug_dump(DebugDump&) { return false; }
 };
 
 std::unique_ptr<EncodePipeline> make_cpu_jpeg_pipeline(const CaptureSettings&);

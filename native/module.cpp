@@ -247,7 +247,7 @@ PYBIND11_MODULE(_native, m) {
   m.def(
       "_pipeline_encode",
       [](const std::string& kind, py::list frames, int w, int h, int qp,
-         int stripe_h, int output_mode) {
+         int stripe_h, int output_mode, bool dump) -> py::object {
         CaptureSettings s;
         s.capture_width = w;
         s.capture_height = h;
@@ -300,11 +300,33 @@ PYBIND11_MODULE(_native, m) {
           }
           result.append(frame_out);
         }
-        return result;
+        if (dump) {
+          EncodePipeline::DebugDump d;
+          if (!p->debug_dump(d))
+            throw std::runtime_error("pipeline has no debug_dump");
+          py::dict dd;
+          dd["w"] = d.w;
+          dd["h"] = d.h;
+          dd["ypitch"] = d.ypitch;
+          dd["cpitch"] = d.cpitch;
+          dd["y"] = py::bytes(reinterpret_cast<char*>(d.y.data()),
+                              d.y.size());
+          dd["cb"] = py::bytes(reinterpret_cast<char*>(d.cb.data()),
+                               d.cb.size());
+          dd["cr"] = py::bytes(reinterpret_cast<char*>(d.cr.data()),
+                               d.cr.size());
+          dd["levels"] = py::bytes(
+              reinterpret_cast<char*>(d.levels.data()),
+              d.levels.size() * sizeof(int16_t));
+          dd["meta"] = py::bytes(reinterpret_cast<char*>(d.meta.data()),
+                                 d.meta.size() * sizeof(int));
+          return py::object(py::make_tuple(result, dd));
+        }
+        return py::object(result);
       },
       py::arg("kind"), py::arg("frames"), py::arg("w"), py::arg("h"),
       py::arg("qp") = 26, py::arg("stripe_h") = 64,
-      py::arg("output_mode") = 1,
+      py::arg("output_mode") = 1, py::arg("dump") = false,
       "Test hook: run frames through a named encode pipeline.");
 
   // ---- persistent pipeline handle for benchmarking ------------------------

@@ -509,6 +509,34 @@ class HipH264Pipeline : public EncodePipeline {
                              hipMemcpyDeviceToDevice, stream_));
   }
 
+ public:
+  bool debug_dump(DebugDump& d) override {
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    d.w = w_;
+    d.h = h_;
+    d.ypitch = ypitch_;
+    d.cpitch = cpitch_;
+    size_t ysz = static_cast<size_t>(ypitch_) * mbh_ * 16;
+    size_t csz = static_cast<size_t>(cpitch_) * mbh_ * 8;
+    d.y.resize(ysz);
+    d.cb.resize(csz);
+    d.cr.resize(csz);
+    // after encode_frame the recon has been copied cur -> ref
+    HIP_CHECK(hipMemcpy(d.y.data(), d_refY_, ysz, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(d.cb.data(), d_refCb_, csz, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(d.cr.data(), d_refCr_, csz, hipMemcpyDeviceToHost));
+    size_t nlv = static_cast<size_t>(mbw_) * mbh_ * h264gpu::kLevelsPerMb;
+    size_t nmt = static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb;
+    d.levels.resize(nlv);
+    d.meta.resize(nmt);
+    HIP_CHECK(hipMemcpy(d.levels.data(), d_levels_, nlv * sizeof(int16_t),
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(d.meta.data(), d_meta_, nmt * sizeof(int),
+                        hipMemcpyDeviceToHost));
+    return true;
+  }
+
+ private:
   void alloc_for(int w, int h) {
     HIP_CHECK(hipStreamSynchronize(stream_));
     for (void* p : device_ptrs_)

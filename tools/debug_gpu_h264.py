@@ -106,6 +106,51 @@ def run(w=256, h=128, n=1, qp=28, stripe=64):
                 print(f"  headers identical ({ng} mbs)")
 
 
+def run_recon_check(w=256, h=128, qp=28, stripe=64):
+    """Kernel recon vs decoder recon of the kernel's own stream."""
+    rng = np.random.default_rng(7)
+    f = rng.integers(0, 256, (h, w, 4), dtype=np.uint8)
+    f[:, :, 3] = 255
+    out, dump = _native._pipeline_encode(
+        "gpu", [np.ascontiguousarray(f)], w, h, qp, stripe, 1, True)
+    yp, cp = dump["ypitch"], dump["cpitch"]
+    ky = np.frombuffer(dump["y"], np.uint8).reshape(-1, yp)[:h, :w]
+    kcb = np.frombuffer(dump["cb"], np.uint8).reshape(-1, cp)[:h//2, :w//2]
+    kcr = np.frombuffer(dump["cr"], np.uint8).reshape(-1, cp)[:h//2, :w//2]
+    rows = {}
+    for fr in out:
+        for data, y, _, _ in fr:
+            rows.setdefault(y, bytearray()).extend(bytes(data))
+    for y0 in sorted(rows):
+        dec = R.Decoder().decode(bytes(rows[y0]))
+        dy, dcb, dcr = dec[0]
+        sh = dy.shape[0]
+        sky = ky[y0:y0+sh]
+        skb = kcb[y0//2:y0//2+sh//2]
+        skr = kcr[y0//2:y0//2+sh//2]
+        dpy = np.abs(sky.astype(int) - dy.astype(int))
+        dpb = np.abs(skb.astype(int) - dcb.astype(int))
+        dpr = np.abs(skr.astype(int) - dcr.astype(int))
+        print(f"stripe y={y0}: kernel-vs-decoder maxdiff "
+              f"Y={dpy.max()} Cb={dpb.max()} Cr={dpr.max()}")
+        if dpy.max() > 0:
+            ys, xs = np.nonzero(dpy)
+            i = np.argmin(ys * 10000 + xs)
+            py_, px_ = int(ys[i]), int(xs[i])
+            print(f"  first Y diff at ({py_},{px_}) mb=({py_//16},{px_//16})"
+                  f" blkpos=({py_%16},{px_%16})"
+                  f" kernel={sky[py_,px_]} decoder={dy[py_,px_]}")
+            mby, mbx = py_ // 16, px_ // 16
+            print("  kernel MB:")
+            print(sky[mby*16:mby*16+16, mbx*16:mbx*16+16])
+            print("  decoder MB:")
+            print(dy[mby*16:mby*16+16, mbx*16:mbx*16+16])
+            break
+
+
 if __name__ == "__main__":
-    n = int(sys.argv[1]) if len(sys.argv) > 1 else 1
-    run(n=n)
+    if len(sys.argv) > 1 and sys.argv[1] == "recon":
+        run_recon_check()
+    else:
+        n = int(sys.argv[1]) if len(sys.argv) > 1 else 1
+        run(n=n)
