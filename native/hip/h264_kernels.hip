@@ -331,9 +331,12 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
     for (int pass = 0; pass < 4; ++pass) {
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
+      // shuffle OUTSIDE the divergent branch: ds_bpermute from a lane that
+      // is not executing the instruction returns undefined data
+      int dcb = __shfl(dcrec, blk);
       int d = 0;
       if (c == 0)
-        d = __shfl(dcrec, blk);
+        d = dcb;
       else if (cbp_luma)
         d = dequant_c(lvl_p[pass], qp, coeff_cls(c));
       int rec = idct4_wave(d, lane);
