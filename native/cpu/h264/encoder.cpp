@@ -60,6 +60,23 @@ struct MbTotals {
   uint8_t cb[4] = {}, cr[4] = {};
 };
 
+// coded_block_pattern me(v) codeNum for Inter prediction (Table 9-4,
+// ChromaArrayType=1). We only ever emit cbp_luma in {0,15} and
+// cbp_chroma in {0,1,2}, so only these six of the 48 entries are
+// reachable; the subset decoder (tests/h264_ref_decoder.py) carries the
+// same six and asserts nothing else appears.
+inline int inter_cbp_codenum(int cbp) {
+  switch (cbp) {
+    case 0: return 0;
+    case 16: return 1;
+    case 32: return 6;
+    case 15: return 11;
+    case 47: return 12;
+    case 31: return 19;
+  }
+  return -1;  // unreachable
+}
+
 }  // namespace
 
 struct StripeEncoder::Impl {
@@ -458,6 +475,190 @@ struct StripeEncoder::Impl {
     return 0;
   }
 
+  // ---- P_L0_16x16 with coded residual -------------------------------------
+  // Luma residual = src - MC pred, coded as 16 full 4x4 blocks (no DC
+  // Hadamard for inter); chroma as 2x2-Hadamard DC + AC. Inter quant
+  // rounding (f = 2^qbits/6). MVs are even integers so the chroma MC is
+  // an integer copy (mv/2). Mirrors the GPU row kernel's inter path.
+  void encode_p16(BitWriter& bw, int mbx, int mby, int qp, int mvx, int mvy,
+                  RowCtx& ctx, MbTotals& tot) {
+    const int x0 = mbx * 16, y0 = mby * 16;
+    const int cx0 = mbx * 8, cy0 = mby * 8;
+    const int cmx = mvx / 2, cmy = mvy / 2;
+
+    // ----- luma: 16 blocks, full 16-coeff zigzag, inter quant
+    int zz[16][16];
+    int dq_coef[16][16];
+    bool any_l = false;
+    for (int by = 0; by < 4; ++by)
+      for (int bx = 0; bx < 4; ++bx) {
+        int b = by * 4 + bx;
+        int resid[16], coef[16];
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = y0 + by * 4 + r, xx = x0 + bx * 4 + c;
+            resid[4 * r + c] = int(src.y.row(yy)[xx]) -
+                               int(ref.y.row(yy + mvy)[xx + mvx]);
+          }
+        fdct4x4(resid, coef);
+        for (int i = 0; i < 16; ++i) {
+          int pos = kZigzag4[i];
+          zz[b][i] = quant_coeff(coef[pos], qp,
+                                 coeff_class(pos >> 2, pos & 3), false);
+        }
+        cap_coeffs(zz[b], 16);
+        for (int i = 0; i < 16; ++i) any_l |= zz[b][i] != 0;
+      }
+    const int cbp_luma = any_l ? 15 : 0;
+
+    // ----- chroma: DC 2x2 Hadamard + AC, inter quant
+    const int qpc = chroma_qp(qp);
+    int cdc[2][4], czz[2][4][15], cqdc[2][4];
+    bool c_any_ac = false, c_any_dc = false;
+    for (int comp = 0; comp < 2; ++comp) {
+      Plane& sp = comp ? src.cr : src.cb;
+      Plane& rp = comp ? ref.cr : ref.cb;
+      for (int sub = 0; sub < 4; ++sub) {
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int resid[16], coef[16];
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = cy0 + scy + r, xx = cx0 + scx + c;
+            resid[4 * r + c] =
+                int(sp.row(yy)[xx]) - int(rp.row(yy + cmy)[xx + cmx]);
+          }
+        fdct4x4(resid, coef);
+        cdc[comp][sub] = coef[0];
+        for (int i = 1; i < 16; ++i) {
+          int pos = kZigzag4[i];
+          czz[comp][sub][i - 1] = quant_coeff(
+              coef[pos], qpc, coeff_class(pos >> 2, pos & 3), false);
+        }
+        cap_coeffs(czz[comp][sub], 15);
+        for (int i = 0; i < 15; ++i) c_any_ac |= czz[comp][sub][i] != 0;
+      }
+      int* d = cdc[comp];
+      int w0 = d[0] + d[1] + d[2] + d[3], w1 = d[0] - d[1] + d[2] - d[3];
+      int w2 = d[0] + d[1] - d[2] - d[3], w3 = d[0] - d[1] - d[2] + d[3];
+      cqdc[comp][0] = quant_dc(w0, qpc, false);
+      cqdc[comp][1] = quant_dc(w1, qpc, false);
+      cqdc[comp][2] = quant_dc(w2, qpc, false);
+      cqdc[comp][3] = quant_dc(w3, qpc, false);
+      for (int i = 0; i < 4; ++i) c_any_dc |= cqdc[comp][i] != 0;
+    }
+    const int cbp_chroma = c_any_ac ? 2 : (c_any_dc ? 1 : 0);
+    const int cbp = (cbp_chroma << 4) | cbp_luma;
+
+    // ----- bitstream
+    bw.ue(0);  // mb_type P_L0_16x16
+    int mvpx = ctx.have_left && ctx.left_is_inter ? ctx.left_mvx : 0;
+    int mvpy = ctx.have_left && ctx.left_is_inter ? ctx.left_mvy : 0;
+    bw.se(mvx * 4 - mvpx);
+    bw.se(mvy * 4 - mvpy);
+    bw.ue(inter_cbp_codenum(cbp));
+    MbTotals newtot;
+    if (cbp) {
+      bw.se(0);  // mb_qp_delta
+      if (cbp_luma) {
+        for (int blk = 0; blk < 16; ++blk) {
+          int bx, by;
+          blk_xy(blk, bx, by);
+          int r = by * 4 + bx;
+          int nC = luma_nc_for(ctx, newtot, bx, by);
+          int tc = cavlc_residual(bw, zz[r], 16, nC);
+          newtot.luma[r] = static_cast<uint8_t>(tc);
+        }
+      }
+      if (cbp_chroma > 0) {
+        int zdc[4];
+        for (int comp = 0; comp < 2; ++comp) {
+          for (int i = 0; i < 4; ++i) zdc[i] = cqdc[comp][i];
+          cavlc_residual(bw, zdc, 4, -1);
+        }
+      }
+      if (cbp_chroma == 2) {
+        for (int comp = 0; comp < 2; ++comp) {
+          uint8_t* t = comp ? newtot.cr : newtot.cb;
+          for (int sub = 0; sub < 4; ++sub) {
+            int cx = sub & 1, cy = sub >> 1;
+            int nC = chroma_nc_for(ctx, t, comp, cx, cy);
+            int tc = cavlc_residual(bw, czz[comp][sub], 15, nC);
+            t[cy * 2 + cx] = static_cast<uint8_t>(tc);
+          }
+        }
+      }
+    }
+    tot = newtot;
+
+    // ----- recon: MC pred + idct(dequant)
+    for (int by = 0; by < 4; ++by)
+      for (int bx = 0; bx < 4; ++bx) {
+        int b = by * 4 + bx;
+        if (!cbp_luma) {
+          for (int r = 0; r < 4; ++r) {
+            int yy = y0 + by * 4 + r, xx = x0 + bx * 4;
+            std::memcpy(cur.y.row(yy) + xx, ref.y.row(yy + mvy) + xx + mvx,
+                        4);
+          }
+          continue;
+        }
+        int dqb[16] = {};
+        for (int i = 0; i < 16; ++i) {
+          int pos = kZigzag4[i];
+          dqb[pos] =
+              dequant_coeff(zz[b][i], qp, coeff_class(pos >> 2, pos & 3));
+        }
+        int rec[16];
+        idct4x4(dqb, rec);
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = y0 + by * 4 + r, xx = x0 + bx * 4 + c;
+            cur.y.row(yy)[xx] =
+                clip8(rec[4 * r + c] + int(ref.y.row(yy + mvy)[xx + mvx]));
+          }
+      }
+    for (int comp = 0; comp < 2; ++comp) {
+      Plane& rc = comp ? cur.cr : cur.cb;
+      Plane& rp = comp ? ref.cr : ref.cb;
+      int dcq[4] = {0, 0, 0, 0};
+      if (cbp_chroma >= 1) {
+        int* q = cqdc[comp];
+        int w0 = q[0] + q[1] + q[2] + q[3], w1 = q[0] - q[1] + q[2] - q[3];
+        int w2 = q[0] + q[1] - q[2] - q[3], w3 = q[0] - q[1] - q[2] + q[3];
+        dcq[0] = dequant_chroma_dc(w0, qpc);
+        dcq[1] = dequant_chroma_dc(w1, qpc);
+        dcq[2] = dequant_chroma_dc(w2, qpc);
+        dcq[3] = dequant_chroma_dc(w3, qpc);
+      }
+      for (int sub = 0; sub < 4; ++sub) {
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        if (cbp_chroma == 0) {
+          for (int r = 0; r < 4; ++r) {
+            int yy = cy0 + scy + r, xx = cx0 + scx;
+            std::memcpy(rc.row(yy) + xx, rp.row(yy + cmy) + xx + cmx, 4);
+          }
+          continue;
+        }
+        int dqb[16] = {};
+        dqb[0] = dcq[sub];
+        if (cbp_chroma == 2)
+          for (int i = 1; i < 16; ++i) {
+            int pos = kZigzag4[i];
+            dqb[pos] = dequant_coeff(czz[comp][sub][i - 1], qpc,
+                                     coeff_class(pos >> 2, pos & 3));
+          }
+        int rec[16];
+        idct4x4(dqb, rec);
+        for (int r = 0; r < 4; ++r)
+          for (int c = 0; c < 4; ++c) {
+            int yy = cy0 + scy + r, xx = cx0 + scx + c;
+            rc.row(yy)[xx] =
+                clip8(rec[4 * r + c] + int(rp.row(yy + cmy)[xx + cmx]));
+          }
+      }
+    }
+  }
+
   // ---- P macroblock helpers -----------------------------------------------
   long sad16(const uint8_t* a, int ap, const uint8_t* b, int bp) const {
     long s = 0;
@@ -519,7 +720,10 @@ struct StripeEncoder::Impl {
     }
 
     const long skip_thresh = 48L << (qp / 6);       // tuned-for-screen default
-    const long inter_thresh = 2 * skip_thresh;
+    // With coded inter residuals the MC only has to be a decent predictor,
+    // not a near-match: fall back to intra only when even the best MV
+    // leaves mostly-uncorrelated content (fresh content / occlusion).
+    const long inter_thresh = 6 * skip_thresh;
 
     for (int mb_row = 0; mb_row < mbh; ++mb_row) {
       BitWriter b;
@@ -583,23 +787,20 @@ struct StripeEncoder::Impl {
           }
           if (!improved) break;
         }
-        if (best <= inter_thresh && !(best_mvx == 0 && best_mvy == 0)) {
-          // P_L0_16x16, zero residual
+        if (best <= inter_thresh) {
+          // P_L0_16x16 with coded residual (cbp may still come out 0)
           flush_skip_run(b, ctx);
-          b.ue(0);  // mb_type P_L0_16x16
-          int mvpx = ctx.have_left && ctx.left_is_inter ? ctx.left_mvx : 0;
-          int mvpy = ctx.have_left && ctx.left_is_inter ? ctx.left_mvy : 0;
-          b.se(best_mvx * 4 - mvpx);
-          b.se(best_mvy * 4 - mvpy);
-          b.ue(0);  // coded_block_pattern = 0 (inter me(v): codeNum 0)
-          copy_mb_from_ref(mbx, mb_row, best_mvx, best_mvy);
+          encode_p16(b, mbx, mb_row, qp, best_mvx, best_mvy, ctx, tot);
           ctx.have_left = true;
           ctx.left_is_inter = true;
           ctx.left_mvx = best_mvx * 4;
           ctx.left_mvy = best_mvy * 4;
-          std::memset(ctx.left_luma_nc, 0, 4);
-          std::memset(ctx.left_cb_nc, 0, 2);
-          std::memset(ctx.left_cr_nc, 0, 2);
+          for (int by = 0; by < 4; ++by)
+            ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
+          for (int cy = 0; cy < 2; ++cy) {
+            ctx.left_cb_nc[cy] = tot.cb[cy * 2 + 1];
+            ctx.left_cr_nc[cy] = tot.cr[cy * 2 + 1];
+          }
           if (stats) ++stats->mb_inter;
         } else {
           flush_skip_run(b, ctx);

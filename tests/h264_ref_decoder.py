@@ -408,7 +408,7 @@ class Decoder:
             mb_type = br.ue()
             if is_p and mb_type < 5:
                 assert mb_type == 0, f"subset: P mb_type {mb_type}"
-                self.decode_p16(br, mbx, mb_row, ctx)
+                self.decode_p16(br, mbx, mb_row, ctx, qp)
             else:
                 it = mb_type - 5 if is_p else mb_type
                 assert 1 <= it <= 24, f"subset: mb_type {mb_type}"
@@ -427,10 +427,18 @@ class Decoder:
         self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = self.ref_cr[cy0:cy0 + 8,
                                                         cx0:cx0 + 8]
 
-    def decode_p16(self, br, mbx, mby, ctx):
+    # coded_block_pattern me(v) inverse for Inter prediction: the six
+    # Table 9-4 entries our encoder can emit (cbp_luma in {0,15},
+    # cbp_chroma in {0,1,2}).
+    INTER_CBP_FROM_CODENUM = {0: 0, 1: 16, 6: 32, 11: 15, 12: 47, 19: 31}
+
+    def decode_p16(self, br, mbx, mby, ctx, qp):
         mvdx, mvdy = br.se(), br.se()
         cbp_cn = br.ue()
-        assert cbp_cn == 0, "subset: inter residual not supported"
+        assert cbp_cn in self.INTER_CBP_FROM_CODENUM, \
+            f"subset: inter cbp codeNum {cbp_cn}"
+        cbp = self.INTER_CBP_FROM_CODENUM[cbp_cn]
+        cbp_luma, cbp_chroma = cbp & 15, cbp >> 4
         mvpx, mvpy = ctx["left_mv"] if (ctx["left_avail"] and
                                         ctx["left_inter"]) else (0, 0)
         mvx, mvy = mvdx + mvpx, mvdy + mvpy
@@ -438,17 +446,85 @@ class Decoder:
         ix, iy = mvx // 4, mvy // 4
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
-        self.y[y0:y0 + 16, x0:x0 + 16] = \
-            self.ref_y[y0 + iy:y0 + iy + 16, x0 + ix:x0 + ix + 16]
-        self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = \
-            self.ref_cb[cy0 + iy // 2:cy0 + iy // 2 + 8,
-                        cx0 + ix // 2:cx0 + ix // 2 + 8]
-        self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = \
-            self.ref_cr[cy0 + iy // 2:cy0 + iy // 2 + 8,
-                        cx0 + ix // 2:cx0 + ix // 2 + 8]
+        pred_y = self.ref_y[y0 + iy:y0 + iy + 16,
+                            x0 + ix:x0 + ix + 16].astype(np.int64)
+        pred_cb = self.ref_cb[cy0 + iy // 2:cy0 + iy // 2 + 8,
+                              cx0 + ix // 2:cx0 + ix // 2 + 8].astype(
+                                  np.int64)
+        pred_cr = self.ref_cr[cy0 + iy // 2:cy0 + iy // 2 + 8,
+                              cx0 + ix // 2:cx0 + ix // 2 + 8].astype(
+                                  np.int64)
+        new_luma_nc = np.zeros((4, 4), np.int32)
+        new_cb_nc = np.zeros((2, 2), np.int32)
+        new_cr_nc = np.zeros((2, 2), np.int32)
+        if cbp:
+            qp = qp + br.se()  # mb_qp_delta
+        if cbp_luma:
+            # 16 full 4x4 blocks, Z-order, no DC Hadamard for inter
+            for blk in range(16):
+                bx, by = blk_xy(blk)
+                if bx > 0:
+                    nC = int(new_luma_nc[by, bx - 1])
+                elif ctx["left_avail"]:
+                    nC = ctx["left_luma_nc"][by]
+                else:
+                    nC = 0
+                zz, tc = residual_cavlc(br, 16, nC)
+                new_luma_nc[by, bx] = tc
+                coeffs = np.zeros(16, np.int64)
+                for i in range(16):
+                    coeffs[ZZ4[i]] = dequant_ac(zz[i], qp,
+                                                coeff_class(ZZ4[i] >> 2,
+                                                            ZZ4[i] & 3))
+                rec = idct4(coeffs).reshape(4, 4)
+                pred_y[by * 4:by * 4 + 4, bx * 4:bx * 4 + 4] += rec
+        qpc = chroma_qp(qp + self.pps["chroma_qp_offset"])
+        cdc = {"cb": [0] * 4, "cr": [0] * 4}
+        if cbp_chroma >= 1:
+            for comp in ("cb", "cr"):
+                zz, _ = residual_cavlc(br, 4, -1)
+                q = zz
+                w0 = q[0] + q[1] + q[2] + q[3]
+                w1 = q[0] - q[1] + q[2] - q[3]
+                w2 = q[0] + q[1] - q[2] - q[3]
+                w3 = q[0] - q[1] - q[2] + q[3]
+                cdc[comp] = [dequant_chroma_dc(w, qpc)
+                             for w in (w0, w1, w2, w3)]
+        cac = {"cb": np.zeros((4, 16), np.int64),
+               "cr": np.zeros((4, 16), np.int64)}
+        if cbp_chroma == 2:
+            for comp, newnc, leftnc in (("cb", new_cb_nc, ctx["left_cb_nc"]),
+                                        ("cr", new_cr_nc,
+                                         ctx["left_cr_nc"])):
+                for sub in range(4):
+                    cx, cy = sub & 1, sub >> 1
+                    if cx > 0:
+                        nC = int(newnc[cy, 0])
+                    elif ctx["left_avail"]:
+                        nC = leftnc[cy]
+                    else:
+                        nC = 0
+                    zz, tc = residual_cavlc(br, 15, nC)
+                    newnc[cy, cx] = tc
+                    for i in range(1, 16):
+                        cac[comp][sub][ZZ4[i]] = dequant_ac(
+                            zz[i - 1], qpc,
+                            coeff_class(ZZ4[i] >> 2, ZZ4[i] & 3))
+        if cbp_chroma >= 1:
+            for comp, pred in (("cb", pred_cb), ("cr", pred_cr)):
+                for sub in range(4):
+                    scx, scy = (sub & 1) * 4, (sub >> 1) * 4
+                    coeffs = cac[comp][sub].copy()
+                    coeffs[0] = cdc[comp][sub]
+                    rec = idct4(coeffs).reshape(4, 4)
+                    pred[scy:scy + 4, scx:scx + 4] += rec
+        self.y[y0:y0 + 16, x0:x0 + 16] = np.clip(pred_y, 0, 255)
+        self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cb, 0, 255)
+        self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cr, 0, 255)
         ctx.update(left_avail=True, left_inter=True, left_mv=(mvx, mvy),
-                   left_luma_nc=[0] * 4, left_cb_nc=[0] * 2,
-                   left_cr_nc=[0] * 2)
+                   left_luma_nc=[int(new_luma_nc[by, 3]) for by in range(4)],
+                   left_cb_nc=[int(new_cb_nc[cy, 1]) for cy in range(2)],
+                   left_cr_nc=[int(new_cr_nc[cy, 1]) for cy in range(2)])
 
     def decode_i16(self, br, mbx, mby, i16_type, qp, ctx):
         t = i16_type - 1

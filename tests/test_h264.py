@@ -138,3 +138,68 @@ def test_static_content_all_skip():
     nmb = (w // 16) * (h // 16)
     assert r2["mb_skip"] == nmb, f"static frame: {r2['mb_skip']}/{nmb} skip"
     assert len(r2["data"]) < 60  # a few bytes of slice headers only
+
+
+def test_inter_residuals_bit_exact_and_efficient():
+    """A global small brightness step is too large to skip but an excellent
+    MC candidate: every MB must be coded P_L0_16x16 WITH residual (not
+    intra), the decode must match the encoder recon bit-exactly, and the
+    P frame must be far smaller than an intra refresh of the same change."""
+    w, h = 192, 96
+    rng = np.random.default_rng(21)
+    base = noise_frame(rng, w, h)
+    lifted = base.copy()
+    lifted[:, :, :3] = np.clip(base[:, :, :3].astype(np.int16) + 6,
+                               0, 255).astype(np.uint8)
+    enc = hipflux.H264Encoder(w, h)
+    r0 = enc.encode(base.tobytes(), qp=28, idr=True)
+    r1 = enc.encode(lifted.tobytes(), qp=28)
+    rec1 = recon_planes(enc, w, h)
+    nmb = (w // 16) * (h // 16)
+    assert r1["mb_inter"] == nmb, \
+        f"expected all-inter: {r1['mb_inter']}/{nmb} (skip {r1['mb_skip']}," \
+        f" intra {r1['mb_intra']})"
+    frames = Decoder().decode(r0["data"] + r1["data"])
+    assert len(frames) == 2
+    for d, r, name in zip(frames[1], rec1, "y cb cr".split()):
+        assert np.array_equal(d, r), f"P resid frame plane {name} mismatch"
+    # the inter-coded step costs a small fraction of the IDR
+    assert len(r1["data"]) < len(r0["data"]) * 0.35, \
+        f"P {len(r1['data'])} vs IDR {len(r0['data'])}"
+    # quality must track the IDR's (both bounded by the QP28 quant noise;
+    # without coded residuals the +6 step alone would cap PSNR at ~32.5)
+    y0, _, _ = hipflux.bgrx_to_yuv420(base.tobytes(), w, h)
+    y1, _, _ = hipflux.bgrx_to_yuv420(lifted.tobytes(), w, h)
+    p_idr = psnr(frames[0][0], np.frombuffer(y0, np.uint8).reshape(h, w))
+    p_p = psnr(frames[1][0], np.frombuffer(y1, np.uint8).reshape(h, w))
+    assert p_p > p_idr - 2.0, \
+        f"P-frame luma PSNR {p_p:.1f} far below IDR {p_idr:.1f}"
+
+
+def test_inter_residual_moving_content_chain():
+    """Rolling noise: MC aligns perfectly after the shift, small residuals
+    at block seams. Chain of 6 P frames must stay bit-exact vs recon."""
+    w, h = 160, 96
+    rng = np.random.default_rng(23)
+    base = noise_frame(rng, w, h)
+    enc = hipflux.H264Encoder(w, h)
+    stream = b""
+    recons = []
+    inter_seen = 0
+    for i in range(6):
+        img = np.roll(base, 2 * i, axis=1)
+        # add a small local perturbation so MC is imperfect
+        img[40:48, 40:48, :3] = np.clip(
+            img[40:48, 40:48, :3].astype(np.int16) + 12, 0, 255
+        ).astype(np.uint8)
+        r = enc.encode(img.tobytes(), qp=26, idr=(i == 0))
+        stream += r["data"]
+        recons.append(recon_planes(enc, w, h))
+        if i:
+            inter_seen += r["mb_inter"]
+    assert inter_seen > 0
+    frames = Decoder().decode(stream)
+    assert len(frames) == 6
+    for i, (dec, rec) in enumerate(zip(frames, recons)):
+        for d, r, name in zip(dec, rec, "y cb cr".split()):
+            assert np.array_equal(d, r), f"frame {i} plane {name} mismatch"

@@ -33,9 +33,9 @@ class TraceDecoder(R.Decoder):
         self.mbs.append((mbx, mby, "SKIP", {}))
         return super().decode_skip(mbx, mby)
 
-    def decode_p16(self, br, mbx, mby, ctx):
+    def decode_p16(self, br, mbx, mby, ctx, qp):
         self.mbs.append((mbx, mby, "P16", {}))
-        return super().decode_p16(br, mbx, mby, ctx)
+        return super().decode_p16(br, mbx, mby, ctx, qp)
 
 
 def psnr(a, b):
