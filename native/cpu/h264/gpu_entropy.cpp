@@ -118,6 +118,101 @@ void entropy_i16(BitWriter& bw, const int16_t* L, int m0, int m1, int qp,
   }
 }
 
+// Inter column of Table 9-4 (six reachable values only).
+inline int inter_cbp_codenum(int cbp) {
+  switch (cbp) {
+    case 0: return 0;
+    case 16: return 1;
+    case 32: return 6;
+    case 15: return 11;
+    case 47: return 12;
+    default: return 19;  // 31
+  }
+}
+
+// Encode one P_L0_16x16 MB (with residual) from the GPU level buffer.
+// Inter luma blocks are full 16-coeff zigzags (no DC Hadamard); mirrors
+// encoder.cpp::encode_p16's entropy section exactly.
+void entropy_p16(BitWriter& bw, const int16_t* L, int mvx, int mvy,
+                 RowCtx& ctx) {
+  int zz[16][16];
+  bool any_l = false;
+  for (int r = 0; r < 16; ++r)
+    for (int i = 0; i < 16; ++i) {
+      zz[r][i] = L[kLumaAcOff + r * 16 + kZigzag4[i]];
+      any_l |= zz[r][i] != 0;
+    }
+  const int cbp_luma = any_l ? 15 : 0;
+  int cdc[2][4];
+  bool any_cdc = false;
+  for (int comp = 0; comp < 2; ++comp)
+    for (int i = 0; i < 4; ++i) {
+      cdc[comp][i] = L[kChromaDcOff + comp * 4 + i];
+      any_cdc |= cdc[comp][i] != 0;
+    }
+  int czz[8][15];
+  bool any_cac = false;
+  for (int b = 0; b < 8; ++b)
+    for (int i = 1; i < 16; ++i) {
+      czz[b][i - 1] = L[kChromaAcOff + b * 16 + kZigzag4[i]];
+      any_cac |= czz[b][i - 1] != 0;
+    }
+  const int cbp_chroma = any_cac ? 2 : (any_cdc ? 1 : 0);
+  const int cbp = (cbp_chroma << 4) | cbp_luma;
+
+  bw.ue(0);  // P_L0_16x16
+  int mvpx = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvx : 0;
+  int mvpy = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvy : 0;
+  bw.se(mvx - mvpx);
+  bw.se(mvy - mvpy);
+  bw.ue(inter_cbp_codenum(cbp));
+
+  uint8_t new_luma[16] = {};
+  uint8_t new_cb[4] = {}, new_cr[4] = {};
+  if (cbp) {
+    bw.se(0);  // mb_qp_delta
+    if (cbp_luma) {
+      for (int blk = 0; blk < 16; ++blk) {
+        int bx, by;
+        blk_xy(blk, bx, by);
+        int r = by * 4 + bx;
+        int n;
+        if (bx > 0)
+          n = new_luma[by * 4 + bx - 1];
+        else
+          n = ctx.have_left ? ctx.left_luma_nc[by] : 0;
+        int tc = cavlc_residual(bw, zz[r], 16, n);
+        new_luma[r] = static_cast<uint8_t>(tc);
+      }
+    }
+    if (cbp_chroma > 0) {
+      cavlc_residual(bw, cdc[0], 4, -1);
+      cavlc_residual(bw, cdc[1], 4, -1);
+    }
+    if (cbp_chroma == 2) {
+      for (int comp = 0; comp < 2; ++comp) {
+        uint8_t* t = comp ? new_cr : new_cb;
+        const uint8_t* lt = comp ? ctx.left_cr_nc : ctx.left_cb_nc;
+        for (int sub = 0; sub < 4; ++sub) {
+          int cx = sub & 1, cy = sub >> 1;
+          int n = cx > 0 ? t[cy * 2] : (ctx.have_left ? lt[cy] : 0);
+          int tc = cavlc_residual(bw, czz[comp * 4 + sub], 15, n);
+          t[cy * 2 + cx] = static_cast<uint8_t>(tc);
+        }
+      }
+    }
+  }
+  ctx.have_left = true;
+  ctx.left_is_inter = true;
+  ctx.left_mvx = mvx;
+  ctx.left_mvy = mvy;
+  for (int by = 0; by < 4; ++by) ctx.left_luma_nc[by] = new_luma[by * 4 + 3];
+  for (int cy = 0; cy < 2; ++cy) {
+    ctx.left_cb_nc[cy] = new_cb[cy * 2 + 1];
+    ctx.left_cr_nc[cy] = new_cr[cy * 2 + 1];
+  }
+}
+
 }  // namespace
 
 void encode_stripe_from_gpu(const GpuStripeParams& p,
@@ -164,19 +259,7 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
       if (!p.idr && mode == h264gpu::kInter) {
         int mvx = (int16_t)(m1 & 0xFFFF);
         int mvy = m1 >> 16;
-        b.ue(0);  // P_L0_16x16
-        int mvpx = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvx : 0;
-        int mvpy = (ctx.have_left && ctx.left_is_inter) ? ctx.left_mvy : 0;
-        b.se(mvx - mvpx);
-        b.se(mvy - mvpy);
-        b.ue(0);  // cbp = 0
-        ctx.have_left = true;
-        ctx.left_is_inter = true;
-        ctx.left_mvx = mvx;
-        ctx.left_mvy = mvy;
-        std::fill(ctx.left_luma_nc, ctx.left_luma_nc + 4, 0);
-        ctx.left_cb_nc[0] = ctx.left_cb_nc[1] = 0;
-        ctx.left_cr_nc[0] = ctx.left_cr_nc[1] = 0;
+        entropy_p16(b, L, mvx, mvy, ctx);
         continue;
       }
       entropy_i16(b, L, m0, m1, p.qp, !p.idr, ctx);

@@ -190,6 +190,19 @@ __device__ inline int blk_total(const int16_t* L, int off, int first,
   return t;
 }
 
+// coded_block_pattern me(v) codeNum, Inter column of Table 9-4 — only
+// the six values our encoder emits (see cpu/h264/encoder.cpp).
+__device__ inline int dev_inter_cbp_codenum(int cbp) {
+  switch (cbp) {
+    case 0: return 0;
+    case 16: return 1;
+    case 32: return 6;
+    case 15: return 11;
+    case 47: return 12;
+    default: return 19;  // 31
+  }
+}
+
 // Per-MB info, precomputed once per row into LDS:
 //   flags: mode(2) | luma_mode<<2 | chroma_mode<<5 | cbp_luma1<<8 |
 //          cbp_chroma<<9
@@ -215,11 +228,14 @@ __device__ void precompute_mb(const int16_t* levels, const int* meta,
   out->mvx = (short)(m1 & 0xFFFF);
   out->mvy = (short)(m1 >> 16);
   int cbp_luma1 = 0, cbp_chroma = 0;
-  if (mode == kIntra) {
+  if (mode == kIntra || mode == kInter) {
+    // intra: luma totals over the 15 AC coeffs; inter: full 16-coeff
+    // blocks (no DC Hadamard for inter residuals)
+    const int first = (mode == kIntra) ? 1 : 0;
     const int16_t* L = levels + mb_index * kLevelsPerMb;
     int any_ac = 0;
     for (int b = 0; b < 16; ++b) {
-      int t = blk_total(L, kLumaAcOff + b * 16, 1, 15);
+      int t = blk_total(L, kLumaAcOff + b * 16, first, 16 - first);
       out->ltot[b] = (uint8_t)t;
       any_ac |= t;
     }
@@ -250,7 +266,7 @@ __device__ inline int lds_luma_nc(const MbInfo* info, int mbx, int bx,
   }
   if (mbx == 0) return 0;
   const MbInfo& l = info[mbx - 1];
-  if (MB_MODE(l.flags) != kIntra || !MB_CBPL(l.flags)) return 0;
+  if (!MB_CBPL(l.flags)) return 0;  // skip/uncoded neighbors -> 0 totals
   return l.ltot[by * 4 + 3];
 }
 
@@ -262,7 +278,7 @@ __device__ inline int lds_chroma_nc(const MbInfo* info, int mbx, int comp,
   }
   if (mbx == 0) return 0;
   const MbInfo& l = info[mbx - 1];
-  if (MB_MODE(l.flags) != kIntra || MB_CBPC(l.flags) != 2) return 0;
+  if (MB_CBPC(l.flags) != 2) return 0;
   return l.ctot[comp * 4 + cy * 2 + 1];
 }
 
@@ -355,7 +371,9 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
           }
           bw.se(m.mvx - mvpx);
           bw.se(m.mvy - mvpy);
-          bw.ue(0);  // cbp = 0
+          int cbp = (cbp_chroma << 4) | (MB_CBPL(m.flags) ? 15 : 0);
+          bw.ue(dev_inter_cbp_codenum(cbp));
+          if (cbp) bw.se(0);  // mb_qp_delta
         } else {
           int i16 = 1 + MB_LMODE(m.flags) + 4 * cbp_chroma +
                     12 * MB_CBPL(m.flags);
@@ -364,26 +382,35 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
           bw.se(0);  // mb_qp_delta
         }
         bits = bw.flush();
-      } else if (mode != kIntra) {
+      } else if (mode == kSkip) {
         bits = 0;
       } else if (slot == 1) {
-        // luma DC
-        int zz[16];
-        for (int i = 0; i < 16; ++i)
-          zz[i] = L[kLumaDcOff + c_zig4[i]];
-        int nC = lds_luma_nc(s_mb, mb, 0, 0);
-        dev_cavlc_residual<16>(bw, zz, nC);
-        bits = bw.flush();
+        // luma DC (intra only; inter has no DC Hadamard)
+        if (mode == kIntra) {
+          int zz[16];
+          for (int i = 0; i < 16; ++i)
+            zz[i] = L[kLumaDcOff + c_zig4[i]];
+          int nC = lds_luma_nc(s_mb, mb, 0, 0);
+          dev_cavlc_residual<16>(bw, zz, nC);
+          bits = bw.flush();
+        }
       } else if (slot < 18) {
         if (MB_CBPL(m.flags)) {
           int blk = slot - 2;                      // Z-order index
           int r = c_zorder_raster[blk];
           int bx = r & 3, by = r >> 2;
-          int zz[15];
-          for (int i = 1; i < 16; ++i)
-            zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
           int nC = lds_luma_nc(s_mb, mb, bx, by);
-          dev_cavlc_residual<15>(bw, zz, nC);
+          if (mode == kIntra) {
+            int zz[15];
+            for (int i = 1; i < 16; ++i)
+              zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
+            dev_cavlc_residual<15>(bw, zz, nC);
+          } else {
+            int zz[16];
+            for (int i = 0; i < 16; ++i)
+              zz[i] = L[kLumaAcOff + r * 16 + c_zig4[i]];
+            dev_cavlc_residual<16>(bw, zz, nC);
+          }
           bits = bw.flush();
         }
       } else if (slot < 20) {

@@ -46,18 +46,19 @@ __device__ inline int coeff_cls(int pos) {
   return (ei && ej) ? 0 : ((!ei && !ej) ? 1 : 2);
 }
 
-__device__ inline int quant_coeff(int w, int qp, int cls) {
+__device__ inline int quant_coeff(int w, int qp, int cls,
+                                  bool intra = true) {
   int qbits = 15 + qp / 6;
-  int f = (1 << qbits) / 3;  // intra
+  int f = (1 << qbits) / (intra ? 3 : 6);
   int az = abs(w);
   int level = (az * c_quant_mf[qp % 6][cls] + f) >> qbits;
   level = min(level, 2063);
   return w < 0 ? -level : level;
 }
 
-__device__ inline int quant_dc_v(int w, int qp) {
+__device__ inline int quant_dc_v(int w, int qp, bool intra = true) {
   int qbits = 15 + qp / 6;
-  int f = (1 << qbits) / 3;
+  int f = (1 << qbits) / (intra ? 3 : 6);
   int az = abs(w);
   int level = (az * c_quant_mf[qp % 6][0] + 2 * f) >> (qbits + 1);
   level = min(level, 2063);
@@ -250,12 +251,76 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       mvy = m1 >> 16;
     }
 
-    if (mode != kIntra) {
-      int ix = mvx >> 2, iy = mvy >> 2;
-      const uint8_t* s = refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
+    if (mode == kSkip) {
+      const uint8_t* s = refY + (size_t)(y0 + r) * ypitch + x0 + cq;
       uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
       d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
       if (cq == 12) s_left_y[r] = s[3];
+      wave_lds_fence();
+      have_left = true;
+      continue;
+    }
+
+    if (mode == kInter) {
+      // P_L0_16x16 with coded residual: full 16-coeff blocks, no DC
+      // Hadamard, inter quant rounding (f = 2^qbits/6). cbp==0 falls out
+      // naturally (idct of zeros is zero -> recon == MC pred).
+      const int ix = mvx >> 2, iy = mvy >> 2;
+      uint32_t psrc, ppred;
+      {
+        const uint8_t* sr = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+        psrc = (uint32_t)sr[min(x0 + cq + 0, w - 1)] |
+               ((uint32_t)sr[min(x0 + cq + 1, w - 1)] << 8) |
+               ((uint32_t)sr[min(x0 + cq + 2, w - 1)] << 16) |
+               ((uint32_t)sr[min(x0 + cq + 3, w - 1)] << 24);
+        const uint8_t* pr =
+            refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
+        ppred = (uint32_t)pr[0] | ((uint32_t)pr[1] << 8) |
+                ((uint32_t)pr[2] << 16) | ((uint32_t)pr[3] << 24);
+      }
+      auto pix_at = [&](int py, int px) -> int {
+        uint32_t v = __shfl(psrc, py * 4 + (px >> 2));
+        return (v >> (8 * (px & 3))) & 0xFF;
+      };
+      auto pred_at = [&](int py, int px) -> int {
+        uint32_t v = __shfl(ppred, py * 4 + (px >> 2));
+        return (v >> (8 * (px & 3))) & 0xFF;
+      };
+      int lvl_p[4];
+#pragma unroll
+      for (int pass = 0; pass < 4; ++pass) {
+        int blk = pass * 4 + g;
+        int bx = blk & 3, by = blk >> 2;
+        int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
+        int resid = pix_at(py, px) - pred_at(py, px);
+        int coefv = fdct4_wave(resid, lane);
+        int lvl = quant_coeff(coefv, qp, coeff_cls(c), false);
+        lvl = cap12_group(lvl, zz, true, lane);
+        lvl_p[pass] = lvl;
+        store_lvl_pair(L + kLumaAcOff + blk * 16, c, lvl, lane);
+      }
+      int anyl = (lvl_p[0] | lvl_p[1] | lvl_p[2] | lvl_p[3]) != 0;
+      const int cbp_luma = __ballot(anyl) ? 15 : 0;
+#pragma unroll
+      for (int pass = 0; pass < 4; ++pass) {
+        int blk = pass * 4 + g;
+        int bx = blk & 3, by = blk >> 2;
+        int d = cbp_luma ? dequant_c(lvl_p[pass], qp, coeff_cls(c)) : 0;
+        int rec = idct4_wave(d, lane);
+        int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
+        int pix = clip8(rec + pred_at(py, px));
+        int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
+            p3 = __shfl(pix, lane + 3);
+        if ((c & 3) == 0) {
+          *reinterpret_cast<uint32_t*>(
+              curY + (size_t)(y0 + py) * ypitch + x0 + px) =
+              (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
+              ((uint32_t)p3 << 24);
+        }
+        if (px == 15) s_left_y[py + 16] = (uint8_t)pix;
+      }
+      wave_lds_fence();
+      if (lane < 16) s_left_y[lane] = s_left_y[lane + 16];
       wave_lds_fence();
       have_left = true;
       continue;
@@ -395,20 +460,141 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
       mvy = m1 >> 16;
     }
 
-    if (mode != kIntra) {
-      int ix = (mvx >> 2) >> 1, iy = (mvy >> 2) >> 1;
+    if (mode == kSkip) {
       if (lane < 32) {
         int comp = lane >> 4;
         int r = (lane & 15) >> 1, cq = (lane & 1) * 4;
         const uint8_t* sp = comp ? refCr : refCb;
         uint8_t* dp = comp ? curCr : curCb;
-        const uint8_t* s =
-            sp + (size_t)(cy0 + iy + r) * cpitch + cx0 + ix + cq;
+        const uint8_t* s = sp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
         uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
         d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
         if (cq == 4) {
           if (comp) s_left_cr[r] = s[3]; else s_left_cb[r] = s[3];
         }
+      }
+      wave_lds_fence();
+      have_left = true;
+      continue;
+    }
+
+    if (mode == kInter) {
+      // chroma inter residual: MC pred (integer, mv/2) + DC Hadamard + AC,
+      // inter quant rounding. cbp falls out of the quantized levels.
+      const int ix = (mvx >> 2) >> 1, iy = (mvy >> 2) >> 1;
+      uint32_t csrc = 0, cprd = 0;
+      if (lane < 32) {
+        int comp = lane >> 4;
+        int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
+        const uint8_t* sp = comp ? srcCr : srcCb;
+        int sr = min(cy0 + rr, chh - 1);
+        csrc = (uint32_t)sp[(size_t)sr * cpitch + min(cx0 + ccq + 0, cw - 1)] |
+               ((uint32_t)sp[(size_t)sr * cpitch +
+                             min(cx0 + ccq + 1, cw - 1)] << 8) |
+               ((uint32_t)sp[(size_t)sr * cpitch +
+                             min(cx0 + ccq + 2, cw - 1)] << 16) |
+               ((uint32_t)sp[(size_t)sr * cpitch +
+                             min(cx0 + ccq + 3, cw - 1)] << 24);
+        const uint8_t* rp = (comp ? refCr : refCb) +
+                            (size_t)(cy0 + iy + rr) * cpitch + cx0 + ix + ccq;
+        cprd = (uint32_t)rp[0] | ((uint32_t)rp[1] << 8) |
+               ((uint32_t)rp[2] << 16) | ((uint32_t)rp[3] << 24);
+      }
+      auto cpix_at = [&](int comp, int rr, int cc) -> int {
+        uint32_t v = __shfl(csrc, (comp << 4) | (rr << 1) | (cc >> 2));
+        return (v >> (8 * (cc & 3))) & 0xFF;
+      };
+      auto cprd_at = [&](int comp, int rr, int cc) -> int {
+        uint32_t v = __shfl(cprd, (comp << 4) | (rr << 1) | (cc >> 2));
+        return (v >> (8 * (cc & 3))) & 0xFF;
+      };
+      int lvl_p[2];
+      int dcpass[2];
+#pragma unroll
+      for (int pass = 0; pass < 2; ++pass) {
+        int comp = pass;
+        int sub = g;
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int rr = scy + (c >> 2), cc2 = scx + (c & 3);
+        int resid = cpix_at(comp, rr, cc2) - cprd_at(comp, rr, cc2);
+        int coefv = fdct4_wave(resid, lane);
+        dcpass[pass] = __shfl(coefv, (lane & 3) * 16);
+        int lvl = (c == 0) ? 0 : quant_coeff(coefv, qpc, coeff_cls(c),
+                                             false);
+        lvl = cap12_group(lvl, zz, c != 0, lane);
+        lvl_p[pass] = lvl;
+        store_lvl_pair(L + kChromaAcOff + (comp * 4 + sub) * 16, c, lvl,
+                       lane);
+      }
+      int qdc_all[8], dcr_all[8];
+      bool any_cdc = false;
+#pragma unroll
+      for (int comp = 0; comp < 2; ++comp) {
+        int d0 = __shfl(dcpass[comp], 0);
+        int d1 = __shfl(dcpass[comp], 1);
+        int d2 = __shfl(dcpass[comp], 2);
+        int d3 = __shfl(dcpass[comp], 3);
+        int w0 = d0 + d1 + d2 + d3, w1 = d0 - d1 + d2 - d3;
+        int w2 = d0 + d1 - d2 - d3, w3 = d0 - d1 - d2 + d3;
+        int q0 = quant_dc_v(w0, qpc, false), q1 = quant_dc_v(w1, qpc, false);
+        int q2 = quant_dc_v(w2, qpc, false), q3 = quant_dc_v(w3, qpc, false);
+        qdc_all[comp * 4 + 0] = q0;
+        qdc_all[comp * 4 + 1] = q1;
+        qdc_all[comp * 4 + 2] = q2;
+        qdc_all[comp * 4 + 3] = q3;
+        any_cdc |= (q0 | q1 | q2 | q3) != 0;
+      }
+      if (lane < 8)
+        L[kChromaDcOff + lane] = (int16_t)qdc_all[lane];
+      int anyc = (lvl_p[0] | lvl_p[1]) != 0 && c != 0;
+      const int cbp_chroma = __ballot(anyc) ? 2 : (any_cdc ? 1 : 0);
+#pragma unroll
+      for (int comp = 0; comp < 2; ++comp) {
+        int dq0 = 0, dq1 = 0, dq2 = 0, dq3 = 0;
+        if (cbp_chroma >= 1) {
+          int* q = &qdc_all[comp * 4];
+          int w0 = q[0] + q[1] + q[2] + q[3], w1 = q[0] - q[1] + q[2] - q[3];
+          int w2 = q[0] + q[1] - q[2] - q[3], w3 = q[0] - q[1] - q[2] + q[3];
+          dq0 = dequant_chroma_dc_v(w0, qpc);
+          dq1 = dequant_chroma_dc_v(w1, qpc);
+          dq2 = dequant_chroma_dc_v(w2, qpc);
+          dq3 = dequant_chroma_dc_v(w3, qpc);
+        }
+        dcr_all[comp * 4 + 0] = dq0;
+        dcr_all[comp * 4 + 1] = dq1;
+        dcr_all[comp * 4 + 2] = dq2;
+        dcr_all[comp * 4 + 3] = dq3;
+      }
+#pragma unroll
+      for (int pass = 0; pass < 2; ++pass) {
+        int comp = pass;
+        int sub = g;
+        int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
+        int rr = scy + (c >> 2), cc2 = scx + (c & 3);
+        int d = (c == 0) ? dcr_all[comp * 4 + sub]
+                         : (cbp_chroma == 2
+                                ? dequant_c(lvl_p[pass], qpc, coeff_cls(c))
+                                : 0);
+        int rec = idct4_wave(d, lane);
+        int pix = clip8(rec + cprd_at(comp, rr, cc2));
+        uint8_t* dp = comp ? curCr : curCb;
+        int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
+            p3 = __shfl(pix, lane + 3);
+        if ((c & 3) == 0) {
+          *reinterpret_cast<uint32_t*>(
+              dp + (size_t)(cy0 + rr) * cpitch + cx0 + cc2) =
+              (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
+              ((uint32_t)p3 << 24);
+        }
+        if (cc2 == 7) {
+          if (comp) s_left_cr[rr + 8] = (uint8_t)pix;
+          else s_left_cb[rr + 8] = (uint8_t)pix;
+        }
+      }
+      wave_lds_fence();
+      if (lane < 8) {
+        s_left_cb[lane] = s_left_cb[lane + 8];
+        s_left_cr[lane] = s_left_cr[lane + 8];
       }
       wave_lds_fence();
       have_left = true;
@@ -646,7 +832,8 @@ __global__ void __launch_bounds__(64) k_h264_me(
 
   const int qp = job.qp;
   const int skip_thresh = 48 << (qp / 6);
-  const int inter_thresh = 2 * skip_thresh;
+  // residual coding makes inter viable whenever MC is a decent predictor
+  const int inter_thresh = 6 * skip_thresh;
 
   int sad0 = sad_at(0, 0);
   int mode, bmx = 0, bmy = 0;
@@ -675,8 +862,8 @@ __global__ void __launch_bounds__(64) k_h264_me(
       }
       if (!improved) break;
     }
-    if (best <= inter_thresh && !(bmx == 0 && bmy == 0))
-      mode = kInter;
+    if (best <= inter_thresh)
+      mode = kInter;   // mv may be (0,0): residuals carry the change
     else
       mode = kIntra;
   }
@@ -848,7 +1035,7 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   };
   const int qp = job.qp;
   const int skip_thresh = 48 << (qp / 6);
-  const int inter_thresh = 2 * skip_thresh;
+  const int inter_thresh = 6 * skip_thresh;
   int sad0 = sad_at(0, 0);
   int mode, omvx = 0, omvy = 0;
   if (sad0 <= skip_thresh) {
@@ -856,8 +1043,8 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   } else {
     int best_sad = (best_mvx || best_mvy) ? sad_at(best_mvx, best_mvy)
                                           : sad0;
-    if (best_sad <= inter_thresh && !(best_mvx == 0 && best_mvy == 0)) {
-      mode = kInter;
+    if (best_sad <= inter_thresh) {
+      mode = kInter;   // mv may be (0,0): residuals carry the change
       omvx = best_mvx;
       omvy = best_mvy;
     } else {
