@@ -39,6 +39,10 @@ class CentralizedStreamServer:
             settings.upload_dir,
             allow_upload="upload" in dirs,
             allow_download="download" in dirs)
+        from .transfers import UplinkAllowance
+        self.allowance = UplinkAllowance(self.transfers.pacer)
+        self._uplink_task = None
+        self._last_video_bytes = 0
         self.app = web.Application(middlewares=[self._auth_middleware])
         self.webrtc = None
         self.started_at = time.time()
@@ -48,6 +52,23 @@ class CentralizedStreamServer:
         self._register_routes()
 
     # ---- routes -------------------------------------------------------------
+    async def _uplink_loop(self, interval: float = 2.0):
+        """Sample video throughput + client ACK RTT; adapt the transfer
+        pacer so file transfers only use spare uplink (UplinkAllowance)."""
+        while True:
+            await asyncio.sleep(interval)
+            try:
+                clients = list(self.streaming.clients.values())
+                total = sum(c.relay.sent_bytes for c in clients)
+                video_bps = max(0.0,
+                                (total - self._last_video_bytes) / interval)
+                self._last_video_bytes = total
+                rtts = [c.ack_rtt_ms for c in clients if c.ack_rtt_ms > 0]
+                self.allowance.observe(video_bps,
+                                       max(rtts) if rtts else None)
+            except Exception:
+                logger.debug("uplink sampling failed", exc_info=True)
+
     def _register_routes(self):
         app = self.app
         app.router.add_get("/websockets", self._ws_entry)
@@ -276,9 +297,13 @@ class CentralizedStreamServer:
         await site.start()
         if self._ssl_ctx is not None:
             asyncio.get_running_loop().create_task(self._watch_certs())
+        self._uplink_task = asyncio.get_running_loop().create_task(
+            self._uplink_loop())
         logger.info("serving on %s:%s (mode=%s)", s.addr, s.port, s.mode)
 
     async def stop(self):
+        if self._uplink_task is not None:
+            self._uplink_task.cancel()
         self.streaming.stop_capture()
         self.streaming.stop_audio()
         if self.webrtc is not None:

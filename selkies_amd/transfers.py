@@ -46,6 +46,53 @@ class TransferPacer:
                 await asyncio.sleep(min(0.25, deficit / self.rate))
 
 
+class UplinkAllowance:
+    """Adaptive transfer budget that protects the media stream.
+
+    Mirrors the reference's upload-pacing intent (transfers
+    must never starve video): AIMD on the shared :class:`TransferPacer`
+    rate, driven by the observed video send rate and the clients' frame-ACK
+    RTT. When RTT inflates above its rolling floor while transfers run, the
+    link is saturated -> multiplicative back-off; while RTT stays near the
+    floor the allowance creeps back up additively.
+
+    Pure-python and clock-injectable for unit tests.
+    """
+
+    RTT_INFLATE = 1.8          # x floor considered congested
+    BACKOFF = 0.5
+    CREEP_BYTES = 1_000_000    # +1 MB/s per healthy observation
+
+    def __init__(self, pacer: TransferPacer,
+                 floor_rate: float = 256_000.0,
+                 cap_rate: float = 100_000_000.0):
+        self.pacer = pacer
+        self.floor_rate = floor_rate
+        self.cap_rate = cap_rate
+        self.rate = pacer.rate
+        self._rtt_floor: Optional[float] = None
+
+    def observe(self, video_bps: float, rtt_ms: Optional[float]) -> float:
+        """Feed one sampling interval; returns the new transfer rate."""
+        if rtt_ms is not None and rtt_ms > 0:
+            self._rtt_floor = (rtt_ms if self._rtt_floor is None
+                               else min(self._rtt_floor * 1.02, rtt_ms))
+        congested = (rtt_ms is not None and self._rtt_floor is not None
+                     and rtt_ms > self._rtt_floor * self.RTT_INFLATE)
+        if congested:
+            self.rate = max(self.floor_rate, self.rate * self.BACKOFF)
+        else:
+            self.rate = min(self.cap_rate, self.rate + self.CREEP_BYTES)
+        # regardless of probing, never budget into the video's share:
+        # assume the link is at least video + current allowance when healthy
+        if video_bps > 0:
+            headroom_cap = max(self.floor_rate,
+                               (video_bps + self.rate) * 0.35)
+            self.rate = min(self.rate, headroom_cap)
+        self.pacer.set_rate(self.rate)
+        return self.rate
+
+
 class TransferManager:
     PART_TTL_S = 3600          # stale staging parts reaped after 1 h
     CHUNK = 256 * 1024

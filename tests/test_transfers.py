@@ -105,3 +105,31 @@ def test_http_endpoints(tmp_path, loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_uplink_allowance_aimd():
+    """RTT inflation while transferring backs the rate off multiplicatively;
+    healthy intervals creep it back up; floor/cap clamped; video share
+    bounds the budget."""
+    from selkies_amd.transfers import TransferPacer, UplinkAllowance
+    pacer = TransferPacer(rate_bytes_per_s=10_000_000)
+    a = UplinkAllowance(pacer)
+    # healthy: establishes the RTT floor and creeps up
+    r1 = a.observe(video_bps=0, rtt_ms=20)
+    r2 = a.observe(video_bps=0, rtt_ms=21)
+    assert r2 >= r1
+    # congestion: rtt >> floor -> multiplicative backoff, pacer follows
+    r3 = a.observe(video_bps=0, rtt_ms=200)
+    assert r3 <= r2 * 0.6
+    assert pacer.rate == r3
+    # repeated congestion floors out
+    for _ in range(30):
+        r = a.observe(video_bps=0, rtt_ms=500)
+    assert r == a.floor_rate
+    # recovery is gradual (additive)
+    r4 = a.observe(video_bps=0, rtt_ms=20)
+    assert r4 <= a.floor_rate + a.CREEP_BYTES
+    # a heavy video stream caps the transfer share
+    a2 = UplinkAllowance(TransferPacer(rate_bytes_per_s=50_000_000))
+    r5 = a2.observe(video_bps=8_000_000, rtt_ms=20)
+    assert r5 <= (8_000_000 + 50_000_000 + a2.CREEP_BYTES) * 0.35 + 1
