@@ -187,6 +187,105 @@ def set_dpi(dpi: int, display: Optional[str] = None) -> bool:
         return False
 
 
+def cursor_size_for_dpi(dpi: float, base_size: int = 24) -> int:
+    """Scale the cursor theme size with DPI (reference
+    cursor_size_for_dpi, display_utils.py:553)."""
+    return max(1, int(round(float(dpi) / 96.0 * base_size)))
+
+
+def monitor_geometry(w: int, h: int, x: int, y: int,
+                     dpi: float = 96.0) -> str:
+    """RandR --setmonitor geometry `w/mmWxh/mmH+x+y` with physical mm
+    derived from the session DPI (reference builds mm the same way,
+    display_utils.py:352-358)."""
+    mm_w = max(1, round(w * 25.4 / dpi))
+    mm_h = max(1, round(h * 25.4 / dpi))
+    return f"{w}/{mm_w}x{h}/{mm_h}+{x}+{y}"
+
+
+def set_logical_monitor(name: str, w: int, h: int, x: int, y: int,
+                        output: str = "none", dpi: float = 96.0,
+                        display: Optional[str] = None) -> bool:
+    """Create/replace a RandR 1.5 logical monitor (reference
+    display_utils.py:1001-1010: xrandr --setmonitor fallback path). Used
+    to present dual logical displays on a single large framebuffer."""
+    geom = monitor_geometry(w, h, x, y, dpi)
+    return _run(["xrandr", "--setmonitor", name, geom, output],
+                display) is not None
+
+
+def delete_logical_monitor(name: str,
+                           display: Optional[str] = None) -> bool:
+    """Remove a logical monitor (reference display_utils.py:1022)."""
+    return _run(["xrandr", "--delmonitor", name], display) is not None
+
+
+def apply_logical_dual_layout(w1: int, h1: int, w2: int, h2: int,
+                              orientation: str = "right",
+                              dpi: float = 96.0,
+                              display: Optional[str] = None) -> bool:
+    """Expose the dual-display layout (compute_dual_layout) as two RandR
+    logical monitors on the shared framebuffer."""
+    m1, m2 = compute_dual_layout(w1, h1, w2, h2, orientation)
+    fb_w, fb_h = framebuffer_bounds([m1, m2])
+    if _run(["xrandr", "--fb", f"{fb_w}x{fb_h}"], display) is None:
+        return False
+    ok1 = set_logical_monitor("selkies-0", m1.width, m1.height, m1.x, m1.y,
+                              dpi=dpi, display=display)
+    ok2 = set_logical_monitor("selkies-1", m2.width, m2.height, m2.x, m2.y,
+                              dpi=dpi, display=display)
+    return ok1 and ok2
+
+
+def _persist_xresources_dpi(dpi: int, path: Optional[str] = None) -> bool:
+    """Rewrite only the Xft.dpi resource in ~/.Xresources (reference
+    _write_xresources_dpi, display_utils.py:1574-1590)."""
+    path = path or os.path.expanduser("~/.Xresources")
+    try:
+        lines: list[str] = []
+        if os.path.exists(path):
+            with open(path, "r", encoding="utf-8", errors="replace") as f:
+                lines = [ln for ln in f.read().splitlines()
+                         if not re.match(r"^\s*Xft\.dpi\s*:", ln)]
+        lines.append(f"Xft.dpi:   {dpi}")
+        tmp = path + ".tmp"
+        with open(tmp, "w", encoding="utf-8") as f:
+            f.write("\n".join(lines) + "\n")
+        os.replace(tmp, path)
+        return True
+    except OSError:
+        return False
+
+
+def apply_dpi_ladder(dpi: int, display: Optional[str] = None,
+                     xresources_path: Optional[str] = None) -> dict:
+    """Apply DPI through every available desktop-environment channel
+    (reference ladder: xrdb live merge, Xresources persistence,
+    xfconf-query for XFCE, gsettings for MATE/GNOME — display_utils.py:
+    1767-1891). Returns which channels were applied."""
+    applied = {"xrdb": set_dpi(dpi, display),
+               "xresources": _persist_xresources_dpi(dpi, xresources_path)}
+    cursor = cursor_size_for_dpi(dpi)
+    if shutil.which("xfconf-query"):
+        applied["xfconf"] = (
+            _run(["xfconf-query", "-c", "xsettings", "-p", "/Xft/DPI",
+                  "-n", "-t", "int", "-s", str(dpi)], display) is not None)
+        _run(["xfconf-query", "-c", "xsettings", "-p",
+              "/Gtk/CursorThemeSize", "-n", "-t", "int", "-s",
+              str(cursor)], display)
+    if shutil.which("gsettings"):
+        scale = dpi / 96.0
+        whole = int(scale) if float(scale).is_integer() else 1
+        text_scale = scale / whole
+        applied["gsettings"] = (
+            _run(["gsettings", "set", "org.gnome.desktop.interface",
+                  "text-scaling-factor", f"{text_scale:.4f}"],
+                 display) is not None)
+        _run(["gsettings", "set", "org.gnome.desktop.interface",
+              "cursor-size", str(cursor)], display)
+    return applied
+
+
 def resize_entrypoint(argv=None) -> int:
     """`selkies-resize WxH` CLI (reference entrypoint,
     display_utils.py:2200)."""

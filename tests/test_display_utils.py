@@ -46,3 +46,47 @@ def test_parse_dri_node():
     assert D.parse_dri_node_to_index("/dev/dri/renderD128") == 0
     assert D.parse_dri_node_to_index("/dev/dri/renderD129") == 1
     assert D.parse_dri_node_to_index("bogus") == -1
+
+
+def test_monitor_geometry_and_cursor_scale():
+    from selkies_amd import display_utils as du
+    # 1920x1080 at 96 dpi -> 508x286 mm
+    assert du.monitor_geometry(1920, 1080, 0, 0, 96) == \
+        "1920/508x1080/286+0+0"
+    g = du.monitor_geometry(2560, 1440, 1920, 0, 192)
+    assert g.endswith("+1920+0")
+    assert "2560/339" in g
+    assert du.cursor_size_for_dpi(96) == 24
+    assert du.cursor_size_for_dpi(192) == 48
+    assert du.cursor_size_for_dpi(120, 32) == 40
+
+
+def test_logical_monitor_commands(monkeypatch):
+    from selkies_amd import display_utils as du
+    calls = []
+
+    def fake_run(cmd, display):
+        calls.append(cmd)
+        return ""
+    monkeypatch.setattr(du, "_run", fake_run)
+    assert du.set_logical_monitor("selkies-0", 1920, 1080, 0, 0)
+    assert calls[-1][:3] == ["xrandr", "--setmonitor", "selkies-0"]
+    assert du.delete_logical_monitor("selkies-0")
+    assert calls[-1] == ["xrandr", "--delmonitor", "selkies-0"]
+    assert du.apply_logical_dual_layout(1920, 1080, 1280, 1024)
+    fbs = [c for c in calls if c[1] == "--fb"]
+    assert fbs and fbs[-1][2] == "3200x1072"  # heights 16-aligned (1080->1072)
+    mons = [c for c in calls if c[1] == "--setmonitor"]
+    assert len(mons) == 3
+
+
+def test_xresources_dpi_persist(tmp_path):
+    from selkies_amd import display_utils as du
+    p = tmp_path / "Xresources"
+    p.write_text("Xft.antialias: 1\nXft.dpi: 96\nXcursor.size: 24\n")
+    assert du._persist_xresources_dpi(144, str(p))
+    txt = p.read_text()
+    assert "Xft.dpi:   144" in txt
+    assert txt.count("Xft.dpi") == 1
+    assert "Xft.antialias: 1" in txt
+    assert "Xcursor.size: 24" in txt
