@@ -140,9 +140,10 @@ def test_gpu_p_chain_no_drift():
 
 def test_gpu_mfma_motion_search_exact():
     """Frame 1 = frame 0 rolled by (dx=4, dy=-2). The MFMA full-search ME
-    must find mv=(-4,+2) for interior MBs; with cbp=0 inter coding the
-    decoded P frame then equals the rolled decoded IDR EXACTLY wherever
-    inter/skip applies. Any MFMA layout/scoring bug breaks this."""
+    must find mv=(-4,+2) for interior MBs — verified by parsing the coded
+    MVs straight out of the bitstream. (Since inter MBs code residuals,
+    the decoded P frame is no longer byte-equal to the rolled IDR; the MV
+    check is the direct probe of the MFMA scoring/layout.)"""
     require_gpu()
     from scipy.ndimage import uniform_filter
     w, h = 256, 128
@@ -159,30 +160,46 @@ def test_gpu_mfma_motion_search_exact():
         "gpu", [np.ascontiguousarray(f0), np.ascontiguousarray(f1)],
         w, h, 30, 64, 1)
     rows = reassemble(out)
-    dec = decode_rows(rows, w, 64)
-    ys = sorted(dec)
-    d0 = np.concatenate([dec[y][0][0] for y in ys], 0)
-    d1 = np.concatenate([dec[y][1][0] for y in ys], 0)
-    expected = np.roll(np.roll(d0, dy, axis=0), dx, axis=1)
-    # valid MBs: not the first MB column (mv x out of frame) and not the
-    # last MB row of each stripe (mv y out of the stripe)
-    match = 0
-    total = 0
-    for mby in range(h // 16):
-        if (mby * 16) % 64 == 48:
-            continue
-        for mbx in range(1, w // 16):
-            a = d1[mby * 16:mby * 16 + 16, mbx * 16:mbx * 16 + 16]
-            b = expected[mby * 16:mby * 16 + 16, mbx * 16:mbx * 16 + 16]
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = {}
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs[(mbx, mby)] = tuple(ctx["left_mv"])
+
+        def decode_skip(self, mbx, mby):
+            super().decode_skip(mbx, mby)
+            self.mvs[(mbx, mby)] = (0, 0)
+
+    want = (dx * -4, dy * -4)  # quarter-pel units: (-16, +8)
+    match = total = 0
+    psnrs = []
+    for y0, stream in rows.items():
+        d = MvDecoder()
+        frames = d.decode(bytes(stream))
+        assert len(frames) == 2
+        # interior MBs of this stripe: skip first column (mv x leaves the
+        # frame) and the last MB row (mv y leaves the stripe)
+        stripe_mbh = frames[0][0].shape[0] // 16
+        for (mbx, mby), mv in d.mvs.items():
+            if mbx == 0 or mby == stripe_mbh - 1:
+                continue
             total += 1
-            match += int(np.array_equal(a, b))
+            match += int(mv == want)
+        # decoded P must track the rolled source closely (MC + residual)
+        sy = np.roll(np.roll(smooth, dy, axis=0), dx, axis=1)
+        psnrs.append(psnr(frames[1][0],
+                          sy[y0:y0 + frames[1][0].shape[0]]))
     assert total > 0
     frac = match / total
-    assert frac > 0.9, f"only {match}/{total} MBs motion-matched exactly"
-    # and the P frame must be far smaller than the IDR (pure MC copies)
-    # P frame is mostly MC copies; border MBs legitimately go intra
+    assert frac > 0.9, f"only {match}/{total} interior MBs found {want}"
+    assert min(psnrs) > 32, f"P-frame quality too low: {psnrs}"
+    # and the P frame must be far smaller than the IDR
     sizes = [sum(len(t[0]) for t in fr) for fr in out]
-    assert sizes[1] < sizes[0] * 0.5, f"P frame too large: {sizes}"
+    assert sizes[1] < sizes[0] * 0.6, f"P frame too large: {sizes}"
 
 
 def test_gpu_cavlc_matches_cpu_entropy():
