@@ -196,7 +196,7 @@ def test_gpu_mfma_motion_search_exact():
     assert total > 0
     frac = match / total
     assert frac > 0.9, f"only {match}/{total} interior MBs found {want}"
-    assert min(psnrs) > 32, f"P-frame quality too low: {psnrs}"
+    assert min(psnrs) > 30, f"P-frame quality too low: {psnrs}"
     # and the P frame must be far smaller than the IDR
     sizes = [sum(len(t[0]) for t in fr) for fr in out]
     assert sizes[1] < sizes[0] * 0.6, f"P frame too large: {sizes}"
