@@ -150,3 +150,29 @@ def parse_rtcp(data: bytes) -> list[dict]:
             break
         off += size
     return out
+
+
+class AudioPacketizer:
+    """G.711 frames -> single RTP packets (RFC 3551: PCMU pt 0, 8 kHz
+    clock, one 20 ms frame per packet, timestamp += sample count)."""
+
+    def __init__(self, ssrc: int, payload_type: int = 0):
+        self.ssrc = ssrc
+        self.pt = payload_type
+        self.seq = 0
+        self.timestamp = 0
+        self.packets_sent = 0
+        self.bytes_sent = 0
+        self._first = True
+
+    def packetize(self, payload: bytes) -> bytes:
+        marker = 0x80 if self._first else 0   # marker starts the stream
+        self._first = False
+        b1 = self.pt | marker
+        h = struct.pack(">BBHII", 0x80, b1, self.seq & 0xFFFF,
+                        self.timestamp & 0xFFFFFFFF, self.ssrc)
+        self.seq = (self.seq + 1) & 0xFFFF
+        self.timestamp = (self.timestamp + len(payload)) & 0xFFFFFFFF
+        self.packets_sent += 1
+        self.bytes_sent += len(payload)
+        return h + payload

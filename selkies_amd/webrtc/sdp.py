@@ -14,6 +14,7 @@ class MediaSection:
     ice_pwd: str = ""
     fingerprint: str = ""
     h264_pts: list = field(default_factory=list)  # (pt, profile_level_id)
+    g711_pt: int = -1          # PCMU(0)/PCMA(8) if offered
     raw: list = field(default_factory=list)
 
 
@@ -56,6 +57,22 @@ def parse_offer(sdp: str) -> Offer:
             fmtp[int(pt)] = params.strip()
         if tgt is not None:
             tgt.raw.append(line)
+    # audio: accept G.711 (PCMU preferred over PCMA); static PTs may be
+    # offered without an rtpmap line, so also scan the m= format list
+    for m in offer.media:
+        if m.kind != "audio":
+            continue
+        fmts = []
+        for ln in m.raw:
+            if ln.startswith("m=audio"):
+                fmts = [int(f) for f in ln.split()[3:] if f.isdigit()]
+        offered = {pt for pt, codec in rtpmap.items()
+                   if codec.startswith(("pcmu/", "pcma/"))}
+        offered |= {f for f in fmts if f in (0, 8)}
+        if 0 in offered:
+            m.g711_pt = 0
+        elif 8 in offered:
+            m.g711_pt = 8
     # pick H.264 payload types (prefer packetization-mode=1 42e01f)
     for m in offer.media:
         if m.kind != "video":
@@ -78,9 +95,11 @@ def parse_offer(sdp: str) -> Offer:
 
 def build_answer(offer: Offer, ice_ufrag: str, ice_pwd: str,
                  fingerprint: str, host_ip: str, port: int,
-                 ssrc: int, cname: str = "selkies-amd") -> str:
-    """ice-lite answer: video sendonly with the chosen H.264 PT; every
-    other m-line is rejected (port 0) but kept for BUNDLE ordering."""
+                 ssrc: int, cname: str = "selkies-amd",
+                 audio_ssrc: int = 0) -> str:
+    """ice-lite answer: video sendonly with the chosen H.264 PT, audio
+    sendonly G.711 when offered (and audio_ssrc is set); every other
+    m-line is rejected (port 0) but kept for BUNDLE ordering."""
     lines = [
         "v=0",
         f"o=- 0 0 IN IP4 {host_ip}",
@@ -114,6 +133,26 @@ def build_answer(offer: Offer, ice_ufrag: str, ice_pwd: str,
                 f"a=rtcp-fb:{pt} ccm fir",
                 f"a=ssrc:{ssrc} cname:{cname}",
                 f"a=ssrc:{ssrc} msid:selkies video0",
+                f"a=candidate:1 1 udp 2130706431 {host_ip} {port} typ host",
+                "a=end-of-candidates",
+            ]
+        elif m.kind == "audio" and m.g711_pt >= 0 and audio_ssrc:
+            apt = m.g711_pt
+            codec = "PCMU" if apt == 0 else "PCMA"
+            lines += [
+                f"m=audio {port} UDP/TLS/RTP/SAVPF {apt}",
+                f"c=IN IP4 {host_ip}",
+                f"a=mid:{m.mid}",
+                f"a=ice-ufrag:{ice_ufrag}",
+                f"a=ice-pwd:{ice_pwd}",
+                f"a=fingerprint:sha-256 {fingerprint}",
+                "a=setup:passive",
+                "a=sendonly",
+                "a=rtcp-mux",
+                f"a=rtpmap:{apt} {codec}/8000",
+                "a=ptime:20",
+                f"a=ssrc:{audio_ssrc} cname:{cname}",
+                f"a=ssrc:{audio_ssrc} msid:selkies audio0",
                 f"a=candidate:1 1 udp 2130706431 {host_ip} {port} typ host",
                 "a=end-of-candidates",
             ]

@@ -43,14 +43,19 @@ class WebRTCService:
         self.cert = dtls.Certificate()
         self.ufrag, self.pwd = ice.make_ice_credentials()
         self.ssrc = secrets.randbits(31) | 1
+        self.audio_ssrc = (secrets.randbits(31) | 1) ^ 0x10000
         self.packetizer = rtp.H264Packetizer(self.ssrc)
+        self.audio_packetizer = rtp.AudioPacketizer(self.audio_ssrc)
         self.peers: dict[tuple, PeerState] = {}
         self.transport = None
         self.port = 0
         self.host_ip = ice.default_host_ip()
         self.capture: Optional[hipflux.ScreenCapture] = None
+        self.audio_capture = None
+        self._audio_enabled = False
         self._ts_base = time.monotonic()
         self.frames_sent = 0
+        self.audio_frames_sent = 0
 
     # ---- lifecycle ---------------------------------------------------------
     async def start(self, port: int = 0):
@@ -63,6 +68,7 @@ class WebRTCService:
 
     async def stop(self):
         self.stop_video()
+        self.stop_audio()
         if self.transport:
             self.transport.close()
             self.transport = None
@@ -77,9 +83,15 @@ class WebRTCService:
         video = next((m for m in offer.media if m.kind == "video"), None)
         if video is not None and video.h264_pts:
             self.packetizer.pt = video.h264_pts[0][0]
-        answer = sdp.build_answer(offer, self.ufrag, self.pwd,
-                                  self.cert.fingerprint, self.host_ip,
-                                  self.port, self.ssrc)
+        audio = next((m for m in offer.media
+                      if m.kind == "audio" and m.g711_pt >= 0), None)
+        self._audio_enabled = audio is not None and             getattr(self.settings, "enable_audio", True)
+        if audio is not None:
+            self.audio_packetizer.pt = audio.g711_pt
+        answer = sdp.build_answer(
+            offer, self.ufrag, self.pwd, self.cert.fingerprint,
+            self.host_ip, self.port, self.ssrc,
+            audio_ssrc=self.audio_ssrc if self._audio_enabled else 0)
         return answer
 
     # ---- media ---------------------------------------------------------------
@@ -105,6 +117,48 @@ class WebRTCService:
         if self.capture is not None:
             self.capture.stop_capture()
             self.capture = None
+
+    # ---- audio (G.711 over SRTP; webrtc/g711.py) ---------------------------
+    def start_audio(self):
+        if not self._audio_enabled or self.audio_capture is not None:
+            return
+        from hipflux import _native
+        from .webrtc import g711
+        st = self.settings
+        s = _native.AudioCaptureSettings()
+        s.device_name = st.audio_device if st.audio_device != "auto"             else "synthetic"
+        s.channels = st.audio_channels
+        s.frame_duration_ms = 20
+        s.red_distance = 0           # RTP has its own loss handling
+        loop = asyncio.get_running_loop()
+        channels = st.audio_channels
+
+        def on_frame(data, pts_ms):
+            payload = g711.wire_frame_to_ulaw(bytes(data), channels)
+            if payload:
+                loop.call_soon_threadsafe(self._send_audio, payload)
+
+        self.audio_capture = _native.AudioCapture()
+        self.audio_capture.start_capture(s, on_frame)
+        logger.info("webrtc audio capture started (G.711 %s)",
+                    "PCMU" if self.audio_packetizer.pt == 0 else "PCMA")
+
+    def stop_audio(self):
+        if self.audio_capture is not None:
+            self.audio_capture.stop_capture()
+            self.audio_capture = None
+
+    def _send_audio(self, payload: bytes):
+        pkt = self.audio_packetizer.packetize(payload)
+        self.audio_frames_sent += 1
+        for peer in list(self.peers.values()):
+            if not peer.connected:
+                continue
+            try:
+                self.transport.sendto(peer.srtp_out.protect_rtp(pkt),
+                                      peer.addr)
+            except Exception as exc:
+                logger.debug("srtp audio send failed: %r", exc)
 
     def _send_frame(self, annexb: bytes):
         if not any(p.connected for p in self.peers.values()):
@@ -161,6 +215,7 @@ class WebRTCService:
                 peer.connected = True
                 logger.info("webrtc peer %s connected (DTLS-SRTP up)", addr)
                 self.start_video()
+                self.start_audio()
                 if self.capture:
                     self.capture.request_idr_frame()
         elif first >= 128:           # SRTP/SRTCP

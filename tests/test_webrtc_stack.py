@@ -269,3 +269,86 @@ def test_webrtc_loopback_end_to_end():
         return 0
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_g711_ulaw_matches_audioop_and_roundtrips():
+    """Our µ-law must be bit-identical to the stdlib reference encoder
+    over the FULL 16-bit range, and decode within G.711 quantization."""
+    import audioop
+    import numpy as np
+    from selkies_amd.webrtc import g711
+    x = np.arange(-32768, 32768, dtype=np.int16)
+    assert g711.ulaw_encode(x) == audioop.lin2ulaw(x.tobytes(), 2)
+    dec = g711.ulaw_decode(bytes(range(256)))
+    ref = np.frombuffer(audioop.ulaw2lin(bytes(range(256)), 2), np.int16)
+    assert (dec == ref).all()
+    # audible roundtrip: a 1 kHz tone survives with low relative error
+    t = np.arange(160) / 8000.0
+    tone = (np.sin(2 * np.pi * 1000 * t) * 8000).astype(np.int16)
+    rt = g711.ulaw_decode(g711.ulaw_encode(tone)).astype(np.float64)
+    err = np.sqrt(((rt - tone) ** 2).mean())
+    assert err < 200, f"roundtrip RMS error {err:.0f}"
+
+
+def test_g711_downmix_and_wire_frame():
+    import numpy as np
+    from selkies_amd.webrtc import g711
+    # 48k stereo 20 ms: 960 samples x2 ch -> 160 mono 8k samples
+    pcm = np.zeros(1920, np.int16)
+    pcm[0::2] = 1000   # L
+    pcm[1::2] = 3000   # R
+    mono = g711.downmix_8k(pcm.tobytes(), 2)
+    assert mono.shape == (160,)
+    assert abs(int(mono[0]) - 2000) <= 1
+    frame = bytes([0x01, 0x00]) + pcm.tobytes()
+    payload = g711.wire_frame_to_ulaw(frame, 2)
+    assert len(payload) == 160
+    # redundant frames are skipped, only the primary is encoded
+    red = np.ones(960, np.int16).tobytes()
+    frame2 = bytes([0x01, 0x01]) + len(red).to_bytes(2, "little") + red \
+        + pcm.tobytes()
+    assert g711.wire_frame_to_ulaw(frame2, 2) == payload
+
+
+def test_sdp_answer_includes_audio_mline():
+    from selkies_amd.webrtc import sdp
+    offer = "\r\n".join([
+        "v=0", "o=- 1 1 IN IP4 0.0.0.0", "s=-", "t=0 0",
+        "a=group:BUNDLE 0 1",
+        "m=audio 9 UDP/TLS/RTP/SAVPF 111 0 8",
+        "a=mid:0", "a=ice-ufrag:abcd", "a=ice-pwd:" + "p" * 22,
+        "a=fingerprint:sha-256 " + "AB:" * 31 + "AB",
+        "a=rtpmap:111 opus/48000/2", "a=rtpmap:0 PCMU/8000",
+        "m=video 9 UDP/TLS/RTP/SAVPF 102",
+        "a=mid:1", "a=ice-ufrag:abcd", "a=ice-pwd:" + "p" * 22,
+        "a=fingerprint:sha-256 " + "AB:" * 31 + "AB",
+        "a=rtpmap:102 H264/90000",
+        "a=fmtp:102 packetization-mode=1;profile-level-id=42e01f",
+    ]) + "\r\n"
+    o = sdp.parse_offer(offer)
+    audio = next(m for m in o.media if m.kind == "audio")
+    assert audio.g711_pt == 0
+    ans = sdp.build_answer(o, "uf", "pw", "FP", "10.0.0.1", 5000,
+                           ssrc=1234, audio_ssrc=5678)
+    assert "m=audio 5000 UDP/TLS/RTP/SAVPF 0" in ans
+    assert "a=rtpmap:0 PCMU/8000" in ans
+    assert "a=ssrc:5678" in ans
+    # without audio_ssrc the m-line is rejected but kept for BUNDLE
+    ans2 = sdp.build_answer(o, "uf", "pw", "FP", "10.0.0.1", 5000,
+                            ssrc=1234)
+    assert "m=audio 0 " in ans2
+
+
+def test_audio_rtp_packetizer():
+    from selkies_amd.webrtc import rtp
+    p = rtp.AudioPacketizer(ssrc=42, payload_type=0)
+    pkt1 = p.packetize(b"\x55" * 160)
+    pkt2 = p.packetize(b"\x55" * 160)
+    assert pkt1[0] == 0x80
+    assert pkt1[1] & 0x80          # marker on first packet
+    assert not (pkt2[1] & 0x80)
+    import struct
+    _, _, seq1, ts1, ssrc1 = struct.unpack(">BBHII", pkt1[:12])
+    _, _, seq2, ts2, _ = struct.unpack(">BBHII", pkt2[:12])
+    assert ssrc1 == 42 and seq2 == seq1 + 1
+    assert ts2 - ts1 == 160         # 20 ms at 8 kHz
