@@ -480,11 +480,28 @@ struct StripeEncoder::Impl {
   // Hadamard for inter); chroma as 2x2-Hadamard DC + AC. Inter quant
   // rounding (f = 2^qbits/6). MVs are even integers so the chroma MC is
   // an integer copy (mv/2). Mirrors the GPU row kernel's inter path.
-  void encode_p16(BitWriter& bw, int mbx, int mby, int qp, int mvx, int mvy,
-                  RowCtx& ctx, MbTotals& tot) {
+  void encode_p16(BitWriter& bw, int mbx, int mby, int qp, int mvq_x,
+                  int mvq_y, RowCtx& ctx, MbTotals& tot) {
     const int x0 = mbx * 16, y0 = mby * 16;
     const int cx0 = mbx * 8, cy0 = mby * 8;
-    const int cmx = mvx / 2, cmy = mvy / 2;
+    const int ix = mvq_x >> 2, iy = mvq_y >> 2;
+    const int fx = mvq_x & 3, fy = mvq_y & 3;
+    const int cix = mvq_x >> 3, ciy = mvq_y >> 3;
+    const int cdx = mvq_x & 7, cdy = mvq_y & 7;
+    // prediction planes (interpolated once, used for residual AND recon)
+    uint8_t pred_y[256];
+    uint8_t pred_c[2][64];
+    for (int r = 0; r < 16; ++r)
+      for (int c = 0; c < 16; ++c)
+        pred_y[16 * r + c] = static_cast<uint8_t>(
+            luma_pred_px(ref.y, x0 + ix + c, y0 + iy + r, fx, fy));
+    for (int comp = 0; comp < 2; ++comp) {
+      const Plane& rp = comp ? ref.cr : ref.cb;
+      for (int r = 0; r < 8; ++r)
+        for (int c = 0; c < 8; ++c)
+          pred_c[comp][8 * r + c] = static_cast<uint8_t>(
+              chroma_pred_px(rp, cx0 + cix + c, cy0 + ciy + r, cdx, cdy));
+    }
 
     // ----- luma: 16 blocks, full 16-coeff zigzag, inter quant
     int zz[16][16];
@@ -496,9 +513,9 @@ struct StripeEncoder::Impl {
         int resid[16], coef[16];
         for (int r = 0; r < 4; ++r)
           for (int c = 0; c < 4; ++c) {
-            int yy = y0 + by * 4 + r, xx = x0 + bx * 4 + c;
-            resid[4 * r + c] = int(src.y.row(yy)[xx]) -
-                               int(ref.y.row(yy + mvy)[xx + mvx]);
+            int yy = by * 4 + r, xx = bx * 4 + c;
+            resid[4 * r + c] = int(src.y.row(y0 + yy)[x0 + xx]) -
+                               int(pred_y[16 * yy + xx]);
           }
         fdct4x4(resid, coef);
         for (int i = 0; i < 16; ++i) {
@@ -517,15 +534,14 @@ struct StripeEncoder::Impl {
     bool c_any_ac = false, c_any_dc = false;
     for (int comp = 0; comp < 2; ++comp) {
       Plane& sp = comp ? src.cr : src.cb;
-      Plane& rp = comp ? ref.cr : ref.cb;
       for (int sub = 0; sub < 4; ++sub) {
         int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
         int resid[16], coef[16];
         for (int r = 0; r < 4; ++r)
           for (int c = 0; c < 4; ++c) {
-            int yy = cy0 + scy + r, xx = cx0 + scx + c;
-            resid[4 * r + c] =
-                int(sp.row(yy)[xx]) - int(rp.row(yy + cmy)[xx + cmx]);
+            int yy = scy + r, xx = scx + c;
+            resid[4 * r + c] = int(sp.row(cy0 + yy)[cx0 + xx]) -
+                               int(pred_c[comp][8 * yy + xx]);
           }
         fdct4x4(resid, coef);
         cdc[comp][sub] = coef[0];
@@ -553,8 +569,8 @@ struct StripeEncoder::Impl {
     bw.ue(0);  // mb_type P_L0_16x16
     int mvpx = ctx.have_left && ctx.left_is_inter ? ctx.left_mvx : 0;
     int mvpy = ctx.have_left && ctx.left_is_inter ? ctx.left_mvy : 0;
-    bw.se(mvx * 4 - mvpx);
-    bw.se(mvy * 4 - mvpy);
+    bw.se(mvq_x - mvpx);
+    bw.se(mvq_y - mvpy);
     bw.ue(inter_cbp_codenum(cbp));
     MbTotals newtot;
     if (cbp) {
@@ -596,9 +612,9 @@ struct StripeEncoder::Impl {
         int b = by * 4 + bx;
         if (!cbp_luma) {
           for (int r = 0; r < 4; ++r) {
-            int yy = y0 + by * 4 + r, xx = x0 + bx * 4;
-            std::memcpy(cur.y.row(yy) + xx, ref.y.row(yy + mvy) + xx + mvx,
-                        4);
+            int yy = by * 4 + r, xx = bx * 4;
+            std::memcpy(cur.y.row(y0 + yy) + x0 + xx,
+                        pred_y + 16 * yy + xx, 4);
           }
           continue;
         }
@@ -612,14 +628,13 @@ struct StripeEncoder::Impl {
         idct4x4(dqb, rec);
         for (int r = 0; r < 4; ++r)
           for (int c = 0; c < 4; ++c) {
-            int yy = y0 + by * 4 + r, xx = x0 + bx * 4 + c;
-            cur.y.row(yy)[xx] =
-                clip8(rec[4 * r + c] + int(ref.y.row(yy + mvy)[xx + mvx]));
+            int yy = by * 4 + r, xx = bx * 4 + c;
+            cur.y.row(y0 + yy)[x0 + xx] =
+                clip8(rec[4 * r + c] + int(pred_y[16 * yy + xx]));
           }
       }
     for (int comp = 0; comp < 2; ++comp) {
       Plane& rc = comp ? cur.cr : cur.cb;
-      Plane& rp = comp ? ref.cr : ref.cb;
       int dcq[4] = {0, 0, 0, 0};
       if (cbp_chroma >= 1) {
         int* q = cqdc[comp];
@@ -634,8 +649,9 @@ struct StripeEncoder::Impl {
         int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
         if (cbp_chroma == 0) {
           for (int r = 0; r < 4; ++r) {
-            int yy = cy0 + scy + r, xx = cx0 + scx;
-            std::memcpy(rc.row(yy) + xx, rp.row(yy + cmy) + xx + cmx, 4);
+            int yy = scy + r, xx = scx;
+            std::memcpy(rc.row(cy0 + yy) + cx0 + xx,
+                        pred_c[comp] + 8 * yy + xx, 4);
           }
           continue;
         }
@@ -651,12 +667,77 @@ struct StripeEncoder::Impl {
         idct4x4(dqb, rec);
         for (int r = 0; r < 4; ++r)
           for (int c = 0; c < 4; ++c) {
-            int yy = cy0 + scy + r, xx = cx0 + scx + c;
-            rc.row(yy)[xx] =
-                clip8(rec[4 * r + c] + int(rp.row(yy + cmy)[xx + cmx]));
+            int yy = scy + r, xx = scx + c;
+            rc.row(cy0 + yy)[cx0 + xx] =
+                clip8(rec[4 * r + c] + int(pred_c[comp][8 * yy + xx]));
           }
       }
     }
+  }
+
+  // ---- sub-pel motion compensation (8.4.2.2.1 subset) ---------------------
+  // MVs live on the HALF-pel grid (quarter-pel units, mv % 2 == 0): luma
+  // uses the 6-tap (1,-5,20,20,-5,1) filter at half positions, chroma the
+  // spec bilinear with eighth-pel weights (mv & 7 in {0,2,4,6}).
+  static inline int tap6(int a, int b, int c, int d, int e, int f) {
+    return a - 5 * b + 20 * c + 20 * d - 5 * e + f;
+  }
+
+  // unclipped horizontal 6-tap sum at integer (x,y), half offset in x
+  inline int hsum6(const Plane& p, int x, int y) const {
+    const uint8_t* r = p.row(y);
+    return tap6(r[x - 2], r[x - 1], r[x], r[x + 1], r[x + 2], r[x + 3]);
+  }
+
+  // predicted luma sample at integer base (x,y) with frac (fx,fy) in {0,2}
+  inline int luma_pred_px(const Plane& p, int x, int y, int fx,
+                          int fy) const {
+    if (fx == 0 && fy == 0) return p.row(y)[x];
+    if (fy == 0) return clip8((hsum6(p, x, y) + 16) >> 5);
+    if (fx == 0) {
+      int v = tap6(p.row(y - 2)[x], p.row(y - 1)[x], p.row(y)[x],
+                   p.row(y + 1)[x], p.row(y + 2)[x], p.row(y + 3)[x]);
+      return clip8((v + 16) >> 5);
+    }
+    // j position: vertical 6-tap over unclipped horizontal sums
+    int v = tap6(hsum6(p, x, y - 2), hsum6(p, x, y - 1), hsum6(p, x, y),
+                 hsum6(p, x, y + 1), hsum6(p, x, y + 2),
+                 hsum6(p, x, y + 3));
+    return clip8((v + 512) >> 10);
+  }
+
+  // predicted chroma sample: integer base (cx,cy), eighth-pel (dx,dy)
+  inline int chroma_pred_px(const Plane& p, int cx, int cy, int dx,
+                            int dy) const {
+    if (dx == 0 && dy == 0) return p.row(cy)[cx];
+    int a = p.row(cy)[cx], b = p.row(cy)[cx + 1];
+    int c = p.row(cy + 1)[cx], d = p.row(cy + 1)[cx + 1];
+    return ((8 - dx) * (8 - dy) * a + dx * (8 - dy) * b +
+            (8 - dx) * dy * c + dx * dy * d + 32) >> 6;
+  }
+
+  // A candidate MV (quarter-pel) is usable iff the full interpolation
+  // window stays inside the padded plane (uniform margin rule, applied
+  // to integer candidates too so refinements around them stay legal).
+  inline bool mv_window_ok(int x0, int y0, int mvq_x, int mvq_y) const {
+    int ix = mvq_x >> 2, iy = mvq_y >> 2;
+    return x0 + ix - 2 >= 0 && x0 + ix + 19 <= yw && y0 + iy - 2 >= 0 &&
+           y0 + iy + 19 <= yh;
+  }
+
+  // interpolated-SAD of the 16x16 at (x0,y0) for quarter-pel mv
+  long sad16_q(int x0, int y0, int mvq_x, int mvq_y) const {
+    int ix = mvq_x >> 2, iy = mvq_y >> 2;
+    int fx = mvq_x & 3, fy = mvq_y & 3;
+    long s = 0;
+    for (int r = 0; r < 16; ++r) {
+      const uint8_t* sp = src.y.row(y0 + r) + x0;
+      for (int c = 0; c < 16; ++c)
+        s += std::abs(int(sp[c]) -
+                      luma_pred_px(ref.y, x0 + ix + c, y0 + iy + r, fx,
+                                   fy));
+    }
+    return s;
   }
 
   // ---- P macroblock helpers -----------------------------------------------
@@ -763,18 +844,20 @@ struct StripeEncoder::Impl {
           if (stats) ++stats->mb_skip;
           continue;
         }
-        // small even-step diamond search around (0,0), clamped to frame
+        // integer diamond search around (0,0) (unit steps now that the
+        // chroma MC interpolates), then half-pel refinement
         int best_mvx = 0, best_mvy = 0;
         long best = sad0;
-        static const int pat[8][2] = {{-2, 0}, {2, 0},  {0, -2}, {0, 2},
-                                      {-2, -2}, {2, 2}, {-2, 2}, {2, -2}};
-        for (int iter = 0; iter < 8; ++iter) {
+        static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
+                                      {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
+        for (int iter = 0; iter < 16; ++iter) {
           int bmx = best_mvx, bmy = best_mvy;
           bool improved = false;
           for (auto& p : pat) {
             int mx = bmx + p[0], my = bmy + p[1];
-            if (x0 + mx < 0 || y0 + my < 0 || x0 + mx + 16 > yw ||
-                y0 + my + 16 > yh || std::abs(mx) > 16 || std::abs(my) > 16)
+            if (std::abs(mx) > 16 || std::abs(my) > 16) continue;
+            if (!(mx == 0 && my == 0) &&
+                !mv_window_ok(x0, y0, mx * 4, my * 4))
               continue;
             long s = sad16(src.y.row(y0) + x0, src.y.pitch,
                            ref.y.row(y0 + my) + x0 + mx, ref.y.pitch);
@@ -787,14 +870,29 @@ struct StripeEncoder::Impl {
           }
           if (!improved) break;
         }
+        // half-pel refinement (quarter-pel units, grid step 2)
+        int best_q_x = best_mvx * 4, best_q_y = best_mvy * 4;
+        {
+          int cqx = best_q_x, cqy = best_q_y;
+          for (auto& p : pat) {
+            int qx = cqx + 2 * p[0], qy = cqy + 2 * p[1];
+            if (!mv_window_ok(x0, y0, qx, qy)) continue;
+            long s = sad16_q(x0, y0, qx, qy);
+            if (s < best) {
+              best = s;
+              best_q_x = qx;
+              best_q_y = qy;
+            }
+          }
+        }
         if (best <= inter_thresh) {
           // P_L0_16x16 with coded residual (cbp may still come out 0)
           flush_skip_run(b, ctx);
-          encode_p16(b, mbx, mb_row, qp, best_mvx, best_mvy, ctx, tot);
+          encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
           ctx.have_left = true;
           ctx.left_is_inter = true;
-          ctx.left_mvx = best_mvx * 4;
-          ctx.left_mvy = best_mvy * 4;
+          ctx.left_mvx = best_q_x;
+          ctx.left_mvy = best_q_y;
           for (int by = 0; by < 4; ++by)
             ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
           for (int cy = 0; cy < 2; ++cy) {

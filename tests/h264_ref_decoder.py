@@ -41,6 +41,50 @@ def coeff_class(i, j):
     return 0 if (ei and ej) else (1 if (not ei and not ej) else 2)
 
 
+def _tap6(a, b, c, d, e, f):
+    return a - 5 * b + 20 * c + 20 * d - 5 * e + f
+
+
+def luma_mc(ref, x0, y0, mvx, mvy):
+    """16x16 luma MC at quarter-pel mv restricted to the half-pel grid
+    (mv % 2 == 0): H.264 6-tap half-sample interpolation (8.4.2.2.1)."""
+    ix, iy = mvx >> 2, mvy >> 2
+    fx, fy = mvx & 3, mvy & 3
+    bx, by = x0 + ix, y0 + iy
+    if fx == 0 and fy == 0:
+        return ref[by:by + 16, bx:bx + 16].astype(np.int64)
+    win = ref[by - 2:by + 19, bx - 2:bx + 19].astype(np.int64)
+    if fy == 0:
+        r = win[2:18]
+        s = _tap6(r[:, 0:16], r[:, 1:17], r[:, 2:18], r[:, 3:19],
+                  r[:, 4:20], r[:, 5:21])
+        return np.clip((s + 16) >> 5, 0, 255)
+    if fx == 0:
+        c = win[:, 2:18]
+        s = _tap6(c[0:16], c[1:17], c[2:18], c[3:19], c[4:20], c[5:21])
+        return np.clip((s + 16) >> 5, 0, 255)
+    hs = _tap6(win[:, 0:16], win[:, 1:17], win[:, 2:18], win[:, 3:19],
+               win[:, 4:20], win[:, 5:21])
+    s = _tap6(hs[0:16], hs[1:17], hs[2:18], hs[3:19], hs[4:20], hs[5:21])
+    return np.clip((s + 512) >> 10, 0, 255)
+
+
+def chroma_mc(ref, cx0, cy0, mvx, mvy):
+    """8x8 chroma MC: spec bilinear with eighth-pel weights."""
+    cix, ciy = mvx >> 3, mvy >> 3
+    dx, dy = mvx & 7, mvy & 7
+    bx, by = cx0 + cix, cy0 + ciy
+    h_ext = 9 if dy else 8
+    w_ext = 9 if dx else 8
+    W = ref[by:by + h_ext, bx:bx + w_ext].astype(np.int64)
+    a = W[0:8, 0:8]
+    b = W[0:8, 1:9] if dx else a
+    c = W[1:9, 0:8] if dy else a
+    d = (W[1:9, 1:9] if dx else c) if dy else b
+    return ((8 - dx) * (8 - dy) * a + dx * (8 - dy) * b +
+            (8 - dx) * dy * c + dx * dy * d + 32) >> 6
+
+
 def build_decode_tree(entries):
     """entries: list of (symbol, len, bits) -> dict[(len,bits)] = symbol"""
     return {(ln, bits): sym for sym, ln, bits in entries}
@@ -442,18 +486,12 @@ class Decoder:
         mvpx, mvpy = ctx["left_mv"] if (ctx["left_avail"] and
                                         ctx["left_inter"]) else (0, 0)
         mvx, mvy = mvdx + mvpx, mvdy + mvpy
-        assert mvx % 8 == 0 and mvy % 8 == 0, "subset: even integer MVs"
-        ix, iy = mvx // 4, mvy // 4
+        assert mvx % 2 == 0 and mvy % 2 == 0, "subset: half-pel MV grid"
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
-        pred_y = self.ref_y[y0 + iy:y0 + iy + 16,
-                            x0 + ix:x0 + ix + 16].astype(np.int64)
-        pred_cb = self.ref_cb[cy0 + iy // 2:cy0 + iy // 2 + 8,
-                              cx0 + ix // 2:cx0 + ix // 2 + 8].astype(
-                                  np.int64)
-        pred_cr = self.ref_cr[cy0 + iy // 2:cy0 + iy // 2 + 8,
-                              cx0 + ix // 2:cx0 + ix // 2 + 8].astype(
-                                  np.int64)
+        pred_y = luma_mc(self.ref_y, x0, y0, mvx, mvy)
+        pred_cb = chroma_mc(self.ref_cb, cx0, cy0, mvx, mvy)
+        pred_cr = chroma_mc(self.ref_cr, cx0, cy0, mvx, mvy)
         new_luma_nc = np.zeros((4, 4), np.int32)
         new_cb_nc = np.zeros((2, 2), np.int32)
         new_cr_nc = np.zeros((2, 2), np.int32)
