@@ -83,6 +83,49 @@ __device__ inline int dequant_chroma_dc_v(int c, int qp) {
 
 __device__ inline uint8_t clip8(int v) { return (uint8_t)max(0, min(255, v)); }
 
+// ---- H.264 half-sample interpolation (8.4.2.2.1 subset) -------------------
+// MVs live on the half-pel grid (quarter-pel units, mv % 2 == 0).
+__device__ inline int tap6i(int a, int b, int c, int d, int e, int f) {
+  return a - 5 * b + 20 * c + 20 * d - 5 * e + f;
+}
+
+__device__ inline int hsum6g(const uint8_t* row, int x) {
+  return tap6i(row[x - 2], row[x - 1], row[x], row[x + 1], row[x + 2],
+               row[x + 3]);
+}
+
+// predicted luma sample at integer (x,y) with frac (fx,fy) in {0,2}
+__device__ inline int luma_interp(const uint8_t* p, int pitch, int x, int y,
+                                  int fx, int fy) {
+  if ((fx | fy) == 0) return p[(size_t)y * pitch + x];
+  if (fy == 0) return clip8((hsum6g(p + (size_t)y * pitch, x) + 16) >> 5);
+  if (fx == 0) {
+    const uint8_t* c = p + (size_t)(y - 2) * pitch + x;
+    int v = tap6i(c[0], c[pitch], c[2 * pitch], c[3 * pitch], c[4 * pitch],
+                  c[5 * pitch]);
+    return clip8((v + 16) >> 5);
+  }
+  int v = tap6i(hsum6g(p + (size_t)(y - 2) * pitch, x),
+                hsum6g(p + (size_t)(y - 1) * pitch, x),
+                hsum6g(p + (size_t)y * pitch, x),
+                hsum6g(p + (size_t)(y + 1) * pitch, x),
+                hsum6g(p + (size_t)(y + 2) * pitch, x),
+                hsum6g(p + (size_t)(y + 3) * pitch, x));
+  return clip8((v + 512) >> 10);
+}
+
+// predicted chroma sample: bilinear with eighth-pel weights
+__device__ inline int chroma_interp(const uint8_t* p, int pitch, int x,
+                                    int y, int dx, int dy) {
+  const uint8_t* r0 = p + (size_t)y * pitch + x;
+  int a = r0[0];
+  int b = dx ? r0[1] : a;
+  int c = dy ? r0[pitch] : a;
+  int d = dy ? (dx ? r0[pitch + 1] : c) : b;
+  return ((8 - dx) * (8 - dy) * a + dx * (8 - dy) * b + (8 - dx) * dy * c +
+          dx * dy * d + 32) >> 6;
+}
+
 // ---------------------------------------------------------------------------
 // 16-lane-group 4x4 transforms via shuffles. lane c in [0,16): r=c>>2, x=c&3.
 // base = (lane & ~15) is the group's first lane in the wave.
@@ -266,6 +309,7 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       // Hadamard, inter quant rounding (f = 2^qbits/6). cbp==0 falls out
       // naturally (idct of zeros is zero -> recon == MC pred).
       const int ix = mvx >> 2, iy = mvy >> 2;
+      const int fx = mvx & 3, fy = mvy & 3;
       uint32_t psrc, ppred;
       {
         const uint8_t* sr = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
@@ -273,10 +317,17 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
                ((uint32_t)sr[min(x0 + cq + 1, w - 1)] << 8) |
                ((uint32_t)sr[min(x0 + cq + 2, w - 1)] << 16) |
                ((uint32_t)sr[min(x0 + cq + 3, w - 1)] << 24);
-        const uint8_t* pr =
-            refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
-        ppred = (uint32_t)pr[0] | ((uint32_t)pr[1] << 8) |
-                ((uint32_t)pr[2] << 16) | ((uint32_t)pr[3] << 24);
+        if ((fx | fy) == 0) {
+          const uint8_t* pr =
+              refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
+          ppred = (uint32_t)pr[0] | ((uint32_t)pr[1] << 8) |
+                  ((uint32_t)pr[2] << 16) | ((uint32_t)pr[3] << 24);
+        } else {
+          ppred = 0;
+          for (int j = 0; j < 4; ++j)
+            ppred |= (uint32_t)luma_interp(refY, ypitch, x0 + ix + cq + j,
+                                           y0 + iy + r, fx, fy) << (8 * j);
+        }
       }
       auto pix_at = [&](int py, int px) -> int {
         uint32_t v = __shfl(psrc, py * 4 + (px >> 2));
@@ -481,7 +532,8 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
     if (mode == kInter) {
       // chroma inter residual: MC pred (integer, mv/2) + DC Hadamard + AC,
       // inter quant rounding. cbp falls out of the quantized levels.
-      const int ix = (mvx >> 2) >> 1, iy = (mvy >> 2) >> 1;
+      const int cix = mvx >> 3, ciy = mvy >> 3;
+      const int cdx = mvx & 7, cdy = mvy & 7;
       uint32_t csrc = 0, cprd = 0;
       if (lane < 32) {
         int comp = lane >> 4;
@@ -495,10 +547,11 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                              min(cx0 + ccq + 2, cw - 1)] << 16) |
                ((uint32_t)sp[(size_t)sr * cpitch +
                              min(cx0 + ccq + 3, cw - 1)] << 24);
-        const uint8_t* rp = (comp ? refCr : refCb) +
-                            (size_t)(cy0 + iy + rr) * cpitch + cx0 + ix + ccq;
-        cprd = (uint32_t)rp[0] | ((uint32_t)rp[1] << 8) |
-               ((uint32_t)rp[2] << 16) | ((uint32_t)rp[3] << 24);
+        const uint8_t* rp = comp ? refCr : refCb;
+        for (int j = 0; j < 4; ++j)
+          cprd |= (uint32_t)chroma_interp(rp, cpitch, cx0 + cix + ccq + j,
+                                          cy0 + ciy + rr, cdx, cdy)
+                  << (8 * j);
       }
       auto cpix_at = [&](int comp, int rr, int cc) -> int {
         uint32_t v = __shfl(csrc, (comp << 4) | (rr << 1) | (cc >> 2));
@@ -829,6 +882,21 @@ __global__ void __launch_bounds__(64) k_h264_me(
             abs(sp[3] - rp[3]);
     return wave_sum_i(s);
   };
+  // interpolated SAD at a quarter-pel mv (half-pel grid)
+  auto sad_q = [&](int qx, int qy) -> int {
+    int ixq = qx >> 2, iyq = qy >> 2, fxq = qx & 3, fyq = qy & 3;
+    int s = 0;
+    for (int j = 0; j < 4; ++j)
+      s += abs(sp[j] - luma_interp(refY, ypitch, x0 + ixq + cq + j,
+                                   y0 + iyq + r, fxq, fyq));
+    return wave_sum_i(s);
+  };
+  // the full interpolation window must stay inside frame + stripe
+  auto window_ok = [&](int qx, int qy) -> bool {
+    int ixq = qx >> 2, iyq = qy >> 2;
+    return x0 + ixq - 2 >= 0 && x0 + ixq + 19 <= frame_w_mb16 &&
+           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 19 <= job.stripe_y1;
+  };
 
   const int qp = job.qp;
   const int skip_thresh = 48 << (qp / 6);
@@ -836,22 +904,21 @@ __global__ void __launch_bounds__(64) k_h264_me(
   const int inter_thresh = 6 * skip_thresh;
 
   int sad0 = sad_at(0, 0);
-  int mode, bmx = 0, bmy = 0;
+  int mode, bqx = 0, bqy = 0;
   if (sad0 <= skip_thresh) {
     mode = kSkip;
   } else {
     int best = sad0;
-    static const int pat[8][2] = {{-2, 0}, {2, 0},  {0, -2}, {0, 2},
-                                  {-2, -2}, {2, 2}, {-2, 2}, {2, -2}};
-    for (int iter = 0; iter < 8; ++iter) {
+    int bmx = 0, bmy = 0;
+    static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
+                                  {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
+    for (int iter = 0; iter < 16; ++iter) {
       int cx = bmx, cy = bmy;
       bool improved = false;
       for (int pi = 0; pi < 8; ++pi) {
         int mx = cx + pat[pi][0], my = cy + pat[pi][1];
-        if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
-            y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1 ||
-            abs(mx) > 16 || abs(my) > 16)
-          continue;
+        if (abs(mx) > 16 || abs(my) > 16) continue;
+        if (!(mx == 0 && my == 0) && !window_ok(mx * 4, my * 4)) continue;
         int s = sad_at(mx, my);
         if (s < best) {
           best = s;
@@ -862,6 +929,22 @@ __global__ void __launch_bounds__(64) k_h264_me(
       }
       if (!improved) break;
     }
+    // half-pel refinement ring
+    bqx = bmx * 4;
+    bqy = bmy * 4;
+    {
+      int cqx = bqx, cqy = bqy;
+      for (int pi = 0; pi < 8; ++pi) {
+        int qx = cqx + 2 * pat[pi][0], qy = cqy + 2 * pat[pi][1];
+        if (!window_ok(qx, qy)) continue;
+        int s = sad_q(qx, qy);
+        if (s < best) {
+          best = s;
+          bqx = qx;
+          bqy = qy;
+        }
+      }
+    }
     if (best <= inter_thresh)
       mode = kInter;   // mv may be (0,0): residuals carry the change
     else
@@ -869,8 +952,7 @@ __global__ void __launch_bounds__(64) k_h264_me(
   }
   if (lane == 0) {
     meta[mb_index * kMetaPerMb + 0] = mode;
-    meta[mb_index * kMetaPerMb + 1] =
-        ((bmx * 4) & 0xFFFF) | ((bmy * 4) << 16);
+    meta[mb_index * kMetaPerMb + 1] = (bqx & 0xFFFF) | (bqy << 16);
   }
 }
 
@@ -1033,28 +1115,72 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
             abs(sp[3] - rp[3]);
     return wave_sum_i(s);
   };
+  // interpolated SAD at a quarter-pel mv (half-pel grid)
+  auto sad_q = [&](int qx, int qy) -> int {
+    int ixq = qx >> 2, iyq = qy >> 2, fxq = qx & 3, fyq = qy & 3;
+    int s = 0;
+    for (int j = 0; j < 4; ++j)
+      s += abs(sp[j] - luma_interp(refY, ypitch, x0 + ixq + cq + j,
+                                   y0 + iyq + r, fxq, fyq));
+    return wave_sum_i(s);
+  };
+  auto window_ok = [&](int qx, int qy) -> bool {
+    int ixq = qx >> 2, iyq = qy >> 2;
+    return x0 + ixq - 2 >= 0 && x0 + ixq + 19 <= frame_w_mb16 &&
+           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 19 <= job.stripe_y1;
+  };
+
   const int qp = job.qp;
   const int skip_thresh = 48 << (qp / 6);
   const int inter_thresh = 6 * skip_thresh;
   int sad0 = sad_at(0, 0);
-  int mode, omvx = 0, omvy = 0;
+  int mode, oqx = 0, oqy = 0;
   if (sad0 <= skip_thresh) {
     mode = kSkip;
   } else {
     int best_sad = (best_mvx || best_mvy) ? sad_at(best_mvx, best_mvy)
                                           : sad0;
+    // refine the even-grid SSD winner: integer +-1 ring, then half-pel
+    int bqx = best_mvx * 4, bqy = best_mvy * 4;
+    static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
+                                  {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
+    {
+      int cqx = bqx, cqy = bqy;
+      for (int pi = 0; pi < 8; ++pi) {
+        int qx = cqx + 4 * pat[pi][0], qy = cqy + 4 * pat[pi][1];
+        if (!window_ok(qx, qy)) continue;
+        int s = sad_at(qx >> 2, qy >> 2);
+        if (s < best_sad) {
+          best_sad = s;
+          bqx = qx;
+          bqy = qy;
+        }
+      }
+    }
+    {
+      int cqx = bqx, cqy = bqy;
+      for (int pi = 0; pi < 8; ++pi) {
+        int qx = cqx + 2 * pat[pi][0], qy = cqy + 2 * pat[pi][1];
+        if (!window_ok(qx, qy)) continue;
+        int s = sad_q(qx, qy);
+        if (s < best_sad) {
+          best_sad = s;
+          bqx = qx;
+          bqy = qy;
+        }
+      }
+    }
     if (best_sad <= inter_thresh) {
       mode = kInter;   // mv may be (0,0): residuals carry the change
-      omvx = best_mvx;
-      omvy = best_mvy;
+      oqx = bqx;
+      oqy = bqy;
     } else {
       mode = kIntra;
     }
   }
   if (lane == 0) {
     meta[mb_index * kMetaPerMb + 0] = mode;
-    meta[mb_index * kMetaPerMb + 1] =
-        ((omvx * 4) & 0xFFFF) | ((omvy * 4) << 16);
+    meta[mb_index * kMetaPerMb + 1] = (oqx & 0xFFFF) | (oqy << 16);
   }
 }
 

@@ -237,3 +237,51 @@ def test_gpu_h264_1080p_throughput():
     dt = (time.monotonic() - t0) / len(frames)
     print(f"1080p noise encode: {dt*1000:.2f} ms/frame")
     assert dt < 0.05, f"1080p encode too slow: {dt*1000:.1f} ms"
+
+
+def test_gpu_halfpel_and_odd_mvs():
+    """GPU pipeline codes odd-integer and half-pel MVs and the streams
+    stay decodable bit-exactly (the decoder asserts the half-pel grid and
+    interpolates per 8.4.2.2.1 — any GPU/decoder interp mismatch breaks
+    the P-chain PSNR)."""
+    require_gpu()
+    w, h = 192, 96
+    rng = np.random.default_rng(31)
+    base = rng.integers(0, 256, (h, w, 4), dtype=np.uint8)
+    f = base.astype(np.float32)
+    for _ in range(2):
+        f = (f + np.roll(f, 1, 1) + np.roll(f, -1, 1) +
+             np.roll(f, 1, 0) + np.roll(f, -1, 0)) / 5
+    f0 = f.astype(np.uint8)
+    f0[:, :, 3] = 255
+    f1 = np.roll(f0, 1, axis=1)
+    fh = ((f0.astype(np.uint16) + np.roll(f0, 1, 1)) // 2).astype(np.uint8)
+    fh[:, :, 3] = 255
+    frames = [np.ascontiguousarray(x) for x in (f0, f1, fh)]
+    out = _native._pipeline_encode("gpu", frames, w, h, 22, 96, 1)
+    rows = reassemble(out)
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = []
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs.append(tuple(ctx["left_mv"]))
+
+    all_mvs = []
+    for y0, stream in rows.items():
+        d = MvDecoder()
+        decoded = d.decode(bytes(stream))
+        assert len(decoded) == 3
+        all_mvs += d.mvs
+        # quality must track the source for the half-shifted frame
+        sy, _, _ = hipflux.bgrx_to_yuv420(fh.tobytes(), w, h)
+        syn = np.frombuffer(sy, np.uint8).reshape(h, w)
+        p = psnr(decoded[2][0], syn[y0:y0 + decoded[2][0].shape[0]])
+        assert p > 34, f"half-shift frame quality {p:.1f}"
+    odd = [mv for mv in all_mvs if (mv[0] // 4) % 2 == 1 and mv[0] % 4 == 0]
+    half = [mv for mv in all_mvs if mv[0] % 4 == 2 or mv[1] % 4 == 2]
+    assert odd, "no odd-integer MVs coded by the GPU pipeline"
+    assert half, "no half-pel MVs coded by the GPU pipeline"
