@@ -186,7 +186,14 @@ function canvasPos(ev) {
           Math.max(0, Math.min(canvas.height - 1, y))];
 }
 
-function send(msg) { if (ws && ws.readyState === 1) ws.send(msg); }
+function send(msg) {
+  /* input verbs prefer the WebRTC data channel (lower latency path);
+     everything else (and fallback) uses the control WebSocket */
+  if (dcInput && dcInput.readyState === "open") {
+    try { dcInput.send(msg); return; } catch (e) { /* fall through */ }
+  }
+  if (ws && ws.readyState === 1) ws.send(msg);
+}
 window.skSend = send;   /* dashboard hook */
 
 function hookInput() {
@@ -343,11 +350,18 @@ window.addEventListener("resize", () => {
 });
 
 /* ---------------- WebRTC transport (?transport=webrtc) ----------------
- * Video arrives over DTLS-SRTP from the ice-lite endpoint; input/control
- * still ride the WebSocket (connected with display=none). */
+ * Video (and audio, G.711) arrive over DTLS-SRTP from the ice-lite
+ * endpoint; input rides the SCTP data channel when it opens, with the
+ * control WebSocket (display=none) as fallback + settings/clipboard
+ * channel. */
+let dcInput = null;            /* open RTCDataChannel, used by sendInput */
 async function connectWebRTC() {
   const pc = new RTCPeerConnection();
   pc.addTransceiver("video", { direction: "recvonly" });
+  pc.addTransceiver("audio", { direction: "recvonly" });
+  const dc = pc.createDataChannel("input", { ordered: true });
+  dc.onopen = () => { dcInput = dc; };
+  dc.onclose = () => { if (dcInput === dc) dcInput = null; };
   pc.ontrack = (ev) => {
     const vid = document.createElement("video");
     vid.autoplay = true; vid.playsInline = true; vid.muted = true;

@@ -165,6 +165,25 @@ class DtlsEndpoint:
             if err != SSL_ERROR_WANT_READ:
                 raise RuntimeError(f"DTLS handshake error {err}")
 
+    # -- application data (SCTP rides inside DTLS, RFC 8261) -----------------
+    def send_app(self, data: bytes) -> None:
+        """Encrypt one application datagram (drain with take_datagrams)."""
+        rc = api.SSL_write(self._ssl, data, len(data))
+        if rc <= 0:
+            err = api.SSL_get_error(self._ssl, rc)
+            raise RuntimeError(f"SSL_write error {err}")
+
+    def recv_app(self) -> list:
+        """Decrypted application datagrams received so far."""
+        out = []
+        while True:
+            buf = ctypes.create_string_buffer(8192)
+            n = api.SSL_read(self._ssl, buf, 8192)
+            if n <= 0:
+                break
+            out.append(buf.raw[:n])
+        return out
+
     # -- SRTP keying (RFC 5764 §4.2) -----------------------------------------
     def export_srtp_keys(self):
         """Returns ((client_key, client_salt), (server_key, server_salt))."""

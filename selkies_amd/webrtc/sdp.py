@@ -136,6 +136,21 @@ def build_answer(offer: Offer, ice_ufrag: str, ice_pwd: str,
                 f"a=candidate:1 1 udp 2130706431 {host_ip} {port} typ host",
                 "a=end-of-candidates",
             ]
+        elif m.kind == "application" and "webrtc-datachannel" in \
+                " ".join(m.raw):
+            lines += [
+                f"m=application {port} UDP/DTLS/SCTP webrtc-datachannel",
+                f"c=IN IP4 {host_ip}",
+                f"a=mid:{m.mid}",
+                f"a=ice-ufrag:{ice_ufrag}",
+                f"a=ice-pwd:{ice_pwd}",
+                f"a=fingerprint:sha-256 {fingerprint}",
+                "a=setup:passive",
+                "a=sctp-port:5000",
+                "a=max-message-size:65536",
+                f"a=candidate:1 1 udp 2130706431 {host_ip} {port} typ host",
+                "a=end-of-candidates",
+            ]
         elif m.kind == "audio" and m.g711_pt >= 0 and audio_ssrc:
             apt = m.g711_pt
             codec = "PCMU" if apt == 0 else "PCMA"

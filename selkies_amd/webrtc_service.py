@@ -19,6 +19,7 @@ from typing import Optional
 import hipflux
 
 from .webrtc import dtls, ice, rtp, sdp
+from .webrtc.sctp import PPID_STRING, PPID_STRING_EMPTY, SctpAssociation
 from .webrtc.srtp import SrtpSession, is_rtcp
 
 logger = logging.getLogger("selkies.webrtc")
@@ -30,6 +31,7 @@ class PeerState:
         self.dtls: Optional[dtls.DtlsEndpoint] = None
         self.srtp_out: Optional[SrtpSession] = None
         self.srtp_in: Optional[SrtpSession] = None
+        self.sctp: Optional[SctpAssociation] = None
         self.connected = False
         self.last_sr = 0.0
 
@@ -206,6 +208,8 @@ class WebRTCService:
             peer.dtls.put_datagram(data)
             for out in peer.dtls.take_datagrams():
                 self.transport.sendto(out, addr)
+            if peer.dtls.handshake_done:
+                self._pump_sctp(peer)
             if peer.dtls.handshake_done and not peer.connected:
                 (ck, cs), (sk, ss) = peer.dtls.export_srtp_keys()
                 # we are the DTLS server: send with server keys,
@@ -229,11 +233,74 @@ class WebRTCService:
                         if self.capture:
                             self.capture.request_idr_frame()
 
+    # ---- data channel (SCTP over DTLS; webrtc/sctp.py) ---------------------
+    def _pump_sctp(self, peer: PeerState):
+        """Feed decrypted app datagrams into the peer's SCTP association
+        and flush its outbound packets back through DTLS."""
+        inbound = peer.dtls.recv_app()
+        if inbound and peer.sctp is None:
+            peer.sctp = SctpAssociation(
+                True,
+                on_message=lambda sid, ppid, d, p=peer:
+                    self._on_dc_message(p, sid, ppid, d),
+                on_channel_open=lambda ch:
+                    logger.info("datachannel '%s' open (sid %d)", ch.label,
+                                ch.sid))
+        if peer.sctp is None:
+            return
+        now = time.monotonic()
+        for pkt in inbound:
+            peer.sctp.receive(pkt, now)
+        peer.sctp.poll(now)
+        out = peer.sctp.outbound()
+        if out:
+            for pkt in out:
+                peer.dtls.send_app(pkt)
+            for rec in peer.dtls.take_datagrams():
+                self.transport.sendto(rec, peer.addr)
+
+    def _on_dc_message(self, peer: PeerState, sid: int, ppid: int,
+                       data: bytes):
+        """Data-channel messages carry the same text verbs as the control
+        WebSocket (kd/ku/m/mb/... — docs/protocol.md)."""
+        if ppid not in (PPID_STRING, PPID_STRING_EMPTY):
+            return
+        try:
+            text = data.decode("utf-8")
+        except UnicodeDecodeError:
+            return
+        try:
+            self.streaming.input.on_message(text)
+        except Exception:
+            logger.debug("datachannel verb failed", exc_info=True)
+
+    def dc_broadcast(self, text: str):
+        """Send a control message to every open data channel."""
+        for peer in self.peers.values():
+            if peer.sctp is None:
+                continue
+            for sid, ch in peer.sctp.channels.items():
+                if ch.open:
+                    try:
+                        peer.sctp.send(sid, text, now=time.monotonic())
+                    except Exception:
+                        pass
+            self._flush_sctp(peer)
+
+    def _flush_sctp(self, peer: PeerState):
+        for pkt in peer.sctp.outbound():
+            peer.dtls.send_app(pkt)
+        for rec in peer.dtls.take_datagrams():
+            self.transport.sendto(rec, peer.addr)
+
     def stats(self) -> dict:
         return {
             "port": self.port,
             "peers": len(self.peers),
             "connected": sum(p.connected for p in self.peers.values()),
+            "datachannels": sum(
+                len([c for c in p.sctp.channels.values() if c.open])
+                for p in self.peers.values() if p.sctp),
             "frames_sent": self.frames_sent,
             "packets_sent": self.packetizer.packets_sent,
         }

@@ -260,6 +260,9 @@ def test_webrtc_loopback_end_to_end():
             assert saw_idr, "PLI did not produce an IDR"
             transport.close()
         finally:
+            # stop native capture threads BEFORE interpreter teardown:
+            # a callback firing after Py_Finalize aborts the process
+            await server.stop()
             await runner.cleanup()
 
     def ans_ssrc(answer):
@@ -352,3 +355,91 @@ def test_audio_rtp_packetizer():
     _, _, seq2, ts2, _ = struct.unpack(">BBHII", pkt2[:12])
     assert ssrc1 == 42 and seq2 == seq1 + 1
     assert ts2 - ts1 == 160         # 20 ms at 8 kHz
+
+
+def test_sctp_crc32c_check_value():
+    from selkies_amd.webrtc.sctp import crc32c
+    assert crc32c(b"123456789") == 0xE3069283   # published check value
+    assert crc32c(b"") == 0
+
+
+def _pump(a, b, drop_first_data_from=None):
+    """Exchange outbound packets until both queues drain."""
+    from selkies_amd.webrtc import sctp as S
+    dropped = [False]
+    for _ in range(50):
+        pa, pb = a.outbound(), b.outbound()
+        if not pa and not pb:
+            break
+        for p in pa:
+            if (drop_first_data_from is a and not dropped[0]
+                    and p[12] == S.CT_DATA):
+                dropped[0] = True
+                continue
+            b.receive(p)
+        for p in pb:
+            a.receive(p)
+
+
+def test_sctp_association_and_datachannel():
+    """Full loopback: INIT handshake, DCEP open/ack, string+binary
+    messages both ways, fragmentation/reassembly of a large message."""
+    from selkies_amd.webrtc.sctp import (SctpAssociation, PPID_STRING,
+                                         PPID_BINARY)
+    got_srv, got_cli, opened = [], [], []
+    srv = SctpAssociation(True,
+                          on_message=lambda s, p, d: got_srv.append((s, p, d)),
+                          on_channel_open=lambda ch: opened.append(ch))
+    cli = SctpAssociation(False,
+                          on_message=lambda s, p, d: got_cli.append((s, p, d)))
+    cli.start()
+    _pump(cli, srv)
+    assert srv.established and cli.established
+    cli.open_channel(1, "input", "")
+    _pump(cli, srv)
+    assert opened and opened[0].label == "input"
+    assert srv.channels[1].open
+    cli.send(1, "kd,65")
+    cli.send(1, b"\x01\x02\x03")
+    _pump(cli, srv)
+    assert (1, PPID_STRING, b"kd,65") in got_srv
+    assert (1, PPID_BINARY, b"\x01\x02\x03") in got_srv
+    # server -> client
+    srv.send(1, "SETTINGS_PAYLOAD,{}")
+    _pump(srv, cli)
+    assert got_cli and got_cli[-1][2] == b"SETTINGS_PAYLOAD,{}"
+    # large message fragments (3 x 1100 < 3500) and reassembles
+    big = bytes(range(256)) * 14
+    cli.send(1, big)
+    _pump(cli, srv)
+    assert got_srv[-1][2] == big
+
+
+def test_sctp_retransmit_on_loss():
+    from selkies_amd.webrtc.sctp import SctpAssociation
+    got = []
+    srv = SctpAssociation(True, on_message=lambda s, p, d: got.append(d))
+    cli = SctpAssociation(False)
+    cli.start()
+    _pump(cli, srv)
+    cli.open_channel(1, "x")
+    _pump(cli, srv)
+    # drop the first DATA packet of the message, then retransmit via poll
+    cli.send(1, "hello", now=100.0)
+    _pump(cli, srv, drop_first_data_from=cli)
+    assert b"hello" not in got
+    cli.poll(102.0)      # past RTO -> retransmit
+    _pump(cli, srv)
+    assert b"hello" in got
+
+
+def test_sctp_rejects_bad_checksum():
+    from selkies_amd.webrtc.sctp import SctpAssociation
+    srv = SctpAssociation(True)
+    cli = SctpAssociation(False)
+    cli.start()
+    pkt = bytearray(cli.outbound()[0])
+    pkt[-1] ^= 0xFF
+    srv.receive(bytes(pkt))
+    assert not srv.outbound()
+    assert srv.errors == 1
