@@ -870,9 +870,10 @@ struct StripeEncoder::Impl {
           }
           if (!improved) break;
         }
-        // half-pel refinement (quarter-pel units, grid step 2)
+        // half-pel refinement (quarter-pel units, grid step 2); skipped
+        // when the MB is headed to intra regardless
         int best_q_x = best_mvx * 4, best_q_y = best_mvy * 4;
-        {
+        if (best <= 2 * inter_thresh) {
           int cqx = best_q_x, cqy = best_q_y;
           for (auto& p : pat) {
             int qx = cqx + 2 * p[0], qy = cqy + 2 * p[1];

@@ -929,10 +929,11 @@ __global__ void __launch_bounds__(64) k_h264_me(
       }
       if (!improved) break;
     }
-    // half-pel refinement ring
+    // half-pel refinement ring (skipped when the MB is headed to intra
+    // regardless: refinement cannot halve a hopeless SAD)
     bqx = bmx * 4;
     bqy = bmy * 4;
-    {
+    if (best <= 2 * inter_thresh) {
       int cqx = bqx, cqy = bqy;
       for (int pi = 0; pi < 8; ++pi) {
         int qx = cqx + 2 * pat[pi][0], qy = cqy + 2 * pat[pi][1];
@@ -1141,10 +1142,11 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     int best_sad = (best_mvx || best_mvy) ? sad_at(best_mvx, best_mvy)
                                           : sad0;
     // refine the even-grid SSD winner: integer +-1 ring, then half-pel
+    // (skipped when the MB is headed to intra regardless)
     int bqx = best_mvx * 4, bqy = best_mvy * 4;
     static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
                                   {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
-    {
+    if (best_sad <= 2 * inter_thresh) {
       int cqx = bqx, cqy = bqy;
       for (int pi = 0; pi < 8; ++pi) {
         int qx = cqx + 4 * pat[pi][0], qy = cqy + 4 * pat[pi][1];
@@ -1156,11 +1158,9 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
           bqy = qy;
         }
       }
-    }
-    {
-      int cqx = bqx, cqy = bqy;
+      int cqx2 = bqx, cqy2 = bqy;
       for (int pi = 0; pi < 8; ++pi) {
-        int qx = cqx + 2 * pat[pi][0], qy = cqy + 2 * pat[pi][1];
+        int qx = cqx2 + 2 * pat[pi][0], qy = cqy2 + 2 * pat[pi][1];
         if (!window_ok(qx, qy)) continue;
         int s = sad_q(qx, qy);
         if (s < best_sad) {
