@@ -369,9 +369,19 @@ class HipH264Pipeline : public EncodePipeline {
       h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs, d_jobs_,
                                  d_stage_, d_nbits_, d_entout_,
                                  ent_stride_words_, d_outbits_, stream_);
-      HIP_CHECK(hipMemcpyAsync(h_entout_, d_entout_,
-                               (size_t)n_jobs * ent_stride_words_ * 4,
-                               hipMemcpyDeviceToHost, stream_));
+      // compaction: copy only ~the used prefix of each row's bitstream
+      // (adaptive cap = 2x last frame's max row, full stride first frame;
+      // rows that overflow the cap are re-copied exactly below)
+      if (ent_copy_words_ >= ent_stride_words_) {
+        HIP_CHECK(hipMemcpyAsync(h_entout_, d_entout_,
+                                 (size_t)n_jobs * ent_stride_words_ * 4,
+                                 hipMemcpyDeviceToHost, stream_));
+      } else {
+        HIP_CHECK(hipMemcpy2DAsync(
+            h_entout_, (size_t)ent_stride_words_ * 4, d_entout_,
+            (size_t)ent_stride_words_ * 4, (size_t)ent_copy_words_ * 4,
+            n_jobs, hipMemcpyDeviceToHost, stream_));
+      }
       HIP_CHECK(hipMemcpyAsync(h_outbits_, d_outbits_,
                                sizeof(int) * n_jobs, hipMemcpyDeviceToHost,
                                stream_));
@@ -461,6 +471,18 @@ class HipH264Pipeline : public EncodePipeline {
       HIP_CHECK(hipStreamSynchronize(stream_));
     } else {
       HIP_CHECK(hipStreamSynchronize(stream_));
+      int max_words = 0;
+      for (int j = 0; j < n_jobs; ++j) {
+        int wds = (h_outbits_[j] + 31) / 32 + 1;
+        max_words = std::max(max_words, wds);
+        if (wds > ent_copy_words_) {
+          // rare: this row outgrew the adaptive cap; fetch it exactly
+          HIP_CHECK(hipMemcpy(h_entout_ + (size_t)j * ent_stride_words_,
+                              d_entout_ + (size_t)j * ent_stride_words_,
+                              (size_t)wds * 4, hipMemcpyDeviceToHost));
+        }
+      }
+      ent_copy_words_ = std::min(ent_stride_words_, max_words * 2 + 64);
       for (int j = 0; j < n_jobs; ++j) {
         auto [si, r] = job_map[j];
         const uint32_t* words = h_entout_ + (size_t)j * ent_stride_words_;
@@ -588,6 +610,7 @@ class HipH264Pipeline : public EncodePipeline {
     cpu_entropy_ = std::getenv("HIPFLUX_CPU_ENTROPY") != nullptr;
     const int nitems = h264gpu::items_per_row(mbw_);
     ent_stride_words_ = nitems * h264gpu::kStageWordsPerItem;
+    ent_copy_words_ = 1 << 30;     // first frame after (re)alloc: full copy
     size_t stage_bytes =
         (size_t)mbh_ * nitems * h264gpu::kStageWordsPerItem * 4;
     d_stage_ = reinterpret_cast<uint32_t*>(dalloc(stage_bytes));
@@ -631,6 +654,7 @@ class HipH264Pipeline : public EncodePipeline {
   uint32_t* h_entout_ = nullptr;
   int* h_outbits_ = nullptr;
   int ent_stride_words_ = 0;
+  int ent_copy_words_ = 1 << 30;   // adaptive D2H cap (words per row)
   std::vector<hipEvent_t> batch_events_;
   std::vector<void*> device_ptrs_;
   std::vector<StripeState> stripes_;
