@@ -182,11 +182,17 @@ __device__ __forceinline__ void dev_cavlc_residual(DevBW& bw,
 }
 
 // ---- per-MB derived state ---------------------------------------------------
-// totals of a block (0 if the MB is not intra)
-__device__ inline int blk_total(const int16_t* L, int off, int first,
-                                int count) {
+// totals of one 16-coeff block via two int4 (8 x int16) loads; `first`
+// is 0 (full block) or 1 (AC-only: position 0 excluded)
+__device__ inline int blk_total16v(const int16_t* L, int off, int first) {
+  union U { int4 v; short s[8]; };
+  U a, b;
+  a.v = *reinterpret_cast<const int4*>(L + off);
+  b.v = *reinterpret_cast<const int4*>(L + off + 8);
   int t = 0;
-  for (int i = first; i < first + count; ++i) t += L[off + i] != 0;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) t += (a.s[i] != 0) + (b.s[i] != 0);
+  if (first) t -= a.s[0] != 0;
   return t;
 }
 
@@ -235,13 +241,13 @@ __device__ void precompute_mb(const int16_t* levels, const int* meta,
     const int16_t* L = levels + mb_index * kLevelsPerMb;
     int any_ac = 0;
     for (int b = 0; b < 16; ++b) {
-      int t = blk_total(L, kLumaAcOff + b * 16, first, 16 - first);
+      int t = blk_total16v(L, kLumaAcOff + b * 16, first);
       out->ltot[b] = (uint8_t)t;
       any_ac |= t;
     }
     int any_cac = 0, any_cdc = 0;
     for (int b = 0; b < 8; ++b) {
-      int t = blk_total(L, kChromaAcOff + b * 16, 1, 15);
+      int t = blk_total16v(L, kChromaAcOff + b * 16, 1);
       out->ctot[b] = (uint8_t)t;
       any_cac |= t;
     }
@@ -283,7 +289,7 @@ __device__ inline int lds_chroma_nc(const MbInfo* info, int mbx, int comp,
 }
 
 // ---- the kernel -------------------------------------------------------------
-__global__ void __launch_bounds__(256) k_h264_cavlc_rows(
+__global__ void __launch_bounds__(512) k_h264_cavlc_rows(
     const int16_t* __restrict__ levels, const int* __restrict__ meta,
     int mbw, const RowJob* __restrict__ jobs,
     uint32_t* __restrict__ stage,       // [row][item][kStageWordsPerItem]
@@ -292,7 +298,7 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
     int out_stride_words,
     int* __restrict__ out_bits) {       // [row]
   const RowJob job = jobs[blockIdx.x];
-  const int tid = threadIdx.x;            // 256 threads (4 waves)
+  const int tid = threadIdx.x;            // 512 threads (8 waves)
   const bool i_slice = (job.flags & 1) != 0;
   const size_t row_base = (size_t)job.mb_row * mbw;
   const int nitems = items_per_row(mbw);
@@ -305,7 +311,7 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
   __shared__ MbInfo s_mb[512];
   __shared__ short s_skiprun[512];
   __shared__ int s_trailing;
-  for (int mb = tid; mb < mbw; mb += 256)
+  for (int mb = tid; mb < mbw; mb += 512)
     precompute_mb(levels, meta, row_base + mb, i_slice, &s_mb[mb]);
   __syncthreads();
   if (tid == 0) {
@@ -324,7 +330,7 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
   __syncthreads();
 
   // ---- phase A: encode items
-  for (int item = tid; item < nitems; item += 256) {
+  for (int item = tid; item < nitems; item += 512) {
     DevBW bw;
     bw.out = row_stage + (size_t)item * kStageWordsPerItem;
     int bits = 0;
@@ -439,8 +445,8 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
   __syncthreads();
 
   // ---- phase B: exclusive prefix sum via per-thread segments
-  const int seg_lo = (int)((long)nitems * tid / 256);
-  const int seg_hi = (int)((long)nitems * (tid + 1) / 256);
+  const int seg_lo = (int)((long)nitems * tid / 512);
+  const int seg_hi = (int)((long)nitems * (tid + 1) / 512);
   int seg_sum = 0;
   for (int i = seg_lo; i < seg_hi; ++i) seg_sum += row_nbits[i];
   __shared__ int s_total;
@@ -451,13 +457,13 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
       int other = __shfl_up(inc, d);
       if (lane >= d) inc += other;
     }
-    __shared__ int s_wsum[4];
+    __shared__ int s_wsum[8];
     if (lane == 63) s_wsum[wid] = inc;
     __syncthreads();
     int wbase = 0;
     for (int w = 0; w < wid; ++w) wbase += s_wsum[w];
     int seg_off = wbase + inc - seg_sum;
-    if (tid == 255) s_total = wbase + inc;
+    if (tid == 511) s_total = wbase + inc;
     // rewrite nbits -> offsets for this thread's segment
     int off = seg_off;
     for (int i = seg_lo; i < seg_hi; ++i) {
@@ -472,10 +478,10 @@ __global__ void __launch_bounds__(256) k_h264_cavlc_rows(
 
   // ---- zero the needed output words, then scatter
   int total_words = (total_bits + 31) / 32 + 1;
-  for (int wdx = tid; wdx < total_words; wdx += 256) row_out[wdx] = 0;
+  for (int wdx = tid; wdx < total_words; wdx += 512) row_out[wdx] = 0;
   __syncthreads();
 
-  for (int item = tid; item < nitems; item += 256) {
+  for (int item = tid; item < nitems; item += 512) {
     int off = row_nbits[item];
     int next_off = item + 1 < nitems
                        ? row_nbits[item + 1]
@@ -513,7 +519,7 @@ void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
                        int* d_nbits, uint32_t* d_out, int out_stride_words,
                        int* d_out_bits, hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(256), 0, stream,
+  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(512), 0, stream,
                      d_levels, d_meta, mbw, d_jobs, d_stage, d_nbits, d_out,
                      out_stride_words, d_out_bits);
 }
