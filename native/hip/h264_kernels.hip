@@ -257,26 +257,23 @@ __device__ inline int wave_min_i(int v) {
 // of LDS staging; the only LDS state is the left-neighbor recon columns,
 // updated once per MB behind a wave-level fence.
 
-__device__ inline void wave_lds_fence() {
-  // order our own LDS writes before subsequent cross-lane LDS reads
-  __builtin_amdgcn_wave_barrier();
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_wave_barrier();
-}
-
 // ---- luma row pipeline (one wave) ----------------------------------------
 __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
                          int h, const uint8_t* __restrict__ refY,
                          uint8_t* __restrict__ curY, int mbw, int mby,
                          int qp, bool i_slice,
                          const int16_t* __restrict__ levels_base,
-                         int* __restrict__ meta, int lane,
-                         uint8_t* s_left_y) {
+                         int* __restrict__ meta, int lane) {
   const int y0 = mby * 16;
   const int r = lane >> 2, cq = (lane & 3) * 4;
   const int g = lane >> 4, c = lane & 15;
   const int zz = c_zz_of_pos[c];
   bool have_left = false;
+  // left-neighbor recon column lives in REGISTERS: lane l (<16) holds the
+  // reconstructed pixel of row l, column 15 of the previous MB. All state
+  // movement is shuffles -> the serial per-MB chain has no LDS round
+  // trips and no fences.
+  int leftreg = 0;
 
   for (int mbx = 0; mbx < mbw; ++mbx) {
     const int x0 = mbx * 16;
@@ -298,8 +295,8 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       const uint8_t* s = refY + (size_t)(y0 + r) * ypitch + x0 + cq;
       uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
       d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
-      if (cq == 12) s_left_y[r] = s[3];
-      wave_lds_fence();
+      if (lane < 16)
+        leftreg = refY[(size_t)(y0 + lane) * ypitch + x0 + 15];
       have_left = true;
       continue;
     }
@@ -352,6 +349,7 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       }
       int anyl = (lvl_p[0] | lvl_p[1] | lvl_p[2] | lvl_p[3]) != 0;
       const int cbp_luma = __ballot(anyl) ? 15 : 0;
+      int newleft = leftreg;
 #pragma unroll
       for (int pass = 0; pass < 4; ++pass) {
         int blk = pass * 4 + g;
@@ -368,11 +366,12 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
               (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
               ((uint32_t)p3 << 24);
         }
-        if (px == 15) s_left_y[py + 16] = (uint8_t)pix;
+        // collect the MB's right-edge recon into the register column:
+        // row l's col-15 pixel lives on lane 51 + 4*(l&3) in pass l>>2
+        int cand = __shfl(pix, 51 + 4 * (lane & 3));
+        if (lane < 16 && (lane >> 2) == pass) newleft = cand;
       }
-      wave_lds_fence();
-      if (lane < 16) s_left_y[lane] = s_left_y[lane + 16];
-      wave_lds_fence();
+      leftreg = newleft;
       have_left = true;
       continue;
     }
@@ -389,10 +388,11 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
     // DC value + mode costs
     int dcval = 128;
     if (have_left) {
-      int part = (lane < 16) ? s_left_y[lane] : 0;
+      int part = (lane < 16) ? leftreg : 0;
       dcval = (wave_sum_i(part) + 8) >> 4;
     }
-    int lv = have_left ? s_left_y[r] : 0;
+    int lv_all = __shfl(leftreg, r);
+    int lv = have_left ? lv_all : 0;
     int costH = 0, costDC = 0;
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
@@ -418,7 +418,8 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
-      int pred = luma_mode == 1 ? s_left_y[py] : dcval;
+      int lcol = __shfl(leftreg, py);
+      int pred = luma_mode == 1 ? lcol : dcval;
       int resid = pix_at(py, px) - pred;
       int coefv = fdct4_wave(resid, lane);
       // collect DC of block (pass*4+gg) into lane (pass*4+gg)
@@ -443,6 +444,7 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
     const int cbp_luma = __ballot(anyl) ? 15 : 0;
 
     // ---- recon passes
+    int newleft = leftreg;
 #pragma unroll
     for (int pass = 0; pass < 4; ++pass) {
       int blk = pass * 4 + g;
@@ -457,7 +459,8 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
         d = dequant_c(lvl_p[pass], qp, coeff_cls(c));
       int rec = idct4_wave(d, lane);
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
-      int pred = luma_mode == 1 ? s_left_y[py] : dcval;
+      int lcol = __shfl(leftreg, py);
+      int pred = luma_mode == 1 ? lcol : dcval;
       int pix = clip8(rec + pred);
       int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
           p3 = __shfl(pix, lane + 3);
@@ -467,11 +470,10 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
             (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
             ((uint32_t)p3 << 24);
       }
-      if (px == 15) s_left_y[py + 16] = (uint8_t)pix;  // staging half
+      int cand = __shfl(pix, 51 + 4 * (lane & 3));
+      if (lane < 16 && (lane >> 2) == pass) newleft = cand;
     }
-    wave_lds_fence();
-    if (lane < 16) s_left_y[lane] = s_left_y[lane + 16];
-    wave_lds_fence();
+    leftreg = newleft;
 
     if (lane == 0) M[0] = kIntra | (luma_mode << 2);
     have_left = true;
@@ -487,13 +489,15 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                            uint8_t* __restrict__ curCr, int mbw, int mby,
                            int qpc, bool i_slice,
                            const int16_t* __restrict__ levels_base,
-                           int* __restrict__ meta, int lane,
-                           uint8_t* s_left_cb, uint8_t* s_left_cr) {
+                           int* __restrict__ meta, int lane) {
   const int cy0 = mby * 8;
   const int g = lane >> 4, c = lane & 15;
   const int zz = c_zz_of_pos[c];
   const int cw = (w + 1) / 2, chh = (h + 1) / 2;
   bool have_left = false;
+  // left-neighbor recon columns in registers: lane l (<8) holds row l,
+  // col 7 of the previous MB for each component (no LDS, no fences).
+  int leftcb = 0, leftcr = 0;
 
   for (int mbx = 0; mbx < mbw; ++mbx) {
     const int cx0 = mbx * 8;
@@ -520,11 +524,11 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
         const uint8_t* s = sp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
         uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
         d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
-        if (cq == 4) {
-          if (comp) s_left_cr[r] = s[3]; else s_left_cb[r] = s[3];
-        }
       }
-      wave_lds_fence();
+      if (lane < 8) {
+        leftcb = refCb[(size_t)(cy0 + lane) * cpitch + cx0 + 7];
+        leftcr = refCr[(size_t)(cy0 + lane) * cpitch + cx0 + 7];
+      }
       have_left = true;
       continue;
     }
@@ -601,6 +605,7 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
         L[kChromaDcOff + lane] = (int16_t)qdc_all[lane];
       int anyc = (lvl_p[0] | lvl_p[1]) != 0 && c != 0;
       const int cbp_chroma = __ballot(anyc) ? 2 : (any_cdc ? 1 : 0);
+      int ncb = leftcb, ncr = leftcr;
 #pragma unroll
       for (int comp = 0; comp < 2; ++comp) {
         int dq0 = 0, dq1 = 0, dq2 = 0, dq3 = 0;
@@ -639,17 +644,16 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
               (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
               ((uint32_t)p3 << 24);
         }
-        if (cc2 == 7) {
-          if (comp) s_left_cr[rr + 8] = (uint8_t)pix;
-          else s_left_cb[rr + 8] = (uint8_t)pix;
+        // right-edge recon of row l lives on lane 19+4l (rows 0-3) /
+        // 35+4l (rows 4-7); collect into the register columns
+        int cand = __shfl(pix, lane < 4 ? 19 + 4 * lane : 35 + 4 * lane);
+        if (lane < 8) {
+          if (pass == 0) ncb = cand;
+          else ncr = cand;
         }
       }
-      wave_lds_fence();
-      if (lane < 8) {
-        s_left_cb[lane] = s_left_cb[lane + 8];
-        s_left_cr[lane] = s_left_cr[lane + 8];
-      }
-      wave_lds_fence();
+      leftcb = ncb;
+      leftcr = ncr;
       have_left = true;
       continue;
     }
@@ -675,18 +679,29 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
       return (v >> (8 * (cc & 3))) & 0xFF;
     };
 
+    // per-sub-row DC-prediction averages from the register columns
+    // (uniform across the wave; default 128 when no left neighbor)
+    int cbt = 128, cbb = 128, crt = 128, crb = 128;
+    if (have_left) {
+      cbt = (__shfl(leftcb, 0) + __shfl(leftcb, 1) + __shfl(leftcb, 2) +
+             __shfl(leftcb, 3) + 2) >> 2;
+      cbb = (__shfl(leftcb, 4) + __shfl(leftcb, 5) + __shfl(leftcb, 6) +
+             __shfl(leftcb, 7) + 2) >> 2;
+      crt = (__shfl(leftcr, 0) + __shfl(leftcr, 1) + __shfl(leftcr, 2) +
+             __shfl(leftcr, 3) + 2) >> 2;
+      crb = (__shfl(leftcr, 4) + __shfl(leftcr, 5) + __shfl(leftcr, 6) +
+             __shfl(leftcr, 7) + 2) >> 2;
+    }
+
     // ---- mode decision: DC(0) vs H(1), summed across both components
     int ccH = 0, ccDC = 0;
     {
       int comp = lane >> 5;
       int idx = lane & 31;
       int rr = idx >> 2, ccq = (idx & 3) * 2;
-      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
-      int scy4 = (rr >> 2) * 4;
-      int dd = 128;
-      if (have_left)
-        dd = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2) >> 2;
-      int lvv = have_left ? lc[rr] : 0;
+      int lcb = __shfl(leftcb, rr), lcr = __shfl(leftcr, rr);
+      int dd = comp ? ((rr & 4) ? crb : crt) : ((rr & 4) ? cbb : cbt);
+      int lvv = have_left ? (comp ? lcr : lcb) : 0;
       for (int k = 0; k < 2; ++k) {
         int sv = cpix_at(comp, rr, ccq + k);
         ccDC += abs(sv - dd);
@@ -706,18 +721,9 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
       int sub = g;
       int scx = (sub & 1) * 4, scy = (sub >> 1) * 4;
       int rr = scy + (c >> 2), cc2 = scx + (c & 3);
-      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
-      int pred;
-      if (chroma_mode == 1) {
-        pred = lc[rr];
-      } else {
-        pred = 128;
-        if (have_left) {
-          int scy4 = (rr >> 2) * 4;
-          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2)
-                 >> 2;
-        }
-      }
+      int lH = comp ? __shfl(leftcr, rr) : __shfl(leftcb, rr);
+      int pdc = comp ? ((rr & 4) ? crb : crt) : ((rr & 4) ? cbb : cbt);
+      int pred = chroma_mode == 1 ? lH : pdc;
       int resid = cpix_at(comp, rr, cc2) - pred;
       int coefv = fdct4_wave(resid, lane);
       // every lane grabs all 4 sub-block DCs of this pass (sub k from
@@ -773,6 +779,7 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
     }
 
     // ---- recon
+    int ncb2 = leftcb, ncr2 = leftcr;
 #pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
       int comp = pass;
@@ -784,18 +791,9 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                               ? dequant_c(lvl_p[pass], qpc, coeff_cls(c))
                               : 0);
       int rec = idct4_wave(d, lane);
-      const uint8_t* lc = comp ? s_left_cr : s_left_cb;
-      int pred;
-      if (chroma_mode == 1) {
-        pred = lc[rr];
-      } else {
-        pred = 128;
-        if (have_left) {
-          int scy4 = (rr >> 2) * 4;
-          pred = (lc[scy4] + lc[scy4 + 1] + lc[scy4 + 2] + lc[scy4 + 3] + 2)
-                 >> 2;
-        }
-      }
+      int lH = comp ? __shfl(leftcr, rr) : __shfl(leftcb, rr);
+      int pdc = comp ? ((rr & 4) ? crb : crt) : ((rr & 4) ? cbb : cbt);
+      int pred = chroma_mode == 1 ? lH : pdc;
       int pix = clip8(rec + pred);
       uint8_t* dp = comp ? curCr : curCb;
       int p1 = __shfl(pix, lane + 1), p2 = __shfl(pix, lane + 2),
@@ -806,17 +804,14 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
             (uint32_t)pix | ((uint32_t)p1 << 8) | ((uint32_t)p2 << 16) |
             ((uint32_t)p3 << 24);
       }
-      if (cc2 == 7) {
-        if (comp) s_left_cr[rr + 8] = (uint8_t)pix;
-        else s_left_cb[rr + 8] = (uint8_t)pix;
+      int cand = __shfl(pix, lane < 4 ? 19 + 4 * lane : 35 + 4 * lane);
+      if (lane < 8) {
+        if (pass == 0) ncb2 = cand;
+        else ncr2 = cand;
       }
     }
-    wave_lds_fence();
-    if (lane < 8) {
-      s_left_cb[lane] = s_left_cb[lane + 8];
-      s_left_cr[lane] = s_left_cr[lane + 8];
-    }
-    wave_lds_fence();
+    leftcb = ncb2;
+    leftcr = ncr2;
 
     if (lane == 0) M[1] = chroma_mode;   // m1 is unused for intra MBs
     have_left = true;
@@ -837,17 +832,12 @@ __global__ void __launch_bounds__(128) k_h264_rows(
   const int mby = job.mb_row;
   const int lane = threadIdx.x & 63;
 
-  __shared__ uint8_t s_left_y[32];       // [0..15] current, [16..31] staging
-  __shared__ uint8_t s_left_cb[16];      // [0..7] current, [8..15] staging
-  __shared__ uint8_t s_left_cr[16];
-
   if (threadIdx.x < 64) {
     luma_row(srcY, ypitch, w, h, refY, curY, mbw, mby, qp, i_slice, levels,
-             meta, lane, s_left_y);
+             meta, lane);
   } else {
     chroma_row(srcCb, srcCr, cpitch, w, h, refCb, refCr, curCb, curCr, mbw,
-               mby, dev_chroma_qp(qp), i_slice, levels, meta, lane,
-               s_left_cb, s_left_cr);
+               mby, dev_chroma_qp(qp), i_slice, levels, meta, lane);
   }
 }
 
