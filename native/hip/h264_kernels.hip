@@ -275,6 +275,36 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
   // trips and no fences.
   int leftreg = 0;
 
+  // The serial MB chain is latency-bound: issue the NEXT MB's loads
+  // (source pixels, zero-mv reference, meta) while computing the current
+  // one so global-memory latency overlaps the transform work.
+  auto load_psrc = [&](int mbx2) -> uint32_t {
+    const uint8_t* sr = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
+    int xb = mbx2 * 16 + cq;
+    return (uint32_t)sr[min(xb + 0, w - 1)] |
+           ((uint32_t)sr[min(xb + 1, w - 1)] << 8) |
+           ((uint32_t)sr[min(xb + 2, w - 1)] << 16) |
+           ((uint32_t)sr[min(xb + 3, w - 1)] << 24);
+  };
+  auto load_refp = [&](int mbx2) -> uint32_t {
+    const uint8_t* s2 = refY + (size_t)(y0 + r) * ypitch + mbx2 * 16 + cq;
+    return (uint32_t)s2[0] | ((uint32_t)s2[1] << 8) |
+           ((uint32_t)s2[2] << 16) | ((uint32_t)s2[3] << 24);
+  };
+  auto load_refcol = [&](int mbx2) -> int {
+    return lane < 16
+               ? refY[(size_t)(y0 + lane) * ypitch + mbx2 * 16 + 15]
+               : 0;
+  };
+  uint32_t psrc_cur = load_psrc(0);
+  uint32_t refp_cur = i_slice ? 0 : load_refp(0);
+  int refcol_cur = i_slice ? 0 : load_refcol(0);
+  int m0_cur = 0, m1_cur = 0;
+  if (!i_slice) {
+    m0_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 0];
+    m1_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 1];
+  }
+
   for (int mbx = 0; mbx < mbw; ++mbx) {
     const int x0 = mbx * 16;
     const size_t mb_index = (size_t)mby * mbw + mbx;
@@ -282,39 +312,41 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
                  mb_index * kLevelsPerMb;
     int* M = meta + mb_index * kMetaPerMb;
 
+    uint32_t psrc_nxt = 0, refp_nxt = 0;
+    int refcol_nxt = 0, m0_nxt = 0, m1_nxt = 0;
+    if (mbx + 1 < mbw) {
+      psrc_nxt = load_psrc(mbx + 1);
+      if (!i_slice) {
+        refp_nxt = load_refp(mbx + 1);
+        refcol_nxt = load_refcol(mbx + 1);
+        m0_nxt = M[kMetaPerMb + 0];
+        m1_nxt = M[kMetaPerMb + 1];
+      }
+    }
+
     int mode = kIntra, mvx = 0, mvy = 0;
     if (!i_slice) {
-      int m0 = M[0];
-      mode = m0 & 3;
-      int m1 = M[1];
-      mvx = (short)(m1 & 0xFFFF);
-      mvy = m1 >> 16;
+      mode = m0_cur & 3;
+      mvx = (short)(m1_cur & 0xFFFF);
+      mvy = m1_cur >> 16;
     }
 
     if (mode == kSkip) {
-      const uint8_t* s = refY + (size_t)(y0 + r) * ypitch + x0 + cq;
       uint8_t* d = curY + (size_t)(y0 + r) * ypitch + x0 + cq;
-      d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
-      if (lane < 16)
-        leftreg = refY[(size_t)(y0 + lane) * ypitch + x0 + 15];
+      *reinterpret_cast<uint32_t*>(d) = refp_cur;
+      leftreg = refcol_cur;
       have_left = true;
-      continue;
-    }
-
-    if (mode == kInter) {
+    } else if (mode == kInter) {
       // P_L0_16x16 with coded residual: full 16-coeff blocks, no DC
       // Hadamard, inter quant rounding (f = 2^qbits/6). cbp==0 falls out
       // naturally (idct of zeros is zero -> recon == MC pred).
       const int ix = mvx >> 2, iy = mvy >> 2;
       const int fx = mvx & 3, fy = mvy & 3;
-      uint32_t psrc, ppred;
+      uint32_t psrc = psrc_cur, ppred;
       {
-        const uint8_t* sr = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
-        psrc = (uint32_t)sr[min(x0 + cq + 0, w - 1)] |
-               ((uint32_t)sr[min(x0 + cq + 1, w - 1)] << 8) |
-               ((uint32_t)sr[min(x0 + cq + 2, w - 1)] << 16) |
-               ((uint32_t)sr[min(x0 + cq + 3, w - 1)] << 24);
-        if ((fx | fy) == 0) {
+        if ((fx | fy | ix | iy) == 0) {
+          ppred = refp_cur;
+        } else if ((fx | fy) == 0) {
           const uint8_t* pr =
               refY + (size_t)(y0 + iy + r) * ypitch + x0 + ix + cq;
           ppred = (uint32_t)pr[0] | ((uint32_t)pr[1] << 8) |
@@ -373,18 +405,9 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
       }
       leftreg = newleft;
       have_left = true;
-      continue;
-    }
-
-    // ---- source pixels: lane (r, cq..cq+3) as packed u32 (for shuffles)
-    uint32_t psrc;
-    {
-      const uint8_t* s = srcY + (size_t)min(y0 + r, h - 1) * ypitch;
-      psrc = (uint32_t)s[min(x0 + cq + 0, w - 1)] |
-             ((uint32_t)s[min(x0 + cq + 1, w - 1)] << 8) |
-             ((uint32_t)s[min(x0 + cq + 2, w - 1)] << 16) |
-             ((uint32_t)s[min(x0 + cq + 3, w - 1)] << 24);
-    }
+    } else {
+    // ---- source pixels come from the prefetched register
+    uint32_t psrc = psrc_cur;
     // DC value + mode costs
     int dcval = 128;
     if (have_left) {
@@ -477,6 +500,14 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
 
     if (lane == 0) M[0] = kIntra | (luma_mode << 2);
     have_left = true;
+    }
+
+    // rotate the prefetched state
+    psrc_cur = psrc_nxt;
+    refp_cur = refp_nxt;
+    refcol_cur = refcol_nxt;
+    m0_cur = m0_nxt;
+    m1_cur = m1_nxt;
   }
 }
 
@@ -499,6 +530,44 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
   // col 7 of the previous MB for each component (no LDS, no fences).
   int leftcb = 0, leftcr = 0;
 
+  // prefetch helpers (see luma_row): next MB's loads issue early so the
+  // serial chain overlaps global latency with compute
+  auto cload_src = [&](int mbx2) -> uint32_t {
+    if (lane >= 32) return 0;
+    int comp = lane >> 4;
+    int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
+    const uint8_t* sp = comp ? srcCr : srcCb;
+    int sr = min(cy0 + rr, chh - 1);
+    int xb = mbx2 * 8 + ccq;
+    return (uint32_t)sp[(size_t)sr * cpitch + min(xb + 0, cw - 1)] |
+           ((uint32_t)sp[(size_t)sr * cpitch + min(xb + 1, cw - 1)] << 8) |
+           ((uint32_t)sp[(size_t)sr * cpitch + min(xb + 2, cw - 1)] << 16) |
+           ((uint32_t)sp[(size_t)sr * cpitch + min(xb + 3, cw - 1)] << 24);
+  };
+  auto cload_ref = [&](int mbx2) -> uint32_t {
+    if (lane >= 32) return 0;
+    int comp = lane >> 4;
+    int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
+    const uint8_t* sp = comp ? refCr : refCb;
+    const uint8_t* s2 = sp + (size_t)(cy0 + rr) * cpitch + mbx2 * 8 + ccq;
+    return (uint32_t)s2[0] | ((uint32_t)s2[1] << 8) |
+           ((uint32_t)s2[2] << 16) | ((uint32_t)s2[3] << 24);
+  };
+  auto cload_col = [&](int mbx2, const uint8_t* plane) -> int {
+    return lane < 8
+               ? plane[(size_t)(cy0 + lane) * cpitch + mbx2 * 8 + 7]
+               : 0;
+  };
+  uint32_t csrc_cur = cload_src(0);
+  uint32_t cref_cur = i_slice ? 0 : cload_ref(0);
+  int colcb_cur = i_slice ? 0 : cload_col(0, refCb);
+  int colcr_cur = i_slice ? 0 : cload_col(0, refCr);
+  int m0_cur = 0, m1_cur = 0;
+  if (!i_slice) {
+    m0_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 0];
+    m1_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 1];
+  }
+
   for (int mbx = 0; mbx < mbw; ++mbx) {
     const int cx0 = mbx * 8;
     const size_t mb_index = (size_t)mby * mbw + mbx;
@@ -506,51 +575,48 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                  mb_index * kLevelsPerMb;
     int* M = meta + mb_index * kMetaPerMb;
 
+    uint32_t csrc_nxt = 0, cref_nxt = 0;
+    int colcb_nxt = 0, colcr_nxt = 0, m0_nxt = 0, m1_nxt = 0;
+    if (mbx + 1 < mbw) {
+      csrc_nxt = cload_src(mbx + 1);
+      if (!i_slice) {
+        cref_nxt = cload_ref(mbx + 1);
+        colcb_nxt = cload_col(mbx + 1, refCb);
+        colcr_nxt = cload_col(mbx + 1, refCr);
+        m0_nxt = M[kMetaPerMb + 0];
+        m1_nxt = M[kMetaPerMb + 1];
+      }
+    }
+
     int mode = kIntra, mvx = 0, mvy = 0;
     if (!i_slice) {
-      int m0 = M[0];
-      mode = m0 & 3;
-      int m1 = M[1];
-      mvx = (short)(m1 & 0xFFFF);
-      mvy = m1 >> 16;
+      mode = m0_cur & 3;
+      mvx = (short)(m1_cur & 0xFFFF);
+      mvy = m1_cur >> 16;
     }
 
     if (mode == kSkip) {
       if (lane < 32) {
         int comp = lane >> 4;
         int r = (lane & 15) >> 1, cq = (lane & 1) * 4;
-        const uint8_t* sp = comp ? refCr : refCb;
         uint8_t* dp = comp ? curCr : curCb;
-        const uint8_t* s = sp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
         uint8_t* d = dp + (size_t)(cy0 + r) * cpitch + cx0 + cq;
-        d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3];
+        *reinterpret_cast<uint32_t*>(d) = cref_cur;
       }
-      if (lane < 8) {
-        leftcb = refCb[(size_t)(cy0 + lane) * cpitch + cx0 + 7];
-        leftcr = refCr[(size_t)(cy0 + lane) * cpitch + cx0 + 7];
-      }
+      leftcb = colcb_cur;
+      leftcr = colcr_cur;
       have_left = true;
-      continue;
-    }
-
-    if (mode == kInter) {
+    } else if (mode == kInter) {
       // chroma inter residual: MC pred (integer, mv/2) + DC Hadamard + AC,
       // inter quant rounding. cbp falls out of the quantized levels.
       const int cix = mvx >> 3, ciy = mvy >> 3;
       const int cdx = mvx & 7, cdy = mvy & 7;
-      uint32_t csrc = 0, cprd = 0;
-      if (lane < 32) {
+      uint32_t csrc = csrc_cur, cprd = 0;
+      if ((cix | ciy | cdx | cdy) == 0) {
+        cprd = cref_cur;
+      } else if (lane < 32) {
         int comp = lane >> 4;
         int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
-        const uint8_t* sp = comp ? srcCr : srcCb;
-        int sr = min(cy0 + rr, chh - 1);
-        csrc = (uint32_t)sp[(size_t)sr * cpitch + min(cx0 + ccq + 0, cw - 1)] |
-               ((uint32_t)sp[(size_t)sr * cpitch +
-                             min(cx0 + ccq + 1, cw - 1)] << 8) |
-               ((uint32_t)sp[(size_t)sr * cpitch +
-                             min(cx0 + ccq + 2, cw - 1)] << 16) |
-               ((uint32_t)sp[(size_t)sr * cpitch +
-                             min(cx0 + ccq + 3, cw - 1)] << 24);
         const uint8_t* rp = comp ? refCr : refCb;
         for (int j = 0; j < 4; ++j)
           cprd |= (uint32_t)chroma_interp(rp, cpitch, cx0 + cix + ccq + j,
@@ -655,25 +721,9 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
       leftcb = ncb;
       leftcr = ncr;
       have_left = true;
-      continue;
-    }
-
-    // ---- source: loader lane (comp = lane>>4, row = (lane&15)>>1,
-    // 4 cols at (lane&1)*4) as packed u32; shuffles serve the passes.
-    uint32_t csrc = 0;
-    if (lane < 32) {
-      int comp = lane >> 4;
-      int rr = (lane & 15) >> 1, ccq = (lane & 1) * 4;
-      const uint8_t* sp = comp ? srcCr : srcCb;
-      int sr = min(cy0 + rr, chh - 1);
-      csrc = (uint32_t)sp[(size_t)sr * cpitch + min(cx0 + ccq + 0, cw - 1)] |
-             ((uint32_t)sp[(size_t)sr * cpitch +
-                           min(cx0 + ccq + 1, cw - 1)] << 8) |
-             ((uint32_t)sp[(size_t)sr * cpitch +
-                           min(cx0 + ccq + 2, cw - 1)] << 16) |
-             ((uint32_t)sp[(size_t)sr * cpitch +
-                           min(cx0 + ccq + 3, cw - 1)] << 24);
-    }
+    } else {
+    // source pixels come from the prefetched register
+    uint32_t csrc = csrc_cur;
     auto cpix_at = [&](int comp, int rr, int cc) -> int {
       uint32_t v = __shfl(csrc, (comp << 4) | (rr << 1) | (cc >> 2));
       return (v >> (8 * (cc & 3))) & 0xFF;
@@ -815,6 +865,15 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
 
     if (lane == 0) M[1] = chroma_mode;   // m1 is unused for intra MBs
     have_left = true;
+    }
+
+    // rotate the prefetched state
+    csrc_cur = csrc_nxt;
+    cref_cur = cref_nxt;
+    colcb_cur = colcb_nxt;
+    colcr_cur = colcr_nxt;
+    m0_cur = m0_nxt;
+    m1_cur = m1_nxt;
   }
 }
 
