@@ -13,6 +13,14 @@ void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
                     const RowJob* d_jobs, int* d_meta, hipStream_t stream,
                     bool use_mfma = true);
 
+void launch_h264_rows4(const uint8_t* srcY, const uint8_t* srcCb,
+                       const uint8_t* srcCr, int ypitch, int cpitch, int w,
+                       int h, const uint8_t* refY, const uint8_t* refCb,
+                       const uint8_t* refCr, uint8_t* curY, uint8_t* curCb,
+                       uint8_t* curCr, int mbw, int n_jobs,
+                       const struct RowJob* d_jobs, int16_t* d_levels,
+                       int* d_meta, hipStream_t stream);
+
 void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
                       const uint8_t* srcCr, int ypitch, int cpitch, int w,
                       int h, const uint8_t* refY, const uint8_t* refCb,

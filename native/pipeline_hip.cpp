@@ -349,10 +349,20 @@ class HipH264Pipeline : public EncodePipeline {
         static_cast<size_t>(mbw_) * h264gpu::kMetaPerMb * sizeof(int);
     for (size_t b = 0; b < batches.size(); ++b) {
       const Batch& bt = batches[b];
-      h264gpu::launch_h264_rows(
-          d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
-          d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
-          bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_, stream_);
+      // 4-wave row kernel by default (issue-rate-bound chain split across
+      // waves); HIPFLUX_ROWS_V1=1 falls back to the 2-wave variant
+      if (rows_v1_)
+        h264gpu::launch_h264_rows(
+            d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
+            d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
+            bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_,
+            stream_);
+      else
+        h264gpu::launch_h264_rows4(
+            d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
+            d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
+            bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_,
+            stream_);
       if (cpu_entropy_) {
         HIP_CHECK(hipMemcpyAsync(
             reinterpret_cast<uint8_t*>(h_levels_) + bt.row0 * lvl_row,
@@ -655,6 +665,7 @@ class HipH264Pipeline : public EncodePipeline {
   int* h_outbits_ = nullptr;
   int ent_stride_words_ = 0;
   int ent_copy_words_ = 1 << 30;   // adaptive D2H cap (words per row)
+  bool rows_v1_ = std::getenv("HIPFLUX_ROWS_V1") != nullptr;
   std::vector<hipEvent_t> batch_events_;
   std::vector<void*> device_ptrs_;
   std::vector<StripeState> stripes_;
