@@ -1,5 +1,7 @@
 #include "gpu_entropy.h"
 
+#include <cstring>
+
 #include "../../hip/h264_gpu_layout.h"
 #include "bitwriter.h"
 #include "cavlc.h"
@@ -272,30 +274,45 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
 
 void assemble_gpu_row_nal(const uint32_t* words, int bits, bool idr,
                           bool long_startcode, std::vector<uint8_t>& out) {
+  // big-endian unpack word-wise (the GPU packs MSB-first u32 words)
   int nbytes = (bits + 7) / 8;
-  std::vector<uint8_t> rbsp(nbytes + 1, 0);
-  for (int i = 0; i < nbytes; ++i)
-    rbsp[i] = static_cast<uint8_t>(words[i / 4] >> (24 - 8 * (i % 4)));
+  int nwords = nbytes / 4 + 1;
+  std::vector<uint32_t> swapped(nwords + 1, 0);
+  for (int i = 0; i < nwords; ++i)
+    swapped[i] = __builtin_bswap32(words[i]);
+  uint8_t* rbsp = reinterpret_cast<uint8_t*>(swapped.data());
   // rbsp stop bit (trailing bits beyond `bits` are zero by construction)
   int stop_byte = bits / 8;
   rbsp[stop_byte] |= 0x80 >> (bits % 8);
-  int total = stop_byte + 1;
+  size_t total = static_cast<size_t>(stop_byte) + 1;
 
+  out.reserve(out.size() + total + total / 64 + 8);
   if (long_startcode) out.push_back(0);
   out.push_back(0);
   out.push_back(0);
   out.push_back(1);
   out.push_back(static_cast<uint8_t>(
       idr ? ((3 << 5) | 5) : ((2 << 5) | 1)));
+  // emulation prevention: bulk-copy runs between zero bytes (memchr),
+  // handle the 00 00 0x escape rule only at the zeros themselves
+  size_t p = 0;
   int zeros = 0;
-  for (int i = 0; i < total; ++i) {
-    uint8_t b = rbsp[i];
-    if (zeros >= 2 && b <= 3) {
+  while (p < total) {
+    if (zeros >= 2 && rbsp[p] <= 3) {
       out.push_back(3);
       zeros = 0;
     }
-    out.push_back(b);
-    zeros = (b == 0) ? zeros + 1 : 0;
+    const void* z = memchr(rbsp + p, 0, total - p);
+    size_t zi = z ? static_cast<const uint8_t*>(z) - rbsp : total;
+    if (zi > p) {
+      out.insert(out.end(), rbsp + p, rbsp + zi);
+      zeros = 0;
+      p = zi;
+      continue;
+    }
+    out.push_back(0);   // rbsp[p] == 0
+    ++zeros;
+    ++p;
   }
 }
 
