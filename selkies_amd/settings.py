@@ -94,6 +94,18 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                "UDP port for the WebRTC ICE-lite endpoint (0 = ephemeral).",
                value_range=(0, 65535)),
     SettingDef("metrics_http_token", str, "", "Bearer token guarding /metrics."),
+    # TURN relay (coturn use-auth-secret scheme; reference settings.py:774-800)
+    SettingDef("turn_host", str, "", "TURN server hostname/IP for clients."),
+    SettingDef("turn_port", int, 3478, "TURN server port.",
+               value_range=(1, 65535)),
+    SettingDef("turn_shared_secret", str, "",
+               "coturn static-auth-secret for HMAC short-term credentials."),
+    SettingDef("turn_protocol", str, "udp", "TURN transport for clients.",
+               allowed=("udp", "tcp")),
+    SettingDef("turn_tls", bool, False, "Use turns: (TLS) TURN URLs."),
+    SettingDef("stun_host", str, "", "Extra STUN host to advertise first."),
+    SettingDef("stun_port", int, 3478, "Port for stun_host.",
+               value_range=(1, 65535)),
 
     # ---- display / capture ----
     SettingDef("display", str, ":0", "X DISPLAY to capture/inject into.",

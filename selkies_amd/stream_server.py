@@ -78,6 +78,7 @@ class CentralizedStreamServer:
         app.router.add_get("/api/stats", self.handle_stats)
         app.router.add_get("/api/settings", self.handle_settings)
         app.router.add_get("/metrics", self.handle_metrics)
+        app.router.add_get("/api/turn", self.handle_turn)
         from .computer_use import ComputerUseAPI
         ComputerUseAPI(self.settings, self.streaming.input).register(app)
         app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
@@ -281,6 +282,21 @@ class CentralizedStreamServer:
                     self._cert_mtimes = mt
                 except Exception as exc:
                     logger.error("cert reload failed: %r", exc)
+
+    async def handle_turn(self, request):
+        """RTC config with minted coturn HMAC credentials (reference
+        handle_turn_req, signaling_server.py:1030)."""
+        s = self.settings
+        if not (s.turn_shared_secret and s.turn_host):
+            return web.json_response({"error": "no TURN configured"},
+                                     status=404)
+        from .webrtc.turn import generate_rtc_config
+        cfg = generate_rtc_config(
+            s.turn_host, s.turn_port, s.turn_shared_secret,
+            user=request.query.get("user"), protocol=s.turn_protocol,
+            turn_tls=s.turn_tls, stun_host=s.stun_host or None,
+            stun_port=s.stun_port)
+        return web.json_response(cfg)
 
     # ---- lifecycle ----------------------------------------------------------
     async def start(self):

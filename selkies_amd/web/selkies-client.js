@@ -356,7 +356,15 @@ window.addEventListener("resize", () => {
  * channel. */
 let dcInput = null;            /* open RTCDataChannel, used by sendInput */
 async function connectWebRTC() {
-  const pc = new RTCPeerConnection();
+  let rtcCfg = {};
+  try {
+    const tr = await fetch("/api/turn" + location.search);
+    if (tr.ok) {
+      const cfg = await tr.json();
+      if (cfg.iceServers) rtcCfg.iceServers = cfg.iceServers;
+    }
+  } catch (e) { /* no TURN configured: host candidates only */ }
+  const pc = new RTCPeerConnection(rtcCfg);
   pc.addTransceiver("video", { direction: "recvonly" });
   pc.addTransceiver("audio", { direction: "recvonly" });
   const dc = pc.createDataChannel("input", { ordered: true });

@@ -192,3 +192,51 @@ def test_audio_broadcast(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_turn_endpoint_and_credentials():
+    """/api/turn mints coturn use-auth-secret credentials; 404 without
+    configuration."""
+    import asyncio
+    import base64
+    import hashlib
+    import hmac as hmac_mod
+    import aiohttp
+    from test_server import make_server, start_on_free_port
+
+    async def main():
+        srv = make_server()
+        runner, port = await start_on_free_port(srv)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                r = await sess.get(f"http://127.0.0.1:{port}/api/turn")
+                assert r.status == 404
+        finally:
+            await srv.stop()
+            await runner.cleanup()
+
+        srv = make_server(SELKIES_TURN_HOST="relay.example",
+                          SELKIES_TURN_PORT="3478",
+                          SELKIES_TURN_SHARED_SECRET="s3cret")
+        runner, port = await start_on_free_port(srv)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                r = await sess.get(
+                    f"http://127.0.0.1:{port}/api/turn?user=alice")
+                assert r.status == 200
+                cfg = await r.json()
+        finally:
+            await srv.stop()
+            await runner.cleanup()
+        turn = cfg["iceServers"][1]
+        assert turn["urls"] == ["turn:relay.example:3478?transport=udp"]
+        expiry, user = turn["username"].split(":")
+        assert user == "alice" and int(expiry) > 0
+        want = base64.b64encode(hmac_mod.new(
+            b"s3cret", turn["username"].encode(),
+            hashlib.sha1).digest()).decode()
+        assert turn["credential"] == want
+        assert any(u.startswith("stun:relay.example")
+                   for u in cfg["iceServers"][0]["urls"])
+
+    asyncio.new_event_loop().run_until_complete(main())
