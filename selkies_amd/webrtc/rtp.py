@@ -130,7 +130,22 @@ def parse_rtcp(data: bytes) -> list[dict]:
         size = 4 * (length + 1)
         chunk = data[off:off + size]
         if pt == 201:
-            out.append({"type": "RR"})
+            rep = {"type": "RR", "blocks": []}
+            rc = v_p_rc & 0x1F
+            boff = off + 8          # header + reporter SSRC
+            for _ in range(rc):
+                if boff + 24 > off + size:
+                    break
+                (ssrc, fl_cum, ext_seq, jitter, lsr,
+                 dlsr) = struct.unpack_from(">IIIIII", data, boff)
+                rep["blocks"].append({
+                    "ssrc": ssrc,
+                    "fraction_lost": (fl_cum >> 24) / 256.0,
+                    "cum_lost": fl_cum & 0xFFFFFF,
+                    "jitter": jitter,
+                })
+                boff += 24
+            out.append(rep)
         elif pt == 200:
             out.append({"type": "SR"})
         elif pt == 206:  # PSFB

@@ -232,6 +232,8 @@ class WebRTCService:
                     if report["type"] in ("PLI", "FIR"):
                         if self.capture:
                             self.capture.request_idr_frame()
+                    elif report["type"] == "RR" and report.get("blocks"):
+                        self._on_receiver_report(report["blocks"])
 
     # ---- data channel (SCTP over DTLS; webrtc/sctp.py) ---------------------
     def _pump_sctp(self, peer: PeerState):
@@ -273,6 +275,36 @@ class WebRTCService:
             self.streaming.input.on_message(text)
         except Exception:
             logger.debug("datachannel verb failed", exc_info=True)
+
+    # ---- loss-based congestion control (reference runs GCC/TWCC in
+    # webrtc_mode._congestion_control_loop; this is the loss-driven AIMD
+    # slice of it: back off multiplicatively above 2% loss, probe up
+    # additively when clean) ------------------------------------------------
+    LOSS_BACKOFF_AT = 0.02
+    BITRATE_MIN_KBPS = 1000
+    BITRATE_MAX_KBPS = 60000
+    PROBE_STEP_KBPS = 1000
+
+    def _on_receiver_report(self, blocks):
+        loss = max(b.get("fraction_lost", 0.0) for b in blocks)
+        now = time.monotonic()
+        if now - getattr(self, "_last_cc", 0.0) < 0.5:
+            return
+        self._last_cc = now
+        kbps = getattr(self, "_video_kbps",
+                       getattr(self.settings, "video_bitrate_kbps", 16000))
+        if loss > self.LOSS_BACKOFF_AT:
+            kbps = max(self.BITRATE_MIN_KBPS,
+                       int(kbps * (1.0 - 0.5 * min(1.0, loss * 4))))
+        else:
+            kbps = min(self.BITRATE_MAX_KBPS, kbps + self.PROBE_STEP_KBPS)
+        if kbps != getattr(self, "_video_kbps", None):
+            self._video_kbps = kbps
+            if self.capture is not None:
+                try:
+                    self.capture.update_video_bitrate(kbps)
+                except Exception:
+                    pass
 
     def dc_broadcast(self, text: str):
         """Send a control message to every open data channel."""
