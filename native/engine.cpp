@@ -1,5 +1,7 @@
 #include "engine.h"
 
+#include "cpu/scale.h"
+
 #include <algorithm>
 #include <chrono>
 #include <cstdio>
@@ -180,8 +182,11 @@ void ScreenCapture::run() {
   }
   pipeline_name_ = pipeline->name();
 
+  const int scale_div =
+      std::min(4, std::max(1, settings_.capture_scale_div));
   DamageTracker damage;
-  damage.reset(src->width(), src->height());
+  damage.reset(src->width() / scale_div, src->height() / scale_div);
+  std::vector<uint8_t> scale_buf;
 
   Watermark watermark;
   std::vector<uint8_t> wm_scratch;
@@ -263,6 +268,19 @@ void ScreenCapture::run() {
       last_cursor_serial = cursor.serial;
       cursor_cb_(cursor.width, cursor.height, cursor.hot_x, cursor.hot_y,
                  cursor.argb.data(), cursor.argb.size());
+    }
+
+    // integer box downscale (capture_scale_div): encode a w/div x h/div
+    // stream; cursor/watermark were composited at native size above
+    if (scale_div > 1) {
+      int ow, oh, ostride;
+      box_downscale_bgrx(frame.data, frame.stride, frame.width,
+                         frame.height, scale_div, scale_buf, ow, oh,
+                         ostride);
+      frame.data = scale_buf.data();
+      frame.width = ow;
+      frame.height = oh;
+      frame.stride = ostride;
     }
 
     // damage update (skipped in fullframe mode to save the diff cost)

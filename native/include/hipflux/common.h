@@ -27,6 +27,9 @@ struct CaptureSettings {
   bool capture_cursor = false;
 
   int output_mode = 1;              // 0 = JPEG, 1 = H.264
+  int capture_scale_div = 1;        // integer box-downscale (1..4): encode
+                                    // at capture/div (e.g. 4K capture -> 2
+                                    // -> 1080p stream)
   bool video_fullframe = false;     // disable damage gating
   bool use_cpu = false;             // force CPU encode path
   int gpu_id = 0;                   // HIP device ordinal (encode_node_index)

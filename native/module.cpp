@@ -32,6 +32,7 @@ PYBIND11_MODULE(_native, m) {
       .def_readwrite("target_fps", &CaptureSettings::target_fps)
       .def_readwrite("capture_cursor", &CaptureSettings::capture_cursor)
       .def_readwrite("output_mode", &CaptureSettings::output_mode)
+      .def_readwrite("capture_scale_div", &CaptureSettings::capture_scale_div)
       .def_readwrite("video_fullframe", &CaptureSettings::video_fullframe)
       .def_readwrite("use_cpu", &CaptureSettings::use_cpu)
       .def_readwrite("gpu_id", &CaptureSettings::gpu_id)

@@ -138,6 +138,7 @@ class StreamingService:
         cs.keyframe_interval_s = s.keyframe_interval_s
         cs.video_fullcolor = s.video_fullcolor
         cs.video_fullframe = s.video_fullframe
+        cs.capture_scale_div = s.capture_scale_div
         cs.use_paint_over_quality = s.use_paint_over_quality
         cs.paint_over_trigger_frames = s.paint_over_trigger_frames
         cs.video_paintover_crf = s.video_paintover_crf
