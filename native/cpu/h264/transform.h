@@ -144,7 +144,8 @@ inline int quant_coeff(int w, int qp, int cls, bool intra) {
 }
 
 inline int dequant_coeff(int level, int qp, int cls) {
-  return (level * kDequantV[qp % 6][cls]) << (qp / 6);
+  // multiply, not shift: level may be negative (<< on negatives is UB)
+  return level * kDequantV[qp % 6][cls] * (1 << (qp / 6));
 }
 
 // DC (Hadamard-domain) quant: double shift, doubled rounding.
@@ -161,14 +162,14 @@ inline int quant_dc(int w, int qp, bool intra) {
 // Luma DC dequant (after inverse Hadamard).
 inline int dequant_luma_dc(int c, int qp) {
   int v = kDequantV[qp % 6][0];
-  if (qp >= 12) return (c * v) << (qp / 6 - 2);
+  if (qp >= 12) return c * v * (1 << (qp / 6 - 2));
   return (c * v + (1 << (1 - qp / 6))) >> (2 - qp / 6);
 }
 
 // Chroma DC dequant (after inverse 2x2 Hadamard).
 inline int dequant_chroma_dc(int c, int qp) {
   int v = kDequantV[qp % 6][0];
-  if (qp >= 6) return (c * v) << (qp / 6 - 1);
+  if (qp >= 6) return c * v * (1 << (qp / 6 - 1));
   return (c * v) >> 1;
 }
 

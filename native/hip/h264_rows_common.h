@@ -51,18 +51,19 @@ __device__ inline int quant_dc_v(int w, int qp, bool intra = true) {
 }
 
 __device__ inline int dequant_c(int level, int qp, int cls) {
-  return (level * c_dequant_v[qp % 6][cls]) << (qp / 6);
+  // multiply, not shift: level may be negative (<< on negatives is UB)
+  return level * c_dequant_v[qp % 6][cls] * (1 << (qp / 6));
 }
 
 __device__ inline int dequant_luma_dc_v(int c, int qp) {
   int v = c_dequant_v[qp % 6][0];
-  if (qp >= 12) return (c * v) << (qp / 6 - 2);
+  if (qp >= 12) return c * v * (1 << (qp / 6 - 2));
   return (c * v + (1 << (1 - qp / 6))) >> (2 - qp / 6);
 }
 
 __device__ inline int dequant_chroma_dc_v(int c, int qp) {
   int v = c_dequant_v[qp % 6][0];
-  if (qp >= 6) return (c * v) << (qp / 6 - 1);
+  if (qp >= 6) return c * v * (1 << (qp / 6 - 1));
   return (c * v) >> 1;
 }
 

@@ -183,6 +183,17 @@ class CentralizedStreamServer:
         g2 = Gauge("selkies_last_encode_ms", "last frame encode ms",
                    registry=reg)
         g2.set(st["last_encode_ms"])
+        g3 = Gauge("selkies_transfer_rate_bytes", "file-transfer budget",
+                   registry=reg)
+        g3.set(self.transfers.pacer.rate)
+        if self.webrtc is not None:
+            ws = self.webrtc.stats()
+            g4 = Gauge("selkies_webrtc_peers", "connected WebRTC peers",
+                       registry=reg)
+            g4.set(ws.get("connected", 0))
+            g5 = Gauge("selkies_webrtc_video_kbps",
+                       "congestion-controlled video bitrate", registry=reg)
+            g5.set(getattr(self.webrtc, "_video_kbps", 0) or 0)
         for gpu in gpu_stats_snapshot():
             gg = Gauge(f"selkies_gpu_busy_percent_{gpu['card']}",
                        "amdgpu busy", registry=reg)

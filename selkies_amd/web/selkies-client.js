@@ -292,13 +292,25 @@ function connect() {
   ws = new WebSocket(`${proto}://${location.host}/websockets` +
                      location.search);
   ws.binaryType = "arraybuffer";
-  ws.onopen = () => { stateEl.textContent = "connected"; };
+
   ws.onclose = () => {
     stateEl.textContent = "reconnecting…";
     for (const row of h264Rows.values())
       if (row.decoder) try { row.decoder.close(); } catch (e) {}
     h264Rows.clear();
     setTimeout(connect, 1000);
+  };
+  ws.onopen = () => {
+    stateEl.textContent = "connected";
+    /* replay persisted client preferences (unlocked settings only:
+       the server sanitizes every proposal) */
+    try {
+      const saved = JSON.parse(
+          localStorage.getItem("selkies.prefs") || "{}");
+      for (const [k, v] of Object.entries(saved)) {
+        ws.send("SETTINGS," + JSON.stringify({ [k]: v }));
+      }
+    } catch (e) {}
   };
   ws.onmessage = (ev) => {
     if (typeof ev.data === "string") {
@@ -309,6 +321,12 @@ function connect() {
         try {
           const payload = JSON.parse(rest);
           applyServerSettings(payload);
+          /* persist client-changeable settings across sessions
+             (namespaced, quota-safe — reference selkies-core.js:14-40) */
+          try {
+            localStorage.setItem("selkies.settings",
+                                 JSON.stringify(payload));
+          } catch (e) { /* quota exceeded / privacy mode: skip */ }
           if (window.skOnSettings) window.skOnSettings(payload);
         } catch (e) {}
       } else if (verb === "SYSTEM_STATS") {
