@@ -106,6 +106,10 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("stun_host", str, "", "Extra STUN host to advertise first."),
     SettingDef("stun_port", int, 3478, "Port for stun_host.",
                value_range=(1, 65535)),
+    SettingDef("enable_webrtc_statistics", bool, False,
+               "Record client-reported WebRTC getStats payloads to CSV."),
+    SettingDef("webrtc_statistics_dir", str, "/tmp/selkies-webrtc-stats",
+               "Directory for the per-day WebRTC statistics CSV files."),
     SettingDef("capture_scale_div", int, 1,
                "Integer capture downscale: encode at capture/div "
                "(2 = 4K capture -> 1080p stream).",

@@ -42,6 +42,9 @@ class CentralizedStreamServer:
         from .transfers import UplinkAllowance
         self.allowance = UplinkAllowance(self.transfers.pacer)
         self._uplink_task = None
+        from concurrent.futures import ThreadPoolExecutor
+        # single worker => rows land in arrival order (reference contract)
+        self._stats_writer = ThreadPoolExecutor(max_workers=1)
         self._last_video_bytes = 0
         self.app = web.Application(middlewares=[self._auth_middleware])
         self.webrtc = None
@@ -79,6 +82,7 @@ class CentralizedStreamServer:
         app.router.add_get("/api/settings", self.handle_settings)
         app.router.add_get("/metrics", self.handle_metrics)
         app.router.add_get("/api/turn", self.handle_turn)
+        app.router.add_post("/api/webrtc-stats", self.handle_webrtc_stats)
         from .computer_use import ComputerUseAPI
         ComputerUseAPI(self.settings, self.streaming.input).register(app)
         app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
@@ -293,6 +297,59 @@ class CentralizedStreamServer:
                     self._cert_mtimes = mt
                 except Exception as exc:
                     logger.error("cert reload failed: %r", exc)
+
+    async def handle_webrtc_stats(self, request):
+        """Append client-reported getStats rows to a per-day CSV
+        (reference webrtc_utils.py:1201-1533: sanitized, ordered via a
+        dedicated single-worker pool)."""
+        s = self.settings
+        if not s.enable_webrtc_statistics:
+            return web.json_response({"error": "disabled"}, status=404)
+        try:
+            entries = await request.json()
+        except Exception:
+            raise web.HTTPBadRequest(text="invalid json")
+        if not isinstance(entries, list):
+            raise web.HTTPBadRequest(text="expected a list")
+        rows = []
+        for e in entries[:512]:
+            if not isinstance(e, dict):
+                continue
+            row = {}
+            for k, v in list(e.items())[:48]:
+                if not isinstance(k, str) or len(k) > 64:
+                    continue
+                if isinstance(v, bool):
+                    v = int(v)
+                if isinstance(v, (int, float)):
+                    row[k] = v
+                elif isinstance(v, str) and len(v) <= 128:
+                    # CSV-safe: no separators/newlines survive
+                    row[k] = v.replace(",", ";").replace("\n", " ")
+            if row:
+                rows.append(row)
+        if rows:
+            loop = asyncio.get_running_loop()
+            await loop.run_in_executor(self._stats_writer,
+                                       self._append_stats_rows, rows)
+        return web.json_response({"accepted": len(rows)})
+
+    def _append_stats_rows(self, rows):
+        import csv
+        import datetime
+        s = self.settings
+        os.makedirs(s.webrtc_statistics_dir, exist_ok=True)
+        day = datetime.date.today().isoformat()
+        path = os.path.join(s.webrtc_statistics_dir, f"webrtc-{day}.csv")
+        keys = sorted({k for r in rows for k in r})
+        new = not os.path.exists(path)
+        with open(path, "a", newline="") as f:
+            w = csv.writer(f)
+            if new:
+                w.writerow(["ts"] + keys)
+            ts = time.time()
+            for r in rows:
+                w.writerow([f"{ts:.3f}"] + [r.get(k, "") for k in keys])
 
     async def handle_turn(self, request):
         """RTC config with minted coturn HMAC credentials (reference

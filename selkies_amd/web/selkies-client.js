@@ -418,6 +418,25 @@ async function connectWebRTC() {
   });
   const ans = await r.json();
   await pc.setRemoteDescription({ type: "answer", sdp: ans.sdp });
+  /* opt-in stats recording: flatten getStats and ship to the server */
+  setInterval(async () => {
+    try {
+      const stats = [];
+      (await pc.getStats()).forEach((report) => {
+        const row = { id: report.id, type: report.type };
+        for (const [k, v] of Object.entries(report)) {
+          if (typeof v === "number" || typeof v === "string") row[k] = v;
+        }
+        stats.push(row);
+      });
+      if (stats.length)
+        fetch("/api/webrtc-stats" + location.search, {
+          method: "POST",
+          headers: { "Content-Type": "application/json" },
+          body: JSON.stringify(stats),
+        }).catch(() => {});
+    } catch (e) {}
+  }, 5000);
 }
 
 hookInput();

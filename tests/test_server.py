@@ -240,3 +240,51 @@ def test_turn_endpoint_and_credentials():
                    for u in cfg["iceServers"][0]["urls"])
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_webrtc_stats_recorder(tmp_path):
+    """POSTed getStats rows land sanitized + ordered in the per-day CSV;
+    disabled by default (404)."""
+    import asyncio
+    import aiohttp
+    from test_server import make_server, start_on_free_port
+
+    async def main():
+        srv = make_server()
+        runner, port = await start_on_free_port(srv)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                r = await sess.post(
+                    f"http://127.0.0.1:{port}/api/webrtc-stats", json=[])
+                assert r.status == 404
+        finally:
+            await srv.stop()
+            await runner.cleanup()
+
+        srv = make_server(SELKIES_ENABLE_WEBRTC_STATISTICS="true",
+                          SELKIES_WEBRTC_STATISTICS_DIR=str(tmp_path))
+        runner, port = await start_on_free_port(srv)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                rows = [{"id": "t1", "type": "inbound-rtp",
+                         "framesDecoded": 42, "bad,key" * 40: "x",
+                         "note": "a,b\nc"},
+                        {"id": "t2", "type": "candidate-pair",
+                         "rtt": 0.012}]
+                r = await sess.post(
+                    f"http://127.0.0.1:{port}/api/webrtc-stats", json=rows)
+                assert r.status == 200
+                assert (await r.json())["accepted"] == 2
+        finally:
+            await srv.stop()
+            await runner.cleanup()
+        files = list(tmp_path.glob("webrtc-*.csv"))
+        assert len(files) == 1
+        lines = files[0].read_text().splitlines()
+        assert lines[0].startswith("ts,")
+        assert len(lines) == 3
+        assert "framesDecoded" in lines[0]
+        assert "a;b c" in lines[1]          # sanitized separators
+        assert "bad,key" not in lines[0]    # oversized key dropped
+
+    asyncio.new_event_loop().run_until_complete(main())
