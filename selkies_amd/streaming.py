@@ -87,6 +87,7 @@ class StreamingService:
         )
         self._frame_clock = 0
         self._stats_task: Optional[asyncio.Task] = None
+        self._watchdog_task: Optional[asyncio.Task] = None
         self.frames_relayed = 0
         self.audio: Optional[object] = None
         self._audio_queue: Optional[asyncio.Queue] = None
@@ -349,6 +350,9 @@ class StreamingService:
             if self._stats_task is None or self._stats_task.done():
                 self._stats_task = asyncio.get_running_loop().create_task(
                     self._stats_pusher())
+            if self._watchdog_task is None or self._watchdog_task.done():
+                self._watchdog_task = asyncio.get_running_loop().create_task(
+                    self._capture_watchdog())
 
             async for msg in ws:
                 if msg.type == WSMsgType.TEXT:
@@ -518,6 +522,28 @@ class StreamingService:
                 except Exception:
                     pass
             await asyncio.sleep(2.0)
+
+    async def _capture_watchdog(self, interval: float = 5.0):
+        """Rebuild stale capture instances (reference behavior: stale
+        captures detected via is_capturing and rebuilt, selkies.py:
+        5167-5190). A capture thread that died (encoder fault, display
+        loss) is restarted as long as clients are connected."""
+        while True:
+            await asyncio.sleep(interval)
+            if not self.clients:
+                continue
+            displays = {c.display for c in self.clients.values()
+                        if c.display != "none"}
+            for d in displays:
+                cap = self.captures.get(d)
+                if cap is not None and not cap.is_capturing:
+                    logger.warning("capture for %s died; rebuilding", d)
+                    self.stop_capture(d)
+                    try:
+                        self.start_capture(d)
+                        self.request_idr(d)
+                    except Exception:
+                        logger.exception("capture rebuild failed for %s", d)
 
     # ---- stats --------------------------------------------------------------
     def stats(self) -> dict:

@@ -195,3 +195,46 @@ def test_capture_scale_div_halves_stream(tmp_path):
     r = enc.encode(np.ascontiguousarray(expect).tobytes(), qp=14, idr=True)
     y_direct = Decoder().decode(r["data"])[0][0]
     assert np.array_equal(y_dec, y_direct),         "engine scale path diverges from the exact box average"
+
+
+def test_capture_watchdog_rebuilds_dead_capture():
+    """A capture whose native thread died is rebuilt by the watchdog
+    while clients are connected (reference selkies.py:5167-5190)."""
+    import asyncio
+    from selkies_amd.settings import load_settings
+    from selkies_amd.streaming import StreamingService
+
+    async def main():
+        s = load_settings(argv=[], env={
+            "SELKIES_CAPTURE_BACKEND": "synthetic:static",
+            "SELKIES_RESOLUTION": "128x64",
+            "SELKIES_USE_CPU": "true",
+            "SELKIES_ENABLE_AUDIO": "false",
+        })
+        svc = StreamingService(s)
+        svc.loop = asyncio.get_running_loop()
+
+        class FakeClient:
+            display = "primary"
+
+        svc.clients["x"] = FakeClient()
+        svc.start_capture("primary")
+        cap = svc.captures["primary"]
+        assert cap.is_capturing
+        # kill the capture out from under the service
+        cap.stop_capture()
+        assert not cap.is_capturing
+        task = asyncio.get_running_loop().create_task(
+            svc._capture_watchdog(interval=0.1))
+        for _ in range(40):
+            await asyncio.sleep(0.1)
+            c2 = svc.captures.get("primary")
+            if c2 is not None and c2.is_capturing:
+                break
+        task.cancel()
+        c2 = svc.captures.get("primary")
+        alive = c2 is not None and c2.is_capturing
+        svc.stop_capture()
+        assert alive, "watchdog did not rebuild the dead capture"
+
+    asyncio.new_event_loop().run_until_complete(main())
