@@ -26,6 +26,31 @@ namespace h264gpu {
 
 __constant__ int c_zig4[16] = {0, 1, 4, 8, 5, 2, 3, 6,
                                9, 12, 13, 10, 7, 11, 14, 15};
+// compile-time copy of the zigzag so unrolled loops index registers,
+// not memory (a __constant__ lookup defeats the register promotion)
+__device__ constexpr int kZig4[16] = {0, 1, 4, 8, 5, 2, 3, 6,
+                                      9, 12, 13, 10, 7, 11, 14, 15};
+
+// one 4x4 block's levels via two int4 loads; zigzag applied with
+// compile-time indices so everything stays in registers. `first` = 1
+// drops position 0 (AC-only blocks).
+template <int FIRST>
+__device__ __forceinline__ void load_zz16(const int16_t* L, int off,
+                                          int* zz) {
+  union U {
+    int4 v;
+    short s[8];
+  } a, b;
+  a.v = *reinterpret_cast<const int4*>(L + off);
+  b.v = *reinterpret_cast<const int4*>(L + off + 8);
+#pragma unroll
+  for (int i = FIRST; i < 16; ++i) {
+    constexpr int kZ[16] = {0, 1, 4, 8, 5, 2, 3, 6,
+                            9, 12, 13, 10, 7, 11, 14, 15};
+    int pos = kZ[i];
+    zz[i - FIRST] = pos < 8 ? a.s[pos] : b.s[pos - 8];
+  }
+}
 
 // Z-order block -> raster index within the MB
 __constant__ int c_zorder_raster[16] = {0, 1, 4, 5, 2, 3, 6, 7,
@@ -394,8 +419,7 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
         // luma DC (intra only; inter has no DC Hadamard)
         if (mode == kIntra) {
           int zz[16];
-          for (int i = 0; i < 16; ++i)
-            zz[i] = L[kLumaDcOff + c_zig4[i]];
+          load_zz16<0>(L, kLumaDcOff, zz);
           int nC = lds_luma_nc(s_mb, mb, 0, 0);
           dev_cavlc_residual<16>(bw, zz, nC);
           bits = bw.flush();
@@ -408,13 +432,11 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
           int nC = lds_luma_nc(s_mb, mb, bx, by);
           if (mode == kIntra) {
             int zz[15];
-            for (int i = 1; i < 16; ++i)
-              zz[i - 1] = L[kLumaAcOff + r * 16 + c_zig4[i]];
+            load_zz16<1>(L, kLumaAcOff + r * 16, zz);
             dev_cavlc_residual<15>(bw, zz, nC);
           } else {
             int zz[16];
-            for (int i = 0; i < 16; ++i)
-              zz[i] = L[kLumaAcOff + r * 16 + c_zig4[i]];
+            load_zz16<0>(L, kLumaAcOff + r * 16, zz);
             dev_cavlc_residual<16>(bw, zz, nC);
           }
           bits = bw.flush();
@@ -432,8 +454,7 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
           int b = slot - 20;                       // cb0..3 then cr0..3
           int comp = b >> 2, sub = b & 3;
           int zz[15];
-          for (int i = 1; i < 16; ++i)
-            zz[i - 1] = L[kChromaAcOff + (comp * 4 + sub) * 16 + c_zig4[i]];
+          load_zz16<1>(L, kChromaAcOff + (comp * 4 + sub) * 16, zz);
           int nC = lds_chroma_nc(s_mb, mb, comp, sub & 1, sub >> 1);
           dev_cavlc_residual<15>(bw, zz, nC);
           bits = bw.flush();
