@@ -37,7 +37,8 @@ struct Shared {
   uint8_t ccol[2][16];
 };
 
-// ---- luma: wave w handles passes {2w, 2w+1} -------------------------------
+// ---- luma: wave w handles NP consecutive passes starting at w*NP ---------
+template <int NP>
 __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
                           int w_, int h, const uint8_t* __restrict__ refY,
                           uint8_t* __restrict__ curY, int mbw, int mby,
@@ -72,8 +73,8 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
     if (mode == kSkip) {
       // both waves cover all 16 rows: wave w copies rows where r parity
       // is irrelevant — split by halves: wave 0 rows 0..7, wave 1 8..15
-      if (lane < 32) {
-        int rr = w * 8 + (lane >> 2);
+      if (lane < 16 * NP) {
+        int rr = w * 4 * NP + (lane >> 2);   // wave covers 4*NP rows
         int qq = (lane & 3) * 4;
         const uint8_t* s = refY + (size_t)(y0 + rr) * ypitch + x0 + qq;
         uint8_t* d = curY + (size_t)(y0 + rr) * ypitch + x0 + qq;
@@ -81,8 +82,8 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
             ((uint32_t)s[1] << 8) | ((uint32_t)s[2] << 16) |
             ((uint32_t)s[3] << 24);
       }
-      if (lane < 8) {
-        int row = w * 8 + lane;
+      if (lane < 4 * NP) {
+        int row = w * 4 * NP + lane;
         lcol_n[row] = refY[(size_t)(y0 + row) * ypitch + x0 + 15];
       }
       __syncthreads();
@@ -120,10 +121,10 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
         uint32_t v = __shfl(ppred, py * 4 + (px >> 2));
         return (v >> (8 * (px & 3))) & 0xFF;
       };
-      int lvl_p[2];
+      int lvl_p[NP];
 #pragma unroll
-      for (int pi = 0; pi < 2; ++pi) {
-        int pass = 2 * w + pi;
+      for (int pi = 0; pi < NP; ++pi) {
+        int pass = NP * w + pi;
         int blk = pass * 4 + g;
         int bx = blk & 3, by = blk >> 2;
         int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
@@ -136,8 +137,8 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
       }
       // recon: no cbp gate needed — zero levels dequantize to zero
 #pragma unroll
-      for (int pi = 0; pi < 2; ++pi) {
-        int pass = 2 * w + pi;
+      for (int pi = 0; pi < NP; ++pi) {
+        int pass = NP * w + pi;
         int blk = pass * 4 + g;
         int bx = blk & 3, by = blk >> 2;
         int d = dequant_c(lvl_p[pi], qp, coeff_cls(c));
@@ -189,10 +190,10 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
       return (v >> (8 * (px & 3))) & 0xFF;
     };
 
-    int lvl_p[2];
+    int lvl_p[NP];
 #pragma unroll
-    for (int pi = 0; pi < 2; ++pi) {
-      int pass = 2 * w + pi;
+    for (int pi = 0; pi < NP; ++pi) {
+      int pass = NP * w + pi;
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
@@ -218,8 +219,8 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
 
     // recon (AC gate dropped: zero levels dequantize to zero)
 #pragma unroll
-    for (int pi = 0; pi < 2; ++pi) {
-      int pass = 2 * w + pi;
+    for (int pi = 0; pi < NP; ++pi) {
+      int pass = NP * w + pi;
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
       int dcb = __shfl(dcrec, blk);
@@ -402,7 +403,7 @@ __device__ void chroma_wave(const uint8_t* __restrict__ srcC,
 
 }  // namespace rows4
 
-__global__ void __launch_bounds__(256) k_h264_rows4(
+__global__ void __launch_bounds__(384) k_h264_rows4(
     const uint8_t* __restrict__ srcY, const uint8_t* __restrict__ srcCb,
     const uint8_t* __restrict__ srcCr, int ypitch, int cpitch, int w, int h,
     const uint8_t* __restrict__ refY, const uint8_t* __restrict__ refCb,
@@ -419,10 +420,10 @@ __global__ void __launch_bounds__(256) k_h264_rows4(
 
   __shared__ rows4::Shared sh;
 
-  if (wave < 2) {
-    rows4::luma_wave(srcY, ypitch, w, h, refY, curY, mbw, mby, qp, i_slice,
-                     levels, meta, lane, wave, &sh);
-  } else if (wave == 2) {
+  if (wave < 4) {
+    rows4::luma_wave<1>(srcY, ypitch, w, h, refY, curY, mbw, mby, qp,
+                        i_slice, levels, meta, lane, wave, &sh);
+  } else if (wave == 4) {
     rows4::chroma_wave(srcCb, refCb, curCb, cpitch, w, h, mbw, mby,
                        dev_chroma_qp(qp), i_slice, levels, meta, lane, 0,
                        &sh);
@@ -441,7 +442,7 @@ void launch_h264_rows4(const uint8_t* srcY, const uint8_t* srcCb,
                        const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
                        hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_rows4, dim3(n_jobs), dim3(256), 0, stream, srcY,
+  hipLaunchKernelGGL(k_h264_rows4, dim3(n_jobs), dim3(384), 0, stream, srcY,
                      srcCb, srcCr, ypitch, cpitch, w, h, refY, refCb, refCr,
                      curY, curCb, curCr, mbw, d_jobs, d_levels, d_meta);
 }
