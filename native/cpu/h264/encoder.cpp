@@ -675,10 +675,11 @@ struct StripeEncoder::Impl {
     }
   }
 
-  // ---- sub-pel motion compensation (8.4.2.2.1 subset) ---------------------
-  // MVs live on the HALF-pel grid (quarter-pel units, mv % 2 == 0): luma
-  // uses the 6-tap (1,-5,20,20,-5,1) filter at half positions, chroma the
-  // spec bilinear with eighth-pel weights (mv & 7 in {0,2,4,6}).
+  // ---- sub-pel motion compensation (8.4.2.2.1/8.4.2.2.2) ------------------
+  // MVs live on the full QUARTER-pel grid: half samples from the 6-tap
+  // (1,-5,20,20,-5,1) filter, quarter samples as the spec's rounded
+  // averages of the two nearest half/integer samples (Table 8-12);
+  // chroma uses the bilinear with eighth-pel weights.
   static inline int tap6(int a, int b, int c, int d, int e, int f) {
     return a - 5 * b + 20 * c + 20 * d - 5 * e + f;
   }
@@ -689,21 +690,55 @@ struct StripeEncoder::Impl {
     return tap6(r[x - 2], r[x - 1], r[x], r[x + 1], r[x + 2], r[x + 3]);
   }
 
-  // predicted luma sample at integer base (x,y) with frac (fx,fy) in {0,2}
-  inline int luma_pred_px(const Plane& p, int x, int y, int fx,
-                          int fy) const {
-    if (fx == 0 && fy == 0) return p.row(y)[x];
-    if (fy == 0) return clip8((hsum6(p, x, y) + 16) >> 5);
-    if (fx == 0) {
-      int v = tap6(p.row(y - 2)[x], p.row(y - 1)[x], p.row(y)[x],
-                   p.row(y + 1)[x], p.row(y + 2)[x], p.row(y + 3)[x]);
-      return clip8((v + 16) >> 5);
-    }
-    // j position: vertical 6-tap over unclipped horizontal sums
+  // half samples at integer base (x,y)
+  inline int half_b(const Plane& p, int x, int y) const {   // (x+1/2, y)
+    return clip8((hsum6(p, x, y) + 16) >> 5);
+  }
+  inline int half_h(const Plane& p, int x, int y) const {   // (x, y+1/2)
+    int v = tap6(p.row(y - 2)[x], p.row(y - 1)[x], p.row(y)[x],
+                 p.row(y + 1)[x], p.row(y + 2)[x], p.row(y + 3)[x]);
+    return clip8((v + 16) >> 5);
+  }
+  inline int half_j(const Plane& p, int x, int y) const {   // (x+1/2,y+1/2)
     int v = tap6(hsum6(p, x, y - 2), hsum6(p, x, y - 1), hsum6(p, x, y),
                  hsum6(p, x, y + 1), hsum6(p, x, y + 2),
                  hsum6(p, x, y + 3));
     return clip8((v + 512) >> 10);
+  }
+
+  // predicted luma sample at integer base (x,y), frac (fx,fy) in 0..3
+  // (Table 8-12: quarter samples average the two nearest half/integer
+  // samples with upward rounding)
+  inline int luma_pred_px(const Plane& p, int x, int y, int fx,
+                          int fy) const {
+    if (fx == 0 && fy == 0) return p.row(y)[x];
+    if (fy == 0) {
+      if (fx == 2) return half_b(p, x, y);
+      int b = half_b(p, x, y);
+      int g = fx == 1 ? p.row(y)[x] : p.row(y)[x + 1];
+      return (g + b + 1) >> 1;
+    }
+    if (fx == 0) {
+      if (fy == 2) return half_h(p, x, y);
+      int hh = half_h(p, x, y);
+      int g = fy == 1 ? p.row(y)[x] : p.row(y + 1)[x];
+      return (g + hh + 1) >> 1;
+    }
+    if (fx == 2 && fy == 2) return half_j(p, x, y);
+    if (fx == 2) {            // f (fy=1) / q (fy=3): avg of b and j
+      int j = half_j(p, x, y);
+      int b = fy == 1 ? half_b(p, x, y) : half_b(p, x, y + 1);
+      return (b + j + 1) >> 1;
+    }
+    if (fy == 2) {            // i (fx=1) / k (fx=3): avg of h and j
+      int j = half_j(p, x, y);
+      int hh = fx == 1 ? half_h(p, x, y) : half_h(p, x + 1, y);
+      return (hh + j + 1) >> 1;
+    }
+    // diagonal quarters: avg of nearest b (above/below) and h (left/right)
+    int b = fy == 1 ? half_b(p, x, y) : half_b(p, x, y + 1);
+    int hh = fx == 1 ? half_h(p, x, y) : half_h(p, x + 1, y);
+    return (b + hh + 1) >> 1;
   }
 
   // predicted chroma sample: integer base (cx,cy), eighth-pel (dx,dy)
@@ -721,8 +756,9 @@ struct StripeEncoder::Impl {
   // to integer candidates too so refinements around them stay legal).
   inline bool mv_window_ok(int x0, int y0, int mvq_x, int mvq_y) const {
     int ix = mvq_x >> 2, iy = mvq_y >> 2;
-    return x0 + ix - 2 >= 0 && x0 + ix + 19 <= yw && y0 + iy - 2 >= 0 &&
-           y0 + iy + 19 <= yh;
+    // +20: quarter positions may also read the x+1 / y+1 half sample
+    return x0 + ix - 2 >= 0 && x0 + ix + 20 <= yw && y0 + iy - 2 >= 0 &&
+           y0 + iy + 20 <= yh;
   }
 
   // interpolated-SAD of the 16x16 at (x0,y0) for quarter-pel mv
@@ -870,19 +906,21 @@ struct StripeEncoder::Impl {
           }
           if (!improved) break;
         }
-        // half-pel refinement (quarter-pel units, grid step 2); skipped
-        // when the MB is headed to intra regardless
+        // half- then quarter-pel refinement rings; skipped when the MB
+        // is headed to intra regardless
         int best_q_x = best_mvx * 4, best_q_y = best_mvy * 4;
         if (best <= 2 * inter_thresh) {
-          int cqx = best_q_x, cqy = best_q_y;
-          for (auto& p : pat) {
-            int qx = cqx + 2 * p[0], qy = cqy + 2 * p[1];
-            if (!mv_window_ok(x0, y0, qx, qy)) continue;
-            long s = sad16_q(x0, y0, qx, qy);
-            if (s < best) {
-              best = s;
-              best_q_x = qx;
-              best_q_y = qy;
+          for (int step = 2; step >= 1; --step) {
+            int cqx = best_q_x, cqy = best_q_y;
+            for (auto& p : pat) {
+              int qx = cqx + step * p[0], qy = cqy + step * p[1];
+              if (!mv_window_ok(x0, y0, qx, qy)) continue;
+              long s = sad16_q(x0, y0, qx, qy);
+              if (s < best) {
+                best = s;
+                best_q_x = qx;
+                best_q_y = qy;
+              }
             }
           }
         }

@@ -721,8 +721,9 @@ __global__ void __launch_bounds__(64) k_h264_me(
   // the full interpolation window must stay inside frame + stripe
   auto window_ok = [&](int qx, int qy) -> bool {
     int ixq = qx >> 2, iyq = qy >> 2;
-    return x0 + ixq - 2 >= 0 && x0 + ixq + 19 <= frame_w_mb16 &&
-           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 19 <= job.stripe_y1;
+    // +20: quarter positions may also read the x+1 / y+1 half sample
+    return x0 + ixq - 2 >= 0 && x0 + ixq + 20 <= frame_w_mb16 &&
+           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 20 <= job.stripe_y1;
   };
 
   const int qp = job.qp;
@@ -756,20 +757,22 @@ __global__ void __launch_bounds__(64) k_h264_me(
       }
       if (!improved) break;
     }
-    // half-pel refinement ring (skipped when the MB is headed to intra
-    // regardless: refinement cannot halve a hopeless SAD)
+    // half- then quarter-pel refinement rings (skipped when the MB is
+    // headed to intra regardless)
     bqx = bmx * 4;
     bqy = bmy * 4;
     if (best <= 2 * inter_thresh) {
-      int cqx = bqx, cqy = bqy;
-      for (int pi = 0; pi < 8; ++pi) {
-        int qx = cqx + 2 * pat[pi][0], qy = cqy + 2 * pat[pi][1];
-        if (!window_ok(qx, qy)) continue;
-        int s = sad_q(qx, qy);
-        if (s < best) {
-          best = s;
-          bqx = qx;
-          bqy = qy;
+      for (int step = 2; step >= 1; --step) {
+        int cqx = bqx, cqy = bqy;
+        for (int pi = 0; pi < 8; ++pi) {
+          int qx = cqx + step * pat[pi][0], qy = cqy + step * pat[pi][1];
+          if (!window_ok(qx, qy)) continue;
+          int s = sad_q(qx, qy);
+          if (s < best) {
+            best = s;
+            bqx = qx;
+            bqy = qy;
+          }
         }
       }
     }
@@ -954,8 +957,9 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   };
   auto window_ok = [&](int qx, int qy) -> bool {
     int ixq = qx >> 2, iyq = qy >> 2;
-    return x0 + ixq - 2 >= 0 && x0 + ixq + 19 <= frame_w_mb16 &&
-           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 19 <= job.stripe_y1;
+    // +20: quarter positions may also read the x+1 / y+1 half sample
+    return x0 + ixq - 2 >= 0 && x0 + ixq + 20 <= frame_w_mb16 &&
+           y0 + iyq - 2 >= job.stripe_y0 && y0 + iyq + 20 <= job.stripe_y1;
   };
 
   const int qp = job.qp;
@@ -985,15 +989,17 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
           bqy = qy;
         }
       }
-      int cqx2 = bqx, cqy2 = bqy;
-      for (int pi = 0; pi < 8; ++pi) {
-        int qx = cqx2 + 2 * pat[pi][0], qy = cqy2 + 2 * pat[pi][1];
-        if (!window_ok(qx, qy)) continue;
-        int s = sad_q(qx, qy);
-        if (s < best_sad) {
-          best_sad = s;
-          bqx = qx;
-          bqy = qy;
+      for (int step = 2; step >= 1; --step) {
+        int cqx2 = bqx, cqy2 = bqy;
+        for (int pi = 0; pi < 8; ++pi) {
+          int qx = cqx2 + step * pat[pi][0], qy = cqy2 + step * pat[pi][1];
+          if (!window_ok(qx, qy)) continue;
+          int s = sad_q(qx, qy);
+          if (s < best_sad) {
+            best_sad = s;
+            bqx = qx;
+            bqy = qy;
+          }
         }
       }
     }

@@ -80,17 +80,19 @@ __device__ inline int hsum6g(const uint8_t* row, int x) {
                row[x + 3]);
 }
 
-// predicted luma sample at integer (x,y) with frac (fx,fy) in {0,2}
-__device__ inline int luma_interp(const uint8_t* p, int pitch, int x, int y,
-                                  int fx, int fy) {
-  if ((fx | fy) == 0) return p[(size_t)y * pitch + x];
-  if (fy == 0) return clip8((hsum6g(p + (size_t)y * pitch, x) + 16) >> 5);
-  if (fx == 0) {
-    const uint8_t* c = p + (size_t)(y - 2) * pitch + x;
-    int v = tap6i(c[0], c[pitch], c[2 * pitch], c[3 * pitch], c[4 * pitch],
-                  c[5 * pitch]);
-    return clip8((v + 16) >> 5);
-  }
+// half samples at integer base (x,y)
+__device__ inline int half_b(const uint8_t* p, int pitch, int x, int y) {
+  return clip8((hsum6g(p + (size_t)y * pitch, x) + 16) >> 5);
+}
+
+__device__ inline int half_h(const uint8_t* p, int pitch, int x, int y) {
+  const uint8_t* c = p + (size_t)(y - 2) * pitch + x;
+  int v = tap6i(c[0], c[pitch], c[2 * pitch], c[3 * pitch], c[4 * pitch],
+                c[5 * pitch]);
+  return clip8((v + 16) >> 5);
+}
+
+__device__ inline int half_j(const uint8_t* p, int pitch, int x, int y) {
   int v = tap6i(hsum6g(p + (size_t)(y - 2) * pitch, x),
                 hsum6g(p + (size_t)(y - 1) * pitch, x),
                 hsum6g(p + (size_t)y * pitch, x),
@@ -98,6 +100,39 @@ __device__ inline int luma_interp(const uint8_t* p, int pitch, int x, int y,
                 hsum6g(p + (size_t)(y + 2) * pitch, x),
                 hsum6g(p + (size_t)(y + 3) * pitch, x));
   return clip8((v + 512) >> 10);
+}
+
+// predicted luma sample at integer (x,y), frac (fx,fy) in 0..3: 6-tap
+// half samples (8.4.2.2.1) + rounded-average quarters (Table 8-12)
+__device__ inline int luma_interp(const uint8_t* p, int pitch, int x, int y,
+                                  int fx, int fy) {
+  if ((fx | fy) == 0) return p[(size_t)y * pitch + x];
+  if (fy == 0) {
+    int b = half_b(p, pitch, x, y);
+    if (fx == 2) return b;
+    int g = p[(size_t)y * pitch + x + (fx == 1 ? 0 : 1)];
+    return (g + b + 1) >> 1;
+  }
+  if (fx == 0) {
+    int hh = half_h(p, pitch, x, y);
+    if (fy == 2) return hh;
+    int g = p[(size_t)(y + (fy == 1 ? 0 : 1)) * pitch + x];
+    return (g + hh + 1) >> 1;
+  }
+  if (fx == 2 && fy == 2) return half_j(p, pitch, x, y);
+  if (fx == 2) {                  // f / q: avg of b (above/below) and j
+    int j = half_j(p, pitch, x, y);
+    int b = half_b(p, pitch, x, y + (fy == 1 ? 0 : 1));
+    return (b + j + 1) >> 1;
+  }
+  if (fy == 2) {                  // i / k: avg of h (left/right) and j
+    int j = half_j(p, pitch, x, y);
+    int hh = half_h(p, pitch, x + (fx == 1 ? 0 : 1), y);
+    return (hh + j + 1) >> 1;
+  }
+  int b = half_b(p, pitch, x, y + (fy == 1 ? 0 : 1));
+  int hh = half_h(p, pitch, x + (fx == 1 ? 0 : 1), y);
+  return (b + hh + 1) >> 1;
 }
 
 // predicted chroma sample: bilinear with eighth-pel weights

@@ -46,27 +46,47 @@ def _tap6(a, b, c, d, e, f):
 
 
 def luma_mc(ref, x0, y0, mvx, mvy):
-    """16x16 luma MC at quarter-pel mv restricted to the half-pel grid
-    (mv % 2 == 0): H.264 6-tap half-sample interpolation (8.4.2.2.1)."""
+    """16x16 luma MC at any quarter-pel mv: 6-tap half samples
+    (8.4.2.2.1) and rounded-average quarter samples (8.4.2.2.2,
+    Table 8-12)."""
     ix, iy = mvx >> 2, mvy >> 2
     fx, fy = mvx & 3, mvy & 3
     bx, by = x0 + ix, y0 + iy
     if fx == 0 and fy == 0:
         return ref[by:by + 16, bx:bx + 16].astype(np.int64)
-    win = ref[by - 2:by + 19, bx - 2:bx + 19].astype(np.int64)
+    win = ref[by - 2:by + 20, bx - 2:bx + 20].astype(np.int64)  # 22x22
+
+    def htap(w):          # horizontal 6-tap sums: (rows, cols-5)
+        return _tap6(w[:, 0:-5], w[:, 1:-4], w[:, 2:-3], w[:, 3:-2],
+                     w[:, 4:-1], w[:, 5:])
+
+    def vtap(w):          # vertical 6-tap sums: (rows-5, cols)
+        return _tap6(w[0:-5], w[1:-4], w[2:-3], w[3:-2], w[4:-1], w[5:])
+
+    G = win[2:19, 2:19]                                     # 17x17 ints
+    b = np.clip((htap(win)[2:19] + 16) >> 5, 0, 255)        # 17x17
+    h = np.clip((vtap(win)[:, 2:19] + 16) >> 5, 0, 255)     # 17x17
+    j = np.clip((vtap(htap(win)) + 512) >> 10, 0, 255)      # 17x17
+    S = slice(0, 16)
+    G0, Gx, Gy = G[S, S], G[S, 1:17], G[1:17, S]
+    b0, s_ = b[S, S], b[1:17, S]        # s = b one row below
+    h0, m_ = h[S, S], h[S, 1:17]        # m = h one column right
+    j0 = j[S, S]
+
+    def avg(a, c):
+        return (a + c + 1) >> 1
+
     if fy == 0:
-        r = win[2:18]
-        s = _tap6(r[:, 0:16], r[:, 1:17], r[:, 2:18], r[:, 3:19],
-                  r[:, 4:20], r[:, 5:21])
-        return np.clip((s + 16) >> 5, 0, 255)
+        return b0 if fx == 2 else avg(G0 if fx == 1 else Gx, b0)
     if fx == 0:
-        c = win[:, 2:18]
-        s = _tap6(c[0:16], c[1:17], c[2:18], c[3:19], c[4:20], c[5:21])
-        return np.clip((s + 16) >> 5, 0, 255)
-    hs = _tap6(win[:, 0:16], win[:, 1:17], win[:, 2:18], win[:, 3:19],
-               win[:, 4:20], win[:, 5:21])
-    s = _tap6(hs[0:16], hs[1:17], hs[2:18], hs[3:19], hs[4:20], hs[5:21])
-    return np.clip((s + 512) >> 10, 0, 255)
+        return h0 if fy == 2 else avg(G0 if fy == 1 else Gy, h0)
+    if fx == 2 and fy == 2:
+        return j0
+    if fx == 2:                          # f / q
+        return avg(b0 if fy == 1 else s_, j0)
+    if fy == 2:                          # i / k
+        return avg(h0 if fx == 1 else m_, j0)
+    return avg(b0 if fy == 1 else s_, h0 if fx == 1 else m_)
 
 
 def chroma_mc(ref, cx0, cy0, mvx, mvy):
@@ -486,7 +506,6 @@ class Decoder:
         mvpx, mvpy = ctx["left_mv"] if (ctx["left_avail"] and
                                         ctx["left_inter"]) else (0, 0)
         mvx, mvy = mvdx + mvpx, mvdy + mvpy
-        assert mvx % 2 == 0 and mvy % 2 == 0, "subset: half-pel MV grid"
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
         pred_y = luma_mc(self.ref_y, x0, y0, mvx, mvy)

@@ -251,3 +251,48 @@ def test_halfpel_and_odd_integer_mvs():
     half = [mv for mv in d.mvs if mv[0] % 4 == 2 or mv[1] % 4 == 2]
     assert odd, "no odd-integer MVs coded for the 1-px shift"
     assert half, "no half-pel MVs coded for the half-shifted frame"
+
+
+def test_quarterpel_mvs():
+    """A ~quarter-pixel shift (3:1 blend of adjacent columns) must produce
+    quarter-pel MVs (mv & 3 odd) and stay bit-exact decoder-vs-recon."""
+    from h264_ref_decoder import Decoder
+    w, h = 192, 96
+    rng = np.random.default_rng(41)
+    base = rng.integers(0, 256, (h, w, 4), dtype=np.uint8)
+    f = base.astype(np.float32)
+    for _ in range(2):
+        f = (f + np.roll(f, 1, 1) + np.roll(f, -1, 1) +
+             np.roll(f, 1, 0) + np.roll(f, -1, 0)) / 5
+    f0 = f.astype(np.uint8)
+    f0[:, :, 3] = 255
+    fq = ((3 * f0.astype(np.uint16) + np.roll(f0, 1, 1)) // 4).astype(
+        np.uint8)
+    fq[:, :, 3] = 255
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = []
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs.append(tuple(ctx["left_mv"]))
+
+    enc = hipflux.H264Encoder(w, h)
+    stream = b""
+    recons = []
+    for img in (f0, fq):
+        r = enc.encode(np.ascontiguousarray(img).tobytes(), qp=22,
+                       idr=(img is f0))
+        stream += r["data"]
+        recons.append(recon_planes(enc, w, h))
+    d = MvDecoder()
+    frames = d.decode(stream)
+    assert len(frames) == 2
+    for i, (dec, rec) in enumerate(zip(frames, recons)):
+        for a, b, name in zip(dec, rec, "y cb cr".split()):
+            assert np.array_equal(a, b), f"frame {i} plane {name} mismatch"
+    quarter = [mv for mv in d.mvs if (mv[0] & 3) in (1, 3)
+               or (mv[1] & 3) in (1, 3)]
+    assert quarter, "no quarter-pel MVs coded for the quarter shift"
