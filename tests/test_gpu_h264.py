@@ -285,3 +285,44 @@ def test_gpu_halfpel_and_odd_mvs():
     half = [mv for mv in all_mvs if mv[0] % 4 == 2 or mv[1] % 4 == 2]
     assert odd, "no odd-integer MVs coded by the GPU pipeline"
     assert half, "no half-pel MVs coded by the GPU pipeline"
+
+
+def test_gpu_quarterpel_mvs():
+    """Quarter-shifted content (3:1 column blend) produces quarter-pel
+    MVs from the GPU pipeline and the stream decodes bit-exactly."""
+    require_gpu()
+    w, h = 192, 96
+    rng = np.random.default_rng(41)
+    base = rng.integers(0, 256, (h, w, 4), dtype=np.uint8)
+    f = base.astype(np.float32)
+    for _ in range(2):
+        f = (f + np.roll(f, 1, 1) + np.roll(f, -1, 1) +
+             np.roll(f, 1, 0) + np.roll(f, -1, 0)) / 5
+    f0 = f.astype(np.uint8)
+    f0[:, :, 3] = 255
+    fq = ((3 * f0.astype(np.uint16) + np.roll(f0, 1, 1)) // 4).astype(
+        np.uint8)
+    fq[:, :, 3] = 255
+    out = _native._pipeline_encode(
+        "gpu", [np.ascontiguousarray(f0), np.ascontiguousarray(fq)],
+        w, h, 22, 96, 1)
+    rows = reassemble(out)
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = []
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs.append(tuple(ctx["left_mv"]))
+
+    all_mvs = []
+    for y0, stream in rows.items():
+        d = MvDecoder()
+        decoded = d.decode(bytes(stream))
+        assert len(decoded) == 2
+        all_mvs += d.mvs
+    quarter = [mv for mv in all_mvs if (mv[0] & 3) in (1, 3)
+               or (mv[1] & 3) in (1, 3)]
+    assert quarter, "no quarter-pel MVs coded by the GPU pipeline"
