@@ -66,6 +66,7 @@ def main():
     from hipflux import _native
 
     kind = "gpu" if use_gpu else "cpu"
+    out_mode = 0 if args.encoder == "jpeg" else 1
     tile = args.mode == "tile" and world > 1
     if tile:
         # strong scaling: rank encodes rows [band0, band1) of the SAME
@@ -80,7 +81,7 @@ def main():
         enc_h = args.height
         band0 = 0
     pipe = _native.BenchPipeline(kind, args.width, enc_h, qp=args.qp,
-                                 stripe_height=64, output_mode=1,
+                                 stripe_height=64, output_mode=out_mode,
                                  gpu_id=local_rank if use_gpu else -1)
 
     # synthetic capture source: pre-generated random BGRX frames, cycled.
@@ -135,7 +136,8 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "encoded_fps_1080p60_h264",
+            "metric": ("encoded_fps_1080p60_h264" if out_mode == 1
+                       else "encoded_fps_1080p60_jpeg"),
             "value": round(fps_job, 2),
             "unit": "frames/s",
             "n_gpus": world,
