@@ -49,12 +49,18 @@ def parse_offer(sdp: str) -> Offer:
             tgt.mid = line.split(":", 1)[1].strip()
         elif line.startswith("a=rtpmap:") and tgt is not None:
             body = line.split(":", 1)[1]
-            pt, codec = body.split(" ", 1)
-            rtpmap[int(pt)] = codec.strip().lower()
+            try:
+                pt, codec = body.split(" ", 1)
+                rtpmap[int(pt)] = codec.strip().lower()
+            except ValueError:
+                pass   # client-controlled input: ignore malformed lines
         elif line.startswith("a=fmtp:") and tgt is not None:
             body = line.split(":", 1)[1]
-            pt, params = body.split(" ", 1)
-            fmtp[int(pt)] = params.strip()
+            try:
+                pt, params = body.split(" ", 1)
+                fmtp[int(pt)] = params.strip()
+            except ValueError:
+                pass
         if tgt is not None:
             tgt.raw.append(line)
     # audio: accept G.711 (PCMU preferred over PCMA); static PTs may be
