@@ -230,14 +230,20 @@ void encode_stripe_from_gpu(const GpuStripeParams& p,
 
 void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
                              std::vector<uint8_t>& out) {
+  encode_seg_nal_from_gpu(p, row, 0, (p.width + 15) / 16, row == 0, out);
+}
+
+void encode_seg_nal_from_gpu(const GpuStripeParams& p, int row, int mbx0,
+                             int seg_mbw, bool long_startcode,
+                             std::vector<uint8_t>& out) {
   const int mbw_stripe = (p.width + 15) / 16;
   {
     BitWriter b;
-    write_slice_header_bits(b, p.idr, row * mbw_stripe, p.frame_num,
+    write_slice_header_bits(b, p.idr, row * mbw_stripe + mbx0, p.frame_num,
                             p.idr_pic_id, p.qp);
     RowCtx ctx;
     const size_t mb_base = (size_t)(p.mb_row0 + row) * p.mbw;
-    for (int mbx = 0; mbx < mbw_stripe; ++mbx) {
+    for (int mbx = mbx0; mbx < mbx0 + seg_mbw; ++mbx) {
       const int16_t* L = p.levels + (mb_base + mbx) * kLevelsPerMb;
       const int* M = p.meta + (mb_base + mbx) * kMetaPerMb;
       int m0 = M[0];
@@ -268,7 +274,7 @@ void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
     }
     if (!p.idr && ctx.skip_run > 0) b.ue(ctx.skip_run);
     b.rbsp_trailing();
-    b.emit_nal(out, p.idr ? 3 : 2, p.idr ? 5 : 1, row == 0);
+    b.emit_nal(out, p.idr ? 3 : 2, p.idr ? 5 : 1, long_startcode);
   }
 }
 

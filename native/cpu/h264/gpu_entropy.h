@@ -29,6 +29,10 @@ void encode_stripe_from_gpu(const GpuStripeParams& p,
 // One MB-row slice NAL (row is stripe-relative). Used for row-parallel
 // entropy; concatenating SPS/PPS (idr) + all rows reproduces
 // encode_stripe_from_gpu's output exactly.
+void encode_seg_nal_from_gpu(const GpuStripeParams& p, int row, int mbx0,
+                             int seg_mbw, bool long_startcode,
+                             std::vector<uint8_t>& out);
+
 void encode_row_nal_from_gpu(const GpuStripeParams& p, int row,
                              std::vector<uint8_t>& out);
 

@@ -325,8 +325,9 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
   const RowJob job = jobs[blockIdx.x];
   const int tid = threadIdx.x;            // 512 threads (8 waves)
   const bool i_slice = (job.flags & 1) != 0;
-  const size_t row_base = (size_t)job.mb_row * mbw;
-  const int nitems = items_per_row(mbw);
+  const int seg_mbw = job.seg_mbw;
+  const size_t row_base = (size_t)job.mb_row * mbw + job.mbx0;
+  const int nitems = items_per_row(seg_mbw);
   uint32_t* row_stage =
       stage + (size_t)blockIdx.x * nitems * kStageWordsPerItem;
   int* row_nbits = nbits + (size_t)blockIdx.x * nitems;
@@ -336,12 +337,12 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
   __shared__ MbInfo s_mb[512];
   __shared__ short s_skiprun[512];
   __shared__ int s_trailing;
-  for (int mb = tid; mb < mbw; mb += 512)
+  for (int mb = tid; mb < seg_mbw; mb += 512)
     precompute_mb(levels, meta, row_base + mb, i_slice, &s_mb[mb]);
   __syncthreads();
   if (tid == 0) {
     int run = 0;
-    for (int mb = 0; mb < mbw; ++mb) {
+    for (int mb = 0; mb < seg_mbw; ++mb) {
       if (!i_slice && MB_MODE(s_mb[mb].flags) == kSkip) {
         s_skiprun[mb] = -1;  // skipped MB: no items
         ++run;

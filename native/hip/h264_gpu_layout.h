@@ -27,9 +27,15 @@ constexpr int kMetaPerMb = 2;
 
 enum MbMode : int { kSkip = 0, kInter = 1, kIntra = 2 };
 
+// One job = one SLICE = a horizontal SEGMENT of an MB row. Narrow frames
+// use one segment per row; wide frames split rows into multiple slices
+// (H.264 allows a slice to start at any MB) so the serial left-neighbor
+// chain — the row kernels' latency bound — halves at 4K and quarters at
+// 8K. Segment boundaries reset intra/MVP/skip context exactly like row
+// starts do.
 struct RowJob {
   int mb_row;      // absolute MB row in the frame
-  int qp;          // luma QP for this row's slice
+  int qp;          // luma QP for this slice
   int flags;       // bit0: I slice (IDR)
   int stripe_y0;   // stripe pixel bounds (for ME clamping)
   int stripe_y1;
@@ -37,7 +43,12 @@ struct RowJob {
   int first_mb;    // first_mb_in_slice (stripe-relative)
   int frame_num;
   int idr_pic_id;
+  int mbx0;        // first MB column of this segment
+  int seg_mbw;     // segment width in MBs
 };
+
+// rows wider than this are split into ceil(mbw / kMaxSegMbw) slices
+constexpr int kMaxSegMbw = 128;
 
 // GPU entropy staging layout: per row, items =
 //   [0] slice header, [1 + mb*28 + slot] per-MB items, [last] trailing

@@ -41,7 +41,8 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
                          uint8_t* __restrict__ curY, int mbw, int mby,
                          int qp, bool i_slice,
                          const int16_t* __restrict__ levels_base,
-                         int* __restrict__ meta, int lane) {
+                         int* __restrict__ meta, int lane, int mbx0,
+                         int seg_mbw) {
   const int y0 = mby * 16;
   const int r = lane >> 2, cq = (lane & 3) * 4;
   const int g = lane >> 4, c = lane & 15;
@@ -74,16 +75,17 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
                ? refY[(size_t)(y0 + lane) * ypitch + mbx2 * 16 + 15]
                : 0;
   };
-  uint32_t psrc_cur = load_psrc(0);
-  uint32_t refp_cur = i_slice ? 0 : load_refp(0);
-  int refcol_cur = i_slice ? 0 : load_refcol(0);
+  uint32_t psrc_cur = load_psrc(mbx0);
+  uint32_t refp_cur = i_slice ? 0 : load_refp(mbx0);
+  int refcol_cur = i_slice ? 0 : load_refcol(mbx0);
   int m0_cur = 0, m1_cur = 0;
   if (!i_slice) {
-    m0_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 0];
-    m1_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 1];
+    m0_cur = meta[((size_t)mby * mbw + mbx0) * kMetaPerMb + 0];
+    m1_cur = meta[((size_t)mby * mbw + mbx0) * kMetaPerMb + 1];
   }
 
-  for (int mbx = 0; mbx < mbw; ++mbx) {
+  const int mbx_end = mbx0 + seg_mbw;
+  for (int mbx = mbx0; mbx < mbx_end; ++mbx) {
     const int x0 = mbx * 16;
     const size_t mb_index = (size_t)mby * mbw + mbx;
     int16_t* L = const_cast<int16_t*>(levels_base) +
@@ -92,7 +94,7 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
 
     uint32_t psrc_nxt = 0, refp_nxt = 0;
     int refcol_nxt = 0, m0_nxt = 0, m1_nxt = 0;
-    if (mbx + 1 < mbw) {
+    if (mbx + 1 < mbx_end) {
       psrc_nxt = load_psrc(mbx + 1);
       if (!i_slice) {
         refp_nxt = load_refp(mbx + 1);
@@ -298,7 +300,8 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                            uint8_t* __restrict__ curCr, int mbw, int mby,
                            int qpc, bool i_slice,
                            const int16_t* __restrict__ levels_base,
-                           int* __restrict__ meta, int lane) {
+                           int* __restrict__ meta, int lane, int mbx0,
+                           int seg_mbw) {
   const int cy0 = mby * 8;
   const int g = lane >> 4, c = lane & 15;
   const int zz = c_zz_of_pos[c];
@@ -336,17 +339,18 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
                ? plane[(size_t)(cy0 + lane) * cpitch + mbx2 * 8 + 7]
                : 0;
   };
-  uint32_t csrc_cur = cload_src(0);
-  uint32_t cref_cur = i_slice ? 0 : cload_ref(0);
-  int colcb_cur = i_slice ? 0 : cload_col(0, refCb);
-  int colcr_cur = i_slice ? 0 : cload_col(0, refCr);
+  uint32_t csrc_cur = cload_src(mbx0);
+  uint32_t cref_cur = i_slice ? 0 : cload_ref(mbx0);
+  int colcb_cur = i_slice ? 0 : cload_col(mbx0, refCb);
+  int colcr_cur = i_slice ? 0 : cload_col(mbx0, refCr);
   int m0_cur = 0, m1_cur = 0;
   if (!i_slice) {
-    m0_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 0];
-    m1_cur = meta[((size_t)mby * mbw) * kMetaPerMb + 1];
+    m0_cur = meta[((size_t)mby * mbw + mbx0) * kMetaPerMb + 0];
+    m1_cur = meta[((size_t)mby * mbw + mbx0) * kMetaPerMb + 1];
   }
 
-  for (int mbx = 0; mbx < mbw; ++mbx) {
+  const int mbx_end = mbx0 + seg_mbw;
+  for (int mbx = mbx0; mbx < mbx_end; ++mbx) {
     const int cx0 = mbx * 8;
     const size_t mb_index = (size_t)mby * mbw + mbx;
     int16_t* L = const_cast<int16_t*>(levels_base) +
@@ -355,7 +359,7 @@ __device__ void chroma_row(const uint8_t* __restrict__ srcCb,
 
     uint32_t csrc_nxt = 0, cref_nxt = 0;
     int colcb_nxt = 0, colcr_nxt = 0, m0_nxt = 0, m1_nxt = 0;
-    if (mbx + 1 < mbw) {
+    if (mbx + 1 < mbx_end) {
       csrc_nxt = cload_src(mbx + 1);
       if (!i_slice) {
         cref_nxt = cload_ref(mbx + 1);
@@ -671,10 +675,11 @@ __global__ void __launch_bounds__(128) k_h264_rows(
 
   if (threadIdx.x < 64) {
     luma_row(srcY, ypitch, w, h, refY, curY, mbw, mby, qp, i_slice, levels,
-             meta, lane);
+             meta, lane, job.mbx0, job.seg_mbw);
   } else {
     chroma_row(srcCb, srcCr, cpitch, w, h, refCb, refCr, curCb, curCr, mbw,
-               mby, dev_chroma_qp(qp), i_slice, levels, meta, lane);
+               mby, dev_chroma_qp(qp), i_slice, levels, meta, lane,
+               job.mbx0, job.seg_mbw);
   }
 }
 
@@ -683,11 +688,13 @@ __global__ void __launch_bounds__(128) k_h264_rows(
 // (64 lanes) per MB of each P row. Thresholds match the CPU encoder.
 __global__ void __launch_bounds__(64) k_h264_me(
     const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
-    const uint8_t* __restrict__ refY, int mbw, int frame_w_mb16,
-    const RowJob* __restrict__ jobs, int* __restrict__ meta) {
-  const int job_idx = blockIdx.x / mbw;
-  const int mbx = blockIdx.x % mbw;
+    const uint8_t* __restrict__ refY, int mbw, int seg_max,
+    int frame_w_mb16, const RowJob* __restrict__ jobs,
+    int* __restrict__ meta) {
+  const int job_idx = blockIdx.x / seg_max;
   const RowJob job = jobs[job_idx];
+  const int mbx = job.mbx0 + blockIdx.x % seg_max;
+  if (blockIdx.x % seg_max >= job.seg_mbw) return;
   if (job.flags & 1) return;  // I rows have no ME
   const int lane = threadIdx.x;
   const int mby = job.mb_row;
@@ -802,11 +809,13 @@ using i32x4 = __attribute__((__vector_size__(16))) int;
 
 __global__ void __launch_bounds__(64) k_h264_me_mfma(
     const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
-    const uint8_t* __restrict__ refY, int mbw, int frame_w_mb16,
-    const RowJob* __restrict__ jobs, int* __restrict__ meta) {
-  const int job_idx = blockIdx.x / mbw;
-  const int mbx = blockIdx.x % mbw;
+    const uint8_t* __restrict__ refY, int mbw, int seg_max,
+    int frame_w_mb16, const RowJob* __restrict__ jobs,
+    int* __restrict__ meta) {
+  const int job_idx = blockIdx.x / seg_max;
   const RowJob job = jobs[job_idx];
+  const int mbx = job.mbx0 + blockIdx.x % seg_max;
+  if (blockIdx.x % seg_max >= job.seg_mbw) return;
   if (job.flags & 1) return;
   const int lane = threadIdx.x;
   const int mby = job.mb_row;
@@ -1024,14 +1033,15 @@ void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
                     const RowJob* d_jobs, int* d_meta, hipStream_t stream,
                     bool use_mfma) {
   if (n_jobs == 0) return;
+  const int seg_max = mbw < kMaxSegMbw ? mbw : kMaxSegMbw;
   if (use_mfma) {
-    hipLaunchKernelGGL(k_h264_me_mfma, dim3(n_jobs * mbw), dim3(64), 0,
-                       stream, srcY, ypitch, w, h, refY, mbw, mbw * 16,
-                       d_jobs, d_meta);
+    hipLaunchKernelGGL(k_h264_me_mfma, dim3(n_jobs * seg_max), dim3(64), 0,
+                       stream, srcY, ypitch, w, h, refY, mbw, seg_max,
+                       mbw * 16, d_jobs, d_meta);
   } else {
-    hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * mbw), dim3(64), 0, stream,
-                       srcY, ypitch, w, h, refY, mbw, mbw * 16, d_jobs,
-                       d_meta);
+    hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * seg_max), dim3(64), 0,
+                       stream, srcY, ypitch, w, h, refY, mbw, seg_max,
+                       mbw * 16, d_jobs, d_meta);
   }
 }
 

@@ -45,14 +45,15 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
                           int qp, bool i_slice,
                           const int16_t* __restrict__ levels_base,
                           int* __restrict__ meta, int lane, int w,
-                          Shared* sh) {
+                          int mbx0, int seg_mbw, Shared* sh) {
   const int y0 = mby * 16;
   const int r = lane >> 2, cq = (lane & 3) * 4;
   const int g = lane >> 4, c = lane & 15;
   const int zz = c_zz_of_pos[c];
   bool have_left = false;
 
-  for (int mbx = 0; mbx < mbw; ++mbx) {
+  const int mbx_end = mbx0 + seg_mbw;
+  for (int mbx = mbx0; mbx < mbx_end; ++mbx) {
     const int x0 = mbx * 16;
     const size_t mb_index = (size_t)mby * mbw + mbx;
     int16_t* L = const_cast<int16_t*>(levels_base) + mb_index * kLevelsPerMb;
@@ -252,14 +253,15 @@ __device__ void chroma_wave(const uint8_t* __restrict__ srcC,
                             int h, int mbw, int mby, int qpc, bool i_slice,
                             const int16_t* __restrict__ levels_base,
                             int* __restrict__ meta, int lane, int comp,
-                            Shared* sh) {
+                            int mbx0, int seg_mbw, Shared* sh) {
   const int cy0 = mby * 8;
   const int g = lane >> 4, c = lane & 15;   // sub-block g, coeff c
   const int zz = c_zz_of_pos[c];
   const int cw = (w_ + 1) / 2, chh = (h + 1) / 2;
   bool have_left = false;
 
-  for (int mbx = 0; mbx < mbw; ++mbx) {
+  const int mbx_end = mbx0 + seg_mbw;
+  for (int mbx = mbx0; mbx < mbx_end; ++mbx) {
     const int cx0 = mbx * 8;
     const size_t mb_index = (size_t)mby * mbw + mbx;
     int16_t* L = const_cast<int16_t*>(levels_base) + mb_index * kLevelsPerMb;
@@ -422,15 +424,16 @@ __global__ void __launch_bounds__(384) k_h264_rows4(
 
   if (wave < 4) {
     rows4::luma_wave<1>(srcY, ypitch, w, h, refY, curY, mbw, mby, qp,
-                        i_slice, levels, meta, lane, wave, &sh);
+                        i_slice, levels, meta, lane, wave, job.mbx0,
+                        job.seg_mbw, &sh);
   } else if (wave == 4) {
     rows4::chroma_wave(srcCb, refCb, curCb, cpitch, w, h, mbw, mby,
                        dev_chroma_qp(qp), i_slice, levels, meta, lane, 0,
-                       &sh);
+                       job.mbx0, job.seg_mbw, &sh);
   } else {
     rows4::chroma_wave(srcCr, refCr, curCr, cpitch, w, h, mbw, mby,
                        dev_chroma_qp(qp), i_slice, levels, meta, lane, 1,
-                       &sh);
+                       job.mbx0, job.seg_mbw, &sh);
   }
 }
 

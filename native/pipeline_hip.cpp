@@ -308,15 +308,22 @@ class HipH264Pipeline : public EncodePipeline {
       // stripe's padded pixel bounds for ME clamping
       int sy0 = st.y0, sy1 = (row0 + rows) * 16;
       for (int r = 0; r < rows; ++r) {
-        h_jobs_[n_jobs].mb_row = row0 + r;
-        h_jobs_[n_jobs].qp = qp;
-        h_jobs_[n_jobs].flags = sj.idr ? 1 : 0;
-        h_jobs_[n_jobs].stripe_y0 = sy0;
-        h_jobs_[n_jobs].stripe_y1 = sy1;
-        h_jobs_[n_jobs].first_mb = r * mbw_;
-        h_jobs_[n_jobs].frame_num = static_cast<int>(sj.frame_num);
-        h_jobs_[n_jobs].idr_pic_id = static_cast<int>(sj.idr_pic_id);
-        ++n_jobs;
+        int x = 0;
+        for (int sg = 0; sg < segs_; ++sg) {
+          int segw = std::min(seg_w0_, mbw_ - x);
+          h_jobs_[n_jobs].mb_row = row0 + r;
+          h_jobs_[n_jobs].qp = qp;
+          h_jobs_[n_jobs].flags = sj.idr ? 1 : 0;
+          h_jobs_[n_jobs].stripe_y0 = sy0;
+          h_jobs_[n_jobs].stripe_y1 = sy1;
+          h_jobs_[n_jobs].first_mb = r * mbw_ + x;
+          h_jobs_[n_jobs].frame_num = static_cast<int>(sj.frame_num);
+          h_jobs_[n_jobs].idr_pic_id = static_cast<int>(sj.idr_pic_id);
+          h_jobs_[n_jobs].mbx0 = x;
+          h_jobs_[n_jobs].seg_mbw = segw;
+          x += segw;
+          ++n_jobs;
+        }
       }
     }
     if (n_jobs == 0) return;
@@ -456,24 +463,29 @@ class HipH264Pipeline : public EncodePipeline {
         p.idr = sj.idr;
         p.frame_num = sj.frame_num;
         p.idr_pic_id = sj.idr_pic_id;
-        outs[i].rows.resize(p.n_mb_rows);
+        outs[i].rows.resize(p.n_mb_rows * segs_);
         if (p.idr) {
           h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
                               p.n_mb_rows, p.width, p.height);
           h264::write_pps_nal(outs[i].header);
         }
-        for (int r = 0; r < p.n_mb_rows; ++r) job_map[j++] = {int(i), r};
+        for (int r = 0; r < p.n_mb_rows * segs_; ++r)
+          job_map[j++] = {int(i), r};
       }
     }
     if (cpu_entropy_) {
       for (size_t b = 0; b < batches.size(); ++b) {
         HIP_CHECK(hipEventSynchronize(batch_events_[b]));
         for (int j = batches[b].job0; j < batches[b].jobn; ++j) {
-          auto [si, r] = job_map[j];
-          auto* dst = &outs[si].rows[r];
+          auto [si, sl] = job_map[j];
+          auto* dst = &outs[si].rows[sl];
           const h264::GpuStripeParams* pp = &outs[si].p;
-          pool_.submit([pp, r, dst] {
-            h264::encode_row_nal_from_gpu(*pp, r, *dst);
+          int r = sl / segs_;
+          int mbx0 = h_jobs_[j].mbx0, segw = h_jobs_[j].seg_mbw;
+          bool long_sc = sl == 0;
+          pool_.submit([pp, r, mbx0, segw, long_sc, dst] {
+            h264::encode_seg_nal_from_gpu(*pp, r, mbx0, segw, long_sc,
+                                          *dst);
           });
         }
       }
@@ -494,12 +506,12 @@ class HipH264Pipeline : public EncodePipeline {
       }
       ent_copy_words_ = std::min(ent_stride_words_, max_words * 2 + 64);
       for (int j = 0; j < n_jobs; ++j) {
-        auto [si, r] = job_map[j];
+        auto [si, sl] = job_map[j];
         const uint32_t* words = h_entout_ + (size_t)j * ent_stride_words_;
         int bits = h_outbits_[j];
         bool idr = outs[si].idr;
-        auto* dst = &outs[si].rows[r];
-        bool long_sc = r == 0;
+        auto* dst = &outs[si].rows[sl];
+        bool long_sc = sl == 0;
         pool_.submit([words, bits, idr, long_sc, dst] {
           h264::assemble_gpu_row_nal(words, bits, idr, long_sc, *dst);
         });
@@ -607,34 +619,37 @@ class HipH264Pipeline : public EncodePipeline {
         static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb * sizeof(int);
     d_levels_ = reinterpret_cast<int16_t*>(dalloc(level_bytes));
     d_meta_ = reinterpret_cast<int*>(dalloc(meta_bytes));
+    segs_ = (mbw_ + h264gpu::kMaxSegMbw - 1) / h264gpu::kMaxSegMbw;
+    seg_w0_ = (mbw_ + segs_ - 1) / segs_;   // widest segment
+    const int max_jobs = mbh_ * segs_;
     d_jobs_ = reinterpret_cast<h264gpu::RowJob*>(
-        dalloc(sizeof(h264gpu::RowJob) * mbh_));
+        dalloc(sizeof(h264gpu::RowJob) * max_jobs));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_levels_), level_bytes,
                             hipHostMallocDefault));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_meta_), meta_bytes,
                             hipHostMallocDefault));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_jobs_),
-                            sizeof(h264gpu::RowJob) * mbh_,
+                            sizeof(h264gpu::RowJob) * max_jobs,
                             hipHostMallocDefault));
-    // GPU entropy buffers
+    // GPU entropy buffers (one slot per SLICE = row segment)
     cpu_entropy_ = std::getenv("HIPFLUX_CPU_ENTROPY") != nullptr;
-    const int nitems = h264gpu::items_per_row(mbw_);
+    const int nitems = h264gpu::items_per_row(seg_w0_);
     ent_stride_words_ = nitems * h264gpu::kStageWordsPerItem;
     ent_copy_words_ = 1 << 30;     // first frame after (re)alloc: full copy
     size_t stage_bytes =
-        (size_t)mbh_ * nitems * h264gpu::kStageWordsPerItem * 4;
+        (size_t)max_jobs * nitems * h264gpu::kStageWordsPerItem * 4;
     d_stage_ = reinterpret_cast<uint32_t*>(dalloc(stage_bytes));
-    d_nbits_ = reinterpret_cast<int*>(dalloc((size_t)mbh_ * nitems * 4));
+    d_nbits_ = reinterpret_cast<int*>(dalloc((size_t)max_jobs * nitems * 4));
     d_entout_ = reinterpret_cast<uint32_t*>(
-        dalloc((size_t)mbh_ * ent_stride_words_ * 4));
-    d_outbits_ = reinterpret_cast<int*>(dalloc(sizeof(int) * mbh_));
+        dalloc((size_t)max_jobs * ent_stride_words_ * 4));
+    d_outbits_ = reinterpret_cast<int*>(dalloc(sizeof(int) * max_jobs));
     if (h_entout_) (void)hipHostFree(h_entout_);
     if (h_outbits_) (void)hipHostFree(h_outbits_);
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_entout_),
-                            (size_t)mbh_ * ent_stride_words_ * 4,
+                            (size_t)max_jobs * ent_stride_words_ * 4,
                             hipHostMallocDefault));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_outbits_),
-                            sizeof(int) * mbh_, hipHostMallocDefault));
+                            sizeof(int) * max_jobs, hipHostMallocDefault));
     stripes_.assign((h + stripe_h_ - 1) / stripe_h_, StripeState{});
   }
 
@@ -643,6 +658,7 @@ class HipH264Pipeline : public EncodePipeline {
   hipStream_t stream_{};
   int stripe_h_ = 64;
   int w_ = 0, h_ = 0, mbw_ = 0, mbh_ = 0, ypitch_ = 0, cpitch_ = 0;
+  int segs_ = 1, seg_w0_ = 0;     // slices per MB row, widest segment
   uint8_t *d_frame_ = nullptr, *d_srcY_ = nullptr, *d_srcCb_ = nullptr,
           *d_srcCr_ = nullptr, *d_refY_ = nullptr, *d_refCb_ = nullptr,
           *d_refCr_ = nullptr, *d_curY_ = nullptr, *d_curCb_ = nullptr,

@@ -444,7 +444,6 @@ class Decoder:
             raise ValueError("P slice before any IDR")
 
         mb_row = first_mb // mbw
-        assert first_mb % mbw == 0, "subset: slice starts at row boundary"
 
         ctx = {
             "left_avail": False,
@@ -454,7 +453,7 @@ class Decoder:
             "left_cb_nc": [0] * 2,
             "left_cr_nc": [0] * 2,
         }
-        mbx = 0
+        mbx = first_mb % mbw        # slices may start mid-row (segments)
         skip_left = 0
         if is_p:
             skip_left = br.ue()
@@ -467,8 +466,8 @@ class Decoder:
                 skip_left -= 1
                 mbx += 1
                 continue
-            if is_p and not br.more_rbsp_data():
-                break  # trailing skips consumed the rest
+            if not br.more_rbsp_data():
+                break  # end of this slice (trailing skips / segment end)
             mb_type = br.ue()
             if is_p and mb_type < 5:
                 assert mb_type == 0, f"subset: P mb_type {mb_type}"

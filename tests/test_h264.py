@@ -296,3 +296,4 @@ def test_quarterpel_mvs():
     quarter = [mv for mv in d.mvs if (mv[0] & 3) in (1, 3)
                or (mv[1] & 3) in (1, 3)]
     assert quarter, "no quarter-pel MVs coded for the quarter shift"
+
