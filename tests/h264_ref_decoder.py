@@ -479,7 +479,7 @@ class Decoder:
             mbx += 1
             if is_p and mbx < mbw and br.more_rbsp_data():
                 skip_left = br.ue()
-        return mb_row
+        return mb_row, mbx - first_mb % mbw
 
     def decode_skip(self, mbx, mby):
         x0, y0 = mbx * 16, mby * 16
@@ -712,22 +712,22 @@ class Decoder:
     def decode(self, data: bytes):
         """Decode an Annex-B stream; returns list of (y, cb, cr) uint8 frames
         (cropped to the SPS-declared size)."""
-        rows_done = set()
-        mbh = None
+        mbs_done = 0
+        total_mbs = None
         for nal in split_nals(data):
             nal_type = nal[0] & 0x1F
             br = BitReader(nal[1:])
             if nal_type == 7:
                 self.parse_sps(br)
-                mbh = self.sps["mbh"]
+                total_mbs = self.sps["mbw"] * self.sps["mbh"]
             elif nal_type == 8:
                 self.parse_pps(br)
             elif nal_type in (1, 5):
-                row = self.decode_slice(br, nal_type)
-                rows_done.add(row)
-                if len(rows_done) == mbh:
+                _, n_mbs = self.decode_slice(br, nal_type)
+                mbs_done += n_mbs
+                if mbs_done == total_mbs:
                     self.finish_frame()
-                    rows_done = set()
+                    mbs_done = 0
         return self.frames
 
     def finish_frame(self):
