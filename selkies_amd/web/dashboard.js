@@ -48,6 +48,11 @@
     <h3>Files</h3>
     <input type="file" id="sk-upload">
     <div id="sk-files"></div>
+    <h3>Share</h3>
+    <div>
+      <button id="sk-share-view">copy viewer link</button>
+      <button id="sk-share-rtc">copy WebRTC link</button>
+    </div>
   `;
   document.body.appendChild(panel);
   tab.onclick = () => panel.classList.toggle("open");
@@ -57,7 +62,8 @@
   /* settings widgets built from the server's client-settings contract */
   const WIDGETS = ["encoder", "framerate", "video_crf",
                    "video_bitrate_kbps", "video_cbr_mode", "jpeg_quality",
-                   "use_paint_over_quality", "video_fullframe"];
+                   "use_paint_over_quality", "video_fullframe",
+                   "capture_scale_div", "enable_audio"];
   window.skOnSettings = (payload) => {
     const host = document.getElementById("sk-settings");
     host.innerHTML = "";
@@ -90,6 +96,13 @@
         const v = input.type === "checkbox" ? input.checked
             : (input.type === "number" ? +input.value : input.value);
         send("SETTINGS," + JSON.stringify({ [name]: v }));
+        /* remember the preference so reconnects replay it */
+        try {
+          const prefs = JSON.parse(
+              localStorage.getItem("selkies.prefs") || "{}");
+          prefs[name] = v;
+          localStorage.setItem("selkies.prefs", JSON.stringify(prefs));
+        } catch (e) { /* private mode */ }
       };
       label.appendChild(input);
       host.appendChild(label);
@@ -147,4 +160,17 @@
   };
   setInterval(refreshFiles, 5000);
   refreshFiles();
+
+  /* sharing links (reference dashboard "sharing links" section) */
+  function shareUrl(params) {
+    const u = new URL(location.href);
+    for (const [k, v] of Object.entries(params)) u.searchParams.set(k, v);
+    return u.toString();
+  }
+  document.getElementById("sk-share-view").onclick = () =>
+      navigator.clipboard.writeText(shareUrl({ role: "viewer" }))
+          .catch(() => {});
+  document.getElementById("sk-share-rtc").onclick = () =>
+      navigator.clipboard.writeText(shareUrl({ transport: "webrtc" }))
+          .catch(() => {});
 })();
