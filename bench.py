@@ -40,6 +40,10 @@ def main():
     ap.add_argument("--encoder", default="h264enc-striped")
     ap.add_argument("--cpu", action="store_true",
                     help="force the CPU pipeline (debug only)")
+    ap.add_argument("--sessions", type=int, default=1,
+                    help="independent encode sessions per rank/GPU, run on "
+                         "concurrent threads (the BASELINE metric's "
+                         "concurrent-sessions axis)")
     ap.add_argument("--mode", choices=["sessions", "tile"],
                     default="sessions",
                     help="sessions = one independent session per GPU "
@@ -80,9 +84,12 @@ def main():
     else:
         enc_h = args.height
         band0 = 0
-    pipe = _native.BenchPipeline(kind, args.width, enc_h, qp=args.qp,
-                                 stripe_height=64, output_mode=out_mode,
-                                 gpu_id=local_rank if use_gpu else -1)
+    n_sess = max(1, args.sessions)
+    pipes = [_native.BenchPipeline(kind, args.width, enc_h, qp=args.qp,
+                                   stripe_height=64, output_mode=out_mode,
+                                   gpu_id=local_rank if use_gpu else -1)
+             for _ in range(n_sess)]
+    pipe = pipes[0]
 
     # synthetic capture source: pre-generated random BGRX frames, cycled.
     # Every frame differs everywhere (worst case for a screen encoder).
@@ -104,21 +111,50 @@ def main():
             dist.barrier()
 
     # warmup (untimed): includes the IDR and allocator/registration warmup
-    pipe.encode(frames[0], True)
-    for i in range(max(1, args.warmup - 1)):
-        pipe.encode(frames[(i + 1) % n_src], False)
+    for p_ in pipes:
+        p_.encode(frames[0], True)
+        for i in range(max(1, args.warmup - 1)):
+            p_.encode(frames[(i + 1) % n_src], False)
 
     sync()
     lat_ms = []
     total_bytes = 0
-    t0 = time.perf_counter()
-    for i in range(args.steps):
-        ts = time.perf_counter()
-        nbytes, _ = pipe.encode(frames[i % n_src], False)
-        lat_ms.append((time.perf_counter() - ts) * 1e3)
-        total_bytes += nbytes
-    sync()
-    t1 = time.perf_counter()
+    if n_sess == 1:
+        t0 = time.perf_counter()
+        for i in range(args.steps):
+            ts = time.perf_counter()
+            nbytes, _ = pipe.encode(frames[i % n_src], False)
+            lat_ms.append((time.perf_counter() - ts) * 1e3)
+            total_bytes += nbytes
+        sync()
+        t1 = time.perf_counter()
+    else:
+        # concurrent sessions: one thread per session, each runs `steps`
+        # frames; aggregate fps = steps * sessions / wall
+        import threading
+        per = [[] for _ in range(n_sess)]
+        nb = [0] * n_sess
+
+        def run_one(si):
+            p_ = pipes[si]
+            for i in range(args.steps):
+                ts = time.perf_counter()
+                b, _ = p_.encode(frames[(i + si) % n_src], False)
+                per[si].append((time.perf_counter() - ts) * 1e3)
+                nb[si] += b
+
+        threads = [threading.Thread(target=run_one, args=(si,))
+                   for si in range(n_sess)]
+        t0 = time.perf_counter()
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        sync()
+        t1 = time.perf_counter()
+        for q in per:
+            lat_ms.extend(q)
+        total_bytes = sum(nb)
 
     elapsed = t1 - t0
     # MAX elapsed over ranks (slowest rank defines job throughput)
@@ -126,7 +162,7 @@ def main():
         t = torch.tensor([elapsed], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
-    fps_job = args.steps * (1 if tile else world) / elapsed
+    fps_job = args.steps * n_sess * (1 if tile else world) / elapsed
     p50 = float(np.percentile(lat_ms, 50))
     p95 = float(np.percentile(lat_ms, 95))
     if distributed:
@@ -157,7 +193,7 @@ def main():
                 "global_batch": world,
                 "seq_len": args.width * args.height,
                 "parallelism": (f"tile{world}" if tile
-                                else f"sessions{world}"),
+                                else f"sessions{world * n_sess}"),
                 "resolution": f"{args.width}x{args.height}",
                 "qp": args.qp,
                 "pipeline": pipe.pipeline,
