@@ -162,7 +162,8 @@ void put16(std::vector<uint8_t>& o, int v) {
 }
 
 void write_headers(std::vector<uint8_t>& o, int width, int height,
-                   const uint8_t qy[64], const uint8_t qc[64], bool fullcolor) {
+                   const uint8_t qy[64], const uint8_t qc[64], bool fullcolor,
+                   int restart_interval = 0) {
   // SOI
   o.push_back(0xFF); o.push_back(0xD8);
   // APP0 JFIF
@@ -204,6 +205,11 @@ void write_headers(std::vector<uint8_t>& o, int width, int height,
   dht(0x10, kAcLumaBits, kAcLumaVals);
   dht(0x01, kDcChromaBits, kDcChromaVals);
   dht(0x11, kAcChromaBits, kAcChromaVals);
+  if (restart_interval > 0) {             // DRI
+    o.push_back(0xFF); o.push_back(0xDD);
+    put16(o, 4);
+    put16(o, restart_interval);
+  }
   // SOS
   o.push_back(0xFF); o.push_back(0xDA);
   put16(o, 6 + 2 * 3);
@@ -276,11 +282,16 @@ void jpeg_quality_tables(int quality, uint8_t qy[64], uint8_t qc[64]) {
 }
 
 void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
-                      int quality, bool fullcolor, std::vector<uint8_t>& out) {
+                      int quality, bool fullcolor, std::vector<uint8_t>& out,
+                      bool restart_rows) {
   uint8_t qy[64], qc[64];
   jpeg_quality_tables(quality, qy, qc);
-  write_headers(out, width, height, qy, qc, fullcolor);
+  int mcu_w = fullcolor ? 8 : 16;
+  int mcux_hdr = (width + mcu_w - 1) / mcu_w;
+  write_headers(out, width, height, qy, qc, fullcolor,
+                restart_rows ? mcux_hdr : 0);
   BitWriter bw(out);
+  int rst = 0;
 
   int dcY = 0, dcCb = 0, dcCr = 0;
   int16_t blk[64];
@@ -297,6 +308,12 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
     float Y[256], Cb[256], Cr[256];  // 16x16 samples
     float plane[64];
     for (int my = 0; my < mcuy; ++my) {
+      if (restart_rows && my > 0) {
+        bw.flush();
+        out.push_back(0xFF);
+        out.push_back(static_cast<uint8_t>(0xD0 + (rst++ & 7)));
+        dcY = dcCb = dcCr = 0;
+      }
       for (int mx = 0; mx < mcux; ++mx) {
         for (int yy = 0; yy < 16; ++yy)
           for (int xx = 0; xx < 16; ++xx) {
@@ -339,6 +356,12 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
     int mcux = (width + 7) / 8, mcuy = (height + 7) / 8;
     float plane[3][64];
     for (int my = 0; my < mcuy; ++my) {
+      if (restart_rows && my > 0) {
+        bw.flush();
+        out.push_back(0xFF);
+        out.push_back(static_cast<uint8_t>(0xD0 + (rst++ & 7)));
+        dcY = dcCb = dcCr = 0;
+      }
       for (int mx = 0; mx < mcux; ++mx) {
         for (int yy = 0; yy < 8; ++yy)
           for (int xx = 0; xx < 8; ++xx) {
@@ -366,14 +389,15 @@ void jpeg_encode_bgrx(const uint8_t* bgrx, int stride, int width, int height,
 void jpeg_entropy_from_blocks(const int16_t* blocks, int mcu_count_x,
                               int mcu_count_y, int width, int height,
                               int quality, bool fullcolor,
-                              std::vector<uint8_t>& out) {
+                              std::vector<uint8_t>& out,
+                              bool restart_rows) {
   uint8_t qy[64], qc[64];
   jpeg_quality_tables(quality, qy, qc);
-  write_headers(out, width, height, qy, qc, fullcolor);
-  BitWriter bw(out);
-  int dcY = 0, dcCb = 0, dcCr = 0;
+  write_headers(out, width, height, qy, qc, fullcolor,
+                restart_rows ? mcu_count_x : 0);
   const int per_mcu = fullcolor ? 3 : 6;
-  for (int m = 0; m < mcu_count_x * mcu_count_y; ++m) {
+  int dcY = 0, dcCb = 0, dcCr = 0;
+  auto encode_mcu = [&](BitWriter& bw, int m) {
     const int16_t* mcu = blocks + static_cast<size_t>(m) * per_mcu * 64;
     if (!fullcolor) {
       for (int b = 0; b < 4; ++b)
@@ -385,10 +409,47 @@ void jpeg_entropy_from_blocks(const int16_t* blocks, int mcu_count_x,
       dcCb = encode_block(bw, mcu + 1 * 64, dcCb, dc_chroma(), ac_chroma());
       dcCr = encode_block(bw, mcu + 2 * 64, dcCr, dc_chroma(), ac_chroma());
     }
+  };
+  if (!restart_rows) {
+    BitWriter bw(out);
+    for (int m = 0; m < mcu_count_x * mcu_count_y; ++m) encode_mcu(bw, m);
+    bw.flush();
+  } else {
+    int rst = 0;
+    for (int my = 0; my < mcu_count_y; ++my) {
+      if (my > 0) {
+        // byte-aligned RSTn between restart intervals; DC preds reset
+        out.push_back(0xFF);
+        out.push_back(static_cast<uint8_t>(0xD0 + (rst & 7)));
+        ++rst;
+        dcY = dcCb = dcCr = 0;
+      }
+      BitWriter bw(out);
+      for (int mx = 0; mx < mcu_count_x; ++mx)
+        encode_mcu(bw, my * mcu_count_x + mx);
+      bw.flush();   // 1-fill to byte boundary at interval end
+    }
   }
-  bw.flush();
   out.push_back(0xFF);
   out.push_back(0xD9);
+}
+
+// export the Huffman code tables for the GPU entropy kernel: packed
+// code | (size << 16), indexed by symbol
+void jpeg_export_huff(uint32_t dcl[12], uint32_t acl[256], uint32_t dcc[12],
+                      uint32_t acc[256]) {
+  const HuffTable& a = dc_luma();
+  const HuffTable& b = ac_luma();
+  const HuffTable& c = dc_chroma();
+  const HuffTable& d = ac_chroma();
+  for (int i = 0; i < 12; ++i) {
+    dcl[i] = a.code[i] | (uint32_t(a.size[i]) << 16);
+    dcc[i] = c.code[i] | (uint32_t(c.size[i]) << 16);
+  }
+  for (int i = 0; i < 256; ++i) {
+    acl[i] = b.code[i] | (uint32_t(b.size[i]) << 16);
+    acc[i] = d.code[i] | (uint32_t(d.size[i]) << 16);
+  }
 }
 
 }  // namespace hipflux

@@ -148,6 +148,24 @@ PYBIND11_MODULE(_native, m) {
       py::arg("quality") = 80, py::arg("fullcolor") = false,
       "Encode a BGRX buffer as baseline JPEG (CPU reference path).");
 
+  m.def(
+      "_jpeg_encode_restart",
+      [](py::buffer bgrx, int width, int height, int quality,
+         bool fullcolor) {
+        py::buffer_info info = bgrx.request();
+        std::vector<uint8_t> out;
+        {
+          py::gil_scoped_release rel;
+          jpeg_encode_bgrx(static_cast<const uint8_t*>(info.ptr), width * 4,
+                           width, height, quality, fullcolor, out, true);
+        }
+        return py::bytes(reinterpret_cast<const char*>(out.data()),
+                         out.size());
+      },
+      py::arg("bgrx"), py::arg("width"), py::arg("height"),
+      py::arg("quality") = 80, py::arg("fullcolor") = false,
+      "Test hook: restart-row JPEG framing (the GPU entropy structure).");
+
   // ---- H.264 stripe encoder (direct access for tests + conformance) ------
   struct PyH264 {
     h264::StripeEncoder enc;
