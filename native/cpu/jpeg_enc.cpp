@@ -434,6 +434,30 @@ void jpeg_entropy_from_blocks(const int16_t* blocks, int mcu_count_x,
   out.push_back(0xD9);
 }
 
+// public header writer for the GPU entropy path
+void jpeg_write_headers(std::vector<uint8_t>& out, int width, int height,
+                        int quality, bool fullcolor,
+                        int restart_interval) {
+  uint8_t qy[64], qc[64];
+  jpeg_quality_tables(quality, qy, qc);
+  write_headers(out, width, height, qy, qc, fullcolor, restart_interval);
+}
+
+// append one restart interval's bit run: big-endian u32 words from the
+// GPU scatter -> bytes with 1-fill padding and 0xFF00 stuffing
+void jpeg_append_row_bits(const uint32_t* words, int bits,
+                          std::vector<uint8_t>& out) {
+  int nbytes = (bits + 7) / 8;
+  out.reserve(out.size() + nbytes + nbytes / 64 + 4);
+  for (int i = 0; i < nbytes; ++i) {
+    uint8_t b = static_cast<uint8_t>(words[i / 4] >> (24 - 8 * (i % 4)));
+    if (i == nbytes - 1 && (bits & 7))
+      b |= static_cast<uint8_t>(0xFF >> (bits & 7));  // 1-fill padding
+    out.push_back(b);
+    if (b == 0xFF) out.push_back(0x00);               // byte stuffing
+  }
+}
+
 // export the Huffman code tables for the GPU entropy kernel: packed
 // code | (size << 16), indexed by symbol
 void jpeg_export_huff(uint32_t dcl[12], uint32_t acl[256], uint32_t dcc[12],

@@ -33,6 +33,15 @@ void jpeg_entropy_from_blocks(const int16_t* blocks, int mcu_count_x,
                               std::vector<uint8_t>& out,
                               bool restart_rows = false);
 
+// JFIF headers only (SOI..SOS), for the GPU entropy assembly.
+void jpeg_write_headers(std::vector<uint8_t>& out, int width, int height,
+                        int quality, bool fullcolor,
+                        int restart_interval);
+
+// One restart interval's GPU bit run -> padded, stuffed scan bytes.
+void jpeg_append_row_bits(const uint32_t* words, int bits,
+                          std::vector<uint8_t>& out);
+
 // Huffman code tables as code | (size << 16) for the GPU kernel.
 void jpeg_export_huff(uint32_t dcl[12], uint32_t acl[256], uint32_t dcc[12],
                       uint32_t acc[256]);

@@ -20,3 +20,21 @@ void launch_dct_quant(const uint8_t* plane, int pw, int ph, int pitch,
                       int stripe_mcu_count, hipStream_t stream);
 
 }  // namespace hipflux
+
+namespace hipflux {
+namespace jpeggpu {
+// GPU JPEG Huffman entropy (jpeg_entropy.hip): one workgroup per MCU
+// row, restart-interval framing. Tables: 12+256+12+256 packed
+// code|(size<<16) words (dc luma, ac luma, dc chroma, ac chroma).
+struct JRow {
+  int coeff_off;
+};
+constexpr int kJStageWords = 56;
+void launch_jpeg_entropy(const int16_t* d_coeff, const JRow* d_jobs,
+                         int n_rows, int mcux, int per_mcu,
+                         const uint32_t* d_tabs, uint32_t* d_stage,
+                         int* d_nbits, uint32_t* d_out,
+                         int out_stride_words, int* d_out_bits,
+                         hipStream_t stream);
+}  // namespace jpeggpu
+}  // namespace hipflux

@@ -40,12 +40,29 @@ class HipJpegPipeline : public EncodePipeline {
     upload_dct_tables(stream_);
     HIP_CHECK(hipMalloc(&d_rqy_, 64 * sizeof(float)));
     HIP_CHECK(hipMalloc(&d_rqc_, 64 * sizeof(float)));
+    cpu_jpeg_entropy_ = std::getenv("HIPFLUX_CPU_JPEG_ENTROPY") != nullptr;
+    {
+      uint32_t tabs[12 + 256 + 12 + 256];
+      jpeg_export_huff(tabs, tabs + 12, tabs + 268, tabs + 280);
+      HIP_CHECK(hipMalloc(&d_jtabs_, sizeof(tabs)));
+      HIP_CHECK(hipMemcpy(d_jtabs_, tabs, sizeof(tabs),
+                          hipMemcpyHostToDevice));
+    }
     alloc_for(s.capture_width, s.capture_height);
   }
 
   ~HipJpegPipeline() override {
     (void)hipStreamSynchronize(stream_);
     for (auto& kv : registered_) (void)hipHostUnregister(kv.first);
+    if (d_jtabs_) (void)hipFree(d_jtabs_);
+    if (d_jrows_) (void)hipFree(d_jrows_);
+    if (d_jstage_) (void)hipFree(d_jstage_);
+    if (d_jnbits_) (void)hipFree(d_jnbits_);
+    if (d_jout_) (void)hipFree(d_jout_);
+    if (d_joutbits_) (void)hipFree(d_joutbits_);
+    if (h_jout_) (void)hipHostFree(h_jout_);
+    if (h_joutbits_) (void)hipHostFree(h_joutbits_);
+    if (h_jrows_) (void)hipHostFree(h_jrows_);
     if (d_frame_) (void)hipFree(d_frame_);
     if (d_y_) (void)hipFree(d_y_);
     if (d_cb_) (void)hipFree(d_cb_);
@@ -103,16 +120,10 @@ class HipJpegPipeline : public EncodePipeline {
                        mcux, rows_per_stripe, stripe_mcu_count, stream_);
     }
 
-    size_t coeff_count =
-        static_cast<size_t>(mcux) * mcuy * per_mcu_real * 64;
-    HIP_CHECK(hipMemcpyAsync(h_coeff_, d_coeff_, coeff_count * sizeof(int16_t),
-                             hipMemcpyDeviceToHost, stream_));
-    HIP_CHECK(hipStreamSynchronize(stream_));
-
-    // stripe-parallel entropy packing on the CPU pool
     struct Out {
       std::vector<uint8_t> bytes;
       int y0 = 0, h = 0;
+      int row0 = 0, rows = 0;     // MCU-row range (GPU entropy path)
       bool encode = false;
     };
     std::vector<Out> outs(ctx.stripes.size());
@@ -121,19 +132,113 @@ class HipJpegPipeline : public EncodePipeline {
       outs[i].y0 = job.y0;
       outs[i].h = job.y1 - job.y0;
       outs[i].encode = job.encode;
-      if (!job.encode) continue;
-      pool_.submit([&, i] {
-        const auto& j = ctx.stripes[i];
-        int stripe_idx = j.y0 / stripe_h;
-        int rows = std::min(rows_per_stripe, mcuy - stripe_idx * rows_per_stripe);
-        const int16_t* blocks =
-            h_coeff_ + static_cast<size_t>(stripe_idx) * stripe_mcu_count *
-                           per_mcu_real * 64;
-        jpeg_entropy_from_blocks(blocks, mcux, rows, w_, j.y1 - j.y0,
-                                 ctx.jpeg_quality, fullcolor, outs[i].bytes);
-      });
     }
-    pool_.wait_all();
+
+    if (cpu_jpeg_entropy_) {
+      size_t coeff_count =
+          static_cast<size_t>(mcux) * mcuy * per_mcu_real * 64;
+      HIP_CHECK(hipMemcpyAsync(h_coeff_, d_coeff_,
+                               coeff_count * sizeof(int16_t),
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+      // stripe-parallel entropy packing on the CPU pool (restart-row
+      // framing: bit-identical to the GPU kernel's streams)
+      for (size_t i = 0; i < ctx.stripes.size(); ++i) {
+        if (!ctx.stripes[i].encode) continue;
+        pool_.submit([&, i] {
+          const auto& j = ctx.stripes[i];
+          int stripe_idx = j.y0 / stripe_h;
+          int rows = std::min(rows_per_stripe,
+                              mcuy - stripe_idx * rows_per_stripe);
+          const int16_t* blocks =
+              h_coeff_ + static_cast<size_t>(stripe_idx) *
+                             stripe_mcu_count * per_mcu_real * 64;
+          jpeg_entropy_from_blocks(blocks, mcux, rows, w_, j.y1 - j.y0,
+                                   ctx.jpeg_quality, fullcolor,
+                                   outs[i].bytes, true);
+        });
+      }
+      pool_.wait_all();
+    } else {
+      // GPU entropy: one kernel row per MCU row of every scheduled
+      // stripe; the 6 MB coefficient readback disappears entirely
+      auto* jrows = static_cast<jpeggpu::JRow*>(h_jrows_);
+      int n_rows = 0;
+      for (size_t i = 0; i < ctx.stripes.size(); ++i) {
+        if (!ctx.stripes[i].encode) continue;
+        int stripe_idx = outs[i].y0 / stripe_h;
+        int rows = std::min(rows_per_stripe,
+                            mcuy - stripe_idx * rows_per_stripe);
+        outs[i].row0 = n_rows;
+        outs[i].rows = rows;
+        size_t base = static_cast<size_t>(stripe_idx) * stripe_mcu_count *
+                      per_mcu_real * 64;
+        for (int r = 0; r < rows; ++r)
+          jrows[n_rows++].coeff_off = static_cast<int>(
+              base + static_cast<size_t>(r) * mcux * per_mcu_real * 64);
+      }
+      if (n_rows) {
+        HIP_CHECK(hipMemcpyAsync(d_jrows_, h_jrows_,
+                                 sizeof(jpeggpu::JRow) * n_rows,
+                                 hipMemcpyHostToDevice, stream_));
+        jpeggpu::launch_jpeg_entropy(
+            d_coeff_, static_cast<jpeggpu::JRow*>(d_jrows_), n_rows, mcux,
+            per_mcu_real, d_jtabs_, static_cast<uint32_t*>(d_jstage_),
+            static_cast<int*>(d_jnbits_), static_cast<uint32_t*>(d_jout_),
+            jout_stride_, static_cast<int*>(d_joutbits_), stream_);
+        // adaptive compacted readback (2x last frame's max row)
+        if (jout_copy_words_ >= jout_stride_) {
+          HIP_CHECK(hipMemcpyAsync(h_jout_, d_jout_,
+                                   (size_t)n_rows * jout_stride_ * 4,
+                                   hipMemcpyDeviceToHost, stream_));
+        } else {
+          HIP_CHECK(hipMemcpy2DAsync(
+              h_jout_, (size_t)jout_stride_ * 4, d_jout_,
+              (size_t)jout_stride_ * 4, (size_t)jout_copy_words_ * 4,
+              n_rows, hipMemcpyDeviceToHost, stream_));
+        }
+        HIP_CHECK(hipMemcpyAsync(h_joutbits_, d_joutbits_, 4 * n_rows,
+                                 hipMemcpyDeviceToHost, stream_));
+        HIP_CHECK(hipStreamSynchronize(stream_));
+        auto* outbits = static_cast<int*>(h_joutbits_);
+        int max_words = 0;
+        for (int r = 0; r < n_rows; ++r) {
+          int wds = (outbits[r] + 31) / 32 + 1;
+          max_words = std::max(max_words, wds);
+          if (wds > jout_copy_words_) {
+            HIP_CHECK(hipMemcpy(
+                static_cast<uint32_t*>(h_jout_) +
+                    (size_t)r * jout_stride_,
+                static_cast<uint32_t*>(d_jout_) +
+                    (size_t)r * jout_stride_,
+                (size_t)wds * 4, hipMemcpyDeviceToHost));
+          }
+        }
+        jout_copy_words_ = std::min(jout_stride_, max_words * 2 + 64);
+        for (size_t i = 0; i < ctx.stripes.size(); ++i) {
+          if (!ctx.stripes[i].encode) continue;
+          pool_.submit([&, i] {
+            Out& o = outs[i];
+            jpeg_write_headers(o.bytes, w_, o.h, ctx.jpeg_quality,
+                               fullcolor, mcux);
+            int rst = 0;
+            for (int r = 0; r < o.rows; ++r) {
+              if (r > 0) {
+                o.bytes.push_back(0xFF);
+                o.bytes.push_back(
+                    static_cast<uint8_t>(0xD0 + (rst++ & 7)));
+              }
+              const uint32_t* words = static_cast<uint32_t*>(h_jout_) +
+                                      (size_t)(o.row0 + r) * jout_stride_;
+              jpeg_append_row_bits(words, outbits[o.row0 + r], o.bytes);
+            }
+            o.bytes.push_back(0xFF);
+            o.bytes.push_back(0xD9);
+          });
+        }
+        pool_.wait_all();
+      }
+    }
     for (auto& o : outs) {
       if (!o.encode) continue;
       EncodedStripe s;
@@ -176,6 +281,34 @@ class HipJpegPipeline : public EncodePipeline {
     size_t coeff_bytes = mcux * mcuy * 6 * 64 * sizeof(int16_t);
     HIP_CHECK(hipMalloc(&d_coeff_, coeff_bytes));
     HIP_CHECK(hipHostMalloc(&h_coeff_, coeff_bytes, hipHostMallocDefault));
+    // GPU entropy buffers: one slot per MCU row (444 row count is the max)
+    if (d_jrows_) (void)hipFree(d_jrows_);
+    if (d_jstage_) (void)hipFree(d_jstage_);
+    if (d_jnbits_) (void)hipFree(d_jnbits_);
+    if (d_jout_) (void)hipFree(d_jout_);
+    if (d_joutbits_) (void)hipFree(d_joutbits_);
+    if (h_jout_) (void)hipHostFree(h_jout_);
+    if (h_joutbits_) (void)hipHostFree(h_joutbits_);
+    if (h_jrows_) (void)hipHostFree(h_jrows_);
+    size_t max_rows = mcuy;
+    size_t max_items = mcux * 3 > ((w + 15) / 16) * 6
+                           ? mcux * 3
+                           : ((size_t)(w + 15) / 16) * 6;
+    jout_stride_ = (int)(max_items * jpeggpu::kJStageWords);
+    HIP_CHECK(hipMalloc(&d_jrows_,
+                        sizeof(jpeggpu::JRow) * max_rows));
+    HIP_CHECK(hipMalloc(&d_jstage_, max_rows * max_items *
+                                        jpeggpu::kJStageWords * 4));
+    HIP_CHECK(hipMalloc(&d_jnbits_, max_rows * max_items * 4));
+    HIP_CHECK(hipMalloc(&d_jout_, max_rows * (size_t)jout_stride_ * 4));
+    HIP_CHECK(hipMalloc(&d_joutbits_, max_rows * 4));
+    HIP_CHECK(hipHostMalloc(&h_jout_, max_rows * (size_t)jout_stride_ * 4,
+                            hipHostMallocDefault));
+    HIP_CHECK(hipHostMalloc(&h_joutbits_, max_rows * 4,
+                            hipHostMallocDefault));
+    HIP_CHECK(hipHostMalloc(&h_jrows_, sizeof(jpeggpu::JRow) * max_rows,
+                            hipHostMallocDefault));
+    jout_copy_words_ = 1 << 30;
   }
 
   void ensure_quality(int q) {
@@ -203,6 +336,18 @@ class HipJpegPipeline : public EncodePipeline {
   uint8_t* d_cb_ = nullptr;
   uint8_t* d_cr_ = nullptr;
   int16_t* d_coeff_ = nullptr;
+  uint32_t* d_jtabs_ = nullptr;
+  void* d_jrows_ = nullptr;
+  void* d_jstage_ = nullptr;
+  void* d_jnbits_ = nullptr;
+  void* d_jout_ = nullptr;
+  void* d_joutbits_ = nullptr;
+  void* h_jout_ = nullptr;
+  void* h_joutbits_ = nullptr;
+  void* h_jrows_ = nullptr;
+  int jout_stride_ = 0;
+  int jout_copy_words_ = 1 << 30;
+  bool cpu_jpeg_entropy_ = false;
   int16_t* h_coeff_ = nullptr;
   float* d_rqy_ = nullptr;
   float* d_rqc_ = nullptr;
