@@ -266,7 +266,8 @@ PYBIND11_MODULE(_native, m) {
   m.def(
       "_pipeline_encode",
       [](const std::string& kind, py::list frames, int w, int h, int qp,
-         int stripe_h, int output_mode, bool dump) -> py::object {
+         int stripe_h, int output_mode, bool dump,
+         bool fullcolor) -> py::object {
         CaptureSettings s;
         s.capture_width = w;
         s.capture_height = h;
@@ -274,6 +275,7 @@ PYBIND11_MODULE(_native, m) {
         s.stripe_height = stripe_h;
         s.video_crf = qp;
         s.jpeg_quality = qp;
+        s.video_fullcolor = fullcolor;
         s.use_cpu = kind == "cpu";
         s.gpu_id = kind == "cpu" ? -1 : 0;
         std::unique_ptr<EncodePipeline> p;
@@ -346,6 +348,7 @@ PYBIND11_MODULE(_native, m) {
       py::arg("kind"), py::arg("frames"), py::arg("w"), py::arg("h"),
       py::arg("qp") = 26, py::arg("stripe_h") = 64,
       py::arg("output_mode") = 1, py::arg("dump") = false,
+      py::arg("fullcolor") = false,
       "Test hook: run frames through a named encode pipeline.");
 
   // ---- persistent pipeline handle for benchmarking ------------------------

@@ -152,3 +152,18 @@ def test_gpu_jpeg_entropy_matches_cpu_packer():
     assert img.size == (w, 64)
     arr = np.asarray(img.convert("L"))
     assert arr.std() > 10
+    # 4:4:4 (fullcolor) path: same byte-equality contract
+    os.environ["HIPFLUX_CPU_JPEG_ENTROPY"] = "1"
+    c444 = _native._pipeline_encode("gpu", frames[:2], w, h, 80, 64, 0,
+                                    False, True)
+    del os.environ["HIPFLUX_CPU_JPEG_ENTROPY"]
+    g444 = _native._pipeline_encode("gpu", frames[:2], w, h, 80, 64, 0,
+                                    False, True)
+    for fa, fb in zip(c444, g444):
+        for (da, ya, _, _), (db, yb, _, _) in zip(
+                sorted(fa, key=lambda t: t[1]),
+                sorted(fb, key=lambda t: t[1])):
+            assert bytes(da) == bytes(db), "fullcolor GPU entropy differs"
+    img = Image.open(io.BytesIO(bytes(sorted(g444[0],
+                                             key=lambda t: t[1])[0][0])))
+    assert img.size == (w, 64)
