@@ -87,6 +87,7 @@ struct StripeEncoder::Impl {
   uint32_t idr_pic_id = 0;
   bool need_idr = true;
   int level_idc = 0;
+  std::vector<int> prev_mv;   // per-MB quarter-pel mv of the last frame
 
   Impl(int width, int height) : w(width), h(height) {
     mbw = (w + 15) / 16;
@@ -96,6 +97,7 @@ struct StripeEncoder::Impl {
     ref.alloc(yw, yh);
     cur.alloc(yw, yh);
     src.alloc(yw, yh);
+    prev_mv.assign(static_cast<size_t>(mbw) * mbh, 0);
     // level from MB count (MaxFS) and 60 fps MaxMBPS, generous
     int fs = mbw * mbh;
     if (fs <= 1620) level_idc = 31;        // <= 480p
@@ -868,6 +870,7 @@ struct StripeEncoder::Impl {
         if (sad0 <= skip_thresh) {
           // P_Skip (skip MV is always (0,0) here: top neighbor is in
           // another slice, which forces the zero-MV rule)
+          prev_mv[static_cast<size_t>(mb_row) * mbw + mbx] = 0;
           copy_mb_from_ref(mbx, mb_row, 0, 0);
           ++ctx.skip_run;
           ctx.have_left = true;
@@ -880,31 +883,79 @@ struct StripeEncoder::Impl {
           if (stats) ++stats->mb_skip;
           continue;
         }
-        // integer diamond search around (0,0) (unit steps now that the
-        // chroma MC interpolates), then half-pel refinement
+        // integer diamond search seeded at (0,0) and at the previous
+        // frame's MV for this MB (tracks sustained motion); a coarse
+        // +-16 step-4 scan acquires fast new motion when the local walk
+        // fails, followed by another walk
         int best_mvx = 0, best_mvy = 0;
         long best = sad0;
+        const size_t mb_idx = static_cast<size_t>(mb_row) * mbw + mbx;
+        {
+          int pm = prev_mv[mb_idx];
+          int pvx = (static_cast<int16_t>(pm & 0xFFFF)) >> 2;
+          int pvy = (pm >> 16) >> 2;
+          pvx = std::clamp(pvx, -48, 48);
+          pvy = std::clamp(pvy, -48, 48);
+          if ((pvx | pvy) != 0) {
+            // fine grid around the hint (mirrors the MFMA kernel's
+            // predictor pass): acquisition beyond the local walk
+            for (int dy = -8; dy <= 8; dy += 2)
+              for (int dx = -8; dx <= 8; dx += 2) {
+                int cx2 = pvx + dx, cy2 = pvy + dy;
+                if (!mv_window_ok(x0, y0, cx2 * 4, cy2 * 4)) continue;
+                long s = sad16(src.y.row(y0) + x0, src.y.pitch,
+                               ref.y.row(y0 + cy2) + x0 + cx2,
+                               ref.y.pitch);
+                if (s < best) {
+                  best = s;
+                  best_mvx = cx2;
+                  best_mvy = cy2;
+                }
+              }
+          }
+        }
         static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
                                       {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
-        for (int iter = 0; iter < 16; ++iter) {
-          int bmx = best_mvx, bmy = best_mvy;
-          bool improved = false;
-          for (auto& p : pat) {
-            int mx = bmx + p[0], my = bmy + p[1];
-            if (std::abs(mx) > 16 || std::abs(my) > 16) continue;
-            if (!(mx == 0 && my == 0) &&
-                !mv_window_ok(x0, y0, mx * 4, my * 4))
-              continue;
-            long s = sad16(src.y.row(y0) + x0, src.y.pitch,
-                           ref.y.row(y0 + my) + x0 + mx, ref.y.pitch);
-            if (s < best) {
-              best = s;
-              best_mvx = mx;
-              best_mvy = my;
-              improved = true;
+        auto walk = [&]() {
+          for (int iter = 0; iter < 16; ++iter) {
+            int bmx = best_mvx, bmy = best_mvy;
+            bool improved = false;
+            for (auto& p : pat) {
+              int mx = bmx + p[0], my = bmy + p[1];
+              if (std::abs(mx) > 48 || std::abs(my) > 48) continue;
+              if (!(mx == 0 && my == 0) &&
+                  !mv_window_ok(x0, y0, mx * 4, my * 4))
+                continue;
+              long s = sad16(src.y.row(y0) + x0, src.y.pitch,
+                             ref.y.row(y0 + my) + x0 + mx, ref.y.pitch);
+              if (s < best) {
+                best = s;
+                best_mvx = mx;
+                best_mvy = my;
+                improved = true;
+              }
             }
+            if (!improved) break;
           }
-          if (!improved) break;
+        };
+        walk();
+        if (best > 2 * skip_thresh) {
+          // coarse acquisition over +-16 whenever the local match is
+          // mediocre (a permissive accept must not mask real motion),
+          // then walk from the best cell
+          for (int cy = -16; cy <= 16; cy += 4)
+            for (int cx = -16; cx <= 16; cx += 4) {
+              if ((cx | cy) == 0) continue;
+              if (!mv_window_ok(x0, y0, cx * 4, cy * 4)) continue;
+              long s = sad16(src.y.row(y0) + x0, src.y.pitch,
+                             ref.y.row(y0 + cy) + x0 + cx, ref.y.pitch);
+              if (s < best) {
+                best = s;
+                best_mvx = cx;
+                best_mvy = cy;
+              }
+            }
+          walk();
         }
         // half- then quarter-pel refinement rings; skipped when the MB
         // is headed to intra regardless
@@ -926,6 +977,8 @@ struct StripeEncoder::Impl {
         }
         if (best <= inter_thresh) {
           // P_L0_16x16 with coded residual (cbp may still come out 0)
+          prev_mv[mb_idx] =
+              (best_q_x & 0xFFFF) | (best_q_y << 16);
           flush_skip_run(b, ctx);
           encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
           ctx.have_left = true;
@@ -940,6 +993,10 @@ struct StripeEncoder::Impl {
           }
           if (stats) ++stats->mb_inter;
         } else {
+          // keep the best-found mv as a tracking hint across the intra
+          // fallback (fast new motion locks on within a frame or two)
+          prev_mv[mb_idx] = ((best_mvx * 4) & 0xFFFF) |
+                            ((best_mvy * 4) << 16);
           flush_skip_run(b, ctx);
           encode_i16(b, mbx, mb_row, qp, true, ctx, tot);
           ctx.have_left = true;

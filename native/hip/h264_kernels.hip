@@ -278,7 +278,8 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
     }
     leftreg = newleft;
 
-    if (lane == 0) M[0] = kIntra | (luma_mode << 2);
+    if (lane == 0)
+      M[0] = kIntra | (luma_mode << 2) | (m0_cur & 0x3FFF00);
     have_left = true;
     }
 
@@ -745,6 +746,29 @@ __global__ void __launch_bounds__(64) k_h264_me(
   } else {
     int best = sad0;
     int bmx = 0, bmy = 0;
+    // seed with the previous frame's MV/hint (meta still holds it)
+    {
+      int pm0 = meta[mb_index * kMetaPerMb + 0];
+      int pm1 = meta[mb_index * kMetaPerMb + 1];
+      int pvx, pvy;
+      if ((pm0 & 3) == kInter) {
+        pvx = ((int)(short)(pm1 & 0xFFFF)) >> 2;
+        pvy = (pm1 >> 16) >> 2;
+      } else {
+        pvx = ((pm0 >> 8) & 127) - 64;
+        pvy = ((pm0 >> 15) & 127) - 64;
+      }
+      pvx = max(-48, min(48, pvx));
+      pvy = max(-48, min(48, pvy));
+      if ((pvx | pvy) != 0 && window_ok(pvx * 4, pvy * 4)) {
+        int s = sad_at(pvx, pvy);
+        if (s < best) {
+          best = s;
+          bmx = pvx;
+          bmy = pvy;
+        }
+      }
+    }
     static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
                                   {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
     for (int iter = 0; iter < 16; ++iter) {
@@ -789,7 +813,9 @@ __global__ void __launch_bounds__(64) k_h264_me(
       mode = kIntra;
   }
   if (lane == 0) {
-    meta[mb_index * kMetaPerMb + 0] = mode;
+    int hx = max(-48, min(48, bqx >> 2)) + 64;
+    int hy = max(-48, min(48, bqy >> 2)) + 64;
+    meta[mb_index * kMetaPerMb + 0] = mode | (hx << 8) | (hy << 15);
     meta[mb_index * kMetaPerMb + 1] = (bqx & 0xFFFF) | (bqy << 16);
   }
 }
@@ -863,81 +889,107 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     }
   }
 
-  // candidate grid: 81 candidates in 6 groups of up to 16
+  // candidate grids: 81 candidates (9x9) in 6 groups of up to 16, scored
+  // by MFMA SSD around an arbitrary center and step
   const int NC = 81;
   int best_score = INT_MAX, best_mvx = 0, best_mvy = 0;
   const int col = lane & 15, kg = lane >> 4;
 
-  for (int g0 = 0; g0 < NC; g0 += 16) {
-    // this lane's candidate (column) in the group
-    int cand = g0 + col;
-    int mx = 0, my = 0;
-    bool valid = cand < NC;
-    if (valid) {
-      mx = (cand % 9) * 2 - 8;
-      my = (cand / 9) * 2 - 8;
-      if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
-          y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1)
-        valid = false;
-    }
-    i32x4 cross = {0, 0, 0, 0};
-    i32x4 norm = {0, 0, 0, 0};
-    for (int q = 0; q < 4; ++q) {
-      // B fragment: lane holds B[k = kg*16 + j][col] = candidate col's
-      // chunk-q byte (row kg of chunk, cols 0..15 -> j)
-      int8_t bb[16];
+  auto score_grid = [&](int cx0, int cy0, int step) {
+    for (int g0 = 0; g0 < NC; g0 += 16) {
+      int cand = g0 + col;
+      int mx = 0, my = 0;
+      bool valid = cand < NC;
       if (valid) {
-        const uint8_t* rp =
-            refY + (size_t)(y0 + my + 4 * q + kg) * ypitch + x0 + mx;
-        for (int j = 0; j < 16; ++j) bb[j] = (int8_t)((int)rp[j] - 128);
-      } else {
-        for (int j = 0; j < 16; ++j) bb[j] = 0;
+        mx = cx0 + ((cand % 9) - 4) * step;
+        my = cy0 + ((cand / 9) - 4) * step;
+        if (x0 + mx < 0 || x0 + mx + 16 > frame_w_mb16 ||
+            y0 + my < job.stripe_y0 || y0 + my + 16 > job.stripe_y1)
+          valid = false;
       }
-      int words[4];
-      for (int t = 0; t < 4; ++t)
-        words[t] = (uint8_t)bb[4 * t] | ((uint8_t)bb[4 * t + 1] << 8) |
-                   ((uint8_t)bb[4 * t + 2] << 16) |
-                   ((uint8_t)bb[4 * t + 3] << 24);
-      i32x4 bfrag = i32x4{words[0], words[1], words[2], words[3]};
-      cross = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[q], bfrag, cross,
-                                                    0, 0, 0);
-      norm = __builtin_amdgcn_mfma_i32_16x16x64_i8(bfrag, bfrag, norm,
-                                                   0, 0, 0);
-    }
-    // D mapping (16x16): col = lane&15, row = (lane>>4)*4 + i.
-    // X_n lives in row 0 -> lanes kg==0, acc idx 0, col n.
-    // R2_n is the diagonal (n,n) -> lane ((n>>2)<<4)|n, acc idx n&3.
-    // Gather per-candidate scores into the lane owning candidate n = col.
-    for (int n_base = 0; n_base < 16; ++n_base) {
-      // broadcast X_n and R2_n from their owner lanes to everyone
-      int xl = __shfl(cross[0], n_base);                  // lane n_base
-      int diag_lane = ((n_base >> 2) << 4) | n_base;
-      int rl;
-      switch (n_base & 3) {
-        case 0: rl = __shfl(norm[0], diag_lane); break;
-        case 1: rl = __shfl(norm[1], diag_lane); break;
-        case 2: rl = __shfl(norm[2], diag_lane); break;
-        default: rl = __shfl(norm[3], diag_lane); break;
+      i32x4 cross = {0, 0, 0, 0};
+      i32x4 norm = {0, 0, 0, 0};
+      for (int q = 0; q < 4; ++q) {
+        // B fragment: lane holds B[k = kg*16 + j][col] = candidate col's
+        // chunk-q byte (row kg of chunk, cols 0..15 -> j)
+        int8_t bb[16];
+        if (valid) {
+          const uint8_t* rp =
+              refY + (size_t)(y0 + my + 4 * q + kg) * ypitch + x0 + mx;
+          for (int j = 0; j < 16; ++j) bb[j] = (int8_t)((int)rp[j] - 128);
+        } else {
+          for (int j = 0; j < 16; ++j) bb[j] = 0;
+        }
+        int words[4];
+        for (int t = 0; t < 4; ++t)
+          words[t] = (uint8_t)bb[4 * t] | ((uint8_t)bb[4 * t + 1] << 8) |
+                     ((uint8_t)bb[4 * t + 2] << 16) |
+                     ((uint8_t)bb[4 * t + 3] << 24);
+        i32x4 bfrag = i32x4{words[0], words[1], words[2], words[3]};
+        cross = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[q], bfrag,
+                                                      cross, 0, 0, 0);
+        norm = __builtin_amdgcn_mfma_i32_16x16x64_i8(bfrag, bfrag, norm,
+                                                     0, 0, 0);
       }
-      int cand2 = g0 + n_base;
-      if (cand2 < NC && lane == 0) {
-        int mx2 = (cand2 % 9) * 2 - 8, my2 = (cand2 / 9) * 2 - 8;
-        bool v2 = !(x0 + mx2 < 0 || x0 + mx2 + 16 > frame_w_mb16 ||
-                    y0 + my2 < job.stripe_y0 ||
-                    y0 + my2 + 16 > job.stripe_y1);
-        if (v2) {
-          int ssd = s2 - 2 * xl + rl;
-          // small center bias keeps MVs compact on flat content
-          int score = ssd + (abs(mx2) + abs(my2)) * 4;
-          if (score < best_score) {
-            best_score = score;
-            best_mvx = mx2;
-            best_mvy = my2;
+      // D mapping (16x16): X_n in row 0 (lane n, acc 0); R2_n on the
+      // diagonal (lane ((n>>2)<<4)|n, acc n&3). Reduce on lane 0.
+      for (int n_base = 0; n_base < 16; ++n_base) {
+        int xl = __shfl(cross[0], n_base);
+        int diag_lane = ((n_base >> 2) << 4) | n_base;
+        int rl;
+        switch (n_base & 3) {
+          case 0: rl = __shfl(norm[0], diag_lane); break;
+          case 1: rl = __shfl(norm[1], diag_lane); break;
+          case 2: rl = __shfl(norm[2], diag_lane); break;
+          default: rl = __shfl(norm[3], diag_lane); break;
+        }
+        int cand2 = g0 + n_base;
+        if (cand2 < NC && lane == 0) {
+          int mx2 = cx0 + ((cand2 % 9) - 4) * step;
+          int my2 = cy0 + ((cand2 / 9) - 4) * step;
+          bool v2 = !(x0 + mx2 < 0 || x0 + mx2 + 16 > frame_w_mb16 ||
+                      y0 + my2 < job.stripe_y0 ||
+                      y0 + my2 + 16 > job.stripe_y1);
+          if (v2) {
+            int ssd = s2 - 2 * xl + rl;
+            // small center bias keeps MVs compact on flat content
+            int score = ssd + (abs(mx2) + abs(my2)) * 4;
+            if (score < best_score) {
+              best_score = score;
+              best_mvx = mx2;
+              best_mvy = my2;
+            }
           }
         }
       }
     }
+  };
+
+  // pass 1: fine grid at (0,0)
+  score_grid(0, 0, 2);
+  // pass 2: fine grid at the previous frame's MV/hint for this MB (meta
+  // still holds last frame's values here) — tracks sustained motion
+  // beyond +-8 even across intra fallbacks
+  {
+    int pm0 = meta[mb_index * kMetaPerMb + 0];
+    int pm1 = meta[mb_index * kMetaPerMb + 1];
+    int pvx = 0, pvy = 0;
+    if ((pm0 & 3) == kInter) {
+      pvx = ((int)(short)(pm1 & 0xFFFF)) >> 2;
+      pvy = (pm1 >> 16) >> 2;
+    } else {
+      pvx = ((pm0 >> 8) & 127) - 64;    // intra/skip hint (bits 8..21)
+      pvy = ((pm0 >> 15) & 127) - 64;
+    }
+    pvx = max(-48, min(48, pvx));
+    pvy = max(-48, min(48, pvy));
+    if ((pvx | pvy) != 0 && (abs(pvx) > 2 || abs(pvy) > 2))
+      score_grid(pvx, pvy, 2);
   }
+  // pass 3: coarse acquisition (+-16, step 4) when nothing fits yet —
+  // first frame of a fast scroll
+  if (__shfl(best_score, 0) > 256 * 180) score_grid(0, 0, 4);
+
   best_score = __shfl(best_score, 0);
   best_mvx = __shfl(best_mvx, 0);
   best_mvy = __shfl(best_mvy, 0);
@@ -976,6 +1028,7 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   const int inter_thresh = 6 * skip_thresh;
   int sad0 = sad_at(0, 0);
   int mode, oqx = 0, oqy = 0;
+  int hintx = 0, hinty = 0;
   if (sad0 <= skip_thresh) {
     mode = kSkip;
   } else {
@@ -987,16 +1040,21 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     static const int pat[8][2] = {{-1, 0}, {1, 0},  {0, -1}, {0, 1},
                                   {-1, -1}, {1, 1}, {-1, 1}, {1, -1}};
     if (best_sad <= 2 * inter_thresh) {
-      int cqx = bqx, cqy = bqy;
-      for (int pi = 0; pi < 8; ++pi) {
-        int qx = cqx + 4 * pat[pi][0], qy = cqy + 4 * pat[pi][1];
-        if (!window_ok(qx, qy)) continue;
-        int s = sad_at(qx >> 2, qy >> 2);
-        if (s < best_sad) {
-          best_sad = s;
-          bqx = qx;
-          bqy = qy;
+      for (int iter = 0; iter < 8; ++iter) {
+        int cqx = bqx, cqy = bqy;
+        bool improved = false;
+        for (int pi = 0; pi < 8; ++pi) {
+          int qx = cqx + 4 * pat[pi][0], qy = cqy + 4 * pat[pi][1];
+          if (!window_ok(qx, qy)) continue;
+          int s = sad_at(qx >> 2, qy >> 2);
+          if (s < best_sad) {
+            best_sad = s;
+            bqx = qx;
+            bqy = qy;
+            improved = true;
+          }
         }
+        if (!improved) break;
       }
       for (int step = 2; step >= 1; --step) {
         int cqx2 = bqx, cqy2 = bqy;
@@ -1019,9 +1077,16 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     } else {
       mode = kIntra;
     }
+    hintx = bqx >> 2;
+    hinty = bqy >> 2;
   }
   if (lane == 0) {
-    meta[mb_index * kMetaPerMb + 0] = mode;
+    // best-found integer mv persists as a tracking hint even when the MB
+    // goes intra; the luma row wave preserves these bits on its M[0]
+    // rewrite
+    int hx = max(-48, min(48, hintx)) + 64;
+    int hy = max(-48, min(48, hinty)) + 64;
+    meta[mb_index * kMetaPerMb + 0] = mode | (hx << 8) | (hy << 15);
     meta[mb_index * kMetaPerMb + 1] = (oqx & 0xFFFF) | (oqy << 16);
   }
 }

@@ -297,3 +297,52 @@ def test_quarterpel_mvs():
                or (mv[1] & 3) in (1, 3)]
     assert quarter, "no quarter-pel MVs coded for the quarter shift"
 
+
+
+def test_fast_scroll_motion_lock():
+    """12 px/frame scroll (beyond the ±8 fine grid): the ±16 coarse
+    acquisition finds it and the temporal predictor/hint keeps tracking;
+    P frames must be mostly inter with mv x = -48 quarter-pel, stay
+    bit-exact, and cost far less than intra refreshes. (Acquisition range
+    is ±16 like the reference x264 default; a resolution pyramid for
+    faster motion is round-2 — profiles/NOTES.md.)"""
+    from h264_ref_decoder import Decoder
+    w, h, n = 320, 96, 4
+    rng = np.random.default_rng(61)
+    base = rng.integers(0, 256, (h, w, 4), dtype=np.uint8).astype(
+        np.float32)
+    for _ in range(2):
+        base = (base + np.roll(base, 1, 1) + np.roll(base, -1, 1) +
+                np.roll(base, 1, 0) + np.roll(base, -1, 0)) / 5
+    f0 = base.astype(np.uint8)
+    f0[:, :, 3] = 255
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = []
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs.append(tuple(ctx["left_mv"]))
+
+    enc = hipflux.H264Encoder(w, h)
+    stream = b""
+    sizes = []
+    recons = []
+    for i in range(n):
+        img = np.ascontiguousarray(np.roll(f0, 12 * i, axis=1))
+        r = enc.encode(img.tobytes(), qp=24, idr=(i == 0))
+        stream += r["data"]
+        sizes.append(len(r["data"]))
+        recons.append(recon_planes(enc, w, h))
+    d = MvDecoder()
+    frames = d.decode(stream)
+    assert len(frames) == n
+    for i, (dec, rec) in enumerate(zip(frames, recons)):
+        for a, b, name in zip(dec, rec, "y cb cr".split()):
+            assert np.array_equal(a, b), f"frame {i} plane {name} mismatch"
+    locked = [mv for mv in d.mvs if mv[0] == -48]
+    assert len(locked) > len(d.mvs) * 0.4, \
+        f"scroll not locked: {len(locked)}/{len(d.mvs)} at -48"
+    assert max(sizes[2:]) < sizes[0] * 0.6, f"P frames too large: {sizes}"
