@@ -88,6 +88,7 @@ struct StripeEncoder::Impl {
   bool need_idr = true;
   int level_idc = 0;
   std::vector<int> prev_mv;   // per-MB quarter-pel mv of the last frame
+  std::vector<uint8_t> prev_bad;  // MB was hopeless-intra last frame
 
   Impl(int width, int height) : w(width), h(height) {
     mbw = (w + 15) / 16;
@@ -98,6 +99,7 @@ struct StripeEncoder::Impl {
     cur.alloc(yw, yh);
     src.alloc(yw, yh);
     prev_mv.assign(static_cast<size_t>(mbw) * mbh, 0);
+    prev_bad.assign(static_cast<size_t>(mbw) * mbh, 0);
     // level from MB count (MaxFS) and 60 fps MaxMBPS, generous
     int fs = mbw * mbh;
     if (fs <= 1620) level_idc = 31;        // <= 480p
@@ -896,7 +898,7 @@ struct StripeEncoder::Impl {
           int pvy = (pm >> 16) >> 2;
           pvx = std::clamp(pvx, -48, 48);
           pvy = std::clamp(pvy, -48, 48);
-          if ((pvx | pvy) != 0) {
+          if ((pvx | pvy) != 0 && !prev_bad[mb_idx]) {
             // fine grid around the hint (mirrors the MFMA kernel's
             // predictor pass): acquisition beyond the local walk
             for (int dy = -8; dy <= 8; dy += 2)
@@ -939,7 +941,8 @@ struct StripeEncoder::Impl {
           }
         };
         walk();
-        if (best > 2 * skip_thresh) {
+        if (best > 2 * skip_thresh &&
+            !(prev_bad[mb_idx] && (frame_num & 7) != 0)) {
           // coarse acquisition over +-16 whenever the local match is
           // mediocre (a permissive accept must not mask real motion),
           // then walk from the best cell
@@ -975,10 +978,12 @@ struct StripeEncoder::Impl {
             }
           }
         }
+        prev_bad[mb_idx] = best > 3 * inter_thresh ? 1 : 0;
         if (best <= inter_thresh) {
           // P_L0_16x16 with coded residual (cbp may still come out 0)
           prev_mv[mb_idx] =
-              (best_q_x & 0xFFFF) | (best_q_y << 16);
+              (best_q_x & 0xFFFF) |
+              static_cast<int>(static_cast<uint32_t>(best_q_y) << 16);
           flush_skip_run(b, ctx);
           encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
           ctx.have_left = true;
@@ -996,7 +1001,8 @@ struct StripeEncoder::Impl {
           // keep the best-found mv as a tracking hint across the intra
           // fallback (fast new motion locks on within a frame or two)
           prev_mv[mb_idx] = ((best_mvx * 4) & 0xFFFF) |
-                            ((best_mvy * 4) << 16);
+                            static_cast<int>(
+                                static_cast<uint32_t>(best_mvy * 4) << 16);
           flush_skip_run(b, ctx);
           encode_i16(b, mbx, mb_row, qp, true, ctx, tot);
           ctx.have_left = true;

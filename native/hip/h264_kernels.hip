@@ -279,7 +279,7 @@ __device__ void luma_row(const uint8_t* __restrict__ srcY, int ypitch, int w,
     leftreg = newleft;
 
     if (lane == 0)
-      M[0] = kIntra | (luma_mode << 2) | (m0_cur & 0x3FFF00);
+      M[0] = kIntra | (luma_mode << 2) | (m0_cur & 0x7FFF00);
     have_left = true;
     }
 
@@ -816,7 +816,8 @@ __global__ void __launch_bounds__(64) k_h264_me(
     int hx = max(-48, min(48, bqx >> 2)) + 64;
     int hy = max(-48, min(48, bqy >> 2)) + 64;
     meta[mb_index * kMetaPerMb + 0] = mode | (hx << 8) | (hy << 15);
-    meta[mb_index * kMetaPerMb + 1] = (bqx & 0xFFFF) | (bqy << 16);
+    meta[mb_index * kMetaPerMb + 1] =
+        (bqx & 0xFFFF) | (int)((unsigned)bqy << 16);
   }
 }
 
@@ -970,9 +971,11 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   // pass 2: fine grid at the previous frame's MV/hint for this MB (meta
   // still holds last frame's values here) — tracks sustained motion
   // beyond +-8 even across intra fallbacks
+  bool prev_hopeless;
   {
     int pm0 = meta[mb_index * kMetaPerMb + 0];
     int pm1 = meta[mb_index * kMetaPerMb + 1];
+    prev_hopeless = (pm0 >> 22) & 1;
     int pvx = 0, pvy = 0;
     if ((pm0 & 3) == kInter) {
       pvx = ((int)(short)(pm1 & 0xFFFF)) >> 2;
@@ -987,8 +990,13 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
       score_grid(pvx, pvy, 2);
   }
   // pass 3: coarse acquisition (+-16, step 4) when nothing fits yet —
-  // first frame of a fast scroll
-  if (__shfl(best_score, 0) > 256 * 180) score_grid(0, 0, 4);
+  // first frame of a fast scroll. MBs whose LAST frame was hopeless
+  // (intra with nothing remotely matching — e.g. noise) skip the coarse
+  // pass except for a periodic probe every 8 frames, so worst-case
+  // content does not pay for acquisition it can never use.
+  if (!(prev_hopeless && (job.frame_num & 7) != 0) &&
+      __shfl(best_score, 0) > 256 * 180)
+    score_grid(0, 0, 4);
 
   best_score = __shfl(best_score, 0);
   best_mvx = __shfl(best_mvx, 0);
@@ -1080,14 +1088,18 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     hintx = bqx >> 2;
     hinty = bqy >> 2;
   }
+  bool hopeless = mode == kIntra && __shfl(best_score, 0) > 256 * 1200;
+  if (hopeless) hintx = hinty = 0;
   if (lane == 0) {
     // best-found integer mv persists as a tracking hint even when the MB
     // goes intra; the luma row wave preserves these bits on its M[0]
     // rewrite
     int hx = max(-48, min(48, hintx)) + 64;
     int hy = max(-48, min(48, hinty)) + 64;
-    meta[mb_index * kMetaPerMb + 0] = mode | (hx << 8) | (hy << 15);
-    meta[mb_index * kMetaPerMb + 1] = (oqx & 0xFFFF) | (oqy << 16);
+    meta[mb_index * kMetaPerMb + 0] =
+        mode | (hx << 8) | (hy << 15) | ((hopeless ? 1 : 0) << 22);
+    meta[mb_index * kMetaPerMb + 1] =
+        (oqx & 0xFFFF) | (int)((unsigned)oqy << 16);
   }
 }
 

@@ -243,7 +243,7 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
     if (w == 0 && lane == 0) {
       // preserve the ME tracking hint (bits 8..21) across the rewrite
       int pm0 = i_slice ? 0 : M[0];
-      M[0] = kIntra | (luma_mode << 2) | (pm0 & 0x3FFF00);
+      M[0] = kIntra | (luma_mode << 2) | (pm0 & 0x7FFF00);
     }
     __syncthreads();
     have_left = true;
