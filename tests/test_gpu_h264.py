@@ -369,3 +369,44 @@ def test_gpu_wide_frame_segmented_slices():
     sy = np.frombuffer(src_y, np.uint8).reshape(h, w)
     p = psnr(decoded[0][0], sy)
     assert p > 34, f"segmented wide-frame IDR PSNR {p:.1f}"
+
+
+def test_gpu_fast_scroll_motion_lock():
+    """12 px/frame scroll through the GPU pipeline: coarse acquisition +
+    the meta hint channel must lock most MBs onto mv x = -48 quarter-pel
+    within a couple of P frames."""
+    require_gpu()
+    w, h, n = 320, 96, 5
+    rng = np.random.default_rng(61)
+    base = rng.integers(0, 256, (h, w, 4), dtype=np.uint8).astype(
+        np.float32)
+    for _ in range(2):
+        base = (base + np.roll(base, 1, 1) + np.roll(base, -1, 1) +
+                np.roll(base, 1, 0) + np.roll(base, -1, 0)) / 5
+    f0 = base.astype(np.uint8)
+    f0[:, :, 3] = 255
+    frames = [np.ascontiguousarray(np.roll(f0, 12 * i, axis=1))
+              for i in range(n)]
+    out = _native._pipeline_encode("gpu", frames, w, h, 24, 96, 1)
+    rows = reassemble(out)
+
+    class MvDecoder(Decoder):
+        def __init__(self):
+            super().__init__()
+            self.mvs = []
+
+        def decode_p16(self, br, mbx, mby, ctx, qp):
+            super().decode_p16(br, mbx, mby, ctx, qp)
+            self.mvs.append(tuple(ctx["left_mv"]))
+
+    late_mvs = []
+    for y0, stream in rows.items():
+        d = MvDecoder()
+        decoded = d.decode(bytes(stream))
+        assert len(decoded) == n
+        # frames 3..n: acquisition done, tracking should hold
+        per_frame = len(d.mvs) // (n - 1) if d.mvs else 0
+        late_mvs += d.mvs[2 * per_frame:]
+    locked = [mv for mv in late_mvs if mv[0] == -48]
+    assert late_mvs and len(locked) > len(late_mvs) * 0.4, \
+        f"GPU scroll not locked: {len(locked)}/{len(late_mvs)}"
