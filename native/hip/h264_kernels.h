@@ -8,8 +8,12 @@
 namespace hipflux {
 namespace h264gpu {
 
+void launch_downsample2(const uint8_t* src, int spitch, int sw, int sh,
+                        uint8_t* dst, int dpitch, hipStream_t stream);
+
 void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
-                    const uint8_t* refY, int mbw, int n_jobs,
+                    const uint8_t* refY, const uint8_t* srcY2,
+                    const uint8_t* refY2, int ypitch2, int mbw, int n_jobs,
                     const RowJob* d_jobs, int* d_meta, hipStream_t stream,
                     bool use_mfma = true);
 

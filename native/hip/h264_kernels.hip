@@ -836,7 +836,8 @@ using i32x4 = __attribute__((__vector_size__(16))) int;
 
 __global__ void __launch_bounds__(64) k_h264_me_mfma(
     const uint8_t* __restrict__ srcY, int ypitch, int w, int h,
-    const uint8_t* __restrict__ refY, int mbw, int seg_max,
+    const uint8_t* __restrict__ refY, const uint8_t* __restrict__ srcY2,
+    const uint8_t* __restrict__ refY2, int ypitch2, int mbw, int seg_max,
     int frame_w_mb16, const RowJob* __restrict__ jobs,
     int* __restrict__ meta) {
   const int job_idx = blockIdx.x / seg_max;
@@ -989,14 +990,60 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     if ((pvx | pvy) != 0 && (abs(pvx) > 2 || abs(pvy) > 2))
       score_grid(pvx, pvy, 2);
   }
-  // pass 3: coarse acquisition (+-16, step 4) when nothing fits yet —
-  // first frame of a fast scroll. MBs whose LAST frame was hopeless
-  // (intra with nothing remotely matching — e.g. noise) skip the coarse
-  // pass except for a periodic probe every 8 frames, so worst-case
-  // content does not pay for acquisition it can never use.
+  // pass 3: pyramid acquisition when nothing fits yet — first frame of
+  // fast new motion. A +-12 step-1 SAD search on the QUARTER-res luma
+  // pyramid covers +-48 at full res with correlation intact at that
+  // scale; the winner seeds a fine full-res grid. MBs whose LAST frame
+  // was hopeless (noise) skip this except a 1-in-8-frame probe.
   if (!(prev_hopeless && (job.frame_num & 7) != 0) &&
-      __shfl(best_score, 0) > 256 * 180)
-    score_grid(0, 0, 4);
+      __shfl(best_score, 0) > 256 * 180) {
+    // this MB at quarter res: 4x4 proxy block
+    const int qx0 = x0 >> 2, qy0 = y0 >> 2;
+    const int qw = frame_w_mb16 >> 2;
+    const int qs_y0 = job.stripe_y0 >> 2, qs_y1 = job.stripe_y1 >> 2;
+    int px[4];
+    {
+      int rr = lane & 3;
+      const uint8_t* sr = srcY2 + (size_t)min(qy0 + rr, (h >> 2) - 1) *
+                                      ypitch2;
+      for (int j = 0; j < 4; ++j)
+        px[j] = sr[min(qx0 + j, qw - 1)];
+    }
+    int bs = INT_MAX, bmx2 = 0, bmy2 = 0;
+    // 25x25 candidates, distributed over lanes (lane covers cands with
+    // index % 16 == lane>>2 pattern via strided loop)
+    for (int ci = lane >> 2; ci < 625; ci += 16) {
+      int mx = (ci % 25) - 12, my = (ci / 25) - 12;
+      if (qx0 + mx < 0 || qx0 + mx + 4 > qw || qy0 + my < qs_y0 ||
+          qy0 + my + 4 > qs_y1)
+        continue;
+      const uint8_t* rp =
+          refY2 + (size_t)(qy0 + my + (lane & 3)) * ypitch2 + qx0 + mx;
+      int sad = abs(px[0] - rp[0]) + abs(px[1] - rp[1]) +
+                abs(px[2] - rp[2]) + abs(px[3] - rp[3]);
+      // sum the 4 row-partials within the 4-lane group
+      sad += __shfl_xor(sad, 1);
+      sad += __shfl_xor(sad, 2);
+      if (sad < bs) {
+        bs = sad;
+        bmx2 = mx;
+        bmy2 = my;
+      }
+    }
+    // wave-reduce the best candidate
+    for (int d = 4; d < 64; d <<= 1) {
+      int os = __shfl_xor(bs, d);
+      int ox = __shfl_xor(bmx2, d);
+      int oy = __shfl_xor(bmy2, d);
+      if (os < bs || (os == bs && (oy < bmy2 || (oy == bmy2 && ox < bmx2)))) {
+        bs = os;
+        bmx2 = ox;
+        bmy2 = oy;
+      }
+    }
+    // fine full-res verification grid at the pyramid seed
+    if (bs < INT_MAX) score_grid(bmx2 * 4, bmy2 * 4, 2);
+  }
 
   best_score = __shfl(best_score, 0);
   best_mvx = __shfl(best_mvx, 0);
@@ -1106,15 +1153,16 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
 // ---------------------------------------------------------------------------
 // host launchers
 void launch_h264_me(const uint8_t* srcY, int ypitch, int w, int h,
-                    const uint8_t* refY, int mbw, int n_jobs,
+                    const uint8_t* refY, const uint8_t* srcY2,
+                    const uint8_t* refY2, int ypitch2, int mbw, int n_jobs,
                     const RowJob* d_jobs, int* d_meta, hipStream_t stream,
                     bool use_mfma) {
   if (n_jobs == 0) return;
   const int seg_max = mbw < kMaxSegMbw ? mbw : kMaxSegMbw;
   if (use_mfma) {
     hipLaunchKernelGGL(k_h264_me_mfma, dim3(n_jobs * seg_max), dim3(64), 0,
-                       stream, srcY, ypitch, w, h, refY, mbw, seg_max,
-                       mbw * 16, d_jobs, d_meta);
+                       stream, srcY, ypitch, w, h, refY, srcY2, refY2,
+                       ypitch2, mbw, seg_max, mbw * 16, d_jobs, d_meta);
   } else {
     hipLaunchKernelGGL(k_h264_me, dim3(n_jobs * seg_max), dim3(64), 0,
                        stream, srcY, ypitch, w, h, refY, mbw, seg_max,
@@ -1133,6 +1181,31 @@ void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
   hipLaunchKernelGGL(k_h264_rows, dim3(n_jobs), dim3(128), 0, stream, srcY,
                      srcCb, srcCr, ypitch, cpitch, w, h, refY, refCb, refCr,
                      curY, curCb, curCr, mbw, d_jobs, d_levels, d_meta);
+}
+
+
+// ---------------------------------------------------------------------------
+// 2x2 box downsample (luma pyramid for ME acquisition). One thread per
+// output pixel.
+__global__ void k_downsample2(const uint8_t* __restrict__ src, int spitch,
+                              int sw, int sh, uint8_t* __restrict__ dst,
+                              int dpitch) {
+  int x = blockIdx.x * blockDim.x + threadIdx.x;
+  int y = blockIdx.y * blockDim.y + threadIdx.y;
+  int dw = sw >> 1, dh = sh >> 1;
+  if (x >= dw || y >= dh) return;
+  const uint8_t* r0 = src + (size_t)(2 * y) * spitch + 2 * x;
+  const uint8_t* r1 = r0 + spitch;
+  dst[(size_t)y * dpitch + x] =
+      (uint8_t)((r0[0] + r0[1] + r1[0] + r1[1] + 2) >> 2);
+}
+
+void launch_downsample2(const uint8_t* src, int spitch, int sw, int sh,
+                        uint8_t* dst, int dpitch, hipStream_t stream) {
+  dim3 b(16, 16);
+  dim3 g(((sw >> 1) + 15) / 16, ((sh >> 1) + 15) / 16);
+  hipLaunchKernelGGL(k_downsample2, g, b, 0, stream, src, spitch, sw, sh,
+                     dst, dpitch);
 }
 
 }  // namespace h264gpu

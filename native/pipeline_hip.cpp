@@ -476,8 +476,22 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipMemcpyAsync(d_jobs_, h_jobs_,
                              sizeof(h264gpu::RowJob) * n_jobs,
                              hipMemcpyHostToDevice, stream_));
-    h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, mbw_, n_jobs,
-                            d_jobs_, d_meta_, stream_);
+    // luma pyramid (quarter res) for ME acquisition
+    if (!ctx.idr) {
+      h264gpu::launch_downsample2(d_srcY_, ypitch_, ypitch_, mbh_ * 16,
+                                  d_mip1_, ypitch_ / 2, stream_);
+      h264gpu::launch_downsample2(d_mip1_, ypitch_ / 2, ypitch_ / 2,
+                                  mbh_ * 8, d_srcY2_, ypitch_ / 4,
+                                  stream_);
+      h264gpu::launch_downsample2(d_refY_, ypitch_, ypitch_, mbh_ * 16,
+                                  d_mip1_, ypitch_ / 2, stream_);
+      h264gpu::launch_downsample2(d_mip1_, ypitch_ / 2, ypitch_ / 2,
+                                  mbh_ * 8, d_refY2_, ypitch_ / 4,
+                                  stream_);
+    }
+    h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, d_srcY2_,
+                            d_refY2_, ypitch_ / 4, mbw_, n_jobs, d_jobs_,
+                            d_meta_, stream_);
 
     // Single batch: the row kernel's cost is per-row LATENCY (all rows run
     // concurrently), so splitting into sequential batches multiplies GPU
@@ -763,6 +777,9 @@ class HipH264Pipeline : public EncodePipeline {
     size_t meta_bytes =
         static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb * sizeof(int);
     d_levels_ = reinterpret_cast<int16_t*>(dalloc(level_bytes));
+    d_mip1_ = dalloc(static_cast<size_t>(ypitch_ / 2) * mbh_ * 8);
+    d_srcY2_ = dalloc(static_cast<size_t>(ypitch_ / 4) * mbh_ * 4);
+    d_refY2_ = dalloc(static_cast<size_t>(ypitch_ / 4) * mbh_ * 4);
     d_meta_ = reinterpret_cast<int*>(dalloc(meta_bytes));
     segs_ = (mbw_ + h264gpu::kMaxSegMbw - 1) / h264gpu::kMaxSegMbw;
     seg_w0_ = (mbw_ + segs_ - 1) / segs_;   // widest segment
@@ -809,6 +826,7 @@ class HipH264Pipeline : public EncodePipeline {
           *d_refCr_ = nullptr, *d_curY_ = nullptr, *d_curCb_ = nullptr,
           *d_curCr_ = nullptr;
   int16_t* d_levels_ = nullptr;
+  uint8_t *d_mip1_ = nullptr, *d_srcY2_ = nullptr, *d_refY2_ = nullptr;
   int* d_meta_ = nullptr;
   h264gpu::RowJob* d_jobs_ = nullptr;
   int16_t* h_levels_ = nullptr;
