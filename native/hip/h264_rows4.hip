@@ -209,14 +209,20 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
     }
     __syncthreads();
 
-    // ---- luma DC chain (redundant in both luma waves; identical result)
-    int dcreg = (lane < 16) ? sh->dc[lane] : 0;
-    int had = hadamard4_wave(dcreg, lane, true);
-    int qdc = quant_dc_v(had, qp);
-    qdc = cap12_group(qdc, zz, lane < 16, lane);
-    if (w == 0 && lane < 16) store_lvl_pair(L + kLumaDcOff, lane, qdc, lane);
-    int ih = hadamard4_wave(qdc, lane, false);
-    int dcrec = dequant_luma_dc_v(ih, qp);
+    // ---- luma DC chain: computed once on wave 0; the reconstructed DCs
+    // are broadcast back through sh->dc (the raw DCs it held are dead
+    // after this). Saves the redundant hadamard/quant/cap chain on the
+    // other three luma waves — they wait at the barrier instead.
+    if (w == 0) {
+      int dcreg = (lane < 16) ? sh->dc[lane] : 0;
+      int had = hadamard4_wave(dcreg, lane, true);
+      int qdc = quant_dc_v(had, qp);
+      qdc = cap12_group(qdc, zz, lane < 16, lane);
+      if (lane < 16) store_lvl_pair(L + kLumaDcOff, lane, qdc, lane);
+      int ih = hadamard4_wave(qdc, lane, false);
+      if (lane < 16) sh->dc[lane] = dequant_luma_dc_v(ih, qp);
+    }
+    __syncthreads();
 
     // recon (AC gate dropped: zero levels dequantize to zero)
 #pragma unroll
@@ -224,7 +230,7 @@ __device__ void luma_wave(const uint8_t* __restrict__ srcY, int ypitch,
       int pass = NP * w + pi;
       int blk = pass * 4 + g;
       int bx = blk & 3, by = blk >> 2;
-      int dcb = __shfl(dcrec, blk);
+      int dcb = sh->dc[blk];
       int d = (c == 0) ? dcb : dequant_c(lvl_p[pi], qp, coeff_cls(c));
       int rec = idct4_wave(d, lane);
       int py = by * 4 + (c >> 2), px = bx * 4 + (c & 3);
@@ -342,6 +348,7 @@ __device__ void chroma_wave(const uint8_t* __restrict__ srcC,
       int th = sh->ccost[0][0] + sh->ccost[1][0];
       int tdc = sh->ccost[0][1] + sh->ccost[1][1];
       chroma_mode = (have_left && th < tdc) ? 1 : 0;
+      __syncthreads();   // pairs with the luma DC-broadcast barrier
     }
 
     const int scx = (g & 1) * 4, scy = (g >> 1) * 4;
