@@ -53,11 +53,32 @@
       <button id="sk-share-view">copy viewer link</button>
       <button id="sk-share-rtc">copy WebRTC link</button>
     </div>
+    <h3>Touch gamepad</h3>
+    <div>
+      <select id="sk-touch-profile">
+        <option value="modern">modern (dual stick)</option>
+        <option value="classic">classic (d-pad)</option>
+      </select>
+      <button id="sk-touch-toggle">show / hide</button>
+    </div>
   `;
   document.body.appendChild(panel);
   tab.onclick = () => panel.classList.toggle("open");
 
   const send = (msg) => window.skSend && window.skSend(msg);
+
+  /* touch gamepad overlay toggle (reference universal-touch-gamepad) */
+  const touchSel = panel.querySelector("#sk-touch-profile");
+  touchSel.value = localStorage.getItem("selkies.touchProfile") || "modern";
+  touchSel.onchange = () => {
+    try {
+      localStorage.setItem("selkies.touchProfile", touchSel.value);
+    } catch (e) {}
+  };
+  panel.querySelector("#sk-touch-toggle").onclick = () => {
+    if (window.skToggleTouchGamepad)
+      window.skToggleTouchGamepad(touchSel.value);
+  };
 
   /* settings widgets built from the server's client-settings contract */
   const WIDGETS = ["encoder", "framerate", "video_crf",

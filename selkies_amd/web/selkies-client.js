@@ -439,8 +439,39 @@ async function connectWebRTC() {
   }, 5000);
 }
 
+/* ---- gamepads: physical polling + touch overlay (gamepad.js) ---- */
+let touchPad = null;
+function hookGamepads() {
+  const G = window.SelkiesGamepad;
+  if (!G) return;
+  const enc = new G.GamepadEncoder(send);
+  let polling = false;
+  function poll() {
+    G.pollPhysical(enc, navigator.getGamepads ? navigator.getGamepads() : []);
+    if (polling) requestAnimationFrame(poll);
+  }
+  window.addEventListener("gamepadconnected", () => {
+    if (!polling) { polling = true; poll(); }
+  });
+  window.addEventListener("gamepaddisconnected", () => {
+    /* keep polling; encoder emits js,d when the slot empties */
+  });
+  /* touch overlay: ?touch=1 forces it on; dashboard can toggle.
+     Touch pads claim index 3 so they never fight a physical pad. */
+  window.skToggleTouchGamepad = (profile) => {
+    if (touchPad) { touchPad.destroy(); touchPad = null; return false; }
+    touchPad = G.attachOverlay(document.body, send,
+                               profile || localStorage.getItem(
+                                   "selkies.touchProfile") || "modern", 3);
+    return true;
+  };
+  if (new URLSearchParams(location.search).get("touch") === "1")
+    window.skToggleTouchGamepad();
+}
+
 hookInput();
 hookHud();
+hookGamepads();
 const params = new URLSearchParams(location.search);
 if (params.get("transport") === "webrtc") {
   connectWebRTC().catch((e) => {
