@@ -202,6 +202,12 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("enable_gamepad", bool, True, "Gamepad passthrough (interposer/uinput)."),
     SettingDef("enable_shared", bool, False,
                "Allow multiple controlling clients (shared input)."),
+    SettingDef("enable_player2", bool, False,
+               "Allow a second client to claim gamepad seat 1.", client=True),
+    SettingDef("enable_player3", bool, False,
+               "Allow a third client to claim gamepad seat 2.", client=True),
+    SettingDef("enable_player4", bool, False,
+               "Allow a fourth client to claim gamepad seat 3.", client=True),
     SettingDef("enable_command_input", bool, False,
                "Allow the 'cmd' wire verb to run shell commands."),
 

@@ -35,6 +35,25 @@ from .settings import AppSettings
 logger = logging.getLogger("selkies.streaming")
 
 
+def _js_index(text: str) -> Optional[int]:
+    """Pad index of a js,* wire verb (None if malformed)."""
+    parts = text.split(",", 3)
+    if len(parts) < 3:
+        return None
+    try:
+        return int(parts[2])
+    except ValueError:
+        return None
+
+
+def _remap_js(text: str, seat: int) -> str:
+    """Rewrite the pad index of a js,* verb to the holder's seat slot."""
+    parts = text.split(",")
+    if len(parts) >= 3:
+        parts[2] = str(seat)
+    return ",".join(parts)
+
+
 class ClientState:
     def __init__(self, ws, relay: VideoRelay, display: str = "primary",
                  role: str = "viewer"):
@@ -42,6 +61,10 @@ class ClientState:
         self.relay = relay
         self.display = display
         self.role = role
+        # gamepad seat held by this client: None, or 1..3 (player2..4).
+        # The controller implicitly owns seat 0. (Reference: signaling
+        # player2-4 slots, signaling_server.py allowed_client_slots.)
+        self.player_seat: Optional[int] = None
         self.last_acked_frame = -1
         self.last_sent_frame = -1
         self.ack_rtt_ms = 50.0
@@ -366,6 +389,12 @@ class StreamingService:
         finally:
             self.clients.pop(ws, None)
             await relay.stop()
+            if state.player_seat is not None and self.gamepads is not None:
+                # free the seat and unplug its pad slot
+                try:
+                    await self.gamepads.handle(f"js,d,{state.player_seat}")
+                except Exception:
+                    pass
             # promote the oldest remaining client when the controller leaves
             if state.role == "controller" and self.clients:
                 nxt = next(iter(self.clients.values()))
@@ -400,9 +429,34 @@ class StreamingService:
         if verb == "REQUEST_IDR":
             self.request_idr()
             return None
+        if verb == "CLAIM_SEAT":
+            # a viewer claims gamepad seat n (1..3 = player2..4); granted
+            # when enable_player{n+1} is on and the seat is free
+            try:
+                n = int(rest)
+            except (TypeError, ValueError):
+                return P.encode_control("SEAT", "-1")
+            free = not any(c.player_seat == n
+                           for c in self.clients.values())
+            allowed = (1 <= n <= 3 and free and state.role != "controller"
+                       and getattr(self.settings,
+                                   f"enable_player{n + 1}", False))
+            if allowed:
+                state.player_seat = n
+            return P.encode_control("SEAT", str(n if allowed else -1))
+        if verb == "RELEASE_SEAT":
+            state.player_seat = None
+            return P.encode_control("SEAT", "-1")
         if verb == "js" and self.gamepads is not None:
-            if state.role == "controller" or self.settings.enable_shared:
-                await self.gamepads.handle(text)
+            if state.player_seat is not None:
+                # a seated player's pads all map onto their one seat slot
+                await self.gamepads.handle(
+                    _remap_js(text, state.player_seat))
+            elif state.role == "controller" or self.settings.enable_shared:
+                claimed = {c.player_seat for c in self.clients.values()
+                           if c.player_seat is not None}
+                if _js_index(text) not in claimed:
+                    await self.gamepads.handle(text)
             return None
         if state.role != "controller" and not self.settings.enable_shared:
             return None          # viewers cannot inject input

@@ -61,6 +61,7 @@
       </select>
       <button id="sk-touch-toggle">show / hide</button>
     </div>
+    <div id="sk-seats"></div>
   `;
   document.body.appendChild(panel);
   tab.onclick = () => panel.classList.toggle("open");
@@ -86,6 +87,20 @@
                    "use_paint_over_quality", "video_fullframe",
                    "capture_scale_div", "enable_audio"];
   window.skOnSettings = (payload) => {
+    /* player-seat claim buttons appear when the server enables them
+       (reference dashboard PlayerGamepadButton) */
+    const seatHost = document.getElementById("sk-seats");
+    if (seatHost) {
+      seatHost.innerHTML = "";
+      for (let p = 2; p <= 4; p++) {
+        const def = payload["enable_player" + p];
+        if (!def || !def.value) continue;
+        const b = document.createElement("button");
+        b.textContent = "claim player " + p + " gamepad";
+        b.onclick = () => send("CLAIM_SEAT," + (p - 1));
+        seatHost.appendChild(b);
+      }
+    }
     const host = document.getElementById("sk-settings");
     host.innerHTML = "";
     for (const name of WIDGETS) {

@@ -333,6 +333,9 @@ function connect() {
         try {
           if (window.skOnStats) window.skOnStats(JSON.parse(rest));
         } catch (e) {}
+      } else if (verb === "SEAT") {
+        /* player-seat grant/deny (reference PlayerGamepadButton) */
+        if (window.skOnSeat) window.skOnSeat(parseInt(rest, 10));
       } else if (verb === "CURSOR") {
         try { applyCursor(JSON.parse(rest)); } catch (e) {}
       } else if (verb === "clipboard") {
