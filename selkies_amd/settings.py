@@ -103,6 +103,15 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("turn_protocol", str, "udp", "TURN transport for clients.",
                allowed=("udp", "tcp")),
     SettingDef("turn_tls", bool, False, "Use turns: (TLS) TURN URLs."),
+    SettingDef("turn_username", str, "",
+               "Legacy long-term TURN username (when no shared secret)."),
+    SettingDef("turn_password", str, "",
+               "Legacy long-term TURN password (when no shared secret)."),
+    SettingDef("turn_rest_uri", str, "",
+               "TURN-REST service URI for fetched ICE configs."),
+    SettingDef("rtc_config_json", str, "",
+               "Path to a JSON RTC config file (highest-priority ICE "
+               "source; hot-reloaded on change)."),
     SettingDef("stun_host", str, "", "Extra STUN host to advertise first."),
     SettingDef("stun_port", int, 3478, "Port for stun_host.",
                value_range=(1, 65535)),

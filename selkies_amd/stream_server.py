@@ -14,6 +14,7 @@ from __future__ import annotations
 
 import asyncio
 import base64
+import json
 import logging
 import os
 import ssl
@@ -352,19 +353,13 @@ class CentralizedStreamServer:
                 w.writerow([f"{ts:.3f}"] + [r.get(k, "") for k in keys])
 
     async def handle_turn(self, request):
-        """RTC config with minted coturn HMAC credentials (reference
-        handle_turn_req, signaling_server.py:1030)."""
-        s = self.settings
-        if not (s.turn_shared_secret and s.turn_host):
-            return web.json_response({"error": "no TURN configured"},
-                                     status=404)
-        from .webrtc.turn import generate_rtc_config
-        cfg = generate_rtc_config(
-            s.turn_host, s.turn_port, s.turn_shared_secret,
-            user=request.query.get("user"), protocol=s.turn_protocol,
-            turn_tls=s.turn_tls, stun_host=s.stun_host or None,
-            stun_port=s.stun_port)
-        return web.json_response(cfg)
+        """RTC config via the resolution chain (reference
+        webrtc_utils.get_rtc_configuration: JSON file -> TURN-REST ->
+        coturn HMAC -> legacy static -> STUN-only)."""
+        from .webrtc.turn import resolve_rtc_config
+        cfg, source = await resolve_rtc_config(
+            self.settings, request.query.get("user"))
+        return web.json_response(cfg, headers={"X-RTC-Source": source})
 
     # ---- lifecycle ----------------------------------------------------------
     async def start(self):
@@ -383,9 +378,27 @@ class CentralizedStreamServer:
             asyncio.get_running_loop().create_task(self._watch_certs())
         self._uplink_task = asyncio.get_running_loop().create_task(
             self._uplink_loop())
+        if s.rtc_config_json:
+            # hot-reload the file-sourced ICE config and push it to
+            # connected clients (reference RTCConfigFileMonitor)
+            from .webrtc.turn import RTCConfigFileMonitor
+            self._rtc_monitor = RTCConfigFileMonitor(
+                s.rtc_config_json, self._on_rtc_config_change)
+            self._rtc_monitor.start()
         logger.info("serving on %s:%s (mode=%s)", s.addr, s.port, s.mode)
 
+    async def _on_rtc_config_change(self, cfg):
+        payload = "RTC_CONFIG," + json.dumps(cfg)
+        for state in list(self.streaming.clients.values()):
+            try:
+                await state.ws.send_str(payload)
+            except Exception:
+                pass
+
     async def stop(self):
+        if getattr(self, "_rtc_monitor", None) is not None:
+            self._rtc_monitor.stop()
+            self._rtc_monitor = None
         if self._uplink_task is not None:
             self._uplink_task.cancel()
         self.streaming.stop_capture()

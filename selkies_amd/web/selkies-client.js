@@ -333,6 +333,10 @@ function connect() {
         try {
           if (window.skOnStats) window.skOnStats(JSON.parse(rest));
         } catch (e) {}
+      } else if (verb === "RTC_CONFIG") {
+        /* hot-reloaded ICE config (server file monitor); used by the
+           next RTCPeerConnection */
+        try { window.skRtcConfig = JSON.parse(rest); } catch (e) {}
       } else if (verb === "SEAT") {
         /* player-seat grant/deny (reference PlayerGamepadButton) */
         if (window.skOnSeat) window.skOnSeat(parseInt(rest, 10));

@@ -195,8 +195,9 @@ def test_audio_broadcast(loop):
 
 
 def test_turn_endpoint_and_credentials():
-    """/api/turn mints coturn use-auth-secret credentials; 404 without
-    configuration."""
+    """/api/turn mints coturn use-auth-secret credentials; falls back
+    to a STUN-only config when no TURN is configured (resolution
+    chain, reference webrtc_utils.get_rtc_configuration)."""
     import asyncio
     import base64
     import hashlib
@@ -210,7 +211,10 @@ def test_turn_endpoint_and_credentials():
         try:
             async with aiohttp.ClientSession() as sess:
                 r = await sess.get(f"http://127.0.0.1:{port}/api/turn")
-                assert r.status == 404
+                assert r.status == 200
+                assert r.headers["X-RTC-Source"] == "stun"
+                cfg = await r.json()
+                assert cfg["iceServers"][0]["urls"][0].startswith("stun:")
         finally:
             await srv.stop()
             await runner.cleanup()
