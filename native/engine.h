@@ -46,6 +46,16 @@ class EncodePipeline {
                             const Emit& emit) = 0;
   virtual const char* name() const = 0;
 
+  // Throughput mode (reference has no equivalent; x264-style frame
+  // pipelining for the MI355X pipeline): with depth 2 an implementation
+  // may defer emission by one frame — encode_frame(N) submits N's GPU
+  // work and emits frame N-1's stripes, overlapping host bitstream
+  // assembly with GPU compute of the next frame. flush() emits any
+  // deferred frame. Depth 1 (default) is strictly synchronous; the
+  // streaming engine uses depth 1 so interactive latency is unchanged.
+  virtual void set_pipeline_depth(int) {}
+  virtual void flush(const Emit&) {}
+
   // Test-only introspection: reconstruction planes + (GPU) level/meta
   // buffers of the last encoded frame. Returns false when unsupported.
   struct DebugDump {

@@ -266,8 +266,8 @@ PYBIND11_MODULE(_native, m) {
   m.def(
       "_pipeline_encode",
       [](const std::string& kind, py::list frames, int w, int h, int qp,
-         int stripe_h, int output_mode, bool dump,
-         bool fullcolor) -> py::object {
+         int stripe_h, int output_mode, bool dump, bool fullcolor,
+         int pipeline_depth) -> py::object {
         CaptureSettings s;
         s.capture_width = w;
         s.capture_height = h;
@@ -286,8 +286,17 @@ PYBIND11_MODULE(_native, m) {
           p = output_mode == 1 ? make_cpu_h264_pipeline(s)
                                : make_cpu_jpeg_pipeline(s);
         }
+        p->set_pipeline_depth(pipeline_depth);
+        // emissions land in result[st.frame_id] so depth-2 pipelining
+        // (stripes of frame N emitted during the call for N+1) yields the
+        // same per-frame grouping as the synchronous path
+        std::vector<py::list> per_frame(frames.size());
+        auto sink = [&](EncodedStripe& st) {
+          per_frame[st.frame_id].append(py::make_tuple(
+              py::bytes(reinterpret_cast<const char*>(st.data), st.size),
+              st.y, st.height, st.is_keyframe));
+        };
         py::list result;
-        std::vector<std::vector<uint8_t>> copies;
         for (size_t i = 0; i < frames.size(); ++i) {
           py::buffer buf = frames[i].cast<py::buffer>();
           py::buffer_info info = buf.request();
@@ -311,16 +320,10 @@ PYBIND11_MODULE(_native, m) {
             j.encode = true;
             ctx.stripes.push_back(j);
           }
-          py::list frame_out;
-          {
-            p->encode_frame(f, ctx, [&](EncodedStripe& st) {
-              frame_out.append(py::make_tuple(
-                  py::bytes(reinterpret_cast<const char*>(st.data), st.size),
-                  st.y, st.height, st.is_keyframe));
-            });
-          }
-          result.append(frame_out);
+          p->encode_frame(f, ctx, sink);
         }
+        p->flush(sink);
+        for (auto& fo : per_frame) result.append(fo);
         if (dump) {
           EncodePipeline::DebugDump d;
           if (!p->debug_dump(d))
@@ -348,7 +351,7 @@ PYBIND11_MODULE(_native, m) {
       py::arg("kind"), py::arg("frames"), py::arg("w"), py::arg("h"),
       py::arg("qp") = 26, py::arg("stripe_h") = 64,
       py::arg("output_mode") = 1, py::arg("dump") = false,
-      py::arg("fullcolor") = false,
+      py::arg("fullcolor") = false, py::arg("pipeline_depth") = 1,
       "Test hook: run frames through a named encode pipeline.");
 
   // ---- persistent pipeline handle for benchmarking ------------------------
@@ -356,8 +359,10 @@ PYBIND11_MODULE(_native, m) {
     std::unique_ptr<EncodePipeline> p;
     int w, h, qp, stripe_h;
     uint32_t frame_id = 0;
+    long last_frame_id = -1;   // frame whose stripes the last call emitted
     BenchPipeline(const std::string& kind, int width, int height, int qp_,
-                  int stripe, int output_mode, int gpu_id)
+                  int stripe, int output_mode, int gpu_id,
+                  int pipeline_depth)
         : w(width), h(height), qp(qp_), stripe_h(stripe) {
       CaptureSettings s;
       s.capture_width = w;
@@ -375,15 +380,38 @@ PYBIND11_MODULE(_native, m) {
         p = output_mode == 1 ? make_cpu_h264_pipeline(s)
                              : make_cpu_jpeg_pipeline(s);
       }
+      p->set_pipeline_depth(pipeline_depth);
     }
   };
   py::class_<BenchPipeline>(m, "BenchPipeline")
-      .def(py::init<const std::string&, int, int, int, int, int, int>(),
+      .def(py::init<const std::string&, int, int, int, int, int, int,
+                    int>(),
            py::arg("kind"), py::arg("width"), py::arg("height"),
            py::arg("qp") = 28, py::arg("stripe_height") = 64,
-           py::arg("output_mode") = 1, py::arg("gpu_id") = 0)
+           py::arg("output_mode") = 1, py::arg("gpu_id") = 0,
+           py::arg("pipeline_depth") = 1)
       .def_property_readonly("pipeline",
                              [](BenchPipeline& b) { return b.p->name(); })
+      .def_readonly("last_frame_id", &BenchPipeline::last_frame_id)
+      .def(
+          "flush",
+          [](BenchPipeline& b) {
+            size_t total = 0;
+            int stripes = 0;
+            long done = -1;
+            {
+              py::gil_scoped_release rel;
+              b.p->flush([&](EncodedStripe& st) {
+                total += st.size;
+                ++stripes;
+                done = st.frame_id;
+              });
+            }
+            b.last_frame_id = done;
+            return py::make_tuple(total, stripes);
+          },
+          "Emit any pipelined frame still in flight (depth 2); returns "
+          "(bitstream_bytes, stripes).")
       .def(
           "encode",
           [](BenchPipeline& b, py::buffer bgrx, bool idr) {
@@ -412,10 +440,13 @@ PYBIND11_MODULE(_native, m) {
                 j.encode = true;
                 ctx.stripes.push_back(j);
               }
+              long done = -1;
               b.p->encode_frame(f, ctx, [&](EncodedStripe& st) {
                 total += st.size;
                 ++stripes;
+                done = st.frame_id;
               });
+              b.last_frame_id = done;
             }
             return py::make_tuple(total, stripes);
           },

@@ -371,27 +371,70 @@ class HipH264Pipeline : public EncodePipeline {
     batch_events_.resize(4);
     for (auto& e : batch_events_)
       HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    for (int i = 0; i < 2; ++i) {
+      HIP_CHECK(hipEventCreateWithFlags(&ev_done_[i],
+                                        hipEventDisableTiming));
+      HIP_CHECK(hipEventCreateWithFlags(&ev_h2d_[i],
+                                        hipEventDisableTiming));
+    }
     alloc_for(s.capture_width, s.capture_height);
   }
 
   ~HipH264Pipeline() override {
     (void)hipStreamSynchronize(stream_);
     for (auto& e : batch_events_) (void)hipEventDestroy(e);
+    for (int i = 0; i < 2; ++i) {
+      (void)hipEventDestroy(ev_done_[i]);
+      (void)hipEventDestroy(ev_h2d_[i]);
+    }
     for (auto& kv : registered_) (void)hipHostUnregister(kv.first);
     for (void* p : device_ptrs_)
       if (p) (void)hipFree(p);
     if (h_levels_) (void)hipHostFree(h_levels_);
     if (h_meta_) (void)hipHostFree(h_meta_);
-    if (h_jobs_) (void)hipHostFree(h_jobs_);
     if (h_stage_) (void)hipHostFree(h_stage_);
-    if (h_entout_) (void)hipHostFree(h_entout_);
-    if (h_outbits_) (void)hipHostFree(h_outbits_);
+    for (int i = 0; i < 2; ++i) {
+      if (h_jobs_[i]) (void)hipHostFree(h_jobs_[i]);
+      if (h_entout_[i]) (void)hipHostFree(h_entout_[i]);
+      if (h_outbits_[i]) (void)hipHostFree(h_outbits_[i]);
+    }
   }
 
-  void encode_frame(const RawFrame& frame, const FrameContext& ctx,
-                    const Emit& emit) override {
+  // ---- frame pipeline -----------------------------------------------------
+  // submit_frame enqueues one frame's full GPU sequence (upload -> CSC ->
+  // pyramid -> ME -> rows -> CAVLC -> compacted D2H -> ref refresh) with
+  // NO sync; collect_pending waits on the frame's completion event and
+  // does the host-side bitstream assembly + emission. Depth 1 runs them
+  // back to back (today's latency behavior, bit-identical); depth 2 keeps
+  // one frame in flight so host assembly of frame N overlaps the GPU
+  // compute of frame N+1 (x264-style frame pipelining; throughput mode).
+  struct Out {
+    std::vector<uint8_t> header;            // SPS/PPS on IDR
+    std::vector<std::vector<uint8_t>> rows;
+    std::vector<uint8_t> bytes;
+    h264::GpuStripeParams p{};
+    int y0 = 0, h = 0;
+    bool idr = false;
+  };
+  struct Batch {
+    int job0, jobn;        // job index range
+    int row0, rown;        // absolute MB row range [row0, rown)
+  };
+  struct Pending {
+    bool active = false;
+    int par = 0;
+    int n_jobs = 0;
+    int copy_words = 0;    // the D2H cap this frame was submitted with
+    uint32_t frame_id = 0;
+    std::vector<Out> outs;
+    std::vector<std::pair<int, int>> job_map;
+    std::vector<Batch> batches;
+  };
+
+  Pending submit_frame(const RawFrame& frame, const FrameContext& ctx) {
     if (frame.width != w_ || frame.height != h_)
       alloc_for(frame.width, frame.height);
+    const int par = parity_;
     const int qp = std::min(51, std::max(0, ctx.crf));
 
     // upload + CSC. Prefer zero-copy DMA from the (registered) capture
@@ -419,6 +462,8 @@ class HipH264Pipeline : public EncodePipeline {
     }
     HIP_CHECK(hipMemcpyAsync(d_frame_, src, frame_bytes,
                              hipMemcpyHostToDevice, stream_));
+    // the caller may reuse its frame buffer once this event completes
+    HIP_CHECK(hipEventRecord(ev_h2d_[par], stream_));
     launch_bgrx_to_planes(d_frame_, w_, h_, frame.stride / 4, d_srcY_,
                           d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
                           stream_);
@@ -456,24 +501,25 @@ class HipH264Pipeline : public EncodePipeline {
         int x = 0;
         for (int sg = 0; sg < segs_; ++sg) {
           int segw = std::min(seg_w0_, mbw_ - x);
-          h_jobs_[n_jobs].mb_row = row0 + r;
-          h_jobs_[n_jobs].qp = qp;
-          h_jobs_[n_jobs].flags = sj.idr ? 1 : 0;
-          h_jobs_[n_jobs].stripe_y0 = sy0;
-          h_jobs_[n_jobs].stripe_y1 = sy1;
-          h_jobs_[n_jobs].first_mb = r * mbw_ + x;
-          h_jobs_[n_jobs].frame_num = static_cast<int>(sj.frame_num);
-          h_jobs_[n_jobs].idr_pic_id = static_cast<int>(sj.idr_pic_id);
-          h_jobs_[n_jobs].mbx0 = x;
-          h_jobs_[n_jobs].seg_mbw = segw;
+          h_jobs_[par][n_jobs].mb_row = row0 + r;
+          h_jobs_[par][n_jobs].qp = qp;
+          h_jobs_[par][n_jobs].flags = sj.idr ? 1 : 0;
+          h_jobs_[par][n_jobs].stripe_y0 = sy0;
+          h_jobs_[par][n_jobs].stripe_y1 = sy1;
+          h_jobs_[par][n_jobs].first_mb = r * mbw_ + x;
+          h_jobs_[par][n_jobs].frame_num = static_cast<int>(sj.frame_num);
+          h_jobs_[par][n_jobs].idr_pic_id = static_cast<int>(sj.idr_pic_id);
+          h_jobs_[par][n_jobs].mbx0 = x;
+          h_jobs_[par][n_jobs].seg_mbw = segw;
           x += segw;
           ++n_jobs;
         }
       }
     }
-    if (n_jobs == 0) return;
+    if (n_jobs == 0) return Pending{};
+    parity_ ^= 1;
 
-    HIP_CHECK(hipMemcpyAsync(d_jobs_, h_jobs_,
+    HIP_CHECK(hipMemcpyAsync(d_jobs_[par], h_jobs_[par],
                              sizeof(h264gpu::RowJob) * n_jobs,
                              hipMemcpyHostToDevice, stream_));
     // luma pyramid (quarter res) for ME acquisition
@@ -490,24 +536,20 @@ class HipH264Pipeline : public EncodePipeline {
                                   stream_);
     }
     h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, d_srcY2_,
-                            d_refY2_, ypitch_ / 4, mbw_, n_jobs, d_jobs_,
-                            d_meta_, stream_);
+                            d_refY2_, ypitch_ / 4, mbw_, n_jobs,
+                            d_jobs_[par], d_meta_, stream_);
 
     // Single batch: the row kernel's cost is per-row LATENCY (all rows run
     // concurrently), so splitting into sequential batches multiplies GPU
     // time (measured: 4 batches regressed 278->155 fps). Entropy overlap
-    // would need per-row completion signaling; revisit with stream-per-batch.
+    // is cross-FRAME instead (depth-2 pipelining).
     const int kBatches = 1;
-    struct Batch {
-      int job0, jobn;        // job index range
-      int row0, rown;        // absolute MB row range [row0, rown)
-    };
     std::vector<Batch> batches;
     for (int b = 0; b < kBatches; ++b) {
       int j0 = n_jobs * b / kBatches, j1 = n_jobs * (b + 1) / kBatches;
       if (j0 == j1) continue;
-      batches.push_back({j0, j1, h_jobs_[j0].mb_row,
-                         h_jobs_[j1 - 1].mb_row + 1});
+      batches.push_back({j0, j1, h_jobs_[par][j0].mb_row,
+                         h_jobs_[par][j1 - 1].mb_row + 1});
     }
     const size_t lvl_row =
         static_cast<size_t>(mbw_) * h264gpu::kLevelsPerMb * sizeof(int16_t);
@@ -521,13 +563,13 @@ class HipH264Pipeline : public EncodePipeline {
         h264gpu::launch_h264_rows(
             d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
             d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
-            bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_,
+            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_, d_meta_,
             stream_);
       else
         h264gpu::launch_h264_rows4(
             d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
             d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
-            bt.jobn - bt.job0, d_jobs_ + bt.job0, d_levels_, d_meta_,
+            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_, d_meta_,
             stream_);
       if (cpu_entropy_) {
         HIP_CHECK(hipMemcpyAsync(
@@ -541,24 +583,26 @@ class HipH264Pipeline : public EncodePipeline {
       }
       HIP_CHECK(hipEventRecord(batch_events_[b], stream_));
     }
+    const int copy_words = ent_copy_words_;
     if (!cpu_entropy_) {
-      h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs, d_jobs_,
-                                 d_stage_, d_nbits_, d_entout_,
-                                 ent_stride_words_, d_outbits_, stream_);
+      h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs,
+                                 d_jobs_[par], d_stage_, d_nbits_,
+                                 d_entout_[par], ent_stride_words_,
+                                 d_outbits_[par], stream_);
       // compaction: copy only ~the used prefix of each row's bitstream
       // (adaptive cap = 2x last frame's max row, full stride first frame;
-      // rows that overflow the cap are re-copied exactly below)
-      if (ent_copy_words_ >= ent_stride_words_) {
-        HIP_CHECK(hipMemcpyAsync(h_entout_, d_entout_,
+      // rows that overflow the cap are re-copied exactly in collect)
+      if (copy_words >= ent_stride_words_) {
+        HIP_CHECK(hipMemcpyAsync(h_entout_[par], d_entout_[par],
                                  (size_t)n_jobs * ent_stride_words_ * 4,
                                  hipMemcpyDeviceToHost, stream_));
       } else {
         HIP_CHECK(hipMemcpy2DAsync(
-            h_entout_, (size_t)ent_stride_words_ * 4, d_entout_,
-            (size_t)ent_stride_words_ * 4, (size_t)ent_copy_words_ * 4,
+            h_entout_[par], (size_t)ent_stride_words_ * 4, d_entout_[par],
+            (size_t)ent_stride_words_ * 4, (size_t)copy_words * 4,
             n_jobs, hipMemcpyDeviceToHost, stream_));
       }
-      HIP_CHECK(hipMemcpyAsync(h_outbits_, d_outbits_,
+      HIP_CHECK(hipMemcpyAsync(h_outbits_[par], d_outbits_[par],
                                sizeof(int) * n_jobs, hipMemcpyDeviceToHost,
                                stream_));
     }
@@ -585,62 +629,64 @@ class HipH264Pipeline : public EncodePipeline {
       }
       flush_span();
     }
-    // NOTE: no stream sync here — entropy overlaps the remaining batches;
-    // the final sync below covers the ref-refresh copies.
+    HIP_CHECK(hipEventRecord(ev_done_[par], stream_));
 
-    // entropy: GPU CAVLC by default (k_h264_cavlc_rows packs every row
-    // slice on-device; CPU only adds stop bit + emulation prevention +
-    // NAL headers). HIPFLUX_CPU_ENTROPY=1 selects the CPU packer (kept
-    // for tests and as reference).
-    struct Out {
-      std::vector<uint8_t> header;            // SPS/PPS on IDR
-      std::vector<std::vector<uint8_t>> rows;
-      std::vector<uint8_t> bytes;
-      h264::GpuStripeParams p{};
-      int y0 = 0, h = 0;
-      bool idr = false;
-    };
-    std::vector<Out> outs(sjobs.size());
-    // job index -> (stripe index in outs, row within stripe)
-    std::vector<std::pair<int, int>> job_map(n_jobs);
+    // host-side output prep (headers, job -> stripe/row mapping)
+    Pending pd;
+    pd.active = true;
+    pd.par = par;
+    pd.n_jobs = n_jobs;
+    pd.copy_words = copy_words;
+    pd.frame_id = ctx.frame_id;
+    pd.batches = std::move(batches);
+    pd.outs.resize(sjobs.size());
+    pd.job_map.resize(n_jobs);
     {
       int j = 0;
       for (size_t i = 0; i < sjobs.size(); ++i) {
         const auto& sj = sjobs[i];
-        outs[i].y0 = sj.y0;
-        outs[i].h = std::min(sj.y1, h_) - sj.y0;
-        outs[i].idr = sj.idr;
-        h264::GpuStripeParams& p = outs[i].p;
+        Out& o = pd.outs[i];
+        o.y0 = sj.y0;
+        o.h = std::min(sj.y1, h_) - sj.y0;
+        o.idr = sj.idr;
+        h264::GpuStripeParams& p = o.p;
         p.levels = h_levels_;
         p.meta = h_meta_;
         p.mbw = mbw_;
         p.mb_row0 = sj.y0 / 16;
         p.n_mb_rows = (std::min(sj.y1, mbh_ * 16) - sj.y0 + 15) / 16;
         p.width = w_;
-        p.height = outs[i].h;
+        p.height = o.h;
         p.qp = qp;
         p.idr = sj.idr;
         p.frame_num = sj.frame_num;
         p.idr_pic_id = sj.idr_pic_id;
-        outs[i].rows.resize(p.n_mb_rows * segs_);
+        o.rows.resize(p.n_mb_rows * segs_);
         if (p.idr) {
-          h264::write_sps_nal(outs[i].header, (p.width + 15) / 16,
+          h264::write_sps_nal(o.header, (p.width + 15) / 16,
                               p.n_mb_rows, p.width, p.height);
-          h264::write_pps_nal(outs[i].header);
+          h264::write_pps_nal(o.header);
         }
         for (int r = 0; r < p.n_mb_rows * segs_; ++r)
-          job_map[j++] = {int(i), r};
+          pd.job_map[j++] = {int(i), r};
       }
     }
+    return pd;
+  }
+
+  void collect_pending(Pending& pd, const Emit& emit) {
+    if (!pd.active) return;
+    pd.active = false;
+    const int par = pd.par;
     if (cpu_entropy_) {
-      for (size_t b = 0; b < batches.size(); ++b) {
+      for (size_t b = 0; b < pd.batches.size(); ++b) {
         HIP_CHECK(hipEventSynchronize(batch_events_[b]));
-        for (int j = batches[b].job0; j < batches[b].jobn; ++j) {
-          auto [si, sl] = job_map[j];
-          auto* dst = &outs[si].rows[sl];
-          const h264::GpuStripeParams* pp = &outs[si].p;
+        for (int j = pd.batches[b].job0; j < pd.batches[b].jobn; ++j) {
+          auto [si, sl] = pd.job_map[j];
+          auto* dst = &pd.outs[si].rows[sl];
+          const h264::GpuStripeParams* pp = &pd.outs[si].p;
           int r = sl / segs_;
-          int mbx0 = h_jobs_[j].mbx0, segw = h_jobs_[j].seg_mbw;
+          int mbx0 = h_jobs_[par][j].mbx0, segw = h_jobs_[par][j].seg_mbw;
           bool long_sc = sl == 0;
           pool_.submit([pp, r, mbx0, segw, long_sc, dst] {
             h264::encode_seg_nal_from_gpu(*pp, r, mbx0, segw, long_sc,
@@ -651,25 +697,29 @@ class HipH264Pipeline : public EncodePipeline {
       pool_.wait_all();
       HIP_CHECK(hipStreamSynchronize(stream_));
     } else {
-      HIP_CHECK(hipStreamSynchronize(stream_));
+      HIP_CHECK(hipEventSynchronize(ev_done_[par]));
       int max_words = 0;
-      for (int j = 0; j < n_jobs; ++j) {
-        int wds = (h_outbits_[j] + 31) / 32 + 1;
+      for (int j = 0; j < pd.n_jobs; ++j) {
+        int wds = (h_outbits_[par][j] + 31) / 32 + 1;
         max_words = std::max(max_words, wds);
-        if (wds > ent_copy_words_) {
+        if (wds > pd.copy_words) {
           // rare: this row outgrew the adaptive cap; fetch it exactly
-          HIP_CHECK(hipMemcpy(h_entout_ + (size_t)j * ent_stride_words_,
-                              d_entout_ + (size_t)j * ent_stride_words_,
-                              (size_t)wds * 4, hipMemcpyDeviceToHost));
+          // (d_entout_[par] is parity-preserved until this frame's
+          // collect, even with the next frame already submitted)
+          HIP_CHECK(hipMemcpy(
+              h_entout_[par] + (size_t)j * ent_stride_words_,
+              d_entout_[par] + (size_t)j * ent_stride_words_,
+              (size_t)wds * 4, hipMemcpyDeviceToHost));
         }
       }
       ent_copy_words_ = std::min(ent_stride_words_, max_words * 2 + 64);
-      for (int j = 0; j < n_jobs; ++j) {
-        auto [si, sl] = job_map[j];
-        const uint32_t* words = h_entout_ + (size_t)j * ent_stride_words_;
-        int bits = h_outbits_[j];
-        bool idr = outs[si].idr;
-        auto* dst = &outs[si].rows[sl];
+      for (int j = 0; j < pd.n_jobs; ++j) {
+        auto [si, sl] = pd.job_map[j];
+        const uint32_t* words =
+            h_entout_[par] + (size_t)j * ent_stride_words_;
+        int bits = h_outbits_[par][j];
+        bool idr = pd.outs[si].idr;
+        auto* dst = &pd.outs[si].rows[sl];
         bool long_sc = sl == 0;
         pool_.submit([words, bits, idr, long_sc, dst] {
           h264::assemble_gpu_row_nal(words, bits, idr, long_sc, *dst);
@@ -677,7 +727,7 @@ class HipH264Pipeline : public EncodePipeline {
       }
       pool_.wait_all();
     }
-    for (auto& o : outs) {
+    for (auto& o : pd.outs) {
       o.bytes = std::move(o.header);
       for (auto& r : o.rows)
         o.bytes.insert(o.bytes.end(), r.begin(), r.end());
@@ -686,13 +736,39 @@ class HipH264Pipeline : public EncodePipeline {
       s.type = StripeType::kH264;
       s.data = o.bytes.data();
       s.size = o.bytes.size();
-      s.frame_id = ctx.frame_id;
+      s.frame_id = pd.frame_id;
       s.y = o.y0;
       s.width = w_;
       s.height = o.h;
       s.is_keyframe = o.idr;
       emit(s);
     }
+  }
+
+  void encode_frame(const RawFrame& frame, const FrameContext& ctx,
+                    const Emit& emit) override {
+    Pending cur = submit_frame(frame, ctx);
+    if (!cur.active) {
+      // nothing scheduled this frame; keep the pipe draining
+      collect_pending(prev_, emit);
+      return;
+    }
+    if (depth_ < 2 || cpu_entropy_) {
+      collect_pending(cur, emit);
+      return;
+    }
+    // depth 2: emit LAST frame's stripes (host assembly overlaps this
+    // frame's GPU work), keep this one in flight
+    collect_pending(prev_, emit);
+    prev_ = std::move(cur);
+    // the caller may reuse its frame buffer after we return
+    HIP_CHECK(hipEventSynchronize(ev_h2d_[prev_.par]));
+  }
+
+  void flush(const Emit& emit) override { collect_pending(prev_, emit); }
+
+  void set_pipeline_depth(int d) override {
+    depth_ = std::max(1, std::min(2, d));
   }
 
   const char* name() const override { return "hip-h264"; }
@@ -741,13 +817,15 @@ class HipH264Pipeline : public EncodePipeline {
 
  private:
   void alloc_for(int w, int h) {
+    // resolution change: any pipelined frame still in flight references
+    // the old buffers — drop it (streams restart with a fresh IDR anyway)
+    prev_.active = false;
     HIP_CHECK(hipStreamSynchronize(stream_));
     for (void* p : device_ptrs_)
       if (p) (void)hipFree(p);
     device_ptrs_.clear();
     if (h_levels_) (void)hipHostFree(h_levels_);
     if (h_meta_) (void)hipHostFree(h_meta_);
-    if (h_jobs_) (void)hipHostFree(h_jobs_);
     w_ = w;
     h_ = h;
     mbw_ = (w + 15) / 16;
@@ -784,14 +862,9 @@ class HipH264Pipeline : public EncodePipeline {
     segs_ = (mbw_ + h264gpu::kMaxSegMbw - 1) / h264gpu::kMaxSegMbw;
     seg_w0_ = (mbw_ + segs_ - 1) / segs_;   // widest segment
     const int max_jobs = mbh_ * segs_;
-    d_jobs_ = reinterpret_cast<h264gpu::RowJob*>(
-        dalloc(sizeof(h264gpu::RowJob) * max_jobs));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_levels_), level_bytes,
                             hipHostMallocDefault));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_meta_), meta_bytes,
-                            hipHostMallocDefault));
-    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_jobs_),
-                            sizeof(h264gpu::RowJob) * max_jobs,
                             hipHostMallocDefault));
     // GPU entropy buffers (one slot per SLICE = row segment)
     cpu_entropy_ = std::getenv("HIPFLUX_CPU_ENTROPY") != nullptr;
@@ -802,16 +875,24 @@ class HipH264Pipeline : public EncodePipeline {
         (size_t)max_jobs * nitems * h264gpu::kStageWordsPerItem * 4;
     d_stage_ = reinterpret_cast<uint32_t*>(dalloc(stage_bytes));
     d_nbits_ = reinterpret_cast<int*>(dalloc((size_t)max_jobs * nitems * 4));
-    d_entout_ = reinterpret_cast<uint32_t*>(
-        dalloc((size_t)max_jobs * ent_stride_words_ * 4));
-    d_outbits_ = reinterpret_cast<int*>(dalloc(sizeof(int) * max_jobs));
-    if (h_entout_) (void)hipHostFree(h_entout_);
-    if (h_outbits_) (void)hipHostFree(h_outbits_);
-    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_entout_),
-                            (size_t)max_jobs * ent_stride_words_ * 4,
-                            hipHostMallocDefault));
-    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_outbits_),
-                            sizeof(int) * max_jobs, hipHostMallocDefault));
+    for (int i = 0; i < 2; ++i) {
+      d_jobs_[i] = reinterpret_cast<h264gpu::RowJob*>(
+          dalloc(sizeof(h264gpu::RowJob) * max_jobs));
+      d_entout_[i] = reinterpret_cast<uint32_t*>(
+          dalloc((size_t)max_jobs * ent_stride_words_ * 4));
+      d_outbits_[i] = reinterpret_cast<int*>(dalloc(sizeof(int) * max_jobs));
+      if (h_jobs_[i]) (void)hipHostFree(h_jobs_[i]);
+      if (h_entout_[i]) (void)hipHostFree(h_entout_[i]);
+      if (h_outbits_[i]) (void)hipHostFree(h_outbits_[i]);
+      HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_jobs_[i]),
+                              sizeof(h264gpu::RowJob) * max_jobs,
+                              hipHostMallocDefault));
+      HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_entout_[i]),
+                              (size_t)max_jobs * ent_stride_words_ * 4,
+                              hipHostMallocDefault));
+      HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_outbits_[i]),
+                              sizeof(int) * max_jobs, hipHostMallocDefault));
+    }
     stripes_.assign((h + stripe_h_ - 1) / stripe_h_, StripeState{});
   }
 
@@ -828,20 +909,28 @@ class HipH264Pipeline : public EncodePipeline {
   int16_t* d_levels_ = nullptr;
   uint8_t *d_mip1_ = nullptr, *d_srcY2_ = nullptr, *d_refY2_ = nullptr;
   int* d_meta_ = nullptr;
-  h264gpu::RowJob* d_jobs_ = nullptr;
   int16_t* h_levels_ = nullptr;
   int* h_meta_ = nullptr;
-  h264gpu::RowJob* h_jobs_ = nullptr;
   uint8_t* h_stage_ = nullptr;
   size_t h_stage_bytes_ = 0;
-  // GPU entropy buffers
+  // GPU entropy buffers. Buffers the host writes before (h_jobs_/d_jobs_)
+  // or reads after (h_entout_/h_outbits_, d_entout_ for overflow re-copy)
+  // a frame's GPU work are parity-double-buffered so depth-2 pipelining
+  // can submit frame N+1 while frame N's results are still being read;
+  // everything device-only is hazard-free by single-stream ordering.
   bool cpu_entropy_ = false;
+  h264gpu::RowJob* d_jobs_[2] = {nullptr, nullptr};
+  h264gpu::RowJob* h_jobs_[2] = {nullptr, nullptr};
   uint32_t* d_stage_ = nullptr;
   int* d_nbits_ = nullptr;
-  uint32_t* d_entout_ = nullptr;
-  int* d_outbits_ = nullptr;
-  uint32_t* h_entout_ = nullptr;
-  int* h_outbits_ = nullptr;
+  uint32_t* d_entout_[2] = {nullptr, nullptr};
+  int* d_outbits_[2] = {nullptr, nullptr};
+  uint32_t* h_entout_[2] = {nullptr, nullptr};
+  int* h_outbits_[2] = {nullptr, nullptr};
+  hipEvent_t ev_done_[2] = {}, ev_h2d_[2] = {};
+  int depth_ = 1;                  // 1 = sync (latency mode), 2 = pipelined
+  int parity_ = 0;
+  Pending prev_;                   // the in-flight frame (depth 2)
   int ent_stride_words_ = 0;
   int ent_copy_words_ = 1 << 30;   // adaptive D2H cap (words per row)
   bool rows_v1_ = std::getenv("HIPFLUX_ROWS_V1") != nullptr;
