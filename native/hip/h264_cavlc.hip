@@ -314,7 +314,7 @@ __device__ inline int lds_chroma_nc(const MbInfo* info, int mbx, int comp,
 }
 
 // ---- the kernel -------------------------------------------------------------
-__global__ void __launch_bounds__(512) k_h264_cavlc_rows(
+__global__ void __launch_bounds__(1024) k_h264_cavlc_rows(
     const int16_t* __restrict__ levels, const int* __restrict__ meta,
     int mbw, const RowJob* __restrict__ jobs,
     uint32_t* __restrict__ stage,       // [row][item][kStageWordsPerItem]
@@ -323,7 +323,9 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
     int out_stride_words,
     int* __restrict__ out_bits) {       // [row]
   const RowJob job = jobs[blockIdx.x];
-  const int tid = threadIdx.x;            // 512 threads (8 waves)
+  const int tid = threadIdx.x;
+  const int NT = blockDim.x;   // 512 (many blocks: throughput regime)
+                               // or 1024 (few blocks: latency regime)
   const bool i_slice = (job.flags & 1) != 0;
   const int seg_mbw = job.seg_mbw;
   const size_t row_base = (size_t)job.mb_row * mbw + job.mbx0;
@@ -334,10 +336,10 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
   uint32_t* row_out = out + (size_t)blockIdx.x * out_stride_words;
 
   // ---- precompute per-MB info into LDS (totals, cbp, modes, mvs)
-  __shared__ MbInfo s_mb[512];
-  __shared__ short s_skiprun[512];
+  __shared__ MbInfo s_mb[kMaxSegMbw];
+  __shared__ short s_skiprun[kMaxSegMbw];
   __shared__ int s_trailing;
-  for (int mb = tid; mb < seg_mbw; mb += 512)
+  for (int mb = tid; mb < seg_mbw; mb += NT)
     precompute_mb(levels, meta, row_base + mb, i_slice, &s_mb[mb]);
   __syncthreads();
   if (tid == 0) {
@@ -356,7 +358,7 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
   __syncthreads();
 
   // ---- phase A: encode items
-  for (int item = tid; item < nitems; item += 512) {
+  for (int item = tid; item < nitems; item += NT) {
     DevBW bw;
     bw.out = row_stage + (size_t)item * kStageWordsPerItem;
     int bits = 0;
@@ -467,8 +469,8 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
   __syncthreads();
 
   // ---- phase B: exclusive prefix sum via per-thread segments
-  const int seg_lo = (int)((long)nitems * tid / 512);
-  const int seg_hi = (int)((long)nitems * (tid + 1) / 512);
+  const int seg_lo = (int)((long)nitems * tid / NT);
+  const int seg_hi = (int)((long)nitems * (tid + 1) / NT);
   int seg_sum = 0;
   for (int i = seg_lo; i < seg_hi; ++i) seg_sum += row_nbits[i];
   __shared__ int s_total;
@@ -479,13 +481,13 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
       int other = __shfl_up(inc, d);
       if (lane >= d) inc += other;
     }
-    __shared__ int s_wsum[8];
+    __shared__ int s_wsum[16];
     if (lane == 63) s_wsum[wid] = inc;
     __syncthreads();
     int wbase = 0;
     for (int w = 0; w < wid; ++w) wbase += s_wsum[w];
     int seg_off = wbase + inc - seg_sum;
-    if (tid == 511) s_total = wbase + inc;
+    if (tid == NT - 1) s_total = wbase + inc;
     // rewrite nbits -> offsets for this thread's segment
     int off = seg_off;
     for (int i = seg_lo; i < seg_hi; ++i) {
@@ -500,10 +502,10 @@ __global__ void __launch_bounds__(512) k_h264_cavlc_rows(
 
   // ---- zero the needed output words, then scatter
   int total_words = (total_bits + 31) / 32 + 1;
-  for (int wdx = tid; wdx < total_words; wdx += 512) row_out[wdx] = 0;
+  for (int wdx = tid; wdx < total_words; wdx += NT) row_out[wdx] = 0;
   __syncthreads();
 
-  for (int item = tid; item < nitems; item += 512) {
+  for (int item = tid; item < nitems; item += NT) {
     int off = row_nbits[item];
     int next_off = item + 1 < nitems
                        ? row_nbits[item + 1]
@@ -541,7 +543,12 @@ void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
                        int* d_nbits, uint32_t* d_out, int out_stride_words,
                        int* d_out_bits, hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(512), 0, stream,
+  // few blocks (chip underfilled, per-row latency-bound): widest blocks
+  // shorten each row's serial item chain. Many blocks (throughput-bound,
+  // e.g. 4K/8K): 512-thread blocks co-reside 3-per-CU and win on
+  // occupancy (measured: 1024 threads = +16% @1080p but -13% @8K).
+  const int nt = n_jobs <= 192 ? 1024 : 512;
+  hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(nt), 0, stream,
                      d_levels, d_meta, mbw, d_jobs, d_stage, d_nbits, d_out,
                      out_stride_words, d_out_bits);
 }
