@@ -860,6 +860,33 @@ class HipH264Pipeline : public EncodePipeline {
     d_refY2_ = dalloc(static_cast<size_t>(ypitch_ / 4) * mbh_ * 4);
     d_meta_ = reinterpret_cast<int*>(dalloc(meta_bytes));
     segs_ = (mbw_ + h264gpu::kMaxSegMbw - 1) / h264gpu::kMaxSegMbw;
+    // Latency shaping: the row kernels are serial in the MB chain of one
+    // slice, and a frame whose rows alone underfill the 256-CU chip is
+    // LATENCY-bound — so split each row into more slices until there is
+    // ~one workgroup per CU, as long as chains keep >=20 MBs (slice
+    // restarts reset intra/MVP context, so ultra-short chains waste bits
+    // and pipeline warmup). 1080p: 68 rows -> 4 slices of 30 MBs.
+    // Only frames with long rows benefit: each extra slice costs a NAL
+    // (host assembly + ~0.1% bits at 1080p noise), and at <=720p the
+    // host side becomes the bound before the GPU chain does (measured:
+    // 4 slices/row = -24% at 720p). Keep total jobs <= 192 so the CAVLC
+    // kernel stays in its wide 1024-thread regime — at 1080p that makes
+    // 2 slices of 60 MBs the sweet spot (864 -> 1252 fps; 4 slices with
+    // the 512-thread CAVLC measured slower on the same box).
+    {
+      int fill = (256 + mbh_ - 1) / std::max(1, mbh_);
+      int chain_cap = std::max(1, mbw_ / 30);
+      int job_cap = std::max(1, 192 / std::max(1, mbh_));
+      if (mbw_ >= 96 && job_cap > 1)
+        segs_ = std::max(segs_, std::min(std::min(fill, job_cap),
+                                         std::min(chain_cap, 8)));
+      if (const char* e = std::getenv("HIPFLUX_SEGS")) {
+        int v = std::atoi(e);
+        if (v >= 1)
+          segs_ = std::max((mbw_ + h264gpu::kMaxSegMbw - 1) /
+                               h264gpu::kMaxSegMbw, v);
+      }
+    }
     seg_w0_ = (mbw_ + segs_ - 1) / segs_;   // widest segment
     const int max_jobs = mbh_ * segs_;
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&h_levels_), level_bytes,
