@@ -866,18 +866,19 @@ class HipH264Pipeline : public EncodePipeline {
     // ~one workgroup per CU, as long as chains keep >=20 MBs (slice
     // restarts reset intra/MVP context, so ultra-short chains waste bits
     // and pipeline warmup). 1080p: 68 rows -> 4 slices of 30 MBs.
-    // Only frames with long rows benefit: each extra slice costs a NAL
-    // (host assembly + ~0.1% bits at 1080p noise), and at <=720p the
-    // host side becomes the bound before the GPU chain does (measured:
-    // 4 slices/row = -24% at 720p). Keep total jobs <= 192 so the CAVLC
-    // kernel stays in its wide 1024-thread regime — at 1080p that makes
-    // 2 slices of 60 MBs the sweet spot (864 -> 1252 fps; 4 slices with
-    // the 512-thread CAVLC measured slower on the same box).
+    // Each extra slice costs a NAL (host assembly + ~0.1% bits at
+    // 1080p noise), so the split is bounded three ways: enough segments
+    // to approach one workgroup per CU (fill), total jobs <= 192 so the
+    // CAVLC kernel keeps its wide 1024-thread shape (job_cap; exceeding
+    // it measured SLOWER: 4 slices @1080p fell back to 512-thread CAVLC
+    // and lost 40%), and chains >= 30 MBs so per-slice overhead stays
+    // amortized. Measured: 1080p 864 -> 1170-1252 fps with 2 slices of
+    // 60; 720p 1297 -> 1819 fps with 2 slices of 40.
     {
       int fill = (256 + mbh_ - 1) / std::max(1, mbh_);
       int chain_cap = std::max(1, mbw_ / 30);
       int job_cap = std::max(1, 192 / std::max(1, mbh_));
-      if (mbw_ >= 96 && job_cap > 1)
+      if (job_cap > 1)
         segs_ = std::max(segs_, std::min(std::min(fill, job_cap),
                                          std::min(chain_cap, 8)));
       if (const char* e = std::getenv("HIPFLUX_SEGS")) {
