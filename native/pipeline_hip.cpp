@@ -7,6 +7,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <chrono>
 #include <map>
 #include <stdexcept>
 
@@ -34,7 +35,9 @@ class HipJpegPipeline : public EncodePipeline {
  public:
   explicit HipJpegPipeline(const CaptureSettings& s)
       : settings_(s),
-        pool_(std::max(2u, std::thread::hardware_concurrency() / 2)) {
+        pool_(std::min(16u, std::max(
+                                2u,
+                                std::thread::hardware_concurrency() / 2))) {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     upload_dct_tables(stream_);
@@ -364,9 +367,14 @@ class HipH264Pipeline : public EncodePipeline {
  public:
   explicit HipH264Pipeline(const CaptureSettings& s)
       : settings_(s),
-        pool_(std::max(2u, std::thread::hardware_concurrency() / 2)) {
+        pool_(std::min(16u,
+                       std::max(2u, std::thread::hardware_concurrency() / 2))) {
+    // pool capped at 16: the per-frame batches are small, and waking a
+    // hardware_concurrency/2-sized pool (128 threads on a 256-core box)
+    // costs more in cv wakeups than the work itself
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&up_stream_, hipStreamNonBlocking));
     stripe_h_ = std::max(16, s.stripe_height & ~15);
     batch_events_.resize(4);
     for (auto& e : batch_events_)
@@ -382,6 +390,8 @@ class HipH264Pipeline : public EncodePipeline {
 
   ~HipH264Pipeline() override {
     (void)hipStreamSynchronize(stream_);
+    (void)hipStreamSynchronize(up_stream_);
+    (void)hipStreamDestroy(up_stream_);
     for (auto& e : batch_events_) (void)hipEventDestroy(e);
     for (int i = 0; i < 2; ++i) {
       (void)hipEventDestroy(ev_done_[i]);
@@ -460,11 +470,16 @@ class HipH264Pipeline : public EncodePipeline {
       std::memcpy(h_stage_, src, frame_bytes);
       src = h_stage_;
     }
-    HIP_CHECK(hipMemcpyAsync(d_frame_, src, frame_bytes,
-                             hipMemcpyHostToDevice, stream_));
+    // the ~8 MB BGRX upload rides its own stream so the SDMA engine
+    // overlaps it with the previous frame's kernels (measured ~120 us
+    // serialized on the compute stream); the compute stream's CSC waits
+    // on the copy via event
+    HIP_CHECK(hipMemcpyAsync(d_frame_[par], src, frame_bytes,
+                             hipMemcpyHostToDevice, up_stream_));
     // the caller may reuse its frame buffer once this event completes
-    HIP_CHECK(hipEventRecord(ev_h2d_[par], stream_));
-    launch_bgrx_to_planes(d_frame_, w_, h_, frame.stride / 4, d_srcY_,
+    HIP_CHECK(hipEventRecord(ev_h2d_[par], up_stream_));
+    HIP_CHECK(hipStreamWaitEvent(stream_, ev_h2d_[par], 0));
+    launch_bgrx_to_planes(d_frame_[par], w_, h_, frame.stride / 4, d_srcY_,
                           d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
                           stream_);
 
@@ -677,6 +692,13 @@ class HipH264Pipeline : public EncodePipeline {
   void collect_pending(Pending& pd, const Emit& emit) {
     if (!pd.active) return;
     pd.active = false;
+    const bool timing = std::getenv("HIPFLUX_TIMES") != nullptr;
+    auto tick = [] {
+      return std::chrono::duration<double, std::micro>(
+                 std::chrono::steady_clock::now().time_since_epoch())
+          .count();
+    };
+    double c0 = timing ? tick() : 0, c1 = 0, c2 = 0;
     const int par = pd.par;
     if (cpu_entropy_) {
       for (size_t b = 0; b < pd.batches.size(); ++b) {
@@ -698,6 +720,7 @@ class HipH264Pipeline : public EncodePipeline {
       HIP_CHECK(hipStreamSynchronize(stream_));
     } else {
       HIP_CHECK(hipEventSynchronize(ev_done_[par]));
+      if (timing) c1 = tick();
       int max_words = 0;
       for (int j = 0; j < pd.n_jobs; ++j) {
         int wds = (h_outbits_[par][j] + 31) / 32 + 1;
@@ -713,24 +736,44 @@ class HipH264Pipeline : public EncodePipeline {
         }
       }
       ent_copy_words_ = std::min(ent_stride_words_, max_words * 2 + 64);
-      for (int j = 0; j < pd.n_jobs; ++j) {
-        auto [si, sl] = pd.job_map[j];
-        const uint32_t* words =
-            h_entout_[par] + (size_t)j * ent_stride_words_;
-        int bits = h_outbits_[par][j];
-        bool idr = pd.outs[si].idr;
-        auto* dst = &pd.outs[si].rows[sl];
-        bool long_sc = sl == 0;
-        pool_.submit([words, bits, idr, long_sc, dst] {
-          h264::assemble_gpu_row_nal(words, bits, idr, long_sc, *dst);
+      // one strided task per worker (not one per NAL: 136 tiny submits
+      // per frame serialized on the pool mutex — measured 370 us/frame
+      // of wall time for ~40 us of actual assembly work)
+      const int nw = std::min(pool_.size(), pd.n_jobs);
+      for (int t = 0; t < nw; ++t) {
+        pool_.submit([this, &pd, par, t, nw] {
+          for (int j = t; j < pd.n_jobs; j += nw) {
+            auto [si, sl] = pd.job_map[j];
+            h264::assemble_gpu_row_nal(
+                h_entout_[par] + (size_t)j * ent_stride_words_,
+                h_outbits_[par][j], pd.outs[si].idr, sl == 0,
+                pd.outs[si].rows[sl]);
+          }
         });
       }
       pool_.wait_all();
     }
+    if (timing) c2 = tick();
+    // per-stripe concat on the pool (serial it cost ~90 us/frame)
+    if (pd.outs.size() > 1) {
+      for (auto& o : pd.outs)
+        pool_.submit([&o] {
+          o.bytes = std::move(o.header);
+          size_t total = o.bytes.size();
+          for (auto& r : o.rows) total += r.size();
+          o.bytes.reserve(total);
+          for (auto& r : o.rows)
+            o.bytes.insert(o.bytes.end(), r.begin(), r.end());
+        });
+      pool_.wait_all();
+    } else {
+      for (auto& o : pd.outs) {
+        o.bytes = std::move(o.header);
+        for (auto& r : o.rows)
+          o.bytes.insert(o.bytes.end(), r.begin(), r.end());
+      }
+    }
     for (auto& o : pd.outs) {
-      o.bytes = std::move(o.header);
-      for (auto& r : o.rows)
-        o.bytes.insert(o.bytes.end(), r.begin(), r.end());
       if (o.bytes.empty()) continue;
       EncodedStripe s;
       s.type = StripeType::kH264;
@@ -743,11 +786,24 @@ class HipH264Pipeline : public EncodePipeline {
       s.is_keyframe = o.idr;
       emit(s);
     }
+    if (timing && c1 > 0) {
+      tm_sync_ += c1 - c0;
+      tm_asm_ += c2 - c1;
+      tm_concat_ += tick() - c2;
+    }
   }
 
   void encode_frame(const RawFrame& frame, const FrameContext& ctx,
                     const Emit& emit) override {
+    const bool timing = std::getenv("HIPFLUX_TIMES") != nullptr;
+    auto tick = [] {
+      return std::chrono::duration<double, std::micro>(
+                 std::chrono::steady_clock::now().time_since_epoch())
+          .count();
+    };
+    double t0 = timing ? tick() : 0;
     Pending cur = submit_frame(frame, ctx);
+    double t1 = timing ? tick() : 0;
     if (!cur.active) {
       // nothing scheduled this frame; keep the pipe draining
       collect_pending(prev_, emit);
@@ -760,9 +816,27 @@ class HipH264Pipeline : public EncodePipeline {
     // depth 2: emit LAST frame's stripes (host assembly overlaps this
     // frame's GPU work), keep this one in flight
     collect_pending(prev_, emit);
+    double t2 = timing ? tick() : 0;
     prev_ = std::move(cur);
     // the caller may reuse its frame buffer after we return
     HIP_CHECK(hipEventSynchronize(ev_h2d_[prev_.par]));
+    if (timing) {
+      double t3 = tick();
+      tm_submit_ += t1 - t0;
+      tm_collect_ += t2 - t1;
+      tm_h2d_ += t3 - t2;
+      if (++tm_n_ == 256) {
+        std::fprintf(stderr,
+                     "[times us/frame] submit=%.0f collect=%.0f (sync=%.0f "
+                     "asm=%.0f concat=%.0f) h2dwait=%.0f\n",
+                     tm_submit_ / tm_n_, tm_collect_ / tm_n_,
+                     tm_sync_ / tm_n_, tm_asm_ / tm_n_, tm_concat_ / tm_n_,
+                     tm_h2d_ / tm_n_);
+        tm_submit_ = tm_collect_ = tm_h2d_ = tm_sync_ = tm_asm_ =
+            tm_concat_ = 0;
+        tm_n_ = 0;
+      }
+    }
   }
 
   void flush(const Emit& emit) override { collect_pending(prev_, emit); }
@@ -840,7 +914,8 @@ class HipH264Pipeline : public EncodePipeline {
       device_ptrs_.push_back(p);
       return static_cast<uint8_t*>(p);
     };
-    d_frame_ = dalloc(static_cast<size_t>(w) * h * 4);
+    d_frame_[0] = dalloc(static_cast<size_t>(w) * h * 4);
+    d_frame_[1] = dalloc(static_cast<size_t>(w) * h * 4);
     d_srcY_ = dalloc(ysz);
     d_srcCb_ = dalloc(csz);
     d_srcCr_ = dalloc(csz);
@@ -927,10 +1002,12 @@ class HipH264Pipeline : public EncodePipeline {
   CaptureSettings settings_;
   ThreadPool pool_;
   hipStream_t stream_{};
+  hipStream_t up_stream_{};
   int stripe_h_ = 64;
   int w_ = 0, h_ = 0, mbw_ = 0, mbh_ = 0, ypitch_ = 0, cpitch_ = 0;
   int segs_ = 1, seg_w0_ = 0;     // slices per MB row, widest segment
-  uint8_t *d_frame_ = nullptr, *d_srcY_ = nullptr, *d_srcCb_ = nullptr,
+  uint8_t* d_frame_[2] = {nullptr, nullptr};
+  uint8_t *d_srcY_ = nullptr, *d_srcCb_ = nullptr,
           *d_srcCr_ = nullptr, *d_refY_ = nullptr, *d_refCb_ = nullptr,
           *d_refCr_ = nullptr, *d_curY_ = nullptr, *d_curCb_ = nullptr,
           *d_curCr_ = nullptr;
@@ -959,6 +1036,10 @@ class HipH264Pipeline : public EncodePipeline {
   int depth_ = 1;                  // 1 = sync (latency mode), 2 = pipelined
   int parity_ = 0;
   Pending prev_;                   // the in-flight frame (depth 2)
+  // HIPFLUX_TIMES=1 phase accumulators (microseconds)
+  double tm_submit_ = 0, tm_collect_ = 0, tm_h2d_ = 0, tm_sync_ = 0,
+         tm_asm_ = 0, tm_concat_ = 0;
+  int tm_n_ = 0;
   int ent_stride_words_ = 0;
   int ent_copy_words_ = 1 << 30;   // adaptive D2H cap (words per row)
   bool rows_v1_ = std::getenv("HIPFLUX_ROWS_V1") != nullptr;
