@@ -181,6 +181,7 @@ void ScreenCapture::run() {
     pipeline = make_cpu_jpeg_pipeline(settings_);
   }
   pipeline_name_ = pipeline->name();
+  pipeline->set_pipeline_depth(settings_.pipeline_depth);
 
   const int scale_div =
       std::min(4, std::max(1, settings_.capture_scale_div));
@@ -357,26 +358,34 @@ void ScreenCapture::run() {
       if (f) std::fwrite(s.data, 1, s.size, f);
     };
 
+    size_t frame_bytes = 0;
+    auto handle_stripe = [&](EncodedStripe& s) {
+      frame_bytes += s.size;
+      record_stripe(s);
+      // pipelined emission stamps its own capture time (the stripe may
+      // belong to the PREVIOUS frame); only backfill when unset
+      if (s.capture_ts_ms == 0) s.capture_ts_ms = frame.ts_ms;
+      s.encode_done_ms = now_ms();
+      if (settings_.omit_stripe_headers) {
+        if (cb_) cb_(s);
+      } else {
+        wire.clear();
+        pack_wire_stripe(s, wire);
+        EncodedStripe ws = s;
+        ws.data = wire.data();
+        ws.size = wire.size();
+        if (cb_) cb_(ws);
+      }
+      stripes_emitted_.fetch_add(1);
+    };
+    if (!any) {
+      // no damage this frame: drain any pipelined frame promptly so
+      // depth-2 emission latency is bounded by the frame interval
+      pipeline->flush(handle_stripe);
+    }
     if (any) {
       double t0 = now_ms();
-      size_t frame_bytes = 0;
-      pipeline->encode_frame(frame, ctx, [&](EncodedStripe& s) {
-        frame_bytes += s.size;
-        record_stripe(s);
-        s.capture_ts_ms = frame.ts_ms;
-        s.encode_done_ms = now_ms();
-        if (settings_.omit_stripe_headers) {
-          if (cb_) cb_(s);
-        } else {
-          wire.clear();
-          pack_wire_stripe(s, wire);
-          EncodedStripe ws = s;
-          ws.data = wire.data();
-          ws.size = wire.size();
-          if (cb_) cb_(ws);
-        }
-        stripes_emitted_.fetch_add(1);
-      });
+      pipeline->encode_frame(frame, ctx, handle_stripe);
       last_encode_ms_.store(now_ms() - t0);
       frames_encoded_.fetch_add(1);
       ++frame_id;

@@ -42,6 +42,9 @@ struct CaptureSettings {
   double vbv_multiplier = 1.5;
   double keyframe_interval_s = 0.0; // 0 = infinite GOP, IDR on demand
   bool video_streaming_mode = false;
+  int pipeline_depth = 1;           // 2 = one encode frame in flight
+                                    // (throughput mode; emission lags one
+                                    // frame — recording/transcode use)
   bool video_fullcolor = false;     // 4:4:4
   bool use_paint_over_quality = true;
   int paint_over_trigger_frames = 15;

@@ -33,6 +33,7 @@ PYBIND11_MODULE(_native, m) {
       .def_readwrite("capture_cursor", &CaptureSettings::capture_cursor)
       .def_readwrite("output_mode", &CaptureSettings::output_mode)
       .def_readwrite("capture_scale_div", &CaptureSettings::capture_scale_div)
+      .def_readwrite("pipeline_depth", &CaptureSettings::pipeline_depth)
       .def_readwrite("video_fullframe", &CaptureSettings::video_fullframe)
       .def_readwrite("use_cpu", &CaptureSettings::use_cpu)
       .def_readwrite("gpu_id", &CaptureSettings::gpu_id)

@@ -435,6 +435,7 @@ class HipH264Pipeline : public EncodePipeline {
     int par = 0;
     int n_jobs = 0;
     int copy_words = 0;    // the D2H cap this frame was submitted with
+    double ts_ms = 0;      // capture timestamp of this frame
     uint32_t frame_id = 0;
     std::vector<Out> outs;
     std::vector<std::pair<int, int>> job_map;
@@ -652,6 +653,7 @@ class HipH264Pipeline : public EncodePipeline {
     pd.par = par;
     pd.n_jobs = n_jobs;
     pd.copy_words = copy_words;
+    pd.ts_ms = frame.ts_ms;
     pd.frame_id = ctx.frame_id;
     pd.batches = std::move(batches);
     pd.outs.resize(sjobs.size());
@@ -780,6 +782,7 @@ class HipH264Pipeline : public EncodePipeline {
       s.data = o.bytes.data();
       s.size = o.bytes.size();
       s.frame_id = pd.frame_id;
+      s.capture_ts_ms = pd.ts_ms;
       s.y = o.y0;
       s.width = w_;
       s.height = o.h;

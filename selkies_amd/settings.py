@@ -162,6 +162,11 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                value_range=(0.0, 600.0)),
     SettingDef("video_fullcolor", bool, False, "4:4:4 chroma (I444) instead of 4:2:0.",
                client=True),
+    SettingDef("video_pipeline_depth", int, 1,
+               "Encoder frame pipelining depth: 1 = synchronous (lowest "
+               "interactive latency), 2 = one frame in flight (throughput "
+               "mode for recording/transcode; emission lags one frame).",
+               value_range=(1, 2)),
     SettingDef("video_fullframe", bool, False,
                "Always encode the full frame (disable damage gating).", client=True),
     SettingDef("video_streaming_mode", bool, False,

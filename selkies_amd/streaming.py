@@ -163,6 +163,7 @@ class StreamingService:
         cs.video_fullcolor = s.video_fullcolor
         cs.video_fullframe = s.video_fullframe
         cs.capture_scale_div = s.capture_scale_div
+        cs.pipeline_depth = s.video_pipeline_depth
         cs.use_paint_over_quality = s.use_paint_over_quality
         cs.paint_over_trigger_frames = s.paint_over_trigger_frames
         cs.video_paintover_crf = s.video_paintover_crf

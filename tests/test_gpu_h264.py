@@ -511,3 +511,41 @@ def test_pipelined_bench_pipeline_contract():
     assert p2.last_frame_id == n - 1
     sizes2.append(b)
     assert sizes1 == sizes2
+
+
+def test_gpu_engine_pipelined_depth2():
+    """Engine path with pipeline_depth=2 (throughput mode): stripes keep
+    flowing with per-stripe capture timestamps stamped by the pipeline
+    (emission lags one frame), and the stream still decodes."""
+    require_gpu()
+    import threading
+    import time
+
+    s = _native.CaptureSettings()
+    s.capture_width = 640
+    s.capture_height = 384
+    s.target_fps = 120
+    s.output_mode = 1
+    s.use_cpu = False
+    s.gpu_id = 0
+    s.capture_backend = "synthetic:desktop"
+    s.stripe_height = 64
+    s.pipeline_depth = 2
+    s.video_fullframe = True          # exercise the steady pipelined path
+    got = {"n": 0, "key": 0}
+    done = threading.Event()
+
+    def cb(data, frame_id, y, width, height, key, *a):
+        got["n"] += 1
+        if key:
+            got["key"] += 1
+        if got["n"] > 600:
+            done.set()
+
+    cap = _native.ScreenCapture()
+    cap.start_capture(cb, s)
+    done.wait(10)
+    assert cap.is_capturing
+    assert cap.pipeline.startswith("hip")
+    cap.stop_capture()
+    assert got["n"] > 600 and got["key"] >= 1
