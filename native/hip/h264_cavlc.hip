@@ -543,11 +543,12 @@ void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
                        int* d_nbits, uint32_t* d_out, int out_stride_words,
                        int* d_out_bits, hipStream_t stream) {
   if (n_jobs == 0) return;
-  // few blocks (chip underfilled, per-row latency-bound): widest blocks
-  // shorten each row's serial item chain. Many blocks (throughput-bound,
-  // e.g. 4K/8K): 512-thread blocks co-reside 3-per-CU and win on
-  // occupancy (measured: 1024 threads = +16% @1080p but -13% @8K).
-  const int nt = n_jobs <= 192 ? 1024 : 512;
+  // few blocks (one 1024-thread workgroup per CU fits up to the 256-CU
+  // count, per-row latency-bound): widest blocks shorten each row's
+  // serial item chain. Many blocks (throughput-bound, e.g. 4K/8K):
+  // 512-thread blocks co-reside 3-per-CU and win on occupancy
+  // (measured: 1024 threads = +16% @1080p but -13% @8K).
+  const int nt = n_jobs <= 256 ? 1024 : 512;
   hipLaunchKernelGGL(k_h264_cavlc_rows, dim3(n_jobs), dim3(nt), 0, stream,
                      d_levels, d_meta, mbw, d_jobs, d_stage, d_nbits, d_out,
                      out_stride_words, d_out_bits);

@@ -946,16 +946,17 @@ class HipH264Pipeline : public EncodePipeline {
     // and pipeline warmup). 1080p: 68 rows -> 4 slices of 30 MBs.
     // Each extra slice costs a NAL (host assembly + ~0.1% bits at
     // 1080p noise), so the split is bounded three ways: enough segments
-    // to approach one workgroup per CU (fill), total jobs <= 192 so the
-    // CAVLC kernel keeps its wide 1024-thread shape (job_cap; exceeding
-    // it measured SLOWER: 4 slices @1080p fell back to 512-thread CAVLC
-    // and lost 40%), and chains >= 30 MBs so per-slice overhead stays
-    // amortized. Measured: 1080p 864 -> 1170-1252 fps with 2 slices of
-    // 60; 720p 1297 -> 1819 fps with 2 slices of 40.
+    // to approach one workgroup per CU (fill), total jobs <= 256 so the
+    // CAVLC kernel keeps its wide 1024-thread shape at one block per CU
+    // (job_cap; exceeding it measured SLOWER: 272 jobs @1080p fell back
+    // to 512-thread CAVLC and lost 40%), and chains >= 30 MBs so
+    // per-slice overhead stays amortized. Measured: 1080p 864 -> 1252
+    // fps with 2 slices, ~1580 with upload overlap; 720p 1297 -> 1819
+    // with 2 slices of 40.
     {
       int fill = (256 + mbh_ - 1) / std::max(1, mbh_);
       int chain_cap = std::max(1, mbw_ / 30);
-      int job_cap = std::max(1, 192 / std::max(1, mbh_));
+      int job_cap = std::max(1, 256 / std::max(1, mbh_));
       if (job_cap > 1)
         segs_ = std::max(segs_, std::min(std::min(fill, job_cap),
                                          std::min(chain_cap, 8)));
