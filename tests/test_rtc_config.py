@@ -152,3 +152,54 @@ def test_file_monitor_detects_change(tmp_path):
             mon.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_rtc_config_push_to_clients(tmp_path):
+    """End to end: changing the rtc_config_json file pushes RTC_CONFIG
+    to connected WS clients via the server's file monitor."""
+    import pytest
+    hipflux = pytest.importorskip("hipflux")
+    if not hipflux.native_available():
+        pytest.skip("hipflux native module not built")
+    import aiohttp
+    from aiohttp import WSMsgType
+    from test_server import make_server, start_on_free_port
+
+    p = tmp_path / "rtc.json"
+    p.write_text(json.dumps({"iceServers": [{"urls": ["stun:a:1"]}]}))
+
+    async def main():
+        server = make_server(SELKIES_RTC_CONFIG_JSON=str(p))
+        runner, port = await start_on_free_port(server)
+        # start_on_free_port bypasses server.start(); run the monitor
+        # ourselves at a test-friendly poll interval
+        server._rtc_monitor = T.RTCConfigFileMonitor(
+            str(p), server._on_rtc_config_change, interval_s=0.05)
+        server._rtc_monitor.start()
+        try:
+            async with aiohttp.ClientSession() as sess:
+                ws = await sess.ws_connect(
+                    f"http://127.0.0.1:{port}/websockets")
+                await asyncio.sleep(0.2)
+                p.write_text(json.dumps(
+                    {"iceServers": [{"urls": ["turn:new:5349"],
+                                     "username": "u",
+                                     "credential": "c"}]}))
+                deadline = asyncio.get_event_loop().time() + 5
+                got = None
+                while asyncio.get_event_loop().time() < deadline:
+                    msg = await ws.receive(timeout=5)
+                    if (msg.type == WSMsgType.TEXT and
+                            msg.data.startswith("RTC_CONFIG,")):
+                        got = json.loads(msg.data.split(",", 1)[1])
+                        break
+                assert got is not None, "no RTC_CONFIG push received"
+                assert got["iceServers"][0]["urls"] == ["turn:new:5349"]
+                await ws.close()
+        finally:
+            await server.stop()
+            server.streaming.stop_capture()
+            server.streaming.stop_audio()
+            await runner.cleanup()
+
+    asyncio.new_event_loop().run_until_complete(main())
