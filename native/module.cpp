@@ -395,6 +395,15 @@ PYBIND11_MODULE(_native, m) {
                              [](BenchPipeline& b) { return b.p->name(); })
       .def_readonly("last_frame_id", &BenchPipeline::last_frame_id)
       .def(
+          "resize",
+          [](BenchPipeline& b, int w, int h) {
+            b.w = w;
+            b.h = h;
+          },
+          py::arg("w"), py::arg("h"),
+          "Change frame dimensions (next encode reallocates; an "
+          "in-flight pipelined frame is dropped, streams restart IDR).")
+      .def(
           "flush",
           [](BenchPipeline& b) {
             size_t total = 0;
