@@ -45,21 +45,27 @@ def test_controller_assignment_and_promotion(loop):
                 ws2 = await sess.ws_connect(url)
                 assert await read_role(ws2) == "viewer"
 
+                async def wait_event(ev, timeout=5.0):
+                    deadline = asyncio.get_event_loop().time() + timeout
+                    while asyncio.get_event_loop().time() < deadline:
+                        if ev in backend.events:
+                            return True
+                        await asyncio.sleep(0.02)
+                    return False
+
                 # viewer input is ignored
                 await ws2.send_str("kd,120")
-                await asyncio.sleep(0.2)
-                assert ("key", 120, True) not in backend.events
-                # controller input lands
+                # controller input lands (also orders the viewer check:
+                # by the time 121 arrived, 120 would have too)
                 await ws1.send_str("kd,121")
-                await asyncio.sleep(0.2)
-                assert ("key", 121, True) in backend.events
+                assert await wait_event(("key", 121, True))
+                assert ("key", 120, True) not in backend.events
 
                 # controller leaves -> viewer promoted
                 await ws1.close()
                 assert await read_role(ws2) == "controller"
                 await ws2.send_str("kd,122")
-                await asyncio.sleep(0.2)
-                assert ("key", 122, True) in backend.events
+                assert await wait_event(("key", 122, True))
                 await ws2.close()
         finally:
             server.streaming.stop_capture()
