@@ -92,14 +92,20 @@ def test_seat_claim_remap_and_release(loop):
                 # the player's pads remap onto seat slot 1
                 await ws2.send_str("js,c,0,UGFk,18,4")
                 await ws2.send_str("js,b,0,3,1")
-                await asyncio.sleep(0.2)
+                deadline = asyncio.get_event_loop().time() + 5
+                while (len(hub.msgs) < 2 and
+                       asyncio.get_event_loop().time() < deadline):
+                    await asyncio.sleep(0.02)
                 assert hub.msgs == ["js,c,1,UGFk,18,4", "js,b,1,3,1"]
 
                 # controller still drives slot 0, but claimed slot 1
                 # is protected from it
                 await ws1.send_str("js,b,0,0,1")
                 await ws1.send_str("js,b,1,0,1")
-                await asyncio.sleep(0.2)
+                deadline = asyncio.get_event_loop().time() + 5
+                while (hub.msgs[-1] != "js,b,0,0,1" and
+                       asyncio.get_event_loop().time() < deadline):
+                    await asyncio.sleep(0.02)
                 assert hub.msgs[-1] == "js,b,0,0,1"
 
                 # disconnect releases the seat and unplugs its pad
