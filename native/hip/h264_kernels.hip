@@ -967,16 +967,22 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     }
   };
 
+  // pass 0: previous-frame tracking state (meta still holds last
+  // frame's values). An MB whose LAST frame was hopeless (unmatchable
+  // at any candidate — noise) skips every search pass except a
+  // 1-in-8-frame probe; the sad(0,0) skip/inter test below still runs
+  // every frame, so content that settles becomes P_Skip immediately —
+  // and going skip clears the hopeless bit, resuming full search.
+  const int pm0 = meta[mb_index * kMetaPerMb + 0];
+  const int pm1 = meta[mb_index * kMetaPerMb + 1];
+  const bool prev_hopeless = (pm0 >> 22) & 1;
+  if (!(prev_hopeless && (job.frame_num & 7) != 0)) {
   // pass 1: fine grid at (0,0)
   score_grid(0, 0, 2);
   // pass 2: fine grid at the previous frame's MV/hint for this MB (meta
   // still holds last frame's values here) — tracks sustained motion
   // beyond +-8 even across intra fallbacks
-  bool prev_hopeless;
   {
-    int pm0 = meta[mb_index * kMetaPerMb + 0];
-    int pm1 = meta[mb_index * kMetaPerMb + 1];
-    prev_hopeless = (pm0 >> 22) & 1;
     int pvx = 0, pvy = 0;
     if ((pm0 & 3) == kInter) {
       pvx = ((int)(short)(pm1 & 0xFFFF)) >> 2;
@@ -993,10 +999,8 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
   // pass 3: pyramid acquisition when nothing fits yet — first frame of
   // fast new motion. A +-12 step-1 SAD search on the QUARTER-res luma
   // pyramid covers +-48 at full res with correlation intact at that
-  // scale; the winner seeds a fine full-res grid. MBs whose LAST frame
-  // was hopeless (noise) skip this except a 1-in-8-frame probe.
-  if (!(prev_hopeless && (job.frame_num & 7) != 0) &&
-      __shfl(best_score, 0) > 256 * 180) {
+  // scale; the winner seeds a fine full-res grid.
+  if (__shfl(best_score, 0) > 256 * 180) {
     // this MB at quarter res: 4x4 proxy block
     const int qx0 = x0 >> 2, qy0 = y0 >> 2;
     const int qw = frame_w_mb16 >> 2;
@@ -1044,6 +1048,7 @@ __global__ void __launch_bounds__(64) k_h264_me_mfma(
     // fine full-res verification grid at the pyramid seed
     if (bs < INT_MAX) score_grid(bmx2 * 4, bmy2 * 4, 2);
   }
+  }  // end non-hopeless search passes
 
   best_score = __shfl(best_score, 0);
   best_mvx = __shfl(best_mvx, 0);
