@@ -84,6 +84,7 @@ class CentralizedStreamServer:
         app.router.add_get("/metrics", self.handle_metrics)
         app.router.add_get("/api/turn", self.handle_turn)
         app.router.add_post("/api/webrtc-stats", self.handle_webrtc_stats)
+        app.router.add_post("/api/mode", self.handle_mode)
         from .computer_use import ComputerUseAPI
         ComputerUseAPI(self.settings, self.streaming.input).register(app)
         app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
@@ -205,6 +206,33 @@ class CentralizedStreamServer:
             gg.set(gpu.get("busy_percent", 0))
         return web.Response(body=generate_latest(reg),
                             content_type="text/plain; version=0.0.4")
+
+    async def handle_mode(self, request):
+        """Runtime transport-mode switch (reference switch_to_mode,
+        stream_server.py:1300). POST {"mode": "websockets"|"webrtc"}:
+        switching to webrtc eagerly brings the RTC stack up (UDP
+        endpoint live before clients arrive); switching back tears it
+        down and drops its peers. The WS plane always stays up — it
+        carries control/signaling for both modes."""
+        try:
+            body = await request.json()
+            mode = body["mode"]
+        except Exception:
+            raise web.HTTPBadRequest(reason='JSON {"mode": ...} required')
+        if mode not in ("websockets", "webrtc"):
+            raise web.HTTPBadRequest(reason="unknown mode")
+        if mode == "webrtc":
+            if self.webrtc is None:
+                from .webrtc_service import WebRTCService
+                self.webrtc = WebRTCService(self.settings, self.streaming)
+                await self.webrtc.start(self.settings.webrtc_udp_port)
+        else:
+            if self.webrtc is not None:
+                await self.webrtc.stop()
+                self.webrtc = None
+        self.settings.mode = mode
+        return web.json_response({"mode": mode,
+                                  "webrtc_active": self.webrtc is not None})
 
     # ---- WebRTC signaling ---------------------------------------------------
     async def handle_webrtc_offer(self, request):

@@ -292,3 +292,42 @@ def test_webrtc_stats_recorder(tmp_path):
         assert "bad,key" not in lines[0]    # oversized key dropped
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_mode_switch_runtime():
+    """POST /api/mode switches transports at runtime (reference
+    switch_to_mode): webrtc brings the RTC stack up eagerly, websockets
+    tears it down; /api/status reflects the active mode."""
+    import asyncio
+    import aiohttp
+
+    async def main():
+        srv = make_server()
+        runner, port = await start_on_free_port(srv)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                base = f"http://127.0.0.1:{port}"
+                r = await sess.post(base + "/api/mode",
+                                    json={"mode": "webrtc"})
+                assert r.status == 200
+                body = await r.json()
+                assert body == {"mode": "webrtc", "webrtc_active": True}
+                assert srv.webrtc is not None
+                st = await (await sess.get(base + "/api/status")).json()
+                assert st["mode"] == "webrtc"
+
+                r = await sess.post(base + "/api/mode",
+                                    json={"mode": "websockets"})
+                assert (await r.json())["webrtc_active"] is False
+                assert srv.webrtc is None
+
+                r = await sess.post(base + "/api/mode",
+                                    json={"mode": "bogus"})
+                assert r.status == 400
+        finally:
+            await srv.stop()
+            srv.streaming.stop_capture()
+            srv.streaming.stop_audio()
+            await runner.cleanup()
+
+    asyncio.new_event_loop().run_until_complete(main())
