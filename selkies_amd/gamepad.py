@@ -22,6 +22,7 @@ from __future__ import annotations
 
 import asyncio
 import base64
+import binascii
 import logging
 import os
 import struct
@@ -275,6 +276,12 @@ class GamepadHub:
         self.pads: dict[int, object] = {}
 
     async def handle(self, msg: str):
+        try:
+            await self._handle(msg)
+        except (ValueError, OverflowError, IndexError, binascii.Error):
+            logger.debug("malformed js verb dropped: %r", msg[:80])
+
+    async def _handle(self, msg: str):
         parts = msg.split(",")
         if len(parts) < 3 or parts[0] != "js":
             return

@@ -89,3 +89,38 @@ def test_wire_binary_handler_survives_garbage():
                 P.inflate_gz_bounded(data[1:], limit=1 << 20)
             except (OSError, EOFError, ValueError):
                 pass  # rejecting is fine; raising odd errors is not
+
+
+def test_streaming_text_verbs_survive_garbage():
+    """The streaming service's own control verbs (seats, gamepad remap,
+    settings, acks) must shrug off malformed input from a hostile
+    client without raising."""
+    import asyncio
+
+    from selkies_amd.streaming import _js_index, _remap_js
+
+    # pure helpers
+    for t in ("", "js", "js,b", "js,b,,1", "js,b,-5,0,1", "x" * 4096,
+              "js,b,99999999999999999999,0,1"):
+        _js_index(t)
+        _remap_js(t, 1)
+
+    # GamepadHub full-verb fuzz (drives the same parse paths a seated
+    # client's remapped messages hit)
+    import selkies_amd.gamepad as G
+
+    async def main():
+        hub = G.GamepadHub(socket_dir="/tmp/selkies_fuzz_js",
+                           prefer_uinput=False)
+        bad = [
+            "js,c,0", "js,c,0,!!!notb64!!!,x,y", "js,c,-1,QQ==,1,1",
+            "js,c,99,QQ==,1,1", "js,b,0,0,1", "js,a,0,0,nan",
+            "js,a,0,0,1e309", "js,b,0,abc,1", "js,d,0", "js,d,0",
+            "js,,,", "js,zzz,0",
+        ]
+        for m in bad:
+            await hub.handle(m)   # never raises: a raise here would
+                                  # disconnect the sending client
+        await hub.close()
+
+    asyncio.new_event_loop().run_until_complete(main())
