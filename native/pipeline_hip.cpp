@@ -945,14 +945,17 @@ class HipH264Pipeline : public EncodePipeline {
     // restarts reset intra/MVP context, so ultra-short chains waste bits
     // and pipeline warmup). 1080p: 68 rows -> 4 slices of 30 MBs.
     // Each extra slice costs a NAL (host assembly + ~0.1% bits at
-    // 1080p noise), so the split is bounded three ways: enough segments
-    // to approach one workgroup per CU (fill), total jobs <= 256 so the
-    // CAVLC kernel keeps its wide 1024-thread shape at one block per CU
-    // (job_cap; exceeding it measured SLOWER: 272 jobs @1080p fell back
-    // to 512-thread CAVLC and lost 40%), and chains >= 30 MBs so
-    // per-slice overhead stays amortized. Measured: 1080p 864 -> 1252
-    // fps with 2 slices, ~1580 with upload overlap; 720p 1297 -> 1819
-    // with 2 slices of 40.
+    // 1080p noise). Two regimes, both measured:
+    //  * latency (rows alone underfill the chip, job_cap > 1): split
+    //    toward one workgroup per CU but keep total jobs <= 256 so the
+    //    CAVLC kernel stays in its wide 1024-thread shape (exceeding it
+    //    fell back to 512 threads and lost 20-40%), chains >= 30 MBs.
+    //    1080p -> 3 slices of 40 (864 -> 2127 fps with the rest of v25+).
+    //  * throughput (4K/8K: enough rows to fill the chip): the tail of
+    //    each wave-round is still chain-latency-bound, so short chains
+    //    win regardless of CAVLC width — cap chains at ~30 MBs, at most
+    //    8 slices/row. 4K 513 -> 711 fps (8 slices of 30), 8K 193 -> 205
+    //    (8 slices of 60; 12 slices measured slightly worse).
     {
       int fill = (256 + mbh_ - 1) / std::max(1, mbh_);
       int chain_cap = std::max(1, mbw_ / 30);
@@ -960,6 +963,8 @@ class HipH264Pipeline : public EncodePipeline {
       if (job_cap > 1)
         segs_ = std::max(segs_, std::min(std::min(fill, job_cap),
                                          std::min(chain_cap, 8)));
+      else
+        segs_ = std::max(segs_, std::min(chain_cap, 8));
       if (const char* e = std::getenv("HIPFLUX_SEGS")) {
         int v = std::atoi(e);
         if (v >= 1)
