@@ -137,7 +137,8 @@ def main():
         # pipelining a frame completes during a later call and its true
         # submit->done latency is reported
         starts = {}
-        next_fid = args.warmup
+        # warmup performed 1 + max(1, warmup-1) encodes (frame ids from 0)
+        next_fid = 1 + max(1, args.warmup - 1)
         t0 = time.perf_counter()
         for i in range(args.steps):
             starts[next_fid] = time.perf_counter()
@@ -184,6 +185,8 @@ def main():
             lat_ms.extend(q)
         total_bytes = sum(nb)
 
+    if not lat_ms:
+        lat_ms = [0.0]
     elapsed = t1 - t0
     # MAX elapsed over ranks (slowest rank defines job throughput)
     if distributed:
