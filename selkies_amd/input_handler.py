@@ -219,7 +219,10 @@ class InputDispatcher:
             parts = rest.split(",")
             button = {"u": BTN_SCROLL_UP, "d": BTN_SCROLL_DOWN,
                       "l": BTN_SCROLL_LEFT, "r": BTN_SCROLL_RIGHT}[parts[0]]
-            for _ in range(int(parts[1]) if len(parts) > 1 else 1):
+            # clamp: the count is client-supplied and each tick is a
+            # synchronous XTEST round-trip on the event loop
+            count = min(int(parts[1]) if len(parts) > 1 else 1, 100)
+            for _ in range(count):
                 if self.enable_input:
                     self.backend.mouse_button(button, True)
                     self.backend.mouse_button(button, False)

@@ -53,7 +53,6 @@ class VideoRelay:
         # stats
         self.sent_frames = 0
         self.sent_bytes = 0
-        self.sent_bytes = 0
         self.dropped_frames = 0
 
     def set_bitrate(self, bitrate_bps: float) -> None:
@@ -114,7 +113,6 @@ class VideoRelay:
                 self._mark_dead()
                 return
             self.sent_frames += 1
-            self.sent_bytes += len(payload)
             self.sent_bytes += len(payload)
             # long (but sub-timeout) sends show congestion; let backpressure
             # logic observe it via stats rather than acting here

@@ -14,6 +14,7 @@ from __future__ import annotations
 
 import asyncio
 import base64
+import hmac
 import json
 import logging
 import os
@@ -122,8 +123,15 @@ class CentralizedStreamServer:
         token = s.auth_token
         if token:
             supplied = request.headers.get("Authorization", "")
-            qtoken = request.query.get("token", "")
-            if supplied != f"Bearer {token}" and qtoken != token:
+            ok = hmac.compare_digest(supplied, f"Bearer {token}")
+            if not ok:
+                # query-parameter tokens leak into logs/history; accept
+                # them only where headers are impossible: the WS upgrade
+                is_ws = (request.headers.get("Upgrade", "").lower()
+                         == "websocket")
+                qtoken = request.query.get("token", "") if is_ws else ""
+                ok = bool(qtoken) and hmac.compare_digest(qtoken, token)
+            if not ok:
                 raise web.HTTPUnauthorized(reason="token required")
         elif s.enable_basic_auth:
             hdr = request.headers.get("Authorization", "")
@@ -132,8 +140,8 @@ class CentralizedStreamServer:
                 try:
                     user, _, pw = base64.b64decode(
                         hdr[6:]).decode().partition(":")
-                    ok = (user == s.basic_auth_user and
-                          pw == s.basic_auth_password)
+                    ok = (hmac.compare_digest(user, s.basic_auth_user) and
+                          hmac.compare_digest(pw, s.basic_auth_password))
                 except Exception:
                     ok = False
             if not ok:

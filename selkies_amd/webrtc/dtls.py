@@ -66,6 +66,11 @@ for name, restype, argtypes in [
      [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_int]),
     ("BIO_ctrl_pending", ctypes.c_size_t, [ctypes.c_void_p]),
     ("BIO_set_mem_eof_return", None, []),
+    # peer-certificate fingerprint (OpenSSL 3.x name first, 1.1 fallback)
+    ("SSL_get1_peer_certificate", ctypes.c_void_p, [ctypes.c_void_p]),
+    ("SSL_get_peer_certificate", ctypes.c_void_p, [ctypes.c_void_p]),
+    ("i2d_X509", ctypes.c_int, [ctypes.c_void_p, ctypes.c_void_p]),
+    ("X509_free", None, [ctypes.c_void_p]),
 ]:
     fn = None
     for lib in (_ssl, _crypto):
@@ -81,7 +86,16 @@ for name, restype, argtypes in [
     setattr(api, name, fn)
 
 SSL_ERROR_WANT_READ = 2
+SSL_VERIFY_PEER = 0x01
+SSL_VERIFY_FAIL_IF_NO_PEER_CERT = 0x02
 SRTP_PROFILE = b"SRTP_AES128_CM_SHA1_80"
+
+# verify callback that accepts any chain: WebRTC certs are self-signed and
+# authenticated by comparing the cert fingerprint against the signaled
+# a=fingerprint (RFC 8122), which peer_fingerprint() enables after the
+# handshake. The callback only forces the peer to PRESENT a certificate.
+_VERIFY_CB_T = ctypes.CFUNCTYPE(ctypes.c_int, ctypes.c_int, ctypes.c_void_p)
+_accept_any_cert_cb = _VERIFY_CB_T(lambda preverify, store_ctx: 1)
 BIO_CTRL_DGRAM_SET_MTU = 42  # not required; memory BIOs fragment for us
 
 
@@ -123,7 +137,13 @@ class DtlsEndpoint:
         assert api.SSL_CTX_use_PrivateKey_file(
             self._ctx, cert.key.encode(), 1) == 1
         assert api.SSL_CTX_set_tlsext_use_srtp(self._ctx, SRTP_PROFILE) == 0
-        api.SSL_CTX_set_verify(self._ctx, 0, None)  # fingerprint is in SDP
+        # Require the peer to present a certificate so its fingerprint can
+        # be checked against the SDP a=fingerprint (servers with
+        # verify off accepted any client). Chain validation stays off — the
+        # cert is self-signed by design.
+        api.SSL_CTX_set_verify(
+            self._ctx, SSL_VERIFY_PEER | SSL_VERIFY_FAIL_IF_NO_PEER_CERT,
+            _accept_any_cert_cb)
 
         self._ssl = api.SSL_new(self._ctx)
         self._rbio = api.BIO_new(api.BIO_s_mem())
@@ -204,6 +224,28 @@ class DtlsEndpoint:
         cs = m[32:46]
         ss = m[46:60]
         return (ck, cs), (sk, ss)
+
+    def peer_fingerprint(self) -> Optional[str]:
+        """SHA-256 fingerprint of the peer's certificate, in the SDP
+        colon-separated uppercase form, or None if no cert was presented."""
+        get = getattr(api, "SSL_get1_peer_certificate",
+                      getattr(api, "SSL_get_peer_certificate", None))
+        if get is None or not self._ssl:
+            return None
+        x509 = get(self._ssl)
+        if not x509:
+            return None
+        try:
+            n = api.i2d_X509(x509, None)
+            if n <= 0:
+                return None
+            buf = ctypes.create_string_buffer(n)
+            p = ctypes.c_void_p(ctypes.addressof(buf))
+            api.i2d_X509(x509, ctypes.byref(p))
+            digest = hashlib.sha256(buf.raw[:n]).hexdigest().upper()
+            return ":".join(digest[i:i + 2] for i in range(0, len(digest), 2))
+        finally:
+            api.X509_free(x509)
 
     def close(self):
         if self._ssl:

@@ -76,6 +76,40 @@ def binding_response(request: bytes, addr, local_pwd: str) -> bytes:
     return hdr(len(attrs)) + attrs
 
 
+def binding_request(username: str, remote_pwd: str,
+                    txid: bytes = b"") -> bytes:
+    """Build an authenticated binding request (USERNAME + MESSAGE-INTEGRITY
+    + FINGERPRINT) as an ICE client would send it (RFC 8445 §7.2.2)."""
+    import secrets
+    txid = txid or secrets.token_bytes(12)
+    attrs = _attr(ATTR_USERNAME, username.encode())
+
+    def hdr(length):
+        return struct.pack(">HHI", 0x0001, length, MAGIC) + txid
+
+    mi = hmac.new(remote_pwd.encode(), hdr(len(attrs) + 24) + attrs,
+                  hashlib.sha1).digest()
+    attrs += _attr(ATTR_MESSAGE_INTEGRITY, mi)
+    crc = (zlib.crc32(hdr(len(attrs) + 8) + attrs) ^ 0x5354554E) & 0xFFFFFFFF
+    attrs += _attr(ATTR_FINGERPRINT, struct.pack(">I", crc))
+    return hdr(len(attrs)) + attrs
+
+
+def parse_username(request: bytes) -> str:
+    """The USERNAME attribute of a binding request ('' if absent)."""
+    attrs_raw = request[20:]
+    off = 0
+    while off + 4 <= len(attrs_raw):
+        atype, alen = struct.unpack_from(">HH", attrs_raw, off)
+        if atype == ATTR_USERNAME:
+            try:
+                return attrs_raw[off + 4:off + 4 + alen].decode()
+            except UnicodeDecodeError:
+                return ""
+        off += 4 + ((alen + 3) & ~3)
+    return ""
+
+
 def verify_request_integrity(request: bytes, local_pwd: str) -> bool:
     """Check the browser's binding-request MESSAGE-INTEGRITY."""
     attrs_raw = request[20:]

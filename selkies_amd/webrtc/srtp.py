@@ -86,10 +86,18 @@ class SrtpSession:
         self.rtcp_salt = srtp_kdf(master_key, master_salt, 5, 14)
         self._rtp_aes = _AesEcb(self.rtp_key)
         self._rtcp_aes = _AesEcb(self.rtcp_key)
-        # replay/ROC state per SSRC (receiver side derives ROC from seq)
+        # sender state per SSRC
         self.roc: dict[int, int] = {}
         self.last_seq: dict[int, int] = {}
         self.rtcp_index = 0
+        # receiver state per SSRC: [roc, s_l, window_mask] where bit i of
+        # window_mask marks index (roc<<16|s_l) - i as already received
+        # (RFC 3711 §3.3.2 replay list, window REPLAY_WINDOW wide)
+        self._rx: dict[int, list] = {}
+        self._rtcp_rx_high = -1
+        self._rtcp_rx_mask = 0
+
+    REPLAY_WINDOW = 64
 
     # ---- RTP ---------------------------------------------------------------
     @staticmethod
@@ -105,29 +113,62 @@ class SrtpSession:
         salt = int.from_bytes(self.rtp_salt + b"\x00\x00", "big")
         return (salt ^ (ssrc << 64) ^ (index << 16)).to_bytes(16, "big")
 
-    def _index_for(self, ssrc: int, seq: int, sender: bool) -> int:
+    def _index_for(self, ssrc: int, seq: int) -> int:
+        """Sender-side packet index (monotone local counter)."""
         roc = self.roc.get(ssrc, 0)
-        if sender:
-            last = self.last_seq.get(ssrc)
-            if last is not None and seq < last:  # wrapped
-                roc += 1
-                self.roc[ssrc] = roc
-            self.last_seq[ssrc] = seq
-            return (roc << 16) | seq
-        # receiver: pick the roc candidate closest to the last seq
         last = self.last_seq.get(ssrc)
-        if last is not None:
-            if seq < 0x2000 and last > 0xE000:
-                roc += 1
-                self.roc[ssrc] = roc
-        if last is None or seq > last or (seq < 0x2000 and last > 0xE000):
-            self.last_seq[ssrc] = seq
+        if last is not None and seq < last:  # wrapped
+            roc += 1
+            self.roc[ssrc] = roc
+        self.last_seq[ssrc] = seq
         return (roc << 16) | seq
+
+    def _rx_estimate(self, ssrc: int, seq: int):
+        """RFC 3711 Appendix A index estimation; returns (v, index) without
+        mutating state (state commits only after authentication)."""
+        st = self._rx.get(ssrc)
+        if st is None:
+            return 0, seq
+        roc, s_l, _ = st
+        if s_l < 0x8000:
+            v = roc - 1 if seq - s_l > 0x8000 and roc > 0 else roc
+        else:
+            v = roc + 1 if s_l - 0x8000 > seq else roc
+        return v, (v << 16) | seq
+
+    def _rx_replay_check(self, ssrc: int, index: int):
+        """Raise if index is outside the window or already received."""
+        st = self._rx.get(ssrc)
+        if st is None:
+            return
+        roc, s_l, mask = st
+        highest = (roc << 16) | s_l
+        if index <= highest:
+            delta = highest - index
+            if delta >= self.REPLAY_WINDOW:
+                raise ValueError("SRTP replay: index too old")
+            if (mask >> delta) & 1:
+                raise ValueError("SRTP replay: duplicate index")
+
+    def _rx_commit(self, ssrc: int, v: int, seq: int, index: int):
+        """Record a successfully authenticated index (§3.3.2 step 5)."""
+        st = self._rx.get(ssrc)
+        if st is None:
+            self._rx[ssrc] = [v, seq, 1]
+            return
+        roc, s_l, mask = st
+        highest = (roc << 16) | s_l
+        if index > highest:
+            shift = index - highest
+            mask = ((mask << shift) | 1) & ((1 << self.REPLAY_WINDOW) - 1)
+            st[0], st[1], st[2] = v, seq, mask
+        else:
+            st[2] = mask | (1 << (highest - index))
 
     def protect_rtp(self, pkt: bytes) -> bytes:
         seq = struct.unpack_from(">H", pkt, 2)[0]
         ssrc = struct.unpack_from(">I", pkt, 8)[0]
-        index = self._index_for(ssrc, seq, True)
+        index = self._index_for(ssrc, seq)
         off = self._payload_offset(pkt)
         ks = _keystream(self._rtp_aes, self._rtp_iv(ssrc, index),
                         len(pkt) - off)
@@ -143,12 +184,14 @@ class SrtpSession:
         body, tag = pkt[:-10], pkt[-10:]
         seq = struct.unpack_from(">H", body, 2)[0]
         ssrc = struct.unpack_from(">I", body, 8)[0]
-        index = self._index_for(ssrc, seq, False)
+        v, index = self._rx_estimate(ssrc, seq)
+        self._rx_replay_check(ssrc, index)
         roc = index >> 16
         expect = hmac.new(self.rtp_auth, body + struct.pack(">I", roc),
                           hashlib.sha1).digest()[:10]
         if not hmac.compare_digest(tag, expect):
             raise ValueError("SRTP auth failed")
+        self._rx_commit(ssrc, v, seq, index)
         off = self._payload_offset(body)
         ks = _keystream(self._rtp_aes, self._rtp_iv(ssrc, index),
                         len(body) - off)
@@ -180,9 +223,20 @@ class SrtpSession:
         if not hmac.compare_digest(tag, expect):
             raise ValueError("SRTCP auth failed")
         idx = struct.unpack(">I", e_index)[0]
+        index = idx & 0x7FFFFFFF
+        # SRTCP replay list (§3.3.2 applies to SRTCP via the explicit index)
+        if index <= self._rtcp_rx_high:
+            delta = self._rtcp_rx_high - index
+            if delta >= self.REPLAY_WINDOW or (self._rtcp_rx_mask >> delta) & 1:
+                raise ValueError("SRTCP replay")
+            self._rtcp_rx_mask |= 1 << delta
+        else:
+            shift = index - self._rtcp_rx_high
+            self._rtcp_rx_mask = ((self._rtcp_rx_mask << shift) | 1) \
+                & ((1 << self.REPLAY_WINDOW) - 1)
+            self._rtcp_rx_high = index
         if not idx & 0x80000000:
             return body      # unencrypted SRTCP
-        index = idx & 0x7FFFFFFF
         ssrc = struct.unpack_from(">I", body, 4)[0]
         ks = _keystream(self._rtcp_aes, self._rtcp_iv(ssrc, index),
                         len(body) - 8)

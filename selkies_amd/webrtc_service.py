@@ -26,8 +26,9 @@ logger = logging.getLogger("selkies.webrtc")
 
 
 class PeerState:
-    def __init__(self, addr):
+    def __init__(self, addr, role: str = "controller"):
         self.addr = addr
+        self.role = role
         self.dtls: Optional[dtls.DtlsEndpoint] = None
         self.srtp_out: Optional[SrtpSession] = None
         self.srtp_in: Optional[SrtpSession] = None
@@ -53,6 +54,9 @@ class WebRTCService:
         self.port = 0
         self.host_ip = ice.default_host_ip()
         self.capture: Optional[hipflux.ScreenCapture] = None
+        # certificate fingerprints signaled in accepted offers; DTLS clients
+        # whose cert does not hash to one of these are rejected (RFC 8122)
+        self.allowed_fingerprints: set[str] = set()
         self.audio_capture = None
         self._audio_enabled = False
         self._ts_base = time.monotonic()
@@ -82,6 +86,12 @@ class WebRTCService:
     # ---- signaling -----------------------------------------------------------
     def handle_offer(self, offer_sdp: str) -> str:
         offer = sdp.parse_offer(offer_sdp)
+        for fp in ([offer.session_fingerprint]
+                   + [m.fingerprint for m in offer.media]):
+            if fp:
+                # drop the "sha-256" hash-algo prefix, keep hex only
+                digest = fp.split()[-1].upper().replace(":", "")
+                self.allowed_fingerprints.add(digest)
         video = next((m for m in offer.media if m.kind == "video"), None)
         if video is not None and video.h264_pts:
             self.packetizer.pt = video.h264_pts[0][0]
@@ -192,11 +202,23 @@ class WebRTCService:
     # ---- datagram demux -------------------------------------------------------
     def on_datagram(self, data: bytes, addr):
         if ice.is_stun(data):
+            # ICE gate: the binding request must carry OUR
+            # ufrag in USERNAME and a MESSAGE-INTEGRITY keyed with OUR
+            # password (RFC 8445 §7.3) before the sender becomes a peer.
+            username = ice.parse_username(data)
+            if not username.startswith(self.ufrag + ":"):
+                return
+            if not ice.verify_request_integrity(data, self.pwd):
+                return
             resp = ice.binding_response(data, addr, self.pwd)
             self.transport.sendto(resp, addr)
             if addr not in self.peers:
-                self.peers[addr] = PeerState(addr)
-                logger.info("webrtc peer candidate %s", addr)
+                role = ("controller"
+                        if not any(p.role == "controller"
+                                   for p in self.peers.values())
+                        else "viewer")
+                self.peers[addr] = PeerState(addr, role)
+                logger.info("webrtc peer candidate %s (%s)", addr, role)
             return
         peer = self.peers.get(addr)
         if peer is None:
@@ -211,6 +233,17 @@ class WebRTCService:
             if peer.dtls.handshake_done:
                 self._pump_sctp(peer)
             if peer.dtls.handshake_done and not peer.connected:
+                # bind the DTLS client to the signaled offer: its cert must
+                # hash to an a=fingerprint we answered (RFC 8122)
+                fp = peer.dtls.peer_fingerprint()
+                if (fp is None or
+                        fp.replace(":", "") not in self.allowed_fingerprints):
+                    logger.warning(
+                        "webrtc peer %s rejected: certificate fingerprint "
+                        "not in any signaled offer", addr)
+                    peer.dtls.close()
+                    self.peers.pop(addr, None)
+                    return
                 (ck, cs), (sk, ss) = peer.dtls.export_srtp_keys()
                 # we are the DTLS server: send with server keys,
                 # receive with client keys
@@ -270,6 +303,12 @@ class WebRTCService:
         try:
             text = data.decode("utf-8")
         except UnicodeDecodeError:
+            return
+        # same role policy as the WebSocket path (streaming._on_text):
+        # only the controller — or anyone when sharing is enabled — may
+        # inject input (the DC bypassed this gate).
+        if (peer.role != "controller"
+                and not getattr(self.settings, "enable_shared", False)):
             return
         try:
             self.streaming.input.on_message(text)

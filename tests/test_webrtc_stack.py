@@ -181,15 +181,24 @@ def test_webrtc_loopback_end_to_end():
         runner, port = await start_on_free_port(server)
         loop = asyncio.get_running_loop()
         try:
+            # a conformant client: its certificate fingerprint rides in the
+            # offer, and binding requests are authenticated (the server now
+            # rejects both anonymous STUN and unsignaled DTLS certs)
+            cert = dtls.Certificate()
+            offer_sdp = BROWSER_OFFER.replace(
+                "AA:BB:CC:DD:EE:FF:00:11:22:33:44:55:66:77:88:99:"
+                "AA:BB:CC:DD:EE:FF:00:11:22:33:44:55:66:77:88:99",
+                cert.fingerprint)
             async with aiohttp.ClientSession() as sess:
                 r = await sess.post(
                     f"http://127.0.0.1:{port}/api/webrtc/offer",
-                    json={"sdp": BROWSER_OFFER})
+                    json={"sdp": offer_sdp})
                 assert r.status == 200, await r.text()
                 answer = (await r.json())["sdp"]
             udp_port = int(answer.split("m=video ")[1].split()[0])
             ans = sdp.parse_offer(answer)   # reuse parser for answer fields
             server_pwd = ans.media[0].ice_pwd
+            server_ufrag = ans.media[0].ice_ufrag
 
             # --- our 'browser': UDP socket + STUN + DTLS client
             recv_q = asyncio.Queue()
@@ -201,13 +210,15 @@ def test_webrtc_loopback_end_to_end():
             transport, _ = await loop.create_datagram_endpoint(
                 Cli, remote_addr=("127.0.0.1", udp_port))
 
-            txid = secrets.token_bytes(12)
-            req = struct.pack(">HHI", 0x0001, 0, ice.MAGIC) + txid
+            # anonymous STUN must be ignored now
+            bare = struct.pack(">HHI", 0x0001, 0, ice.MAGIC) \
+                + secrets.token_bytes(12)
+            transport.sendto(bare)
+            req = ice.binding_request(f"{server_ufrag}:abcd", server_pwd)
             transport.sendto(req)
             resp = await asyncio.wait_for(recv_q.get(), 5)
             assert ice.is_stun(resp)
 
-            cert = dtls.Certificate()
             cli = dtls.DtlsEndpoint(cert, server=False)
             cli.start()
             for d in cli.take_datagrams():
