@@ -3,6 +3,7 @@
 #pragma once
 
 #include <cassert>
+#include <cstddef>
 #include <cstdint>
 #include <vector>
 
@@ -78,7 +79,7 @@ class BitWriter {
 
  private:
   std::vector<uint8_t> buf_;
-  uint32_t acc_ = 0;
+  uint64_t acc_ = 0;
   int nacc_ = 0;
 };
 

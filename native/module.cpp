@@ -9,6 +9,8 @@
 #include "cpu/damage.h"
 #include "cpu/h264/cavlc.h"
 #include "cpu/h264/encoder.h"
+#include "cpu/hevc/cabac.h"
+#include "cpu/hevc/encoder.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
 
@@ -246,6 +248,90 @@ PYBIND11_MODULE(_native, m) {
              return py::make_tuple(y, cb, cr, self.enc.recon_ypitch(),
                                    self.enc.recon_cpitch());
            });
+
+  // ---- HEVC CPU reference encoder (BASELINE config 3 path) --------------
+  struct PyHevc {
+    hevc::StripeEncoder enc;
+    int w, h, ypitch, cpitch;
+    std::vector<uint8_t> yuv;
+    PyHevc(int width, int height, int slices_per_row)
+        : enc(width, height, slices_per_row), w(width), h(height) {
+      ypitch = (width + 15) & ~15;
+      cpitch = ypitch / 2;
+      int yh = (height + 15) & ~15;
+      yuv.resize(static_cast<size_t>(ypitch) * yh * 3 / 2);
+    }
+  };
+  py::class_<PyHevc>(m, "HevcEncoder")
+      .def(py::init<int, int, int>(), py::arg("width"), py::arg("height"),
+           py::arg("slices_per_row") = 1)
+      .def(
+          "encode",
+          [](PyHevc& self, py::buffer bgrx, int qp) {
+            py::buffer_info info = bgrx.request();
+            if (info.size < static_cast<ssize_t>(self.w) * self.h * 4)
+              throw std::runtime_error("buffer too small");
+            std::vector<uint8_t> out;
+            hevc::EncodeStats st;
+            {
+              py::gil_scoped_release rel;
+              uint8_t* y = self.yuv.data();
+              int yh = (self.h + 15) & ~15;
+              uint8_t* cb = y + static_cast<size_t>(self.ypitch) * yh;
+              uint8_t* cr = cb + static_cast<size_t>(self.cpitch) * (yh / 2);
+              h264::bgrx_to_yuv420(static_cast<const uint8_t*>(info.ptr),
+                                   self.w * 4, self.w, self.h, y, self.ypitch,
+                                   cb, cr, self.cpitch);
+              self.enc.encode_frame(y, self.ypitch, cb, cr, self.cpitch, qp,
+                                    out, &st);
+            }
+            py::dict d;
+            d["data"] = py::bytes(reinterpret_cast<const char*>(out.data()),
+                                  out.size());
+            d["qp"] = st.frame_qp;
+            d["ctus"] = st.ctu_count;
+            return d;
+          },
+          py::arg("bgrx"), py::arg("qp") = 30)
+      .def("recon", [](PyHevc& self) {
+        int yh = (self.h + 15) & ~15;
+        int yp = self.enc.recon_ypitch(), cp = self.enc.recon_cpitch();
+        py::bytes y(reinterpret_cast<const char*>(self.enc.recon_y()),
+                    static_cast<size_t>(yp) * yh);
+        py::bytes cb(reinterpret_cast<const char*>(self.enc.recon_cb()),
+                     static_cast<size_t>(cp) * yh / 2);
+        py::bytes cr(reinterpret_cast<const char*>(self.enc.recon_cr()),
+                     static_cast<size_t>(cp) * yh / 2);
+        return py::make_tuple(y, cb, cr, yp, cp);
+      });
+
+  // CABAC pair-fuzz hook: encode an explicit (kind, ctx, bin) op sequence
+  // with the I-slice context bank; tests decode it back with the from-spec
+  // Python CABAC decoder. kind: 0 = context bin, 1 = bypass, 2 = terminate.
+  m.def(
+      "_hevc_cabac_encode",
+      [](const std::vector<std::tuple<int, int, int>>& ops, int qp) {
+        std::vector<uint8_t> bytes;
+        hevc::CabacEncoder cab(bytes);
+        hevc::ContextBank bank;
+        bank.init(qp);
+        for (const auto& op : ops) {
+          int kind = std::get<0>(op), ctx = std::get<1>(op),
+              bin = std::get<2>(op);
+          if (kind == 0)
+            cab.encode_bin(bank.ctx[ctx % hevc::kNumContexts], bin);
+          else if (kind == 1)
+            cab.encode_bypass(bin);
+          else
+            cab.encode_terminate(bin);
+        }
+        auto tail = cab.finish();
+        return py::make_tuple(
+            py::bytes(reinterpret_cast<const char*>(bytes.data()),
+                      bytes.size()),
+            tail.bits, tail.nbits);
+      },
+      py::arg("ops"), py::arg("qp") = 30);
 
   m.def(
       "bgrx_to_yuv420",
