@@ -114,9 +114,9 @@ struct Watermark {
 }  // namespace
 
 void pack_wire_stripe(const EncodedStripe& s, std::vector<uint8_t>& out) {
-  if (s.type == StripeType::kH264) {
-    // [0x04, keyflag, frame_id:u16be, y:u16be, w:u16be, h:u16be] (10 bytes)
-    out.push_back(0x04);
+  if (s.type == StripeType::kH264 || s.type == StripeType::kHevc) {
+    // [tag, keyflag, frame_id:u16be, y:u16be, w:u16be, h:u16be] (10 bytes)
+    out.push_back(static_cast<uint8_t>(s.type));
     out.push_back(s.is_keyframe ? 1 : 0);
     out.push_back(static_cast<uint8_t>((s.frame_id >> 8) & 0xFF));
     out.push_back(static_cast<uint8_t>(s.frame_id & 0xFF));
@@ -172,8 +172,9 @@ void ScreenCapture::run() {
   if (!settings_.use_cpu && settings_.gpu_id >= 0)
     pipeline = make_hip_pipeline(settings_);
   if (!pipeline) {
-    pipeline = settings_.output_mode == 0 ? make_cpu_jpeg_pipeline(settings_)
-                                          : make_cpu_h264_pipeline(settings_);
+    pipeline = settings_.output_mode == 0   ? make_cpu_jpeg_pipeline(settings_)
+               : settings_.output_mode == 2 ? make_cpu_hevc_pipeline(settings_)
+                                            : make_cpu_h264_pipeline(settings_);
   }
   if (!pipeline) {
     std::fprintf(stderr,
@@ -324,7 +325,7 @@ void ScreenCapture::run() {
     ctx.paintover = paintover_frame;
     ctx.jpeg_quality = paintover_frame ? settings_.jpeg_paintover_quality
                                        : jpeg_quality_.load();
-    if (settings_.video_cbr_mode && settings_.output_mode == 1) {
+    if (settings_.video_cbr_mode && settings_.output_mode >= 1) {
       ctx.crf = static_cast<int>(rc_qp + 0.5);
       if (paintover_frame)
         ctx.crf = std::min(ctx.crf, settings_.video_paintover_crf);
@@ -352,7 +353,9 @@ void ScreenCapture::run() {
       if (!f) {
         std::string path = settings_.recording_path + ".s" +
                            std::to_string(s.y) +
-                           (s.type == StripeType::kH264 ? ".h264" : ".mjpeg");
+                           (s.type == StripeType::kH264   ? ".h264"
+                            : s.type == StripeType::kHevc ? ".h265"
+                                                          : ".mjpeg");
         f = std::fopen(path.c_str(), "wb");
       }
       if (f) std::fwrite(s.data, 1, s.size, f);
@@ -390,7 +393,7 @@ void ScreenCapture::run() {
       frames_encoded_.fetch_add(1);
       ++frame_id;
 
-      if (settings_.video_cbr_mode && settings_.output_mode == 1) {
+      if (settings_.video_cbr_mode && settings_.output_mode >= 1) {
         double target_bpf =
             bitrate_kbps_.load() * 1000.0 / 8.0 / std::max(1.0, fps);
         // VBV buffer: vbv_multiplier x ~quarter-second of stream

@@ -69,6 +69,7 @@ class EncodePipeline {
 
 std::unique_ptr<EncodePipeline> make_cpu_jpeg_pipeline(const CaptureSettings&);
 std::unique_ptr<EncodePipeline> make_cpu_h264_pipeline(const CaptureSettings&);
+std::unique_ptr<EncodePipeline> make_cpu_hevc_pipeline(const CaptureSettings&);
 // GPU (HIP/gfx950) pipeline; returns nullptr if no usable HIP device.
 std::unique_ptr<EncodePipeline> make_hip_pipeline(const CaptureSettings&);
 

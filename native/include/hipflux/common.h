@@ -10,10 +10,11 @@
 
 namespace hipflux {
 
-// Output frame/stripe type tags (wire protocol).
-enum class StripeType : uint8_t { kJpeg = 0x03, kH264 = 0x04 };
+// Output frame/stripe type tags (wire protocol; 0x06 is
+// this framework's HEVC extension — the reference has no HEVC encoder).
+enum class StripeType : uint8_t { kJpeg = 0x03, kH264 = 0x04, kHevc = 0x06 };
 
-enum class OutputMode : int { kJpeg = 0, kH264 = 1 };
+enum class OutputMode : int { kJpeg = 0, kH264 = 1, kHevc = 2 };
 
 // Mirrors the pixelflux CaptureSettings contract consumed by the reference
 // control plane (reference display_utils.py:2250-2354, selkies.py:5364-5426;
@@ -26,7 +27,7 @@ struct CaptureSettings {
   double target_fps = 60.0;
   bool capture_cursor = false;
 
-  int output_mode = 1;              // 0 = JPEG, 1 = H.264
+  int output_mode = 1;              // 0 = JPEG, 1 = H.264, 2 = HEVC
   int capture_scale_div = 1;        // integer box-downscale (1..4): encode
                                     // at capture/div (e.g. 4K capture -> 2
                                     // -> 1080p stream)

@@ -10,6 +10,8 @@ server → client binary frames
   0x04  H.264    [0x04, keyflag:u8, frame_id:u16be, y:u16be, w:u16be,
                   h:u16be] + AnnexB NALs               (10-byte header)
   0x05  gzip'd control text (whole payload after the tag is gzip)
+  0x06  HEVC     same 10-byte header layout as 0x04, AnnexB H.265 NALs
+                 (this framework's extension; the reference has no HEVC)
 
 client → server binary frames
   0x02  mic PCM s16le
@@ -32,6 +34,7 @@ TAG_MIC_PCM = 0x02
 TAG_JPEG = 0x03
 TAG_H264 = 0x04
 TAG_GZIP = 0x05
+TAG_HEVC = 0x06
 
 H264_HEADER = struct.Struct(">BBHHHH")   # tag, keyflag, frame_id, y, w, h
 JPEG_HEADER = struct.Struct(">BBHH")     # tag, flags, frame_id, y

@@ -71,7 +71,7 @@ def _parse_bool(raw: str) -> bool:
 # Ours: h264enc (full-frame H.264), h264enc-striped (stripe-parallel
 # independent bitstreams), jpeg (striped MJPEG). All three run on the
 # hipflux HIP path on MI355X with a CPU fallback of the same bitstream.
-VIDEO_ENCODERS = ("h264enc", "h264enc-striped", "jpeg")
+VIDEO_ENCODERS = ("h264enc", "h264enc-striped", "hevcenc-striped", "jpeg")
 
 SETTING_DEFINITIONS: list[SettingDef] = [
     # ---- process / transport ----

@@ -141,7 +141,8 @@ class StreamingService:
         cs.capture_width = w
         cs.capture_height = h
         cs.target_fps = float(s.framerate)
-        cs.output_mode = 0 if s.encoder == "jpeg" else 1
+        cs.output_mode = (0 if s.encoder == "jpeg"
+                          else 2 if "hevc" in s.encoder else 1)
         cs.use_cpu = bool(s.use_cpu)
         # per-session GPU placement: explicit session_gpus list round-robins
         # displays/sessions over devices (SURVEY.md §5.8 multi-GPU facility)

@@ -3,7 +3,8 @@
  * Speaks the binary stripe wire protocol (SURVEY.md §3.2):
  *   0x03 JPEG stripe  [tag, flags, frame_id u16, y u16] + JFIF
  *   0x04 H.264 stripe [tag, key, frame_id u16, y u16, w u16, h u16] + AnnexB
- * H.264 stripes feed one WebCodecs VideoDecoder PER STRIPE ROW (each row is
+ *   0x06 HEVC stripe  same header as 0x04, H.265 AnnexB
+ * H.264/HEVC stripes feed one WebCodecs VideoDecoder PER STRIPE ROW (each row is
  * an independent bitstream); decoded rows composite onto the canvas.
  * JPEG stripes decode via createImageBitmap. Input is captured and sent as
  * the text verbs the server's InputDispatcher understands.
@@ -23,8 +24,9 @@ let lastAckedFrame = -1;
 /* ---------------- decode sinks ---------------- */
 
 class H264Row {
-  constructor(y) {
+  constructor(y, codec) {
     this.y = y;
+    this.codec = codec || "avc1.42e028";
     this.decoder = null;
     this.width = 0;
     this.height = 0;
@@ -42,7 +44,7 @@ class H264Row {
                       this.decoder = null; requestIdr(); },
     });
     this.decoder.configure({
-      codec: "avc1.42e028",
+      codec: this.codec,
       optimizeForLatency: true,
     });
   }
@@ -70,15 +72,19 @@ function onBinary(buf) {
   const d = new Uint8Array(buf);
   byteCount += d.length;
   const tag = d[0];
-  if (tag === 0x04) {
+  if (tag === 0x04 || tag === 0x06) {
     const key = d[1] === 1;
     const frameId = (d[2] << 8) | d[3];
     const y = (d[4] << 8) | d[5];
     const w = (d[6] << 8) | d[7];
     const h = (d[8] << 8) | d[9];
     ensureCanvas(w, y + h);
+    // 0x06 = HEVC Main (WebCodecs hev1 codec string); per-row decoders
+    const codec = tag === 0x06 ? "hev1.1.6.L123.B0" : "avc1.42e028";
     let row = h264Rows.get(y);
-    if (!row) { row = new H264Row(y); h264Rows.set(y, row); }
+    if (!row || row.codec !== codec) {
+      row = new H264Row(y, codec); h264Rows.set(y, row);
+    }
     if (key) row.configure(w, h);
     row.push(d.subarray(10), key, frameId);
     noteFrame(frameId);

@@ -238,3 +238,28 @@ def test_capture_watchdog_rebuilds_dead_capture():
         assert alive, "watchdog did not rebuild the dead capture"
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_hevc_mode_stripes_decode():
+    """output_mode 2 emits 0x06 HEVC stripes; each stripe is an IDR whose
+    bitstream the from-spec HEVC decoder reconstructs."""
+    from hevc_ref_decoder import Decoder
+    s = make_settings(output_mode=2, video_fullframe=True, video_crf=30)
+    col, cap = run_capture(s, 0.4)
+    assert cap.pipeline == "cpu-hevc"
+    assert col.count() > 0
+    seen_rows = set()
+    for st in col.stripes:
+        data = bytes(st["data"])
+        assert data[0] == 0x06
+        assert st["key"]            # all-intra: every stripe is an IDR
+        key, fid, y, w, h = data[1] == 1, (data[2] << 8) | data[3], \
+            (data[4] << 8) | data[5], (data[6] << 8) | data[7], \
+            (data[8] << 8) | data[9]
+        assert w == 320
+        if y in seen_rows:
+            continue
+        seen_rows.add(y)
+        frames = Decoder().decode(data[10:])
+        assert frames and frames[0][0].shape == (h, w)
+    assert seen_rows == {0, 64, 128}
