@@ -49,6 +49,12 @@ class BitWriter {
 
   void put_bit(int b) { u(b & 1, 1); }
 
+  // Bulk-append whole bytes (writer must be byte-aligned).
+  void append_bytes(const uint8_t* data, size_t n) {
+    assert(nacc_ == 0 && "append_bytes requires byte alignment");
+    buf_.insert(buf_.end(), data, data + n);
+  }
+
   void rbsp_trailing() {
     put_bit(1);
     if (nacc_ != 0) u(0, 8 - nacc_);

@@ -29,8 +29,8 @@ class CpuHevcPipeline : public EncodePipeline {
       for (size_t i = 0; i < n_stripes; ++i) {
         int y0 = static_cast<int>(i) * stripe_h;
         int hgt = std::min(stripe_h, frame.height - y0);
-        encoders_.push_back(
-            std::make_unique<hevc::StripeEncoder>(frame.width, hgt));
+        encoders_.push_back(std::make_unique<hevc::StripeEncoder>(
+            frame.width, hgt, hevc::default_slices_per_row(frame.width)));
       }
       ypitch_ = (frame.width + 15) & ~15;
       cpitch_ = ypitch_ / 2;

@@ -1060,12 +1060,15 @@ class HipH264Pipeline : public EncodePipeline {
 
 }  // namespace
 
+std::unique_ptr<EncodePipeline> make_hip_hevc_pipeline(const CaptureSettings&);
+
 std::unique_ptr<EncodePipeline> make_hip_pipeline(const CaptureSettings& s) {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess || n == 0) return nullptr;
   if (s.gpu_id >= n) return nullptr;
   try {
     if (s.output_mode == 0) return std::make_unique<HipJpegPipeline>(s);
+    if (s.output_mode == 2) return make_hip_hevc_pipeline(s);
     return std::make_unique<HipH264Pipeline>(s);
   } catch (const std::exception& e) {
     std::fprintf(stderr, "hipflux: HIP pipeline init failed: %s\n", e.what());
