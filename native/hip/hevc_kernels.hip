@@ -158,14 +158,14 @@ __global__ __launch_bounds__(256) void k_hevc_rows(
     pm[1] = lty;
     // Vertical: top is l0 everywhere; column-0 edge adjust
     pm[2] = (tx == 0) ? clip8d(l0 + ((lty - l0) >> 1)) : l0;
-    // Planar on filtered refs: top' = topright' = l0' = lfY[0],
-    // below-left' = lfY[16]
+    // Planar on filtered refs. The [1 2 1] filter runs over the LINEAR
+    // reference array, where the replicated top/corner/below-left arms
+    // are flat — their filtered values stay the RAW l0 / l15; only the
+    // real left column smooths.
     {
       int lf = has_left ? sh.lfY[ty] : 128;
-      int t0 = has_left ? sh.lfY[0] : 128;
-      int bl = has_left ? sh.lfY[16] : 128;
-      pm[3] = ((15 - tx) * lf + (tx + 1) * t0 + (15 - ty) * t0 +
-               (ty + 1) * bl + 16) >> 5;
+      pm[3] = ((15 - tx) * lf + (tx + 1) * l0 + (15 - ty) * l0 +
+               (ty + 1) * l15 + 16) >> 5;
     }
     // wave-reduce SADs, then pick first strict-min in order DC,H,V,Planar
     for (int m = 0; m < 4; ++m) {

@@ -20,6 +20,11 @@ using std::min;
 namespace hipflux {
 
 // BT.601 full-range (JFIF) coefficients — match cpu/jpeg_enc.cpp exactly.
+// fp contract OFF: the host reference compiles without FMA (x86-64
+// baseline), so fused mul-adds here would produce rare ±1 rounding
+// differences on noise content and break the HEVC/H.264 byte-equality
+// contract between the CPU and GPU pipelines.
+#pragma clang fp contract(off)
 __device__ inline float dev_y(float r, float g, float b) {
   return 0.299f * r + 0.587f * g + 0.114f * b;
 }
