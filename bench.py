@@ -78,7 +78,8 @@ def main():
     from hipflux import _native
 
     kind = "gpu" if use_gpu else "cpu"
-    out_mode = 0 if args.encoder == "jpeg" else 1
+    out_mode = (0 if args.encoder == "jpeg"
+                else 2 if "hevc" in args.encoder else 1)
     tile = args.mode == "tile" and world > 1
     if tile:
         # strong scaling: rank encodes rows [band0, band1) of the SAME
@@ -204,6 +205,7 @@ def main():
     if rank == 0:
         result = {
             "metric": ("encoded_fps_1080p60_h264" if out_mode == 1
+                       else "encoded_fps_4k60_hevc" if out_mode == 2
                        else "encoded_fps_1080p60_jpeg"),
             "value": round(fps_job, 2),
             "unit": "frames/s",
@@ -219,8 +221,8 @@ def main():
             "dtype": "uint8",
             "data": "synthetic",
             "config": {
-                "model": "hipflux h264enc-striped (HIP gfx950)"
-                         if kind == "gpu" else "hipflux h264enc-striped (CPU)",
+                "model": f"hipflux {args.encoder} "
+                         + ("(HIP gfx950)" if kind == "gpu" else "(CPU)"),
                 "global_batch": world,
                 "seq_len": args.width * args.height,
                 "parallelism": (f"tile{world}" if tile

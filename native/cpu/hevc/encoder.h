@@ -36,7 +36,7 @@ struct EncodeStats {
 // left-neighbor chain (GPU wavefront) and multiply CABAC parallelism, at
 // ~0.1-0.5% bitrate cost per extra segment boundary.
 inline int default_slices_per_row(int width) {
-  return width >= 2560 ? 4 : width >= 1280 ? 2 : 1;
+  return width >= 3840 ? 16 : width >= 2560 ? 4 : width >= 1280 ? 2 : 1;
 }
 
 class StripeEncoder {

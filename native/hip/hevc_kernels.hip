@@ -473,6 +473,7 @@ struct DevCabac {
   uint32_t low, range, buffered;
   int bits_left, num_buffered, count;
   uint8_t* out;
+  const uint32_t* ptab;  // packed per-(state,q) entry, see k_hevc_cabac
 
   __device__ void init(uint8_t* o) {
     low = 0;
@@ -505,22 +506,23 @@ struct DevCabac {
       buffered = lead & 0xFF;
     }
   }
-  __device__ void bin(uint8_t* st, int b) {
-    int state = *st >> 1, mps = *st & 1;
-    uint32_t lps = cRangeTabLps[state][(range >> 6) & 3];
+  // packed entry: lps | transIdxLps<<8 | transIdxMps<<14 | renorm<<20
+  __device__ void bin(uint32_t* st, int b) {
+    const int state = (int)(*st >> 1);
+    int mps = (int)(*st & 1);
+    const uint32_t e = ptab[(state << 2) | ((range >> 6) & 3)];
+    const uint32_t lps = e & 0xFF;
     range -= lps;
     if (b != mps) {
-      int n = cRenorm[lps >> 3];
+      const int n = (e >> 20) & 7;
       low = (low + range) << n;
       range = lps << n;
       if (state == 0) mps ^= 1;
-      state = cTransIdxLps[state];
+      *st = ((((e >> 8) & 63)) << 1) | (uint32_t)mps;
       bits_left -= n;
-      *st = (uint8_t)((state << 1) | mps);
       test_write();
     } else {
-      state = state < 62 ? state + 1 : state;
-      *st = (uint8_t)((state << 1) | mps);
+      *st = (((e >> 14) & 63) << 1) | (uint32_t)mps;
       if (range >= 256) return;
       low <<= 1;
       range <<= 1;
@@ -534,8 +536,26 @@ struct DevCabac {
     --bits_left;
     test_write();
   }
+  // n bypass bins at once: low' = low*2^n + range*value (associativity of
+  // the per-bit recurrence low = 2*low + b*range). Chunked so bits_left
+  // stays positive between flushes; byte-identical to bit-at-a-time.
   __device__ void bypass_bins(uint32_t v, int n) {
-    for (int i = n - 1; i >= 0; --i) bypass((v >> i) & 1);
+    while (n > 0) {
+      const int take = n < 10 ? n : 10;
+      n -= take;
+      low = (low << take) + range * ((v >> n) & ((1u << take) - 1));
+      bits_left -= take;
+      test_write();
+    }
+  }
+  // a run of `ones` 1-bins followed by one 0-bin (unary prefix)
+  __device__ void bypass_unary(int ones) {
+    while (ones > 0) {
+      const int take = ones < 10 ? ones : 10;
+      ones -= take;
+      bypass_bins((1u << take) - 1, take);
+    }
+    bypass(0);
   }
   __device__ void terminate(int b) {
     range -= 2;
@@ -578,28 +598,71 @@ struct DevCabac {
   }
 };
 
-// transliteration of entropy.h code_residual<N>
+// Transliteration of entropy.h code_residual<N>, restructured for the
+// GPU: contexts live in LDS (lane-strided words, no scratch), the
+// sub-block significance map is a register bitmask with constant
+// neighbor tables, and the in-sub-block scans are fully unrolled so
+// every level access has a static offset (the 16 loads of a sub-block
+// issue together and pipeline instead of serializing).
+
+// scan-index of the right / below neighbor sub-block (-1 = none)
+__constant__ int8_t cSbRight4[16] = {2, 4, 5, 7, 8, 9, 10, 11,
+                                     12, -1, 13, 14, -1, 15, -1, -1};
+__constant__ int8_t cSbBelow4[16] = {1, 3, 4, 6, 7, 8, -1, 10,
+                                     11, 12, -1, 13, 14, -1, 15, -1};
+__constant__ int8_t cSbRight2[4] = {2, 3, -1, -1};
+__constant__ int8_t cSbBelow2[4] = {1, -1, 3, -1};
+// sig_coeff_flag base context by prevCsbf pattern and scan position
+// (§9.3.4.2.5; xp/yp are the static in-sub-block coordinates)
+__constant__ uint8_t cSigPat[4][16] = {
+    // prev 0: (xp+yp==0)?2:(xp+yp<3)?1:0
+    {2, 1, 1, 1, 1, 1, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0},
+    // prev 1: yp==0?2:yp==1?1:0
+    {2, 1, 2, 0, 1, 2, 0, 0, 1, 2, 0, 0, 1, 0, 0, 0},
+    // prev 2: xp==0?2:xp==1?1:0
+    {2, 2, 1, 2, 1, 0, 2, 1, 0, 0, 1, 0, 0, 0, 0, 0},
+    // prev 3
+    {2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2, 2}};
+
+// per-block LDS context bank (one slice segment per block; only lane 0
+// runs the bin-serial coder, so jobs never serialize each other through
+// wave lockstep — that costs ~64x with one job per LANE)
+#define CTX_LDS(i) ctx_lds[(i)]
+
 template <int N>
-__device__ void dev_code_residual(DevCabac& cab, uint8_t* ctx,
-                                  const int16_t* level, int cidx) {
+__device__ void dev_code_residual(DevCabac& cab, uint32_t* ctx_lds, int lane,
+                                  const int16_t* __restrict__ level,
+                                  int cidx) {
   const int log2n = N == 8 ? 3 : 4;
-  const int n_sb = (N / 4) * (N / 4);
+  constexpr int kSbW = N / 4;
+  constexpr int n_sb = kSbW * kSbW;
+  const int8_t* sb_right = (N == 8) ? cSbRight2 : cSbRight4;
+  const int8_t* sb_below = (N == 8) ? cSbBelow2 : cSbBelow4;
   const uint8_t* sb_scan = (N == 8) ? cScan2 : cScan4;
 
-  int last_sb = -1, last_pos = -1, last_x = 0, last_y = 0;
-  for (int i = n_sb - 1; i >= 0 && last_sb < 0; --i) {
-    int sx = (sb_scan[i] & 15) * 4, sy = (sb_scan[i] >> 4) * 4;
-    for (int n = 15; n >= 0; --n) {
-      int x = sx + (cScan4[n] & 15), y = sy + (cScan4[n] >> 4);
-      if (level[y * N + x]) {
-        last_sb = i;
-        last_pos = n;
-        last_x = x;
-        last_y = y;
-        break;
-      }
+  // ---- pass 1: sub-block significance map + last position.
+  // Each sub-block's 16 levels load with STATIC offsets (unrolled scan),
+  // so they live in registers — no per-lane scratch arrays.
+  uint32_t csbf_mask = 0;
+  int last_sb = -1, last_pos = -1;
+  for (int i = 0; i < n_sb; ++i) {
+    const int sx = (sb_scan[i] & 15) * 4, sy = (sb_scan[i] >> 4) * 4;
+    const int16_t* base = level + sy * N + sx;
+    uint32_t nzm = 0;
+#pragma unroll
+    for (int n = 0; n < 16; ++n) {
+      int16_t v = base[(cScan4[n] >> 4) * N + (cScan4[n] & 15)];
+      nzm |= (v != 0 ? 1u : 0u) << n;
+    }
+    if (nzm) {
+      csbf_mask |= 1u << i;
+      last_sb = i;
+      last_pos = 31 - __builtin_clz(nzm);
     }
   }
+  const int last_x = (sb_scan[last_sb] & 15) * 4 + (cScan4[last_pos] & 15);
+  const int last_y = (sb_scan[last_sb] >> 4) * 4 + (cScan4[last_pos] >> 4);
+
   int ctx_off, ctx_shift;
   if (cidx == 0) {
     ctx_off = 3 * (log2n - 2) + ((log2n - 1) >> 2);
@@ -621,13 +684,13 @@ __device__ void dev_code_residual(DevCabac& cab, uint8_t* ctx,
   };
   int px = group_idx(last_x), py = group_idx(last_y);
   for (int b = 0; b < (px < g_max ? px : g_max); ++b)
-    cab.bin(&ctx[HG_CTX_LASTX + ctx_off + (b >> ctx_shift)], 1);
+    cab.bin(&CTX_LDS(HG_CTX_LASTX + ctx_off + (b >> ctx_shift)), 1);
   if (px < g_max)
-    cab.bin(&ctx[HG_CTX_LASTX + ctx_off + (px >> ctx_shift)], 0);
+    cab.bin(&CTX_LDS(HG_CTX_LASTX + ctx_off + (px >> ctx_shift)), 0);
   for (int b = 0; b < (py < g_max ? py : g_max); ++b)
-    cab.bin(&ctx[HG_CTX_LASTY + ctx_off + (b >> ctx_shift)], 1);
+    cab.bin(&CTX_LDS(HG_CTX_LASTY + ctx_off + (b >> ctx_shift)), 1);
   if (py < g_max)
-    cab.bin(&ctx[HG_CTX_LASTY + ctx_off + (py >> ctx_shift)], 0);
+    cab.bin(&CTX_LDS(HG_CTX_LASTY + ctx_off + (py >> ctx_shift)), 0);
   if (px > 3) {
     int nbits = (px >> 1) - 1;
     cab.bypass_bins(last_x - ((2 + (px & 1)) << nbits), nbits);
@@ -637,121 +700,107 @@ __device__ void dev_code_residual(DevCabac& cab, uint8_t* ctx,
     cab.bypass_bins(last_y - ((2 + (py & 1)) << nbits), nbits);
   }
 
-  bool csbf[16];
-  for (int i = 0; i < n_sb; ++i) csbf[i] = false;
-  for (int i = 0; i <= last_sb; ++i) {
-    int sx = (sb_scan[i] & 15) * 4, sy = (sb_scan[i] >> 4) * 4;
-    for (int n = 0; n < 16; ++n) {
-      int x = sx + (cScan4[n] & 15), y = sy + (cScan4[n] >> 4);
-      if (level[y * N + x]) {
-        csbf[i] = true;
-        break;
-      }
-    }
-  }
-
   int prev_g1_zero = -1;
   for (int i = last_sb; i >= 0; --i) {
-    const int sbx = sb_scan[i] & 15, sby = sb_scan[i] >> 4;
-    const int sx = sbx * 4, sy = sby * 4;
+    const bool sb_nz = (csbf_mask >> i) & 1;
     bool coded_explicit = false;
+    // prevCsbf pattern from the constant neighbor tables
+    const int ri = sb_right[i], bi = sb_below[i];
+    const int right = ri >= 0 ? ((csbf_mask >> ri) & 1) : 0;
+    const int below = bi >= 0 ? ((csbf_mask >> bi) & 1) : 0;
     if (i < last_sb && i > 0) {
-      int right = 0, below = 0;
-      for (int j = 0; j < n_sb; ++j) {
-        if ((sb_scan[j] & 15) == sbx + 1 && (sb_scan[j] >> 4) == sby)
-          right = csbf[j];
-        if ((sb_scan[j] & 15) == sbx && (sb_scan[j] >> 4) == sby + 1)
-          below = csbf[j];
-      }
       int c = (right + below > 0 ? 1 : 0) + (cidx ? 2 : 0);
-      cab.bin(&ctx[HG_CTX_CSBF + c], csbf[i] ? 1 : 0);
+      cab.bin(&CTX_LDS(HG_CTX_CSBF + c), sb_nz ? 1 : 0);
       coded_explicit = true;
     }
-    if (!csbf[i] && i != last_sb && i != 0) continue;
+    if (!sb_nz && i != last_sb && i != 0) continue;
     bool infer_dc = coded_explicit;
 
-    int sig_pos[16], n_sig = 0;
-    int start = (i == last_sb) ? last_pos - 1 : 15;
-    if (i == last_sb) sig_pos[n_sig++] = last_pos;
-    for (int n = start; n >= 0; --n) {
-      int x = sx + (cScan4[n] & 15), y = sy + (cScan4[n] >> 4);
-      int sig = level[y * N + x] != 0;
+    // reload this sub-block's levels (static offsets -> registers)
+    int lvn[16];
+    {
+      const int sx = (sb_scan[i] & 15) * 4, sy = (sb_scan[i] >> 4) * 4;
+      const int16_t* base = level + sy * N + sx;
+#pragma unroll
+      for (int n = 0; n < 16; ++n)
+        lvn[n] = base[(cScan4[n] >> 4) * N + (cScan4[n] & 15)];
+    }
+
+    const int prev = right + (below << 1);
+    // context adders shared by every coefficient of this sub-block
+    int sig_add;
+    if (cidx == 0)
+      sig_add = (i > 0 ? 3 : 0) + ((log2n == 3) ? 9 : 21);
+    else
+      sig_add = 27 + ((log2n == 3) ? 9 : 12);
+    const int sig_luma_off = cidx == 0 ? 0 : 0;  // folded into sig_add
+    (void)sig_luma_off;
+
+    // significance flags: unrolled reverse scan, mask accumulates
+    uint32_t sig_mask = 0;
+    const int start = (i == last_sb) ? last_pos - 1 : 15;
+    if (i == last_sb) sig_mask |= 1u << last_pos;
+#pragma unroll
+    for (int n = 15; n >= 0; --n) {
+      if (n > start) continue;
+      const int sig = lvn[n] != 0;
       if (n > 0 || !infer_dc) {
-        int sig_ctx;
-        if (x == 0 && y == 0) {
-          sig_ctx = 0;
+        int sctx;
+        if (i == 0 && n == 0) {
+          sctx = cidx == 0 ? 0 : 27;
         } else {
-          int right = 0, below = 0;
-          for (int j = 0; j < n_sb; ++j) {
-            if ((sb_scan[j] & 15) == sbx + 1 && (sb_scan[j] >> 4) == sby)
-              right = csbf[j];
-            if ((sb_scan[j] & 15) == sbx && (sb_scan[j] >> 4) == sby + 1)
-              below = csbf[j];
-          }
-          int prev = right + (below << 1);
-          int xp = x & 3, yp = y & 3;
-          if (prev == 0)
-            sig_ctx = (xp + yp == 0) ? 2 : (xp + yp < 3) ? 1 : 0;
-          else if (prev == 1)
-            sig_ctx = (yp == 0) ? 2 : (yp == 1) ? 1 : 0;
-          else if (prev == 2)
-            sig_ctx = (xp == 0) ? 2 : (xp == 1) ? 1 : 0;
-          else
-            sig_ctx = 2;
-          if (cidx == 0) {
-            if ((x >> 2) + (y >> 2) > 0) sig_ctx += 3;
-            sig_ctx += (log2n == 3) ? 9 : 21;
-          } else {
-            sig_ctx += (log2n == 3) ? 9 : 12;
-          }
+          sctx = cSigPat[prev][n] + sig_add;
         }
-        cab.bin(&ctx[HG_CTX_SIG + (cidx ? 27 : 0) + sig_ctx], sig);
+        cab.bin(&CTX_LDS(HG_CTX_SIG + sctx), sig);
         if (sig) infer_dc = false;
       }
-      if (sig) sig_pos[n_sig++] = n;
+      if (sig) sig_mask |= 1u << n;
     }
-    if (n_sig == 0) continue;
+    if (!sig_mask) continue;
 
+    // greater1 / greater2 (reverse scan over significant positions)
     int ctx_set = (i == 0 || cidx > 0) ? 0 : 2;
     if (i != last_sb && prev_g1_zero == 1) ctx_set += 1;
-    int g1_ctx = 1, first_g1 = -1;
-    int abs_lvl[16];
-    for (int k = 0; k < n_sig; ++k) {
-      int n = sig_pos[k];
-      int x = sx + (cScan4[n] & 15), y = sy + (cScan4[n] >> 4);
-      int l = level[y * N + x];
-      abs_lvl[k] = l < 0 ? -l : l;
-    }
-    for (int k = 0; k < n_sig && k < 8; ++k) {
-      int g1 = abs_lvl[k] > 1;
-      int gc = g1_ctx < 3 ? g1_ctx : 3;
-      cab.bin(&ctx[HG_CTX_GT1 + ctx_set * 4 + gc + (cidx ? 16 : 0)], g1);
-      if (g1) {
-        g1_ctx = 0;
-        if (first_g1 < 0) first_g1 = k;
-      } else if (g1_ctx > 0 && g1_ctx < 3) {
-        ++g1_ctx;
+    int g1_ctx = 1, first_g1 = -1, k = 0, abs_first_g1 = 0;
+#pragma unroll
+    for (int n = 15; n >= 0; --n) {
+      if (!((sig_mask >> n) & 1)) continue;
+      if (k < 8) {
+        const int a = lvn[n] < 0 ? -lvn[n] : lvn[n];
+        const int g1 = a > 1;
+        const int gc = g1_ctx < 3 ? g1_ctx : 3;
+        cab.bin(&CTX_LDS(HG_CTX_GT1 + ctx_set * 4 + gc + (cidx ? 16 : 0)),
+                g1);
+        if (g1) {
+          g1_ctx = 0;
+          if (first_g1 < 0) { first_g1 = n; abs_first_g1 = a; }
+        } else if (g1_ctx > 0 && g1_ctx < 3) {
+          ++g1_ctx;
+        }
       }
+      ++k;
     }
     prev_g1_zero = (g1_ctx == 0) ? 1 : 0;
     if (first_g1 >= 0) {
-      int g2 = abs_lvl[first_g1] > 2;
-      cab.bin(&ctx[HG_CTX_GT2 + ctx_set + (cidx ? 4 : 0)], g2);
+      const int a = abs_first_g1;
+      cab.bin(&CTX_LDS(HG_CTX_GT2 + ctx_set + (cidx ? 4 : 0)), a > 2);
     }
-    for (int k = 0; k < n_sig; ++k) {
-      int n = sig_pos[k];
-      int x = sx + (cScan4[n] & 15), y = sy + (cScan4[n] >> 4);
-      cab.bypass(level[y * N + x] < 0 ? 1 : 0);
-    }
+    // signs (bypass)
+#pragma unroll
+    for (int n = 15; n >= 0; --n)
+      if ((sig_mask >> n) & 1) cab.bypass(lvn[n] < 0 ? 1 : 0);
+    // remaining levels
     int rice = 0;
-    for (int k = 0; k < n_sig; ++k) {
-      int base = (k < 8) ? ((k == first_g1) ? 3 : 2) : 1;
-      if (abs_lvl[k] >= base) {
-        uint32_t rem = abs_lvl[k] - base;
+    k = 0;
+#pragma unroll
+    for (int n = 15; n >= 0; --n) {
+      if (!((sig_mask >> n) & 1)) continue;
+      const int a = lvn[n] < 0 ? -lvn[n] : lvn[n];
+      const int base = (k < 8) ? ((n == first_g1) ? 3 : 2) : 1;
+      if (a >= base) {
+        uint32_t rem = a - base;
         if (rem < (3u << rice)) {
-          int len = rem >> rice;
-          for (int b = 0; b < len; ++b) cab.bypass(1);
-          cab.bypass(0);
+          cab.bypass_unary(rem >> rice);
           if (rice) cab.bypass_bins(rem & ((1 << rice) - 1), rice);
         } else {
           int len = rice;
@@ -760,20 +809,20 @@ __device__ void dev_code_residual(DevCabac& cab, uint8_t* ctx,
             v -= 1u << len;
             ++len;
           }
-          for (int b = 0; b < 3 + len - rice; ++b) cab.bypass(1);
-          cab.bypass(0);
+          cab.bypass_unary(3 + len - rice);
           cab.bypass_bins(v, len);
         }
-        if (abs_lvl[k] > (3 << rice) && rice < 4) ++rice;
+        if (a > (3 << rice) && rice < 4) ++rice;
       }
+      ++k;
     }
   }
 }
 
-__device__ void dev_code_ctu(DevCabac& cab, uint8_t* ctx, int mode,
-                             int left_mode, int cbf_mask,
-                             const int16_t* lv) {
-  cab.bin(&ctx[HG_CTX_SPLIT], 0);
+__device__ void dev_code_ctu(DevCabac& cab, uint32_t* ctx_lds, int lane,
+                             int mode, int left_mode, int cbf_mask,
+                             const int16_t* __restrict__ lv) {
+  cab.bin(&CTX_LDS(HG_CTX_SPLIT), 0);
   int cand_a = left_mode >= 0 ? left_mode : 1;
   int cand_b = 1;
   int list[3];
@@ -800,7 +849,7 @@ __device__ void dev_code_ctu(DevCabac& cab, uint8_t* ctx, int mode,
       mpm_idx = i;
       break;
     }
-  cab.bin(&ctx[HG_CTX_PREVINTRA], mpm_idx >= 0 ? 1 : 0);
+  cab.bin(&CTX_LDS(HG_CTX_PREVINTRA), mpm_idx >= 0 ? 1 : 0);
   if (mpm_idx >= 0) {
     cab.bypass(mpm_idx > 0 ? 1 : 0);
     if (mpm_idx > 0) cab.bypass(mpm_idx - 1);
@@ -815,46 +864,58 @@ __device__ void dev_code_ctu(DevCabac& cab, uint8_t* ctx, int mode,
     if (mode > a) --rem;
     cab.bypass_bins(rem, 5);
   }
-  cab.bin(&ctx[HG_CTX_CHROMA], 0);
-  cab.bin(&ctx[HG_CTX_CBFC], (cbf_mask >> 1) & 1);
-  cab.bin(&ctx[HG_CTX_CBFC], (cbf_mask >> 2) & 1);
-  cab.bin(&ctx[HG_CTX_CBFY + 1], cbf_mask & 1);
-  if (cbf_mask & 1) dev_code_residual<16>(cab, ctx, lv, 0);
-  if (cbf_mask & 2) dev_code_residual<8>(cab, ctx, lv + 256, 1);
-  if (cbf_mask & 4) dev_code_residual<8>(cab, ctx, lv + 320, 2);
+  cab.bin(&CTX_LDS(HG_CTX_CHROMA), 0);
+  cab.bin(&CTX_LDS(HG_CTX_CBFC), (cbf_mask >> 1) & 1);
+  cab.bin(&CTX_LDS(HG_CTX_CBFC), (cbf_mask >> 2) & 1);
+  cab.bin(&CTX_LDS(HG_CTX_CBFY + 1), cbf_mask & 1);
+  if (cbf_mask & 1) dev_code_residual<16>(cab, ctx_lds, lane, lv, 0);
+  if (cbf_mask & 2) dev_code_residual<8>(cab, ctx_lds, lane, lv + 256, 1);
+  if (cbf_mask & 4) dev_code_residual<8>(cab, ctx_lds, lane, lv + 320, 2);
 }
 
 __global__ __launch_bounds__(64) void k_hevc_cabac(
     const int16_t* __restrict__ levels, const int* __restrict__ meta,
     int ctbw, int n_jobs, const HevcJob* __restrict__ jobs,
     uint8_t* __restrict__ out, int out_stride, int* __restrict__ counts) {
-  const int j = blockIdx.x * 64 + threadIdx.x;
+  __shared__ uint32_t ctx_lds[HG_NUM_CTX];
+  __shared__ uint32_t ptab[256];   // packed (state, qRangeIdx) entries
+  const int lane = threadIdx.x;
+  const int j = blockIdx.x;
   if (j >= n_jobs) return;
-  const HevcJob job = jobs[j];
 
-  // context init (§9.3.2.2)
-  uint8_t ctx[HG_NUM_CTX];
+  const HevcJob job = jobs[j];
   const int qp = job.qp < 0 ? 0 : job.qp > 51 ? 51 : job.qp;
-  for (int i = 0; i < HG_NUM_CTX; ++i) {
+  for (int i = lane; i < 256; i += 64) {
+    const int state = i >> 2, q = i & 3;
+    const uint32_t lps = cRangeTabLps[state][q];
+    const uint32_t nlps = cTransIdxLps[state];
+    const uint32_t nmps = state < 62 ? state + 1 : state;
+    ptab[i] = lps | (nlps << 8) | (nmps << 14)
+              | ((uint32_t)cRenorm[lps >> 3] << 20);
+  }
+  for (int i = lane; i < HG_NUM_CTX; i += 64) {
     int iv = cInitVals[i];
     int slope = (iv >> 4) * 5 - 45;
     int off = ((iv & 15) << 3) - 16;
     int pre = ((slope * qp) >> 4) + off;
     pre = pre < 1 ? 1 : pre > 126 ? 126 : pre;
-    ctx[i] = pre <= 63 ? (uint8_t)((63 - pre) << 1)
-                       : (uint8_t)(((pre - 64) << 1) | 1);
+    ctx_lds[i] = pre <= 63 ? (uint32_t)((63 - pre) << 1)
+                           : (uint32_t)(((pre - 64) << 1) | 1);
   }
+  __syncthreads();
+  if (lane != 0) return;
 
   DevCabac cab;
   cab.init(out + (size_t)j * out_stride);
+  cab.ptab = ptab;
   int left_mode = -1;
   for (int ci = 0; ci < job.seg_w; ++ci) {
     const int cx = job.ctu_x0 + ci;
     const size_t mbase = (size_t)(job.ctu_row * ctbw + cx);
     const int mode = meta[mbase * kHevcMetaPerCtu + 0];
     const int cbf = meta[mbase * kHevcMetaPerCtu + 1];
-    dev_code_ctu(cab, ctx, mode, left_mode,
-                 cbf, levels + mbase * kHevcLevelsPerCtu);
+    dev_code_ctu(cab, ctx_lds, lane, mode, left_mode, cbf,
+                 levels + mbase * kHevcLevelsPerCtu);
     left_mode = mode;
     cab.terminate(ci == job.seg_w - 1 ? 1 : 0);
   }
@@ -869,7 +930,7 @@ void launch_hevc_cabac(const int16_t* d_levels, const int* d_meta, int ctbw,
                        int out_stride_bytes, int* d_counts,
                        hipStream_t stream) {
   if (n_jobs == 0) return;
-  hipLaunchKernelGGL(k_hevc_cabac, dim3((n_jobs + 63) / 64), dim3(64), 0,
+  hipLaunchKernelGGL(k_hevc_cabac, dim3(n_jobs), dim3(64), 0,
                      stream, d_levels, d_meta, ctbw, n_jobs, d_jobs, d_out,
                      out_stride_bytes, d_counts);
 }

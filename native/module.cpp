@@ -465,8 +465,9 @@ PYBIND11_MODULE(_native, m) {
         p = make_hip_pipeline(s);
         if (!p) throw std::runtime_error("no HIP pipeline available");
       } else {
-        p = output_mode == 1 ? make_cpu_h264_pipeline(s)
-                             : make_cpu_jpeg_pipeline(s);
+        p = output_mode == 1   ? make_cpu_h264_pipeline(s)
+            : output_mode == 2 ? make_cpu_hevc_pipeline(s)
+                               : make_cpu_jpeg_pipeline(s);
       }
       p->set_pipeline_depth(pipeline_depth);
     }

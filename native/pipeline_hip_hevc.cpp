@@ -44,6 +44,8 @@ class HipHevcPipeline : public EncodePipeline {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     cpu_entropy_ = std::getenv("HIPFLUX_CPU_HEVC_ENTROPY") != nullptr;
+    timing_ = std::getenv("HIPFLUX_TIMES") != nullptr;
+    for (auto& e : ev_) HIP_CHECK(hipEventCreate(&e));
     stripe_h_ = std::max(16, s.stripe_height & ~15);
     alloc_for(s.capture_width, s.capture_height);
   }
@@ -131,6 +133,7 @@ class HipHevcPipeline : public EncodePipeline {
     }
     if (n_jobs == 0) return;
 
+    if (timing_) HIP_CHECK(hipEventRecord(ev_[0], stream_));
     HIP_CHECK(hipMemcpyAsync(d_jobs_, h_jobs_,
                              sizeof(hevcgpu::HevcJob) * n_jobs,
                              hipMemcpyHostToDevice, stream_));
@@ -144,8 +147,10 @@ class HipHevcPipeline : public EncodePipeline {
                                hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipStreamSynchronize(stream_));
     } else {
+      if (timing_) HIP_CHECK(hipEventRecord(ev_[1], stream_));
       hevcgpu::launch_hevc_cabac(d_levels_, d_meta_, ctbw_, n_jobs, d_jobs_,
                                  d_out_, out_stride_, d_counts_, stream_);
+      if (timing_) HIP_CHECK(hipEventRecord(ev_[2], stream_));
       HIP_CHECK(hipMemcpyAsync(h_counts_, d_counts_,
                                sizeof(int) * 3 * n_jobs,
                                hipMemcpyDeviceToHost, stream_));
@@ -169,6 +174,17 @@ class HipHevcPipeline : public EncodePipeline {
                               hipMemcpyDeviceToHost));
       }
       copy_cap_ = std::max(4096, 2 * max_count);
+      if (timing_) {
+        float t_rows = 0, t_cab = 0;
+        HIP_CHECK(hipEventRecord(ev_[3], stream_));
+        HIP_CHECK(hipEventSynchronize(ev_[3]));
+        (void)hipEventElapsedTime(&t_rows, ev_[0], ev_[1]);
+        (void)hipEventElapsedTime(&t_cab, ev_[1], ev_[2]);
+        float t_d2h = 0;
+        (void)hipEventElapsedTime(&t_d2h, ev_[2], ev_[3]);
+        std::fprintf(stderr, "[hevc-times] rows %.2fms cabac %.2fms d2h %.2fms\n",
+                     t_rows, t_cab, t_d2h);
+      }
     }
 
     // ---- host NAL assembly (parallel over stripes)
@@ -205,6 +221,13 @@ class HipHevcPipeline : public EncodePipeline {
       });
     }
     pool_.wait_all();
+    if (timing_) {
+      static double t_last = 0;
+      double now = now_ms();
+      std::fprintf(stderr, "[hevc-times] frame-to-frame %.2fms\n",
+                   t_last ? now - t_last : 0.0);
+      t_last = now;
+    }
     for (auto& o : outs) {
       EncodedStripe s;
       s.type = StripeType::kHevc;
@@ -321,7 +344,9 @@ class HipHevcPipeline : public EncodePipeline {
   CaptureSettings settings_;
   ThreadPool pool_;
   hipStream_t stream_{};
+  hipEvent_t ev_[4] = {};
   bool cpu_entropy_ = false;
+  bool timing_ = false;
   int stripe_h_ = 64;
   int w_ = 0, h_ = 0, ctbw_ = 0, ctbh_ = 0, ypitch_ = 0, cpitch_ = 0;
   int max_jobs_ = 0, out_stride_ = 0, copy_cap_ = 1 << 30;
