@@ -112,6 +112,10 @@ class StreamingService:
         self._stats_task: Optional[asyncio.Task] = None
         self._watchdog_task: Optional[asyncio.Task] = None
         self.frames_relayed = 0
+        # glass-to-glass accounting: frame id -> capture ts (native
+        # steady-clock ms) and the rolling capture->ACK sample window
+        self._frame_capture_ts: dict[int, float] = {}
+        self._g2g_ms: list[float] = []
         self.audio: Optional[object] = None
         self._audio_queue: Optional[asyncio.Queue] = None
         self._audio_task: Optional[asyncio.Task] = None
@@ -217,7 +221,8 @@ class StreamingService:
                       _display=display):
             # native thread -> loop (the only allowed crossing)
             loop.call_soon_threadsafe(self._fanout, _display, data,
-                                      frame_id, y, is_keyframe)
+                                      frame_id, y, is_keyframe,
+                                      capture_ts_ms)
 
         if display == "primary" and not self.settings.capture_cursor:
             # client-side cursor rendering: push shape updates over control
@@ -296,8 +301,17 @@ class StreamingService:
 
     # ---- fan-out -----------------------------------------------------------
     def _fanout(self, display: str, data: bytes, frame_id: int, y: int,
-                is_keyframe: bool):
+                is_keyframe: bool, capture_ts_ms: float = 0.0):
         self.frames_relayed += 1
+        # capture timestamp per frame id (native steady_clock == Python
+        # time.monotonic epoch) for glass-to-glass accounting at ACK time
+        if capture_ts_ms and display == "primary":
+            fid = frame_id & 0xFFFF
+            if fid not in self._frame_capture_ts:
+                self._frame_capture_ts[fid] = capture_ts_ms
+                if len(self._frame_capture_ts) > 512:
+                    for k in sorted(self._frame_capture_ts)[:256]:
+                        self._frame_capture_ts.pop(k, None)
         for cs in list(self.clients.values()):
             if cs.relay.dead or cs.display != display:
                 continue
@@ -416,7 +430,19 @@ class StreamingService:
     async def _on_text(self, state: ClientState, text: str) -> Optional[str]:
         verb, rest = P.parse_control(text)
         if verb == "CLIENT_FRAME_ACK":
-            state.note_ack(int(rest) & 0xFFFF)
+            fid = int(rest) & 0xFFFF
+            state.note_ack(fid)
+            # glass-to-glass: framebuffer capture -> client ACK received
+            # (the BASELINE latency metric; reference frame-ACK plumbing
+            # selkies.py:2369,3612). Includes encode, WS delivery and the
+            # client's ack turnaround on the measurement path.
+            cap_ts = self._frame_capture_ts.pop(fid, None)
+            if cap_ts is not None:
+                g2g = time.monotonic() * 1e3 - cap_ts
+                if 0 <= g2g < 10_000:
+                    self._g2g_ms.append(g2g)
+                    if len(self._g2g_ms) > 2048:
+                        del self._g2g_ms[:1024]
             self._backpressure_tick()
             return None
         if verb == "SETTINGS":
@@ -602,6 +628,15 @@ class StreamingService:
                         logger.exception("capture rebuild failed for %s", d)
 
     # ---- stats --------------------------------------------------------------
+    def g2g_percentiles(self) -> dict:
+        """Capture->ACK latency percentiles over the rolling window."""
+        if not self._g2g_ms:
+            return {"p50": None, "p95": None, "n": 0}
+        s = sorted(self._g2g_ms)
+        return {"p50": round(s[len(s) // 2], 3),
+                "p95": round(s[int(len(s) * 0.95)], 3),
+                "n": len(s)}
+
     def stats(self) -> dict:
         cap = self.capture
         return {
@@ -612,6 +647,7 @@ class StreamingService:
             "frames_encoded": cap.frames_encoded if cap else 0,
             "stripes_emitted": cap.stripes_emitted if cap else 0,
             "last_encode_ms": cap.last_encode_ms if cap else 0.0,
+            "glass_to_glass_ms": self.g2g_percentiles(),
             "relays": [
                 {"backlog": c.relay.backlog_bytes,
                  "sent": c.relay.sent_frames,

@@ -331,3 +331,48 @@ def test_mode_switch_runtime():
             await runner.cleanup()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_glass_to_glass_stats():
+    """CLIENT_FRAME_ACK drives the capture->ACK latency window exposed in
+    /api/stats (the BASELINE glass-to-glass metric plumbing)."""
+    async def main():
+        server = make_server(SELKIES_VIDEO_FULLFRAME="true")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"http://127.0.0.1:{port}/ws") as ws:
+                    acked = 0
+                    rows = {}
+                    n_rows = 3  # 192 / 64
+                    for _ in range(400):
+                        msg = await asyncio.wait_for(ws.receive(), 10)
+                        if msg.type != aiohttp.WSMsgType.BINARY:
+                            continue
+                        d = msg.data
+                        if d[0] != 0x04:
+                            continue
+                        fid = (d[2] << 8) | d[3]
+                        s = rows.setdefault(fid, set())
+                        s.add((d[4] << 8) | d[5])
+                        if len(s) == n_rows:
+                            rows.pop(fid)
+                            await ws.send_str(f"CLIENT_FRAME_ACK,{fid}")
+                            acked += 1
+                            if acked >= 5:
+                                break
+                    assert acked >= 5
+                    # give the server a beat to process the last ack
+                    await asyncio.sleep(0.1)
+                r = await sess.get(f"http://127.0.0.1:{port}/api/stats")
+                st = await r.json()
+            g2g = st["streaming"]["glass_to_glass_ms"]
+            assert g2g["n"] >= 4
+            assert 0 < g2g["p50"] < 5000
+            assert g2g["p95"] >= g2g["p50"]
+        finally:
+            server.streaming.stop_capture()
+            await runner.cleanup()
+
+    asyncio.run(main())
