@@ -218,6 +218,37 @@ def main():
              for _ in range(n_sess)]
     pipe = pipes[0]
 
+    # ---- tile-boundary collective layer (BASELINE config 5 pattern) -----
+    # GPU: RCCL over xGMI, one-hop p2p all-gather of each band's top+bottom
+    # recon rows (the halo a cross-tile predictor/deblocker consumes).
+    # CPU/world>1: gloo all-gather of the same rows (plumbing parity).
+    comm = None
+    boundary_rows = 16
+    comm_ms = []
+    boundary_bytes = 0
+    if tile and use_gpu and distributed:
+        uid = [_native.TileComm.make_uid() if rank == 0 else None]
+        dist.broadcast_object_list(uid, src=0)
+        comm = _native.TileComm(rank, world, uid[0], local_rank)
+
+    def tile_exchange(frame_np):
+        nonlocal boundary_bytes
+        if not tile or not distributed:
+            return
+        if comm is not None:
+            ptr, nbytes = pipe.boundary_dev(boundary_rows)
+            if nbytes:
+                boundary_bytes = nbytes
+                comm_ms.append(comm.exchange(ptr, nbytes, 1))
+            return
+        t = torch.from_numpy(
+            np.ascontiguousarray(frame_np[:boundary_rows])).flatten()
+        bufs = [torch.empty_like(t) for _ in range(world)]
+        ts = time.perf_counter()
+        dist.all_gather(bufs, t)
+        comm_ms.append((time.perf_counter() - ts) * 1e3)
+        boundary_bytes = t.numel()
+
     # synthetic capture source: pre-generated random BGRX frames, cycled.
     # Every frame differs everywhere (worst case for a screen encoder).
     rng = np.random.default_rng(1234 + rank)
@@ -259,6 +290,7 @@ def main():
             starts[next_fid] = time.perf_counter()
             next_fid += 1
             nbytes, _ = pipe.encode(frames[i % n_src], False)
+            tile_exchange(frames[i % n_src])
             done = pipe.last_frame_id
             if done in starts:
                 lat_ms.append((time.perf_counter() - starts.pop(done)) * 1e3)
@@ -346,6 +378,12 @@ def main():
                 "pipeline": pipe.pipeline,
                 "pipeline_depth": depth,
                 "latency_p50_ms": round(p50, 3),
+                "tile_boundary_exchange": (
+                    {"rows": boundary_rows, "bytes": boundary_bytes,
+                     "p50_ms": round(float(np.percentile(comm_ms, 50)), 3),
+                     "schedule": ("rccl-one-hop-p2p" if comm is not None
+                                  else "gloo-fallback")}
+                    if comm_ms else None),
                 "latency_p95_ms": round(p95, 3),
                 "bitrate_mbps_rank0": round(total_bytes * 8 / elapsed / 1e6,
                                             2),

@@ -56,6 +56,14 @@ class EncodePipeline {
   virtual void set_pipeline_depth(int) {}
   virtual void flush(const Emit&) {}
 
+  // Device pointer to the luma reconstruction/reference plane (GPU
+  // pipelines only). Tile-parallel encode reads boundary rows from here
+  // for the RCCL exchange (rccl_comm.h). Returns false on CPU paths.
+  virtual bool recon_dev(void** y, int* ypitch, int* height) {
+    (void)y; (void)ypitch; (void)height;
+    return false;
+  }
+
   // Test-only introspection: reconstruction planes + (GPU) level/meta
   // buffers of the last encoded frame. Returns false when unsupported.
   struct DebugDump {

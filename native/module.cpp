@@ -13,6 +13,9 @@
 #include "cpu/hevc/encoder.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
+#include <hip/hip_runtime.h>
+
+#include "rccl_comm.h"
 
 namespace py = pybind11;
 using namespace hipflux;
@@ -333,6 +336,58 @@ PYBIND11_MODULE(_native, m) {
       },
       py::arg("ops"), py::arg("qp") = 30);
 
+  // ---- RCCL tile-parallel collective layer (BASELINE config 5) ----------
+  py::class_<TileComm>(m, "TileComm")
+      .def(py::init<int, int, const std::string&, int>(), py::arg("rank"),
+           py::arg("world"), py::arg("uid"), py::arg("device") = 0)
+      .def_static("make_uid",
+                  [] { return py::bytes(TileComm::make_uid()); })
+      .def_property_readonly("rank", &TileComm::rank)
+      .def_property_readonly("world", &TileComm::world)
+      .def(
+          "exchange",
+          [](TileComm& c, uintptr_t src_dev, size_t bytes, int schedule) {
+            py::gil_scoped_release rel;
+            return c.exchange(src_dev, bytes, schedule);
+          },
+          py::arg("src_dev"), py::arg("bytes"), py::arg("schedule") = 1,
+          "All-gather this rank's device payload to every rank; "
+          "schedule 0 = ring (bulk), 1 = one-hop p2p (latency). "
+          "Returns ms.")
+      .def(
+          "exchange_host",
+          [](TileComm& c, py::bytes payload, int schedule) {
+            std::string s = payload;
+            void* d = nullptr;
+            if (hipMalloc(&d, s.size()) != hipSuccess)
+              throw std::runtime_error("alloc failed");
+            (void)hipMemcpy(d, s.data(), s.size(), hipMemcpyHostToDevice);
+            double ms;
+            {
+              py::gil_scoped_release rel;
+              ms = c.exchange(reinterpret_cast<uintptr_t>(d), s.size(),
+                              schedule);
+            }
+            (void)hipFree(d);
+            return ms;
+          },
+          py::arg("payload"), py::arg("schedule") = 1,
+          "Upload host bytes and exchange (tests).")
+      .def("gathered",
+           [](TileComm& c, int rank, size_t bytes) {
+             auto v = c.gathered(rank, bytes);
+             return py::bytes(reinterpret_cast<const char*>(v.data()),
+                              v.size());
+           })
+      .def(
+          "broadcast",
+          [](TileComm& c, uintptr_t buf, size_t bytes, int root) {
+            py::gil_scoped_release rel;
+            return c.broadcast(buf, bytes, root);
+          },
+          py::arg("buf_dev"), py::arg("bytes"), py::arg("root") = 0,
+          "Broadcast a reference-frame buffer from root. Returns ms.");
+
   m.def(
       "bgrx_to_yuv420",
       [](py::buffer bgrx, int width, int height) {
@@ -448,6 +503,8 @@ PYBIND11_MODULE(_native, m) {
     int w, h, qp, stripe_h;
     uint32_t frame_id = 0;
     long last_frame_id = -1;   // frame whose stripes the last call emitted
+    void* d_boundary = nullptr;
+    size_t boundary_cap = 0;
     BenchPipeline(const std::string& kind, int width, int height, int qp_,
                   int stripe, int output_mode, int gpu_id,
                   int pipeline_depth)
@@ -482,6 +539,43 @@ PYBIND11_MODULE(_native, m) {
       .def_property_readonly("pipeline",
                              [](BenchPipeline& b) { return b.p->name(); })
       .def_readonly("last_frame_id", &BenchPipeline::last_frame_id)
+      .def(
+          "boundary_dev",
+          [](BenchPipeline& b, int rows) {
+            // pack the top and bottom `rows` recon rows into a compact
+            // device buffer for the RCCL tile-boundary exchange; returns
+            // (device_ptr, nbytes) or (0, 0) on CPU pipelines
+            void* y = nullptr;
+            int ypitch = 0, height = 0;
+            if (!b.p->recon_dev(&y, &ypitch, &height)) {
+              return py::make_tuple(static_cast<uintptr_t>(0),
+                                    static_cast<size_t>(0));
+            }
+            rows = std::min(rows, height / 2);
+            size_t need = static_cast<size_t>(ypitch) * rows * 2;
+            if (b.boundary_cap < need) {
+              if (b.d_boundary) (void)hipFree(b.d_boundary);
+              if (hipMalloc(&b.d_boundary, need) != hipSuccess)
+                throw std::runtime_error("boundary alloc failed");
+              b.boundary_cap = need;
+            }
+            auto* dst = static_cast<uint8_t*>(b.d_boundary);
+            auto* srcy = static_cast<uint8_t*>(y);
+            size_t half = static_cast<size_t>(ypitch) * rows;
+            if (hipMemcpy(dst, srcy, half, hipMemcpyDeviceToDevice) !=
+                    hipSuccess ||
+                hipMemcpy(dst + half,
+                          srcy + static_cast<size_t>(ypitch) *
+                                     (height - rows),
+                          half, hipMemcpyDeviceToDevice) != hipSuccess)
+              throw std::runtime_error("boundary pack failed");
+            return py::make_tuple(
+                reinterpret_cast<uintptr_t>(b.d_boundary),
+                static_cast<size_t>(need));
+          },
+          py::arg("rows") = 16,
+          "Pack top+bottom recon rows into a device buffer; returns "
+          "(dev_ptr, nbytes) for TileComm.exchange.")
       .def(
           "resize",
           [](BenchPipeline& b, int w, int h) {

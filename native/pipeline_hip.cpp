@@ -866,6 +866,13 @@ class HipH264Pipeline : public EncodePipeline {
   }
 
  public:
+  bool recon_dev(void** y, int* ypitch, int* height) override {
+    *y = d_refY_;
+    *ypitch = ypitch_;
+    *height = mbh_ * 16;
+    return d_refY_ != nullptr;
+  }
+
   bool debug_dump(DebugDump& d) override {
     HIP_CHECK(hipStreamSynchronize(stream_));
     d.w = w_;

@@ -244,6 +244,13 @@ class HipHevcPipeline : public EncodePipeline {
 
   const char* name() const override { return "hip-hevc"; }
 
+  bool recon_dev(void** y, int* ypitch, int* height) override {
+    *y = d_curY_;
+    *ypitch = ypitch_;
+    *height = ctbh_ * 16;
+    return d_curY_ != nullptr;
+  }
+
   bool debug_dump(DebugDump& d) override {
     d.w = w_;
     d.h = h_;
