@@ -158,6 +158,14 @@ def main():
                          "elsewhere). Latency is reported per FRAME "
                          "(submit -> bitstream ready), so depth 2's "
                          "deeper-pipe latency shows up honestly in p50/p95")
+    ap.add_argument("--content", choices=["noise", "scroll"],
+                    default="noise",
+                    help="synthetic content: noise = every pixel random "
+                         "every frame (worst case, all-intra, the ME "
+                         "hopeless-MB gate skips the search); scroll = a "
+                         "random texture scrolling 32 px/frame, 100%% "
+                         "damage but trackable motion, so the MFMA "
+                         "motion search runs EVERY frame")
     ap.add_argument("--sessions", type=int, default=1,
                     help="independent encode sessions per rank/GPU, run on "
                          "concurrent threads (the BASELINE metric's "
@@ -257,10 +265,21 @@ def main():
     n_src = min(24, max(4, args.warmup))
     if tile:
         rng = np.random.default_rng(1234)   # all ranks share the frame
-    frames = [np.ascontiguousarray(
-        rng.integers(0, 256, (args.height, args.width, 4),
-                     dtype=np.uint8)[band0:band0 + enc_h])
-        for _ in range(n_src)]
+    if args.content == "scroll":
+        # tall random texture; frame i is a window scrolled 32 px down —
+        # every pixel changes every frame yet content is trackable
+        dy = 32
+        tex = rng.integers(0, 256,
+                           (args.height + dy * n_src, args.width, 4),
+                           dtype=np.uint8)
+        frames = [np.ascontiguousarray(
+            tex[i * dy:i * dy + args.height][band0:band0 + enc_h])
+            for i in range(n_src)]
+    else:
+        frames = [np.ascontiguousarray(
+            rng.integers(0, 256, (args.height, args.width, 4),
+                         dtype=np.uint8)[band0:band0 + enc_h])
+            for _ in range(n_src)]
 
     def sync():
         if use_gpu:
@@ -375,6 +394,7 @@ def main():
                                 else f"sessions{world * n_sess}"),
                 "resolution": f"{args.width}x{args.height}",
                 "qp": args.qp,
+                "content": args.content,
                 "pipeline": pipe.pipeline,
                 "pipeline_depth": depth,
                 "latency_p50_ms": round(p50, 3),
