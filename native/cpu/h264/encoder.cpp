@@ -7,6 +7,7 @@
 
 #include "bitwriter.h"
 #include "cavlc.h"
+#include "deblock.h"
 #include "transform.h"
 
 namespace hipflux {
@@ -81,6 +82,7 @@ inline int inter_cbp_codenum(int cbp) {
 
 struct StripeEncoder::Impl {
   int w, h, mbw, mbh, yw, yh;
+  bool deblock;          // in-loop deblocking (idc=2: within-slice only)
   Frame ref, cur;        // reference (prev recon) and current recon
   Frame src;             // padded source planes
   uint32_t frame_num = 0;
@@ -90,7 +92,8 @@ struct StripeEncoder::Impl {
   std::vector<int> prev_mv;   // per-MB quarter-pel mv of the last frame
   std::vector<uint8_t> prev_bad;  // MB was hopeless-intra last frame
 
-  Impl(int width, int height) : w(width), h(height) {
+  Impl(int width, int height, bool deblock_on)
+      : w(width), h(height), deblock(deblock_on) {
     mbw = (w + 15) / 16;
     mbh = (h + 15) / 16;
     yw = mbw * 16;
@@ -196,7 +199,7 @@ struct StripeEncoder::Impl {
       b.u(0, 1);                 // adaptive_ref_pic_marking_mode
     }
     b.se(qp - 26);               // slice_qp_delta
-    b.ue(1);                     // disable_deblocking_filter_idc = 1
+    b.ue(deblock ? 2 : 1);       // disable_deblocking_filter_idc
   }
 
   // ---- intra 16x16 macroblock ---------------------------------------------
@@ -846,13 +849,16 @@ struct StripeEncoder::Impl {
     // leaves mostly-uncorrelated content (fresh content / occlusion).
     const long inter_thresh = 6 * skip_thresh;
 
+    std::vector<DbMb> rowmb(mbw);
     for (int mb_row = 0; mb_row < mbh; ++mb_row) {
       BitWriter b;
       write_slice_header(b, idr, mb_row, qp);
       RowCtx ctx;
       MbTotals tot;
       for (int mbx = 0; mbx < mbw; ++mbx) {
+        rowmb[mbx] = DbMb{};
         if (idr) {
+          rowmb[mbx].intra = 1;
           encode_i16(b, mbx, mb_row, qp, false, ctx, tot);
           ctx.have_left = true;
           ctx.left_is_inter = false;
@@ -986,6 +992,10 @@ struct StripeEncoder::Impl {
               static_cast<int>(static_cast<uint32_t>(best_q_y) << 16);
           flush_skip_run(b, ctx);
           encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
+          rowmb[mbx].mvx = static_cast<int16_t>(best_q_x);
+          rowmb[mbx].mvy = static_cast<int16_t>(best_q_y);
+          for (int bi = 0; bi < 16; ++bi)
+            if (tot.luma[bi]) rowmb[mbx].nz |= 1u << bi;
           ctx.have_left = true;
           ctx.left_is_inter = true;
           ctx.left_mvx = best_q_x;
@@ -1004,6 +1014,7 @@ struct StripeEncoder::Impl {
                             static_cast<int>(
                                 static_cast<uint32_t>(best_mvy * 4) << 16);
           flush_skip_run(b, ctx);
+          rowmb[mbx].intra = 1;
           encode_i16(b, mbx, mb_row, qp, true, ctx, tot);
           ctx.have_left = true;
           ctx.left_is_inter = false;
@@ -1022,6 +1033,10 @@ struct StripeEncoder::Impl {
       }
       b.rbsp_trailing();
       b.emit_nal(out, idr ? 3 : 2, idr ? 5 : 1, mb_row == 0);
+      if (deblock)
+        deblock_segment(cur.y.data.data(), cur.y.pitch, cur.cb.data.data(),
+                        cur.cr.data.data(), cur.cb.pitch, mb_row, 0, mbw,
+                        rowmb.data(), qp, chroma_qp(qp));
     }
 
     std::swap(ref, cur);
@@ -1035,8 +1050,9 @@ struct StripeEncoder::Impl {
   }
 };
 
-StripeEncoder::StripeEncoder(int width, int height)
-    : impl_(new Impl(width, height)), width_(width), height_(height) {}
+StripeEncoder::StripeEncoder(int width, int height, bool deblock)
+    : impl_(new Impl(width, height, deblock)),
+      width_(width), height_(height) {}
 StripeEncoder::~StripeEncoder() = default;
 
 void StripeEncoder::encode_frame(const uint8_t* y, int ypitch,

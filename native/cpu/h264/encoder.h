@@ -34,7 +34,7 @@ struct EncodeStats {
 
 class StripeEncoder {
  public:
-  StripeEncoder(int width, int height);
+  StripeEncoder(int width, int height, bool deblock = true);
   ~StripeEncoder();
 
   // Encode one frame from planar YUV420 (pitch in bytes). qp in [0,51].

@@ -240,7 +240,7 @@ void encode_seg_nal_from_gpu(const GpuStripeParams& p, int row, int mbx0,
   {
     BitWriter b;
     write_slice_header_bits(b, p.idr, row * mbw_stripe + mbx0, p.frame_num,
-                            p.idr_pic_id, p.qp);
+                            p.idr_pic_id, p.qp, p.deblock ? 2 : 1);
     RowCtx ctx;
     const size_t mb_base = (size_t)(p.mb_row0 + row) * p.mbw;
     for (int mbx = mbx0; mbx < mbx0 + seg_mbw; ++mbx) {

@@ -19,6 +19,7 @@ struct GpuStripeParams {
   int height;              // stripe logical height (pixels)
   int qp;
   bool idr;
+  bool deblock = false;   // slice headers signal idc=2
   uint32_t frame_num;
   uint32_t idr_pic_id;
 };

@@ -87,7 +87,7 @@ inline void write_pps_nal(std::vector<uint8_t>& out) {
 
 inline void write_slice_header_bits(BitWriter& b, bool idr, int first_mb,
                                     uint32_t frame_num, uint32_t idr_pic_id,
-                                    int qp) {
+                                    int qp, int deblock_idc = 1) {
   b.ue(first_mb);
   b.ue(idr ? 7 : 5);
   b.ue(0);
@@ -104,7 +104,9 @@ inline void write_slice_header_bits(BitWriter& b, bool idr, int first_mb,
     b.u(0, 1);
   }
   b.se(qp - 26);
-  b.ue(1);      // disable_deblocking_filter_idc
+  // 1 = deblocking off; 2 = filter within the slice only (our slices are
+  // single MB rows / row segments, so no cross-row dependency)
+  b.ue(deblock_idc);
 }
 
 }  // namespace h264

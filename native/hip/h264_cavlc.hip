@@ -380,7 +380,7 @@ __global__ void __launch_bounds__(1024) k_h264_cavlc_rows(
         bw.u(0, 1);
       }
       bw.se(job.qp - 26);
-      bw.ue(1);  // disable_deblocking_filter_idc
+      bw.ue((job.flags & 2) ? 2 : 1);  // disable_deblocking_filter_idc
       bits = bw.flush();
     } else if (item == nitems - 1) {
       if (!i_slice && s_trailing > 0) bw.ue(s_trailing);

@@ -33,6 +33,12 @@ void launch_h264_rows(const uint8_t* srcY, const uint8_t* srcCb,
                       const RowJob* d_jobs, int16_t* d_levels, int* d_meta,
                       hipStream_t stream);
 
+void launch_h264_deblock(uint8_t* d_curY, uint8_t* d_curCb,
+                         uint8_t* d_curCr, int ypitch, int cpitch, int mbw,
+                         int n_jobs, const RowJob* d_jobs,
+                         const int16_t* d_levels, const int* d_meta,
+                         hipStream_t stream);
+
 void launch_h264_cavlc(const int16_t* d_levels, const int* d_meta, int mbw,
                        int n_jobs, const RowJob* d_jobs, uint32_t* d_stage,
                        int* d_nbits, uint32_t* d_out, int out_stride_words,

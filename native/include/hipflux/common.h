@@ -43,6 +43,7 @@ struct CaptureSettings {
   double vbv_multiplier = 1.5;
   double keyframe_interval_s = 0.0; // 0 = infinite GOP, IDR on demand
   bool video_streaming_mode = false;
+  bool video_deblock = true;       // in-loop deblocking (idc=2 within-slice)
   int pipeline_depth = 1;           // 2 = one encode frame in flight
                                     // (throughput mode; emission lags one
                                     // frame — recording/transcode use)

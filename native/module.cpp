@@ -50,6 +50,7 @@ PYBIND11_MODULE(_native, m) {
       .def_readwrite("vbv_multiplier", &CaptureSettings::vbv_multiplier)
       .def_readwrite("keyframe_interval_s", &CaptureSettings::keyframe_interval_s)
       .def_readwrite("video_streaming_mode", &CaptureSettings::video_streaming_mode)
+      .def_readwrite("video_deblock", &CaptureSettings::video_deblock)
       .def_readwrite("video_fullcolor", &CaptureSettings::video_fullcolor)
       .def_readwrite("use_paint_over_quality", &CaptureSettings::use_paint_over_quality)
       .def_readwrite("paint_over_trigger_frames", &CaptureSettings::paint_over_trigger_frames)
@@ -177,8 +178,8 @@ PYBIND11_MODULE(_native, m) {
     h264::StripeEncoder enc;
     int w, h, ypitch, cpitch;
     std::vector<uint8_t> yuv;
-    PyH264(int width, int height)
-        : enc(width, height), w(width), h(height) {
+    PyH264(int width, int height, bool deblock)
+        : enc(width, height, deblock), w(width), h(height) {
       ypitch = (w + 15) & ~15;
       cpitch = ypitch / 2;
       int yh = (h + 15) & ~15;
@@ -186,7 +187,8 @@ PYBIND11_MODULE(_native, m) {
     }
   };
   py::class_<PyH264>(m, "H264Encoder")
-      .def(py::init<int, int>(), py::arg("width"), py::arg("height"))
+      .def(py::init<int, int, bool>(), py::arg("width"), py::arg("height"),
+           py::arg("deblock") = true)
       .def(
           "encode",
           [](PyH264& self, py::buffer bgrx, int qp, bool idr) {

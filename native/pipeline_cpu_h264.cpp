@@ -28,7 +28,8 @@ class CpuH264Pipeline : public EncodePipeline {
         int y0 = static_cast<int>(i) * stripe_h;
         int hgt = std::min(stripe_h, frame.height - y0);
         encoders_.push_back(
-            std::make_unique<h264::StripeEncoder>(frame.width, hgt));
+            std::make_unique<h264::StripeEncoder>(
+                frame.width, hgt, settings_.video_deblock));
       }
       // per-stripe scratch YUV planes
       ypitch_ = (frame.width + 15) & ~15;

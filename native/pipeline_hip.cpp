@@ -519,7 +519,8 @@ class HipH264Pipeline : public EncodePipeline {
           int segw = std::min(seg_w0_, mbw_ - x);
           h_jobs_[par][n_jobs].mb_row = row0 + r;
           h_jobs_[par][n_jobs].qp = qp;
-          h_jobs_[par][n_jobs].flags = sj.idr ? 1 : 0;
+          h_jobs_[par][n_jobs].flags =
+              (sj.idr ? 1 : 0) | (settings_.video_deblock ? 2 : 0);
           h_jobs_[par][n_jobs].stripe_y0 = sy0;
           h_jobs_[par][n_jobs].stripe_y1 = sy1;
           h_jobs_[par][n_jobs].first_mb = r * mbw_ + x;
@@ -599,6 +600,13 @@ class HipH264Pipeline : public EncodePipeline {
       }
       HIP_CHECK(hipEventRecord(batch_events_[b], stream_));
     }
+    // in-loop deblock of the current recon (within-slice edges only;
+    // idc=2 is signaled in the slice headers) before it becomes the
+    // reference frame
+    if (settings_.video_deblock)
+      h264gpu::launch_h264_deblock(d_curY_, d_curCb_, d_curCr_, ypitch_,
+                                   cpitch_, mbw_, n_jobs, d_jobs_[par],
+                                   d_levels_, d_meta_, stream_);
     const int copy_words = ent_copy_words_;
     if (!cpu_entropy_) {
       h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs,
@@ -676,6 +684,7 @@ class HipH264Pipeline : public EncodePipeline {
         p.height = o.h;
         p.qp = qp;
         p.idr = sj.idr;
+        p.deblock = settings_.video_deblock;
         p.frame_num = sj.frame_num;
         p.idr_pic_id = sj.idr_pic_id;
         o.rows.resize(p.n_mb_rows * segs_);

@@ -169,6 +169,9 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                value_range=(1, 2)),
     SettingDef("video_fullframe", bool, False,
                "Always encode the full frame (disable damage gating).", client=True),
+    SettingDef("video_deblock", bool, True,
+               "H.264 in-loop deblocking filter (within-slice edges).",
+               client=True),
     SettingDef("video_streaming_mode", bool, False,
                "Motion-optimized mode: bias rate control for smooth motion.", client=True),
     SettingDef("jpeg_quality", int, 80, "JPEG stripe quality.", value_range=(1, 100),

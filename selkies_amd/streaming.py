@@ -166,6 +166,7 @@ class StreamingService:
         cs.vbv_multiplier = s.video_vbv_multiplier
         cs.keyframe_interval_s = s.keyframe_interval_s
         cs.video_fullcolor = s.video_fullcolor
+        cs.video_deblock = s.video_deblock
         cs.video_fullframe = s.video_fullframe
         cs.capture_scale_div = s.capture_scale_div
         cs.pipeline_depth = s.video_pipeline_depth

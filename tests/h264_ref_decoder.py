@@ -357,6 +357,189 @@ def blk_xy(blk):
     return bx, by
 
 
+
+# ---- in-loop deblocking filter (§8.7) --------------------------------------
+# disable_deblocking_filter_idc == 2: slice-boundary edges are skipped, so
+# with one-slice-per-MB-row only row-internal edges filter.
+
+DB_ALPHA = [0]*16 + [4, 4, 5, 6, 7, 8, 9, 10, 12, 13, 15, 17, 20, 22, 25, 28,
+                     32, 36, 40, 45, 50, 56, 63, 71, 80, 90, 101, 113, 127,
+                     144, 162, 182, 203, 226, 255, 255]
+DB_BETA = [0]*16 + [2, 2, 2, 3, 3, 3, 3, 4, 4, 4, 6, 6, 7, 7, 8, 8,
+                    9, 9, 10, 10, 11, 11, 12, 12, 13, 13, 14, 14, 15, 15,
+                    16, 16, 17, 17, 18, 18]
+DB_TC0 = ([ (0,0,0) ]*17 + [(0,0,1)]*4 + [(0,1,1)]*2 + [(1,1,1)]*4 +
+          [(1,1,2)]*4 + [(1,2,3)]*2 + [(2,2,3),(2,2,4),(2,3,4),(2,3,4),
+           (3,3,5),(3,4,6),(3,4,6),(4,5,7),(4,5,8),(5,6,9),(6,7,10),
+           (6,8,11),(7,9,12),(8,10,13),(9,12,15),(10,13,17),(11,16,20),
+           (13,18,23),(14,20,25)])
+
+
+def _db_filter_lines(bs, alpha, beta, tc0, P, Q, chroma):
+    """Filter a batch of independent lines. P/Q: arrays [4][n] (or [2][n]
+    for chroma) of samples away from the edge; returns modified copies."""
+    p = [x.astype(np.int32) for x in P]
+    q = [x.astype(np.int32) for x in Q]
+    fl = ((np.abs(p[0] - q[0]) < alpha) & (np.abs(p[1] - p[0]) < beta)
+          & (np.abs(q[1] - q[0]) < beta))
+    if chroma:
+        if bs < 4:
+            tc = tc0 + 1
+            d = np.clip(((q[0] - p[0]) * 4 + (p[1] - q[1]) + 4) >> 3, -tc, tc)
+            p0n = np.clip(p[0] + d, 0, 255)
+            q0n = np.clip(q[0] - d, 0, 255)
+        else:
+            p0n = (2 * p[1] + p[0] + q[1] + 2) >> 2
+            q0n = (2 * q[1] + q[0] + p[1] + 2) >> 2
+        p[0] = np.where(fl, p0n, p[0])
+        q[0] = np.where(fl, q0n, q[0])
+        return p, q
+    ap = np.abs(p[2] - p[0]) < beta
+    aq = np.abs(q[2] - q[0]) < beta
+    if bs < 4:
+        tc = tc0 + ap.astype(np.int32) + aq.astype(np.int32)
+        d = np.clip(((q[0] - p[0]) * 4 + (p[1] - q[1]) + 4) >> 3, -tc, tc)
+        p0n = np.clip(p[0] + d, 0, 255)
+        q0n = np.clip(q[0] - d, 0, 255)
+        p1n = p[1] + np.clip((p[2] + ((p[0] + q[0] + 1) >> 1) - 2 * p[1])
+                             >> 1, -tc0, tc0)
+        q1n = q[1] + np.clip((q[2] + ((p[0] + q[0] + 1) >> 1) - 2 * q[1])
+                             >> 1, -tc0, tc0)
+        p[1] = np.where(fl & ap, p1n, p[1])
+        q[1] = np.where(fl & aq, q1n, q[1])
+        p[0] = np.where(fl, p0n, p[0])
+        q[0] = np.where(fl, q0n, q[0])
+        return p, q
+    small = np.abs(p[0] - q[0]) < ((alpha >> 2) + 2)
+    strong_p = fl & ap & small
+    strong_q = fl & aq & small
+    p0s = (p[2] + 2 * p[1] + 2 * p[0] + 2 * q[0] + q[1] + 4) >> 3
+    p1s = (p[2] + p[1] + p[0] + q[0] + 2) >> 2
+    p2s = (2 * p[3] + 3 * p[2] + p[1] + p[0] + q[0] + 4) >> 3
+    p0w = (2 * p[1] + p[0] + q[1] + 2) >> 2
+    q0s = (q[2] + 2 * q[1] + 2 * q[0] + 2 * p[0] + p[1] + 4) >> 3
+    q1s = (q[2] + q[1] + q[0] + p[0] + 2) >> 2
+    q2s = (2 * q[3] + 3 * q[2] + q[1] + q[0] + p[0] + 4) >> 3
+    q0w = (2 * q[1] + q[0] + p[1] + 2) >> 2
+    newp0 = np.where(strong_p, p0s, np.where(fl, p0w, p[0]))
+    newp1 = np.where(strong_p, p1s, p[1])
+    newp2 = np.where(strong_p, p2s, p[2])
+    newq0 = np.where(strong_q, q0s, np.where(fl, q0w, q[0]))
+    newq1 = np.where(strong_q, q1s, q[1])
+    newq2 = np.where(strong_q, q2s, q[2])
+    p[0], p[1], p[2] = newp0, newp1, newp2
+    q[0], q[1], q[2] = newq0, newq1, newq2
+    return p, q
+
+
+def _db_bs(mp, pblk, mq, qblk, mb_edge):
+    if mp["intra"] or mq["intra"]:
+        return 4 if mb_edge else 3
+    if ((mp["nz"] >> pblk) & 1) or ((mq["nz"] >> qblk) & 1):
+        return 2
+    if abs(mp["mv"][0] - mq["mv"][0]) >= 4 or             abs(mp["mv"][1] - mq["mv"][1]) >= 4:
+        return 1
+    return 0
+
+
+def deblock_segment_py(Y, Cb, Cr, mb_row, mbx0, infos, qp, qpc):
+    """Independent from-spec implementation of the segment deblock (the
+    encoder's deblock.h counterpart; equality asserted by the tests)."""
+    iA = min(max(qp, 0), 51)
+    alpha, beta = DB_ALPHA[iA], DB_BETA[iA]
+    iAc = min(max(qpc, 0), 51)
+    alpha_c, beta_c = DB_ALPHA[iAc], DB_BETA[iAc]
+    y0, cy0 = mb_row * 16, mb_row * 8
+    for m, cur in enumerate(infos):
+        left = infos[m - 1] if m > 0 else None
+        x0 = (mbx0 + m) * 16
+        # vertical luma edges
+        for e in range(4):
+            if e == 0 and left is None:
+                continue
+            ex = x0 + e * 4
+            for br4 in range(4):       # per 4-row group (bS can differ)
+                rows = slice(y0 + br4 * 4, y0 + br4 * 4 + 4)
+                pblk = br4 * 4 + (3 if e == 0 else e - 1)
+                qblk = br4 * 4 + e
+                bs = _db_bs(left if e == 0 else cur, pblk, cur, qblk,
+                            e == 0)
+                if bs == 0:
+                    continue
+                tc0 = DB_TC0[iA][bs - 1] if bs < 4 else 0
+                P = [Y[rows, ex - 1 - i].copy() for i in range(4)]
+                Q = [Y[rows, ex + i].copy() for i in range(4)]
+                P, Q = _db_filter_lines(bs, alpha, beta, tc0, P, Q, False)
+                for i in range(3):
+                    Y[rows, ex - 1 - i] = P[i]
+                    Y[rows, ex + i] = Q[i]
+        # horizontal luma edges (y=4,8,12)
+        for e in range(1, 4):
+            ey = y0 + e * 4
+            for bc in range(4):
+                cols = slice(x0 + bc * 4, x0 + bc * 4 + 4)
+                pblk = (e - 1) * 4 + bc
+                qblk = e * 4 + bc
+                bs = _db_bs(cur, pblk, cur, qblk, False)
+                if bs == 0:
+                    continue
+                tc0 = DB_TC0[iA][bs - 1] if bs < 4 else 0
+                P = [Y[ey - 1 - i, cols].copy() for i in range(4)]
+                Q = [Y[ey + i, cols].copy() for i in range(4)]
+                P, Q = _db_filter_lines(bs, alpha, beta, tc0, P, Q, False)
+                for i in range(3):
+                    Y[ey - 1 - i, cols] = P[i]
+                    Y[ey + i, cols] = Q[i]
+        # chroma
+        cx0 = (mbx0 + m) * 8
+        for C in (Cb, Cr):
+            for e in range(2):         # vertical, chroma x = 0, 4
+                if e == 0 and left is None:
+                    continue
+                ex = cx0 + e * 4
+                for cy in range(2):    # 4-chroma-row groups -> luma rows
+                    rows = slice(cy0 + cy * 4, cy0 + cy * 4 + 4)
+                    # luma block row for chroma rows cy*4..cy*4+3 varies
+                    # per line; split into 2-line halves (luma rows 0-3 /
+                    # 4-7 per half)
+                    for half in range(2):
+                        r2 = slice(cy0 + cy * 4 + half * 2,
+                                   cy0 + cy * 4 + half * 2 + 2)
+                        lum_brow = cy * 2 + half
+                        pblk = lum_brow * 4 + (3 if e == 0 else 1)
+                        qblk = lum_brow * 4 + (0 if e == 0 else 2)
+                        bs = _db_bs(left if e == 0 else cur, pblk, cur,
+                                    qblk, e == 0)
+                        if bs == 0:
+                            continue
+                        tc0 = DB_TC0[iAc][bs - 1] if bs < 4 else 0
+                        P = [C[r2, ex - 1 - i].copy() for i in range(2)]
+                        Q = [C[r2, ex + i].copy() for i in range(2)]
+                        P, Q = _db_filter_lines(bs, alpha_c, beta_c, tc0,
+                                                P, Q, True)
+                        C[r2, ex - 1] = P[0]
+                        C[r2, ex] = Q[0]
+            # horizontal, chroma y = 4
+            ey = cy0 + 4
+            for bc in range(2):
+                for half in range(2):
+                    c2 = slice(cx0 + bc * 4 + half * 2,
+                               cx0 + bc * 4 + half * 2 + 2)
+                    lum_bcol = bc * 2 + half
+                    pblk = 1 * 4 + lum_bcol
+                    qblk = 2 * 4 + lum_bcol
+                    bs = _db_bs(cur, pblk, cur, qblk, False)
+                    if bs == 0:
+                        continue
+                    tc0 = DB_TC0[iAc][bs - 1] if bs < 4 else 0
+                    P = [C[ey - 1 - i, c2].copy() for i in range(2)]
+                    Q = [C[ey + i, c2].copy() for i in range(2)]
+                    P, Q = _db_filter_lines(bs, alpha_c, beta_c, tc0, P, Q,
+                                            True)
+                    C[ey - 1, c2] = P[0]
+                    C[ey, c2] = Q[0]
+
+
 class Decoder:
     def __init__(self):
         self.sps = None
@@ -432,9 +615,11 @@ class Decoder:
         else:
             assert br.u(1) == 0
         qp = self.pps["pic_init_qp"] + br.se()
+        dbf = 1
         if self.pps["deblocking_control"]:
             dbf = br.ue()
-            assert dbf == 1, "decoder subset requires deblocking disabled"
+            assert dbf in (1, 2), \
+                "decoder subset: deblocking off or within-slice only"
 
         if idr and first_mb == 0:
             self.y = np.zeros((H, W), np.int32)
@@ -454,6 +639,7 @@ class Decoder:
             "left_cr_nc": [0] * 2,
         }
         mbx = first_mb % mbw        # slices may start mid-row (segments)
+        db_infos = []               # per-MB info for the deblocking filter
         skip_left = 0
         if is_p:
             skip_left = br.ue()
@@ -463,6 +649,7 @@ class Decoder:
                 ctx.update(left_avail=True, left_inter=True, left_mv=(0, 0),
                            left_luma_nc=[0] * 4, left_cb_nc=[0] * 2,
                            left_cr_nc=[0] * 2)
+                db_infos.append({"intra": 0, "mv": (0, 0), "nz": 0})
                 skip_left -= 1
                 mbx += 1
                 continue
@@ -471,14 +658,18 @@ class Decoder:
             mb_type = br.ue()
             if is_p and mb_type < 5:
                 assert mb_type == 0, f"subset: P mb_type {mb_type}"
-                self.decode_p16(br, mbx, mb_row, ctx, qp)
+                db_infos.append(self.decode_p16(br, mbx, mb_row, ctx, qp))
             else:
                 it = mb_type - 5 if is_p else mb_type
                 assert 1 <= it <= 24, f"subset: mb_type {mb_type}"
                 self.decode_i16(br, mbx, mb_row, it, qp, ctx)
+                db_infos.append({"intra": 1, "mv": (0, 0), "nz": 0})
             mbx += 1
             if is_p and mbx < mbw and br.more_rbsp_data():
                 skip_left = br.ue()
+        if dbf == 2 and db_infos:
+            deblock_segment_py(self.y, self.cb, self.cr, mb_row,
+                               first_mb % mbw, db_infos, qp, chroma_qp(qp))
         return mb_row, mbx - first_mb % mbw
 
     def decode_skip(self, mbx, mby):
@@ -577,10 +768,16 @@ class Decoder:
         self.y[y0:y0 + 16, x0:x0 + 16] = np.clip(pred_y, 0, 255)
         self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cb, 0, 255)
         self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cr, 0, 255)
+        nz = 0
+        for by in range(4):
+            for bx in range(4):
+                if new_luma_nc[by, bx]:
+                    nz |= 1 << (by * 4 + bx)
         ctx.update(left_avail=True, left_inter=True, left_mv=(mvx, mvy),
                    left_luma_nc=[int(new_luma_nc[by, 3]) for by in range(4)],
                    left_cb_nc=[int(new_cb_nc[cy, 1]) for cy in range(2)],
                    left_cr_nc=[int(new_cr_nc[cy, 1]) for cy in range(2)])
+        return {"intra": 0, "mv": (mvx, mvy), "nz": nz}
 
     def decode_i16(self, br, mbx, mby, i16_type, qp, ctx):
         t = i16_type - 1

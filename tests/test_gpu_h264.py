@@ -167,8 +167,9 @@ def test_gpu_mfma_motion_search_exact():
             self.mvs = {}
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs[(mbx, mby)] = tuple(ctx["left_mv"])
+            return info
 
         def decode_skip(self, mbx, mby):
             super().decode_skip(mbx, mby)
@@ -267,8 +268,9 @@ def test_gpu_halfpel_and_odd_mvs():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     all_mvs = []
     for y0, stream in rows.items():
@@ -314,8 +316,9 @@ def test_gpu_quarterpel_mvs():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     all_mvs = []
     for y0, stream in rows.items():
@@ -396,8 +399,9 @@ def test_gpu_fast_scroll_motion_lock():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     late_mvs = []
     for y0, stream in rows.items():
@@ -616,3 +620,30 @@ def test_gpu_pipelined_resolution_change():
     rows = reassemble(out)
     for y, stream in rows.items():
         assert len(Decoder().decode(stream)) == len(frames_b)
+
+
+def test_gpu_deblock_recon_matches_decoder():
+    """The GPU deblock kernel's reconstruction must equal the from-spec
+    decoder's (which filters per §8.7 from the idc=2 slice headers) —
+    the strongest deblock check: any bS/table/order divergence breaks
+    byte equality, and the P-frame chain amplifies drift."""
+    require_gpu()
+    w, h, n = 320, 192, 5
+    frames = make_frames(w, h, n)
+    out, dump = _native._pipeline_encode("gpu", frames, w, h, 30, 64, 1,
+                                         True)
+    stream = b""
+    for fr in out:
+        for data, y, hgt, key in fr:
+            stream += bytes(data)
+    # reassemble rows separately (independent bitstreams)
+    rows = reassemble(out)
+    ypitch = dump["ypitch"]
+    ry = np.frombuffer(dump["y"], np.uint8).reshape(-1, ypitch)
+    for y0, s in rows.items():
+        decoded = Decoder().decode(s)
+        assert len(decoded) == n
+        dy = decoded[-1][0]
+        gy = ry[y0:y0 + dy.shape[0], :w]
+        assert np.array_equal(dy, gy), (
+            f"stripe y={y0}: decoder recon != GPU deblocked recon")

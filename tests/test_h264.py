@@ -230,8 +230,9 @@ def test_halfpel_and_odd_integer_mvs():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     enc = hipflux.H264Encoder(w, h)
     stream = b""
@@ -276,8 +277,9 @@ def test_quarterpel_mvs():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     enc = hipflux.H264Encoder(w, h)
     stream = b""
@@ -323,8 +325,9 @@ def test_fast_scroll_motion_lock():
             self.mvs = []
 
         def decode_p16(self, br, mbx, mby, ctx, qp):
-            super().decode_p16(br, mbx, mby, ctx, qp)
+            info = super().decode_p16(br, mbx, mby, ctx, qp)
             self.mvs.append(tuple(ctx["left_mv"]))
+            return info
 
     enc = hipflux.H264Encoder(w, h)
     stream = b""
@@ -346,3 +349,46 @@ def test_fast_scroll_motion_lock():
     assert len(locked) > len(d.mvs) * 0.4, \
         f"scroll not locked: {len(locked)}/{len(d.mvs)} at -48"
     assert max(sizes[2:]) < sizes[0] * 0.6, f"P frames too large: {sizes}"
+
+
+def test_deblock_changes_recon_and_signals_idc2():
+    """The in-loop deblocking filter must actually fire (recon differs
+    from the unfiltered encode) and be signaled via
+    disable_deblocking_filter_idc = 2; both variants stay bit-exact
+    under the from-spec decoder."""
+    w, h = 128, 64
+    img = gradient_frame(w, h)
+    # blocky content: coarse quantization creates block-edge steps
+    on = hipflux.H264Encoder(w, h)
+    off = hipflux.H264Encoder(w, h, deblock=False)
+    r_on = on.encode(img.tobytes(), qp=40, idr=True)
+    r_off = off.encode(img.tobytes(), qp=40, idr=True)
+    y_on, _, _ = recon_planes(on, w, h)
+    y_off, _, _ = recon_planes(off, w, h)
+    assert not np.array_equal(y_on, y_off), "deblock had no effect"
+    # decoder agrees with both (parses idc and filters only when 2)
+    dy_on = Decoder().decode(r_on["data"])[0][0]
+    dy_off = Decoder().decode(r_off["data"])[0][0]
+    assert np.array_equal(dy_on, y_on)
+    assert np.array_equal(dy_off, y_off)
+
+
+def test_deblock_p_frame_chain_bit_exact():
+    """Deblocked recon feeds the reference chain: a multi-frame P
+    sequence with motion must stay bit-exact (decoder MC reads the
+    deblocked reference exactly like the encoder)."""
+    w, h = 128, 64
+    rng = np.random.default_rng(17)
+    base = rng.integers(0, 256, (h + 32, w, 4), dtype=np.uint8)
+    enc = hipflux.H264Encoder(w, h)
+    stream = b""
+    recons = []
+    for i in range(4):
+        img = np.ascontiguousarray(base[i * 8:i * 8 + h])
+        r = enc.encode(img.tobytes(), qp=32, idr=(i == 0))
+        stream += r["data"]
+        recons.append(recon_planes(enc, w, h)[0].copy())
+    frames = Decoder().decode(stream)
+    assert len(frames) == 4
+    for i, (dy, _, _) in enumerate(frames):
+        assert np.array_equal(dy, recons[i]), f"frame {i} recon mismatch"
