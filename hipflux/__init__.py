@@ -65,3 +65,24 @@ class Stripe:
 
 def make_stripe(*args) -> Stripe:
     return Stripe(*args)
+
+# ---- Wayland surface (pixelflux module-function contract; SURVEY.md
+# §2.3: ensure_wayland_display / get_wayland_display_name). The
+# compositor is pure control-plane Python (selkies_amd.wayland).
+def ensure_wayland_display(name="selkies-wl-0", width=1920, height=1080):
+    from selkies_amd.wayland import ensure_wayland_display as _e
+    return _e(name, width, height)
+
+
+def get_wayland_display_name():
+    from selkies_amd.wayland import get_wayland_display_name as _g
+    return _g()
+
+
+def probe_wayland_gpu():
+    """Headless compositor needs no GBM/EGL: software path always works;
+    report the GPU count so entrypoints can prefer the HIP encoder."""
+    try:
+        return {"ok": True, "gpus": hip_device_count()}
+    except Exception:
+        return {"ok": True, "gpus": 0}

@@ -136,8 +136,63 @@ class XTestBackend(InputBackend):
             self._dpy = None
 
 
+class WaylandBackend(InputBackend):
+    """Injects into the in-tree headless Wayland compositor's seat
+    (reference ladder: compositor injection before XTEST,
+    input_handler.py:20-47). Keysyms map to evdev codes via the X11
+    keysym tables (evdev code = X keycode - 8 in the compositor's
+    keymap)."""
+
+    def __init__(self):
+        from .wayland import get_compositor
+        comp = get_compositor()
+        if comp is None:
+            raise RuntimeError("no wayland compositor running")
+        self.comp = comp
+        from .wayland.compositor import KEYSYM_TO_XKEYCODE
+        self._map = KEYSYM_TO_XKEYCODE
+        self._xy = [0, 0]
+
+    def key(self, keysym, down):
+        code = self._map.get(keysym)
+        if code is not None:
+            self.comp.inject_key(code - 8, down)   # evdev = X keycode - 8
+
+    def mouse_move(self, x, y):
+        self._xy = [x, y]
+        self.comp.inject_mouse_move(float(x), float(y))
+
+    def mouse_move_rel(self, dx, dy):
+        self._xy = [self._xy[0] + dx, self._xy[1] + dy]
+        self.comp.inject_mouse_move(float(self._xy[0]),
+                                    float(self._xy[1]))
+
+    def mouse_button(self, button, down):
+        # X buttons: 1..3 -> BTN_LEFT/MIDDLE/RIGHT; 4..7 -> scroll
+        if button in (BTN_SCROLL_UP, BTN_SCROLL_DOWN):
+            if down:
+                self.comp.inject_mouse_scroll(
+                    0, -15.0 if button == BTN_SCROLL_UP else 15.0)
+        elif button in (BTN_SCROLL_LEFT, BTN_SCROLL_RIGHT):
+            if down:
+                self.comp.inject_mouse_scroll(
+                    -15.0 if button == BTN_SCROLL_LEFT else 15.0, 0)
+        else:
+            evbtn = {1: 0x110, 2: 0x112, 3: 0x111}.get(button, 0x110)
+            self.comp.inject_mouse_button(evbtn, down)
+
+    def close(self):
+        pass
+
+
 def make_backend(display: Optional[str] = None) -> InputBackend:
-    """XTEST if a display is reachable, else recording."""
+    """Wayland seat if the in-tree compositor runs, else XTEST if a
+    display is reachable, else recording."""
+    if display and display.startswith("wayland"):
+        try:
+            return WaylandBackend()
+        except Exception as exc:
+            logger.info("wayland backend unavailable (%s)", exc)
     try:
         return XTestBackend(display)
     except Exception as exc:
