@@ -56,6 +56,7 @@ struct udev_device {
   int refs;
   int pad;              /* 0..3 */
   int is_parent;        /* the "input device" parent w/ properties */
+  int is_event;         /* evdev node (event1000+pad) instead of jsN */
   char syspath[192];
   char action[16];
 };
@@ -106,6 +107,15 @@ static struct udev_device* make_device(struct udev* u, int pad,
   return d;
 }
 
+static struct udev_device* make_event_device(struct udev* u, int pad) {
+  struct udev_device* d = make_device(u, pad, 0);
+  d->is_event = 1;
+  snprintf(d->syspath, sizeof(d->syspath),
+           "/sys/devices/virtual/input/selkies-input%d/event%d", pad,
+           1000 + pad);
+  return d;
+}
+
 struct udev_device* udev_device_new_from_syspath(struct udev* u,
                                                  const char* syspath) {
   int pad;
@@ -113,6 +123,10 @@ struct udev_device* udev_device_new_from_syspath(struct udev* u,
   const char* js = strstr(syspath, "/js");
   if (js && sscanf(js, "/js%d", &pad) == 1 && pad >= 0 && pad < MAX_PADS)
     return make_device(u, pad, 0);
+  const char* ev = strstr(syspath, "/event");
+  if (ev && sscanf(ev, "/event%d", &pad) == 1 && pad >= 1000 &&
+      pad < 1000 + MAX_PADS)
+    return make_event_device(u, pad - 1000);
   if (sscanf(syspath, "/sys/devices/virtual/input/selkies-input%d", &pad)
           == 1 && pad >= 0 && pad < MAX_PADS)
     return make_device(u, pad, 1);
@@ -137,6 +151,8 @@ const char* udev_device_get_sysname(struct udev_device* d) {
   static __thread char name[32];
   if (d->is_parent)
     snprintf(name, sizeof(name), "selkies-input%d", d->pad);
+  else if (d->is_event)
+    snprintf(name, sizeof(name), "event%d", 1000 + d->pad);
   else
     snprintf(name, sizeof(name), "js%d", d->pad);
   return name;
@@ -152,7 +168,10 @@ const char* udev_device_get_devtype(struct udev_device* d) {
 const char* udev_device_get_devnode(struct udev_device* d) {
   if (!d || d->is_parent) return NULL;
   static __thread char node[32];
-  snprintf(node, sizeof(node), "/dev/input/js%d", d->pad);
+  if (d->is_event)
+    snprintf(node, sizeof(node), "/dev/input/event%d", 1000 + d->pad);
+  else
+    snprintf(node, sizeof(node), "/dev/input/js%d", d->pad);
   return node;
 }
 const char* udev_device_get_action(struct udev_device* d) {
@@ -263,6 +282,13 @@ int udev_enumerate_scan_devices(struct udev_enumerate* e) {
     struct udev_list_entry* ent = calloc(1, sizeof(*ent));
     snprintf(ent->name, sizeof(ent->name),
              "/sys/devices/virtual/input/selkies-input%d/js%d", i, i);
+    *tail = ent;
+    tail = &ent->next;
+    /* the evdev sibling (served by the interposer's event surface) */
+    ent = calloc(1, sizeof(*ent));
+    snprintf(ent->name, sizeof(ent->name),
+             "/sys/devices/virtual/input/selkies-input%d/event%d", i,
+             1000 + i);
     *tail = ent;
     tail = &ent->next;
   }

@@ -54,8 +54,10 @@ def test_enumerate_and_device(libudev, tmp_path, monkeypatch):
     (tmp_path / "selkies_js2.sock").touch()
     u = libudev.udev_new()
     names = enum_names(libudev, u)
-    assert len(names) == 2 and names[0].endswith("js0") and \
-        names[1].endswith("js2")
+    # each present pad enumerates a joydev node AND its evdev sibling
+    assert len(names) == 4
+    assert names[0].endswith("js0") and names[1].endswith("event1000")
+    assert names[2].endswith("js2") and names[3].endswith("event1002")
 
     d = libudev.udev_device_new_from_syspath(
         ctypes.c_void_p(u), names[0].encode())
@@ -91,3 +93,17 @@ def test_monitor_hotplug(libudev, tmp_path, monkeypatch):
     assert r
     d = libudev.udev_monitor_receive_device(ctypes.c_void_p(m))
     assert libudev.udev_device_get_action(ctypes.c_void_p(d)) == b"remove"
+
+
+def test_event_device_from_syspath(libudev):
+    lib = libudev
+    lib.udev_device_get_sysname.restype = ctypes.c_char_p
+    u = lib.udev_new()
+    d = lib.udev_device_new_from_syspath(
+        ctypes.c_void_p(u),
+        b"/sys/devices/virtual/input/selkies-input1/event1001")
+    assert d
+    assert lib.udev_device_get_devnode(
+        ctypes.c_void_p(d)) == b"/dev/input/event1001"
+    assert lib.udev_device_get_sysname(
+        ctypes.c_void_p(d)) == b"event1001"
