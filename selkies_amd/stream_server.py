@@ -95,10 +95,16 @@ class CentralizedStreamServer:
         web_dir = os.path.join(os.path.dirname(__file__), "web")
         if os.path.isdir(web_dir):
             app.router.add_get("/", self._index)
+            app.router.add_get("/dashboard", self._dashboard)
             app.router.add_static("/static", web_dir)
 
     async def _index(self, request):
         path = os.path.join(os.path.dirname(__file__), "web", "index.html")
+        return web.FileResponse(path)
+
+    async def _dashboard(self, request):
+        path = os.path.join(os.path.dirname(__file__), "web",
+                            "dashboard.html")
         return web.FileResponse(path)
 
     async def _ws_entry(self, request):

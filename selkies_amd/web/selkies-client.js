@@ -111,7 +111,8 @@ function onBinary(buf) {
 /* ---------------- audio playback ----------------
  * wire: [0x01, n_red] + n_red x (u16 len + redundant payload) + primary.
  * Payload is s16le PCM (WS is reliable, so RED payloads are skipped). */
-let audioCtx = null, audioTime = 0;
+let audioCtx = null, audioTime = 0, gainNode = null, masterVolume = 1,
+    audioMuted = false;
 const AUDIO_RATE = 48000, AUDIO_CH = 2;
 
 function playAudioFrame(d) {
@@ -124,7 +125,12 @@ function playAudioFrame(d) {
   const pcm = new Int16Array(d.buffer, d.byteOffset + off,
                              (d.length - off) >> 1);
   if (!audioCtx) {
-    try { audioCtx = new AudioContext({ sampleRate: AUDIO_RATE }); }
+    try {
+      audioCtx = new AudioContext({ sampleRate: AUDIO_RATE });
+      gainNode = audioCtx.createGain();
+      gainNode.gain.value = audioMuted ? 0 : masterVolume;
+      gainNode.connect(audioCtx.destination);
+    }
     catch (e) { return; }
   }
   const framesN = pcm.length / AUDIO_CH;
@@ -135,7 +141,7 @@ function playAudioFrame(d) {
   }
   const src = audioCtx.createBufferSource();
   src.buffer = buf;
-  src.connect(audioCtx.destination);
+  src.connect(gainNode || audioCtx.destination);
   const now = audioCtx.currentTime;
   if (audioTime < now + 0.02) audioTime = now + 0.04;  // jitter buffer
   src.start(audioTime);
@@ -496,3 +502,62 @@ if (params.get("transport") === "webrtc") {
 }
 connect();
 canvas.focus();
+
+/* ---- dashboard postMessage bridge (contract:
+   selkies_amd/web/postmessage-bridge.js) ---- */
+let lastServerStats = null, lastServerSettings = null;
+if (window.installPostMessageBridge) {
+  const bridge = window.installPostMessageBridge({
+    send: send,
+    b64encode: (t) => btoa(unescape(encodeURIComponent(t))),
+    getStats: () => ({
+      clientFps: Number(
+          (frameCount / Math.max(0.001,
+              (performance.now() - lastStats) / 1000)).toFixed(1)),
+      videoBuffer: h264Rows.size,
+      audioBuffer: 0,
+      encoderName: (lastServerSettings &&
+                    lastServerSettings.encoder &&
+                    lastServerSettings.encoder.value) || "h264enc-striped",
+      server: lastServerStats,
+      isVideoPipelineActive: !!(ws && ws.readyState === 1),
+      isAudioPipelineActive: !!audioCtx,
+      isMicrophoneActive: false,
+    }),
+    setVolume: (v) => {
+      masterVolume = Math.max(0, Math.min(1, v));
+      if (gainNode && !audioMuted) gainNode.gain.value = masterVolume;
+    },
+    setMute: (m) => {
+      audioMuted = m;
+      if (gainNode) gainNode.gain.value = m ? 0 : masterVolume;
+    },
+    setGamepadEnabled: (on) => { window.skGamepadEnabled = on; },
+    resizeToWindow: () => {
+      const el = document.getElementById("stage");
+      send(`r,${el.clientWidth}x${el.clientHeight}`);
+    },
+    showVirtualKeyboard: () => canvas.focus(),
+    setRenderFlag: (k, v) => {
+      if (k === "setAntiAliasing")
+        canvas.style.imageRendering = v ? "auto" : "pixelated";
+    },
+  }, window);
+  const prevSettings = window.skOnSettings;
+  window.skOnSettings = (p) => {
+    lastServerSettings = p;
+    bridge.onServerSettings(p);
+    if (prevSettings) prevSettings(p);
+  };
+  const prevStats = window.skOnStats;
+  window.skOnStats = (st) => {
+    lastServerStats = st;
+    bridge.onStats(st);
+    if (prevStats) prevStats(st);
+  };
+  const prevClip = window.skOnClipboard;
+  window.skOnClipboard = (t) => {
+    bridge.onClipboard(t);
+    if (prevClip) prevClip(t);
+  };
+}

@@ -376,3 +376,27 @@ def test_glass_to_glass_stats():
             await runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_dashboard_page_served():
+    async def main():
+        server = make_server()
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                r = await sess.get(f"http://127.0.0.1:{port}/dashboard")
+                assert r.status == 200
+                body = await r.text()
+                assert "dashboard-app.js" in body
+                r = await sess.get(
+                    f"http://127.0.0.1:{port}/static/dashboard-app.js")
+                assert r.status == 200
+                assert "postMessage" in await r.text()
+                r = await sess.get(
+                    f"http://127.0.0.1:{port}/static/postmessage-bridge.js")
+                assert r.status == 200
+        finally:
+            server.streaming.stop_capture()
+            await runner.cleanup()
+
+    asyncio.run(main())
