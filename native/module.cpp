@@ -11,6 +11,8 @@
 #include "cpu/h264/encoder.h"
 #include "cpu/hevc/cabac.h"
 #include "cpu/hevc/encoder.h"
+#include "cpu/opus/celt.h"
+#include "cpu/opus/range_coder.h"
 #include "cpu/jpeg_enc.h"
 #include "engine.h"
 #include <hip/hip_runtime.h>
@@ -337,6 +339,58 @@ PYBIND11_MODULE(_native, m) {
             tail.bits, tail.nbits);
       },
       py::arg("ops"), py::arg("qp") = 30);
+
+  // ---- Opus-framed CELT-class audio encoder -----------------------------
+  py::class_<opus::CeltEncoder>(m, "OpusEncoder")
+      .def(py::init<int>(), py::arg("bitrate_bps") = 96000)
+      .def("set_bitrate", &opus::CeltEncoder::set_bitrate)
+      .def(
+          "encode",
+          [](opus::CeltEncoder& e, py::buffer pcm, int channels) {
+            py::buffer_info info = pcm.request();
+            if (info.size <
+                static_cast<ssize_t>(opus::kFrameSamples) * channels)
+              throw std::runtime_error("need 960 frames of s16 PCM");
+            std::vector<uint8_t> out;
+            {
+              py::gil_scoped_release rel;
+              out = e.encode_frame(
+                  static_cast<const int16_t*>(info.ptr), channels);
+            }
+            return py::bytes(reinterpret_cast<const char*>(out.data()),
+                             out.size());
+          },
+          py::arg("pcm"), py::arg("channels") = 2);
+
+  // range-coder pair-fuzz hook (tests/test_opus.py): ops =
+  // (kind, a, b): 0 = bit_logp(bit=a, logp=b), 1 = uint(fl=a, ft=b),
+  // 2 = raw bits(fl=a, nbits=b), 3 = icdf(sym=a over a 4-entry table)
+  m.def("_opus_range_encode",
+        [](const std::vector<std::tuple<int, int, int>>& ops,
+           size_t capacity) {
+          opus::RangeEncoder ec(capacity);
+          static const uint8_t icdf4[4] = {200, 120, 40, 0};
+          for (const auto& op : ops) {
+            int kind = std::get<0>(op), a = std::get<1>(op),
+                b = std::get<2>(op);
+            if (kind == 0)
+              ec.enc_bit_logp(a, static_cast<unsigned>(b));
+            else if (kind == 1)
+              ec.enc_uint(static_cast<uint32_t>(a),
+                          static_cast<uint32_t>(b));
+            else if (kind == 2)
+              ec.enc_bits(static_cast<uint32_t>(a),
+                          static_cast<unsigned>(b));
+            else
+              ec.enc_icdf(a % 4, icdf4, 8);
+          }
+          ec.done();
+          auto s = ec.stream();
+          return py::make_tuple(
+              py::bytes(reinterpret_cast<const char*>(s.data()),
+                        s.size()),
+              ec.error());
+        });
 
   // ---- RCCL tile-parallel collective layer (BASELINE config 5) ----------
   py::class_<TileComm>(m, "TileComm")
