@@ -1,5 +1,7 @@
 #include "audio.h"
 
+#include "cpu/opus/celt.h"
+
 #include <chrono>
 #include <cmath>
 #include <cstdio>
@@ -37,6 +39,10 @@ void AudioCapture::run() {
     src_file = std::fopen(settings_.device_name.c_str() + 5, "rb");
 
   std::vector<int16_t> pcm(static_cast<size_t>(frame_samples) * ch);
+  const bool use_opus = settings_.codec != "pcm" &&
+                        rate == 48000 && frame_samples == 960;
+  opus::CeltEncoder opus_enc(bitrate_.load());
+  int last_bitrate = bitrate_.load();
   // history for RED redundancy
   std::vector<std::vector<uint8_t>> history;
   std::vector<uint8_t> wire;
@@ -71,9 +77,20 @@ void AudioCapture::run() {
       phase += static_cast<double>(frame_samples) / rate;
     }
 
-    // --- "encode": PCM passthrough (pluggable codec stage)
-    std::vector<uint8_t> payload(frame_bytes);
-    std::memcpy(payload.data(), pcm.data(), frame_bytes);
+    // --- encode: CELT-class Opus framing by default (native/cpu/opus),
+    // raw PCM passthrough when codec == "pcm"
+    std::vector<uint8_t> payload;
+    if (use_opus) {
+      int want = bitrate_.load();
+      if (want != last_bitrate) {
+        opus_enc.set_bitrate(want);      // live update_audio_bitrate
+        last_bitrate = want;
+      }
+      payload = opus_enc.encode_frame(pcm.data(), ch);
+    } else {
+      payload.resize(frame_bytes);
+      std::memcpy(payload.data(), pcm.data(), frame_bytes);
+    }
 
     // --- wire frame: [0x01, n_red][u16 len]*red oldest..newest + payloads
     wire.clear();

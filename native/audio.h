@@ -27,7 +27,8 @@ struct AudioCaptureSettings {
   std::string device_name = "synthetic";  // synthetic | silence | file:<path>
   int sample_rate = 48000;
   int channels = 2;                        // 1..6
-  int opus_bitrate = 128000;               // advisory (PCM ignores)
+  std::string codec = "opus";              // "opus" (CELT-class) | "pcm"
+  int opus_bitrate = 128000;
   int frame_duration_ms = 20;
   int red_distance = 0;                    // redundant previous frames
   bool omit_audio_header = false;

@@ -725,6 +725,7 @@ PYBIND11_MODULE(_native, m) {
       .def_readwrite("device_name", &AudioCaptureSettings::device_name)
       .def_readwrite("sample_rate", &AudioCaptureSettings::sample_rate)
       .def_readwrite("channels", &AudioCaptureSettings::channels)
+      .def_readwrite("codec", &AudioCaptureSettings::codec)
       .def_readwrite("opus_bitrate", &AudioCaptureSettings::opus_bitrate)
       .def_readwrite("frame_duration_ms",
                      &AudioCaptureSettings::frame_duration_ms)

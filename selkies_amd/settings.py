@@ -202,6 +202,9 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     # ---- audio ----
     SettingDef("enable_audio", bool, True, "Capture + stream audio.", client=True),
     SettingDef("audio_device", str, "auto", "Audio source (auto/synthetic/none)."),
+    SettingDef("audio_codec", str, "opus",
+               "Audio codec: opus (CELT-class, in-tree) or pcm.",
+               allowed=("opus", "pcm"), client=True),
     SettingDef("audio_bitrate", int, 128000, "Audio codec bitrate.",
                value_range=(16000, 512000), client=True),
     SettingDef("audio_channels", int, 2, "Channel count.", value_range=(1, 6)),

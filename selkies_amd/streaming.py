@@ -252,6 +252,7 @@ class StreamingService:
         s.device_name = self.settings.audio_device if \
             self.settings.audio_device != "auto" else "synthetic"
         s.channels = self.settings.audio_channels
+        s.codec = self.settings.audio_codec
         s.opus_bitrate = self.settings.audio_bitrate
         s.frame_duration_ms = self.settings.audio_frame_duration_ms
         s.red_distance = self.settings.audio_red_distance
