@@ -110,10 +110,35 @@ function onBinary(buf) {
 
 /* ---------------- audio playback ----------------
  * wire: [0x01, n_red] + n_red x (u16 len + redundant payload) + primary.
- * Payload is s16le PCM (WS is reliable, so RED payloads are skipped). */
+ * Payload is an Opus (CELT-class) packet by default — decoded by
+ * opus-decoder.js (SkOpus) — or s16le PCM when audio_codec=pcm. WS is
+ * reliable, so RED payloads are skipped. */
 let audioCtx = null, audioTime = 0, gainNode = null, masterVolume = 1,
     audioMuted = false;
 const AUDIO_RATE = 48000, AUDIO_CH = 2;
+let audioCodec = null;                 /* null = sniff until SETTINGS */
+let opusDec = null;
+
+function ensureAudioCtx() {
+  if (audioCtx) return true;
+  try {
+    audioCtx = new AudioContext({ sampleRate: AUDIO_RATE });
+    gainNode = audioCtx.createGain();
+    gainNode.gain.value = audioMuted ? 0 : masterVolume;
+    gainNode.connect(audioCtx.destination);
+    return true;
+  } catch (e) { return false; }
+}
+
+function schedulePlay(buf) {
+  const src = audioCtx.createBufferSource();
+  src.buffer = buf;
+  src.connect(gainNode || audioCtx.destination);
+  const now = audioCtx.currentTime;
+  if (audioTime < now + 0.02) audioTime = now + 0.04;  // jitter buffer
+  src.start(audioTime);
+  audioTime += buf.length / AUDIO_RATE;
+}
 
 function playAudioFrame(d) {
   let off = 2;
@@ -122,30 +147,32 @@ function playAudioFrame(d) {
     const len = (d[off] << 8) | d[off + 1];
     off += 2 + len;
   }
-  const pcm = new Int16Array(d.buffer, d.byteOffset + off,
-                             (d.length - off) >> 1);
-  if (!audioCtx) {
+  if (!ensureAudioCtx()) return;
+  const payload = d.subarray(off);
+  /* codec: explicit from SETTINGS_PAYLOAD, else sniff — a 20 ms PCM
+     frame is >= 1920 B while CBR Opus at <= 500 kb/s is < 1300 B */
+  const isOpus = audioCodec === "opus" ||
+      (audioCodec === null && payload.length < 1500 &&
+       (payload[0] >> 3) === 31);
+  if (isOpus && typeof SkOpus !== "undefined") {
     try {
-      audioCtx = new AudioContext({ sampleRate: AUDIO_RATE });
-      gainNode = audioCtx.createGain();
-      gainNode.gain.value = audioMuted ? 0 : masterVolume;
-      gainNode.connect(audioCtx.destination);
-    }
-    catch (e) { return; }
+      if (!opusDec) opusDec = new SkOpus.OpusDecoder();
+      const mono = opusDec.decodePacket(payload);
+      const buf = audioCtx.createBuffer(1, mono.length, AUDIO_RATE);
+      buf.getChannelData(0).set(mono);
+      schedulePlay(buf);
+    } catch (e) { console.warn("opus decode", e); }
+    return;
   }
+  const pcm = new Int16Array(payload.buffer, payload.byteOffset,
+                             payload.length >> 1);
   const framesN = pcm.length / AUDIO_CH;
   const buf = audioCtx.createBuffer(AUDIO_CH, framesN, AUDIO_RATE);
   for (let c = 0; c < AUDIO_CH; c++) {
     const chan = buf.getChannelData(c);
     for (let i = 0; i < framesN; i++) chan[i] = pcm[i * AUDIO_CH + c] / 32768;
   }
-  const src = audioCtx.createBufferSource();
-  src.buffer = buf;
-  src.connect(gainNode || audioCtx.destination);
-  const now = audioCtx.currentTime;
-  if (audioTime < now + 0.02) audioTime = now + 0.04;  // jitter buffer
-  src.start(audioTime);
-  audioTime += framesN / AUDIO_RATE;
+  schedulePlay(buf);
 }
 
 function noteFrame(frameId) {
@@ -270,6 +297,11 @@ function applyServerSettings(payload) {
   set("encoder", "encoder");
   set("fps", "framerate");
   set("crf", "video_crf");
+  if (payload.audio_codec && payload.audio_codec.value) {
+    const c = payload.audio_codec.value;
+    if (c !== audioCodec) opusDec = null;   /* codec switch: fresh state */
+    audioCodec = c;
+  }
   const res = payload.resolution && payload.resolution.value;
   if (res) {
     const [w, h] = res.split("x").map(Number);
