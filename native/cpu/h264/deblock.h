@@ -196,7 +196,9 @@ inline void deblock_segment(uint8_t* Y, int ypitch, uint8_t* Cb,
     }
     // ---- chroma edges: vertical at luma x offsets 0 and 8 (chroma
     // 0, 4); horizontal at luma y offset 8 (chroma 4). bS comes from
-    // the corresponding luma edge position.
+    // the corresponding luma edge position. Monochrome callers
+    // (ChromaArrayType == 0 / Hi444 separate planes) pass null chroma.
+    if (Cb == nullptr) continue;
     for (int pl = 0; pl < 2; ++pl) {
       uint8_t* C = pl ? Cr : Cb;
       const int cx0 = mbx * 8;

@@ -83,6 +83,14 @@ inline int inter_cbp_codenum(int cbp) {
 struct StripeEncoder::Impl {
   int w, h, mbw, mbh, yw, yh;
   bool deblock;          // in-loop deblocking (idc=2: within-slice only)
+  // >= 0: this instance codes ONE colour plane of a Hi444
+  // separate_colour_plane stream as monochrome (ChromaArrayType == 0):
+  // no chroma syntax at all, and inter residual falls back to I16x16
+  // (whose luma cbp lives in mb_type) so the only coded_block_pattern
+  // ever emitted is cbp == 0 -> codeNum 0, identical in every Table 9-4
+  // column -- no monochrome cbp table has to be transcribed.
+  int colour_plane = -1;
+  bool mono() const { return colour_plane >= 0; }
   Frame ref, cur;        // reference (prev recon) and current recon
   Frame src;             // padded source planes
   uint32_t frame_num = 0;
@@ -92,8 +100,8 @@ struct StripeEncoder::Impl {
   std::vector<int> prev_mv;   // per-MB quarter-pel mv of the last frame
   std::vector<uint8_t> prev_bad;  // MB was hopeless-intra last frame
 
-  Impl(int width, int height, bool deblock_on)
-      : w(width), h(height), deblock(deblock_on) {
+  Impl(int width, int height, bool deblock_on, int plane = -1)
+      : w(width), h(height), deblock(deblock_on), colour_plane(plane) {
     mbw = (w + 15) / 16;
     mbh = (h + 15) / 16;
     yw = mbw * 16;
@@ -185,6 +193,7 @@ struct StripeEncoder::Impl {
     b.ue(mb_row * mbw);          // first_mb_in_slice
     b.ue(idr ? 7 : 5);           // slice_type: I / P (all-slices-same form)
     b.ue(0);                     // pps id
+    if (mono()) b.u(colour_plane, 2);   // colour_plane_id
     b.u(frame_num & 0xFFFF, 16); // frame_num
     if (idr) b.ue(idr_pic_id);
     if (!idr) {
@@ -306,6 +315,11 @@ struct StripeEncoder::Impl {
 
     // ----- chroma: mode DC (0) vs H (1); shared by Cb/Cr
     const int qpc = chroma_qp(qp);
+    int cbp_chroma = 0;
+    int chroma_mode = 0;
+    int cqdc[2][4] = {};
+    int czz_ac[2][4][15] = {};
+    if (!mono()) {
     uint8_t cpred[2][64];
     long ccostH = 1 << 30, ccostDC = 0;
     uint8_t cpredH[2][64], cpredDC[2][64];
@@ -336,12 +350,11 @@ struct StripeEncoder::Impl {
           }
         }
     }
-    const int chroma_mode = (left && ccostH < ccostDC) ? 1 : 0;
+    chroma_mode = (left && ccostH < ccostDC) ? 1 : 0;
     std::memcpy(cpred[0], chroma_mode ? cpredH[0] : cpredDC[0], 64);
     std::memcpy(cpred[1], chroma_mode ? cpredH[1] : cpredDC[1], 64);
 
     int ccoef[2][4][16], cdc[2][4];
-    int czz_ac[2][4][15];
     bool c_any_ac = false;
     for (int comp = 0; comp < 2; ++comp) {
       Plane& sp = comp ? src.cr : src.cb;
@@ -364,7 +377,6 @@ struct StripeEncoder::Impl {
       }
     }
     // 2x2 Hadamard + quant of chroma DC
-    int cqdc[2][4];
     bool c_any_dc = false;
     for (int comp = 0; comp < 2; ++comp) {
       int* d = cdc[comp];
@@ -378,7 +390,7 @@ struct StripeEncoder::Impl {
       cqdc[comp][3] = quant_dc(w3, qpc, true);
       for (int i = 0; i < 4; ++i) c_any_dc |= cqdc[comp][i] != 0;
     }
-    const int cbp_chroma = c_any_ac ? 2 : (c_any_dc ? 1 : 0);
+    cbp_chroma = c_any_ac ? 2 : (c_any_dc ? 1 : 0);
 
     // ----- chroma recon
     for (int comp = 0; comp < 2; ++comp) {
@@ -413,6 +425,7 @@ struct StripeEncoder::Impl {
                 rec[4 * r + c] + cpred[comp][8 * (scy + r) + scx + c]);
       }
     }
+    }  // !mono()
 
     // ----- entropy
     // I-slice mb_type for I16x16: 1 + predMode + 4*cbp_chroma + 12*cbp_luma01
@@ -421,7 +434,7 @@ struct StripeEncoder::Impl {
     const int i16_type =
         1 + luma_mode + 4 * cbp_chroma + 12 * (cbp_luma ? 1 : 0);
     bw.ue(p_slice ? 5 + i16_type : i16_type);
-    bw.ue(chroma_mode);   // intra_chroma_pred_mode
+    if (!mono()) bw.ue(chroma_mode);   // intra_chroma_pred_mode
     bw.se(0);             // mb_qp_delta
     // DC block: nC from neighboring 4x4 blk(0,0)
     int nC_dc = blk_nc(ctx, tot, 0, true);
@@ -487,7 +500,10 @@ struct StripeEncoder::Impl {
   // Hadamard for inter); chroma as 2x2-Hadamard DC + AC. Inter quant
   // rounding (f = 2^qbits/6). MVs are even integers so the chroma MC is
   // an integer copy (mv/2). Mirrors the GPU row kernel's inter path.
-  void encode_p16(BitWriter& bw, int mbx, int mby, int qp, int mvq_x,
+  // Returns false (nothing written) when mono() and the quantized luma
+  // residual is nonzero -- the caller re-codes the MB as I16x16 so no
+  // monochrome coded_block_pattern codeNum other than 0 is ever needed.
+  bool encode_p16(BitWriter& bw, int mbx, int mby, int qp, int mvq_x,
                   int mvq_y, RowCtx& ctx, MbTotals& tot) {
     const int x0 = mbx * 16, y0 = mby * 16;
     const int cx0 = mbx * 8, cy0 = mby * 8;
@@ -502,17 +518,17 @@ struct StripeEncoder::Impl {
       for (int c = 0; c < 16; ++c)
         pred_y[16 * r + c] = static_cast<uint8_t>(
             luma_pred_px(ref.y, x0 + ix + c, y0 + iy + r, fx, fy));
-    for (int comp = 0; comp < 2; ++comp) {
-      const Plane& rp = comp ? ref.cr : ref.cb;
-      for (int r = 0; r < 8; ++r)
-        for (int c = 0; c < 8; ++c)
-          pred_c[comp][8 * r + c] = static_cast<uint8_t>(
-              chroma_pred_px(rp, cx0 + cix + c, cy0 + ciy + r, cdx, cdy));
-    }
+    if (!mono())
+      for (int comp = 0; comp < 2; ++comp) {
+        const Plane& rp = comp ? ref.cr : ref.cb;
+        for (int r = 0; r < 8; ++r)
+          for (int c = 0; c < 8; ++c)
+            pred_c[comp][8 * r + c] = static_cast<uint8_t>(
+                chroma_pred_px(rp, cx0 + cix + c, cy0 + ciy + r, cdx, cdy));
+      }
 
     // ----- luma: 16 blocks, full 16-coeff zigzag, inter quant
     int zz[16][16];
-    int dq_coef[16][16];
     bool any_l = false;
     for (int by = 0; by < 4; ++by)
       for (int bx = 0; bx < 4; ++bx) {
@@ -533,12 +549,14 @@ struct StripeEncoder::Impl {
         cap_coeffs(zz[b], 16);
         for (int i = 0; i < 16; ++i) any_l |= zz[b][i] != 0;
       }
+    if (mono() && any_l) return false;
     const int cbp_luma = any_l ? 15 : 0;
 
     // ----- chroma: DC 2x2 Hadamard + AC, inter quant
     const int qpc = chroma_qp(qp);
-    int cdc[2][4], czz[2][4][15], cqdc[2][4];
+    int cdc[2][4], czz[2][4][15], cqdc[2][4] = {};
     bool c_any_ac = false, c_any_dc = false;
+    if (!mono())
     for (int comp = 0; comp < 2; ++comp) {
       Plane& sp = comp ? src.cr : src.cb;
       for (int sub = 0; sub < 4; ++sub) {
@@ -569,7 +587,7 @@ struct StripeEncoder::Impl {
       cqdc[comp][3] = quant_dc(w3, qpc, false);
       for (int i = 0; i < 4; ++i) c_any_dc |= cqdc[comp][i] != 0;
     }
-    const int cbp_chroma = c_any_ac ? 2 : (c_any_dc ? 1 : 0);
+    const int cbp_chroma = mono() ? 0 : (c_any_ac ? 2 : (c_any_dc ? 1 : 0));
     const int cbp = (cbp_chroma << 4) | cbp_luma;
 
     // ----- bitstream
@@ -640,6 +658,7 @@ struct StripeEncoder::Impl {
                 clip8(rec[4 * r + c] + int(pred_y[16 * yy + xx]));
           }
       }
+    if (!mono())
     for (int comp = 0; comp < 2; ++comp) {
       Plane& rc = comp ? cur.cr : cur.cb;
       int dcq[4] = {0, 0, 0, 0};
@@ -680,6 +699,7 @@ struct StripeEncoder::Impl {
           }
       }
     }
+    return true;
   }
 
   // ---- sub-pel motion compensation (8.4.2.2.1/8.4.2.2.2) ------------------
@@ -799,6 +819,7 @@ struct StripeEncoder::Impl {
     for (int r = 0; r < 16; ++r)
       std::memcpy(cur.y.row(y0 + r) + x0,
                   ref.y.row(y0 + r + mvy) + x0 + mvx, 16);
+    if (mono()) return;
     int cx0 = mbx * 8, cy0 = mby * 8, cmx = mvx / 2, cmy = mvy / 2;
     for (int r = 0; r < 8; ++r) {
       std::memcpy(cur.cb.row(cy0 + r) + cx0,
@@ -820,20 +841,24 @@ struct StripeEncoder::Impl {
       uint8_t e = src.y.row(r)[w - 1];
       std::memset(src.y.row(r) + w, e, yw - w);
     }
-    int cw = (w + 1) / 2, ch = (h + 1) / 2, cwp = yw / 2, chp = yh / 2;
-    for (int r = 0; r < chp; ++r) {
-      int sr = std::min(r, ch - 1);
-      std::memcpy(src.cb.row(r), scb + static_cast<size_t>(sr) * scp, cw);
-      std::memcpy(src.cr.row(r), scr + static_cast<size_t>(sr) * scp, cw);
-      std::memset(src.cb.row(r) + cw, src.cb.row(r)[cw - 1], cwp - cw);
-      std::memset(src.cr.row(r) + cw, src.cr.row(r)[cw - 1], cwp - cw);
+    if (!mono()) {
+      int cw = (w + 1) / 2, ch = (h + 1) / 2, cwp = yw / 2, chp = yh / 2;
+      for (int r = 0; r < chp; ++r) {
+        int sr = std::min(r, ch - 1);
+        std::memcpy(src.cb.row(r), scb + static_cast<size_t>(sr) * scp, cw);
+        std::memcpy(src.cr.row(r), scr + static_cast<size_t>(sr) * scp, cw);
+        std::memset(src.cb.row(r) + cw, src.cb.row(r)[cw - 1], cwp - cw);
+        std::memset(src.cr.row(r) + cw, src.cr.row(r)[cw - 1], cwp - cw);
+      }
     }
 
     bool idr = force_idr || need_idr;
     if (idr) {
       frame_num = 0;
-      write_sps(out);
-      write_pps(out);
+      if (!mono()) {             // Hi444 wrapper writes the shared headers
+        write_sps(out);
+        write_pps(out);
+      }
       ++idr_pic_id;
       need_idr = false;
     }
@@ -985,35 +1010,43 @@ struct StripeEncoder::Impl {
           }
         }
         prev_bad[mb_idx] = best > 3 * inter_thresh ? 1 : 0;
+        bool wrote_inter = false;
         if (best <= inter_thresh) {
-          // P_L0_16x16 with coded residual (cbp may still come out 0)
+          // P_L0_16x16 with coded residual (cbp may still come out 0);
+          // in mono() a nonzero residual falls through to I16x16
           prev_mv[mb_idx] =
               (best_q_x & 0xFFFF) |
               static_cast<int>(static_cast<uint32_t>(best_q_y) << 16);
           flush_skip_run(b, ctx);
-          encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
-          rowmb[mbx].mvx = static_cast<int16_t>(best_q_x);
-          rowmb[mbx].mvy = static_cast<int16_t>(best_q_y);
-          for (int bi = 0; bi < 16; ++bi)
-            if (tot.luma[bi]) rowmb[mbx].nz |= 1u << bi;
-          ctx.have_left = true;
-          ctx.left_is_inter = true;
-          ctx.left_mvx = best_q_x;
-          ctx.left_mvy = best_q_y;
-          for (int by = 0; by < 4; ++by)
-            ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
-          for (int cy = 0; cy < 2; ++cy) {
-            ctx.left_cb_nc[cy] = tot.cb[cy * 2 + 1];
-            ctx.left_cr_nc[cy] = tot.cr[cy * 2 + 1];
+          wrote_inter =
+              encode_p16(b, mbx, mb_row, qp, best_q_x, best_q_y, ctx, tot);
+          if (wrote_inter) {
+            rowmb[mbx].mvx = static_cast<int16_t>(best_q_x);
+            rowmb[mbx].mvy = static_cast<int16_t>(best_q_y);
+            for (int bi = 0; bi < 16; ++bi)
+              if (tot.luma[bi]) rowmb[mbx].nz |= 1u << bi;
+            ctx.have_left = true;
+            ctx.left_is_inter = true;
+            ctx.left_mvx = best_q_x;
+            ctx.left_mvy = best_q_y;
+            for (int by = 0; by < 4; ++by)
+              ctx.left_luma_nc[by] = tot.luma[by * 4 + 3];
+            for (int cy = 0; cy < 2; ++cy) {
+              ctx.left_cb_nc[cy] = tot.cb[cy * 2 + 1];
+              ctx.left_cr_nc[cy] = tot.cr[cy * 2 + 1];
+            }
+            if (stats) ++stats->mb_inter;
           }
-          if (stats) ++stats->mb_inter;
-        } else {
-          // keep the best-found mv as a tracking hint across the intra
-          // fallback (fast new motion locks on within a frame or two)
-          prev_mv[mb_idx] = ((best_mvx * 4) & 0xFFFF) |
-                            static_cast<int>(
-                                static_cast<uint32_t>(best_mvy * 4) << 16);
-          flush_skip_run(b, ctx);
+        }
+        if (!wrote_inter) {
+          if (best > inter_thresh) {
+            // keep the best-found mv as a tracking hint across the intra
+            // fallback (fast new motion locks on within a frame or two)
+            prev_mv[mb_idx] = ((best_mvx * 4) & 0xFFFF) |
+                              static_cast<int>(
+                                  static_cast<uint32_t>(best_mvy * 4) << 16);
+            flush_skip_run(b, ctx);
+          }
           rowmb[mbx].intra = 1;
           encode_i16(b, mbx, mb_row, qp, true, ctx, tot);
           ctx.have_left = true;
@@ -1034,9 +1067,10 @@ struct StripeEncoder::Impl {
       b.rbsp_trailing();
       b.emit_nal(out, idr ? 3 : 2, idr ? 5 : 1, mb_row == 0);
       if (deblock)
-        deblock_segment(cur.y.data.data(), cur.y.pitch, cur.cb.data.data(),
-                        cur.cr.data.data(), cur.cb.pitch, mb_row, 0, mbw,
-                        rowmb.data(), qp, chroma_qp(qp));
+        deblock_segment(cur.y.data.data(), cur.y.pitch,
+                        mono() ? nullptr : cur.cb.data.data(),
+                        mono() ? nullptr : cur.cr.data.data(), cur.cb.pitch,
+                        mb_row, 0, mbw, rowmb.data(), qp, chroma_qp(qp));
     }
 
     std::swap(ref, cur);
@@ -1050,9 +1084,74 @@ struct StripeEncoder::Impl {
   }
 };
 
-StripeEncoder::StripeEncoder(int width, int height, bool deblock)
-    : impl_(new Impl(width, height, deblock)),
-      width_(width), height_(height) {}
+namespace {
+
+// SPS for High 4:4:4 Predictive with separate colour planes: each plane
+// decodes as a monochrome picture (ChromaArrayType == 0), so CropUnitX/Y
+// are 1 (no /2 on the crop amounts, unlike 4:2:0).
+void write_sps_444(std::vector<uint8_t>& out, int mbw, int mbh, int w,
+                   int h, int level_idc) {
+  BitWriter b;
+  b.u(244, 8);  // profile_idc: High 4:4:4 Predictive
+  b.u(0, 8);    // no constraint flags
+  b.u(level_idc, 8);
+  b.ue(0);      // sps id
+  b.ue(3);      // chroma_format_idc: 4:4:4
+  b.u(1, 1);    // separate_colour_plane_flag
+  b.ue(0);      // bit_depth_luma_minus8
+  b.ue(0);      // bit_depth_chroma_minus8
+  b.u(0, 1);    // qpprime_y_zero_transform_bypass
+  b.u(0, 1);    // seq_scaling_matrix_present
+  b.ue(12);     // log2_max_frame_num_minus4 -> 16-bit frame_num
+  b.ue(2);      // pic_order_cnt_type = 2
+  b.ue(1);      // max_num_ref_frames
+  b.u(0, 1);    // gaps_in_frame_num_value_allowed
+  b.ue(mbw - 1);
+  b.ue(mbh - 1);
+  b.u(1, 1);    // frame_mbs_only
+  b.u(1, 1);    // direct_8x8_inference
+  int crop_r = mbw * 16 - w, crop_b = mbh * 16 - h;
+  if (crop_r || crop_b) {
+    b.u(1, 1);
+    b.ue(0);
+    b.ue(crop_r);
+    b.ue(0);
+    b.ue(crop_b);
+  } else {
+    b.u(0, 1);
+  }
+  // VUI: BT.601 full range, as in the 4:2:0 SPS
+  b.u(1, 1);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.u(1, 1);
+  b.u(5, 3);
+  b.u(1, 1);
+  b.u(1, 1);
+  b.u(6, 8);
+  b.u(6, 8);
+  b.u(6, 8);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.u(0, 1);
+  b.rbsp_trailing();
+  b.emit_nal(out, 3, 7);
+}
+
+}  // namespace
+
+StripeEncoder::StripeEncoder(int width, int height, bool deblock,
+                             bool fullcolor)
+    : impl_(new Impl(width, height, deblock, fullcolor ? 0 : -1)),
+      width_(width), height_(height), fullcolor_(fullcolor) {
+  if (fullcolor) {
+    impl_cb_.reset(new Impl(width, height, deblock, 1));
+    impl_cr_.reset(new Impl(width, height, deblock, 2));
+  }
+}
 StripeEncoder::~StripeEncoder() = default;
 
 void StripeEncoder::encode_frame(const uint8_t* y, int ypitch,
@@ -1060,14 +1159,44 @@ void StripeEncoder::encode_frame(const uint8_t* y, int ypitch,
                                  int cpitch, int qp, bool force_idr,
                                  std::vector<uint8_t>& out,
                                  EncodeStats* stats) {
-  impl_->encode(y, ypitch, cb, cr, cpitch, qp, force_idr, out, stats);
+  if (!fullcolor_) {
+    impl_->encode(y, ypitch, cb, cr, cpitch, qp, force_idr, out, stats);
+    return;
+  }
+  // Hi444 separate planes: shared SPS/PPS, then the three planes'
+  // slices (plane-major) forming ONE access unit per frame.
+  bool idr = force_idr || impl_->need_idr;
+  if (idr) {
+    write_sps_444(out, impl_->mbw, impl_->mbh, width_, height_,
+                  impl_->level_idc);
+    impl_->write_pps(out);
+  }
+  EncodeStats s0, s1, s2;
+  impl_->encode(y, ypitch, nullptr, nullptr, 0, qp, idr, out, &s0);
+  impl_cb_->encode(cb, cpitch, nullptr, nullptr, 0, qp, idr, out, &s1);
+  impl_cr_->encode(cr, cpitch, nullptr, nullptr, 0, qp, idr, out, &s2);
+  if (stats) {
+    *stats = s0;
+    stats->mb_intra = s0.mb_intra + s1.mb_intra + s2.mb_intra;
+    stats->mb_inter = s0.mb_inter + s1.mb_inter + s2.mb_inter;
+    stats->mb_skip = s0.mb_skip + s1.mb_skip + s2.mb_skip;
+    stats->bytes = out.size();
+  }
 }
 
 const uint8_t* StripeEncoder::recon_y() const { return impl_->ref.y.data.data(); }
-const uint8_t* StripeEncoder::recon_cb() const { return impl_->ref.cb.data.data(); }
-const uint8_t* StripeEncoder::recon_cr() const { return impl_->ref.cr.data.data(); }
+const uint8_t* StripeEncoder::recon_cb() const {
+  return fullcolor_ ? impl_cb_->ref.y.data.data()
+                    : impl_->ref.cb.data.data();
+}
+const uint8_t* StripeEncoder::recon_cr() const {
+  return fullcolor_ ? impl_cr_->ref.y.data.data()
+                    : impl_->ref.cr.data.data();
+}
 int StripeEncoder::recon_ypitch() const { return impl_->ref.y.pitch; }
-int StripeEncoder::recon_cpitch() const { return impl_->ref.cb.pitch; }
+int StripeEncoder::recon_cpitch() const {
+  return fullcolor_ ? impl_cb_->ref.y.pitch : impl_->ref.cb.pitch;
+}
 
 void bgrx_to_yuv420(const uint8_t* bgrx, int stride, int width, int height,
                     uint8_t* y, int ypitch, uint8_t* cb, uint8_t* cr,
@@ -1095,6 +1224,26 @@ void bgrx_to_yuv420(const uint8_t* bgrx, int stride, int width, int height,
       cb[static_cast<size_t>(cy) * cpitch + cx] = clampf(cbs * 0.25f);
       cr[static_cast<size_t>(cy) * cpitch + cx] = clampf(crs * 0.25f);
     }
+}
+
+void bgrx_to_yuv444(const uint8_t* bgrx, int stride, int width, int height,
+                    uint8_t* y, uint8_t* cb, uint8_t* cr, int pitch) {
+  auto clampf = [](float v) {
+    return static_cast<uint8_t>(std::lrintf(std::min(std::max(v, 0.f), 255.f)));
+  };
+  for (int yy = 0; yy < height; ++yy) {
+    const uint8_t* rowp = bgrx + static_cast<size_t>(yy) * stride;
+    uint8_t* yr = y + static_cast<size_t>(yy) * pitch;
+    uint8_t* cbr = cb + static_cast<size_t>(yy) * pitch;
+    uint8_t* crr = cr + static_cast<size_t>(yy) * pitch;
+    for (int xx = 0; xx < width; ++xx) {
+      const uint8_t* p = rowp + xx * 4;
+      float r = p[2], g = p[1], b = p[0];
+      yr[xx] = clampf(0.299f * r + 0.587f * g + 0.114f * b);
+      cbr[xx] = clampf(-0.168736f * r - 0.331264f * g + 0.5f * b + 128.f);
+      crr[xx] = clampf(0.5f * r - 0.418688f * g - 0.081312f * b + 128.f);
+    }
+  }
 }
 
 }  // namespace h264

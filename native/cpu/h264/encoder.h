@@ -34,14 +34,22 @@ struct EncodeStats {
 
 class StripeEncoder {
  public:
-  StripeEncoder(int width, int height, bool deblock = true);
+  // fullcolor: Hi444 (profile 244) with separate_colour_plane_flag=1 --
+  // each plane is coded as monochrome through the same luma machinery
+  // (the reference's pixelflux `h264_fullcolor` / I444 streaming mode);
+  // encode_frame then takes FULL-resolution cb/cr planes with `cpitch`.
+  StripeEncoder(int width, int height, bool deblock = true,
+                bool fullcolor = false);
   ~StripeEncoder();
 
-  // Encode one frame from planar YUV420 (pitch in bytes). qp in [0,51].
-  // force_idr resets the stream (SPS/PPS + IDR). Appends Annex-B to `out`.
+  // Encode one frame from planar YUV420 (or I444 when fullcolor; pitch
+  // in bytes). qp in [0,51]. force_idr resets the stream (SPS/PPS +
+  // IDR). Appends Annex-B to `out`.
   void encode_frame(const uint8_t* y, int ypitch, const uint8_t* cb,
                     const uint8_t* cr, int cpitch, int qp, bool force_idr,
                     std::vector<uint8_t>& out, EncodeStats* stats = nullptr);
+
+  bool fullcolor() const { return fullcolor_; }
 
   // Access to the reconstructed reference (for tests and drift checks).
   const uint8_t* recon_y() const;
@@ -55,8 +63,9 @@ class StripeEncoder {
 
  private:
   struct Impl;
-  std::unique_ptr<Impl> impl_;
+  std::unique_ptr<Impl> impl_, impl_cb_, impl_cr_;
   int width_, height_;
+  bool fullcolor_ = false;
 };
 
 // BGRX -> planar YUV420 (BT.601 full range), edge-replicated to MB-aligned
@@ -64,6 +73,10 @@ class StripeEncoder {
 void bgrx_to_yuv420(const uint8_t* bgrx, int stride, int width, int height,
                     uint8_t* y, int ypitch, uint8_t* cb, uint8_t* cr,
                     int cpitch);
+
+// BGRX -> planar I444 (BT.601 full range), for the Hi444 fullcolor mode.
+void bgrx_to_yuv444(const uint8_t* bgrx, int stride, int width, int height,
+                    uint8_t* y, uint8_t* cb, uint8_t* cr, int pitch);
 
 }  // namespace h264
 }  // namespace hipflux

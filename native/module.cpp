@@ -179,18 +179,22 @@ PYBIND11_MODULE(_native, m) {
   struct PyH264 {
     h264::StripeEncoder enc;
     int w, h, ypitch, cpitch;
+    bool fullcolor;
     std::vector<uint8_t> yuv;
-    PyH264(int width, int height, bool deblock)
-        : enc(width, height, deblock), w(width), h(height) {
+    PyH264(int width, int height, bool deblock, bool fc)
+        : enc(width, height, deblock, fc), w(width), h(height),
+          fullcolor(fc) {
       ypitch = (w + 15) & ~15;
-      cpitch = ypitch / 2;
+      cpitch = fullcolor ? ypitch : ypitch / 2;
       int yh = (h + 15) & ~15;
-      yuv.resize(static_cast<size_t>(ypitch) * yh * 3 / 2);
+      yuv.resize(static_cast<size_t>(ypitch) * yh * (fullcolor ? 3 : 2) *
+                 (fullcolor ? 1 : 3) / (fullcolor ? 1 : 4));
     }
   };
   py::class_<PyH264>(m, "H264Encoder")
-      .def(py::init<int, int, bool>(), py::arg("width"), py::arg("height"),
-           py::arg("deblock") = true)
+      .def(py::init<int, int, bool, bool>(), py::arg("width"),
+           py::arg("height"), py::arg("deblock") = true,
+           py::arg("fullcolor") = false)
       .def(
           "encode",
           [](PyH264& self, py::buffer bgrx, int qp, bool idr) {
@@ -204,10 +208,16 @@ PYBIND11_MODULE(_native, m) {
               uint8_t* y = self.yuv.data();
               int yh = (self.h + 15) & ~15;
               uint8_t* cb = y + static_cast<size_t>(self.ypitch) * yh;
-              uint8_t* cr = cb + static_cast<size_t>(self.cpitch) * (yh / 2);
-              h264::bgrx_to_yuv420(static_cast<const uint8_t*>(info.ptr),
-                                   self.w * 4, self.w, self.h, y, self.ypitch,
-                                   cb, cr, self.cpitch);
+              uint8_t* cr = cb + static_cast<size_t>(self.cpitch) *
+                                     (self.fullcolor ? yh : yh / 2);
+              if (self.fullcolor)
+                h264::bgrx_to_yuv444(static_cast<const uint8_t*>(info.ptr),
+                                     self.w * 4, self.w, self.h, y, cb, cr,
+                                     self.ypitch);
+              else
+                h264::bgrx_to_yuv420(static_cast<const uint8_t*>(info.ptr),
+                                     self.w * 4, self.w, self.h, y,
+                                     self.ypitch, cb, cr, self.cpitch);
               self.enc.encode_frame(y, self.ypitch, cb, cr, self.cpitch, qp,
                                     idr, out, &st);
             }
@@ -246,12 +256,13 @@ PYBIND11_MODULE(_native, m) {
            [](PyH264& self) {
              // returns (y, cb, cr, ypitch, cpitch) of the reference frame
              int yh = (self.h + 15) & ~15;
+             int ch = self.fullcolor ? yh : yh / 2;
              py::bytes y(reinterpret_cast<const char*>(self.enc.recon_y()),
                          static_cast<size_t>(self.enc.recon_ypitch()) * yh);
              py::bytes cb(reinterpret_cast<const char*>(self.enc.recon_cb()),
-                          static_cast<size_t>(self.enc.recon_cpitch()) * yh / 2);
+                          static_cast<size_t>(self.enc.recon_cpitch()) * ch);
              py::bytes cr(reinterpret_cast<const char*>(self.enc.recon_cr()),
-                          static_cast<size_t>(self.enc.recon_cpitch()) * yh / 2);
+                          static_cast<size_t>(self.enc.recon_cpitch()) * ch);
              return py::make_tuple(y, cb, cr, self.enc.recon_ypitch(),
                                    self.enc.recon_cpitch());
            });

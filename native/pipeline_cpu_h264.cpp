@@ -29,14 +29,17 @@ class CpuH264Pipeline : public EncodePipeline {
         int hgt = std::min(stripe_h, frame.height - y0);
         encoders_.push_back(
             std::make_unique<h264::StripeEncoder>(
-                frame.width, hgt, settings_.video_deblock));
+                frame.width, hgt, settings_.video_deblock,
+                settings_.video_fullcolor));
       }
-      // per-stripe scratch YUV planes
+      // per-stripe scratch YUV planes (I444 when fullcolor)
       ypitch_ = (frame.width + 15) & ~15;
-      cpitch_ = ypitch_ / 2;
+      fullcolor_ = settings_.video_fullcolor;
+      cpitch_ = fullcolor_ ? ypitch_ : ypitch_ / 2;
       yuv_.resize(n_stripes);
       for (auto& v : yuv_)
-        v.resize(static_cast<size_t>(ypitch_) * stripe_h * 3 / 2);
+        v.resize(static_cast<size_t>(ypitch_) * stripe_h *
+                 (fullcolor_ ? 6 : 3) / 2);
     }
 
     struct Out {
@@ -57,10 +60,17 @@ class CpuH264Pipeline : public EncodePipeline {
         int hgt = j.y1 - j.y0;
         uint8_t* y = yuv_[i].data();
         uint8_t* cb = y + static_cast<size_t>(ypitch_) * stripe_h;
-        uint8_t* cr = cb + static_cast<size_t>(cpitch_) * (stripe_h / 2);
-        h264::bgrx_to_yuv420(
-            frame.data + static_cast<size_t>(j.y0) * frame.stride,
-            frame.stride, frame.width, hgt, y, ypitch_, cb, cr, cpitch_);
+        uint8_t* cr =
+            cb + static_cast<size_t>(cpitch_) *
+                     (fullcolor_ ? stripe_h : stripe_h / 2);
+        if (fullcolor_)
+          h264::bgrx_to_yuv444(
+              frame.data + static_cast<size_t>(j.y0) * frame.stride,
+              frame.stride, frame.width, hgt, y, cb, cr, ypitch_);
+        else
+          h264::bgrx_to_yuv420(
+              frame.data + static_cast<size_t>(j.y0) * frame.stride,
+              frame.stride, frame.width, hgt, y, ypitch_, cb, cr, cpitch_);
         encoders_[i]->encode_frame(y, ypitch_, cb, cr, cpitch_, ctx.crf,
                                    ctx.idr, outs[i].bytes, &outs[i].st);
       });
@@ -89,6 +99,7 @@ class CpuH264Pipeline : public EncodePipeline {
   std::vector<std::unique_ptr<h264::StripeEncoder>> encoders_;
   std::vector<std::vector<uint8_t>> yuv_;
   int w_ = 0, ypitch_ = 0, cpitch_ = 0;
+  bool fullcolor_ = false;
 };
 
 }  // namespace

@@ -1085,6 +1085,13 @@ std::unique_ptr<EncodePipeline> make_hip_pipeline(const CaptureSettings& s) {
   try {
     if (s.output_mode == 0) return std::make_unique<HipJpegPipeline>(s);
     if (s.output_mode == 2) return make_hip_hevc_pipeline(s);
+    if (s.video_fullcolor) {
+      // Hi444 separate-colour-plane H.264 is served by the CPU
+      // stripe-parallel encoder (the GPU row kernels are 4:2:0).
+      std::fprintf(stderr,
+                   "hipflux: h264 fullcolor -> CPU Hi444 encoder\n");
+      return nullptr;
+    }
     return std::make_unique<HipH264Pipeline>(s);
   } catch (const std::exception& e) {
     std::fprintf(stderr, "hipflux: HIP pipeline init failed: %s\n", e.what());

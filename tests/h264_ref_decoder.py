@@ -490,7 +490,9 @@ def deblock_segment_py(Y, Cb, Cr, mb_row, mbx0, infos, qp, qpc):
                 for i in range(3):
                     Y[ey - 1 - i, cols] = P[i]
                     Y[ey + i, cols] = Q[i]
-        # chroma
+        # chroma (skipped for monochrome planes: ChromaArrayType == 0)
+        if Cb is None:
+            continue
         cx0 = (mbx0 + m) * 8
         for C in (Cb, Cr):
             for e in range(2):         # vertical, chroma x = 0, 4
@@ -546,6 +548,8 @@ class Decoder:
         self.pps = None
         self.y = self.cb = self.cr = None
         self.ref_y = self.ref_cb = self.ref_cr = None
+        self.planes = [None, None, None]       # Hi444 separate planes
+        self.ref_planes = [None, None, None]
         self.frames = []  # decoded (cropped) frames as (y, cb, cr)
 
     # ---- headers -----------------------------------------------------------
@@ -555,6 +559,16 @@ class Decoder:
         br.u(8)  # constraint flags + reserved
         s["level_idc"] = br.u(8)
         assert br.ue() == 0
+        s["separate"] = False
+        if s["profile_idc"] in (100, 110, 122, 244):
+            cfi = br.ue()           # chroma_format_idc
+            assert cfi == 3, "subset: high profiles only as Hi444"
+            s["separate"] = br.u(1) == 1
+            assert s["separate"], "subset: separate_colour_plane only"
+            assert br.ue() == 0     # bit_depth_luma_minus8
+            assert br.ue() == 0     # bit_depth_chroma_minus8
+            assert br.u(1) == 0     # qpprime_y_zero_transform_bypass
+            assert br.u(1) == 0     # seq_scaling_matrix_present
         s["log2_max_frame_num"] = br.ue() + 4
         s["poc_type"] = br.ue()
         assert s["poc_type"] == 2
@@ -566,7 +580,10 @@ class Decoder:
         br.u(1)  # direct_8x8
         if br.u(1):  # cropping
             cl, cr_, ct, cb_ = br.ue(), br.ue(), br.ue(), br.ue()
-            s["crop"] = (cl * 2, cr_ * 2, ct * 2, cb_ * 2)
+            # CropUnit is 1 for ChromaArrayType==0 (separate planes), 2
+            # for 4:2:0
+            u = 1 if s["separate"] else 2
+            s["crop"] = (cl * u, cr_ * u, ct * u, cb_ * u)
         else:
             s["crop"] = (0, 0, 0, 0)
         self.sps = s
@@ -601,6 +618,7 @@ class Decoder:
         is_p = slice_type in (0, 5)
         assert is_i or is_p
         assert br.ue() == 0  # pps id
+        cp = br.u(2) if s["separate"] else None   # colour_plane_id
         br.u(s["log2_max_frame_num"])  # frame_num
         if idr:
             br.ue()  # idr_pic_id
@@ -621,7 +639,15 @@ class Decoder:
             assert dbf in (1, 2), \
                 "decoder subset: deblocking off or within-slice only"
 
-        if idr and first_mb == 0:
+        if cp is not None:
+            # route the shared mono machinery at this colour plane:
+            # self.y is the plane, chroma disabled (ChromaArrayType==0)
+            if idr and first_mb == 0:
+                self.planes[cp] = np.zeros((H, W), np.int32)
+            self.y = self.planes[cp]
+            self.cb = self.cr = None
+            self.ref_y = self.ref_planes[cp]
+        elif idr and first_mb == 0:
             self.y = np.zeros((H, W), np.int32)
             self.cb = np.zeros((H // 2, W // 2), np.int32)
             self.cr = np.zeros((H // 2, W // 2), np.int32)
@@ -676,6 +702,8 @@ class Decoder:
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
         self.y[y0:y0 + 16, x0:x0 + 16] = self.ref_y[y0:y0 + 16, x0:x0 + 16]
+        if self.cb is None:
+            return
         self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = self.ref_cb[cy0:cy0 + 8,
                                                         cx0:cx0 + 8]
         self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = self.ref_cr[cy0:cy0 + 8,
@@ -692,6 +720,10 @@ class Decoder:
         assert cbp_cn in self.INTER_CBP_FROM_CODENUM, \
             f"subset: inter cbp codeNum {cbp_cn}"
         cbp = self.INTER_CBP_FROM_CODENUM[cbp_cn]
+        if self.cb is None:
+            # monochrome plane: the encoder only ever emits cbp == 0
+            # (residual MBs fall back to I16x16)
+            assert cbp == 0, f"mono inter cbp {cbp}"
         cbp_luma, cbp_chroma = cbp & 15, cbp >> 4
         mvpx, mvpy = ctx["left_mv"] if (ctx["left_avail"] and
                                         ctx["left_inter"]) else (0, 0)
@@ -699,8 +731,11 @@ class Decoder:
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
         pred_y = luma_mc(self.ref_y, x0, y0, mvx, mvy)
-        pred_cb = chroma_mc(self.ref_cb, cx0, cy0, mvx, mvy)
-        pred_cr = chroma_mc(self.ref_cr, cx0, cy0, mvx, mvy)
+        mono = self.cb is None
+        pred_cb = None if mono else chroma_mc(self.ref_cb, cx0, cy0,
+                                              mvx, mvy)
+        pred_cr = None if mono else chroma_mc(self.ref_cr, cx0, cy0,
+                                              mvx, mvy)
         new_luma_nc = np.zeros((4, 4), np.int32)
         new_cb_nc = np.zeros((2, 2), np.int32)
         new_cr_nc = np.zeros((2, 2), np.int32)
@@ -766,8 +801,9 @@ class Decoder:
                     rec = idct4(coeffs).reshape(4, 4)
                     pred[scy:scy + 4, scx:scx + 4] += rec
         self.y[y0:y0 + 16, x0:x0 + 16] = np.clip(pred_y, 0, 255)
-        self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cb, 0, 255)
-        self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cr, 0, 255)
+        if not mono:
+            self.cb[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cb, 0, 255)
+            self.cr[cy0:cy0 + 8, cx0:cx0 + 8] = np.clip(pred_cr, 0, 255)
         nz = 0
         for by in range(4):
             for bx in range(4):
@@ -784,7 +820,12 @@ class Decoder:
         pred_mode = t % 4
         cbp_chroma = (t // 4) % 3
         cbp_luma = 15 if t >= 12 else 0
-        chroma_mode = br.ue()
+        mono = self.cb is None
+        if mono:
+            assert cbp_chroma == 0, "mono I16 with chroma cbp"
+            chroma_mode = 0          # intra_chroma_pred_mode not present
+        else:
+            chroma_mode = br.ue()
         qp = qp + br.se()  # mb_qp_delta
         x0, y0 = mbx * 16, mby * 16
         cx0, cy0 = mbx * 8, mby * 8
@@ -839,10 +880,15 @@ class Decoder:
                         0, 255)
 
         # chroma
-        qpc = chroma_qp(qp + self.pps["chroma_qp_offset"])
-        assert chroma_mode in (0, 1), "subset: chroma V/Plane not emitted"
         new_cb_nc = np.zeros((2, 2), np.int32)
         new_cr_nc = np.zeros((2, 2), np.int32)
+        if mono:
+            ctx.update(left_avail=True, left_inter=False,
+                       left_luma_nc=[int(new_luma_nc[by, 3])
+                                     for by in range(4)])
+            return
+        qpc = chroma_qp(qp + self.pps["chroma_qp_offset"])
+        assert chroma_mode in (0, 1), "subset: chroma V/Plane not emitted"
         cpred = {}
         for comp, plane in (("cb", self.cb), ("cr", self.cr)):
             p = np.zeros((8, 8), np.int32)
@@ -916,7 +962,8 @@ class Decoder:
             br = BitReader(nal[1:])
             if nal_type == 7:
                 self.parse_sps(br)
-                total_mbs = self.sps["mbw"] * self.sps["mbh"]
+                total_mbs = self.sps["mbw"] * self.sps["mbh"] * (
+                    3 if self.sps["separate"] else 1)
             elif nal_type == 8:
                 self.parse_pps(br)
             elif nal_type in (1, 5):
@@ -932,6 +979,13 @@ class Decoder:
         cl, cr_, ct, cb_ = s["crop"]
         W, H = s["mbw"] * 16, s["mbh"] * 16
         w, h = W - cl - cr_, H - ct - cb_
+        if s["separate"]:
+            # Hi444: three full-resolution planes form the frame
+            self.frames.append(tuple(
+                p[ct:ct + h, cl:cl + w].astype(np.uint8)
+                for p in self.planes))
+            self.ref_planes = [p.copy() for p in self.planes]
+            return
         self.frames.append((
             self.y[ct:ct + h, cl:cl + w].astype(np.uint8),
             self.cb[ct // 2:(ct + h) // 2, cl // 2:(cl + w) // 2]
