@@ -186,8 +186,15 @@ void ScreenCapture::run() {
 
   const int scale_div =
       std::min(4, std::max(1, settings_.capture_scale_div));
+  const float fscale =
+      std::min(1.0f, std::max(0.25f, settings_.capture_scale));
+  const bool frac_scale = fscale < 0.9999f;
   DamageTracker damage;
-  damage.reset(src->width() / scale_div, src->height() / scale_div);
+  if (frac_scale)
+    damage.reset(std::max(1, int(src->width() * fscale + 0.5f)),
+                 std::max(1, int(src->height() * fscale + 0.5f)));
+  else
+    damage.reset(src->width() / scale_div, src->height() / scale_div);
   std::vector<uint8_t> scale_buf;
 
   Watermark watermark;
@@ -272,9 +279,19 @@ void ScreenCapture::run() {
                  cursor.argb.data(), cursor.argb.size());
     }
 
-    // integer box downscale (capture_scale_div): encode a w/div x h/div
-    // stream; cursor/watermark were composited at native size above
-    if (scale_div > 1) {
+    // capture downscale: fractional bilinear (capture_scale) or exact
+    // integer box (capture_scale_div); cursor/watermark were composited
+    // at native size above
+    if (frac_scale) {
+      int ow, oh, ostride;
+      bilinear_downscale_bgrx(frame.data, frame.stride, frame.width,
+                              frame.height, fscale, scale_buf, ow, oh,
+                              ostride);
+      frame.data = scale_buf.data();
+      frame.width = ow;
+      frame.height = oh;
+      frame.stride = ostride;
+    } else if (scale_div > 1) {
       int ow, oh, ostride;
       box_downscale_bgrx(frame.data, frame.stride, frame.width,
                          frame.height, scale_div, scale_buf, ow, oh,

@@ -28,6 +28,9 @@ struct CaptureSettings {
   bool capture_cursor = false;
 
   int output_mode = 1;              // 0 = JPEG, 1 = H.264, 2 = HEVC
+  float capture_scale = 1.0f;       // fractional bilinear downscale
+                                    // (0.25..1.0); takes precedence over
+                                    // capture_scale_div when != 1
   int capture_scale_div = 1;        // integer box-downscale (1..4): encode
                                     // at capture/div (e.g. 4K capture -> 2
                                     // -> 1080p stream)

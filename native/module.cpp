@@ -14,6 +14,7 @@
 #include "cpu/opus/celt.h"
 #include "cpu/opus/range_coder.h"
 #include "cpu/jpeg_enc.h"
+#include "cpu/scale.h"
 #include "engine.h"
 #include <hip/hip_runtime.h>
 
@@ -39,6 +40,7 @@ PYBIND11_MODULE(_native, m) {
       .def_readwrite("target_fps", &CaptureSettings::target_fps)
       .def_readwrite("capture_cursor", &CaptureSettings::capture_cursor)
       .def_readwrite("output_mode", &CaptureSettings::output_mode)
+      .def_readwrite("capture_scale", &CaptureSettings::capture_scale)
       .def_readwrite("capture_scale_div", &CaptureSettings::capture_scale_div)
       .def_readwrite("pipeline_depth", &CaptureSettings::pipeline_depth)
       .def_readwrite("video_fullframe", &CaptureSettings::video_fullframe)
@@ -796,6 +798,23 @@ PYBIND11_MODULE(_native, m) {
              return py::bytes(reinterpret_cast<char*>(out.data()), got);
            })
       .def_property_readonly("buffered", &AudioPlayback::buffered);
+
+  m.def(
+      "_bilinear_downscale",
+      [](py::buffer bgrx, int w, int h, float scale) {
+        py::buffer_info info = bgrx.request();
+        if (info.size < static_cast<ssize_t>(w) * h * 4)
+          throw std::runtime_error("buffer too small");
+        std::vector<uint8_t> out;
+        int ow, oh, ostride;
+        bilinear_downscale_bgrx(static_cast<const uint8_t*>(info.ptr),
+                                w * 4, w, h, scale, out, ow, oh, ostride);
+        return py::make_tuple(
+            py::bytes(reinterpret_cast<const char*>(out.data()), out.size()),
+            ow, oh);
+      },
+      py::arg("bgrx"), py::arg("w"), py::arg("h"), py::arg("scale"),
+      "Test hook: exact fixed-point bilinear downscale (engine path).");
 
   m.def(
       "screenshot",

@@ -119,6 +119,10 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                "Record client-reported WebRTC getStats payloads to CSV."),
     SettingDef("webrtc_statistics_dir", str, "/tmp/selkies-webrtc-stats",
                "Directory for the per-day WebRTC statistics CSV files."),
+    SettingDef("capture_scale", float, 1.0,
+               "Fractional capture downscale (0.25-1.0); overrides "
+               "capture_scale_div when below 1.",
+               value_range=(0.25, 1.0), client=True),
     SettingDef("capture_scale_div", int, 1,
                "Integer capture downscale: encode at capture/div "
                "(2 = 4K capture -> 1080p stream).",

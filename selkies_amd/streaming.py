@@ -168,6 +168,7 @@ class StreamingService:
         cs.video_fullcolor = s.video_fullcolor
         cs.video_deblock = s.video_deblock
         cs.video_fullframe = s.video_fullframe
+        cs.capture_scale = s.capture_scale
         cs.capture_scale_div = s.capture_scale_div
         cs.pipeline_depth = s.video_pipeline_depth
         cs.use_paint_over_quality = s.use_paint_over_quality
@@ -518,6 +519,7 @@ class StreamingService:
 
     # ---- settings application ----------------------------------------------
     STRUCTURAL = {"encoder", "resolution", "video_fullcolor", "use_cpu",
+                  "capture_scale",
                   "video_fullframe"}
 
     async def _apply_client_settings(self, changes: dict):

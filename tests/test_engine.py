@@ -263,3 +263,95 @@ def test_hevc_mode_stripes_decode():
         frames = Decoder().decode(data[10:])
         assert frames and frames[0][0].shape == (h, w)
     assert seen_rows == {0, 64, 128}
+
+
+def test_bilinear_downscale_matches_numpy_fixedpoint():
+    """Fractional capture_scale: the engine's fixed-point bilinear must
+    match an independent numpy mirror of the documented math bit-exactly
+    (16.16 positions, 8-bit weights, round-half-up)."""
+    import numpy as np
+    from hipflux import _native
+
+    rng = np.random.default_rng(21)
+    w, h = 157, 93
+    src = rng.integers(0, 256, (h, w, 4), dtype=np.uint8)
+    for scale in (0.25, 0.5, 2 / 3, 0.8125):
+        data, ow, oh = _native._bilinear_downscale(src.tobytes(), w, h,
+                                                   scale)
+        got = np.frombuffer(data, np.uint8).reshape(oh, ow, 4)
+        assert ow == int(w * scale + 0.5) and oh == int(h * scale + 0.5)
+        xstep = (w << 16) // ow
+        ystep = (h << 16) // oh
+        s64 = src.astype(np.int64)
+        exp = np.zeros((oh, ow, 4), np.int64)
+        xs = ((2 * np.arange(ow) + 1) * xstep - (1 << 16)) // 2
+        xs = np.clip(xs, 0, None)
+        xi = np.minimum(xs >> 16, w - 2)
+        fx = (xs >> 8) & 0xFF
+        for y in range(oh):
+            sy = ((2 * y + 1) * ystep - (1 << 16)) // 2
+            sy = max(sy, 0)
+            iy = min(sy >> 16, h - 2)
+            fy = (sy >> 8) & 0xFF
+            a = s64[iy, xi]
+            b = s64[iy, xi + 1]
+            c = s64[iy + 1, xi]
+            e = s64[iy + 1, xi + 1]
+            top = (a << 8) + (b - a) * fx[:, None]
+            bot = (c << 8) + (e - c) * fx[:, None]
+            exp[y] = ((top << 8) + (bot - top) * fy + (1 << 15)) >> 16
+        assert np.array_equal(got, exp.astype(np.uint8)), scale
+
+
+def test_capture_scale_fractional_stream(tmp_path):
+    """capture_scale=0.5 on a 256x128 capture emits a 128x64 stream whose
+    decode matches the engine's own bilinear of the source (PSNR-tight,
+    both sides exact so only codec loss remains)."""
+    import threading
+    import numpy as np
+    from h264_ref_decoder import Decoder
+    from hipflux import _native
+
+    w, h = 256, 128
+    shot, sw, sh = _native.screenshot("synthetic:static", "", w, h)
+    small, ow, oh = _native._bilinear_downscale(shot, w, h, 0.5)
+    assert (ow, oh) == (128, 64)
+    exp = np.frombuffer(small, np.uint8).reshape(oh, ow, 4)
+
+    s = _native.CaptureSettings()
+    s.capture_width = w
+    s.capture_height = h
+    s.capture_scale = 0.5
+    s.target_fps = 30
+    s.output_mode = 1
+    s.use_cpu = True
+    s.gpu_id = -1
+    s.capture_backend = "synthetic:static"
+    s.video_fullframe = True
+    s.video_crf = 10
+    s.video_cbr_mode = False
+    s.stripe_height = 64
+    got = {}
+    done = threading.Event()
+
+    def cb(data, frame_id, y, width, height, key, *a):
+        got.setdefault(y, []).append((bytes(data), width, height))
+        if len(got.get(0, [])) >= 2:
+            done.set()
+
+    cap = _native.ScreenCapture()
+    cap.start_capture(cb, s)
+    assert done.wait(6)
+    cap.stop_capture()
+    data, width, height = got[0][0]
+    assert (width, height) == (128, 64)
+    stream = data[10:]            # strip the wire header
+    y_dec = Decoder().decode(stream)[0][0]
+    # the engine's scaled input must EQUAL the fixed-point bilinear:
+    # encoding `exp` directly at the same QP must decode identically
+    import hipflux
+    enc = hipflux.H264Encoder(128, 64)
+    r = enc.encode(np.ascontiguousarray(exp).tobytes(), qp=10, idr=True)
+    y_direct = Decoder().decode(r["data"])[0][0]
+    assert np.array_equal(y_dec, y_direct), \
+        "engine fractional-scale path diverges from bilinear_downscale"
