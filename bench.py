@@ -144,8 +144,11 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=300)
     ap.add_argument("--warmup", type=int, default=30)
-    ap.add_argument("--width", type=int, default=1920)
-    ap.add_argument("--height", type=int, default=1080)
+    # default resolution follows the encoder's BASELINE config: 1080p for
+    # h264/jpeg (config 2), 4K for hevc (config 3) — the metric name and
+    # the measured config must agree
+    ap.add_argument("--width", type=int, default=None)
+    ap.add_argument("--height", type=int, default=None)
     ap.add_argument("--qp", type=int, default=28)
     ap.add_argument("--encoder", default="h264enc-striped")
     ap.add_argument("--cpu", action="store_true",
@@ -178,6 +181,11 @@ def main():
                          "scaling of single-stream latency — stripes are "
                          "independent bitstreams, the multi-GPU seam)")
     args = ap.parse_args()
+    if args.width is None or args.height is None:
+        if args.encoder.startswith("hevc"):
+            args.width, args.height = 3840, 2160
+        else:
+            args.width, args.height = 1920, 1080
 
     if args.mode == "e2e":
         run_e2e(args)

@@ -248,6 +248,76 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("ws_max_message_mb", int, 16, "Inbound WS message ceiling.",
                value_range=(1, 256)),
     SettingDef("debug", bool, False, "Verbose logging."),
+    # ---- app lifecycle (reference app_ready/run_after hooks) ----
+    SettingDef("app_ready_file", str, "",
+               "Path whose existence marks the session app as ready."),
+    SettingDef("app_wait_ready", bool, False,
+               "Delay capture start until app_ready_file exists."),
+    SettingDef("run_after_connect", str, "",
+               "Shell command run when the first client connects."),
+    SettingDef("run_after_disconnect", str, "",
+               "Shell command run when the last client disconnects."),
+    # ---- auth extras ----
+    SettingDef("basic_auth_viewonly_password", str, "",
+               "Second basic-auth password granting view-only role."),
+    # ---- infra knobs ----
+    SettingDef("cert_reload_interval", int, 60,
+               "Seconds between TLS cert/key change polls.",
+               value_range=(5, 3600)),
+    SettingDef("file_transfer_limit_mbps", float, 0.0,
+               "Hard ceiling on file-transfer bandwidth (0 = adaptive "
+               "pacer only).", value_range=(0.0, 10000.0)),
+    SettingDef("subfolder", str, "",
+               "URL prefix to serve under (reverse-proxy subpath)."),
+    SettingDef("web_root", str, "",
+               "Override directory for the static web client ('' = "
+               "bundled)."),
+    SettingDef("js_socket_path", str, "/tmp/selkies_js",
+               "Base path for gamepad interposer unix sockets."),
+    SettingDef("webrtc_public_ip", str, "",
+               "Advertise this IP in ICE candidates (NAT'd hosts)."),
+    # ---- display/cursor extras ----
+    SettingDef("force_aligned_resolution", bool, True,
+               "Round client-requested resolutions to even dimensions.",
+               client=True),
+    SettingDef("enable_cursors", bool, True,
+               "Push server cursor shapes to clients.", client=True),
+    SettingDef("use_browser_cursors", bool, True,
+               "Render the cursor via CSS on the client instead of "
+               "compositing it into the video.", client=True),
+    SettingDef("debug_cursors", bool, False, "Log cursor shape updates."),
+    SettingDef("paint_over_jpeg_quality", int, 95,
+               "JPEG quality for static paint-over passes.",
+               value_range=(10, 100), client=True),
+    # ---- web UI contract (client-side visibility; reference ui_*) ----
+    SettingDef("ui_title", str, "Selkies", "Browser tab / UI title.",
+               client=True),
+    SettingDef("ui_show_logo", bool, True, "Show the logo in the UI.",
+               client=True),
+    SettingDef("ui_show_sidebar", bool, True, "Show the sidebar.",
+               client=True),
+    SettingDef("ui_show_core_buttons", bool, True,
+               "Show fullscreen/settings core buttons.", client=True),
+    SettingDef("ui_sidebar_show_apps", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_audio_settings", bool, True, "",
+               client=True),
+    SettingDef("ui_sidebar_show_clipboard", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_files", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_fullscreen", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_gamepads", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_gaming_mode", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_keyboard_button", bool, True, "",
+               client=True),
+    SettingDef("ui_sidebar_show_screen_settings", bool, True, "",
+               client=True),
+    SettingDef("ui_sidebar_show_sharing", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_shortcuts", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_soft_buttons", bool, True, "",
+               client=True),
+    SettingDef("ui_sidebar_show_stats", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_trackpad", bool, True, "", client=True),
+    SettingDef("ui_sidebar_show_video_settings", bool, True, "",
+               client=True),
 
     # ---- multi-GPU / scaling ----
     SettingDef("session_gpus", str, "",

@@ -42,7 +42,10 @@ class CentralizedStreamServer:
             allow_upload="upload" in dirs,
             allow_download="download" in dirs)
         from .transfers import UplinkAllowance
-        self.allowance = UplinkAllowance(self.transfers.pacer)
+        limit = settings.file_transfer_limit_mbps
+        self.allowance = UplinkAllowance(
+            self.transfers.pacer,
+            cap_rate=(limit * 125_000.0 if limit > 0 else 100_000_000.0))
         self._uplink_task = None
         from concurrent.futures import ThreadPoolExecutor
         # single worker => rows land in arrival order (reference contract)
@@ -78,6 +81,10 @@ class CentralizedStreamServer:
         app = self.app
         app.router.add_get("/websockets", self._ws_entry)
         app.router.add_get("/ws", self._ws_entry)
+        sub_ws = self.settings.subfolder.strip("/")
+        if sub_ws:
+            app.router.add_get(f"/{sub_ws}/websockets", self._ws_entry)
+            app.router.add_get(f"/{sub_ws}/ws", self._ws_entry)
         app.router.add_get("/api/status", self.handle_status)
         app.router.add_get("/api/health", self.handle_health)
         app.router.add_get("/api/stats", self.handle_stats)
@@ -92,20 +99,31 @@ class CentralizedStreamServer:
         app.router.add_post("/api/upload", self.handle_upload)
         app.router.add_get("/api/download", self.handle_download)
         app.router.add_get("/api/files", self.handle_files)
-        web_dir = os.path.join(os.path.dirname(__file__), "web")
+        web_dir = self.settings.web_root or os.path.join(
+            os.path.dirname(__file__), "web")
         if os.path.isdir(web_dir):
             app.router.add_get("/", self._index)
             app.router.add_get("/dashboard", self._dashboard)
             app.router.add_static("/static", web_dir)
+        # reverse-proxy subpath: mirror the app under /<subfolder>/
+        sub = self.settings.subfolder.strip("/")
+        if sub:
+            app.router.add_get(f"/{sub}", self._index)
+            app.router.add_get(f"/{sub}/", self._index)
+            if os.path.isdir(web_dir):
+                app.router.add_static(f"/{sub}/static", web_dir)
+
+    def _web_dir(self):
+        return self.settings.web_root or os.path.join(
+            os.path.dirname(__file__), "web")
 
     async def _index(self, request):
-        path = os.path.join(os.path.dirname(__file__), "web", "index.html")
-        return web.FileResponse(path)
+        return web.FileResponse(os.path.join(self._web_dir(),
+                                             "index.html"))
 
     async def _dashboard(self, request):
-        path = os.path.join(os.path.dirname(__file__), "web",
-                            "dashboard.html")
-        return web.FileResponse(path)
+        return web.FileResponse(os.path.join(self._web_dir(),
+                                             "dashboard.html"))
 
     async def _ws_entry(self, request):
         if not self._ws_origin_allowed(request):
@@ -148,6 +166,15 @@ class CentralizedStreamServer:
                         hdr[6:]).decode().partition(":")
                     ok = (hmac.compare_digest(user, s.basic_auth_user) and
                           hmac.compare_digest(pw, s.basic_auth_password))
+                    # second password grants view-only access (reference
+                    # basic_auth_viewonly_password)
+                    if (not ok and s.basic_auth_viewonly_password and
+                            hmac.compare_digest(
+                                user, s.basic_auth_user) and
+                            hmac.compare_digest(
+                                pw, s.basic_auth_viewonly_password)):
+                        ok = True
+                        request["forced_role"] = "viewer"
                 except Exception:
                     ok = False
             if not ok:
@@ -324,7 +351,8 @@ class CentralizedStreamServer:
 
     async def _watch_certs(self):
         while True:
-            await asyncio.sleep(5)
+            await asyncio.sleep(
+                max(5, int(self.settings.cert_reload_interval)))
             s = self.settings
             if not (s.enable_https and s.https_cert):
                 continue

@@ -123,7 +123,8 @@ class StreamingService:
         self.gamepads = None
         if settings.enable_gamepad:
             from .gamepad import GamepadHub
-            self.gamepads = GamepadHub()
+            self.gamepads = GamepadHub(
+                socket_dir=settings.js_socket_path)
 
     @property
     def capture(self):
@@ -175,6 +176,7 @@ class StreamingService:
         cs.paint_over_trigger_frames = s.paint_over_trigger_frames
         cs.video_paintover_crf = s.video_paintover_crf
         cs.video_paintover_burst_frames = s.video_paintover_burst_frames
+        cs.jpeg_paintover_quality = s.paint_over_jpeg_quality
         cs.damage_block_threshold = s.damage_block_threshold
         cs.damage_block_duration = s.damage_block_duration
         cs.jpeg_quality = s.jpeg_quality
@@ -209,11 +211,33 @@ class StreamingService:
             logger.warning("watermark conversion failed: %r", exc)
             return None
 
+    @staticmethod
+    def _run_hook(cmd: str):
+        """Fire-and-forget session lifecycle hook (reference
+        run_after_connect / run_after_disconnect)."""
+        if not cmd:
+            return
+        import subprocess
+        try:
+            subprocess.Popen(cmd, shell=True,
+                             stdout=subprocess.DEVNULL,
+                             stderr=subprocess.DEVNULL)
+        except Exception as exc:
+            logger.warning("lifecycle hook failed: %r", exc)
+
     def start_capture(self, display: str = "primary"):
         cap = self.captures.get(display)
         if cap is not None and cap.is_capturing:
             return
         self.loop = asyncio.get_running_loop()
+        if (self.settings.app_wait_ready and self.settings.app_ready_file
+                and not os.path.exists(self.settings.app_ready_file)):
+            # poll until the session app marks itself ready (reference
+            # app_ready_file/app_wait_ready contract)
+            logger.info("waiting for app ready file %s",
+                        self.settings.app_ready_file)
+            self.loop.call_later(0.5, self.start_capture, display)
+            return
         cap = hipflux.ScreenCapture()
         self.captures[display] = cap
         loop = self.loop
@@ -226,9 +250,13 @@ class StreamingService:
                                       frame_id, y, is_keyframe,
                                       capture_ts_ms)
 
-        if display == "primary" and not self.settings.capture_cursor:
+        if (display == "primary" and not self.settings.capture_cursor
+                and self.settings.enable_cursors):
             # client-side cursor rendering: push shape updates over control
             def on_cursor(w, h, hx, hy, argb):
+                if self.settings.debug_cursors:
+                    logger.info("cursor shape %dx%d hot(%d,%d)", w, h, hx,
+                                hy)
                 loop.call_soon_threadsafe(self._broadcast_cursor, w, h, hx,
                                           hy, argb)
             cap.set_cursor_callback(on_cursor)
@@ -359,6 +387,8 @@ class StreamingService:
         # role/slot policy (reference roles: one controller, shared viewers;
         # signaling_server.py allowed_client_slots behavior)
         want_role = request.query.get("role", "")
+        if request.get("forced_role") == "viewer":
+            want_role = "viewer"
         has_controller = any(c.role == "controller"
                              for c in self.clients.values())
         if want_role == "viewer":
@@ -377,7 +407,10 @@ class StreamingService:
             bitrate_bps=self.settings.video_bitrate_kbps * 1000.0)
         relay.start()
         state = ClientState(ws, relay, display, role)
+        first_client = not self.clients
         self.clients[ws] = state
+        if first_client:
+            self._run_hook(self.settings.run_after_connect)
 
         try:
             await ws.send_str(P.encode_control("MODE", "websockets"))
@@ -407,6 +440,8 @@ class StreamingService:
                     break
         finally:
             self.clients.pop(ws, None)
+            if not self.clients:
+                self._run_hook(self.settings.run_after_disconnect)
             await relay.stop()
             if state.player_seat is not None and self.gamepads is not None:
                 # free the seat and unplug its pad slot
@@ -563,7 +598,10 @@ class StreamingService:
         if not self.settings.enable_resize:
             return
         from . import display_utils
-        w, h = display_utils.align_dims_16(w, h)
+        if self.settings.force_aligned_resolution:
+            w, h = display_utils.align_dims_16(w, h)
+        else:
+            w, h = max(2, w & ~1), max(2, h & ~1)  # even at minimum
         self.settings.set("resolution", f"{w}x{h}")
         # resize the real display when capturing X11
         backend = self.settings.capture_backend

@@ -297,6 +297,16 @@ function applyServerSettings(payload) {
   set("encoder", "encoder");
   set("fps", "framerate");
   set("crf", "video_crf");
+  /* UI contract knobs (reference ui_*): title + chrome visibility */
+  if (payload.ui_title && payload.ui_title.value)
+    document.title = payload.ui_title.value;
+  const hud = document.getElementById("hud");
+  if (hud && payload.ui_show_sidebar)
+    hud.style.display = payload.ui_show_sidebar.value ? "" : "none";
+  const statsEl = document.getElementById("stats");
+  if (statsEl && payload.ui_sidebar_show_stats)
+    statsEl.style.display =
+        payload.ui_sidebar_show_stats.value ? "" : "none";
   if (payload.audio_codec && payload.audio_codec.value) {
     const c = payload.audio_codec.value;
     if (c !== audioCodec) opusDec = null;   /* codec switch: fresh state */

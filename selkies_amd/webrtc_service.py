@@ -52,7 +52,9 @@ class WebRTCService:
         self.peers: dict[tuple, PeerState] = {}
         self.transport = None
         self.port = 0
-        self.host_ip = ice.default_host_ip()
+        # NAT'd hosts advertise the configured public IP in candidates
+        self.host_ip = (settings.webrtc_public_ip or
+                        ice.default_host_ip())
         self.capture: Optional[hipflux.ScreenCapture] = None
         # certificate fingerprints signaled in accepted offers; DTLS clients
         # whose cert does not hash to one of these are rejected (RFC 8122)
