@@ -400,3 +400,113 @@ def test_dashboard_page_served():
             await runner.cleanup()
 
     asyncio.run(main())
+
+
+def test_viewonly_password_forces_viewer(loop):
+    """basic_auth_viewonly_password: valid second password authenticates
+    but the WS session is pinned to the viewer role (reference
+    basic_auth_viewonly_password)."""
+    import base64
+
+    async def main():
+        server = make_server(SELKIES_ENABLE_BASIC_AUTH="true",
+                             SELKIES_BASIC_AUTH_PASSWORD="main",
+                             SELKIES_BASIC_AUTH_VIEWONLY_PASSWORD="view")
+        runner, port = await start_on_free_port(server)
+
+        def hdr(pw):
+            tok = base64.b64encode(f"selkies:{pw}".encode()).decode()
+            return {"Authorization": f"Basic {tok}"}
+
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.get(f"http://127.0.0.1:{port}/api/status",
+                                    headers=hdr("wrong")) as r:
+                    assert r.status == 401
+                async with sess.get(f"http://127.0.0.1:{port}/api/status",
+                                    headers=hdr("view")) as r:
+                    assert r.status == 200
+                # WS under the viewonly password: server assigns viewer
+                # even though this is the first (would-be controller) client
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws",
+                        headers=hdr("view")) as ws:
+                    role = None
+                    for _ in range(20):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if (msg.type == WSMsgType.TEXT and
+                                msg.data.startswith("ROLE,")):
+                            role = msg.data.split(",", 1)[1]
+                            break
+                    assert role == "viewer"
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_subfolder_routes(loop):
+    async def main():
+        server = make_server(SELKIES_SUBFOLDER="desk1")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                for path in ("/desk1", "/desk1/static/selkies-client.js"):
+                    async with sess.get(
+                            f"http://127.0.0.1:{port}{path}") as r:
+                        assert r.status == 200, path
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_run_after_connect_and_disconnect_hooks(loop, tmp_path):
+    mark_c = tmp_path / "connected"
+    mark_d = tmp_path / "disconnected"
+
+    async def main():
+        server = make_server(
+            SELKIES_RUN_AFTER_CONNECT=f"touch {mark_c}",
+            SELKIES_RUN_AFTER_DISCONNECT=f"touch {mark_d}")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    await asyncio.wait_for(ws.receive(), 5)
+            for _ in range(50):
+                if mark_c.exists() and mark_d.exists():
+                    break
+                await asyncio.sleep(0.1)
+            assert mark_c.exists() and mark_d.exists()
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
+
+
+def test_ui_settings_in_client_payload(loop):
+    async def main():
+        server = make_server(SELKIES_UI_TITLE="My Desk")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    payload = None
+                    for _ in range(20):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if (msg.type == WSMsgType.TEXT and
+                                msg.data.startswith("SETTINGS_PAYLOAD,")):
+                            payload = json.loads(
+                                msg.data.split(",", 1)[1])
+                            break
+                    assert payload is not None
+                    assert payload["ui_title"]["value"] == "My Desk"
+                    assert payload["ui_show_sidebar"]["value"] is True
+                    assert payload["ui_sidebar_show_stats"]["value"] is True
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
