@@ -226,7 +226,7 @@ def main():
     n_sess = max(1, args.sessions)
     depth = args.pipeline_depth
     if depth is None:
-        depth = 2 if (use_gpu and out_mode == 1) else 1
+        depth = 2 if (use_gpu and out_mode in (1, 2)) else 1
     pipes = [_native.BenchPipeline(kind, args.width, enc_h, qp=args.qp,
                                    stripe_height=64, output_mode=out_mode,
                                    gpu_id=local_rank if use_gpu else -1,

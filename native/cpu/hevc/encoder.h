@@ -18,6 +18,8 @@
 //    emission) — the stripe-parallel seam of SURVEY.md §5.7.
 #pragma once
 
+#include <cstdlib>
+
 #include <cstdint>
 #include <memory>
 #include <vector>
@@ -36,7 +38,17 @@ struct EncodeStats {
 // left-neighbor chain (GPU wavefront) and multiply CABAC parallelism, at
 // ~0.1-0.5% bitrate cost per extra segment boundary.
 inline int default_slices_per_row(int width) {
-  return width >= 3840 ? 16 : width >= 2560 ? 4 : width >= 1280 ? 2 : 1;
+  // HIPFLUX_HEVC_SPR overrides for tuning runs (CABAC jobs are serial
+  // per slice segment, so more segments = shorter GPU critical path at
+  // ~10 bytes/slice header cost)
+  if (const char* e = std::getenv("HIPFLUX_HEVC_SPR")) {
+    int v = std::atoi(e);
+    if (v >= 1 && v <= 256) return v;
+  }
+  // measured on MI355X (profiles/NOTES.md): the CABAC critical path
+  // is serial per segment, fps plateaus at ~32 segments/row at 4K
+  // for ~2% per-frame bitrate cost
+  return width >= 3840 ? 32 : width >= 2560 ? 8 : width >= 1280 ? 2 : 1;
 }
 
 class StripeEncoder {
