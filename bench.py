@@ -295,6 +295,48 @@ def main():
         if distributed:
             dist.barrier()
 
+    # background gpu_busy sampler (weak-telemetry fix: a single external
+    # SMI probe can miss the burst; sample sysfs through the timed region
+    # and report the mean alongside)
+    class BusySampler:
+        def __init__(self, dev):
+            import glob as _g
+            self.path = None
+            self.samples = []
+            self.stop = False
+            self.thread = None
+            for card in sorted(_g.glob(
+                    "/sys/class/drm/card*/device/gpu_busy_percent")):
+                self.path = card
+                break
+
+        def __enter__(self):
+            if self.path is None or not use_gpu:
+                return self
+            import threading as _t
+
+            def loop():
+                while not self.stop:
+                    try:
+                        with open(self.path) as f:
+                            self.samples.append(int(f.read().strip()))
+                    except OSError:
+                        return
+                    time.sleep(0.02)
+
+            self.thread = _t.Thread(target=loop, daemon=True)
+            self.thread.start()
+            return self
+
+        def __exit__(self, *a):
+            self.stop = True
+            if self.thread:
+                self.thread.join(timeout=1)
+
+        def mean(self):
+            return (round(sum(self.samples) / len(self.samples), 1)
+                    if self.samples else None)
+
     # warmup (untimed): includes the IDR and allocator/registration warmup
     for p_ in pipes:
         p_.encode(frames[0], True)
@@ -302,6 +344,8 @@ def main():
             p_.encode(frames[(i + 1) % n_src], False)
 
     sync()
+    busy = BusySampler(local_rank if use_gpu else -1)
+    busy.__enter__()
     lat_ms = []
     total_bytes = 0
     if n_sess == 1:
@@ -359,6 +403,7 @@ def main():
             lat_ms.extend(q)
         total_bytes = sum(nb)
 
+
     if not lat_ms:
         lat_ms = [0.0]
     elapsed = t1 - t0
@@ -370,6 +415,7 @@ def main():
     fps_job = args.steps * n_sess * (1 if tile else world) / elapsed
     p50 = float(np.percentile(lat_ms, 50))
     p95 = float(np.percentile(lat_ms, 95))
+    busy.__exit__()
     if distributed:
         t = torch.tensor([p50, p95], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
@@ -405,6 +451,7 @@ def main():
                 "content": args.content,
                 "pipeline": pipe.pipeline,
                 "pipeline_depth": depth,
+                "gpu_busy_mean_pct": busy.mean(),
                 "latency_p50_ms": round(p50, 3),
                 "tile_boundary_exchange": (
                     {"rows": boundary_rows, "bytes": boundary_bytes,

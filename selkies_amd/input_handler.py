@@ -63,6 +63,40 @@ class RecordingBackend(InputBackend):
         pass
 
 
+class SpareKeycodePool:
+    """Rotating allocator of spare keycodes for keysyms absent from the
+    current keymap (reference input_handler.py:1052-1160 overlay): any
+    keycode whose mapping is entirely NoSymbol is claimable; entries
+    rotate LRU when the pool is exhausted so rare glyphs keep typing.
+    Pure-python and unit-testable; the X calls are injected."""
+
+    def __init__(self, keycodes: list[int], remap):
+        # keycodes: spare keycode numbers; remap(kc, keysym) performs the
+        # XChangeKeyboardMapping side effect
+        self._free = list(keycodes)
+        self._remap = remap
+        self._by_keysym: dict[int, int] = {}
+        self._order: list[int] = []          # keysyms, oldest first
+
+    def keycode_for(self, keysym: int):
+        kc = self._by_keysym.get(keysym)
+        if kc is not None:
+            self._order.remove(keysym)
+            self._order.append(keysym)
+            return kc
+        if self._free:
+            kc = self._free.pop(0)
+        elif self._order:
+            evict = self._order.pop(0)
+            kc = self._by_keysym.pop(evict)
+        else:
+            return None
+        self._remap(kc, keysym)
+        self._by_keysym[keysym] = kc
+        self._order.append(keysym)
+        return kc
+
+
 class XTestBackend(InputBackend):
     """XTEST injection via ctypes against libX11 + libXtst.
 
@@ -85,11 +119,21 @@ class XTestBackend(InputBackend):
             raise RuntimeError(f"cannot open X display {display!r}")
         self._setup_protos()
         self._keysym_cache: dict[int, int] = {}
+        self._spares = self._build_spare_pool()
 
     def _setup_protos(self):
         d = ctypes.c_void_p
         self._x11.XKeysymToKeycode.argtypes = [d, ctypes.c_ulong]
         self._x11.XKeysymToKeycode.restype = ctypes.c_ubyte
+        self._x11.XDisplayKeycodes.argtypes = [d, ctypes.c_void_p,
+                                               ctypes.c_void_p]
+        self._x11.XGetKeyboardMapping.argtypes = [d, ctypes.c_ubyte,
+                                                  ctypes.c_int,
+                                                  ctypes.c_void_p]
+        self._x11.XChangeKeyboardMapping.argtypes = [d, ctypes.c_int,
+                                                     ctypes.c_int,
+                                                     ctypes.c_void_p,
+                                                     ctypes.c_int]
         self._x11.XFlush.argtypes = [d]
         self._xtst.XTestFakeKeyEvent.argtypes = [d, ctypes.c_uint,
                                                  ctypes.c_int, ctypes.c_ulong]
@@ -104,10 +148,45 @@ class XTestBackend(InputBackend):
                                                             ctypes.c_int,
                                                             ctypes.c_ulong]
 
+    def _build_spare_pool(self):
+        """Scan the keymap for keycodes mapped entirely to NoSymbol."""
+        try:
+            mn, mx = ctypes.c_int(), ctypes.c_int()
+            self._x11.XDisplayKeycodes(self._dpy, ctypes.byref(mn),
+                                       ctypes.byref(mx))
+            n = mx.value - mn.value + 1
+            per = ctypes.c_int()
+            self._x11.XGetKeyboardMapping.restype = ctypes.POINTER(
+                ctypes.c_ulong)
+            syms = self._x11.XGetKeyboardMapping(
+                self._dpy, ctypes.c_ubyte(mn.value), n, ctypes.byref(per))
+            spares = []
+            for i in range(n):
+                row = syms[i * per.value:(i + 1) * per.value]
+                if all(v == 0 for v in row):
+                    spares.append(mn.value + i)
+            self._x11.XFree(ctypes.cast(syms, ctypes.c_void_p))
+
+            def remap(kc, keysym):
+                arr = (ctypes.c_ulong * 1)(keysym)
+                self._x11.XChangeKeyboardMapping(self._dpy, kc, 1, arr, 1)
+                self._x11.XFlush(self._dpy)
+                # keysym lookups may now resolve differently
+                self._keysym_cache.clear()
+
+            return SpareKeycodePool(spares, remap)
+        except Exception as exc:
+            logger.debug("spare keycode scan failed: %r", exc)
+            return SpareKeycodePool([], lambda kc, ks: None)
+
     def _keycode(self, keysym: int) -> int:
         kc = self._keysym_cache.get(keysym)
         if kc is None:
             kc = self._x11.XKeysymToKeycode(self._dpy, keysym)
+            if not kc:
+                # keysym absent from the keymap: overlay it onto a spare
+                # keycode so it becomes typeable
+                kc = self._spares.keycode_for(keysym) or 0
             self._keysym_cache[keysym] = kc
         return kc
 

@@ -237,6 +237,9 @@ window.skSend = send;   /* dashboard hook */
 
 function hookInput() {
   canvas.addEventListener("keydown", (ev) => {
+    /* browser auto-repeat is suppressed: the server X autorepeat is the
+       single repeat source for XTEST-held keys (double-repeat fix) */
+    if (ev.repeat) { ev.preventDefault(); return; }
     const ks = keysymOf(ev);
     if (ks !== undefined) { send("kd," + ks); ev.preventDefault(); }
   });

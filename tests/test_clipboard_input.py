@@ -69,3 +69,29 @@ def test_key_release_all_on_reset():
     d.on_message("kr,")
     ups = [e for e in backend.events if e[0] == "key" and not e[2]]
     assert ("key", 65, False) in ups and ("key", 66, False) in ups
+
+
+def test_spare_keycode_pool_rotation():
+    """Unmapped keysyms claim spare keycodes; LRU rotation when the pool
+    is exhausted (reference keysym->spare-keycode overlay)."""
+    from selkies_amd.input_handler import SpareKeycodePool
+
+    remaps = []
+    pool = SpareKeycodePool([250, 251], lambda kc, ks: remaps.append(
+        (kc, ks)))
+    assert pool.keycode_for(0x1001000) == 250
+    assert pool.keycode_for(0x1001001) == 251
+    # cached: no new remap
+    assert pool.keycode_for(0x1001000) == 250
+    assert remaps == [(250, 0x1001000), (251, 0x1001001)]
+    # pool full: evict the LRU entry (0x1001001 after the re-touch above)
+    assert pool.keycode_for(0x1001002) == 251
+    assert remaps[-1] == (251, 0x1001002)
+    # the evicted keysym re-claims via another eviction
+    assert pool.keycode_for(0x1001001) == 250
+
+
+def test_spare_keycode_pool_empty():
+    from selkies_amd.input_handler import SpareKeycodePool
+    pool = SpareKeycodePool([], lambda kc, ks: None)
+    assert pool.keycode_for(0x100) is None
