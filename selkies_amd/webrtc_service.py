@@ -19,6 +19,7 @@ from typing import Optional
 import hipflux
 
 from .webrtc import dtls, ice, rtp, sdp
+from .webrtc.pacer import AUDIO, VIDEO, Pacer
 from .webrtc.sctp import PPID_STRING, PPID_STRING_EMPTY, SctpAssociation
 from .webrtc.srtp import SrtpSession, is_rtcp
 
@@ -60,6 +61,7 @@ class WebRTCService:
         # whose cert does not hash to one of these are rejected (RFC 8122)
         self.allowed_fingerprints: set[str] = set()
         self.audio_capture = None
+        self.pacer = None
         self._audio_enabled = False
         self._ts_base = time.monotonic()
         self.frames_sent = 0
@@ -71,12 +73,22 @@ class WebRTCService:
         self.transport, _ = await loop.create_datagram_endpoint(
             lambda: _Proto(self), local_addr=("0.0.0.0", port))
         self.port = self.transport.get_extra_info("sockname")[1]
+        # strict-priority token-bucket pacer: smooths keyframe bursts at
+        # 1.25x the video target (+ audio headroom); audio skips ahead
+        self.pacer = Pacer(
+            lambda pkt, addr: self.transport.sendto(pkt, addr),
+            rate_bytes_per_s=self.settings.video_bitrate_kbps * 125.0 *
+            1.25 + 32_000)
+        self.pacer.start()
         logger.info("webrtc ice-lite endpoint on %s:%d", self.host_ip,
                     self.port)
 
     async def stop(self):
         self.stop_video()
         self.stop_audio()
+        if getattr(self, "pacer", None) is not None:
+            await self.pacer.stop()
+            self.pacer = None
         if self.transport:
             self.transport.close()
             self.transport = None
@@ -169,8 +181,8 @@ class WebRTCService:
             if not peer.connected:
                 continue
             try:
-                self.transport.sendto(peer.srtp_out.protect_rtp(pkt),
-                                      peer.addr)
+                self.pacer.enqueue(AUDIO, peer.srtp_out.protect_rtp(pkt),
+                                   peer.addr)
             except Exception as exc:
                 logger.debug("srtp audio send failed: %r", exc)
 
@@ -186,8 +198,9 @@ class WebRTCService:
                 continue
             for pkt in packets:
                 try:
-                    self.transport.sendto(peer.srtp_out.protect_rtp(pkt),
-                                          peer.addr)
+                    self.pacer.enqueue(VIDEO,
+                                       peer.srtp_out.protect_rtp(pkt),
+                                       peer.addr)
                 except Exception as exc:
                     logger.debug("srtp send failed: %r", exc)
             if now - peer.last_sr > 1.0:
@@ -346,6 +359,8 @@ class WebRTCService:
                     self.capture.update_video_bitrate(kbps)
                 except Exception:
                     pass
+            if getattr(self, "pacer", None) is not None:
+                self.pacer.set_rate(kbps * 125.0 * 1.25 + 32_000)
 
     def dc_broadcast(self, text: str):
         """Send a control message to every open data channel."""
