@@ -278,7 +278,65 @@ function hookInput() {
 
 /* ---------------- settings UI ---------------- */
 
+/* ---------------- file transfers ----------------
+ * Upload: streamed POST /api/upload?name= (staging-rename server side);
+ * Download: /api/files listing -> /api/download?name= links
+ * (reference web-core lib/file-upload.js surface). */
+function hookTransfers() {
+  const input = document.getElementById("upload-input");
+  const btn = document.getElementById("upload-btn");
+  const filesBtn = document.getElementById("files-btn");
+  const list = document.getElementById("file-list");
+  const prog = document.getElementById("upload-progress");
+  if (!input || !btn) return;
+  btn.onclick = () => input.click();
+  input.onchange = async () => {
+    for (const f of input.files) {
+      prog.style.display = "";
+      prog.value = 0;
+      try {
+        await new Promise((resolve, reject) => {
+          /* XHR for upload progress events (fetch lacks them) */
+          const xhr = new XMLHttpRequest();
+          xhr.open("POST",
+                   "/api/upload?name=" + encodeURIComponent(f.name));
+          xhr.upload.onprogress = (ev) => {
+            if (ev.lengthComputable)
+              prog.value = (100 * ev.loaded / ev.total) | 0;
+          };
+          xhr.onload = () => xhr.status < 300 ? resolve()
+                                              : reject(xhr.statusText);
+          xhr.onerror = reject;
+          xhr.send(f);
+        });
+      } catch (e) { console.warn("upload failed", e); }
+    }
+    prog.style.display = "none";
+    input.value = "";
+  };
+  if (filesBtn)
+    filesBtn.onclick = async () => {
+      if (list.style.display === "") { list.style.display = "none"; return; }
+      try {
+        const r = await fetch("/api/files");
+        const files = await r.json();
+        list.innerHTML = "";
+        for (const f of files.files || files) {
+          const name = f.name || f;
+          const li = document.createElement("li");
+          const a = document.createElement("a");
+          a.href = "/api/download?name=" + encodeURIComponent(name);
+          a.textContent = name + (f.size != null ? ` (${f.size} B)` : "");
+          li.appendChild(a);
+          list.appendChild(li);
+        }
+        list.style.display = "";
+      } catch (e) { console.warn("file list failed", e); }
+    };
+}
+
 function hookHud() {
+  hookTransfers();
   const enc = document.getElementById("encoder");
   const fps = document.getElementById("fps");
   const crf = document.getElementById("crf");
@@ -310,6 +368,10 @@ function applyServerSettings(payload) {
   if (statsEl && payload.ui_sidebar_show_stats)
     statsEl.style.display =
         payload.ui_sidebar_show_stats.value ? "" : "none";
+  const trEl = document.getElementById("transfers");
+  if (trEl && payload.ui_sidebar_show_files)
+    trEl.style.display =
+        payload.ui_sidebar_show_files.value ? "" : "none";
   if (payload.audio_codec && payload.audio_codec.value) {
     const c = payload.audio_codec.value;
     if (c !== audioCodec) opusDec = null;   /* codec switch: fresh state */

@@ -510,3 +510,32 @@ def test_ui_settings_in_client_payload(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_transfer_ui_and_file_roundtrip(loop, tmp_path):
+    """The index page ships the transfer panel; upload -> list ->
+    download round-trips through the API the panel drives."""
+    async def main():
+        server = make_server(SELKIES_UPLOAD_DIR=str(tmp_path))
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.get(f"http://127.0.0.1:{port}/") as r:
+                    page = await r.text()
+                    assert "upload-btn" in page and "file-list" in page
+                async with sess.post(
+                        f"http://127.0.0.1:{port}/api/upload?name=a.txt",
+                        data=b"hello transfers") as r:
+                    assert r.status == 200
+                async with sess.get(
+                        f"http://127.0.0.1:{port}/api/files") as r:
+                    files = await r.json()
+                    assert any(f["name"] == "a.txt" for f in files)
+                async with sess.get(
+                        f"http://127.0.0.1:{port}/api/download?name=a.txt"
+                ) as r:
+                    assert await r.read() == b"hello transfers"
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
