@@ -335,8 +335,30 @@ function hookTransfers() {
     };
 }
 
+/* clipboard sync worker (reference web-core lib/clipboard-sync.js):
+ * on focus, read the local clipboard (permission-gated) and push it to
+ * the server when it changed; server->client writes land in
+ * lastRemoteClipboard so we don't echo them back. */
+let lastSentClipboard = null, lastRemoteClipboard = null;
+
+function hookClipboardSync() {
+  const push = async () => {
+    if (!navigator.clipboard || !navigator.clipboard.readText) return;
+    try {
+      const text = await navigator.clipboard.readText();
+      if (!text || text === lastSentClipboard ||
+          text === lastRemoteClipboard) return;
+      lastSentClipboard = text;
+      send("cw," + btoa(unescape(encodeURIComponent(text))));
+    } catch (e) { /* permission denied: stay quiet */ }
+  };
+  window.addEventListener("focus", push);
+  canvas.addEventListener("mouseenter", push);
+}
+
 function hookHud() {
   hookTransfers();
+  hookClipboardSync();
   const enc = document.getElementById("encoder");
   const fps = document.getElementById("fps");
   const crf = document.getElementById("crf");
@@ -464,6 +486,7 @@ function connect() {
       } else if (verb === "clipboard") {
         try {
           const text = decodeURIComponent(escape(atob(rest)));
+          lastRemoteClipboard = text;
           navigator.clipboard.writeText(text).catch(() => {});
           if (window.skOnClipboard) window.skOnClipboard(text);
         } catch (e) {}
