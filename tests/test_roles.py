@@ -73,3 +73,46 @@ def test_controller_assignment_and_promotion(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_concurrent_join_exactly_one_controller(loop):
+    """Slot enforcement under a simultaneous join burst: exactly one of
+    N concurrent connects wins the controller role (reference
+    reconnection/slot semantics; weak spot flagged in review)."""
+    async def main():
+        server = make_server()
+        server.streaming.input.backend = RecordingBackend()
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                url = f"http://127.0.0.1:{port}/websockets"
+                wss = await asyncio.gather(
+                    *[sess.ws_connect(url) for _ in range(5)])
+                roles = await asyncio.gather(
+                    *[read_role(ws) for ws in wss])
+                assert roles.count("controller") == 1, roles
+                assert roles.count("viewer") == 4
+                # the controller disconnects; someone else is promoted,
+                # still exactly one
+
+                async def next_role(ws):
+                    try:
+                        return await read_role(ws, timeout=10)
+                    except (asyncio.TimeoutError, Exception):
+                        return None
+
+                idx = roles.index("controller")
+                await wss[idx].close()
+                later = await asyncio.gather(
+                    *[next_role(ws) for i, ws in enumerate(wss)
+                      if i != idx])
+                assert later.count("controller") == 1, later
+                for i, ws in enumerate(wss):
+                    if i != idx:
+                        await ws.close()
+        finally:
+            server.streaming.stop_capture()
+            server.streaming.stop_audio()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
