@@ -104,10 +104,16 @@ static void mdct_forward(const float* in, float* out, int n) {
 // ---- PVQ (textbook CWRS) -------------------------------------------------
 
 // V(n, k): count of integer vectors of dim n with |.|_1 == k, as double
-// (for bit estimates / split decisions).
+// (for bit estimates / split decisions). Memoized: the per-band pulse
+// binary search re-queries the same (n, k) pairs every frame and the
+// O(n*k) DP was a measurable share of the encode budget.
 static double pvq_v_d(int n, int k) {
   if (k == 0) return 1;
   if (n == 0) return 0;
+  static thread_local std::unordered_map<uint32_t, double> memo;
+  const uint32_t key = (uint32_t)n << 16 | (uint32_t)k;
+  auto it = memo.find(key);
+  if (it != memo.end()) return it->second;
   std::vector<double> cur(k + 1, 0.0), prev;
   cur[0] = 1;
   for (int i = 1; i <= k; ++i) cur[i] = 2;   // n = 1
@@ -117,6 +123,7 @@ static double pvq_v_d(int n, int k) {
     for (int i = 1; i <= k; ++i)
       cur[i] = prev[i] + cur[i - 1] + prev[i - 1];
   }
+  memo.emplace(key, cur[k]);
   return cur[k];
 }
 
