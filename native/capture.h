@@ -46,4 +46,8 @@ std::unique_ptr<FrameSource> make_synthetic_source(int width, int height,
 std::unique_ptr<FrameSource> make_x11_source(const std::string& display,
                                              int x, int y, int w, int h);
 
+// Seqlock shared-memory frame source ("shm:<path>" capture backend):
+// the Wayland-compositor capture seam and a generic producer bridge.
+std::unique_ptr<FrameSource> make_shm_source(const std::string& path);
+
 }  // namespace hipflux

@@ -157,6 +157,14 @@ void ScreenCapture::run() {
                    settings_.display.c_str());
     }
   }
+  if (!src && settings_.capture_backend.rfind("shm:", 0) == 0) {
+    // shared-memory producer (the Wayland compositor capture seam)
+    src = make_shm_source(settings_.capture_backend.substr(4));
+    if (!src)
+      std::fprintf(stderr,
+                   "hipflux: shm capture source unavailable, falling back "
+                   "to synthetic\n");
+  }
   if (!src) {
     // synthetic pattern override: "synthetic:<pattern>" in capture_backend
     std::string pattern = "desktop";

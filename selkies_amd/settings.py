@@ -249,6 +249,11 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                value_range=(1, 256)),
     SettingDef("debug", bool, False, "Verbose logging."),
     # ---- app lifecycle (reference app_ready/run_after hooks) ----
+    SettingDef("use_wayland", bool, False,
+               "Capture from the in-tree headless Wayland compositor "
+               "instead of X11 (engine shm capture seam)."),
+    SettingDef("wayland_display", str, "selkies-wl-0",
+               "WAYLAND_DISPLAY name the in-tree compositor binds."),
     SettingDef("app_ready_file", str, "",
                "Path whose existence marks the session app as ready."),
     SettingDef("app_wait_ready", bool, False,

@@ -157,7 +157,12 @@ class StreamingService:
             cs.gpu_id = gpus[idx % len(gpus)]
         else:
             cs.gpu_id = s.gpu_id if s.gpu_id >= 0 else 0
-        cs.capture_backend = s.capture_backend
+        if s.use_wayland:
+            # in-tree compositor -> seqlock shm file -> native engine
+            # (reference use_wayland capture contract)
+            cs.capture_backend = "shm:" + self._ensure_wayland_shm()
+        else:
+            cs.capture_backend = s.capture_backend
         cs.display = s.display
         cs.video_bitrate_kbps = s.video_bitrate_kbps
         cs.video_crf = s.video_crf
@@ -210,6 +215,20 @@ class StreamingService:
         except Exception as exc:
             logger.warning("watermark conversion failed: %r", exc)
             return None
+
+    def _ensure_wayland_shm(self) -> str:
+        if getattr(self, "_wl_shm_path", None):
+            return self._wl_shm_path
+        import tempfile
+        from .wayland import compositor as wl
+        w, h = self.settings.resolution_wh
+        wl.ensure_wayland_display(self.settings.wayland_display, w, h)
+        comp = wl.get_compositor()
+        path = os.path.join(tempfile.gettempdir(),
+                            f"selkies-wl-fb-{os.getpid()}.shm")
+        comp.start_shm_publisher(path, fps=float(self.settings.framerate))
+        self._wl_shm_path = path
+        return path
 
     @staticmethod
     def _run_hook(cmd: str):

@@ -700,6 +700,54 @@ class Compositor:
         os.close(r)
         return out
 
+    # ---- engine capture bridge --------------------------------------------
+    def start_shm_publisher(self, path: str, fps: float = 60.0):
+        """Publish composited frames into a seqlock shm file the native
+        engine maps via its "shm:<path>" capture backend (header: magic
+        'HFSH', seq, w, h; even seq = stable — native/capture_shm.cpp)."""
+        import mmap
+        import struct as _st
+        out = self.outputs[0]
+        size = 16 + out.w * out.h * 4
+        with open(path, "wb") as f:
+            f.write(_st.pack("<IIII", 0x48534648, 0, out.w, out.h))
+            f.truncate(size)
+        f = open(path, "r+b")
+        mm = mmap.mmap(f.fileno(), size)
+        self._shm_stop = False
+
+        def loop():
+            seq = 0
+            period = 1.0 / max(1.0, fps)
+            while not self._shm_stop:
+                t0 = time.monotonic()
+                try:
+                    _, _, fb = self.composite_frame()
+                except Exception:
+                    break
+                seq += 1                       # odd: writing
+                mm[4:8] = _st.pack("<I", seq)
+                mm[16:16 + len(fb)] = fb
+                seq += 1                       # even: stable
+                mm[4:8] = _st.pack("<I", seq)
+                dt = period - (time.monotonic() - t0)
+                if dt > 0:
+                    time.sleep(dt)
+            mm.close()
+            f.close()
+
+        self._shm_thread = threading.Thread(target=loop, daemon=True,
+                                            name="wl-shm-pub")
+        self._shm_thread.start()
+        return path
+
+    def stop_shm_publisher(self):
+        self._shm_stop = True
+        t = getattr(self, "_shm_thread", None)
+        if t is not None:
+            t.join(timeout=2)
+            self._shm_thread = None
+
     # ---- capture seam ------------------------------------------------------
     def composite_frame(self, output_name=None):
         """Render mapped surfaces' committed shm buffers into a BGRX
