@@ -278,6 +278,47 @@ function hookInput() {
 
 /* ---------------- settings UI ---------------- */
 
+/* ---------------- microphone uplink ----------------
+ * getUserMedia -> 48 kHz mono s16 frames over binary [0x02]+PCM
+ * (server writes them into the virtual microphone sink when
+ * enable_microphone is on). */
+let micStream = null, micCtx = null, micNode = null;
+
+async function micStart() {
+  if (micStream) return true;
+  try {
+    micStream = await navigator.mediaDevices.getUserMedia({ audio: {
+      channelCount: 1, sampleRate: 48000 } });
+    micCtx = new AudioContext({ sampleRate: 48000 });
+    const src = micCtx.createMediaStreamSource(micStream);
+    micNode = micCtx.createScriptProcessor(2048, 1, 1);
+    micNode.onaudioprocess = (ev) => {
+      if (!ws || ws.readyState !== 1) return;
+      const f = ev.inputBuffer.getChannelData(0);
+      const out = new Uint8Array(1 + f.length * 2);
+      out[0] = 0x02;
+      const dv = new DataView(out.buffer);
+      for (let i = 0; i < f.length; i++) {
+        let v = Math.max(-1, Math.min(1, f[i]));
+        dv.setInt16(1 + i * 2, v * 32767, true);
+      }
+      ws.send(out);
+    };
+    src.connect(micNode);
+    micNode.connect(micCtx.destination);
+    return true;
+  } catch (e) { console.warn("mic start failed", e); return false; }
+}
+
+function micStop() {
+  if (micNode) { micNode.disconnect(); micNode = null; }
+  if (micCtx) { micCtx.close().catch(() => {}); micCtx = null; }
+  if (micStream) {
+    for (const t of micStream.getTracks()) t.stop();
+    micStream = null;
+  }
+}
+
 /* ---------------- file transfers ----------------
  * Upload: streamed POST /api/upload?name= (staging-rename server side);
  * Download: /api/files listing -> /api/download?name= links
@@ -369,6 +410,12 @@ function hookHud() {
   crf.onchange = () => send('SETTINGS,' +
       JSON.stringify({ video_crf: +crf.value }));
   document.getElementById("idr").onclick = requestIdr;
+  const micBtn = document.getElementById("mic");
+  if (micBtn)
+    micBtn.onclick = async () => {
+      if (micStream) { micStop(); micBtn.textContent = "mic off"; }
+      else if (await micStart()) micBtn.textContent = "mic on";
+    };
 }
 
 function applyServerSettings(payload) {

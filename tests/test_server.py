@@ -539,3 +539,29 @@ def test_transfer_ui_and_file_roundtrip(loop, tmp_path):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_mic_uplink_binary_frames(loop):
+    """Client mic PCM (binary 0x02) lands in the virtual microphone sink
+    when enable_microphone is on."""
+    async def main():
+        server = make_server(SELKIES_ENABLE_MICROPHONE="true")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    await asyncio.wait_for(ws.receive(), 5)
+                    pcm = bytes(2048)
+                    await ws.send_bytes(b"\x02" + pcm)
+                    for _ in range(100):
+                        sink = server.streaming.mic_sink
+                        if sink is not None and sink.buffered >= 2048:
+                            break
+                        await asyncio.sleep(0.02)
+                    assert server.streaming.mic_sink is not None
+                    assert server.streaming.mic_sink.buffered >= 2048
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
