@@ -202,7 +202,7 @@ def main():
     if distributed:
         dist.init_process_group("nccl" if use_gpu else "gloo")
     if use_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
 
     import hipflux
     from hipflux import _native
@@ -227,9 +227,13 @@ def main():
     depth = args.pipeline_depth
     if depth is None:
         depth = 2 if (use_gpu and out_mode in (1, 2)) else 1
+    # ranks wrap over the visible devices (driver runs nproc == n_gpus,
+    # so this is identity there; on smaller boxes extra ranks share)
+    dev = (local_rank % max(1, torch.cuda.device_count())
+           if use_gpu else -1)
     pipes = [_native.BenchPipeline(kind, args.width, enc_h, qp=args.qp,
                                    stripe_height=64, output_mode=out_mode,
-                                   gpu_id=local_rank if use_gpu else -1,
+                                   gpu_id=dev,
                                    pipeline_depth=depth)
              for _ in range(n_sess)]
     pipe = pipes[0]
@@ -245,7 +249,8 @@ def main():
     if tile and use_gpu and distributed:
         uid = [_native.TileComm.make_uid() if rank == 0 else None]
         dist.broadcast_object_list(uid, src=0)
-        comm = _native.TileComm(rank, world, uid[0], local_rank)
+        comm = _native.TileComm(rank, world, uid[0],
+                                local_rank % max(1, torch.cuda.device_count()))
 
     def tile_exchange(frame_np):
         nonlocal boundary_bytes
