@@ -375,6 +375,12 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&up_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&db_stream_, hipStreamNonBlocking));
+    for (int i = 0; i < 2; ++i) {
+      HIP_CHECK(hipEventCreateWithFlags(&ev_rows_[i],
+                                        hipEventDisableTiming));
+      HIP_CHECK(hipEventCreateWithFlags(&ev_db_[i], hipEventDisableTiming));
+    }
     stripe_h_ = std::max(16, s.stripe_height & ~15);
     batch_events_.resize(4);
     for (auto& e : batch_events_)
@@ -392,6 +398,12 @@ class HipH264Pipeline : public EncodePipeline {
     (void)hipStreamSynchronize(stream_);
     (void)hipStreamSynchronize(up_stream_);
     (void)hipStreamDestroy(up_stream_);
+    (void)hipStreamSynchronize(db_stream_);
+    (void)hipStreamDestroy(db_stream_);
+    for (int i = 0; i < 2; ++i) {
+      (void)hipEventDestroy(ev_rows_[i]);
+      (void)hipEventDestroy(ev_db_[i]);
+    }
     for (auto& e : batch_events_) (void)hipEventDestroy(e);
     for (int i = 0; i < 2; ++i) {
       (void)hipEventDestroy(ev_done_[i]);
@@ -602,11 +614,24 @@ class HipH264Pipeline : public EncodePipeline {
     }
     // in-loop deblock of the current recon (within-slice edges only;
     // idc=2 is signaled in the slice headers) before it becomes the
-    // reference frame
-    if (settings_.video_deblock)
+    // reference frame. Deblock (reads recon+levels) and CAVLC (reads
+    // levels) are independent, so deblock runs on its own stream
+    // concurrently with entropy + bitstream D2H; the cur->ref copies
+    // below join both streams before the recon becomes the reference.
+    const bool forked_db = settings_.video_deblock && !cpu_entropy_;
+    if (settings_.video_deblock) {
+      hipStream_t db_s = forked_db ? db_stream_ : stream_;
+      if (forked_db) {
+        HIP_CHECK(hipEventRecord(ev_rows_[par], stream_));
+        HIP_CHECK(hipStreamWaitEvent(db_stream_, ev_rows_[par], 0));
+      }
       h264gpu::launch_h264_deblock(d_curY_, d_curCb_, d_curCr_, ypitch_,
                                    cpitch_, mbw_, n_jobs, d_jobs_[par],
-                                   d_levels_, d_meta_, stream_);
+                                   d_levels_, d_meta_, db_s);
+      if (forked_db) {
+        HIP_CHECK(hipEventRecord(ev_db_[par], db_stream_));
+      }
+    }
     const int copy_words = ent_copy_words_;
     if (!cpu_entropy_) {
       h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs,
@@ -632,6 +657,7 @@ class HipH264Pipeline : public EncodePipeline {
     }
     // refresh ref from cur for encoded stripes (merge contiguous spans so
     // the all-stripes case is 3 copies, not 3 x n_stripes)
+    if (forked_db) HIP_CHECK(hipStreamWaitEvent(stream_, ev_db_[par], 0));
     {
       int span_y0 = -1, span_y1 = -1;
       auto flush_span = [&] {
@@ -1028,6 +1054,7 @@ class HipH264Pipeline : public EncodePipeline {
   ThreadPool pool_;
   hipStream_t stream_{};
   hipStream_t up_stream_{};
+  hipStream_t db_stream_{};
   int stripe_h_ = 64;
   int w_ = 0, h_ = 0, mbw_ = 0, mbh_ = 0, ypitch_ = 0, cpitch_ = 0;
   int segs_ = 1, seg_w0_ = 0;     // slices per MB row, widest segment
@@ -1058,6 +1085,7 @@ class HipH264Pipeline : public EncodePipeline {
   uint32_t* h_entout_[2] = {nullptr, nullptr};
   int* h_outbits_[2] = {nullptr, nullptr};
   hipEvent_t ev_done_[2] = {}, ev_h2d_[2] = {};
+  hipEvent_t ev_rows_[2] = {}, ev_db_[2] = {};
   int depth_ = 1;                  // 1 = sync (latency mode), 2 = pipelined
   int parity_ = 0;
   Pending prev_;                   // the in-flight frame (depth 2)
