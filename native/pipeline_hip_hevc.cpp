@@ -44,6 +44,10 @@ class HipHevcPipeline : public EncodePipeline {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&copy_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&rows_stream_, hipStreamNonBlocking));
+    for (int i = 0; i < 2; ++i)
+      HIP_CHECK(hipEventCreateWithFlags(&ev_rows_[i],
+                                        hipEventDisableTiming));
     cpu_entropy_ = std::getenv("HIPFLUX_CPU_HEVC_ENTROPY") != nullptr;
     timing_ = std::getenv("HIPFLUX_TIMES") != nullptr;
     for (auto& e : ev_) HIP_CHECK(hipEventCreate(&e));
@@ -74,6 +78,9 @@ class HipHevcPipeline : public EncodePipeline {
     if (h_meta_) (void)hipHostFree(h_meta_);
     if (h_stage_) (void)hipHostFree(h_stage_);
     (void)hipStreamDestroy(copy_stream_);
+    (void)hipStreamSynchronize(rows_stream_);
+    (void)hipStreamDestroy(rows_stream_);
+    for (int i = 0; i < 2; ++i) (void)hipEventDestroy(ev_rows_[i]);
   }
 
   struct StripeRef {
@@ -120,12 +127,15 @@ class HipHevcPipeline : public EncodePipeline {
       std::memcpy(h_stage_, src, frame_bytes);
       src = h_stage_;
     }
+    // frame N+1's upload/CSC/rows run on rows_stream_, concurrent with
+    // frame N's CABAC on stream_ (levels/meta are parity
+    // double-buffered; the recon planes have no cross-frame reader)
     HIP_CHECK(hipMemcpyAsync(d_frame_[par], src, frame_bytes,
-                             hipMemcpyHostToDevice, stream_));
-    HIP_CHECK(hipEventRecord(ev_h2d_[par], stream_));
+                             hipMemcpyHostToDevice, rows_stream_));
+    HIP_CHECK(hipEventRecord(ev_h2d_[par], rows_stream_));
     launch_bgrx_to_planes(d_frame_[par], w_, h_, frame.stride / 4, d_srcY_,
                           d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
-                          stream_);
+                          rows_stream_);
 
     // ---- job list (must mirror the CPU StripeEncoder's segmentation)
     Pending pd;
@@ -166,24 +176,26 @@ class HipHevcPipeline : public EncodePipeline {
     pd.active = true;
     parity_ ^= 1;
 
-    if (timing_) HIP_CHECK(hipEventRecord(ev_[0], stream_));
+    if (timing_) HIP_CHECK(hipEventRecord(ev_[0], rows_stream_));
     HIP_CHECK(hipMemcpyAsync(d_jobs_[par], h_jobs_[par],
                              sizeof(hevcgpu::HevcJob) * n_jobs,
-                             hipMemcpyHostToDevice, stream_));
+                             hipMemcpyHostToDevice, rows_stream_));
     hevcgpu::launch_hevc_rows(d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_,
                               w_, h_, d_curY_, d_curCb_, d_curCr_, ctbw_,
-                              n_jobs, d_jobs_[par], d_levels_, d_meta_,
-                              stream_);
+                              n_jobs, d_jobs_[par], d_levels_[par],
+                              d_meta_[par], rows_stream_);
+    HIP_CHECK(hipEventRecord(ev_rows_[par], rows_stream_));
+    HIP_CHECK(hipStreamWaitEvent(stream_, ev_rows_[par], 0));
     if (cpu_entropy_) {
-      HIP_CHECK(hipMemcpyAsync(h_levels_, d_levels_, levels_bytes_,
+      HIP_CHECK(hipMemcpyAsync(h_levels_, d_levels_[par], levels_bytes_,
                                hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipMemcpyAsync(h_meta_, d_meta_, meta_bytes_,
+      HIP_CHECK(hipMemcpyAsync(h_meta_, d_meta_[par], meta_bytes_,
                                hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipEventRecord(ev_done_[par], stream_));
       return pd;
     }
     if (timing_) HIP_CHECK(hipEventRecord(ev_[1], stream_));
-    hevcgpu::launch_hevc_cabac(d_levels_, d_meta_, ctbw_, n_jobs,
+    hevcgpu::launch_hevc_cabac(d_levels_[par], d_meta_[par], ctbw_, n_jobs,
                                d_jobs_[par], d_out_[par], out_stride_,
                                d_counts_[par], stream_);
     if (timing_) HIP_CHECK(hipEventRecord(ev_[2], stream_));
@@ -337,9 +349,10 @@ class HipHevcPipeline : public EncodePipeline {
                         hipMemcpyDeviceToHost));
     d.levels.resize(levels_bytes_ / sizeof(int16_t));
     d.meta.resize(meta_bytes_ / sizeof(int));
-    HIP_CHECK(hipMemcpy(d.levels.data(), d_levels_, levels_bytes_,
-                        hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(d.meta.data(), d_meta_, meta_bytes_,
+    const int last_par = parity_ ^ 1;   // parity of the last submit
+    HIP_CHECK(hipMemcpy(d.levels.data(), d_levels_[last_par],
+                        levels_bytes_, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(d.meta.data(), d_meta_[last_par], meta_bytes_,
                         hipMemcpyDeviceToHost));
     return true;
   }
@@ -394,8 +407,10 @@ class HipHevcPipeline : public EncodePipeline {
     const size_t n_ctu = static_cast<size_t>(ctbw_) * ctbh_;
     levels_bytes_ = n_ctu * hevcgpu::kHevcLevelsPerCtu * sizeof(int16_t);
     meta_bytes_ = n_ctu * hevcgpu::kHevcMetaPerCtu * sizeof(int);
-    d_levels_ = dalloc<int16_t>(n_ctu * hevcgpu::kHevcLevelsPerCtu);
-    d_meta_ = dalloc<int>(n_ctu * hevcgpu::kHevcMetaPerCtu);
+    for (int i = 0; i < 2; ++i) {
+      d_levels_[i] = dalloc<int16_t>(n_ctu * hevcgpu::kHevcLevelsPerCtu);
+      d_meta_[i] = dalloc<int>(n_ctu * hevcgpu::kHevcMetaPerCtu);
+    }
 
     const int spr = hevc::default_slices_per_row(w);
     max_jobs_ = ctbh_ * spr + 8;
@@ -427,9 +442,10 @@ class HipHevcPipeline : public EncodePipeline {
 
   CaptureSettings settings_;
   ThreadPool pool_;
-  hipStream_t stream_{}, copy_stream_{};
+  hipStream_t stream_{}, copy_stream_{}, rows_stream_{};
   hipEvent_t ev_[4] = {};
   hipEvent_t ev_h2d_[2] = {}, ev_done_[2] = {};
+  hipEvent_t ev_rows_[2] = {};
   bool cpu_entropy_ = false;
   bool timing_ = false;
   int stripe_h_ = 64;
@@ -443,8 +459,8 @@ class HipHevcPipeline : public EncodePipeline {
   uint8_t* d_frame_[2] = {};
   uint8_t *d_srcY_ = nullptr, *d_srcCb_ = nullptr, *d_srcCr_ = nullptr;
   uint8_t *d_curY_ = nullptr, *d_curCb_ = nullptr, *d_curCr_ = nullptr;
-  int16_t* d_levels_ = nullptr;
-  int* d_meta_ = nullptr;
+  int16_t* d_levels_[2] = {};
+  int* d_meta_[2] = {};
   hevcgpu::HevcJob* d_jobs_[2] = {};
   uint8_t* d_out_[2] = {};
   int* d_counts_[2] = {};
