@@ -48,7 +48,9 @@ inline int default_slices_per_row(int width) {
   // measured on MI355X (profiles/NOTES.md): the CABAC critical path
   // is serial per segment, fps plateaus at ~32 segments/row at 4K
   // for ~2% per-frame bitrate cost
-  return width >= 3840 ? 32 : width >= 2560 ? 8 : width >= 1280 ? 2 : 1;
+  // ~120 px per segment at 4K+ (32 segments at 3840, 64 at 7680)
+  if (width >= 3840) return width / 120;
+  return width >= 2560 ? 8 : width >= 1280 ? 2 : 1;
 }
 
 class StripeEncoder {
