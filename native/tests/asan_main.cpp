@@ -13,6 +13,7 @@
 #include "../cpu/h264/encoder.h"
 #include "../cpu/h264/gpu_entropy.h"
 #include "../cpu/jpeg_enc.h"
+#include "../cpu/opus/celt.h"
 #include "../cpu/scale.h"
 
 using namespace hipflux;
@@ -68,6 +69,44 @@ int main() {
     int ow, oh, ostride;
     box_downscale_bgrx(src.data(), w * 4, w, h, div, out, ow, oh, ostride);
     if (ow != w / div || oh != h / div) return 3;
+  }
+  // fractional bilinear downscale across awkward ratios
+  for (float sc : {0.26f, 0.5f, 0.667f, 0.99f}) {
+    int w = 103, h = 57;
+    std::vector<uint8_t> src(static_cast<size_t>(w) * h * 4);
+    for (auto& b : src) b = static_cast<uint8_t>(rnd());
+    std::vector<uint8_t> out;
+    int ow, oh, ostride;
+    bilinear_downscale_bgrx(src.data(), w * 4, w, h, sc, out, ow, oh,
+                            ostride);
+    if (ow <= 0 || oh <= 0) return 4;
+  }
+  // Hi444 fullcolor encode (mono planes x3, IDR + P with residual
+  // fallback) under the sanitizers
+  {
+    int w = 80, h = 48;
+    h264::StripeEncoder enc(w, h, true, true);
+    std::vector<uint8_t> y(static_cast<size_t>(w) * h),
+        cb(y.size()), cr(y.size());
+    for (int f = 0; f < 3; ++f) {
+      for (auto& v : y) v = static_cast<uint8_t>(rnd());
+      for (auto& v : cb) v = static_cast<uint8_t>(rnd());
+      for (auto& v : cr) v = static_cast<uint8_t>(rnd());
+      std::vector<uint8_t> out;
+      enc.encode_frame(y.data(), w, cb.data(), cr.data(), w, 24, f == 0,
+                       out);
+      if (out.empty()) return 5;
+    }
+  }
+  // Opus encode (range coder + MDCT + PVQ) under the sanitizers
+  {
+    opus::CeltEncoder enc(128000);
+    std::vector<int16_t> pcm(960 * 2);
+    for (int f = 0; f < 4; ++f) {
+      for (auto& v : pcm) v = static_cast<int16_t>(rnd());
+      auto pkt = enc.encode_frame(pcm.data(), 2);
+      if (pkt.empty()) return 6;
+    }
   }
   // NAL assembly: zero-heavy bit patterns hit the emulation-prevention path
   for (int t = 0; t < 50; ++t) {
