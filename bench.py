@@ -109,7 +109,10 @@ def run_e2e(args):
     g2g = st["glass_to_glass_ms"]
     fps = n_acked / max(elapsed, 1e-9)
     print(json.dumps({
-        "metric": "delivered_fps_glass_to_glass_1080p60",
+        "metric": ("delivered_fps_glass_to_glass_"
+                   + ("2160p" if args.height >= 2160 else
+                      f"{args.height}p") + "60"
+                   + ("_hevc" if args.encoder.startswith("hevc") else "")),
         "value": round(fps, 2),
         "unit": "frames/s",
         "n_gpus": 1,
