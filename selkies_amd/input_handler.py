@@ -295,7 +295,9 @@ class InputDispatcher:
                  on_clipboard: Optional[Callable[[str], None]] = None,
                  clipboard_read: Optional[Callable[[], str]] = None,
                  enable_input: bool = True,
-                 enable_clipboard: bool = True):
+                 enable_clipboard: bool = True,
+                 enable_binary_clipboard: bool = False,
+                 on_clipboard_binary=None):
         self.backend = backend
         self.on_resize = on_resize
         self.on_dpi = on_dpi
@@ -305,6 +307,9 @@ class InputDispatcher:
         self.clipboard_read = clipboard_read
         self.enable_input = enable_input
         self.enable_clipboard = enable_clipboard
+        self.enable_binary_clipboard = enable_binary_clipboard
+        self.on_clipboard_binary = on_clipboard_binary
+        self._mp = None
         self._held: dict[int, float] = {}
         self._button_mask = 0
 
@@ -386,6 +391,8 @@ class InputDispatcher:
                 data = self.clipboard_read() or ""
                 return "clipboard," + base64.b64encode(
                     data.encode()).decode()
+        elif verb in ("cws", "cbs", "cwd", "cbd", "cwe", "cbe", "cb"):
+            return self._clipboard_multipart(verb, rest)
         elif verb == "co":                      # atomic char typing
             # "co,<base64 text>": type text by keysym per char
             import base64
@@ -425,6 +432,91 @@ class InputDispatcher:
             if changed & bit:
                 self.backend.mouse_button(button, bool(mask & bit))
         self._button_mask = mask
+
+    # ---- multipart clipboard (reference cws/cwd/cwe text,
+    # cbs/cbd/cbe binary, cb single binary): chunked base64 transfers
+    # with declared-size bounds and abort-on-mismatch semantics -------------
+    MULTIPART_CLIPBOARD_MAX = 8 * 1024 * 1024
+
+    def _mp_reset(self):
+        self._mp = None
+
+    def _clipboard_multipart(self, verb: str, rest: str):
+        import base64
+        if not self.enable_clipboard:
+            return None
+        toks = rest.split(",") if rest else []
+        if verb == "cb":                    # single binary write
+            if not self.enable_binary_clipboard:
+                return None
+            try:
+                mime, b64 = toks[0], toks[1]
+                data = base64.b64decode(b64)
+            except Exception:
+                return None
+            self._deliver_clipboard(mime, data)
+            return None
+        if verb in ("cws", "cbs"):
+            binary = verb == "cbs"
+            if binary and not self.enable_binary_clipboard:
+                logger.warning("rejecting binary clipboard: disabled")
+                return None
+            try:
+                tid = toks[0]
+                mime = toks[1] if binary else "text/plain"
+                total = int(toks[2] if binary else toks[1])
+            except Exception:
+                return None
+            if total < 0 or total > self.MULTIPART_CLIPBOARD_MAX:
+                logger.error("clipboard transfer size %d out of bounds",
+                             total)
+                return None
+            self._mp = {"id": tid, "binary": binary, "mime": mime,
+                        "total": total, "buf": bytearray()}
+            return None
+        mp = getattr(self, "_mp", None)
+        if mp is None:
+            return None
+        expected_binary = verb in ("cbd", "cbe")
+        if mp["binary"] != expected_binary or len(toks) < 1 or                 toks[0] != mp["id"]:
+            logger.warning("clipboard transfer mismatch; aborting")
+            self._mp_reset()
+            return None
+        if verb in ("cwd", "cbd"):
+            if len(toks) < 2:
+                self._mp_reset()
+                return None
+            try:
+                chunk = base64.b64decode(toks[1])
+            except Exception:
+                self._mp_reset()
+                return None
+            mp["buf"] += chunk
+            if len(mp["buf"]) > mp["total"]:
+                logger.error("clipboard transfer exceeded declared size")
+                self._mp_reset()
+            return None
+        # cwe / cbe: finalize
+        data = bytes(mp["buf"])
+        binary, mime = mp["binary"], mp["mime"]
+        self._mp_reset()
+        if len(data) != mp.get("total") and mp.get("total") is not None                 and len(data) != mp["total"]:
+            logger.warning("clipboard transfer size mismatch (%d != %d)",
+                           len(data), mp["total"])
+        self._deliver_clipboard(mime if binary else "text/plain", data,
+                                binary)
+        return None
+
+    def _deliver_clipboard(self, mime: str, data: bytes,
+                           binary: bool = True):
+        if not binary or mime.startswith("text/"):
+            if self.on_clipboard:
+                self.on_clipboard(data.decode("utf-8", "replace"))
+        elif self.on_clipboard_binary:
+            self.on_clipboard_binary(mime, data)
+        else:
+            logger.info("binary clipboard (%s, %d B) dropped: no sink",
+                        mime, len(data))
 
     def sweep_stale_keys(self):
         now = time.monotonic()

@@ -107,6 +107,7 @@ class StreamingService:
             clipboard_read=self.clipboard.read,
             enable_input=settings.enable_input,
             enable_clipboard=settings.enable_clipboard,
+            enable_binary_clipboard=settings.enable_binary_clipboard,
         )
         self._frame_clock = 0
         self._stats_task: Optional[asyncio.Task] = None

@@ -390,7 +390,19 @@ function hookClipboardSync() {
       if (!text || text === lastSentClipboard ||
           text === lastRemoteClipboard) return;
       lastSentClipboard = text;
-      send("cw," + btoa(unescape(encodeURIComponent(text))));
+      const b64 = btoa(unescape(encodeURIComponent(text)));
+      if (b64.length < 64 * 1024) {
+        send("cw," + b64);
+      } else {
+        /* multipart protocol for large payloads (cws/cwd/cwe) */
+        const tid = Date.now().toString(36);
+        const rawLen = Math.ceil(b64.length / 4) * 3 -
+            (b64.endsWith("==") ? 2 : b64.endsWith("=") ? 1 : 0);
+        send(`cws,${tid},${rawLen}`);
+        for (let i = 0; i < b64.length; i += 64 * 1024)
+          send(`cwd,${tid},${b64.slice(i, i + 64 * 1024)}`);
+        send(`cwe,${tid}`);
+      }
     } catch (e) { /* permission denied: stay quiet */ }
   };
   window.addEventListener("focus", push);

@@ -95,3 +95,75 @@ def test_spare_keycode_pool_empty():
     from selkies_amd.input_handler import SpareKeycodePool
     pool = SpareKeycodePool([], lambda kc, ks: None)
     assert pool.keycode_for(0x100) is None
+
+
+def _mk_dispatcher(**kw):
+    from selkies_amd.input_handler import InputDispatcher, RecordingBackend
+    writes = []
+    binaries = []
+    d = InputDispatcher(RecordingBackend(),
+                        on_clipboard=writes.append,
+                        on_clipboard_binary=lambda m, b: binaries.append(
+                            (m, b)),
+                        **kw)
+    return d, writes, binaries
+
+
+def test_multipart_text_clipboard():
+    """cws/cwd/cwe chunked text transfer (reference multipart clipboard
+    protocol): chunks reassemble, delivered once at cwe."""
+    import base64
+    d, writes, _ = _mk_dispatcher()
+    payload = "hello " * 1000
+    raw = payload.encode()
+    d.on_message(f"cws,t1,{len(raw)}")
+    for i in range(0, len(raw), 1024):
+        chunk = base64.b64encode(raw[i:i + 1024]).decode()
+        d.on_message(f"cwd,t1,{chunk}")
+        assert not writes                  # nothing delivered mid-transfer
+    d.on_message("cwe,t1")
+    assert writes == [payload]
+
+
+def test_multipart_binary_clipboard_gated_and_delivered():
+    import base64
+    # gate off: start rejected, chunks ignored
+    d, writes, binaries = _mk_dispatcher(enable_binary_clipboard=False)
+    d.on_message("cbs,t2,image/png,8")
+    d.on_message("cbd,t2," + base64.b64encode(b"PNGDATA!").decode())
+    d.on_message("cbe,t2")
+    assert not binaries
+    # gate on: delivered with mime
+    d, writes, binaries = _mk_dispatcher(enable_binary_clipboard=True)
+    d.on_message("cbs,t3,image/png,8")
+    d.on_message("cbd,t3," + base64.b64encode(b"PNGDATA!").decode())
+    d.on_message("cbe,t3")
+    assert binaries == [("image/png", b"PNGDATA!")]
+
+
+def test_multipart_clipboard_bounds_and_mismatch():
+    import base64
+    d, writes, _ = _mk_dispatcher()
+    # oversize declared: rejected at start
+    d.on_message(f"cws,big,{64 * 1024 * 1024}")
+    d.on_message("cwd,big," + base64.b64encode(b"x").decode())
+    d.on_message("cwe,big")
+    assert not writes
+    # id mismatch mid-transfer aborts
+    d.on_message("cws,a,4")
+    d.on_message("cwd,b," + base64.b64encode(b"zz").decode())
+    d.on_message("cwe,a")
+    assert not writes
+    # overflow beyond declared size aborts
+    d.on_message("cws,c,2")
+    d.on_message("cwd,c," + base64.b64encode(b"zzzz").decode())
+    d.on_message("cwe,c")
+    assert not writes
+
+
+def test_single_binary_cb_verb():
+    import base64
+    d, _, binaries = _mk_dispatcher(enable_binary_clipboard=True)
+    d.on_message("cb,application/x-test," +
+                 base64.b64encode(b"\x00\x01\x02").decode())
+    assert binaries == [("application/x-test", b"\x00\x01\x02")]
