@@ -69,6 +69,9 @@ class ClientState:
         self.last_sent_frame = -1
         self.ack_rtt_ms = 50.0
         self.paused = False
+        # per-client stream gates (reference START/STOP_VIDEO/AUDIO verbs)
+        self.video_stopped = False
+        self.audio_stopped = False
         self._sent_ts: dict[int, float] = {}
 
     def note_sent(self, frame_id: int):
@@ -338,7 +341,7 @@ class StreamingService:
         while True:
             data = await self._audio_queue.get()
             for cs in list(self.clients.values()):
-                if cs.relay.dead:
+                if cs.relay.dead or cs.audio_stopped:
                     continue
                 try:
                     await asyncio.wait_for(cs.ws.send_bytes(data), 1.0)
@@ -364,7 +367,7 @@ class StreamingService:
                     for k in sorted(self._frame_capture_ts)[:256]:
                         self._frame_capture_ts.pop(k, None)
         for cs in list(self.clients.values()):
-            if cs.relay.dead or cs.display != display:
+            if cs.relay.dead or cs.display != display or cs.video_stopped:
                 continue
             if cs.paused and not is_keyframe:
                 continue
@@ -512,8 +515,33 @@ class StreamingService:
             return P.encode_control(
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload())
-        if verb == "REQUEST_IDR":
+        if verb in ("REQUEST_IDR", "REQUEST_KEYFRAME"):
             self.request_idr()
+            return None
+        if verb == "STOP_VIDEO":
+            state.video_stopped = True
+            return P.encode_control("VIDEO_STOPPED", "")
+        if verb == "START_VIDEO":
+            state.video_stopped = False
+            self.request_idr(state.display)
+            return P.encode_control("VIDEO_STARTED", "")
+        if verb == "STOP_AUDIO":
+            state.audio_stopped = True
+            return P.encode_control("AUDIO_STOPPED", "")
+        if verb == "START_AUDIO":
+            state.audio_stopped = False
+            self.start_audio()
+            return P.encode_control("AUDIO_STARTED", "")
+        if verb == "REQUEST_CLIPBOARD":
+            reply = self.input.on_message("cr,")
+            return reply
+        if verb == "SET_NATIVE_CURSOR_RENDERING":
+            # 1 = composite the cursor into the video (native), 0 = push
+            # shapes for client-side rendering (reference verb)
+            want = rest.strip() in ("1", "true")
+            if want != self.settings.capture_cursor:
+                await self._apply_client_settings(
+                    {"capture_cursor": want})
             return None
         if verb == "CLAIM_SEAT":
             # a viewer claims gamepad seat n (1..3 = player2..4); granted

@@ -565,3 +565,59 @@ def test_mic_uplink_binary_frames(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_stream_control_verbs(loop):
+    """START/STOP_VIDEO and START/STOP_AUDIO gate the per-client streams
+    (reference control verbs); REQUEST_KEYFRAME aliases REQUEST_IDR."""
+    async def main():
+        server = make_server()
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    # receive some video first
+                    saw_video = False
+                    for _ in range(100):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if msg.type == aiohttp.WSMsgType.BINARY and \
+                                msg.data[0] in (3, 4, 6):
+                            saw_video = True
+                            break
+                    assert saw_video
+                    await ws.send_str("STOP_VIDEO,")
+                    # drain until the ack; then confirm video stops
+                    acked = False
+                    for _ in range(200):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if msg.type == aiohttp.WSMsgType.TEXT and \
+                                msg.data.startswith("VIDEO_STOPPED"):
+                            acked = True
+                            break
+                    assert acked
+                    # after the ack no more video frames arrive
+                    stray = 0
+                    try:
+                        for _ in range(30):
+                            msg = await asyncio.wait_for(ws.receive(), 0.2)
+                            if (msg.type == aiohttp.WSMsgType.BINARY and
+                                    msg.data[0] in (3, 4, 6)):
+                                stray += 1
+                    except asyncio.TimeoutError:
+                        pass
+                    assert stray == 0, stray
+                    await ws.send_str("START_VIDEO,")
+                    resumed = False
+                    for _ in range(200):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if msg.type == aiohttp.WSMsgType.BINARY and \
+                                msg.data[0] in (3, 4, 6):
+                            resumed = True
+                            break
+                    assert resumed
+                    await ws.send_str("REQUEST_KEYFRAME,")
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
