@@ -374,6 +374,32 @@ class StreamingService:
             if cs.relay.offer(data, y, is_keyframe):
                 cs.note_sent(frame_id)
 
+    async def _notify_pipeline_reset(self, displays):
+        """PIPELINE_RESETTING <display>: clients drop their per-row
+        decoders and frame-id state before the encoder restarts
+        (reference selkies.py:1955)."""
+        for cs in list(self.clients.values()):
+            for d in displays:
+                try:
+                    await cs.ws.send_str(f"PIPELINE_RESETTING {d}")
+                except Exception:
+                    pass
+
+    async def _notify_display_config(self):
+        """DISPLAY_CONFIG_UPDATE,{json}: current display layout pushed
+        after reconfiguration (reference selkies.py:1580)."""
+        displays = [{"display_id": name,
+                     "width": self.settings.resolution_wh[0],
+                     "height": self.settings.resolution_wh[1]}
+                    for name in (self.captures or {"primary": None})]
+        payload = json.dumps({"type": "display_config_update",
+                              "displays": displays})
+        for cs in list(self.clients.values()):
+            try:
+                await cs.ws.send_str("DISPLAY_CONFIG_UPDATE," + payload)
+            except Exception:
+                pass
+
     def _broadcast_cursor(self, w, h, hx, hy, argb):
         import base64
         msg = P.encode_control("CURSOR", {
@@ -628,10 +654,12 @@ class StreamingService:
         if structural and self.captures:
             logger.info("structural setting changed; restarting captures")
             displays = list(self.captures)
+            await self._notify_pipeline_reset(displays)
             self.stop_capture()
             for d in displays:
                 self.start_capture(d)
             self.request_idr()
+            await self._notify_display_config()
 
     def _tune(self, name: str, value):
         try:
@@ -657,6 +685,8 @@ class StreamingService:
             display_utils.resize_display(w, h, self.settings.display)
         if self.captures:
             displays = list(self.captures)
+            loop = asyncio.get_event_loop()
+            loop.create_task(self._notify_pipeline_reset(displays))
             self.stop_capture()
             for d in displays:
                 self.start_capture(d)

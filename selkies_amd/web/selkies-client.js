@@ -537,6 +537,18 @@ function connect() {
         /* hot-reloaded ICE config (server file monitor); used by the
            next RTCPeerConnection */
         try { window.skRtcConfig = JSON.parse(rest); } catch (e) {}
+      } else if (ev.data.startsWith("PIPELINE_RESETTING")) {
+        /* encoder restarted: drop per-row decoders + frame tracking so
+           the next IDR re-primes cleanly */
+        for (const row of h264Rows.values())
+          if (row.decoder) try { row.decoder.close(); } catch (e) {}
+        h264Rows.clear();
+        lastAckedFrame = -1;
+      } else if (verb === "DISPLAY_CONFIG_UPDATE") {
+        try {
+          if (window.skOnDisplayConfig)
+            window.skOnDisplayConfig(JSON.parse(rest));
+        } catch (e) {}
       } else if (verb === "SEAT") {
         /* player-seat grant/deny (reference PlayerGamepadButton) */
         if (window.skOnSeat) window.skOnSeat(parseInt(rest, 10));

@@ -621,3 +621,37 @@ def test_stream_control_verbs(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_pipeline_reset_and_display_config_notices(loop):
+    """Structural setting changes broadcast PIPELINE_RESETTING before
+    the restart and DISPLAY_CONFIG_UPDATE after (reference notices)."""
+    async def main():
+        server = make_server()
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    await asyncio.wait_for(ws.receive(), 5)
+                    await ws.send_str(
+                        'SETTINGS,{"video_fullframe": false}')
+                    saw_reset = saw_cfg = False
+                    for _ in range(300):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if msg.type != aiohttp.WSMsgType.TEXT:
+                            continue
+                        if msg.data.startswith("PIPELINE_RESETTING"):
+                            saw_reset = True
+                        if msg.data.startswith("DISPLAY_CONFIG_UPDATE,"):
+                            saw_cfg = True
+                            payload = json.loads(
+                                msg.data.split(",", 1)[1])
+                            assert payload["displays"][0]["width"] > 0
+                        if saw_reset and saw_cfg:
+                            break
+                    assert saw_reset and saw_cfg
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
