@@ -224,6 +224,11 @@ SETTING_DEFINITIONS: list[SettingDef] = [
     SettingDef("enable_clipboard", bool, True, "Bidirectional clipboard sync.",
                client=True),
     SettingDef("enable_gamepad", bool, True, "Gamepad passthrough (interposer/uinput)."),
+    SettingDef("enable_collab", bool, False,
+               "Per-user collab tokens (role/seat table with live "
+               "reconciliation)."),
+    SettingDef("master_token", str, "",
+               "Admin token guarding the /api/tokens collab table."),
     SettingDef("enable_shared", bool, False,
                "Allow multiple controlling clients (shared input)."),
     SettingDef("enable_player2", bool, False,

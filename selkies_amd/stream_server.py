@@ -93,6 +93,8 @@ class CentralizedStreamServer:
         app.router.add_get("/api/turn", self.handle_turn)
         app.router.add_post("/api/webrtc-stats", self.handle_webrtc_stats)
         app.router.add_post("/api/mode", self.handle_mode)
+        app.router.add_post("/api/tokens", self.handle_tokens)
+        app.router.add_get("/api/tokens", self.handle_tokens_list)
         from .computer_use import ComputerUseAPI
         ComputerUseAPI(self.settings, self.streaming.input).register(app)
         app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
@@ -274,6 +276,45 @@ class CentralizedStreamServer:
         self.settings.mode = mode
         return web.json_response({"mode": mode,
                                   "webrtc_active": self.webrtc is not None})
+
+    def _check_master_token(self, request):
+        s = self.settings
+        if not (s.enable_collab and s.master_token):
+            raise web.HTTPNotFound()
+        supplied = request.headers.get("Authorization", "")
+        if not hmac.compare_digest(supplied,
+                                   f"Bearer {s.master_token}"):
+            raise web.HTTPUnauthorized(reason="master token required")
+
+    async def handle_tokens(self, request):
+        """Replace/patch the collab token table and reconcile live
+        clients (reference user-token table + reconcile_clients). Body:
+        {"set": {token: {"role": "...", "seat": n|null}},
+         "revoke": [token, ...]}  — roles: controller|viewer."""
+        self._check_master_token(request)
+        try:
+            body = await request.json()
+        except Exception:
+            raise web.HTTPBadRequest(reason="JSON body required")
+        table = self.streaming.user_tokens
+        for tok, perms in (body.get("set") or {}).items():
+            role = perms.get("role", "viewer")
+            if role not in ("controller", "viewer"):
+                raise web.HTTPBadRequest(reason=f"bad role {role!r}")
+            seat = perms.get("seat")
+            if seat is not None and not (1 <= int(seat) <= 3):
+                raise web.HTTPBadRequest(reason="seat must be 1..3")
+            table[str(tok)] = {"role": role,
+                               "seat": int(seat) if seat else None}
+        for tok in body.get("revoke") or []:
+            table.pop(str(tok), None)
+        await self.streaming.reconcile_clients()
+        return web.json_response({"tokens": len(table)})
+
+    async def handle_tokens_list(self, request):
+        self._check_master_token(request)
+        return web.json_response(
+            {t: p for t, p in self.streaming.user_tokens.items()})
 
     # ---- WebRTC signaling ---------------------------------------------------
     async def handle_webrtc_offer(self, request):

@@ -65,6 +65,7 @@ class ClientState:
         # The controller implicitly owns seat 0. (Reference: signaling
         # player2-4 slots, signaling_server.py allowed_client_slots.)
         self.player_seat: Optional[int] = None
+        self.token: Optional[str] = None
         self.last_acked_frame = -1
         self.last_sent_frame = -1
         self.ack_rtt_ms = 50.0
@@ -124,6 +125,9 @@ class StreamingService:
         self._audio_queue: Optional[asyncio.Queue] = None
         self._audio_task: Optional[asyncio.Task] = None
         self.mic_sink: Optional[object] = None
+        # collab token table: token -> {"role": ..., "seat": int|None}
+        # (reference user_tokens + reconcile_clients)
+        self.user_tokens: dict[str, dict] = {}
         self.gamepads = None
         if settings.enable_gamepad:
             from .gamepad import GamepadHub
@@ -400,6 +404,37 @@ class StreamingService:
             except Exception:
                 pass
 
+    async def reconcile_clients(self):
+        """Walk live clients against the collab token table: disconnect
+        revoked tokens / role changes, push ROLE_UPDATE for seat-only
+        changes, re-announce MK_ACCESS (reference reconcile_clients,
+        selkies.py:5963)."""
+        for cs in list(self.clients.values()):
+            if cs.token is None:
+                continue
+            perms = self.user_tokens.get(cs.token)
+            if perms is None or perms.get("role") != cs.role:
+                try:
+                    await cs.ws.close(code=4002,
+                                      message=b"permissions changed")
+                except Exception:
+                    pass
+                continue
+            new_seat = perms.get("seat")
+            if new_seat != cs.player_seat:
+                cs.player_seat = new_seat
+                try:
+                    await cs.ws.send_str("ROLE_UPDATE," + json.dumps(
+                        {"role": cs.role, "slot": new_seat}))
+                except Exception:
+                    pass
+            try:
+                await cs.ws.send_str(
+                    "MK_ACCESS," +
+                    ("1" if cs.role == "controller" else "0"))
+            except Exception:
+                pass
+
     def _broadcast_cursor(self, w, h, hx, hy, argb):
         import base64
         msg = P.encode_control("CURSOR", {
@@ -438,6 +473,17 @@ class StreamingService:
         want_role = request.query.get("role", "")
         if request.get("forced_role") == "viewer":
             want_role = "viewer"
+        user_token = None
+        if self.settings.enable_collab:
+            t = request.query.get("utoken", "")
+            perms = self.user_tokens.get(t) if t else None
+            if self.user_tokens and perms is None:
+                # a populated table makes tokens mandatory
+                import aiohttp.web as _web
+                raise _web.HTTPForbidden(reason="collab token required")
+            if perms is not None:
+                user_token = t
+                want_role = perms.get("role", "viewer")
         has_controller = any(c.role == "controller"
                              for c in self.clients.values())
         if want_role == "viewer":
@@ -456,6 +502,7 @@ class StreamingService:
             bitrate_bps=self.settings.video_bitrate_kbps * 1000.0)
         relay.start()
         state = ClientState(ws, relay, display, role)
+        state.token = user_token
         first_client = not self.clients
         self.clients[ws] = state
         if first_client:
@@ -464,6 +511,8 @@ class StreamingService:
         try:
             await ws.send_str(P.encode_control("MODE", "websockets"))
             await ws.send_str(P.encode_control("ROLE", role))
+            await ws.send_str("MK_ACCESS," +
+                              ("1" if role == "controller" else "0"))
             await ws.send_str(P.encode_control(
                 "SETTINGS_PAYLOAD",
                 self.settings.build_client_settings_payload()))

@@ -116,3 +116,69 @@ def test_concurrent_join_exactly_one_controller(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_collab_tokens_roles_and_reconcile(loop):
+    """Collab token table: per-token role assignment at connect,
+    MK_ACCESS announcements, live revocation disconnects (reference
+    user_tokens + reconcile_clients)."""
+    import aiohttp as _aiohttp
+
+    async def main():
+        server = make_server(SELKIES_ENABLE_COLLAB="true",
+                             SELKIES_MASTER_TOKEN="mt")
+        server.streaming.input.backend = RecordingBackend()
+        runner, port = await start_on_free_port(server)
+        hdr = {"Authorization": "Bearer mt"}
+        try:
+            async with _aiohttp.ClientSession() as sess:
+                # mint tokens
+                r = await sess.post(
+                    f"http://127.0.0.1:{port}/api/tokens",
+                    json={"set": {"alice": {"role": "controller"},
+                                  "bob": {"role": "viewer", "seat": 1}}},
+                    headers=hdr)
+                assert r.status == 200
+                # wrong master token rejected
+                r = await sess.post(
+                    f"http://127.0.0.1:{port}/api/tokens", json={},
+                    headers={"Authorization": "Bearer nope"})
+                assert r.status == 401
+                # tokenless connect rejected while the table is populated
+                try:
+                    ws = await sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws")
+                    msg = await ws.receive(timeout=3)
+                    assert msg.type in (_aiohttp.WSMsgType.CLOSE,
+                                        _aiohttp.WSMsgType.CLOSED,
+                                        _aiohttp.WSMsgType.ERROR)
+                except _aiohttp.WSServerHandshakeError:
+                    pass
+                # bob (viewer token) connects FIRST but stays viewer
+                wsb = await sess.ws_connect(
+                    f"ws://127.0.0.1:{port}/ws?utoken=bob")
+                assert await read_role(wsb) == "viewer"
+                wsa = await sess.ws_connect(
+                    f"ws://127.0.0.1:{port}/ws?utoken=alice")
+                assert await read_role(wsa) == "controller"
+                # revoke alice -> her socket is closed by reconcile
+                r = await sess.post(
+                    f"http://127.0.0.1:{port}/api/tokens",
+                    json={"revoke": ["alice"]}, headers=hdr)
+                assert r.status == 200
+                closed = False
+                for _ in range(100):
+                    msg = await wsa.receive(timeout=5)
+                    if msg.type in (_aiohttp.WSMsgType.CLOSE,
+                                    _aiohttp.WSMsgType.CLOSED,
+                                    _aiohttp.WSMsgType.ERROR):
+                        closed = True
+                        break
+                assert closed
+                await wsb.close()
+        finally:
+            server.streaming.stop_capture()
+            server.streaming.stop_audio()
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
