@@ -109,6 +109,22 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                "Legacy long-term TURN password (when no shared secret)."),
     SettingDef("turn_rest_uri", str, "",
                "TURN-REST service URI for fetched ICE configs."),
+    SettingDef("turn_rest_api_key", str, "",
+               "Bearer token for the TURN-REST service."),
+    SettingDef("turn_rest_username", str, "",
+               "Username requested from the TURN-REST service."),
+    SettingDef("turn_rest_username_auth_header", str, "",
+               "Header name carrying the username to TURN-REST."),
+    SettingDef("turn_rest_protocol_header", str, "",
+               "Header name carrying the TURN protocol hint."),
+    SettingDef("turn_rest_tls_header", str, "",
+               "Header name carrying the TURN TLS hint."),
+    SettingDef("enable_cloudflare_turn", bool, False,
+               "Mint short-lived TURN credentials from Cloudflare."),
+    SettingDef("cloudflare_turn_token_id", str, "",
+               "Cloudflare TURN key id."),
+    SettingDef("cloudflare_turn_api_token", str, "",
+               "Cloudflare API token for TURN credential minting."),
     SettingDef("rtc_config_json", str, "",
                "Path to a JSON RTC config file (highest-priority ICE "
                "source; hot-reloaded on change)."),

@@ -129,20 +129,73 @@ def load_rtc_config_json(path: str) -> Optional[dict]:
 
 
 async def fetch_rest_config(uri: str, user: str = "selkies",
-                            timeout_s: float = 5.0) -> Optional[dict]:
+                            timeout_s: float = 5.0,
+                            api_key: str = "",
+                            username_header: str = "",
+                            protocol: str = "",
+                            protocol_header: str = "",
+                            tls: bool = False,
+                            tls_header: str = "") -> Optional[dict]:
     """RTC config from a TURN-REST service (reference RESTRTCMonitor):
-    GET <uri>?service=turn&username=<user> returning an iceServers doc."""
+    GET <uri>?service=turn&username=<user> returning an iceServers doc.
+    The reference's header knobs are honored: an API key rides as a
+    Bearer token, and the username/protocol/TLS hints go in the
+    service's configured header names."""
     import aiohttp
+    headers = {}
+    if api_key:
+        headers["Authorization"] = f"Bearer {api_key}"
+    if username_header:
+        headers[username_header] = user
+    if protocol_header and protocol:
+        headers[protocol_header] = protocol
+    if tls_header:
+        headers[tls_header] = "true" if tls else "false"
     try:
         async with aiohttp.ClientSession(
                 timeout=aiohttp.ClientTimeout(total=timeout_s)) as sess:
             async with sess.get(uri, params={"service": "turn",
-                                             "username": user}) as resp:
+                                             "username": user},
+                                headers=headers) as resp:
                 if resp.status != 200:
                     return None
                 cfg = await resp.json(content_type=None)
     except Exception:
         return None
+    return cfg if validate_rtc_config(cfg) else None
+
+
+async def fetch_cloudflare_turn(token_id: str, api_token: str,
+                                ttl_s: int = 86400,
+                                timeout_s: float = 5.0,
+                                endpoint: str = "") -> Optional[dict]:
+    """Short-lived TURN credentials from Cloudflare's TURN service
+    (reference enable_cloudflare_turn): POST to the credentials endpoint
+    with the API token; returns an rtc-config iceServers doc."""
+    import aiohttp
+    url = endpoint or (
+        "https://rtc.live.cloudflare.com/v1/turn/keys/"
+        f"{token_id}/credentials/generate")
+    try:
+        async with aiohttp.ClientSession(
+                timeout=aiohttp.ClientTimeout(total=timeout_s)) as sess:
+            async with sess.post(
+                    url, json={"ttl": ttl_s},
+                    headers={"Authorization": f"Bearer {api_token}"}
+            ) as resp:
+                if resp.status != 201 and resp.status != 200:
+                    return None
+                body = await resp.json(content_type=None)
+    except Exception:
+        return None
+    ice = body.get("iceServers") or {}
+    urls = ice.get("urls") or []
+    if not urls:
+        return None
+    cfg = {"lifetimeDuration": f"{ttl_s}s",
+           "iceServers": [{"urls": urls,
+                           "username": ice.get("username", ""),
+                           "credential": ice.get("credential", "")}]}
     return cfg if validate_rtc_config(cfg) else None
 
 
@@ -153,9 +206,25 @@ async def resolve_rtc_config(s, user: Optional[str] = None):
     cfg = load_rtc_config_json(getattr(s, "rtc_config_json", ""))
     if cfg is not None:
         return cfg, "file"
+    if getattr(s, "enable_cloudflare_turn", False) and \
+            getattr(s, "cloudflare_turn_token_id", "") and \
+            getattr(s, "cloudflare_turn_api_token", ""):
+        cfg = await fetch_cloudflare_turn(
+            s.cloudflare_turn_token_id, s.cloudflare_turn_api_token,
+            endpoint=getattr(s, "_cloudflare_endpoint", ""))
+        if cfg is not None:
+            return cfg, "cloudflare"
     rest = getattr(s, "turn_rest_uri", "")
     if rest:
-        cfg = await fetch_rest_config(rest, user)
+        cfg = await fetch_rest_config(
+            rest, getattr(s, "turn_rest_username", "") or user,
+            api_key=getattr(s, "turn_rest_api_key", ""),
+            username_header=getattr(s, "turn_rest_username_auth_header",
+                                    ""),
+            protocol=getattr(s, "turn_protocol", ""),
+            protocol_header=getattr(s, "turn_rest_protocol_header", ""),
+            tls=getattr(s, "turn_tls", False),
+            tls_header=getattr(s, "turn_rest_tls_header", ""))
         if cfg is not None:
             return cfg, "rest"
     if s.turn_host and s.turn_shared_secret:

@@ -203,3 +203,92 @@ def test_rtc_config_push_to_clients(tmp_path):
             await runner.cleanup()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_chain_rest_headers_and_api_key():
+    """TURN-REST header knobs (reference turn_rest_* settings): API key
+    as Bearer, username/protocol/TLS hints in configured headers."""
+    from aiohttp import web
+
+    async def main():
+        served = {"iceServers": [{"urls": ["turn:resthost:443"],
+                                  "username": "r", "credential": "r"}]}
+        seen = {}
+
+        async def handler(request):
+            seen.update({k: v for k, v in request.headers.items()
+                         if k.startswith("X-") or k == "Authorization"})
+            seen["username_q"] = request.query.get("username")
+            return web.json_response(served)
+
+        app = web.Application()
+        app.router.add_get("/turn", handler)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        port = runner.addresses[0][1]
+        try:
+            s = make_settings(
+                SELKIES_TURN_REST_URI=f"http://127.0.0.1:{port}/turn",
+                SELKIES_TURN_REST_API_KEY="sekrit",
+                SELKIES_TURN_REST_USERNAME="alice",
+                SELKIES_TURN_REST_USERNAME_AUTH_HEADER="X-Auth-User",
+                SELKIES_TURN_REST_PROTOCOL_HEADER="X-Turn-Protocol",
+                SELKIES_TURN_REST_TLS_HEADER="X-Turn-Tls",
+                SELKIES_TURN_PROTOCOL="tcp",
+                SELKIES_TURN_TLS="true")
+            cfg, source = await T.resolve_rtc_config(s)
+            assert source == "rest"
+            assert seen["Authorization"] == "Bearer sekrit"
+            assert seen["X-Auth-User"] == "alice"
+            assert seen["X-Turn-Protocol"] == "tcp"
+            assert seen["X-Turn-Tls"] == "true"
+            assert seen["username_q"] == "alice"
+        finally:
+            await runner.cleanup()
+
+    asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_chain_cloudflare_turn():
+    """Cloudflare TURN credential minting beats TURN-REST in the chain
+    and maps the response into an rtc config."""
+    from aiohttp import web
+
+    async def main():
+        async def handler(request):
+            assert request.headers["Authorization"] == "Bearer cftok"
+            body = await request.json()
+            assert body["ttl"] == 86400
+            return web.json_response({
+                "iceServers": {
+                    "urls": ["turn:turn.cloudflare.com:3478?transport=udp"],
+                    "username": "cfu", "credential": "cfc"}}, status=201)
+
+        app = web.Application()
+        app.router.add_post("/gen", handler)
+        runner = web.AppRunner(app)
+        await runner.setup()
+        site = web.TCPSite(runner, "127.0.0.1", 0)
+        await site.start()
+        port = runner.addresses[0][1]
+        try:
+            s = make_settings(
+                SELKIES_ENABLE_CLOUDFLARE_TURN="true",
+                SELKIES_CLOUDFLARE_TURN_TOKEN_ID="kid",
+                SELKIES_CLOUDFLARE_TURN_API_TOKEN="cftok",
+                SELKIES_TURN_HOST="relay.example",
+                SELKIES_TURN_SHARED_SECRET="x")
+            s._cloudflare_endpoint = f"http://127.0.0.1:{port}/gen"
+            cfg, source = await T.resolve_rtc_config(s)
+            assert source == "cloudflare"
+            assert cfg["iceServers"][0]["username"] == "cfu"
+            # minting failure falls through to HMAC
+            s._cloudflare_endpoint = f"http://127.0.0.1:{port}/nope"
+            cfg, source = await T.resolve_rtc_config(s)
+            assert source == "hmac"
+        finally:
+            await runner.cleanup()
+
+    asyncio.new_event_loop().run_until_complete(main())
