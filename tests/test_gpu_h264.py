@@ -702,3 +702,46 @@ def test_gpu_fractional_scale_stream():
     y_direct = Decoder().decode(r["data"])[0][0]
     assert np.array_equal(y_dec, y_direct), \
         "GPU fractional-scale path diverges from the scaled direct encode"
+
+
+def test_gpu_two_captures_concurrent():
+    """Two engine captures (primary + display2 pattern) encode
+    concurrently on one GPU — the multi-display production shape."""
+    require_gpu()
+    import threading
+    import time
+
+    caps, counts, events = [], [], []
+    try:
+        for i in range(2):
+            s = _native.CaptureSettings()
+            s.capture_width = 1280
+            s.capture_height = 720
+            s.target_fps = 60
+            s.output_mode = 1 if i == 0 else 2   # h264 + hevc mixed
+            s.use_cpu = False
+            s.gpu_id = 0
+            s.capture_backend = "synthetic:desktop"
+            s.video_fullframe = True
+            s.stripe_height = 64
+            n = {"v": 0}
+            ev = threading.Event()
+
+            def cb(data, fid, y, w, h, key, *a, _n=n, _ev=ev):
+                _n["v"] += 1
+                if _n["v"] >= 200:
+                    _ev.set()
+            cap = _native.ScreenCapture()
+            cap.start_capture(cb, s)
+            caps.append(cap)
+            counts.append(n)
+            events.append(ev)
+        for ev in events:
+            assert ev.wait(15)
+        for cap in caps:
+            assert cap.is_capturing
+            assert cap.pipeline.startswith("hip"), cap.pipeline
+    finally:
+        for cap in caps:
+            cap.stop_capture()
+    assert all(n["v"] >= 200 for n in counts)
