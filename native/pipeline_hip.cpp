@@ -376,6 +376,7 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&up_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&db_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&rows_stream_, hipStreamNonBlocking));
     for (int i = 0; i < 2; ++i) {
       HIP_CHECK(hipEventCreateWithFlags(&ev_rows_[i],
                                         hipEventDisableTiming));
@@ -400,6 +401,8 @@ class HipH264Pipeline : public EncodePipeline {
     (void)hipStreamDestroy(up_stream_);
     (void)hipStreamSynchronize(db_stream_);
     (void)hipStreamDestroy(db_stream_);
+    (void)hipStreamSynchronize(rows_stream_);
+    (void)hipStreamDestroy(rows_stream_);
     for (int i = 0; i < 2; ++i) {
       (void)hipEventDestroy(ev_rows_[i]);
       (void)hipEventDestroy(ev_db_[i]);
@@ -491,10 +494,12 @@ class HipH264Pipeline : public EncodePipeline {
                              hipMemcpyHostToDevice, up_stream_));
     // the caller may reuse its frame buffer once this event completes
     HIP_CHECK(hipEventRecord(ev_h2d_[par], up_stream_));
-    HIP_CHECK(hipStreamWaitEvent(stream_, ev_h2d_[par], 0));
+    // the whole produce chain (CSC -> ME -> rows) rides rows_stream_ so
+    // it overlaps the PREVIOUS frame's entropy kernel + D2H on stream_
+    HIP_CHECK(hipStreamWaitEvent(rows_stream_, ev_h2d_[par], 0));
     launch_bgrx_to_planes(d_frame_[par], w_, h_, frame.stride / 4, d_srcY_,
                           d_srcCb_, d_srcCr_, ypitch_, cpitch_, false,
-                          stream_);
+                          rows_stream_);
 
     // build row jobs for scheduled stripes (stripe stream state — IDR,
     // frame_num, idr_pic_id — resolves here so the GPU entropy kernel gets
@@ -550,23 +555,29 @@ class HipH264Pipeline : public EncodePipeline {
 
     HIP_CHECK(hipMemcpyAsync(d_jobs_[par], h_jobs_[par],
                              sizeof(h264gpu::RowJob) * n_jobs,
-                             hipMemcpyHostToDevice, stream_));
+                             hipMemcpyHostToDevice, rows_stream_));
+    // meta carries cross-frame tracking state (prev MV hints, hopeless
+    // bits); seed this parity's buffer from the previous one so the
+    // semantics are identical to the single-buffer design even for
+    // damage-skipped stripes
+    HIP_CHECK(hipMemcpyAsync(d_meta_[par], d_meta_[par ^ 1], meta_bytes_,
+                             hipMemcpyDeviceToDevice, rows_stream_));
     // luma pyramid (quarter res) for ME acquisition
     if (!ctx.idr) {
       h264gpu::launch_downsample2(d_srcY_, ypitch_, ypitch_, mbh_ * 16,
-                                  d_mip1_, ypitch_ / 2, stream_);
+                                  d_mip1_, ypitch_ / 2, rows_stream_);
       h264gpu::launch_downsample2(d_mip1_, ypitch_ / 2, ypitch_ / 2,
                                   mbh_ * 8, d_srcY2_, ypitch_ / 4,
-                                  stream_);
+                                  rows_stream_);
       h264gpu::launch_downsample2(d_refY_, ypitch_, ypitch_, mbh_ * 16,
-                                  d_mip1_, ypitch_ / 2, stream_);
+                                  d_mip1_, ypitch_ / 2, rows_stream_);
       h264gpu::launch_downsample2(d_mip1_, ypitch_ / 2, ypitch_ / 2,
                                   mbh_ * 8, d_refY2_, ypitch_ / 4,
-                                  stream_);
+                                  rows_stream_);
     }
     h264gpu::launch_h264_me(d_srcY_, ypitch_, w_, h_, d_refY_, d_srcY2_,
                             d_refY2_, ypitch_ / 4, mbw_, n_jobs,
-                            d_jobs_[par], d_meta_, stream_);
+                            d_jobs_[par], d_meta_[par], rows_stream_);
 
     // Single batch: the row kernel's cost is per-row LATENCY (all rows run
     // concurrently), so splitting into sequential batches multiplies GPU
@@ -592,50 +603,51 @@ class HipH264Pipeline : public EncodePipeline {
         h264gpu::launch_h264_rows(
             d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
             d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
-            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_, d_meta_,
-            stream_);
+            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_[par],
+            d_meta_[par], rows_stream_);
       else
         h264gpu::launch_h264_rows4(
             d_srcY_, d_srcCb_, d_srcCr_, ypitch_, cpitch_, w_, h_, d_refY_,
             d_refCb_, d_refCr_, d_curY_, d_curCb_, d_curCr_, mbw_,
-            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_, d_meta_,
-            stream_);
+            bt.jobn - bt.job0, d_jobs_[par] + bt.job0, d_levels_[par],
+            d_meta_[par], rows_stream_);
       if (cpu_entropy_) {
         HIP_CHECK(hipMemcpyAsync(
             reinterpret_cast<uint8_t*>(h_levels_) + bt.row0 * lvl_row,
-            reinterpret_cast<uint8_t*>(d_levels_) + bt.row0 * lvl_row,
-            (bt.rown - bt.row0) * lvl_row, hipMemcpyDeviceToHost, stream_));
+            reinterpret_cast<uint8_t*>(d_levels_[par]) + bt.row0 * lvl_row,
+            (bt.rown - bt.row0) * lvl_row, hipMemcpyDeviceToHost,
+            rows_stream_));
         HIP_CHECK(hipMemcpyAsync(
             reinterpret_cast<uint8_t*>(h_meta_) + bt.row0 * meta_row,
-            reinterpret_cast<uint8_t*>(d_meta_) + bt.row0 * meta_row,
-            (bt.rown - bt.row0) * meta_row, hipMemcpyDeviceToHost, stream_));
+            reinterpret_cast<uint8_t*>(d_meta_[par]) + bt.row0 * meta_row,
+            (bt.rown - bt.row0) * meta_row, hipMemcpyDeviceToHost,
+            rows_stream_));
       }
-      HIP_CHECK(hipEventRecord(batch_events_[b], stream_));
+      HIP_CHECK(hipEventRecord(batch_events_[b], rows_stream_));
     }
+    HIP_CHECK(hipEventRecord(ev_rows_[par], rows_stream_));
     // in-loop deblock of the current recon (within-slice edges only;
     // idc=2 is signaled in the slice headers) before it becomes the
     // reference frame. Deblock (reads recon+levels) and CAVLC (reads
     // levels) are independent, so deblock runs on its own stream
     // concurrently with entropy + bitstream D2H; the cur->ref copies
     // below join both streams before the recon becomes the reference.
-    const bool forked_db = settings_.video_deblock && !cpu_entropy_;
+    const bool forked_db = settings_.video_deblock;
     if (settings_.video_deblock) {
-      hipStream_t db_s = forked_db ? db_stream_ : stream_;
-      if (forked_db) {
-        HIP_CHECK(hipEventRecord(ev_rows_[par], stream_));
-        HIP_CHECK(hipStreamWaitEvent(db_stream_, ev_rows_[par], 0));
-      }
+      HIP_CHECK(hipStreamWaitEvent(db_stream_, ev_rows_[par], 0));
       h264gpu::launch_h264_deblock(d_curY_, d_curCb_, d_curCr_, ypitch_,
                                    cpitch_, mbw_, n_jobs, d_jobs_[par],
-                                   d_levels_, d_meta_, db_s);
-      if (forked_db) {
-        HIP_CHECK(hipEventRecord(ev_db_[par], db_stream_));
-      }
+                                   d_levels_[par], d_meta_[par],
+                                   db_stream_);
+      HIP_CHECK(hipEventRecord(ev_db_[par], db_stream_));
     }
     const int copy_words = ent_copy_words_;
     if (!cpu_entropy_) {
-      h264gpu::launch_h264_cavlc(d_levels_, d_meta_, mbw_, n_jobs,
-                                 d_jobs_[par], d_stage_, d_nbits_,
+      // entropy rides stream_ so the NEXT frame's produce chain on
+      // rows_stream_ overlaps it (levels/meta are parity-buffered)
+      HIP_CHECK(hipStreamWaitEvent(stream_, ev_rows_[par], 0));
+      h264gpu::launch_h264_cavlc(d_levels_[par], d_meta_[par], mbw_,
+                                 n_jobs, d_jobs_[par], d_stage_, d_nbits_,
                                  d_entout_[par], ent_stride_words_,
                                  d_outbits_[par], stream_);
       // compaction: copy only ~the used prefix of each row's bitstream
@@ -655,9 +667,12 @@ class HipH264Pipeline : public EncodePipeline {
                                sizeof(int) * n_jobs, hipMemcpyDeviceToHost,
                                stream_));
     }
-    // refresh ref from cur for encoded stripes (merge contiguous spans so
-    // the all-stripes case is 3 copies, not 3 x n_stripes)
-    if (forked_db) HIP_CHECK(hipStreamWaitEvent(stream_, ev_db_[par], 0));
+    // refresh ref from cur for encoded stripes (merge contiguous spans
+    // so the all-stripes case is 3 copies, not 3 x n_stripes). They ride
+    // rows_stream_ (gated on deblock) so the NEXT frame's rows see the
+    // filtered reference without waiting on this frame's entropy.
+    if (forked_db)
+      HIP_CHECK(hipStreamWaitEvent(rows_stream_, ev_db_[par], 0));
     {
       int span_y0 = -1, span_y1 = -1;
       auto flush_span = [&] {
@@ -679,7 +694,8 @@ class HipH264Pipeline : public EncodePipeline {
       }
       flush_span();
     }
-    HIP_CHECK(hipEventRecord(ev_done_[par], stream_));
+    HIP_CHECK(hipEventRecord(ev_done_[par],
+                             cpu_entropy_ ? rows_stream_ : stream_));
 
     // host-side output prep (headers, job -> stripe/row mapping)
     Pending pd;
@@ -897,7 +913,7 @@ class HipH264Pipeline : public EncodePipeline {
     HIP_CHECK(hipMemcpyAsync(dst + static_cast<size_t>(y0) * pitch,
                              src + static_cast<size_t>(y0) * pitch,
                              static_cast<size_t>(rows) * pitch,
-                             hipMemcpyDeviceToDevice, stream_));
+                             hipMemcpyDeviceToDevice, rows_stream_));
   }
 
  public:
@@ -927,9 +943,12 @@ class HipH264Pipeline : public EncodePipeline {
     size_t nmt = static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb;
     d.levels.resize(nlv);
     d.meta.resize(nmt);
-    HIP_CHECK(hipMemcpy(d.levels.data(), d_levels_, nlv * sizeof(int16_t),
+    const int last_par = parity_ ^ 1;
+    HIP_CHECK(hipMemcpy(d.levels.data(), d_levels_[last_par],
+                        nlv * sizeof(int16_t),
                         hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(d.meta.data(), d_meta_, nmt * sizeof(int),
+    HIP_CHECK(hipMemcpy(d.meta.data(), d_meta_[last_par],
+                        nmt * sizeof(int),
                         hipMemcpyDeviceToHost));
     return true;
   }
@@ -974,11 +993,17 @@ class HipH264Pipeline : public EncodePipeline {
                          h264gpu::kLevelsPerMb * sizeof(int16_t);
     size_t meta_bytes =
         static_cast<size_t>(mbw_) * mbh_ * h264gpu::kMetaPerMb * sizeof(int);
-    d_levels_ = reinterpret_cast<int16_t*>(dalloc(level_bytes));
+    meta_bytes_ = meta_bytes;
+    for (int i = 0; i < 2; ++i) {
+      d_levels_[i] = reinterpret_cast<int16_t*>(dalloc(level_bytes));
+      d_meta_[i] = reinterpret_cast<int*>(dalloc(meta_bytes));
+      // meta seeds from parity^1 every frame; zero both so the first
+      // seed copy is well-defined
+      HIP_CHECK(hipMemset(d_meta_[i], 0, meta_bytes));
+    }
     d_mip1_ = dalloc(static_cast<size_t>(ypitch_ / 2) * mbh_ * 8);
     d_srcY2_ = dalloc(static_cast<size_t>(ypitch_ / 4) * mbh_ * 4);
     d_refY2_ = dalloc(static_cast<size_t>(ypitch_ / 4) * mbh_ * 4);
-    d_meta_ = reinterpret_cast<int*>(dalloc(meta_bytes));
     segs_ = (mbw_ + h264gpu::kMaxSegMbw - 1) / h264gpu::kMaxSegMbw;
     // Latency shaping: the row kernels are serial in the MB chain of one
     // slice, and a frame whose rows alone underfill the 256-CU chip is
@@ -1055,6 +1080,7 @@ class HipH264Pipeline : public EncodePipeline {
   hipStream_t stream_{};
   hipStream_t up_stream_{};
   hipStream_t db_stream_{};
+  hipStream_t rows_stream_{};
   int stripe_h_ = 64;
   int w_ = 0, h_ = 0, mbw_ = 0, mbh_ = 0, ypitch_ = 0, cpitch_ = 0;
   int segs_ = 1, seg_w0_ = 0;     // slices per MB row, widest segment
@@ -1063,9 +1089,10 @@ class HipH264Pipeline : public EncodePipeline {
           *d_srcCr_ = nullptr, *d_refY_ = nullptr, *d_refCb_ = nullptr,
           *d_refCr_ = nullptr, *d_curY_ = nullptr, *d_curCb_ = nullptr,
           *d_curCr_ = nullptr;
-  int16_t* d_levels_ = nullptr;
+  int16_t* d_levels_[2] = {};
+  size_t meta_bytes_ = 0;
   uint8_t *d_mip1_ = nullptr, *d_srcY2_ = nullptr, *d_refY2_ = nullptr;
-  int* d_meta_ = nullptr;
+  int* d_meta_[2] = {};
   int16_t* h_levels_ = nullptr;
   int* h_meta_ = nullptr;
   uint8_t* h_stage_ = nullptr;
