@@ -229,7 +229,7 @@ def main():
     n_sess = max(1, args.sessions)
     depth = args.pipeline_depth
     if depth is None:
-        depth = 2 if (use_gpu and out_mode in (1, 2)) else 1
+        depth = 2 if use_gpu else 1
     # ranks wrap over the visible devices (driver runs nproc == n_gpus,
     # so this is identity there; on smaller boxes extra ranks share)
     dev = (local_rank % max(1, torch.cuda.device_count())

@@ -40,6 +40,14 @@ class HipJpegPipeline : public EncodePipeline {
                                 std::thread::hardware_concurrency() / 2))) {
     HIP_CHECK(hipSetDevice(std::max(0, s.gpu_id)));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&prod_stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&fix_stream_, hipStreamNonBlocking));
+    for (int i = 0; i < 2; ++i) {
+      HIP_CHECK(hipEventCreateWithFlags(&jev_dct_[i],
+                                        hipEventDisableTiming));
+      HIP_CHECK(hipEventCreateWithFlags(&jev_done_[i],
+                                        hipEventDisableTiming));
+    }
     upload_dct_tables(stream_);
     HIP_CHECK(hipMalloc(&d_rqy_, 64 * sizeof(float)));
     HIP_CHECK(hipMalloc(&d_rqc_, 64 * sizeof(float)));
@@ -56,32 +64,60 @@ class HipJpegPipeline : public EncodePipeline {
 
   ~HipJpegPipeline() override {
     (void)hipStreamSynchronize(stream_);
+    (void)hipStreamSynchronize(prod_stream_);
+    (void)hipStreamSynchronize(fix_stream_);
+    (void)hipStreamDestroy(prod_stream_);
+    (void)hipStreamDestroy(fix_stream_);
+    for (int i = 0; i < 2; ++i) {
+      (void)hipEventDestroy(jev_dct_[i]);
+      (void)hipEventDestroy(jev_done_[i]);
+    }
     for (auto& kv : registered_) (void)hipHostUnregister(kv.first);
     if (d_jtabs_) (void)hipFree(d_jtabs_);
-    if (d_jrows_) (void)hipFree(d_jrows_);
     if (d_jstage_) (void)hipFree(d_jstage_);
     if (d_jnbits_) (void)hipFree(d_jnbits_);
-    if (d_jout_) (void)hipFree(d_jout_);
-    if (d_joutbits_) (void)hipFree(d_joutbits_);
-    if (h_jout_) (void)hipHostFree(h_jout_);
-    if (h_joutbits_) (void)hipHostFree(h_joutbits_);
-    if (h_jrows_) (void)hipHostFree(h_jrows_);
-    if (d_frame_) (void)hipFree(d_frame_);
+    for (int i = 0; i < 2; ++i) {
+      if (d_jrows2_[i]) (void)hipFree(d_jrows2_[i]);
+      if (d_jout2_[i]) (void)hipFree(d_jout2_[i]);
+      if (d_joutbits2_[i]) (void)hipFree(d_joutbits2_[i]);
+      if (h_jout2_[i]) (void)hipHostFree(h_jout2_[i]);
+      if (h_joutbits2_[i]) (void)hipHostFree(h_joutbits2_[i]);
+      if (h_jrows2_[i]) (void)hipHostFree(h_jrows2_[i]);
+      if (d_frame2_[i]) (void)hipFree(d_frame2_[i]);
+      if (d_coeff2_[i]) (void)hipFree(d_coeff2_[i]);
+    }
     if (d_y_) (void)hipFree(d_y_);
     if (d_cb_) (void)hipFree(d_cb_);
     if (d_cr_) (void)hipFree(d_cr_);
-    if (d_coeff_) (void)hipFree(d_coeff_);
     if (d_rqy_) (void)hipFree(d_rqy_);
     if (d_rqc_) (void)hipFree(d_rqc_);
     if (h_coeff_) (void)hipHostFree(h_coeff_);
   }
 
-  void encode_frame(const RawFrame& frame, const FrameContext& ctx,
-                    const Emit& emit) override {
+  struct JOut {
+    std::vector<uint8_t> bytes;
+    int y0 = 0, h = 0;
+    int row0 = 0, rows = 0;     // MCU-row range (GPU entropy path)
+    bool encode = false;
+  };
+  struct JPending {
+    bool active = false;
+    int par = 0;
+    int n_rows = 0;
+    int copy_words = 0;
+    int quality = 80;
+    bool fullcolor = false;
+    uint32_t frame_id = 0;
+    int mcux = 0;
+    std::vector<JOut> outs;
+  };
+
+  JPending submit_frame(const RawFrame& frame, const FrameContext& ctx) {
     const bool fullcolor = settings_.video_fullcolor;
     if (frame.width != w_ || frame.height != h_)
       alloc_for(frame.width, frame.height);
     ensure_quality(ctx.jpeg_quality);
+    const int par = jparity_;
 
     // upload (register the capture buffer once; zero-copy DMA afterwards)
     const uint8_t* src = frame.data;
@@ -92,12 +128,14 @@ class HipJpegPipeline : public EncodePipeline {
       registered_[const_cast<uint8_t*>(src)] = (e == hipSuccess);
       if (e != hipSuccess) (void)hipGetLastError();  // clear; fall back
     }
-    HIP_CHECK(hipMemcpyAsync(d_frame_, src, frame_bytes,
-                             hipMemcpyHostToDevice, stream_));
+    // produce chain (upload, CSC, DCT) on its own stream so it overlaps
+    // the previous frame's entropy kernel + readback on stream_
+    HIP_CHECK(hipMemcpyAsync(d_frame2_[par], src, frame_bytes,
+                             hipMemcpyHostToDevice, prod_stream_));
 
     const int stride_px = frame.stride / 4;
-    launch_bgrx_to_planes(d_frame_, w_, h_, stride_px, d_y_, d_cb_, d_cr_,
-                          ypitch_, cpitch_, fullcolor, stream_);
+    launch_bgrx_to_planes(d_frame2_[par], w_, h_, stride_px, d_y_, d_cb_,
+                          d_cr_, ypitch_, cpitch_, fullcolor, prod_stream_);
 
     const int stripe_h = std::max(16, settings_.stripe_height & ~15);
     const int mcu = fullcolor ? 8 : 16;
@@ -107,148 +145,178 @@ class HipJpegPipeline : public EncodePipeline {
     const int stripe_mcu_count = rows_per_stripe * mcux;
     const int per_mcu_real = fullcolor ? 3 : 6;
 
+    int16_t* d_coeff = d_coeff2_[par];
     if (!fullcolor) {
-      launch_dct_quant(d_y_, w_, h_, ypitch_, d_rqy_, d_coeff_, 0, false,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
-      launch_dct_quant(d_cb_, cw_, ch_, cpitch_, d_rqc_, d_coeff_, 1, false,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
-      launch_dct_quant(d_cr_, cw_, ch_, cpitch_, d_rqc_, d_coeff_, 2, false,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
+      launch_dct_quant(d_y_, w_, h_, ypitch_, d_rqy_, d_coeff, 0, false,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
+      launch_dct_quant(d_cb_, cw_, ch_, cpitch_, d_rqc_, d_coeff, 1, false,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
+      launch_dct_quant(d_cr_, cw_, ch_, cpitch_, d_rqc_, d_coeff, 2, false,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
     } else {
-      launch_dct_quant(d_y_, w_, h_, ypitch_, d_rqy_, d_coeff_, 0, true,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
-      launch_dct_quant(d_cb_, w_, h_, ypitch_, d_rqc_, d_coeff_, 1, true,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
-      launch_dct_quant(d_cr_, w_, h_, ypitch_, d_rqc_, d_coeff_, 2, true,
-                       mcux, rows_per_stripe, stripe_mcu_count, stream_);
+      launch_dct_quant(d_y_, w_, h_, ypitch_, d_rqy_, d_coeff, 0, true,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
+      launch_dct_quant(d_cb_, w_, h_, ypitch_, d_rqc_, d_coeff, 1, true,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
+      launch_dct_quant(d_cr_, w_, h_, ypitch_, d_rqc_, d_coeff, 2, true,
+                       mcux, rows_per_stripe, stripe_mcu_count,
+                       prod_stream_);
     }
+    HIP_CHECK(hipEventRecord(jev_dct_[par], prod_stream_));
 
-    struct Out {
-      std::vector<uint8_t> bytes;
-      int y0 = 0, h = 0;
-      int row0 = 0, rows = 0;     // MCU-row range (GPU entropy path)
-      bool encode = false;
-    };
-    std::vector<Out> outs(ctx.stripes.size());
+    JPending pd;
+    pd.par = par;
+    pd.quality = ctx.jpeg_quality;
+    pd.fullcolor = fullcolor;
+    pd.frame_id = ctx.frame_id;
+    pd.mcux = mcux;
+    pd.outs.resize(ctx.stripes.size());
     for (size_t i = 0; i < ctx.stripes.size(); ++i) {
       const auto& job = ctx.stripes[i];
-      outs[i].y0 = job.y0;
-      outs[i].h = job.y1 - job.y0;
-      outs[i].encode = job.encode;
+      pd.outs[i].y0 = job.y0;
+      pd.outs[i].h = job.y1 - job.y0;
+      pd.outs[i].encode = job.encode;
     }
 
     if (cpu_jpeg_entropy_) {
+      // test/debug path: stays synchronous (depth-1 semantics)
       size_t coeff_count =
           static_cast<size_t>(mcux) * mcuy * per_mcu_real * 64;
-      HIP_CHECK(hipMemcpyAsync(h_coeff_, d_coeff_,
+      HIP_CHECK(hipMemcpyAsync(h_coeff_, d_coeff,
                                coeff_count * sizeof(int16_t),
-                               hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
-      // stripe-parallel entropy packing on the CPU pool (restart-row
-      // framing: bit-identical to the GPU kernel's streams)
-      for (size_t i = 0; i < ctx.stripes.size(); ++i) {
-        if (!ctx.stripes[i].encode) continue;
+                               hipMemcpyDeviceToHost, prod_stream_));
+      HIP_CHECK(hipStreamSynchronize(prod_stream_));
+      for (size_t i = 0; i < pd.outs.size(); ++i) {
+        if (!pd.outs[i].encode) continue;
         pool_.submit([&, i] {
-          const auto& j = ctx.stripes[i];
-          int stripe_idx = j.y0 / stripe_h;
+          const JOut& o = pd.outs[i];
+          int stripe_idx = o.y0 / stripe_h;
           int rows = std::min(rows_per_stripe,
                               mcuy - stripe_idx * rows_per_stripe);
           const int16_t* blocks =
               h_coeff_ + static_cast<size_t>(stripe_idx) *
                              stripe_mcu_count * per_mcu_real * 64;
-          jpeg_entropy_from_blocks(blocks, mcux, rows, w_, j.y1 - j.y0,
-                                   ctx.jpeg_quality, fullcolor,
-                                   outs[i].bytes, true);
+          jpeg_entropy_from_blocks(blocks, mcux, rows, w_, o.h,
+                                   pd.quality, fullcolor,
+                                   pd.outs[i].bytes, true);
         });
       }
       pool_.wait_all();
-    } else {
-      // GPU entropy: one kernel row per MCU row of every scheduled
-      // stripe; the 6 MB coefficient readback disappears entirely
-      auto* jrows = static_cast<jpeggpu::JRow*>(h_jrows_);
-      int n_rows = 0;
-      for (size_t i = 0; i < ctx.stripes.size(); ++i) {
-        if (!ctx.stripes[i].encode) continue;
-        int stripe_idx = outs[i].y0 / stripe_h;
-        int rows = std::min(rows_per_stripe,
-                            mcuy - stripe_idx * rows_per_stripe);
-        outs[i].row0 = n_rows;
-        outs[i].rows = rows;
-        size_t base = static_cast<size_t>(stripe_idx) * stripe_mcu_count *
-                      per_mcu_real * 64;
-        for (int r = 0; r < rows; ++r)
-          jrows[n_rows++].coeff_off = static_cast<int>(
-              base + static_cast<size_t>(r) * mcux * per_mcu_real * 64);
-      }
-      if (n_rows) {
-        HIP_CHECK(hipMemcpyAsync(d_jrows_, h_jrows_,
-                                 sizeof(jpeggpu::JRow) * n_rows,
-                                 hipMemcpyHostToDevice, stream_));
-        jpeggpu::launch_jpeg_entropy(
-            d_coeff_, static_cast<jpeggpu::JRow*>(d_jrows_), n_rows, mcux,
-            per_mcu_real, d_jtabs_, static_cast<uint32_t*>(d_jstage_),
-            static_cast<int*>(d_jnbits_), static_cast<uint32_t*>(d_jout_),
-            jout_stride_, static_cast<int*>(d_joutbits_), stream_);
-        // adaptive compacted readback (2x last frame's max row)
-        if (jout_copy_words_ >= jout_stride_) {
-          HIP_CHECK(hipMemcpyAsync(h_jout_, d_jout_,
-                                   (size_t)n_rows * jout_stride_ * 4,
-                                   hipMemcpyDeviceToHost, stream_));
-        } else {
-          HIP_CHECK(hipMemcpy2DAsync(
-              h_jout_, (size_t)jout_stride_ * 4, d_jout_,
-              (size_t)jout_stride_ * 4, (size_t)jout_copy_words_ * 4,
-              n_rows, hipMemcpyDeviceToHost, stream_));
-        }
-        HIP_CHECK(hipMemcpyAsync(h_joutbits_, d_joutbits_, 4 * n_rows,
-                                 hipMemcpyDeviceToHost, stream_));
-        HIP_CHECK(hipStreamSynchronize(stream_));
-        auto* outbits = static_cast<int*>(h_joutbits_);
-        int max_words = 0;
-        for (int r = 0; r < n_rows; ++r) {
-          int wds = (outbits[r] + 31) / 32 + 1;
-          max_words = std::max(max_words, wds);
-          if (wds > jout_copy_words_) {
-            HIP_CHECK(hipMemcpy(
-                static_cast<uint32_t*>(h_jout_) +
-                    (size_t)r * jout_stride_,
-                static_cast<uint32_t*>(d_jout_) +
-                    (size_t)r * jout_stride_,
-                (size_t)wds * 4, hipMemcpyDeviceToHost));
-          }
-        }
-        jout_copy_words_ = std::min(jout_stride_, max_words * 2 + 64);
-        for (size_t i = 0; i < ctx.stripes.size(); ++i) {
-          if (!ctx.stripes[i].encode) continue;
-          pool_.submit([&, i] {
-            Out& o = outs[i];
-            jpeg_write_headers(o.bytes, w_, o.h, ctx.jpeg_quality,
-                               fullcolor, mcux);
-            int rst = 0;
-            for (int r = 0; r < o.rows; ++r) {
-              if (r > 0) {
-                o.bytes.push_back(0xFF);
-                o.bytes.push_back(
-                    static_cast<uint8_t>(0xD0 + (rst++ & 7)));
-              }
-              const uint32_t* words = static_cast<uint32_t*>(h_jout_) +
-                                      (size_t)(o.row0 + r) * jout_stride_;
-              jpeg_append_row_bits(words, outbits[o.row0 + r], o.bytes);
-            }
-            o.bytes.push_back(0xFF);
-            o.bytes.push_back(0xD9);
-          });
-        }
-        pool_.wait_all();
-      }
+      pd.active = true;
+      pd.n_rows = 0;              // bytes already assembled
+      jparity_ ^= 1;
+      return pd;
     }
-    for (auto& o : outs) {
+
+    // GPU entropy on stream_ behind the DCT event; one kernel row per
+    // MCU row of every scheduled stripe
+    auto* jrows = static_cast<jpeggpu::JRow*>(h_jrows2_[par]);
+    int n_rows = 0;
+    for (size_t i = 0; i < pd.outs.size(); ++i) {
+      if (!pd.outs[i].encode) continue;
+      int stripe_idx = pd.outs[i].y0 / stripe_h;
+      int rows = std::min(rows_per_stripe,
+                          mcuy - stripe_idx * rows_per_stripe);
+      pd.outs[i].row0 = n_rows;
+      pd.outs[i].rows = rows;
+      size_t base = static_cast<size_t>(stripe_idx) * stripe_mcu_count *
+                    per_mcu_real * 64;
+      for (int r = 0; r < rows; ++r)
+        jrows[n_rows++].coeff_off = static_cast<int>(
+            base + static_cast<size_t>(r) * mcux * per_mcu_real * 64);
+    }
+    pd.n_rows = n_rows;
+    pd.active = true;
+    jparity_ ^= 1;
+    if (!n_rows) return pd;
+
+    HIP_CHECK(hipStreamWaitEvent(stream_, jev_dct_[par], 0));
+    HIP_CHECK(hipMemcpyAsync(d_jrows2_[par], h_jrows2_[par],
+                             sizeof(jpeggpu::JRow) * n_rows,
+                             hipMemcpyHostToDevice, stream_));
+    jpeggpu::launch_jpeg_entropy(
+        d_coeff, static_cast<jpeggpu::JRow*>(d_jrows2_[par]), n_rows, mcux,
+        per_mcu_real, d_jtabs_, static_cast<uint32_t*>(d_jstage_),
+        static_cast<int*>(d_jnbits_), static_cast<uint32_t*>(d_jout2_[par]),
+        jout_stride_, static_cast<int*>(d_joutbits2_[par]), stream_);
+    // adaptive compacted readback (2x last frame's max row)
+    pd.copy_words = std::min(jout_stride_, jout_copy_words_);
+    if (pd.copy_words >= jout_stride_) {
+      HIP_CHECK(hipMemcpyAsync(h_jout2_[par], d_jout2_[par],
+                               (size_t)n_rows * jout_stride_ * 4,
+                               hipMemcpyDeviceToHost, stream_));
+    } else {
+      HIP_CHECK(hipMemcpy2DAsync(
+          h_jout2_[par], (size_t)jout_stride_ * 4, d_jout2_[par],
+          (size_t)jout_stride_ * 4, (size_t)pd.copy_words * 4,
+          n_rows, hipMemcpyDeviceToHost, stream_));
+    }
+    HIP_CHECK(hipMemcpyAsync(h_joutbits2_[par], d_joutbits2_[par],
+                             4 * n_rows, hipMemcpyDeviceToHost, stream_));
+    HIP_CHECK(hipEventRecord(jev_done_[par], stream_));
+    return pd;
+  }
+
+  void collect_pending(JPending& pd, const Emit& emit) {
+    if (!pd.active) return;
+    pd.active = false;
+    const int par = pd.par;
+    if (!cpu_jpeg_entropy_ && pd.n_rows) {
+      HIP_CHECK(hipEventSynchronize(jev_done_[par]));
+      auto* outbits = static_cast<int*>(h_joutbits2_[par]);
+      int max_words = 0;
+      bool fix = false;
+      for (int r = 0; r < pd.n_rows; ++r) {
+        int wds = (outbits[r] + 31) / 32 + 1;
+        max_words = std::max(max_words, wds);
+        if (wds > pd.copy_words) {
+          HIP_CHECK(hipMemcpyAsync(
+              static_cast<uint32_t*>(h_jout2_[par]) +
+                  (size_t)r * jout_stride_,
+              static_cast<uint32_t*>(d_jout2_[par]) +
+                  (size_t)r * jout_stride_,
+              (size_t)wds * 4, hipMemcpyDeviceToHost, fix_stream_));
+          fix = true;
+        }
+      }
+      if (fix) HIP_CHECK(hipStreamSynchronize(fix_stream_));
+      jout_copy_words_ = std::min(jout_stride_, max_words * 2 + 64);
+      for (size_t i = 0; i < pd.outs.size(); ++i) {
+        if (!pd.outs[i].encode) continue;
+        pool_.submit([&, i, par, outbits] {
+          JOut& o = pd.outs[i];
+          jpeg_write_headers(o.bytes, w_, o.h, pd.quality,
+                             pd.fullcolor, pd.mcux);
+          int rst = 0;
+          for (int r = 0; r < o.rows; ++r) {
+            if (r > 0) {
+              o.bytes.push_back(0xFF);
+              o.bytes.push_back(
+                  static_cast<uint8_t>(0xD0 + (rst++ & 7)));
+            }
+            const uint32_t* words =
+                static_cast<uint32_t*>(h_jout2_[par]) +
+                (size_t)(o.row0 + r) * jout_stride_;
+            jpeg_append_row_bits(words, outbits[o.row0 + r], o.bytes);
+          }
+          o.bytes.push_back(0xFF);
+          o.bytes.push_back(0xD9);
+        });
+      }
+      pool_.wait_all();
+    }
+    for (auto& o : pd.outs) {
       if (!o.encode) continue;
       EncodedStripe s;
       s.type = StripeType::kJpeg;
       s.data = o.bytes.data();
       s.size = o.bytes.size();
-      s.frame_id = ctx.frame_id;
+      s.frame_id = pd.frame_id;
       s.y = o.y0;
       s.width = w_;
       s.height = o.h;
@@ -257,16 +325,44 @@ class HipJpegPipeline : public EncodePipeline {
     }
   }
 
+  void encode_frame(const RawFrame& frame, const FrameContext& ctx,
+                    const Emit& emit) override {
+    JPending cur = submit_frame(frame, ctx);
+    if (jdepth_ < 2 || cpu_jpeg_entropy_) {
+      collect_pending(cur, emit);
+      collect_pending(jprev_, emit);   // drain any leftover from a mode flip
+      return;
+    }
+    collect_pending(jprev_, emit);
+    jprev_ = std::move(cur);
+    // the caller may reuse its frame buffer once the H2D completes; the
+    // DCT event is recorded after CSC+DCT which transitively covers it
+    HIP_CHECK(hipEventSynchronize(jev_dct_[jprev_.par]));
+  }
+
+  void flush(const Emit& emit) override { collect_pending(jprev_, emit); }
+
+  void set_pipeline_depth(int d) override {
+    jdepth_ = std::max(1, std::min(2, d));
+  }
+
   const char* name() const override { return "hip-jpeg"; }
 
  private:
   void alloc_for(int w, int h) {
     HIP_CHECK(hipStreamSynchronize(stream_));
-    if (d_frame_) (void)hipFree(d_frame_);
+    HIP_CHECK(hipStreamSynchronize(prod_stream_));
+    jprev_ = JPending{};
+    jparity_ = 0;
+    for (int i = 0; i < 2; ++i) {
+      if (d_frame2_[i]) (void)hipFree(d_frame2_[i]);
+      if (d_coeff2_[i]) (void)hipFree(d_coeff2_[i]);
+      d_frame2_[i] = nullptr;
+      d_coeff2_[i] = nullptr;
+    }
     if (d_y_) (void)hipFree(d_y_);
     if (d_cb_) (void)hipFree(d_cb_);
     if (d_cr_) (void)hipFree(d_cr_);
-    if (d_coeff_) (void)hipFree(d_coeff_);
     if (h_coeff_) (void)hipHostFree(h_coeff_);
     w_ = w;
     h_ = h;
@@ -274,7 +370,8 @@ class HipJpegPipeline : public EncodePipeline {
     ch_ = (h + 1) / 2;
     ypitch_ = (w + 255) & ~255;
     cpitch_ = (cw_ + 255) & ~255;
-    HIP_CHECK(hipMalloc(&d_frame_, static_cast<size_t>(w) * h * 4));
+    for (int i = 0; i < 2; ++i)
+      HIP_CHECK(hipMalloc(&d_frame2_[i], static_cast<size_t>(w) * h * 4));
     HIP_CHECK(hipMalloc(&d_y_, static_cast<size_t>(ypitch_) * h));
     // chroma buffers sized for 4:4:4 (the larger case)
     HIP_CHECK(hipMalloc(&d_cb_, static_cast<size_t>(ypitch_) * h));
@@ -282,35 +379,45 @@ class HipJpegPipeline : public EncodePipeline {
     // coefficients: 6 blocks/MCU covers both 420 (6) and 444 (3)
     size_t mcux = (w + 7) / 8, mcuy = (h + 7) / 8;  // worst case 444
     size_t coeff_bytes = mcux * mcuy * 6 * 64 * sizeof(int16_t);
-    HIP_CHECK(hipMalloc(&d_coeff_, coeff_bytes));
+    for (int i = 0; i < 2; ++i)
+      HIP_CHECK(hipMalloc(&d_coeff2_[i], coeff_bytes));
     HIP_CHECK(hipHostMalloc(&h_coeff_, coeff_bytes, hipHostMallocDefault));
     // GPU entropy buffers: one slot per MCU row (444 row count is the max)
-    if (d_jrows_) (void)hipFree(d_jrows_);
     if (d_jstage_) (void)hipFree(d_jstage_);
     if (d_jnbits_) (void)hipFree(d_jnbits_);
-    if (d_jout_) (void)hipFree(d_jout_);
-    if (d_joutbits_) (void)hipFree(d_joutbits_);
-    if (h_jout_) (void)hipHostFree(h_jout_);
-    if (h_joutbits_) (void)hipHostFree(h_joutbits_);
-    if (h_jrows_) (void)hipHostFree(h_jrows_);
+    for (int i = 0; i < 2; ++i) {
+      if (d_jrows2_[i]) (void)hipFree(d_jrows2_[i]);
+      if (d_jout2_[i]) (void)hipFree(d_jout2_[i]);
+      if (d_joutbits2_[i]) (void)hipFree(d_joutbits2_[i]);
+      if (h_jout2_[i]) (void)hipHostFree(h_jout2_[i]);
+      if (h_joutbits2_[i]) (void)hipHostFree(h_joutbits2_[i]);
+      if (h_jrows2_[i]) (void)hipHostFree(h_jrows2_[i]);
+      d_jrows2_[i] = d_jout2_[i] = d_joutbits2_[i] = nullptr;
+      h_jout2_[i] = h_joutbits2_[i] = h_jrows2_[i] = nullptr;
+    }
     size_t max_rows = mcuy;
     size_t max_items = mcux * 3 > ((w + 15) / 16) * 6
                            ? mcux * 3
                            : ((size_t)(w + 15) / 16) * 6;
     jout_stride_ = (int)(max_items * jpeggpu::kJStageWords);
-    HIP_CHECK(hipMalloc(&d_jrows_,
-                        sizeof(jpeggpu::JRow) * max_rows));
     HIP_CHECK(hipMalloc(&d_jstage_, max_rows * max_items *
                                         jpeggpu::kJStageWords * 4));
     HIP_CHECK(hipMalloc(&d_jnbits_, max_rows * max_items * 4));
-    HIP_CHECK(hipMalloc(&d_jout_, max_rows * (size_t)jout_stride_ * 4));
-    HIP_CHECK(hipMalloc(&d_joutbits_, max_rows * 4));
-    HIP_CHECK(hipHostMalloc(&h_jout_, max_rows * (size_t)jout_stride_ * 4,
-                            hipHostMallocDefault));
-    HIP_CHECK(hipHostMalloc(&h_joutbits_, max_rows * 4,
-                            hipHostMallocDefault));
-    HIP_CHECK(hipHostMalloc(&h_jrows_, sizeof(jpeggpu::JRow) * max_rows,
-                            hipHostMallocDefault));
+    for (int i = 0; i < 2; ++i) {
+      HIP_CHECK(hipMalloc(&d_jrows2_[i],
+                          sizeof(jpeggpu::JRow) * max_rows));
+      HIP_CHECK(hipMalloc(&d_jout2_[i],
+                          max_rows * (size_t)jout_stride_ * 4));
+      HIP_CHECK(hipMalloc(&d_joutbits2_[i], max_rows * 4));
+      HIP_CHECK(hipHostMalloc(&h_jout2_[i],
+                              max_rows * (size_t)jout_stride_ * 4,
+                              hipHostMallocDefault));
+      HIP_CHECK(hipHostMalloc(&h_joutbits2_[i], max_rows * 4,
+                              hipHostMallocDefault));
+      HIP_CHECK(hipHostMalloc(&h_jrows2_[i],
+                              sizeof(jpeggpu::JRow) * max_rows,
+                              hipHostMallocDefault));
+    }
     jout_copy_words_ = 1 << 30;
   }
 
@@ -323,31 +430,39 @@ class HipJpegPipeline : public EncodePipeline {
       rqy[i] = 1.0f / qy[i];
       rqc[i] = 1.0f / qc[i];
     }
-    HIP_CHECK(hipMemcpyAsync(d_rqy_, rqy, sizeof(rqy), hipMemcpyHostToDevice,
-                             stream_));
-    HIP_CHECK(hipMemcpyAsync(d_rqc_, rqc, sizeof(rqc), hipMemcpyHostToDevice,
-                             stream_));
+    // quality changes are rare: drain BOTH streams so an in-flight
+    // frame never sees half-updated tables, then upload on the producer
     HIP_CHECK(hipStreamSynchronize(stream_));
+    HIP_CHECK(hipStreamSynchronize(prod_stream_));
+    HIP_CHECK(hipMemcpyAsync(d_rqy_, rqy, sizeof(rqy), hipMemcpyHostToDevice,
+                             prod_stream_));
+    HIP_CHECK(hipMemcpyAsync(d_rqc_, rqc, sizeof(rqc), hipMemcpyHostToDevice,
+                             prod_stream_));
+    HIP_CHECK(hipStreamSynchronize(prod_stream_));
     cur_quality_ = q;
   }
 
   CaptureSettings settings_;
   ThreadPool pool_;
-  hipStream_t stream_{};
-  uint8_t* d_frame_ = nullptr;
+  hipStream_t stream_{}, prod_stream_{}, fix_stream_{};
+  hipEvent_t jev_dct_[2] = {}, jev_done_[2] = {};
+  int jparity_ = 0;
+  int jdepth_ = 1;
+  JPending jprev_;
+  uint8_t* d_frame2_[2] = {};
   uint8_t* d_y_ = nullptr;
   uint8_t* d_cb_ = nullptr;
   uint8_t* d_cr_ = nullptr;
-  int16_t* d_coeff_ = nullptr;
+  int16_t* d_coeff2_[2] = {};
   uint32_t* d_jtabs_ = nullptr;
-  void* d_jrows_ = nullptr;
+  void* d_jrows2_[2] = {};
   void* d_jstage_ = nullptr;
   void* d_jnbits_ = nullptr;
-  void* d_jout_ = nullptr;
-  void* d_joutbits_ = nullptr;
-  void* h_jout_ = nullptr;
-  void* h_joutbits_ = nullptr;
-  void* h_jrows_ = nullptr;
+  void* d_jout2_[2] = {};
+  void* d_joutbits2_[2] = {};
+  void* h_jout2_[2] = {};
+  void* h_joutbits2_[2] = {};
+  void* h_jrows2_[2] = {};
   int jout_stride_ = 0;
   int jout_copy_words_ = 1 << 30;
   bool cpu_jpeg_entropy_ = false;
