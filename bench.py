@@ -75,7 +75,10 @@ def run_e2e(args):
         async with aiohttp.ClientSession() as sess:
             async with sess.ws_connect(
                     f"http://127.0.0.1:{port}/ws") as ws:
-                t_deadline = time.monotonic() + 120
+                # wall cap scales with the requested frames (60 fps
+                # pacing) so long endurance runs measure what they claim
+                t_deadline = time.monotonic() + max(
+                    120, (args.steps + args.warmup) / 60 * 1.3 + 30)
                 while time.monotonic() < t_deadline:
                     msg = await ws.receive(timeout=10)
                     if msg.type != WSMsgType.BINARY:
