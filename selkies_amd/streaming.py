@@ -272,10 +272,15 @@ class StreamingService:
         def on_stripe(data, frame_id, y, width, height, is_keyframe,
                       capture_ts_ms, encode_done_ms, stripe_type,
                       _display=display):
-            # native thread -> loop (the only allowed crossing)
-            loop.call_soon_threadsafe(self._fanout, _display, data,
-                                      frame_id, y, is_keyframe,
-                                      capture_ts_ms)
+            # native thread -> loop (the only allowed crossing). The loop
+            # can close while the capture thread drains its last frames
+            # (shutdown); those stripes are simply dropped.
+            try:
+                loop.call_soon_threadsafe(self._fanout, _display, data,
+                                          frame_id, y, is_keyframe,
+                                          capture_ts_ms)
+            except RuntimeError:
+                pass
 
         if (display == "primary" and not self.settings.capture_cursor
                 and self.settings.enable_cursors):

@@ -22,7 +22,7 @@ def loop():
     loop.close()
 
 
-async def read_role(ws, timeout=5):
+async def read_role(ws, timeout=20):
     deadline = asyncio.get_event_loop().time() + timeout
     while asyncio.get_event_loop().time() < deadline:
         msg = await ws.receive(timeout=timeout)
@@ -97,7 +97,7 @@ def test_concurrent_join_exactly_one_controller(loop):
 
                 async def next_role(ws):
                     try:
-                        return await read_role(ws, timeout=10)
+                        return await read_role(ws, timeout=20)
                     except (asyncio.TimeoutError, Exception):
                         return None
 
