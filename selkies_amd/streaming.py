@@ -289,8 +289,11 @@ class StreamingService:
                 if self.settings.debug_cursors:
                     logger.info("cursor shape %dx%d hot(%d,%d)", w, h, hx,
                                 hy)
-                loop.call_soon_threadsafe(self._broadcast_cursor, w, h, hx,
-                                          hy, argb)
+                try:
+                    loop.call_soon_threadsafe(self._broadcast_cursor, w, h,
+                                              hx, hy, argb)
+                except RuntimeError:
+                    pass
             cap.set_cursor_callback(on_cursor)
         cap.start_capture(on_stripe, self.build_capture_settings(display))
         logger.info("capture started for %s (pipeline=%s)", display,
@@ -331,7 +334,10 @@ class StreamingService:
                     except asyncio.QueueEmpty:
                         pass
                 q.put_nowait(data)
-            loop.call_soon_threadsafe(put)
+            try:
+                loop.call_soon_threadsafe(put)
+            except RuntimeError:
+                pass
 
         self.audio = _native.AudioCapture()
         self.audio.start_capture(s, on_frame)

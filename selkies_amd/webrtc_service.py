@@ -133,7 +133,10 @@ class WebRTCService:
         def on_stripe(data, frame_id, y, width, height, is_keyframe,
                       capture_ts_ms, encode_done_ms, stripe_type):
             payload = bytes(data[10:])  # drop the WS wire header
-            loop.call_soon_threadsafe(self._send_frame, payload)
+            try:
+                loop.call_soon_threadsafe(self._send_frame, payload)
+            except RuntimeError:
+                pass
 
         self.capture.start_capture(on_stripe, cs)
         logger.info("webrtc video capture started (pipeline=%s)",
@@ -162,7 +165,10 @@ class WebRTCService:
         def on_frame(data, pts_ms):
             payload = g711.wire_frame_to_ulaw(bytes(data), channels)
             if payload:
-                loop.call_soon_threadsafe(self._send_audio, payload)
+                try:
+                    loop.call_soon_threadsafe(self._send_audio, payload)
+                except RuntimeError:
+                    pass
 
         self.audio_capture = _native.AudioCapture()
         self.audio_capture.start_capture(s, on_frame)
