@@ -520,6 +520,8 @@ class StreamingService:
             self._run_hook(self.settings.run_after_connect)
 
         try:
+            await ws.send_str("AUTH_SUCCESS," + json.dumps(
+                {"role": role, "slot": state.player_seat}))
             await ws.send_str(P.encode_control("MODE", "websockets"))
             await ws.send_str(P.encode_control("ROLE", role))
             await ws.send_str("MK_ACCESS," +
