@@ -100,6 +100,8 @@ class CentralizedStreamServer:
         app.router.add_post("/api/webrtc/offer", self.handle_webrtc_offer)
         app.router.add_post("/api/upload", self.handle_upload)
         app.router.add_get("/api/download", self.handle_download)
+        # direct-download style (reference '/files/<name>')
+        app.router.add_get("/files/{name:.+}", self.handle_file_direct)
         app.router.add_get("/api/files", self.handle_files)
         web_dir = self.settings.web_root or os.path.join(
             os.path.dirname(__file__), "web")
@@ -368,6 +370,18 @@ class CentralizedStreamServer:
             return resp
         except PermissionError as exc:
             raise web.HTTPForbidden(reason=str(exc))
+
+    async def handle_file_direct(self, request):
+        """GET /files/<name>: direct download path (reference '/files/'
+        surface); same traversal-safe resolution as /api/download."""
+        name = request.match_info.get("name", "")
+        try:
+            path = self.transfers.resolve(name)
+        except PermissionError as exc:
+            raise web.HTTPForbidden(reason=str(exc))
+        if not os.path.isfile(path):
+            raise web.HTTPNotFound()
+        return web.FileResponse(path)
 
     async def handle_files(self, request):
         try:

@@ -655,3 +655,24 @@ def test_pipeline_reset_and_display_config_notices(loop):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_direct_files_route(loop, tmp_path):
+    async def main():
+        server = make_server(SELKIES_UPLOAD_DIR=str(tmp_path))
+        (tmp_path / "doc.txt").write_bytes(b"direct")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.get(
+                        f"http://127.0.0.1:{port}/files/doc.txt") as r:
+                    assert r.status == 200
+                    assert await r.read() == b"direct"
+                async with sess.get(
+                        f"http://127.0.0.1:{port}/files/../etc/passwd"
+                ) as r:
+                    assert r.status in (403, 404)
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
