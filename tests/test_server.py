@@ -676,3 +676,43 @@ def test_direct_files_route(loop, tmp_path):
             await runner.cleanup()
 
     loop.run_until_complete(main())
+
+
+def test_live_audio_is_opus_and_decodes(loop):
+    """The live server's audio fan-out carries Opus packets by default;
+    the from-spec decoder recovers audio from the wire frames."""
+    import numpy as np
+    from opus_ref_decoder import OpusDecoder
+
+    async def main():
+        server = make_server(SELKIES_ENABLE_AUDIO="true")
+        runner, port = await start_on_free_port(server)
+        try:
+            async with aiohttp.ClientSession() as sess:
+                async with sess.ws_connect(
+                        f"ws://127.0.0.1:{port}/ws") as ws:
+                    pkts = []
+                    for _ in range(400):
+                        msg = await asyncio.wait_for(ws.receive(), 5)
+                        if msg.type == aiohttp.WSMsgType.BINARY and \
+                                msg.data[0] == 0x01:
+                            d = msg.data
+                            off = 2
+                            for _ in range(d[1]):
+                                ln = (d[off] << 8) | d[off + 1]
+                                off += 2 + ln
+                            pkts.append(bytes(d[off:]))
+                            if len(pkts) >= 8:
+                                break
+                    assert len(pkts) >= 8
+                    # TOC: CELT fullband 20 ms mono
+                    assert all(p[0] >> 3 == 31 for p in pkts)
+                    dec = OpusDecoder()
+                    for p in pkts:
+                        dec.decode_packet(p)
+                    out = dec.samples()[960:]
+                    assert np.abs(out).max() > 0.01   # audible tone
+        finally:
+            await runner.cleanup()
+
+    loop.run_until_complete(main())
