@@ -308,6 +308,10 @@ SETTING_DEFINITIONS: list[SettingDef] = [
                client=True),
     SettingDef("enable_cursors", bool, True,
                "Push server cursor shapes to clients.", client=True),
+    SettingDef("use_css_scaling", bool, False,
+               "Scale the remote canvas with CSS instead of matching "
+               "the stream resolution 1:1 (sharper text off, fit-to-"
+               "window on).", client=True),
     SettingDef("use_browser_cursors", bool, True,
                "Render the cursor via CSS on the client instead of "
                "compositing it into the video.", client=True),

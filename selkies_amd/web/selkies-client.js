@@ -449,6 +449,11 @@ function applyServerSettings(payload) {
   if (statsEl && payload.ui_sidebar_show_stats)
     statsEl.style.display =
         payload.ui_sidebar_show_stats.value ? "" : "none";
+  if (payload.use_css_scaling) {
+    /* CSS fit-to-window vs native 1:1 (reference use_css_scaling) */
+    canvas.style.width = payload.use_css_scaling.value ? "100%" : "";
+    canvas.style.height = payload.use_css_scaling.value ? "100%" : "";
+  }
   const trEl = document.getElementById("transfers");
   if (trEl && payload.ui_sidebar_show_files)
     trEl.style.display =
